@@ -1,0 +1,77 @@
+// Host-side API shared between the pybind module and kernel TUs.
+#pragma once
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+
+#include "td/device.hpp"
+
+namespace td {
+
+#define TD_CHECK_HIP(expr)                                                   \
+  do {                                                                       \
+    hipError_t _e = (expr);                                                  \
+    if (_e != hipSuccess) {                                                  \
+      throw std::runtime_error(std::string("HIP error at " __FILE__ ":") +   \
+                               std::to_string(__LINE__) + ": " +             \
+                               hipGetErrorString(_e));                       \
+    }                                                                        \
+  } while (0)
+
+// kernels/common.hip ---------------------------------------------------------
+void launch_barrier_all(const PeerTable &pt, int *local_flags, int epoch,
+                        hipStream_t stream);
+void launch_signal_set(int *flag, int val, hipStream_t stream);
+void launch_wait_eq(const int *flags, int n, int expect, hipStream_t stream);
+void launch_copy(void *dst, const void *src, size_t nbytes, hipStream_t stream);
+void launch_put_signal(const PeerTable &pt, void *dst, const void *src,
+                       size_t nbytes, int *flag, int val, int add,
+                       hipStream_t stream);
+void launch_reset_flags(int *flags, int n, int val, hipStream_t stream);
+
+// kernels/gemm.hip -----------------------------------------------------------
+struct GemmArgs {
+  const void *a;   // [M,K] bf16 row-major
+  const void *b;   // [N,K] bf16 row-major (B^T layout: GEMM computes A @ B^T)
+  void *c;         // [M,N] bf16 row-major
+  const void *bias;  // [N] bf16 or nullptr
+  int m, n, k;
+  int lda, ldb, ldc;  // in elements
+};
+
+void launch_gemm_bf16(const GemmArgs &args, hipStream_t stream);
+
+// MFMA operand-layout probe (see gemm.hip): C[16x16] f32 = A[16x32] @ B[32x16].
+void launch_probe_mfma(const void *a, const void *b, void *c, int layout,
+                       hipStream_t stream);
+
+// AG-GEMM consumer: persistent GEMM over the gathered A [world*m_per_rank, K]
+// stored in the local symmetric workspace; each tile spin-waits on the
+// per-chunk flags before consuming rows of A.
+struct AgGemmArgs {
+  GemmArgs g;           // a = symm workspace [M_total, K]
+  const int *flags;     // per-chunk ready flags (local heap)
+  int chunks_per_rank;  // flag granularity along M
+  int m_per_rank;
+  int world;
+  int rank;
+  int expect;           // flag value that means "ready"
+};
+void launch_ag_gemm_consumer_bf16(const AgGemmArgs &args, hipStream_t stream);
+
+// GEMM-RS producer: GEMM whose epilogue scatters each output tile directly
+// into the owner rank's symmetric scatter buffer (remote store over xGMI).
+struct GemmRsArgs {
+  GemmArgs g;       // c unused; m = world * m_per_rank
+  PeerTable pt;
+  size_t scatter_off;  // offset of scatter buffer [world segments, m_per_rank, N]
+  int m_per_rank;
+  int world;
+  int rank;
+};
+void launch_gemm_rs_producer_bf16(const GemmRsArgs &args, hipStream_t stream);
+
+// Reduce over world segments [world, m_per_rank, N] -> [m_per_rank, N].
+void launch_rs_reduce_bf16(const void *segments, void *out, int world,
+                           int rank, int m_per_rank, int n, hipStream_t stream);
+
+}  // namespace td

@@ -1,0 +1,209 @@
+"""Symmetric memory manager — the data plane of triton_dist_amd.
+
+GPU backend: hipIpc heap in `triton_dist_amd._C` — every rank allocates one
+big device heap, IPC handles are exchanged over the bootstrap process group,
+and every rank maps every peer's heap. Allocation is a collective bump
+allocator (all ranks allocate in lockstep, like shmem_malloc), so the same
+offset on every rank names "the same" symmetric buffer and peer pointers are
+base[r] + offset. Capability parity with the reference's rocSHMEM layer
+(Triton-distributed python/triton_dist/utils.py:252-338 nvshmem_create_tensor
+/ rocshmem_create_tensor_list_intra_node / barrier_all_on_stream) with
+hand-written HIP kernels underneath.
+
+CPU backend: POSIX shared-memory mock (cpu_shm.py) with identical API so the
+plumbing runs under gloo on CPU-only CI.
+"""
+from __future__ import annotations
+
+import os
+from dataclasses import dataclass, field
+from typing import List, Optional
+
+import torch
+import torch.distributed as dist
+
+from ..utils.distributed import env_local_rank, has_gpu
+
+# DLPack dtype codes (code, bits)
+_DL_DTYPE = {
+    torch.float32: (2, 32),
+    torch.float16: (2, 16),
+    torch.bfloat16: (4, 16),
+    torch.int64: (0, 64),
+    torch.int32: (0, 32),
+    torch.int8: (0, 8),
+    torch.uint8: (1, 8),
+}
+
+_ALIGN = 256
+
+
+def _numel(shape) -> int:
+    n = 1
+    for s in shape:
+        n *= int(s)
+    return n
+
+
+@dataclass
+class SymmBuffer:
+    """A symmetric allocation: same offset in every rank's heap."""
+    heap: "SymmHeap"
+    offset: int
+    shape: tuple
+    dtype: torch.dtype
+
+    def local(self) -> torch.Tensor:
+        return self.heap.view(self.heap.rank, self.offset, self.shape,
+                              self.dtype)
+
+    def peer(self, rank: int) -> torch.Tensor:
+        return self.heap.view(rank, self.offset, self.shape, self.dtype)
+
+    def all_views(self) -> List[torch.Tensor]:
+        return [self.peer(r) for r in range(self.heap.world)]
+
+    def ptr(self, rank: Optional[int] = None) -> int:
+        return self.heap.ptr(self.heap.rank if rank is None else rank,
+                             self.offset)
+
+    @property
+    def nbytes(self) -> int:
+        return _numel(self.shape) * self.dtype.itemsize
+
+
+class SymmHeap:
+    """One symmetric heap per process. Construct collectively."""
+
+    def __init__(self, group=None, size_mb: Optional[int] = None,
+                 backend: Optional[str] = None):
+        self.group = group
+        self.rank = dist.get_rank(group) if dist.is_initialized() else 0
+        self.world = dist.get_world_size(group) if dist.is_initialized() else 1
+        if backend is None:
+            backend = "hip" if has_gpu() else "cpu"
+        self.backend = backend
+        if size_mb is None:
+            size_mb = int(os.environ.get(
+                "TD_SYMM_HEAP_MB", "4096" if backend == "hip" else "64"))
+        self.size = size_mb * 1024 * 1024
+        self._offset = 0
+        self._epoch = 0
+
+        if backend == "hip":
+            from .. import _C
+            self._C = _C
+            self.device = env_local_rank()
+            handle = _C.heap_init(self.rank, self.world, self.device,
+                                  self.size, True)
+            handles = [None] * self.world
+            if dist.is_initialized():
+                dist.all_gather_object(handles, bytes(handle), group=group)
+            else:
+                handles = [bytes(handle)]
+            _C.heap_open(list(handles))
+            self._offset = _C.heap_scratch_bytes()
+            if dist.is_initialized():
+                dist.barrier(group)
+        else:
+            from . import cpu_shm
+            self._shm = cpu_shm.CpuShmHeap(group, self.size)
+            self._offset = 4096
+
+        # internal flags for barrier_all: world int32 per rank
+        self._barrier_buf = self.alloc_buffer((max(self.world, 8),),
+                                              torch.int32)
+
+    # -- allocation (collective: call in the same order on all ranks) -------
+    def alloc(self, nbytes: int, align: int = _ALIGN) -> int:
+        off = (self._offset + align - 1) // align * align
+        if off + nbytes > self.size:
+            raise MemoryError(
+                f"symmetric heap exhausted: want {nbytes} at {off}, "
+                f"size {self.size} (raise TD_SYMM_HEAP_MB)")
+        self._offset = off + nbytes
+        return off
+
+    def alloc_buffer(self, shape, dtype: torch.dtype) -> SymmBuffer:
+        nbytes = _numel(shape) * dtype.itemsize
+        off = self.alloc(nbytes)
+        return SymmBuffer(self, off, tuple(shape), dtype)
+
+    def create_tensor(self, shape, dtype: torch.dtype) -> torch.Tensor:
+        """Local view of a fresh symmetric allocation (zeroed at heap init)."""
+        return self.alloc_buffer(shape, dtype).local()
+
+    def create_tensor_list(self, shape, dtype: torch.dtype
+                           ) -> List[torch.Tensor]:
+        """Per-rank views of one symmetric allocation (the analog of
+        rocshmem_create_tensor_list_intra_node)."""
+        return self.alloc_buffer(shape, dtype).all_views()
+
+    # -- addressing ---------------------------------------------------------
+    def ptr(self, rank: int, offset: int) -> int:
+        if self.backend == "hip":
+            return self._C.heap_base(rank) + offset
+        raise RuntimeError("cpu heap has no raw pointers")
+
+    def view(self, rank: int, offset: int, shape, dtype: torch.dtype
+             ) -> torch.Tensor:
+        if self.backend == "hip":
+            code, bits = _DL_DTYPE[dtype]
+            cap = self._C.dlpack_from_ptr(self.ptr(rank, offset), list(shape),
+                                          code, bits, self.device)
+            return torch.utils.dlpack.from_dlpack(cap)
+        return self._shm.view(rank, offset, shape, dtype)
+
+    # -- sync ---------------------------------------------------------------
+    def barrier_all_on_stream(self, stream: Optional[torch.cuda.Stream] = None):
+        """Device-side all-to-all barrier (epoch-counting), enqueued on the
+        stream; CPU backend: blocking shared-memory barrier."""
+        self._epoch += 1
+        if self.backend == "hip":
+            s = stream if stream is not None else torch.cuda.current_stream()
+            self._C.barrier_all(self._barrier_buf.ptr(), self._epoch,
+                                s.cuda_stream)
+        else:
+            from . import cpu_shm
+            me = self._barrier_buf.local()
+            for r in range(self.world):
+                cpu_shm.notify(self._barrier_buf.peer(r), self.rank,
+                               self._epoch)
+            for r in range(self.world):
+                cpu_shm.wait_ge(me, r, self._epoch)
+
+    barrier_all = barrier_all_on_stream
+
+    def close(self):
+        if self.backend == "hip":
+            self._C.heap_close()
+        else:
+            self._shm.close()
+
+
+_HEAP: Optional[SymmHeap] = None
+
+
+def init_symm_heap(group=None, size_mb: Optional[int] = None,
+                   backend: Optional[str] = None) -> SymmHeap:
+    global _HEAP
+    if _HEAP is None:
+        _HEAP = SymmHeap(group, size_mb, backend)
+    return _HEAP
+
+
+def get_heap() -> SymmHeap:
+    if _HEAP is None:
+        return init_symm_heap()
+    return _HEAP
+
+
+def heap_initialized() -> bool:
+    return _HEAP is not None
+
+
+def shutdown_heap():
+    global _HEAP
+    if _HEAP is not None:
+        _HEAP.close()
+        _HEAP = None

@@ -1,0 +1,110 @@
+"""Process bootstrap and distributed helpers.
+
+MI355X-native counterpart of the reference runtime layer
+(Triton-distributed python/triton_dist/utils.py:341-372
+`initialize_distributed`, :302 `finalize_distributed`, :445 `dist_print`):
+one process per GPU, torch.distributed with cpu:gloo + cuda:nccl (RCCL on
+ROCm) for bootstrap and golden-reference collectives; the data plane is the
+hipIpc symmetric heap (runtime/symm_mem.py), not RCCL.
+"""
+from __future__ import annotations
+
+import datetime
+import os
+import random
+import sys
+from typing import Optional
+
+import numpy as np
+import torch
+import torch.distributed as dist
+
+_INITIALIZED = False
+
+
+def has_gpu() -> bool:
+    return torch.cuda.is_available()
+
+
+def env_rank() -> int:
+    return int(os.environ.get("RANK", "0"))
+
+
+def env_world_size() -> int:
+    return int(os.environ.get("WORLD_SIZE", "1"))
+
+
+def env_local_rank() -> int:
+    return int(os.environ.get("LOCAL_RANK", str(env_rank())))
+
+
+def initialize_distributed(seed: int = 42, timeout_s: int = 1800,
+                           backend: Optional[str] = None):
+    """Init the process group (gloo on CPU-only hosts, gloo+RCCL on GPU),
+    pin the device for this rank, and seed all RNGs identically per rank.
+
+    Returns the default process group.
+    """
+    global _INITIALIZED
+    if _INITIALIZED:
+        return dist.group.WORLD
+    rank, world = env_rank(), env_world_size()
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29500")
+    if backend is None:
+        backend = "cpu:gloo,cuda:nccl" if has_gpu() else "gloo"
+    if not dist.is_initialized():
+        dist.init_process_group(
+            backend=backend,
+            rank=rank,
+            world_size=world,
+            timeout=datetime.timedelta(seconds=timeout_s),
+        )
+    if has_gpu():
+        torch.cuda.set_device(env_local_rank())
+    random.seed(seed)
+    np.random.seed(seed)
+    torch.manual_seed(seed + rank)
+    if has_gpu():
+        torch.cuda.manual_seed_all(seed + rank)
+    _INITIALIZED = True
+    return dist.group.WORLD
+
+
+def finalize_distributed():
+    global _INITIALIZED
+    from ..runtime.symm_mem import shutdown_heap
+
+    shutdown_heap()
+    if dist.is_initialized():
+        dist.barrier()
+        dist.destroy_process_group()
+    _INITIALIZED = False
+
+
+def rank(group=None) -> int:
+    return dist.get_rank(group) if dist.is_initialized() else 0
+
+
+def world_size(group=None) -> int:
+    return dist.get_world_size(group) if dist.is_initialized() else 1
+
+
+def dist_print(*args, allowed_ranks="0", need_sync: bool = False, **kwargs):
+    """Rank-filtered printing (cf. utils.py:445-476 semantics).
+
+    allowed_ranks: "all" or iterable/str of ranks.
+    """
+    r, w = rank(), world_size()
+    if allowed_ranks == "all":
+        allowed = list(range(w))
+    elif isinstance(allowed_ranks, str):
+        allowed = [int(x) for x in allowed_ranks.split(",") if x != ""]
+    else:
+        allowed = list(allowed_ranks)
+    for i in range(w):
+        if need_sync and dist.is_initialized():
+            dist.barrier()
+        if i == r and r in allowed:
+            print(f"[rank {r}]", *args, **kwargs)
+            sys.stdout.flush()
