@@ -83,13 +83,24 @@ TD_DEV void fence_acquire_sys() {
 // a workgroup barrier closes the op. Acquire semantics make subsequent
 // data reads safe.
 // ---------------------------------------------------------------------------
+// Spin-timeout: a wedged flag traps (kernel aborts with an error the host
+// sees) instead of hanging the GPU forever. ~100 MHz wall clock on CDNA;
+// 3e9 ticks ≈ 30 s.
+#ifndef TD_SPIN_TIMEOUT_TICKS
+#define TD_SPIN_TIMEOUT_TICKS 3000000000ull
+#endif
+
+TD_DEV uint64_t wallclock() { return __builtin_amdgcn_s_memrealtime(); }
+
 template <Scope S = Scope::Sys>
 TD_DEV void wait_eq(const int *flags, int n, int expect) {
   const int tid = threadIdx.x + threadIdx.y * blockDim.x;
   const int nthreads = blockDim.x * blockDim.y * blockDim.z;
   for (int i = tid; i < n; i += nthreads) {
+    uint64_t t0 = wallclock();
     while (ld_acquire<S>(flags + i) != expect) {
       __builtin_amdgcn_s_sleep(2);
+      if (wallclock() - t0 > TD_SPIN_TIMEOUT_TICKS) __builtin_trap();
     }
   }
   __syncthreads();
@@ -98,15 +109,19 @@ TD_DEV void wait_eq(const int *flags, int n, int expect) {
 // Single-thread variant (caller handles divergence / barriers).
 template <Scope S = Scope::Sys>
 TD_DEV void wait_eq_one(const int *flag, int expect) {
+  uint64_t t0 = wallclock();
   while (ld_acquire<S>(flag) != expect) {
     __builtin_amdgcn_s_sleep(2);
+    if (wallclock() - t0 > TD_SPIN_TIMEOUT_TICKS) __builtin_trap();
   }
 }
 
 template <Scope S = Scope::Sys>
 TD_DEV void wait_ge_one(const int *flag, int bound) {
+  uint64_t t0 = wallclock();
   while (ld_acquire<S>(flag) < bound) {
     __builtin_amdgcn_s_sleep(2);
+    if (wallclock() - t0 > TD_SPIN_TIMEOUT_TICKS) __builtin_trap();
   }
 }
 

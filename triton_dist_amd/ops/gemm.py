@@ -1,0 +1,62 @@
+"""Plain persistent MFMA GEMM wrapper: C = A @ W^T (+bias), bf16.
+
+The HIP kernel (csrc/kernels/gemm.hip k_gemm_bf16) replaces the reference's
+Triton GEMM zoo (Triton-distributed python/triton_dist/kernels/amd/gemm.py:
+62-541 — capability only). Weights are stored [N, K] (torch.nn.Linear
+layout), which makes A and W fragment loads symmetric on MFMA.
+"""
+from __future__ import annotations
+
+import torch
+
+from ..utils.distributed import has_gpu
+
+
+def _native():
+    from .. import _C
+
+    if _C is None:
+        raise RuntimeError("triton_dist_amd._C not built (required on GPU)")
+    return _C
+
+
+def gemm_supported(m: int, n: int, k: int) -> bool:
+    return m % 128 == 0 and n % 128 == 0 and k % 64 == 0
+
+
+def gemm(a: torch.Tensor, w: torch.Tensor, bias: torch.Tensor | None = None,
+         out: torch.Tensor | None = None) -> torch.Tensor:
+    """C[M,N] = A[M,K] @ W[N,K]^T + bias."""
+    m, k = a.shape
+    n, k2 = w.shape
+    assert k == k2, (a.shape, w.shape)
+    if a.is_cuda:
+        assert a.dtype == torch.bfloat16 and w.dtype == torch.bfloat16
+        assert a.is_contiguous() and w.is_contiguous()
+        assert gemm_supported(m, n, k), \
+            f"gemm v1 needs M%128==0,N%128==0,K%64==0, got {(m, n, k)}"
+        if out is None:
+            out = torch.empty(m, n, dtype=torch.bfloat16, device=a.device)
+        _native().gemm_bf16(
+            a.data_ptr(), w.data_ptr(), out.data_ptr(),
+            bias.data_ptr() if bias is not None else 0,
+            m, n, k, torch.cuda.current_stream().cuda_stream)
+        return out
+    # CPU reference path
+    c = a.float() @ w.float().t()
+    if bias is not None:
+        c += bias.float()
+    c = c.to(a.dtype)
+    if out is not None:
+        out.copy_(c)
+        return out
+    return c
+
+
+def gemm_ref(a: torch.Tensor, w: torch.Tensor,
+             bias: torch.Tensor | None = None) -> torch.Tensor:
+    """fp32 golden reference."""
+    c = a.float() @ w.float().t()
+    if bias is not None:
+        c += bias.float()
+    return c

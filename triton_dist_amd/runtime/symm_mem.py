@@ -22,7 +22,7 @@ from typing import List, Optional
 import torch
 import torch.distributed as dist
 
-from ..utils.distributed import env_local_rank, has_gpu
+from ..utils.distributed import has_gpu, local_device
 
 # DLPack dtype codes (code, bits)
 _DL_DTYPE = {
@@ -93,7 +93,7 @@ class SymmHeap:
         if backend == "hip":
             from .. import _C
             self._C = _C
-            self.device = env_local_rank()
+            self.device = local_device()
             handle = _C.heap_init(self.rank, self.world, self.device,
                                   self.size, True)
             handles = [None] * self.world

@@ -38,6 +38,14 @@ def env_local_rank() -> int:
     return int(os.environ.get("LOCAL_RANK", str(env_rank())))
 
 
+def local_device() -> int:
+    """Device index for this rank. Clamped modulo visible devices so a
+    2-process world can share one GPU (IPC paths are testable on a 1-GPU
+    box; xGMI paths then light up unchanged on 8)."""
+    n = torch.cuda.device_count()
+    return env_local_rank() % max(n, 1)
+
+
 def initialize_distributed(seed: int = 42, timeout_s: int = 1800,
                            backend: Optional[str] = None):
     """Init the process group (gloo on CPU-only hosts, gloo+RCCL on GPU),
@@ -52,7 +60,12 @@ def initialize_distributed(seed: int = 42, timeout_s: int = 1800,
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ.setdefault("MASTER_PORT", "29500")
     if backend is None:
-        backend = "cpu:gloo,cuda:nccl" if has_gpu() else "gloo"
+        # RCCL needs one rank per device; when ranks share a GPU (1-GPU box
+        # running a 2-rank IPC test) bootstrap over gloo only.
+        if has_gpu() and torch.cuda.device_count() >= world:
+            backend = "cpu:gloo,cuda:nccl"
+        else:
+            backend = "gloo"
     if not dist.is_initialized():
         dist.init_process_group(
             backend=backend,
@@ -61,7 +74,7 @@ def initialize_distributed(seed: int = 42, timeout_s: int = 1800,
             timeout=datetime.timedelta(seconds=timeout_s),
         )
     if has_gpu():
-        torch.cuda.set_device(env_local_rank())
+        torch.cuda.set_device(local_device())
     random.seed(seed)
     np.random.seed(seed)
     torch.manual_seed(seed + rank)
