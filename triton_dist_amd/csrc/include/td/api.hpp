@@ -18,7 +18,7 @@ namespace td {
   } while (0)
 
 // kernels/common.hip ---------------------------------------------------------
-void launch_barrier_all(const PeerTable &pt, int *local_flags, int epoch,
+void launch_barrier_all(const PeerTable &pt, int *local_flags, int *epoch_cell,
                         hipStream_t stream);
 void launch_signal_set(int *flag, int val, hipStream_t stream);
 void launch_wait_eq(const int *flags, int n, int expect, hipStream_t stream);

@@ -22,20 +22,32 @@ unsigned *g_arrive_counter();
 // `local_flags` must point into the symmetric heap so peers' slots resolve
 // via symm_at.
 // ---------------------------------------------------------------------------
-__global__ void k_barrier_all(PeerTable pt, int *local_flags, int epoch) {
+// The epoch counter lives in DEVICE memory (heap scratch) and is bumped by
+// the kernel itself, so the whole op sequence is hipGraph-replayable (no
+// host-side state baked into captured kernel arguments). Peers may already
+// be one barrier ahead, hence wait_ge.
+__global__ void k_barrier_all(PeerTable pt, int *local_flags, int *epoch_cell) {
+  __shared__ int e_sh;
+  if (threadIdx.x == 0) {
+    int e = *epoch_cell + 1;
+    *epoch_cell = e;
+    e_sh = e;
+  }
+  __syncthreads();
+  int e = e_sh;
   int t = threadIdx.x;
   if (t < pt.world) {
     int *peer_flags = symm_at(pt, local_flags, t);
-    st_release<Scope::Sys>(peer_flags + pt.rank, epoch);
-    wait_eq_one<Scope::Sys>(local_flags + t, epoch);
+    st_release<Scope::Sys>(peer_flags + pt.rank, e);
+    wait_ge_one<Scope::Sys>(local_flags + t, e);
   }
   __syncthreads();
 }
 
-void launch_barrier_all(const PeerTable &pt, int *local_flags, int epoch,
+void launch_barrier_all(const PeerTable &pt, int *local_flags, int *epoch_cell,
                         hipStream_t stream) {
   hipLaunchKernelGGL(k_barrier_all, dim3(1), dim3(kWave), 0, stream, pt,
-                     local_flags, epoch);
+                     local_flags, epoch_cell);
 }
 
 __global__ void k_signal_set(int *flag, int val) {

@@ -221,10 +221,11 @@ static void stream_sync(uintptr_t stream) {
 // ---------------------------------------------------------------------------
 // Kernel wrappers
 // ---------------------------------------------------------------------------
-static void barrier_all(uintptr_t flags_ptr, int epoch, uintptr_t stream) {
+static void barrier_all(uintptr_t flags_ptr, uintptr_t epoch_cell,
+                        uintptr_t stream) {
   check_active();
-  launch_barrier_all(g_heap.pt, reinterpret_cast<int *>(flags_ptr), epoch,
-                     as_stream(stream));
+  launch_barrier_all(g_heap.pt, reinterpret_cast<int *>(flags_ptr),
+                     reinterpret_cast<int *>(epoch_cell), as_stream(stream));
   TD_CHECK_HIP(hipGetLastError());
 }
 

@@ -42,10 +42,10 @@ class AGGemmContext:
     chunks_per_rank: int
     ws: SymmBuffer              # [world, max_m_per_rank, K] bf16
     flags: SymmBuffer           # [world * chunks_per_rank] int32
-    epoch: int = 0
+    epoch: int = 0              # CPU backend only; HIP flow is stateless
     comm_streams: List = field(default_factory=list)
-    epoch_src: Optional[SymmBuffer] = None  # staging int32 per comm stream
     ready_ev: Optional[object] = None
+    join_evs: List = field(default_factory=list)
 
     @property
     def world(self) -> int:
@@ -70,8 +70,8 @@ def create_ag_gemm_context(max_m_per_rank: int, k: int,
     if heap.backend == "hip":
         n_streams = min(num_comm_streams, max(world - 1, 1))
         ctx.comm_streams = [torch.cuda.Stream() for _ in range(n_streams)]
-        ctx.epoch_src = heap.alloc_buffer((n_streams,), torch.int32)
         ctx.ready_ev = torch.cuda.Event()
+        ctx.join_evs = [torch.cuda.Event() for _ in range(n_streams)]
     return ctx
 
 
@@ -86,44 +86,48 @@ def ag_gemm(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
     n = w.shape[0]
     assert k == ctx.k and m <= ctx.max_m_per_rank
     world, rank = ctx.world, ctx.rank
-    ctx.epoch += 1
 
     if ctx.heap.backend == "cpu":
+        ctx.epoch += 1
         c = _ag_gemm_cpu(a, w, ctx, m)
         if gathered_out:
             return c, ctx.ws.local()[:, :m].reshape(world * m, k)
         return c
 
+    # Every launch below takes only constant arguments and device-resident
+    # state, so the whole op is hipGraph-capturable (the Engine captures the
+    # decode step; cf. reference engine.py:75-105 requirement).
     assert a.dtype == torch.bfloat16 and a.is_contiguous()
     heap, _C = ctx.heap, ctx.heap._C
     compute = torch.cuda.current_stream()
     chunks = ctx.chunks_per_rank
     rows_per_chunk = ctx.max_m_per_rank // chunks
-    assert m % rows_per_chunk == 0 or m == ctx.max_m_per_rank, \
-        "v1: m must equal max_m_per_rank or a multiple of the chunk rows"
+    assert m == ctx.max_m_per_rank, "v1 consumer requires m == max_m_per_rank"
     m_chunks = (m + rows_per_chunk - 1) // rows_per_chunk
     chunk_bytes = rows_per_chunk * k * 2
 
-    # 1. entry barrier: workspace of the previous call is consumed
+    # 1. reset my flags, then entry barrier (workspace of the previous call
+    #    is consumed; peers push only after the barrier, so the reset can't
+    #    race their signals)
+    _C.reset_flags(ctx.flags.ptr(), world * chunks, 0, compute.cuda_stream)
     heap.barrier_all_on_stream(compute)
 
     # 2. local shard into my segment + my own flags (compute stream order)
     my_seg_ptr = ctx.ws.ptr() + rank * ctx.max_m_per_rank * k * 2
     _C.memcpy_async(my_seg_ptr, a.data_ptr(), m * k * 2, compute.cuda_stream)
-    my_flag_ptr = ctx.flags.ptr() + rank * chunks * 4
-    _C.reset_flags(my_flag_ptr, chunks, ctx.epoch, compute.cuda_stream)
+    _C.reset_flags(ctx.flags.ptr() + rank * chunks * 4, chunks, 1,
+                   compute.cuda_stream)
 
-    # 3. producer: push my shard to every peer on the comm stream pool
+    # 3. producer: push my shard to every peer over the comm stream pool
+    #    (SDMA copies — data chunk then a 4B flag copy from the constant-1
+    #    cell; stream order makes flag-after-data correct)
     ctx.ready_ev.record(compute)
     ns = len(ctx.comm_streams)
     for s in range(ns):
         ctx.comm_streams[s].wait_event(ctx.ready_ev)
-        _C.memset32_async(ctx.epoch_src.ptr() + s * 4, ctx.epoch, 1,
-                          ctx.comm_streams[s].cuda_stream)
     for i in range(world - 1):
         peer = (rank + 1 + i) % world
         stream = ctx.comm_streams[i % ns]
-        src_idx = i % ns
         dst_seg = ctx.ws.ptr(peer) + rank * ctx.max_m_per_rank * k * 2
         dst_flag = ctx.flags.ptr(peer) + rank * chunks * 4
         for c in range(m_chunks):
@@ -131,22 +135,22 @@ def ag_gemm(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
             _C.memcpy_async(dst_seg + c * chunk_bytes,
                             a.data_ptr() + c * chunk_bytes, nbytes,
                             stream.cuda_stream)
-            _C.memcpy_async(dst_flag + c * 4,
-                            ctx.epoch_src.ptr() + src_idx * 4, 4,
+            _C.memcpy_async(dst_flag + c * 4, heap.one_src.ptr(), 4,
                             stream.cuda_stream)
 
-    # 4. consumer GEMM on the compute stream
+    # 4. consumer GEMM on the compute stream (waits per-tile on chunk flags)
     m_total = world * m
     if out is None:
         out = torch.empty(m_total, n, dtype=torch.bfloat16, device=a.device)
-    # workspace rows are laid out [world, max_m_per_rank, K]; for m <
-    # max_m_per_rank the gathered matrix is strided per segment — v1 requires
-    # m == max_m_per_rank for the contiguous consumer.
-    assert m == ctx.max_m_per_rank, "v1 consumer requires m == max_m_per_rank"
     _C.ag_gemm_consumer_bf16(
         ctx.ws.ptr(), w.data_ptr(), out.data_ptr(), m_total, n, k,
-        ctx.flags.ptr(), chunks, m, world, rank, ctx.epoch,
+        ctx.flags.ptr(), chunks, m, world, rank, 1,
         compute.cuda_stream)
+    # join comm streams back into the compute stream (after the consumer
+    # launch: no serialization, but graph capture requires joined forks)
+    for s in range(min(ns, max(world - 1, 1))):
+        ctx.join_evs[s].record(ctx.comm_streams[s])
+        compute.wait_event(ctx.join_evs[s])
     if gathered_out:
         return out, ctx.ws.local().reshape(world * ctx.max_m_per_rank, k)[:m_total]
     return out

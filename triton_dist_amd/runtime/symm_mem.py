@@ -110,9 +110,16 @@ class SymmHeap:
             self._shm = cpu_shm.CpuShmHeap(group, self.size)
             self._offset = 4096
 
-        # internal flags for barrier_all: world int32 per rank
+        # internal flags for barrier_all: world int32 per rank, plus a
+        # device-resident epoch cell (hipGraph-safe: the barrier kernel
+        # increments it on device, so replays stay correct).
         self._barrier_buf = self.alloc_buffer((max(self.world, 8),),
                                               torch.int32)
+        self._barrier_epoch_cell = self.alloc_buffer((1,), torch.int32)
+        # constant int32 == 1 used as the SDMA signal source for producers
+        self.one_src = self.alloc_buffer((1,), torch.int32)
+        if self.backend == "hip":
+            self.one_src.local().fill_(1)
 
     # -- allocation (collective: call in the same order on all ranks) -------
     def alloc(self, nbytes: int, align: int = _ALIGN) -> int:
@@ -158,12 +165,12 @@ class SymmHeap:
     def barrier_all_on_stream(self, stream: Optional[torch.cuda.Stream] = None):
         """Device-side all-to-all barrier (epoch-counting), enqueued on the
         stream; CPU backend: blocking shared-memory barrier."""
-        self._epoch += 1
         if self.backend == "hip":
             s = stream if stream is not None else torch.cuda.current_stream()
-            self._C.barrier_all(self._barrier_buf.ptr(), self._epoch,
-                                s.cuda_stream)
+            self._C.barrier_all(self._barrier_buf.ptr(),
+                                self._barrier_epoch_cell.ptr(), s.cuda_stream)
         else:
+            self._epoch += 1
             from . import cpu_shm
             me = self._barrier_buf.local()
             for r in range(self.world):
