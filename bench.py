@@ -1,0 +1,133 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: Qwen3-32B TP=N end-to-end decode (BASELINE config 5).
+
+Contract: `python bench.py --gpus N --steps K --warmup W` — launched for N>1
+as one rank per GPU via torch.distributed.run. Does W untimed warmup decode
+steps, then times EXACTLY K steps bracketed by barrier+synchronize on both
+sides, takes the MAX over ranks, and rank 0 prints ONE JSON line.
+
+Weak scaling: global batch = 512 * N (per-GPU batch fixed at 512), TP=N,
+synthetic prompts, random-init weights (no network for checkpoints).
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--model", default="qwen3-32b")
+    p.add_argument("--batch-per-gpu", type=int, default=512)
+    p.add_argument("--ctx", type=int, default=128)
+    p.add_argument("--mode", default="ag_rs",
+                   choices=["ag_rs", "allreduce", "torch"])
+    p.add_argument("--no-graph", action="store_true")
+    p.add_argument("--seed", type=int, default=1234)
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    import torch.distributed as dist
+
+    import triton_dist_amd as td
+    from triton_dist_amd.models import DenseLLM, Engine, get_config
+
+    td.initialize_distributed(seed=args.seed)
+    world = td.world_size()
+    rank = td.rank()
+    n_gpus = world if world > 1 else args.gpus
+    assert world == n_gpus or world == 1, \
+        f"launch with torch.distributed.run for --gpus {args.gpus}"
+    on_gpu = torch.cuda.is_available()
+    device = "cuda" if on_gpu else "cpu"
+    heap = td.init_symm_heap()
+
+    batch = args.batch_per_gpu * world
+    max_len = args.ctx + args.warmup + args.steps + 16
+
+    cfg = get_config(args.model, tp_mode=args.mode, max_length=max_len + 64)
+    model = DenseLLM(cfg, device=device)
+    model.init_weights(seed=args.seed)
+    if args.mode == "ag_rs":
+        model.init_dist_ctx(max_m_total=batch)
+
+    eng = Engine(model, batch=batch, max_len=max_len,
+                 use_graph=on_gpu and not args.no_graph)
+
+    # synthetic prompt + prefill (untimed)
+    g = torch.Generator().manual_seed(args.seed)
+    prompt = torch.randint(0, cfg.vocab, (batch, args.ctx), generator=g
+                           ).to(device)
+    eng.kv.reset()
+    tok = model.prefill(prompt, eng.kv)
+
+    # warmup decode steps (untimed)
+    if eng.use_graph:
+        eng._ensure_graph()
+        eng._token_buf.copy_(tok)
+        for _ in range(args.warmup):
+            eng.graph.replay()
+    else:
+        for _ in range(args.warmup):
+            tok = eng.decode_once(tok)
+
+    # timed region
+    if dist.is_initialized():
+        dist.barrier()
+    if on_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    if eng.use_graph:
+        for _ in range(args.steps):
+            eng.graph.replay()
+    else:
+        for _ in range(args.steps):
+            tok = eng.decode_once(tok)
+    if on_gpu:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    if dist.is_initialized():
+        dist.barrier()
+        t = torch.tensor([elapsed])
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    ms_per_step = elapsed * 1e3 / args.steps
+    tok_per_s = batch * args.steps / elapsed
+    if rank == 0:
+        print(json.dumps({
+            "metric": f"{args.model.replace('-', '_')}_tp_decode_tokens_per_s",
+            "value": round(tok_per_s, 2),
+            "unit": "tokens/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": batch,
+                "seq_len": args.ctx,
+                "parallelism": f"tp{n_gpus}",
+                "tp_mode": args.mode,
+                "graph": eng.use_graph,
+            },
+        }))
+    td.finalize_distributed()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,3 @@
+from .norm import rms_norm, Rotary  # noqa: F401
+from .tp_mlp import TP_MLP  # noqa: F401
+from .tp_attn import TP_Attn  # noqa: F401

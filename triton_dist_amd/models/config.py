@@ -1,0 +1,50 @@
+"""Model configs (capability parity with Triton-distributed
+python/triton_dist/models/config.py:30-37 ModelConfig + HF-geometry
+presets; weights here are random-init — no network in this environment).
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+
+
+@dataclass
+class ModelConfig:
+    name: str = "qwen3-32b"
+    hidden: int = 5120
+    intermediate: int = 25600
+    n_layers: int = 64
+    n_heads: int = 64
+    n_kv_heads: int = 8
+    head_dim: int = 128
+    vocab: int = 151936
+    rope_base: float = 1e6
+    rms_eps: float = 1e-6
+    qk_norm: bool = True
+    tie_embeddings: bool = False
+    max_length: int = 2048
+    dtype: str = "bfloat16"
+    tp_mode: str = "ag_rs"  # ag_rs | allreduce | torch
+
+
+PRESETS = {
+    # Qwen3-32B geometry (HF Qwen/Qwen3-32B config.json)
+    "qwen3-32b": dict(hidden=5120, intermediate=25600, n_layers=64,
+                      n_heads=64, n_kv_heads=8, head_dim=128, vocab=151936),
+    # Qwen3-8B geometry
+    "qwen3-8b": dict(hidden=4096, intermediate=12288, n_layers=36,
+                     n_heads=32, n_kv_heads=8, head_dim=128, vocab=151936),
+    # tiny config for CPU tests
+    "tiny": dict(hidden=256, intermediate=512, n_layers=2, n_heads=4,
+                 n_kv_heads=2, head_dim=64, vocab=512),
+    # tiny GPU config: every sharded GEMM dim satisfies the 128/64 tiling
+    # at TP in {1, 2}
+    "tiny-gpu": dict(hidden=512, intermediate=1024, n_layers=2, n_heads=4,
+                     n_kv_heads=2, head_dim=128, vocab=1024),
+}
+
+
+def get_config(name: str, **overrides) -> ModelConfig:
+    cfg = ModelConfig(name=name, **PRESETS[name])
+    for k, v in overrides.items():
+        setattr(cfg, k, v)
+    return cfg

@@ -4,6 +4,7 @@ from .allgather_gemm import (  # noqa: F401
     create_ag_gemm_context,
     ag_gemm,
     ag_gemm_ref,
+    allgather,
 )
 from .gemm_rs import (  # noqa: F401
     GemmRSContext,

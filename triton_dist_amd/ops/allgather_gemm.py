@@ -156,6 +156,69 @@ def ag_gemm(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
     return out
 
 
+def allgather(a: torch.Tensor, ctx: AGGemmContext,
+              out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Standalone push all-gather through the ctx workspace (the consumer is
+    a wait kernel instead of the fused GEMM; cf. reference
+    kernels/amd/allgather.py capability). Returns [world*m, K]."""
+    m, k = a.shape
+    assert k == ctx.k and m == ctx.max_m_per_rank
+    world, rank = ctx.world, ctx.rank
+    chunks = ctx.chunks_per_rank
+
+    if ctx.heap.backend == "cpu":
+        ctx.epoch += 1
+        ctx.heap.barrier_all()
+        for peer in range(world):
+            ctx.ws.peer(peer)[rank, :m].copy_(a)
+            fl = ctx.flags.peer(peer)
+            for c in range(chunks):
+                cpu_shm.notify(fl, rank * chunks + c, ctx.epoch)
+        fl = ctx.flags.local()
+        for i in range(world * chunks):
+            cpu_shm.wait_ge(fl, i, ctx.epoch)
+        gathered = ctx.ws.local()[:, :m].reshape(world * m, k)
+        if out is not None:
+            out.copy_(gathered)
+            return out
+        return gathered.clone()
+
+    heap, _C = ctx.heap, ctx.heap._C
+    compute = torch.cuda.current_stream()
+    rows_per_chunk = ctx.max_m_per_rank // chunks
+    chunk_bytes = rows_per_chunk * k * 2
+    _C.reset_flags(ctx.flags.ptr(), world * chunks, 0, compute.cuda_stream)
+    heap.barrier_all_on_stream(compute)
+    my_seg_ptr = ctx.ws.ptr() + rank * ctx.max_m_per_rank * k * 2
+    _C.memcpy_async(my_seg_ptr, a.data_ptr(), m * k * 2, compute.cuda_stream)
+    _C.reset_flags(ctx.flags.ptr() + rank * chunks * 4, chunks, 1,
+                   compute.cuda_stream)
+    ctx.ready_ev.record(compute)
+    ns = len(ctx.comm_streams)
+    for s in range(ns):
+        ctx.comm_streams[s].wait_event(ctx.ready_ev)
+    for i in range(world - 1):
+        peer = (rank + 1 + i) % world
+        stream = ctx.comm_streams[i % ns]
+        dst_seg = ctx.ws.ptr(peer) + rank * ctx.max_m_per_rank * k * 2
+        dst_flag = ctx.flags.ptr(peer) + rank * chunks * 4
+        for c in range(chunks):
+            _C.memcpy_async(dst_seg + c * chunk_bytes,
+                            a.data_ptr() + c * chunk_bytes, chunk_bytes,
+                            stream.cuda_stream)
+            _C.memcpy_async(dst_flag + c * 4, heap.one_src.ptr(), 4,
+                            stream.cuda_stream)
+    _C.wait_eq(ctx.flags.ptr(), world * chunks, 1, compute.cuda_stream)
+    for s in range(min(ns, max(world - 1, 1))):
+        ctx.join_evs[s].record(ctx.comm_streams[s])
+        compute.wait_event(ctx.join_evs[s])
+    gathered = ctx.ws.local().reshape(world * ctx.max_m_per_rank, k)
+    if out is not None:
+        out.copy_(gathered)
+        return out
+    return gathered
+
+
 def _ag_gemm_cpu(a, w, ctx, m):
     world, rank = ctx.world, ctx.rank
     chunks = ctx.chunks_per_rank
