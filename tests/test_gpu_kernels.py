@@ -128,3 +128,85 @@ def test_rs_reduce():
     torch.cuda.synchronize()
     ref = segs.float().sum(0)
     assert_allclose(out, ref, atol=5e-1, rtol=2e-2)
+
+
+def test_rmsnorm_kernels():
+    from triton_dist_amd.ops.fused import add_rms_norm_op, rms_norm_op
+    from triton_dist_amd.utils.testing import assert_allclose
+
+    torch.manual_seed(0)
+    x = torch.randn(333, 5120, device="cuda").to(torch.bfloat16)
+    w = torch.randn(5120, device="cuda").to(torch.bfloat16)
+    out = rms_norm_op(x, w)
+    x32 = x.float()
+    ref = x32 * torch.rsqrt(x32.pow(2).mean(-1, keepdim=True) + 1e-6) * w.float()
+    assert_allclose(out, ref, atol=3e-2, rtol=3e-2)
+
+    d = torch.randn_like(x)
+    nr, nm = add_rms_norm_op(d, x, w)
+    y = (x.float() + d.float())
+    assert_allclose(nr, y.to(torch.bfloat16), atol=2e-2, rtol=2e-2)
+    ref2 = y * torch.rsqrt(y.pow(2).mean(-1, keepdim=True) + 1e-6) * w.float()
+    assert_allclose(nm, ref2, atol=3e-2, rtol=3e-2)
+
+
+def test_swiglu_kernel():
+    import torch.nn.functional as F
+
+    from triton_dist_amd.ops.fused import swiglu_op
+    from triton_dist_amd.utils.testing import assert_allclose
+
+    torch.manual_seed(1)
+    h = torch.randn(256, 2 * 3200, device="cuda").to(torch.bfloat16)
+    out = swiglu_op(h, 3200)
+    ref = F.silu(h[:, :3200].float()) * h[:, 3200:].float()
+    assert_allclose(out, ref, atol=3e-2, rtol=3e-2)
+
+
+def test_qkv_prologue_and_flash_decode():
+    import torch.nn.functional as F
+
+    from triton_dist_amd.layers.norm import Rotary, rms_norm
+    from triton_dist_amd.ops.fused import (flash_decode_op,
+                                           qkv_prologue_decode_op)
+    from triton_dist_amd.utils.testing import assert_allclose
+
+    torch.manual_seed(2)
+    b, qh, kvh, d, maxlen = 16, 8, 1, 128, 96
+    seq = 37  # existing cache length; new token at position 37
+    rot = Rotary(d, maxlen, device="cuda")
+    qnw = torch.randn(d, device="cuda").abs().to(torch.bfloat16)
+    knw = torch.randn(d, device="cuda").abs().to(torch.bfloat16)
+    qkv = (torch.randn(b, (qh + 2 * kvh) * d, device="cuda") / 4).to(torch.bfloat16)
+    kc = (torch.randn(b, maxlen, kvh, d, device="cuda") / 4).to(torch.bfloat16)
+    vc = (torch.randn(b, maxlen, kvh, d, device="cuda") / 4).to(torch.bfloat16)
+    kc_ref, vc_ref = kc.clone(), vc.clone()
+    offset = torch.tensor(seq, dtype=torch.int64, device="cuda")
+
+    q_rot = qkv_prologue_decode_op(qkv, kc, vc, offset, rot.cos, rot.sin,
+                                   qnw, knw, qh, kvh, 1e-6, True)
+    torch.cuda.synchronize()
+
+    # torch reference of the prologue
+    q = qkv[:, :qh * d].view(b, 1, qh, d)
+    k = qkv[:, qh * d:(qh + kvh) * d].view(b, 1, kvh, d)
+    v = qkv[:, (qh + kvh) * d:].view(b, 1, kvh, d)
+    qn = rms_norm(q, qnw)
+    kn = rms_norm(k, knw)
+    pos = torch.full((b, 1), seq, device="cuda", dtype=torch.int64)
+    qr, kr = rot.apply(qn, kn, pos)
+    assert_allclose(q_rot.view(b, 1, qh, d), qr, atol=4e-2, rtol=4e-2)
+    assert_allclose(kc[:, seq], kr[:, 0], atol=4e-2, rtol=4e-2)
+    assert_allclose(vc[:, seq], v[:, 0], atol=1e-3, rtol=1e-3)
+    # untouched cache rows stay untouched
+    assert torch.equal(kc[:, :seq], kc_ref[:, :seq])
+    assert torch.equal(vc[:, seq + 1:], vc_ref[:, seq + 1:])
+
+    # flash decode vs sdpa over the valid prefix
+    out = flash_decode_op(q_rot, kc, vc, offset, qh, kvh)
+    torch.cuda.synchronize()
+    qs = q_rot.view(b, qh, 1, d).float()
+    ks = kc[:, :seq + 1].transpose(1, 2).float()  # [b, kvh, L, d]
+    vs = vc[:, :seq + 1].transpose(1, 2).float()
+    ref = F.scaled_dot_product_attention(qs, ks, vs, enable_gqa=True)
+    assert_allclose(out.view(b, qh, 1, d), ref, atol=4e-2, rtol=4e-2)

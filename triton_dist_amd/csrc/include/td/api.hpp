@@ -74,4 +74,26 @@ void launch_gemm_rs_producer_bf16(const GemmRsArgs &args, hipStream_t stream);
 void launch_rs_reduce_bf16(const void *segments, void *out, int world,
                            int rank, int m_per_rank, int n, hipStream_t stream);
 
+// kernels/elementwise.hip ----------------------------------------------------
+void launch_rmsnorm(const void *x, const void *w, void *out, int rows,
+                    int cols, float eps, hipStream_t stream);
+void launch_add_rmsnorm(const void *x, const void *resid_in, void *resid_out,
+                        const void *w, void *out, int rows, int cols,
+                        float eps, hipStream_t stream);
+void launch_swiglu(const void *h, void *out, int rows, int inter,
+                   hipStream_t stream);
+void launch_qkv_prologue_decode(const void *qkv, void *q_out, void *kcache,
+                                void *vcache, const void *cos_t,
+                                const void *sin_t, const void *qnw,
+                                const void *knw, const void *offset,
+                                int batch, int qh, int kvh, int max_len,
+                                float eps, bool use_qk_norm,
+                                hipStream_t stream);
+
+// kernels/attention.hip ------------------------------------------------------
+void launch_flash_decode(const void *q, const void *kcache,
+                         const void *vcache, void *out, const void *offset,
+                         int batch, int qh, int kvh, int max_len,
+                         hipStream_t stream);
+
 }  // namespace td

@@ -1,0 +1,158 @@
+// GQA flash-decode kernel (single new token per sequence).
+//
+// Capability parity with the reference's distributed flash-decode family
+// (Triton-distributed python/triton_dist/kernels/nvidia/flash_decode.py:
+// 130-1132 — split-KV online-softmax decode; the inter-rank LSE-merge
+// combine comes with the SP layer). MI355X-native design:
+//
+//   grid = (batch, kv_heads); block = 256 threads (4 waves).
+//   Each block serves ALL G = qh/kvh query heads of one (seq, kv head) —
+//   K/V rows are read once and reused G times (decode is KV-bandwidth
+//   bound; this is the whole game on a 8 TB/s HBM part).
+//   Per 32-position tile: threads (g = tid>>5, t = tid&31) compute one
+//   dot(q[g], k[pos]) each; per-g online-softmax (running max m, sum l)
+//   via 32-lane half-wave shuffles; P tile staged in LDS; every thread
+//   then updates its private (g, d) slice of the fp32 accumulator.
+//   Sequence length = *offset + 1 read from DEVICE memory (hipGraph-safe).
+//
+// Layouts: q [B, qh, 128] bf16 (post-RoPE), cache [B, max_len, kvh, 128].
+#include <stdexcept>
+
+#include "td/api.hpp"
+
+namespace td {
+
+using bf16 = __bf16;
+typedef __attribute__((ext_vector_type(8))) bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+constexpr int kD = 128;
+constexpr int kTile = 32;
+
+__global__ __launch_bounds__(256) void k_flash_decode(
+    const bf16 *__restrict__ q, const bf16 *__restrict__ kcache,
+    const bf16 *__restrict__ vcache, bf16 *__restrict__ out,
+    const long *__restrict__ offset, int qh, int kvh, int max_len,
+    float scale) {
+  const int b = blockIdx.x;
+  const int kh = blockIdx.y;
+  const int G = qh / kvh;  // query heads per kv head (<= 8 supported)
+  const int tid = threadIdx.x;
+  const int g = tid >> 5;       // 0..7 query-head slot
+  const int t = tid & 31;       // position within tile
+  const long seqlen = *offset + 1;
+
+  __shared__ bf16 k_lds[kTile][kD];
+  __shared__ bf16 v_lds[kTile][kD];
+  __shared__ float p_lds[8][kTile];
+  __shared__ float m_lds[8], r_lds[8], l_lds[8];
+  __shared__ bf16 q_lds[8][kD];
+
+  // load q for my block's G heads into LDS (pad missing heads with zeros)
+  for (int i = tid; i < 8 * kD / 8; i += 256) {
+    int hh = i / (kD / 8);
+    int c = (i % (kD / 8)) * 8;
+    bf16x8 v{};
+    if (hh < G)
+      v = *(const bf16x8 *)(q + (((size_t)b * qh) + kh * G + hh) * kD + c);
+    *(bf16x8 *)(&q_lds[hh][c]) = v;
+  }
+  if (tid < 8) {
+    m_lds[tid] = -1e30f;
+    l_lds[tid] = 0.f;
+  }
+  __syncthreads();
+
+  // per-thread accumulator: 4 (g,d) slices — thread owns dims
+  // d = (tid%32)*4 ... for head slot tid/32
+  float acc[4] = {};
+  const int my_d0 = t * 4;
+
+  const long ntiles = (seqlen + kTile - 1) / kTile;
+  for (long tile = 0; tile < ntiles; ++tile) {
+    const long pos0 = tile * kTile;
+    // stage K/V tile: 32 rows x 128 cols, 16B chunks: 32*16=512 chunks
+    __syncthreads();
+    for (int i = tid; i < kTile * kD / 8; i += 256) {
+      int r = i / (kD / 8);
+      int c = (i % (kD / 8)) * 8;
+      long pos = pos0 + r;
+      bf16x8 kv{}, vv{};
+      if (pos < seqlen) {
+        size_t base = (((size_t)b * max_len + pos) * kvh + kh) * kD + c;
+        kv = *(const bf16x8 *)(kcache + base);
+        vv = *(const bf16x8 *)(vcache + base);
+      }
+      *(bf16x8 *)(&k_lds[r][c]) = kv;
+      *(bf16x8 *)(&v_lds[r][c]) = vv;
+    }
+    __syncthreads();
+
+    // scores: thread (g, t) -> dot(q[g], k[t])
+    float s = -1e30f;
+    if (g < G && pos0 + t < seqlen) {
+      float d = 0.f;
+#pragma unroll
+      for (int c = 0; c < kD / 8; ++c) {
+        bf16x8 qv = *(const bf16x8 *)(&q_lds[g][c * 8]);
+        bf16x8 kv = *(const bf16x8 *)(&k_lds[t][c * 8]);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) d += (float)qv[j] * (float)kv[j];
+      }
+      s = d * scale;
+    }
+    // per-g max over the 32 positions (half-wave reduce)
+    float mx = s;
+    for (int off = 16; off > 0; off >>= 1)
+      mx = fmaxf(mx, __shfl_xor(mx, off));
+    // online-softmax update (thread t==0 of each g publishes)
+    float m_old = m_lds[g];
+    float m_new = fmaxf(m_old, mx);
+    float p = (s > -1e29f) ? __expf(s - m_new) : 0.f;
+    p_lds[g][t] = p;
+    float psum = p;
+    for (int off = 16; off > 0; off >>= 1) psum += __shfl_xor(psum, off);
+    if (t == 0) {
+      float r = __expf(m_old - m_new);
+      r_lds[g] = r;
+      l_lds[g] = l_lds[g] * r + psum;
+      m_lds[g] = m_new;
+    }
+    __syncthreads();
+
+    // accumulate: acc[d] = acc[d]*r + sum_t p[t] * V[t][d]
+    const float r = r_lds[g];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[j] *= r;
+    for (int tt = 0; tt < kTile; ++tt) {
+      float p = p_lds[g][tt];
+      if (p != 0.f) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[j] += p * (float)v_lds[tt][my_d0 + j];
+      }
+    }
+  }
+  __syncthreads();
+  if (g < G) {
+    float inv_l = 1.f / l_lds[g];
+    bf16 *dst = out + (((size_t)b * qh) + kh * G + g) * kD + my_d0;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) dst[j] = (bf16)(acc[j] * inv_l);
+  }
+}
+
+void launch_flash_decode(const void *q, const void *kcache,
+                         const void *vcache, void *out, const void *offset,
+                         int batch, int qh, int kvh, int max_len,
+                         hipStream_t stream) {
+  if (qh / kvh > 8 || qh % kvh)
+    throw std::runtime_error("flash_decode: qh/kvh must divide and be <= 8");
+  float scale = 1.f / sqrtf((float)kD);
+  hipLaunchKernelGGL(k_flash_decode, dim3(batch, kvh), dim3(256), 0, stream,
+                     (const bf16 *)q, (const bf16 *)kcache,
+                     (const bf16 *)vcache, (bf16 *)out,
+                     (const long *)offset, qh, kvh, max_len, scale);
+}
+
+}  // namespace td

@@ -1,0 +1,225 @@
+// Fused memory-bound kernels for the decode path: RMSNorm, residual-add +
+// RMSNorm, SwiGLU, and the fused qkv-prologue (q/k head-RMSNorm + RoPE +
+// KV-cache append). All bf16 with fp32 math, short8-vectorized loads
+// (CDNA4 guide G13: scalar bf16 loads are 2-2.5x slower), grid-stride.
+//
+// These replace the reference's reliance on Triton-JIT elementwise fusion
+// (and the eager-torch fp32 chains that an unfused port would inherit).
+#include <stdexcept>
+
+#include "td/api.hpp"
+
+namespace td {
+
+using bf16 = __bf16;
+typedef __attribute__((ext_vector_type(8))) bf16 bf16x8;
+
+TD_DEV float bf2f(bf16 v) { return (float)v; }
+
+// ---------------------------------------------------------------------------
+// RMSNorm: one workgroup per row; cols % 8 == 0, cols <= 256*8*ITER.
+// optionally fused residual add: y = x + resid; out = rmsnorm(y) * w, and y
+// is written back to resid_out (the new residual stream).
+// ---------------------------------------------------------------------------
+template <bool ADD>
+__global__ void k_rmsnorm(const bf16 *__restrict__ x,
+                          const bf16 *__restrict__ resid_in,
+                          bf16 *__restrict__ resid_out,
+                          const bf16 *__restrict__ w, bf16 *__restrict__ out,
+                          int cols, float eps) {
+  const int row = blockIdx.x;
+  const bf16 *xr = x + (size_t)row * cols;
+  const bf16 *rr = ADD ? resid_in + (size_t)row * cols : nullptr;
+  bf16 *ro = ADD ? resid_out + (size_t)row * cols : nullptr;
+  bf16 *orow = out + (size_t)row * cols;
+
+  // pass 1: load 8-wide, accumulate sum of squares (values kept in regs up
+  // to 8 iters = 16K cols; beyond that re-read from L2)
+  constexpr int MAXV = 8;
+  bf16x8 vals[MAXV];
+  float ss = 0.f;
+  const int nv = cols / 8;
+  const int per_thread = (nv + blockDim.x - 1) / blockDim.x;
+  const bool cached = per_thread <= MAXV;
+  for (int i = threadIdx.x, vi = 0; i < nv; i += blockDim.x, ++vi) {
+    bf16x8 v = *(const bf16x8 *)(xr + i * 8);
+    if (ADD) {
+      bf16x8 r = *(const bf16x8 *)(rr + i * 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v[j] = (bf16)(bf2f(v[j]) + bf2f(r[j]));
+      *(bf16x8 *)(ro + i * 8) = v;
+    }
+    if (cached && vi < MAXV) vals[vi] = v;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = bf2f(v[j]);
+      ss += f * f;
+    }
+  }
+  // block reduce
+  __shared__ float red[16];
+  for (int off = 32; off > 0; off >>= 1) ss += __shfl_down(ss, off);
+  if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = ss;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float t = 0.f;
+    for (int i = 0; i < (int)(blockDim.x >> 6); ++i) t += red[i];
+    red[0] = rsqrtf(t / cols + eps);
+  }
+  __syncthreads();
+  const float scale = red[0];
+  // pass 2: scale + weight
+  for (int i = threadIdx.x, vi = 0; i < nv; i += blockDim.x, ++vi) {
+    bf16x8 v = (cached && vi < MAXV) ? vals[vi]
+                                     : *(const bf16x8 *)((ADD ? ro : xr) + i * 8);
+    bf16x8 wv = *(const bf16x8 *)(w + i * 8);
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      o[j] = (bf16)(bf2f(v[j]) * scale * bf2f(wv[j]));
+    *(bf16x8 *)(orow + i * 8) = o;
+  }
+}
+
+void launch_rmsnorm(const void *x, const void *w, void *out, int rows,
+                    int cols, float eps, hipStream_t stream) {
+  if (cols % 8) throw std::runtime_error("rmsnorm: cols % 8 != 0");
+  hipLaunchKernelGGL((k_rmsnorm<false>), dim3(rows), dim3(256), 0, stream,
+                     (const bf16 *)x, nullptr, nullptr, (const bf16 *)w,
+                     (bf16 *)out, cols, eps);
+}
+
+void launch_add_rmsnorm(const void *x, const void *resid_in, void *resid_out,
+                        const void *w, void *out, int rows, int cols,
+                        float eps, hipStream_t stream) {
+  if (cols % 8) throw std::runtime_error("rmsnorm: cols % 8 != 0");
+  hipLaunchKernelGGL((k_rmsnorm<true>), dim3(rows), dim3(256), 0, stream,
+                     (const bf16 *)x, (const bf16 *)resid_in,
+                     (bf16 *)resid_out, (const bf16 *)w, (bf16 *)out, cols,
+                     eps);
+}
+
+// ---------------------------------------------------------------------------
+// SwiGLU: h = [gate | up] rows of 2*inter; out = silu(gate) * up.
+// ---------------------------------------------------------------------------
+__global__ void k_swiglu(const bf16 *__restrict__ h, bf16 *__restrict__ out,
+                         size_t rows, int inter) {
+  size_t i = ((size_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  size_t stride = (size_t)gridDim.x * blockDim.x * 8;
+  size_t total = rows * (size_t)inter;
+  for (; i < total; i += stride) {
+    size_t r = i / inter;
+    size_t c = i % inter;
+    bf16x8 g = *(const bf16x8 *)(h + r * 2 * inter + c);
+    bf16x8 u = *(const bf16x8 *)(h + r * 2 * inter + inter + c);
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gf = bf2f(g[j]);
+      float s = gf / (1.f + __expf(-gf));
+      o[j] = (bf16)(s * bf2f(u[j]));
+    }
+    *(bf16x8 *)(out + i) = o;
+  }
+}
+
+void launch_swiglu(const void *h, void *out, int rows, int inter,
+                   hipStream_t stream) {
+  if (inter % 8) throw std::runtime_error("swiglu: inter % 8 != 0");
+  size_t work = (size_t)rows * inter / 8;
+  int blocks = (int)((work + 255) / 256);
+  if (blocks > 2048) blocks = 2048;
+  hipLaunchKernelGGL(k_swiglu, dim3(blocks), dim3(256), 0, stream,
+                     (const bf16 *)h, (bf16 *)out, (size_t)rows, inter);
+}
+
+// ---------------------------------------------------------------------------
+// Fused decode-qkv prologue: for one token per sequence,
+//   qkv row [B, (qh + 2*kvh) * D]  (D = 128)
+//   -> q/k per-head RMSNorm (Qwen3 qk_norm) -> RoPE at position `*offset`
+//   -> write K,V into the cache at [b, *offset, h, :]
+//   -> write rotated q to q_out [B, qh, D]
+// One wave per (b, head). rot tables: cos/sin [max_pos, D/2] fp32.
+// ---------------------------------------------------------------------------
+__global__ void k_qkv_prologue_decode(
+    const bf16 *__restrict__ qkv, bf16 *__restrict__ q_out,
+    bf16 *__restrict__ kcache, bf16 *__restrict__ vcache,
+    const float *__restrict__ cos_t, const float *__restrict__ sin_t,
+    const bf16 *__restrict__ qnw, const bf16 *__restrict__ knw,
+    const long *__restrict__ offset, int qh, int kvh, int max_len,
+    float eps, int use_qk_norm) {
+  constexpr int D = 128;
+  const int b = blockIdx.x;
+  const int h = blockIdx.y;       // 0..qh+2*kvh-1: q heads, then k, then v
+  const int lane = threadIdx.x;   // 64 lanes, 2 elems each
+  const int nh = qh + 2 * kvh;
+  const long pos = *offset;
+  const bf16 *src = qkv + ((size_t)b * nh + h) * D;
+  float v0 = bf2f(src[lane * 2]);
+  float v1 = bf2f(src[lane * 2 + 1]);
+
+  const bool is_q = h < qh;
+  const bool is_k = h >= qh && h < qh + kvh;
+  if (is_q || is_k) {
+    if (use_qk_norm) {
+      // head RMSNorm over D
+      float ss = v0 * v0 + v1 * v1;
+      for (int off = 32; off > 0; off >>= 1) ss += __shfl_down(ss, off);
+      float scale = rsqrtf(__shfl(ss, 0) / D + eps);
+      const bf16 *nw = is_q ? qnw : knw;
+      v0 *= scale * bf2f(nw[lane * 2]);
+      v1 *= scale * bf2f(nw[lane * 2 + 1]);
+    }
+    // RoPE (half-split layout: pairs are (d, d + D/2))
+    // lanes 0..31 hold d in [0,64): each lane's 2 elems are d=2l, 2l+1
+    // pair element d+64 lives on lane l+32. Use shfl to fetch partner.
+    float p0 = __shfl_xor(v0, 32);
+    float p1 = __shfl_xor(v1, 32);
+    int d2 = (lane & 31) * 2;  // frequency index 0..62
+    float c0 = cos_t[pos * (D / 2) + d2];
+    float s0 = sin_t[pos * (D / 2) + d2];
+    float c1 = cos_t[pos * (D / 2) + d2 + 1];
+    float s1 = sin_t[pos * (D / 2) + d2 + 1];
+    float r0, r1;
+    if (lane < 32) {  // x1' = x1*cos - x2*sin
+      r0 = v0 * c0 - p0 * s0;
+      r1 = v1 * c1 - p1 * s1;
+    } else {          // x2' = x2*cos + x1*sin
+      r0 = v0 * c0 + p0 * s0;
+      r1 = v1 * c1 + p1 * s1;
+    }
+    v0 = r0;
+    v1 = r1;
+  }
+
+  if (is_q) {
+    bf16 *dst = q_out + ((size_t)b * qh + h) * D;
+    dst[lane * 2] = (bf16)v0;
+    dst[lane * 2 + 1] = (bf16)v1;
+  } else {
+    const int kh = h - qh;
+    const bool k_side = kh < kvh;
+    const int hh = k_side ? kh : kh - kvh;
+    bf16 *cache = k_side ? kcache : vcache;
+    bf16 *dst = cache + (((size_t)b * max_len + pos) * kvh + hh) * D;
+    dst[lane * 2] = (bf16)v0;
+    dst[lane * 2 + 1] = (bf16)v1;
+  }
+}
+
+void launch_qkv_prologue_decode(const void *qkv, void *q_out, void *kcache,
+                                void *vcache, const void *cos_t,
+                                const void *sin_t, const void *qnw,
+                                const void *knw, const void *offset,
+                                int batch, int qh, int kvh, int max_len,
+                                float eps, bool use_qk_norm,
+                                hipStream_t stream) {
+  hipLaunchKernelGGL(k_qkv_prologue_decode, dim3(batch, qh + 2 * kvh),
+                     dim3(64), 0, stream, (const bf16 *)qkv, (bf16 *)q_out,
+                     (bf16 *)kcache, (bf16 *)vcache, (const float *)cos_t,
+                     (const float *)sin_t, (const bf16 *)qnw,
+                     (const bf16 *)knw, (const long *)offset, qh, kvh,
+                     max_len, eps, use_qk_norm ? 1 : 0);
+}
+
+}  // namespace td

@@ -320,6 +320,52 @@ static void rs_reduce_bf16(uintptr_t segments, uintptr_t out, int world,
   TD_CHECK_HIP(hipGetLastError());
 }
 
+static void rmsnorm(uintptr_t x, uintptr_t w, uintptr_t out, int rows,
+                    int cols, float eps, uintptr_t stream) {
+  launch_rmsnorm((void *)x, (void *)w, (void *)out, rows, cols, eps,
+                 as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
+static void add_rmsnorm(uintptr_t x, uintptr_t resid_in, uintptr_t resid_out,
+                        uintptr_t w, uintptr_t out, int rows, int cols,
+                        float eps, uintptr_t stream) {
+  launch_add_rmsnorm((void *)x, (void *)resid_in, (void *)resid_out,
+                     (void *)w, (void *)out, rows, cols, eps,
+                     as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
+static void swiglu(uintptr_t h, uintptr_t out, int rows, int inter,
+                   uintptr_t stream) {
+  launch_swiglu((void *)h, (void *)out, rows, inter, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
+static void qkv_prologue_decode(uintptr_t qkv, uintptr_t q_out,
+                                uintptr_t kcache, uintptr_t vcache,
+                                uintptr_t cos_t, uintptr_t sin_t,
+                                uintptr_t qnw, uintptr_t knw,
+                                uintptr_t offset, int batch, int qh, int kvh,
+                                int max_len, float eps, bool use_qk_norm,
+                                uintptr_t stream) {
+  launch_qkv_prologue_decode((void *)qkv, (void *)q_out, (void *)kcache,
+                             (void *)vcache, (void *)cos_t, (void *)sin_t,
+                             (void *)qnw, (void *)knw, (void *)offset, batch,
+                             qh, kvh, max_len, eps, use_qk_norm,
+                             as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
+static void flash_decode(uintptr_t q, uintptr_t kcache, uintptr_t vcache,
+                         uintptr_t out, uintptr_t offset, int batch, int qh,
+                         int kvh, int max_len, uintptr_t stream) {
+  launch_flash_decode((void *)q, (void *)kcache, (void *)vcache, (void *)out,
+                      (void *)offset, batch, qh, kvh, max_len,
+                      as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
 PYBIND11_MODULE(_C, m) {
   m.doc() = "triton_dist_amd native core: hipIpc symmetric heap + gfx950 kernels";
   m.def("heap_init", &heap_init, py::arg("rank"), py::arg("world"),
@@ -346,4 +392,9 @@ PYBIND11_MODULE(_C, m) {
   m.def("ag_gemm_consumer_bf16", &ag_gemm_consumer_bf16);
   m.def("gemm_rs_producer_bf16", &gemm_rs_producer_bf16);
   m.def("rs_reduce_bf16", &rs_reduce_bf16);
+  m.def("rmsnorm", &rmsnorm);
+  m.def("add_rmsnorm", &add_rmsnorm);
+  m.def("swiglu", &swiglu);
+  m.def("qkv_prologue_decode", &qkv_prologue_decode);
+  m.def("flash_decode", &flash_decode);
 }

@@ -80,9 +80,23 @@ class TP_Attn:
         return q, k, v
 
     def _attention(self, qkv: torch.Tensor, kv_cache, layer_idx: int,
-                   pos: torch.Tensor, b: int, s: int, prefill: bool
-                   ) -> torch.Tensor:
+                   pos: torch.Tensor, b: int, s: int, prefill: bool,
+                   native: bool = False) -> torch.Tensor:
         """qkv: [b*s, (qh+2kvh)*D] -> attention output [b*s, qh*D]."""
+        if (native and not prefill and s == 1 and qkv.is_cuda
+                and self.head_dim == 128 and kv_cache is not None):
+            # fused HIP decode path: qk-norm + RoPE + cache append, then
+            # GQA flash-decode (csrc/kernels/{elementwise,attention}.hip)
+            from ..ops.fused import flash_decode_op, qkv_prologue_decode_op
+
+            q_rot = qkv_prologue_decode_op(
+                qkv, kv_cache.k[layer_idx], kv_cache.v[layer_idx],
+                kv_cache.offset, self.rotary.cos, self.rotary.sin,
+                self.q_norm_w, self.k_norm_w, self.qh, self.kvh,
+                self.rms_eps, self.qk_norm)
+            return flash_decode_op(q_rot, kv_cache.k[layer_idx],
+                                   kv_cache.v[layer_idx], kv_cache.offset,
+                                   self.qh, self.kvh)
         q, k, v = self._qkv_split(qkv, b, s)
         if self.qk_norm:
             q = rms_norm(q, self.q_norm_w, self.rms_eps)
@@ -112,12 +126,12 @@ class TP_Attn:
         if self.mode == "ag_rs":
             qkv = ag_gemm(x, self.w_qkv, self.ag_ctx)       # [M, qkv_dim]
             attn = self._attention(qkv, kv_cache, layer_idx, pos, b, s,
-                                   prefill).to(self.dtype)
+                                   prefill, native=True).to(self.dtype)
             return gemm_rs(attn, self.w_o, self.rs_ctx)     # [M/world, H]
         if self.mode == "allreduce":
             qkv = gemm(x, self.w_qkv)
             attn = self._attention(qkv, kv_cache, layer_idx, pos, b, s,
-                                   prefill).to(self.dtype)
+                                   prefill, native=True).to(self.dtype)
             out = gemm(attn, self.w_o)
             dist.all_reduce(out)
             return out
@@ -127,9 +141,17 @@ class TP_Attn:
 
     def torch_fwd(self, x, kv_cache=None, layer_idx=0, pos=None, b=1, s=1,
                   prefill=False):
-        qkv = (x.float() @ self.w_qkv.float().t()).to(self.dtype)
+        # bf16 matmul on GPU (hipBLASLt — CDNA4 has no fp32 MFMA, fp32
+        # matmul would fall to the 157 TF vector ALU); fp32 on CPU
+        if x.is_cuda:
+            qkv = x @ self.w_qkv.t()
+        else:
+            qkv = (x.float() @ self.w_qkv.float().t()).to(self.dtype)
         attn = self._attention(qkv, kv_cache, layer_idx, pos, b, s, prefill)
-        out = (attn.float() @ self.w_o.float().t()).to(self.dtype)
+        if x.is_cuda:
+            out = attn.to(self.dtype) @ self.w_o.t()
+        else:
+            out = (attn.float() @ self.w_o.float().t()).to(self.dtype)
         if dist.is_initialized() and self.world > 1:
             if out.is_cuda and dist.get_backend() == "gloo":
                 cpu = out.cpu()

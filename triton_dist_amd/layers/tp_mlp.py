@@ -66,16 +66,15 @@ class TP_MLP:
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         """ag_rs: x is the batch shard [M/world, hidden] -> [M/world, hidden].
         allreduce/torch: x is replicated [M, hidden] -> [M, hidden]."""
+        from ..ops.fused import swiglu_op
+
         if self.mode == "ag_rs":
             h = ag_gemm(x, self.w_gate_up, self.ag_ctx)     # [M, 2*I_s]
-            act = F.silu(h[:, :self.inter_shard].float()) \
-                * h[:, self.inter_shard:].float()
-            act = act.to(self.dtype)
+            act = swiglu_op(h, self.inter_shard)
             return gemm_rs(act, self.w_down, self.rs_ctx)   # [M/world, hidden]
         if self.mode == "allreduce":
             h = gemm(x, self.w_gate_up)
-            act = (F.silu(h[:, :self.inter_shard].float())
-                   * h[:, self.inter_shard:].float()).to(self.dtype)
+            act = swiglu_op(h, self.inter_shard)
             partial = gemm(act, self.w_down)
             dist.all_reduce(partial)
             return partial
@@ -84,11 +83,17 @@ class TP_MLP:
     __call__ = forward
 
     def torch_fwd(self, x: torch.Tensor) -> torch.Tensor:
-        """Eager golden reference (replicated x [M, hidden])."""
-        h = x.float() @ self.w_gate_up.float().t()
-        act = F.silu(h[:, :self.inter_shard]) * h[:, self.inter_shard:]
-        partial = act @ self.w_down.float().t()
-        out = partial.to(self.dtype)
+        """Eager golden reference (replicated x [M, hidden]). bf16 matmul on
+        GPU (no fp32 MFMA on CDNA4), fp32 on CPU."""
+        if x.is_cuda:
+            h = x @ self.w_gate_up.t()
+            act = (F.silu(h[:, :self.inter_shard].float())
+                   * h[:, self.inter_shard:].float()).to(self.dtype)
+            out = act @ self.w_down.t()
+        else:
+            h = x.float() @ self.w_gate_up.float().t()
+            act = F.silu(h[:, :self.inter_shard]) * h[:, self.inter_shard:]
+            out = (act @ self.w_down.float().t()).to(self.dtype)
         if dist.is_initialized() and self.world > 1:
             if out.is_cuda and dist.get_backend() == "gloo":
                 cpu = out.cpu()

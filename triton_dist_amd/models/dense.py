@@ -115,17 +115,22 @@ class DenseLLM:
             assert m % self.world == 0
             ms = m // self.world
             x = x[self.rank * ms:(self.rank + 1) * ms].contiguous()
+        from ..ops.fused import add_rms_norm_op, rms_norm_op
+
+        eps = self.cfg.rms_eps
+        pending = None  # MLP output whose residual add fuses into next norm
         for li, layer in enumerate(self.layers):
             attn, mlp = layer["attn"], layer["mlp"]
-            h = rms_norm(x, layer["ln1"], self.cfg.rms_eps)
+            if pending is None:
+                h = rms_norm_op(x, layer["ln1"], eps)
+            else:
+                x, h = add_rms_norm_op(pending, x, layer["ln1"], eps)
             a = attn.forward(h, kv, li, pos, b, s, prefill) \
                 if mode == self.mode else \
                 attn.torch_fwd(h, kv, li, pos, b, s, prefill)
-            x = x + a
-            h = rms_norm(x, layer["ln2"], self.cfg.rms_eps)
-            mo = mlp.forward(h) if mode == self.mode else mlp.torch_fwd(h)
-            x = x + mo
-        x = rms_norm(x, self.final_norm_w, self.cfg.rms_eps)
+            x, h = add_rms_norm_op(a, x, layer["ln2"], eps)
+            pending = mlp.forward(h) if mode == self.mode else mlp.torch_fwd(h)
+        _, x = add_rms_norm_op(pending, x, self.final_norm_w, eps)
         if sharded:
             x = allgather(x, self._decode_ag_ctx)
         # lm_head on the last position of each sequence (replicated compute)
