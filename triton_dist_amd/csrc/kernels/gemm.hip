@@ -214,7 +214,20 @@ __global__ __launch_bounds__(NTHREADS) void k_gemm_bf16(GemmArgs args) {
                  pid_n * BN);
 }
 
+// gemm256.hip perf tier (256^2/BK128 K-slice ring); these fall back to the
+// 128^2 kernel for shapes that don't tile by 256.
+bool gemm256_ok(int m, int n, int k);
+void launch_gemm256_bf16(const GemmArgs &args, hipStream_t stream);
+void launch_ag_gemm256_consumer_bf16(const AgGemmArgs &args,
+                                     hipStream_t stream);
+void launch_gemm256_rs_producer_bf16(const GemmRsArgs &args,
+                                     hipStream_t stream);
+
 void launch_gemm_bf16(const GemmArgs &args, hipStream_t stream) {
+  if (gemm256_ok(args.m, args.n, args.k)) {
+    launch_gemm256_bf16(args, stream);
+    return;
+  }
   if (args.m % BM || args.n % BN || args.k % BK)
     throw std::runtime_error("gemm_bf16: M%128/N%128/K%64 must be 0");
   int grid = (args.m / BM) * (args.n / BN);
@@ -270,6 +283,10 @@ void launch_ag_gemm_consumer_bf16(const AgGemmArgs &args, hipStream_t stream) {
     throw std::runtime_error("ag_gemm: shape must tile by 128/128/64");
   if (args.m_per_rank % args.chunks_per_rank)
     throw std::runtime_error("ag_gemm: chunks_per_rank must divide m_per_rank");
+  if (gemm256_ok(g.m, g.n, g.k) && args.m_per_rank % 256 == 0) {
+    launch_ag_gemm256_consumer_bf16(args, stream);
+    return;
+  }
   int grid = (g.m / BM) * (g.n / BN);
   hipLaunchKernelGGL(k_ag_gemm_consumer_bf16, dim3(grid), dim3(NTHREADS), 0,
                      stream, args);
@@ -316,6 +333,10 @@ void launch_gemm_rs_producer_bf16(const GemmRsArgs &args, hipStream_t stream) {
   const GemmArgs &g = args.g;
   if (g.m % BM || g.n % BN || g.k % BK || args.m_per_rank % BM)
     throw std::runtime_error("gemm_rs: shape must tile by 128/128/64");
+  if (gemm256_ok(g.m, g.n, g.k) && args.m_per_rank % 256 == 0) {
+    launch_gemm256_rs_producer_bf16(args, stream);
+    return;
+  }
   int grid = (g.m / BM) * (g.n / BN);
   hipLaunchKernelGGL(k_gemm_rs_producer_bf16, dim3(grid), dim3(NTHREADS), 0,
                      stream, args);

@@ -1,0 +1,282 @@
+// 256x256-tile, BK=128, 8-wave bf16 MFMA GEMM with a K-slice ring pipeline
+// — the perf-tier kernel behind gemm / ag_gemm / gemm_rs (v1 128x128 in
+// gemm.hip stays as the small-shape fallback).
+//
+// Design (derived from the CDNA4 guide's verified 256^2 8-phase template,
+// re-architected as a K-slice ring so staging never overwrites live data):
+//   * LDS: A[4 slices][256 rows][32 k] + B[same] bf16 = 128 KiB, single
+//     buffer. Phase p (p = 0..3 per K-tile) computes MFMA K-slice p and —
+//     after the barrier that retires all reads of slice p — stages slice p
+//     of the NEXT K-tile into the same region. The ring gives a full
+//     K-tile of prefetch with no double buffer.
+//   * Counted vmcnt: 4 global_load_lds per thread per phase; steady state
+//     16 in flight, `s_waitcnt vmcnt(12)` retires exactly the slice about
+//     to be read (drain schedule 12/8/4/0 on the last tile).
+//   * LDS swizzle: within a slice row (4 x 16B chunks), physical chunk =
+//     logical ^ ((row>>1)&3) — 64 lanes of a ds_read_b128 hit 8 distinct
+//     bank groups (2-way = free). Applied on the global SOURCE address at
+//     stage time (global_load_lds writes linearly) and on the read address;
+//     same involution both sides.
+//   * s_setprio(1) around the 32-MFMA cluster (T5: pays off once phases
+//     create wave role-split).
+//   * 8 waves 2x4; per-wave output 128x64 = 8x4 fragments of 16x16x32.
+//   * Epilogue reuses the full 128 KiB LDS as the C tile for 16B stores
+//     (required for the RS variant's remote xGMI stores).
+//
+// Fused-comm variants carry the same semantics as gemm.hip's v1 kernels
+// (reference behavior: Triton-distributed kernels/amd/allgather_gemm.py
+// :552-660, gemm_reduce_scatter.py:128-227 — capability only).
+#include <stdexcept>
+
+#include "td/api.hpp"
+
+namespace td {
+
+using bf16 = __bf16;
+typedef __attribute__((ext_vector_type(8))) bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+namespace g256 {
+
+constexpr int BM = 256, BN = 256, BK = 128;
+constexpr int NTH = 512;             // 8 waves
+constexpr int SLICES = 4;            // K-slices of 32 per K-tile
+constexpr int SLICE_K = 32;
+constexpr int CH_ROW = SLICE_K / 8;  // 4 x 16B chunks per row per slice
+// LDS elems per slice per matrix: 256*32
+constexpr int SLICE_ELEMS = BM * SLICE_K;
+
+TD_DEV f32x4 mfma16(bf16x8 a, bf16x8 b, f32x4 c) {
+  return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+}
+
+TD_DEV int swz(int row, int j) { return j ^ ((row >> 1) & 3); }
+
+// Stage one K-slice (A and B) of tile kt into the slice-p LDS region.
+// 2 loads per thread per matrix. Global src is pre-swizzled.
+TD_DEV void stage_slice(const bf16 *ga, const bf16 *gb, int lda, int ldb,
+                        int k0, bf16 *lds_a, bf16 *lds_b, int p) {
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+#pragma unroll
+  for (int it = 0; it < 2; ++it) {
+    int q = it * NTH + tid;          // chunk id 0..1023 within the slice
+    int row = q >> 2;                // 4 chunks per row
+    int jp = q & 3;                  // physical chunk in row
+    int jg = swz(row, jp);           // source (logical) chunk
+    const bf16 *sa = ga + (size_t)row * lda + k0 + jg * 8;
+    const bf16 *sb = gb + (size_t)row * ldb + k0 + jg * 8;
+    int wave_chunk0 = it * NTH + wave * 64;  // wave-uniform LDS base
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int *)sa,
+        (__attribute__((address_space(3))) unsigned int *)(
+            lds_a + p * SLICE_ELEMS + wave_chunk0 * 8),
+        16, 0, 0);
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int *)sb,
+        (__attribute__((address_space(3))) unsigned int *)(
+            lds_b + p * SLICE_ELEMS + wave_chunk0 * 8),
+        16, 0, 0);
+  }
+}
+
+struct WaveCtx {
+  int lane, wr, wc;
+};
+
+TD_DEV WaveCtx wave_ctx() {
+  WaveCtx w;
+  int tid = threadIdx.x;
+  w.lane = tid & 63;
+  int wave = tid >> 6;
+  w.wr = wave >> 2;  // 0..1  (M halves of 128)
+  w.wc = wave & 3;   // 0..3  (N quarters of 64)
+  return w;
+}
+
+// The full pipelined K loop over `ktiles` tiles of BK.
+TD_DEV void kloop(const bf16 *ga, const bf16 *gb, int lda, int ldb,
+                  int ktiles, bf16 *lds_a, bf16 *lds_b, const WaveCtx &w,
+                  f32x4 acc[8][4]) {
+  // prologue: stage all 4 slices of tile 0
+#pragma unroll
+  for (int p = 0; p < SLICES; ++p)
+    stage_slice(ga, gb, lda, ldb, p * SLICE_K, lds_a, lds_b, p);
+
+  for (int t = 0; t < ktiles; ++t) {
+    const bool has_next = (t + 1) < ktiles;
+    const int knext = (t + 1) * BK;
+#pragma unroll
+    for (int p = 0; p < SLICES; ++p) {
+      // retire the loads for slice p of tile t
+      if (has_next) {
+        asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
+      } else {
+        if (p == 0) asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
+        if (p == 1) asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+        if (p == 2) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        if (p == 3) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      }
+      __builtin_amdgcn_s_barrier();  // slice p landed for ALL waves
+      __builtin_amdgcn_sched_barrier(0);
+      // read fragments now; the barrier AFTER the reads frees the region
+      const int jn = w.lane >> 4;
+      bf16x8 af[8], bf[4];
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        int row = w.wr * 128 + i * 16 + (w.lane & 15);
+        af[i] = *(const bf16x8 *)(lds_a + p * SLICE_ELEMS + row * SLICE_K +
+                                  swz(row, jn) * 8);
+      }
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int row = w.wc * 64 + j * 16 + (w.lane & 15);
+        bf[j] = *(const bf16x8 *)(lds_b + p * SLICE_ELEMS + row * SLICE_K +
+                                  swz(row, jn) * 8);
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_sched_barrier(0);
+      __builtin_amdgcn_s_barrier();  // all waves done READING slice p
+      __builtin_amdgcn_sched_barrier(0);
+      if (has_next)
+        stage_slice(ga + knext, gb + knext, lda, ldb, p * SLICE_K, lds_a,
+                    lds_b, p);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = mfma16(af[i], bf[j], acc[i][j]);
+      __builtin_amdgcn_s_setprio(0);
+    }
+  }
+}
+
+// Epilogue: accumulators -> LDS C tile [256][256] bf16 -> 16B stores.
+TD_DEV void epilogue(f32x4 acc[8][4], const WaveCtx &w, bf16 *lds_c,
+                     bf16 *dst, int ldc) {
+  __syncthreads();  // staging LDS is dead; reuse as C tile
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = w.wr * 128 + i * 16 + ((w.lane >> 4) * 4 + r);
+        int col = w.wc * 64 + j * 16 + (w.lane & 15);
+        lds_c[row * BN + col] = (bf16)acc[i][j][r];
+      }
+  __syncthreads();
+  const int tid = threadIdx.x;
+#pragma unroll
+  for (int it = 0; it < 16; ++it) {
+    int idx = it * NTH + tid;       // 8192 chunks of 16B
+    int row = idx >> 5;             // 32 chunks per row
+    int cc = idx & 31;
+    *(ulonglong2 *)(dst + (size_t)row * ldc + cc * 8) =
+        *(const ulonglong2 *)(lds_c + row * BN + cc * 8);
+  }
+}
+
+}  // namespace g256
+
+// ---------------------------------------------------------------------------
+// Kernels
+// ---------------------------------------------------------------------------
+using namespace g256;
+
+__global__ __launch_bounds__(NTH, 2) void k_gemm256_bf16(GemmArgs args) {
+  __shared__ bf16 lds_a[SLICES * SLICE_ELEMS];
+  __shared__ bf16 lds_b[SLICES * SLICE_ELEMS];
+  const int tiles_n = args.n / BN;
+  const int tiles_m = args.m / BM;
+  int wgid = xcd_remap(blockIdx.x, tiles_m * tiles_n);
+  int pid_m = wgid / tiles_n, pid_n = wgid % tiles_n;
+  WaveCtx w = wave_ctx();
+  f32x4 acc[8][4] = {};
+  const bf16 *ga = (const bf16 *)args.a + (size_t)pid_m * BM * args.lda;
+  const bf16 *gb = (const bf16 *)args.b + (size_t)pid_n * BN * args.ldb;
+  kloop(ga, gb, args.lda, args.ldb, args.k / BK, lds_a, lds_b, w, acc);
+  bf16 *dst = (bf16 *)args.c + (size_t)pid_m * BM * args.ldc + pid_n * BN;
+  epilogue(acc, w, lds_a, dst, args.ldc);
+}
+
+__global__ __launch_bounds__(NTH, 2) void k_ag_gemm256_consumer_bf16(
+    AgGemmArgs args) {
+  __shared__ bf16 lds_a[SLICES * SLICE_ELEMS];
+  __shared__ bf16 lds_b[SLICES * SLICE_ELEMS];
+  GemmArgs &g = args.g;
+  const int tiles_n = g.n / BN;
+  const int tiles_m = g.m / BM;
+  const int tiles_per_rank = args.m_per_rank / BM;
+  int wgid = xcd_remap(blockIdx.x, tiles_m * tiles_n);
+  int pid_m = wgid / tiles_n, pid_n = wgid % tiles_n;
+  pid_m = (pid_m + args.rank * tiles_per_rank) % tiles_m;
+  int rows_per_chunk = args.m_per_rank / args.chunks_per_rank;
+  int c_lo = (pid_m * BM) / rows_per_chunk;
+  int c_hi = (pid_m * BM + BM - 1) / rows_per_chunk;
+  if (threadIdx.x < 64) {
+    for (int c = c_lo + (int)threadIdx.x; c <= c_hi; c += 64)
+      wait_ge_one<Scope::Sys>(args.flags + c, args.expect);
+  }
+  __syncthreads();
+  WaveCtx w = wave_ctx();
+  f32x4 acc[8][4] = {};
+  const bf16 *ga = (const bf16 *)g.a + (size_t)pid_m * BM * g.lda;
+  const bf16 *gb = (const bf16 *)g.b + (size_t)pid_n * BN * g.ldb;
+  kloop(ga, gb, g.lda, g.ldb, g.k / BK, lds_a, lds_b, w, acc);
+  bf16 *dst = (bf16 *)g.c + (size_t)pid_m * BM * g.ldc + pid_n * BN;
+  epilogue(acc, w, lds_a, dst, g.ldc);
+}
+
+__global__ __launch_bounds__(NTH, 2) void k_gemm256_rs_producer_bf16(
+    GemmRsArgs args) {
+  __shared__ bf16 lds_a[SLICES * SLICE_ELEMS];
+  __shared__ bf16 lds_b[SLICES * SLICE_ELEMS];
+  GemmArgs &g = args.g;
+  const int tiles_n = g.n / BN;
+  const int tiles_m = g.m / BM;
+  const int tiles_per_rank = args.m_per_rank / BM;
+  int wgid = xcd_remap(blockIdx.x, tiles_m * tiles_n);
+  int pid_m = wgid / tiles_n, pid_n = wgid % tiles_n;
+  pid_m = (pid_m + (args.rank + 1) * tiles_per_rank) % tiles_m;
+  WaveCtx w = wave_ctx();
+  f32x4 acc[8][4] = {};
+  const bf16 *ga = (const bf16 *)g.a + (size_t)pid_m * BM * g.lda;
+  const bf16 *gb = (const bf16 *)g.b + (size_t)pid_n * BN * g.ldb;
+  kloop(ga, gb, g.lda, g.ldb, g.k / BK, lds_a, lds_b, w, acc);
+  int owner = (pid_m * BM) / args.m_per_rank;
+  int local_row0 = pid_m * BM - owner * args.m_per_rank;
+  bf16 *seg = (bf16 *)((char *)args.pt.bases[owner] + args.scatter_off) +
+              ((size_t)args.rank * args.m_per_rank + local_row0) * g.n +
+              pid_n * BN;
+  epilogue(acc, w, lds_a, seg, g.n);
+}
+
+// ---------------------------------------------------------------------------
+// Dispatch helpers (called from gemm.hip launchers)
+// ---------------------------------------------------------------------------
+bool gemm256_ok(int m, int n, int k) {
+  return m % BM == 0 && n % BN == 0 && k % BK == 0;
+}
+
+void launch_gemm256_bf16(const GemmArgs &args, hipStream_t stream) {
+  int grid = (args.m / BM) * (args.n / BN);
+  hipLaunchKernelGGL(k_gemm256_bf16, dim3(grid), dim3(NTH), 0, stream, args);
+}
+
+void launch_ag_gemm256_consumer_bf16(const AgGemmArgs &args,
+                                     hipStream_t stream) {
+  int grid = (args.g.m / BM) * (args.g.n / BN);
+  hipLaunchKernelGGL(k_ag_gemm256_consumer_bf16, dim3(grid), dim3(NTH), 0,
+                     stream, args);
+}
+
+void launch_gemm256_rs_producer_bf16(const GemmRsArgs &args,
+                                     hipStream_t stream) {
+  int grid = (args.g.m / BM) * (args.g.n / BN);
+  hipLaunchKernelGGL(k_gemm256_rs_producer_bf16, dim3(grid), dim3(NTH), 0,
+                     stream, args);
+}
+
+}  // namespace td
