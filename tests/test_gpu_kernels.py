@@ -210,3 +210,19 @@ def test_qkv_prologue_and_flash_decode():
     vs = vc[:, :seq + 1].transpose(1, 2).float()
     ref = F.scaled_dot_product_attention(qs, ks, vs, enable_gqa=True)
     assert_allclose(out.view(b, qh, 1, d), ref, atol=4e-2, rtol=4e-2)
+
+
+@pytest.mark.parametrize("m,n,k", [(512, 1280, 5120), (512, 6400, 5120),
+                                   (256, 5120, 1024), (128, 128, 4096)])
+def test_gemm_splitk(m, n, k):
+    from triton_dist_amd.ops import gemm
+    from triton_dist_amd.ops.gemm import choose_splits
+    from triton_dist_amd.utils.testing import assert_allclose, bf16_gemm_tol
+
+    torch.manual_seed(m + n + k)
+    a = torch.randn(m, k, device="cuda").to(torch.bfloat16) / 8
+    w = torch.randn(n, k, device="cuda").to(torch.bfloat16) / 8
+    assert choose_splits(m, n, k) > 1  # these shapes must take the splitk path
+    c = gemm(a, w)
+    ref = a.float() @ w.float().t()
+    assert_allclose(c, ref, **bf16_gemm_tol(k))

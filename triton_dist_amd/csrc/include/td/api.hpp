@@ -74,6 +74,14 @@ void launch_gemm_rs_producer_bf16(const GemmRsArgs &args, hipStream_t stream);
 void launch_rs_reduce_bf16(const void *segments, void *out, int world,
                            int rank, int m_per_rank, int n, hipStream_t stream);
 
+// kernels/gemm_splitk.hip ---------------------------------------------------
+void launch_gemm_splitk_bf16(const GemmArgs &g, float *ws, int splits,
+                             hipStream_t stream);
+void launch_ag_gemm_consumer_splitk_bf16(const AgGemmArgs &a, float *ws,
+                                         int splits, hipStream_t stream);
+void launch_gemm_rs_producer_splitk_bf16(const GemmRsArgs &a, float *ws,
+                                         int splits, hipStream_t stream);
+
 // kernels/elementwise.hip ----------------------------------------------------
 void launch_rmsnorm(const void *x, const void *w, void *out, int rows,
                     int cols, float eps, hipStream_t stream);

@@ -224,7 +224,7 @@ void launch_gemm256_rs_producer_bf16(const GemmRsArgs &args,
                                      hipStream_t stream);
 
 void launch_gemm_bf16(const GemmArgs &args, hipStream_t stream) {
-  if (gemm256_ok(args.m, args.n, args.k)) {
+  if (gemm256_ok(args.m, args.n, args.k) && !args.bias) {
     launch_gemm256_bf16(args, stream);
     return;
   }

@@ -320,6 +320,54 @@ static void rs_reduce_bf16(uintptr_t segments, uintptr_t out, int world,
   TD_CHECK_HIP(hipGetLastError());
 }
 
+static void gemm_splitk_bf16(uintptr_t a, uintptr_t b, uintptr_t c,
+                             uintptr_t bias, uintptr_t ws, int m, int n,
+                             int k, int splits, uintptr_t stream) {
+  GemmArgs args{(void *)a, (void *)b, (void *)c, (void *)bias,
+                m, n, k, k, k, n};
+  launch_gemm_splitk_bf16(args, (float *)ws, splits, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
+static void ag_gemm_consumer_splitk_bf16(uintptr_t a, uintptr_t b,
+                                         uintptr_t c, uintptr_t ws, int m,
+                                         int n, int k, uintptr_t flags,
+                                         int chunks_per_rank, int m_per_rank,
+                                         int world, int rank, int expect,
+                                         int splits, uintptr_t stream) {
+  AgGemmArgs args;
+  args.g = GemmArgs{(void *)a, (void *)b, (void *)c, nullptr,
+                    m, n, k, k, k, n};
+  args.flags = (const int *)flags;
+  args.chunks_per_rank = chunks_per_rank;
+  args.m_per_rank = m_per_rank;
+  args.world = world;
+  args.rank = rank;
+  args.expect = expect;
+  launch_ag_gemm_consumer_splitk_bf16(args, (float *)ws, splits,
+                                      as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
+static void gemm_rs_producer_splitk_bf16(uintptr_t a, uintptr_t b,
+                                         uintptr_t ws, int m, int n, int k,
+                                         size_t scatter_off, int m_per_rank,
+                                         int world, int rank, int splits,
+                                         uintptr_t stream) {
+  check_active();
+  GemmRsArgs args;
+  args.g = GemmArgs{(void *)a, (void *)b, nullptr, nullptr,
+                    m, n, k, k, k, n};
+  args.pt = g_heap.pt;
+  args.scatter_off = scatter_off;
+  args.m_per_rank = m_per_rank;
+  args.world = world;
+  args.rank = rank;
+  launch_gemm_rs_producer_splitk_bf16(args, (float *)ws, splits,
+                                      as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
 static void rmsnorm(uintptr_t x, uintptr_t w, uintptr_t out, int rows,
                     int cols, float eps, uintptr_t stream) {
   launch_rmsnorm((void *)x, (void *)w, (void *)out, rows, cols, eps,
@@ -392,6 +440,9 @@ PYBIND11_MODULE(_C, m) {
   m.def("ag_gemm_consumer_bf16", &ag_gemm_consumer_bf16);
   m.def("gemm_rs_producer_bf16", &gemm_rs_producer_bf16);
   m.def("rs_reduce_bf16", &rs_reduce_bf16);
+  m.def("gemm_splitk_bf16", &gemm_splitk_bf16);
+  m.def("ag_gemm_consumer_splitk_bf16", &ag_gemm_consumer_splitk_bf16);
+  m.def("gemm_rs_producer_splitk_bf16", &gemm_rs_producer_splitk_bf16);
   m.def("rmsnorm", &rmsnorm);
   m.def("add_rmsnorm", &add_rmsnorm);
   m.def("swiglu", &swiglu);

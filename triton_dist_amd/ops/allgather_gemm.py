@@ -142,10 +142,20 @@ def ag_gemm(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
     m_total = world * m
     if out is None:
         out = torch.empty(m_total, n, dtype=torch.bfloat16, device=a.device)
-    _C.ag_gemm_consumer_bf16(
-        ctx.ws.ptr(), w.data_ptr(), out.data_ptr(), m_total, n, k,
-        ctx.flags.ptr(), chunks, m, world, rank, 1,
-        compute.cuda_stream)
+    from .gemm import choose_splits, splitk_ws
+
+    splits = choose_splits(m_total, n, k)
+    if splits > 1:
+        ws = splitk_ws(m_total, n, splits, a.device)
+        _C.ag_gemm_consumer_splitk_bf16(
+            ctx.ws.ptr(), w.data_ptr(), out.data_ptr(), ws.data_ptr(),
+            m_total, n, k, ctx.flags.ptr(), chunks, m, world, rank, 1,
+            splits, compute.cuda_stream)
+    else:
+        _C.ag_gemm_consumer_bf16(
+            ctx.ws.ptr(), w.data_ptr(), out.data_ptr(), m_total, n, k,
+            ctx.flags.ptr(), chunks, m, world, rank, 1,
+            compute.cuda_stream)
     # join comm streams back into the compute stream (after the consumer
     # launch: no serialization, but graph capture requires joined forks)
     for s in range(min(ns, max(world - 1, 1))):

@@ -24,6 +24,22 @@ def gemm_supported(m: int, n: int, k: int) -> bool:
     return m % 128 == 0 and n % 128 == 0 and k % 64 == 0
 
 
+def choose_splits(m: int, n: int, k: int) -> int:
+    """Split-K factor for occupancy-starved shapes: the 128-tile grid must
+    reach ~200 workgroups (256 CUs) before split-K stops paying."""
+    grid = (m // 128) * (n // 128)
+    if grid >= 208:
+        return 1
+    for s in (8, 4, 2):
+        if (k // 64) % s == 0 and k // s >= 512 and grid * s <= 2048:
+            return s
+    return 1
+
+
+def splitk_ws(m: int, n: int, splits: int, device) -> torch.Tensor:
+    return torch.empty(splits, m, n, dtype=torch.float32, device=device)
+
+
 def gemm(a: torch.Tensor, w: torch.Tensor, bias: torch.Tensor | None = None,
          out: torch.Tensor | None = None) -> torch.Tensor:
     """C[M,N] = A[M,K] @ W[N,K]^T + bias."""
@@ -37,10 +53,19 @@ def gemm(a: torch.Tensor, w: torch.Tensor, bias: torch.Tensor | None = None,
             f"gemm v1 needs M%128==0,N%128==0,K%64==0, got {(m, n, k)}"
         if out is None:
             out = torch.empty(m, n, dtype=torch.bfloat16, device=a.device)
-        _native().gemm_bf16(
-            a.data_ptr(), w.data_ptr(), out.data_ptr(),
-            bias.data_ptr() if bias is not None else 0,
-            m, n, k, torch.cuda.current_stream().cuda_stream)
+        stream = torch.cuda.current_stream().cuda_stream
+        splits = choose_splits(m, n, k)
+        if splits > 1:
+            ws = splitk_ws(m, n, splits, a.device)
+            _native().gemm_splitk_bf16(
+                a.data_ptr(), w.data_ptr(), out.data_ptr(),
+                bias.data_ptr() if bias is not None else 0,
+                ws.data_ptr(), m, n, k, splits, stream)
+        else:
+            _native().gemm_bf16(
+                a.data_ptr(), w.data_ptr(), out.data_ptr(),
+                bias.data_ptr() if bias is not None else 0,
+                m, n, k, stream)
         return out
     # CPU reference path
     c = a.float() @ w.float().t()

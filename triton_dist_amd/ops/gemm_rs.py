@@ -81,9 +81,19 @@ def gemm_rs(a: torch.Tensor, w: torch.Tensor, ctx: GemmRSContext,
     compute = torch.cuda.current_stream()
     # entry barrier: previous call's reduce has consumed the scatter bufs
     heap.barrier_all_on_stream(compute)
-    _C.gemm_rs_producer_bf16(a.data_ptr(), w.data_ptr(), m, n, k,
-                             ctx.scatter.offset, m_per_rank, world, rank,
-                             compute.cuda_stream)
+    from .gemm import choose_splits, splitk_ws
+
+    splits = choose_splits(m, n, k)
+    if splits > 1:
+        ws = splitk_ws(m, n, splits, a.device)
+        _C.gemm_rs_producer_splitk_bf16(
+            a.data_ptr(), w.data_ptr(), ws.data_ptr(), m, n, k,
+            ctx.scatter.offset, m_per_rank, world, rank, splits,
+            compute.cuda_stream)
+    else:
+        _C.gemm_rs_producer_bf16(a.data_ptr(), w.data_ptr(), m, n, k,
+                                 ctx.scatter.offset, m_per_rank, world, rank,
+                                 compute.cuda_stream)
     heap.barrier_all_on_stream(compute)
     if out is None:
         out = torch.empty(m_per_rank, n, dtype=torch.bfloat16,
