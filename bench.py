@@ -28,7 +28,7 @@ def parse_args():
     p.add_argument("--batch-per-gpu", type=int, default=512)
     p.add_argument("--ctx", type=int, default=128)
     p.add_argument("--mode", default="ag_rs",
-                   choices=["ag_rs", "allreduce", "torch"])
+                   choices=["ag_rs", "gemm_ar", "allreduce", "torch"])
     p.add_argument("--no-graph", action="store_true")
     p.add_argument("--seed", type=int, default=1234)
     return p.parse_args()
@@ -57,7 +57,7 @@ def main():
     cfg = get_config(args.model, tp_mode=args.mode, max_length=max_len + 64)
     model = DenseLLM(cfg, device=device)
     model.init_weights(seed=args.seed)
-    if args.mode == "ag_rs":
+    if args.mode in ("ag_rs", "gemm_ar"):
         model.init_dist_ctx(max_m_total=batch)
 
     eng = Engine(model, batch=batch, max_len=max_len,

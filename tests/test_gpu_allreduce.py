@@ -1,0 +1,61 @@
+"""GPU allreduce tests: 2 ranks sharing a GPU over hipIpc; one-shot and
+two-shot vs the torch.distributed golden."""
+import pytest
+import torch
+
+from tests.conftest import run_distributed
+
+pytestmark = pytest.mark.gpu
+
+
+def _body_ar_gpu(rank, world):
+    from triton_dist_amd.ops import (all_reduce, all_reduce_ref,
+                                     create_allreduce_context)
+    from triton_dist_amd.utils import assert_allclose
+
+    ctx = create_allreduce_context(max_elems=8 << 20)
+    torch.manual_seed(5 + rank)
+    for method, shape in (("one_shot", (512, 5120)),
+                          ("two_shot", (1024, 5120)),
+                          ("one_shot", (8, 1024))):
+        x = (torch.randn(shape, device="cuda") / 4).to(torch.bfloat16)
+        for _ in range(2):
+            out = all_reduce(x, ctx, method=method)
+            torch.cuda.synchronize()
+            ref = all_reduce_ref(x)
+            assert_allclose(out, ref, atol=8e-2, rtol=5e-2,
+                            msg=f"{method} {shape}")
+
+
+def test_allreduce_gpu_2rank():
+    run_distributed(_body_ar_gpu, world_size=2)
+
+
+def _body_model_gemm_ar_gpu(rank, world):
+    from triton_dist_amd.models import DenseLLM, KVCache, get_config
+    from triton_dist_amd.utils import assert_allclose
+
+    cfg = get_config("tiny-gpu", tp_mode="gemm_ar", max_length=128)
+    model = DenseLLM(cfg, device="cuda")
+    model.init_weights(seed=3)
+    b = 128
+    model.init_dist_ctx(max_m_total=b)
+    kvh = cfg.n_kv_heads // world
+    kv1 = KVCache(cfg.n_layers, b, 64, kvh, cfg.head_dim, device="cuda")
+    kv2 = KVCache(cfg.n_layers, b, 64, kvh, cfg.head_dim, device="cuda")
+    tokens = torch.randint(0, cfg.vocab, (b, 4), device="cuda",
+                           generator=torch.Generator("cuda").manual_seed(1))
+    first1 = model.prefill(tokens, kv1)
+    first2 = model.prefill(tokens, kv2)
+    pos = kv1.offset.reshape(1, 1).expand(b, 1)
+    logits_dist = model.step(first1.view(b, 1), kv1, pos, prefill=False)
+    torch.cuda.synchronize()
+    pos2 = kv2.offset.reshape(1, 1).expand(b, 1)
+    logits_ref = model.step(first2.view(b, 1), kv2, pos2, prefill=False,
+                            mode="torch")
+    torch.cuda.synchronize()
+    assert_allclose(logits_dist, logits_ref, atol=1e-1, rtol=5e-2)
+
+
+def test_model_gemm_ar_gpu_2rank():
+    run_distributed(_body_model_gemm_ar_gpu, world_size=2)

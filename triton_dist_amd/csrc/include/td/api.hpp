@@ -82,6 +82,16 @@ void launch_ag_gemm_consumer_splitk_bf16(const AgGemmArgs &a, float *ws,
 void launch_gemm_rs_producer_splitk_bf16(const GemmRsArgs &a, float *ws,
                                          int splits, hipStream_t stream);
 
+// kernels/allreduce.hip -----------------------------------------------------
+void launch_allreduce_oneshot(const PeerTable &pt, const void *x, void *out,
+                              size_t inbox_off, size_t flags_off,
+                              size_t elems, int chunks, int straggler_rank,
+                              unsigned straggler_cycles, hipStream_t stream);
+void launch_allreduce_twoshot(const PeerTable &pt, const void *x, void *out,
+                              size_t inbox_off, size_t outbox_off,
+                              size_t flags_in_off, size_t flags_out_off,
+                              size_t elems, int chunks, hipStream_t stream);
+
 // kernels/elementwise.hip ----------------------------------------------------
 void launch_rmsnorm(const void *x, const void *w, void *out, int rows,
                     int cols, float eps, hipStream_t stream);

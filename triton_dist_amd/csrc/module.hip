@@ -368,6 +368,28 @@ static void gemm_rs_producer_splitk_bf16(uintptr_t a, uintptr_t b,
   TD_CHECK_HIP(hipGetLastError());
 }
 
+static void allreduce_oneshot(uintptr_t x, uintptr_t out, size_t inbox_off,
+                              size_t flags_off, size_t elems, int chunks,
+                              int straggler_rank, unsigned straggler_cycles,
+                              uintptr_t stream) {
+  check_active();
+  launch_allreduce_oneshot(g_heap.pt, (void *)x, (void *)out, inbox_off,
+                           flags_off, elems, chunks, straggler_rank,
+                           straggler_cycles, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
+static void allreduce_twoshot(uintptr_t x, uintptr_t out, size_t inbox_off,
+                              size_t outbox_off, size_t flags_in_off,
+                              size_t flags_out_off, size_t elems, int chunks,
+                              uintptr_t stream) {
+  check_active();
+  launch_allreduce_twoshot(g_heap.pt, (void *)x, (void *)out, inbox_off,
+                           outbox_off, flags_in_off, flags_out_off, elems,
+                           chunks, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
 static void rmsnorm(uintptr_t x, uintptr_t w, uintptr_t out, int rows,
                     int cols, float eps, uintptr_t stream) {
   launch_rmsnorm((void *)x, (void *)w, (void *)out, rows, cols, eps,
@@ -443,6 +465,8 @@ PYBIND11_MODULE(_C, m) {
   m.def("gemm_splitk_bf16", &gemm_splitk_bf16);
   m.def("ag_gemm_consumer_splitk_bf16", &ag_gemm_consumer_splitk_bf16);
   m.def("gemm_rs_producer_splitk_bf16", &gemm_rs_producer_splitk_bf16);
+  m.def("allreduce_oneshot", &allreduce_oneshot);
+  m.def("allreduce_twoshot", &allreduce_twoshot);
   m.def("rmsnorm", &rmsnorm);
   m.def("add_rmsnorm", &add_rmsnorm);
   m.def("swiglu", &swiglu);

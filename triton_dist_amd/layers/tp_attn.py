@@ -19,8 +19,8 @@ import torch
 import torch.distributed as dist
 import torch.nn.functional as F
 
-from ..ops import (ag_gemm, create_ag_gemm_context, create_gemm_rs_context,
-                   gemm, gemm_rs)
+from ..ops import (ag_gemm, create_ag_gemm_context, create_allreduce_context,
+                   create_gemm_rs_context, gemm, gemm_allreduce, gemm_rs)
 from ..runtime.symm_mem import SymmHeap, get_heap
 from .norm import Rotary, rms_norm
 
@@ -49,6 +49,7 @@ class TP_Attn:
         self.k_norm_w = torch.ones(head_dim, device=device, dtype=dtype)
         self.ag_ctx = None
         self.rs_ctx = None
+        self.ar_ctx = None
 
     def init_weights(self, std=0.02, seed: Optional[int] = None):
         g = None
@@ -60,6 +61,12 @@ class TP_Attn:
             w.copy_(tmp.to(self.dtype))
 
     def init_ctx(self, max_m_total: int, ag_ctx=None, rs_ctx=None):
+        if self.mode == "gemm_ar":
+            if ag_ctx is None:
+                ag_ctx = create_allreduce_context(max_m_total * self.hidden,
+                                                  heap=self.heap)
+            self.ar_ctx = ag_ctx
+            return ag_ctx, None
         if self.mode != "ag_rs":
             return None, None
         if ag_ctx is None:
@@ -128,6 +135,11 @@ class TP_Attn:
             attn = self._attention(qkv, kv_cache, layer_idx, pos, b, s,
                                    prefill, native=True).to(self.dtype)
             return gemm_rs(attn, self.w_o, self.rs_ctx)     # [M/world, H]
+        if self.mode == "gemm_ar":
+            qkv = gemm(x, self.w_qkv)
+            attn = self._attention(qkv, kv_cache, layer_idx, pos, b, s,
+                                   prefill, native=True).to(self.dtype)
+            return gemm_allreduce(attn, self.w_o, self.ar_ctx)
         if self.mode == "allreduce":
             qkv = gemm(x, self.w_qkv)
             attn = self._attention(qkv, kv_cache, layer_idx, pos, b, s,

@@ -14,8 +14,8 @@ import torch
 import torch.distributed as dist
 import torch.nn.functional as F
 
-from ..ops import (ag_gemm, create_ag_gemm_context, create_gemm_rs_context,
-                   gemm, gemm_rs)
+from ..ops import (ag_gemm, create_ag_gemm_context, create_allreduce_context,
+                   create_gemm_rs_context, gemm, gemm_allreduce, gemm_rs)
 from ..runtime.symm_mem import SymmHeap, get_heap
 
 
@@ -37,6 +37,7 @@ class TP_MLP:
                                   dtype=dtype)
         self.ag_ctx = None
         self.rs_ctx = None
+        self.ar_ctx = None
 
     def init_weights(self, std=0.02, seed: Optional[int] = None):
         g = None
@@ -51,6 +52,12 @@ class TP_MLP:
         """Create (or alias — layer 0 owns, others share, cf. reference
         dense.py:169-208) the symmetric contexts for sequences up to
         max_m_total gathered tokens."""
+        if self.mode == "gemm_ar":
+            if ag_ctx is None:
+                ag_ctx = create_allreduce_context(max_m_total * self.hidden,
+                                                  heap=self.heap)
+            self.ar_ctx = ag_ctx
+            return ag_ctx, None
         if self.mode != "ag_rs":
             return None, None
         assert max_m_total % self.world == 0
@@ -72,6 +79,10 @@ class TP_MLP:
             h = ag_gemm(x, self.w_gate_up, self.ag_ctx)     # [M, 2*I_s]
             act = swiglu_op(h, self.inter_shard)
             return gemm_rs(act, self.w_down, self.rs_ctx)   # [M/world, hidden]
+        if self.mode == "gemm_ar":
+            h = gemm(x, self.w_gate_up)
+            act = swiglu_op(h, self.inter_shard)
+            return gemm_allreduce(act, self.w_down, self.ar_ctx)
         if self.mode == "allreduce":
             h = gemm(x, self.w_gate_up)
             act = swiglu_op(h, self.inter_shard)

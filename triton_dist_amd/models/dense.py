@@ -92,6 +92,12 @@ class DenseLLM:
     def init_dist_ctx(self, max_m_total: int):
         """Create the shared symmetric contexts for the distributed decode
         path (collective — call on all ranks with the same max_m_total)."""
+        if self.mode == "gemm_ar":
+            ar0 = None
+            for layer in self.layers:
+                ar0, _ = layer["attn"].init_ctx(max_m_total, ar0)
+                layer["mlp"].init_ctx(max_m_total, ar0)
+            return
         if self.mode != "ag_rs":
             return
         ag0 = rs0 = None

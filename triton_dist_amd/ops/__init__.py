@@ -12,3 +12,17 @@ from .gemm_rs import (  # noqa: F401
     gemm_rs,
     gemm_rs_ref,
 )
+from .allreduce import (  # noqa: F401
+    AllReduceContext,
+    create_allreduce_context,
+    all_reduce,
+    all_reduce_ref,
+    gemm_allreduce,
+)
+from .fused import (  # noqa: F401
+    rms_norm_op,
+    add_rms_norm_op,
+    swiglu_op,
+    flash_decode_op,
+    qkv_prologue_decode_op,
+)
