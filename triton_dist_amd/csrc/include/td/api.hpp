@@ -92,6 +92,36 @@ void launch_allreduce_twoshot(const PeerTable &pt, const void *x, void *out,
                               size_t flags_in_off, size_t flags_out_off,
                               size_t elems, int chunks, hipStream_t stream);
 
+// kernels/moe.hip ------------------------------------------------------------
+void launch_moe_count(const void *topk_ids, void *counts, void *send_pos,
+                      void *send_to_dst, int total, int e_num, int e_loc,
+                      int world, hipStream_t stream);
+void launch_moe_layout(const void *all_splits, int rank, int world,
+                       int e_num, int e_loc, void *send_base,
+                       void *expert_base, void *expert_rows,
+                       void *recv_from_src, void *recv_total,
+                       hipStream_t stream);
+void launch_moe_dispatch(const PeerTable &pt, const void *x,
+                         const void *topk_ids, const void *send_pos,
+                         const void *send_base, const void *send_to_dst,
+                         size_t recv_x_off, size_t meta_off, size_t flags_off,
+                         unsigned *arrive, int T, int K, int H, int e_loc,
+                         hipStream_t stream);
+void launch_moe_wait_flags(const void *flags, int world, hipStream_t stream);
+void launch_moe_grouped_gemm(const void *xin, const void *weights, void *out,
+                             const void *expert_base, const void *expert_rows,
+                             int e_loc, int cap_tiles_m, int n, int k,
+                             int cap_rows, hipStream_t stream);
+void launch_moe_combine_send(const PeerTable &pt, const void *expert_out,
+                             const void *meta, const void *recv_total,
+                             const void *recv_from_src, size_t combine_off,
+                             size_t cflags_off, unsigned *arrive, int cap,
+                             int H, hipStream_t stream);
+void launch_moe_combine_reduce(const void *combine_buf, const void *topk_w,
+                               const void *topk_ids, void *out,
+                               const void *cflags, int world, int T, int K,
+                               int H, int e_num, hipStream_t stream);
+
 // kernels/elementwise.hip ----------------------------------------------------
 void launch_rmsnorm(const void *x, const void *w, void *out, int rows,
                     int cols, float eps, hipStream_t stream);

@@ -1,0 +1,398 @@
+// Expert-parallel MoE: token dispatch / grouped expert GEMM / combine over
+// the hipIpc symmetric heap (intra-node EP over xGMI).
+//
+// Capability parity (behavior only) with Triton-distributed's EP stack:
+//   kernels/amd/ep_a2a_intra_node.py:56-513 (dispatch/combine + splits AG +
+//   recv-offset precompute), csrc/lib/moe_utils.cu:61-356 (scatter-align),
+//   kernels/amd/ep_all2all_fused.py (fused dispatch+grouped GEMM),
+//   kernels/amd/low_latency_all_to_all.py (per-segment signaling).
+//
+// MI355X-first redesign:
+//   * NO cross-GPU atomics on the hot path (xGMI atomics are slow): slot
+//     assignment is DETERMINISTIC from the all-gathered splits matrix
+//     [world, E] — every rank derives both its own recv layout and each
+//     destination's layout from the same data.
+//   * recv buffer is EXPERT-SORTED by construction ([expert][src][idx]), so
+//     the grouped GEMM consumes it directly — the reference's
+//     scatter-align/ sorting pass is absorbed into the dispatch.
+//   * per-source completion signals (arrive counters + release flags)
+//     instead of full barriers between phases; one entry barrier per call
+//     protects buffer reuse. Everything reads dynamic sizes from device
+//     memory -> hipGraph-capturable with capacity grids.
+#include <stdexcept>
+
+#include "td/api.hpp"
+
+namespace td {
+
+using bf16 = __bf16;
+typedef __attribute__((ext_vector_type(8))) bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+// ---------------------------------------------------------------------------
+// Phase 0: routing histogram + stable slot-within-expert assignment.
+// counts[E] must be zeroed beforehand. send_pos[t*K+k] = index of this copy
+// within expert e's copies FROM THIS RANK (device-scope atomics, local HBM).
+// ---------------------------------------------------------------------------
+__global__ void k_moe_count(const int *__restrict__ topk_ids,
+                            int *__restrict__ counts,
+                            int *__restrict__ send_pos, int total, int e_num) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= total) return;
+  int e = topk_ids[i];
+  if (e < 0 || e >= e_num) {  // dropped token slot
+    send_pos[i] = -1;
+    return;
+  }
+  send_pos[i] = atomic_add<Scope::Gpu>(counts + e, 1);
+}
+
+// per-destination-rank copy totals (for completion signaling)
+__global__ void k_moe_dst_counts(const int *__restrict__ counts,
+                                 int *__restrict__ send_to_dst, int e_loc,
+                                 int world) {
+  int d = threadIdx.x;
+  if (d >= world) return;
+  int s = 0;
+  for (int le = 0; le < e_loc; ++le) s += counts[d * e_loc + le];
+  send_to_dst[d] = s;
+}
+
+// ---------------------------------------------------------------------------
+// Phase 1 support: after the splits matrix [world, E] has been exchanged
+// (SDMA push + flags, host-side), one block derives:
+//   send_base[e]    — my copies' base slot inside e's region ON ITS OWNER
+//   expert_base[le] — local expert le's region base in MY recv buffer
+//   expert_rows[le] — total rows for my local expert le
+//   recv_from_src[s]— rows I receive from rank s (combine-side signaling)
+//   recv_total[0]
+// ---------------------------------------------------------------------------
+__global__ void k_moe_layout(const int *__restrict__ all_splits, int rank,
+                             int world, int e_num, int e_loc,
+                             int *__restrict__ send_base,
+                             int *__restrict__ expert_base,
+                             int *__restrict__ expert_rows,
+                             int *__restrict__ recv_from_src,
+                             int *__restrict__ recv_total) {
+  // single block; e_num <= 1024 assumed
+  if (threadIdx.x == 0) {
+    for (int d = 0; d < world; ++d) {
+      int base = 0;  // running offset inside rank d's recv buffer
+      for (int le = 0; le < e_loc; ++le) {
+        int e = d * e_loc + le;
+        int off = 0;
+        for (int s = 0; s < world; ++s) {
+          if (s == rank) send_base[e] = base + off;
+          off += all_splits[s * e_num + e];
+        }
+        if (d == rank) {
+          expert_base[le] = base;
+          expert_rows[le] = off;
+        }
+        base += off;
+      }
+      if (d == rank) recv_total[0] = base;
+    }
+    for (int s = 0; s < world; ++s) {
+      int r = 0;
+      for (int le = 0; le < e_loc; ++le)
+        r += all_splits[s * e_num + rank * e_loc + le];
+      recv_from_src[s] = r;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Phase 2: dispatch. One block per token-copy; copies the H-row over xGMI
+// into the owner's expert-sorted recv buffer, writes (src, tok_k) meta, and
+// signals the destination when the LAST of my copies to it lands.
+// ---------------------------------------------------------------------------
+__global__ void k_moe_dispatch(PeerTable pt, const bf16 *__restrict__ x,
+                               const int *__restrict__ topk_ids,
+                               const int *__restrict__ send_pos,
+                               const int *__restrict__ send_base,
+                               const int *__restrict__ send_to_dst,
+                               size_t recv_x_off, size_t meta_off,
+                               size_t flags_off, unsigned *arrive, int T,
+                               int K, int H, int e_loc) {
+  const int i = blockIdx.x;  // copy index t*K+k
+  const int e = topk_ids[i];
+  const int pos = send_pos[i];
+  if (e < 0 || pos < 0) return;
+  const int dst = e / e_loc;
+  const int t = i / K;
+  const int slot = send_base[e] + pos;
+  bf16 *rx = (bf16 *)((char *)pt.bases[dst] + recv_x_off) + (size_t)slot * H;
+  const bf16 *src = x + (size_t)t * H;
+  for (int c = threadIdx.x * 8; c < H; c += blockDim.x * 8)
+    *(bf16x8 *)(rx + c) = *(const bf16x8 *)(src + c);
+  if (threadIdx.x == 0) {
+    int *meta = (int *)((char *)pt.bases[dst] + meta_off);
+    meta[slot * 2] = pt.rank;
+    meta[slot * 2 + 1] = i;  // t*K+k
+    __threadfence_block();
+    fence_release_sys();
+    unsigned prev = atomic_add<Scope::Gpu>(arrive + dst, 1u);
+    if ((int)prev == send_to_dst[dst] - 1) {
+      int *fl = (int *)((char *)pt.bases[dst] + flags_off);
+      st_release<Scope::Sys>(fl + pt.rank, 1);
+    }
+  }
+}
+
+// signal destinations that get ZERO copies from me (they still wait on my
+// flag), and likewise for the combine side.
+__global__ void k_moe_signal_empty(PeerTable pt, const int *__restrict__ cnt,
+                                   size_t flags_off) {
+  int d = threadIdx.x;
+  if (d >= pt.world) return;
+  if (cnt[d] == 0) {
+    int *fl = (int *)((char *)pt.bases[d] + flags_off);
+    st_release<Scope::Sys>(fl + pt.rank, 1);
+  }
+}
+
+// wait for all world dispatch flags (prefix kernel before the expert GEMM)
+__global__ void k_moe_wait_flags(const int *flags, int world) {
+  if (threadIdx.x < (unsigned)world)
+    wait_ge_one<Scope::Sys>(flags + threadIdx.x, 1);
+}
+
+// ---------------------------------------------------------------------------
+// Phase 3: grouped expert GEMM (128x128 tile, BK=64). Capacity grid:
+// grid = (e_loc * cap_tiles_m, tiles_n); blocks beyond expert_rows exit.
+// Edge rows within a tile are masked at the C write; A reads may touch
+// neighbor-region rows (garbage in, garbage rows out — never stored).
+// ---------------------------------------------------------------------------
+namespace gg {
+constexpr int BM = 128, BN = 128, BK = 64, NTH = 256;
+}
+
+__global__ __launch_bounds__(gg::NTH) void k_moe_grouped_gemm(
+    const bf16 *__restrict__ xin, const bf16 *__restrict__ weights,
+    bf16 *__restrict__ out, const int *__restrict__ expert_base,
+    const int *__restrict__ expert_rows, int cap_tiles_m, int n, int k,
+    int cap_rows) {
+  const int tiles_n = n / gg::BN;
+  const int e = blockIdx.x / cap_tiles_m;
+  const int tm = blockIdx.x % cap_tiles_m;
+  const int tn = blockIdx.y;
+  const int rows = expert_rows[e];
+  if (tm * gg::BM >= rows) return;
+  const int base = expert_base[e];
+
+  __shared__ bf16 lds_a[gg::BM * gg::BK];
+  __shared__ bf16 lds_b[gg::BN * gg::BK];
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int wr = wave >> 1, wc = wave & 1;
+  f32x4 acc[4][4] = {};
+  // recv_x is allocated with BM rows of tail slack, so edge-tile reads
+  // never fault (garbage rows compute garbage outputs that are masked).
+  (void)cap_rows;
+  const bf16 *ga = xin + (size_t)(base + tm * gg::BM) * k;
+  const bf16 *gb = weights + (size_t)e * n * k + (size_t)tn * gg::BN * k;
+  for (int k0 = 0; k0 < k; k0 += gg::BK) {
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      int idx = it * gg::NTH + tid;
+      int row = idx >> 3;
+      int kc = idx & 7;
+      int wave_chunk0 = it * gg::NTH + wave * 64;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int *)(
+              ga + (size_t)row * k + k0 + kc * 8),
+          (__attribute__((address_space(3))) unsigned int *)(lds_a +
+                                                             wave_chunk0 * 8),
+          16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int *)(
+              gb + (size_t)row * k + k0 + kc * 8),
+          (__attribute__((address_space(3))) unsigned int *)(lds_b +
+                                                             wave_chunk0 * 8),
+          16, 0, 0);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+#pragma unroll
+    for (int ks = 0; ks < gg::BK / 32; ++ks) {
+      bf16x8 af[4], bfr[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int arow = wr * 64 + i * 16 + (lane & 15);
+        int brow = wc * 64 + i * 16 + (lane & 15);
+        int kk = ks * 32 + (lane >> 4) * 8;
+        af[i] = *(const bf16x8 *)(lds_a + arow * gg::BK + kk);
+        bfr[i] = *(const bf16x8 *)(lds_b + brow * gg::BK + kk);
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[i], bfr[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+  // masked direct store (tail rows beyond the expert's count skipped)
+  const int row_lim = rows - tm * gg::BM;
+  const int actual_base = base + tm * gg::BM;
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = wr * 64 + i * 16 + (lane >> 4) * 4 + r;
+        int col = wc * 64 + j * 16 + (lane & 15);
+        if (row < row_lim) {
+          out[((size_t)actual_base + row) * n + (size_t)tn * gg::BN + col] =
+              (bf16)acc[i][j][r];
+        }
+      }
+}
+
+// ---------------------------------------------------------------------------
+// Phase 4: combine send — return expert outputs to their source ranks.
+// Block per recv row; signals each source when all its rows are returned.
+// ---------------------------------------------------------------------------
+__global__ void k_moe_combine_send(PeerTable pt,
+                                   const bf16 *__restrict__ expert_out,
+                                   const int *__restrict__ meta,
+                                   const int *__restrict__ recv_total,
+                                   const int *__restrict__ recv_from_src,
+                                   size_t combine_off, size_t cflags_off,
+                                   unsigned *arrive, int H) {
+  const int r = blockIdx.x;
+  if (r >= recv_total[0]) return;
+  const int src = meta[r * 2];
+  const int tok_k = meta[r * 2 + 1];
+  bf16 *dst = (bf16 *)((char *)pt.bases[src] + combine_off) +
+              (size_t)tok_k * H;
+  const bf16 *row = expert_out + (size_t)r * H;
+  for (int c = threadIdx.x * 8; c < H; c += blockDim.x * 8)
+    *(bf16x8 *)(dst + c) = *(const bf16x8 *)(row + c);
+  if (threadIdx.x == 0) {
+    fence_release_sys();
+    unsigned prev = atomic_add<Scope::Gpu>(arrive + src, 1u);
+    if ((int)prev == recv_from_src[src] - 1) {
+      int *fl = (int *)((char *)pt.bases[src] + cflags_off);
+      st_release<Scope::Sys>(fl + pt.rank, 1);
+    }
+  }
+}
+
+// Phase 5: weighted reduce of the returned copies:
+// out[t] = sum_k topk_w[t,k] * combine_buf[t*K+k]   (dropped copies add 0)
+__global__ void k_moe_combine_reduce(const bf16 *__restrict__ combine_buf,
+                                     const float *__restrict__ topk_w,
+                                     const int *__restrict__ topk_ids,
+                                     bf16 *__restrict__ out,
+                                     const int *cflags, int world, int T,
+                                     int K, int H, int e_num) {
+  if (threadIdx.x < (unsigned)world)
+    wait_ge_one<Scope::Sys>(cflags + threadIdx.x, 1);
+  __syncthreads();
+  const int t = blockIdx.x;
+  if (t >= T) return;
+  for (int c = threadIdx.x * 8; c < H; c += blockDim.x * 8) {
+    float acc[8] = {};
+    for (int k = 0; k < K; ++k) {
+      int e = topk_ids[t * K + k];
+      if (e < 0 || e >= e_num) continue;
+      float w = topk_w[t * K + k];
+      bf16x8 v = *(const bf16x8 *)(combine_buf + ((size_t)t * K + k) * H + c);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) acc[j] += w * (float)v[j];
+    }
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = (bf16)acc[j];
+    *(bf16x8 *)(out + (size_t)t * H + c) = o;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Launchers
+// ---------------------------------------------------------------------------
+void launch_moe_count(const void *topk_ids, void *counts, void *send_pos,
+                      void *send_to_dst, int total, int e_num, int e_loc,
+                      int world, hipStream_t stream) {
+  int blocks = (total + 255) / 256;
+  hipLaunchKernelGGL(k_moe_count, dim3(blocks), dim3(256), 0, stream,
+                     (const int *)topk_ids, (int *)counts, (int *)send_pos,
+                     total, e_num);
+  hipLaunchKernelGGL(k_moe_dst_counts, dim3(1), dim3(kWave), 0, stream,
+                     (const int *)counts, (int *)send_to_dst, e_loc, world);
+}
+
+void launch_moe_layout(const void *all_splits, int rank, int world,
+                       int e_num, int e_loc, void *send_base,
+                       void *expert_base, void *expert_rows,
+                       void *recv_from_src, void *recv_total,
+                       hipStream_t stream) {
+  hipLaunchKernelGGL(k_moe_layout, dim3(1), dim3(1), 0, stream,
+                     (const int *)all_splits, rank, world, e_num, e_loc,
+                     (int *)send_base, (int *)expert_base,
+                     (int *)expert_rows, (int *)recv_from_src,
+                     (int *)recv_total);
+}
+
+void launch_moe_dispatch(const PeerTable &pt, const void *x,
+                         const void *topk_ids, const void *send_pos,
+                         const void *send_base, const void *send_to_dst,
+                         size_t recv_x_off, size_t meta_off, size_t flags_off,
+                         unsigned *arrive, int T, int K, int H, int e_loc,
+                         hipStream_t stream) {
+  if (H % 8) throw std::runtime_error("moe dispatch: H % 8 != 0");
+  hipLaunchKernelGGL(k_moe_dispatch, dim3(T * K), dim3(256), 0, stream, pt,
+                     (const bf16 *)x, (const int *)topk_ids,
+                     (const int *)send_pos, (const int *)send_base,
+                     (const int *)send_to_dst, recv_x_off, meta_off,
+                     flags_off, arrive, T, K, H, e_loc);
+  hipLaunchKernelGGL(k_moe_signal_empty, dim3(1), dim3(kWave), 0, stream, pt,
+                     (const int *)send_to_dst, flags_off);
+}
+
+void launch_moe_wait_flags(const void *flags, int world, hipStream_t stream) {
+  hipLaunchKernelGGL(k_moe_wait_flags, dim3(1), dim3(kWave), 0, stream,
+                     (const int *)flags, world);
+}
+
+void launch_moe_grouped_gemm(const void *xin, const void *weights, void *out,
+                             const void *expert_base, const void *expert_rows,
+                             int e_loc, int cap_tiles_m, int n, int k,
+                             int cap_rows, hipStream_t stream) {
+  if (n % 128 || k % 64)
+    throw std::runtime_error("grouped gemm: N%128/K%64 required");
+  hipLaunchKernelGGL(k_moe_grouped_gemm,
+                     dim3(e_loc * cap_tiles_m, n / 128), dim3(gg::NTH), 0,
+                     stream, (const bf16 *)xin, (const bf16 *)weights,
+                     (bf16 *)out, (const int *)expert_base,
+                     (const int *)expert_rows, cap_tiles_m, n, k, cap_rows);
+}
+
+void launch_moe_combine_send(const PeerTable &pt, const void *expert_out,
+                             const void *meta, const void *recv_total,
+                             const void *recv_from_src, size_t combine_off,
+                             size_t cflags_off, unsigned *arrive, int cap,
+                             int H, hipStream_t stream) {
+  hipLaunchKernelGGL(k_moe_combine_send, dim3(cap), dim3(256), 0, stream, pt,
+                     (const bf16 *)expert_out, (const int *)meta,
+                     (const int *)recv_total, (const int *)recv_from_src,
+                     combine_off, cflags_off, arrive, H);
+  hipLaunchKernelGGL(k_moe_signal_empty, dim3(1), dim3(kWave), 0, stream, pt,
+                     (const int *)recv_from_src, cflags_off);
+}
+
+void launch_moe_combine_reduce(const void *combine_buf, const void *topk_w,
+                               const void *topk_ids, void *out,
+                               const void *cflags, int world, int T, int K,
+                               int H, int e_num, hipStream_t stream) {
+  hipLaunchKernelGGL(k_moe_combine_reduce, dim3(T), dim3(256), 0, stream,
+                     (const bf16 *)combine_buf, (const float *)topk_w,
+                     (const int *)topk_ids, (bf16 *)out, (const int *)cflags,
+                     world, T, K, H, e_num);
+}
+
+}  // namespace td

@@ -390,6 +390,77 @@ static void allreduce_twoshot(uintptr_t x, uintptr_t out, size_t inbox_off,
   TD_CHECK_HIP(hipGetLastError());
 }
 
+static void moe_count(uintptr_t topk_ids, uintptr_t counts,
+                      uintptr_t send_pos, uintptr_t send_to_dst, int total,
+                      int e_num, int e_loc, int world, uintptr_t stream) {
+  launch_moe_count((void *)topk_ids, (void *)counts, (void *)send_pos,
+                   (void *)send_to_dst, total, e_num, e_loc, world,
+                   as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
+static void moe_layout(uintptr_t all_splits, int rank, int world, int e_num,
+                       int e_loc, uintptr_t send_base, uintptr_t expert_base,
+                       uintptr_t expert_rows, uintptr_t recv_from_src,
+                       uintptr_t recv_total, uintptr_t stream) {
+  launch_moe_layout((void *)all_splits, rank, world, e_num, e_loc,
+                    (void *)send_base, (void *)expert_base,
+                    (void *)expert_rows, (void *)recv_from_src,
+                    (void *)recv_total, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
+static void moe_dispatch(uintptr_t x, uintptr_t topk_ids, uintptr_t send_pos,
+                         uintptr_t send_base, uintptr_t send_to_dst,
+                         size_t recv_x_off, size_t meta_off,
+                         size_t flags_off, uintptr_t arrive, int T, int K,
+                         int H, int e_loc, uintptr_t stream) {
+  check_active();
+  launch_moe_dispatch(g_heap.pt, (void *)x, (void *)topk_ids,
+                      (void *)send_pos, (void *)send_base,
+                      (void *)send_to_dst, recv_x_off, meta_off, flags_off,
+                      (unsigned *)arrive, T, K, H, e_loc, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
+static void moe_wait_flags(uintptr_t flags, int world, uintptr_t stream) {
+  launch_moe_wait_flags((void *)flags, world, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
+static void moe_grouped_gemm(uintptr_t xin, uintptr_t weights, uintptr_t out,
+                             uintptr_t expert_base, uintptr_t expert_rows,
+                             int e_loc, int cap_tiles_m, int n, int k,
+                             int cap_rows, uintptr_t stream) {
+  launch_moe_grouped_gemm((void *)xin, (void *)weights, (void *)out,
+                          (void *)expert_base, (void *)expert_rows, e_loc,
+                          cap_tiles_m, n, k, cap_rows, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
+static void moe_combine_send(uintptr_t expert_out, uintptr_t meta,
+                             uintptr_t recv_total, uintptr_t recv_from_src,
+                             size_t combine_off, size_t cflags_off,
+                             uintptr_t arrive, int cap, int H,
+                             uintptr_t stream) {
+  check_active();
+  launch_moe_combine_send(g_heap.pt, (void *)expert_out, (void *)meta,
+                          (void *)recv_total, (void *)recv_from_src,
+                          combine_off, cflags_off, (unsigned *)arrive, cap,
+                          H, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
+static void moe_combine_reduce(uintptr_t combine_buf, uintptr_t topk_w,
+                               uintptr_t topk_ids, uintptr_t out,
+                               uintptr_t cflags, int world, int T, int K,
+                               int H, int e_num, uintptr_t stream) {
+  launch_moe_combine_reduce((void *)combine_buf, (void *)topk_w,
+                            (void *)topk_ids, (void *)out, (void *)cflags,
+                            world, T, K, H, e_num, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
 static void rmsnorm(uintptr_t x, uintptr_t w, uintptr_t out, int rows,
                     int cols, float eps, uintptr_t stream) {
   launch_rmsnorm((void *)x, (void *)w, (void *)out, rows, cols, eps,
@@ -467,6 +538,13 @@ PYBIND11_MODULE(_C, m) {
   m.def("gemm_rs_producer_splitk_bf16", &gemm_rs_producer_splitk_bf16);
   m.def("allreduce_oneshot", &allreduce_oneshot);
   m.def("allreduce_twoshot", &allreduce_twoshot);
+  m.def("moe_count", &moe_count);
+  m.def("moe_layout", &moe_layout);
+  m.def("moe_dispatch", &moe_dispatch);
+  m.def("moe_wait_flags", &moe_wait_flags);
+  m.def("moe_grouped_gemm", &moe_grouped_gemm);
+  m.def("moe_combine_send", &moe_combine_send);
+  m.def("moe_combine_reduce", &moe_combine_reduce);
   m.def("rmsnorm", &rmsnorm);
   m.def("add_rmsnorm", &add_rmsnorm);
   m.def("swiglu", &swiglu);

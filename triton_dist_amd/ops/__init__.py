@@ -26,3 +26,9 @@ from .fused import (  # noqa: F401
     flash_decode_op,
     qkv_prologue_decode_op,
 )
+from .ep_moe import (  # noqa: F401
+    EPContext,
+    create_ep_context,
+    ep_moe_forward,
+    ep_moe_ref,
+)

@@ -1,0 +1,306 @@
+"""Expert-parallel MoE ops: token dispatch -> grouped expert GEMM (SwiGLU)
+-> combine, over the symmetric heap.
+
+Capability parity with the reference's EP stack (Triton-distributed
+layers/amd/ep_a2a_layer.py:208-548 EPAll2AllLayer dispatch/combine,
+kernels/amd/ep_a2a_intra_node.py, function/amd/ep_moe_fused.py fused_ep_moe
+— behavior only; see csrc/kernels/moe.hip for the MI355X-native design:
+deterministic splits-based layout, expert-sorted recv, per-source
+completion signals).
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Optional
+
+import torch
+
+from ..runtime import cpu_shm
+from ..runtime.symm_mem import SymmBuffer, SymmHeap, get_heap
+
+
+@dataclass
+class EPContext:
+    heap: SymmHeap
+    max_tokens: int     # per-rank tokens per call
+    hidden: int
+    n_experts: int
+    topk: int
+    cap: int            # recv rows capacity = world * max_tokens * topk
+    # symmetric buffers
+    all_splits: SymmBuffer      # [world, E] int32
+    splits_flags: SymmBuffer    # [world] int32
+    recv_x: SymmBuffer          # [cap + 128, H] bf16 (tail slack for GEMM)
+    meta: SymmBuffer            # [cap, 2] int32 (src, tok_k)
+    combine_buf: SymmBuffer     # [max_tokens * topk, H] bf16
+    disp_flags: SymmBuffer      # [world] int32
+    comb_flags: SymmBuffer      # [world] int32
+    # local device state
+    local: dict = field(default_factory=dict)
+    epoch: int = 0
+
+    @property
+    def world(self):
+        return self.heap.world
+
+    @property
+    def rank(self):
+        return self.heap.rank
+
+    @property
+    def e_loc(self):
+        return self.n_experts // self.world
+
+
+def create_ep_context(max_tokens: int, hidden: int, n_experts: int,
+                      topk: int, heap: Optional[SymmHeap] = None
+                      ) -> EPContext:
+    heap = heap or get_heap()
+    world = heap.world
+    assert n_experts % world == 0
+    cap = world * max_tokens * topk
+    ctx = EPContext(
+        heap=heap, max_tokens=max_tokens, hidden=hidden,
+        n_experts=n_experts, topk=topk, cap=cap,
+        all_splits=heap.alloc_buffer((world, n_experts), torch.int32),
+        splits_flags=heap.alloc_buffer((world,), torch.int32),
+        recv_x=heap.alloc_buffer((cap + 128, hidden), torch.bfloat16),
+        meta=heap.alloc_buffer((cap, 2), torch.int32),
+        combine_buf=heap.alloc_buffer((max_tokens * topk, hidden),
+                                      torch.bfloat16),
+        disp_flags=heap.alloc_buffer((world,), torch.int32),
+        comb_flags=heap.alloc_buffer((world,), torch.int32),
+    )
+    if heap.backend == "hip":
+        dev = "cuda"
+        e = n_experts
+        ctx.local = dict(
+            counts=torch.zeros(e, dtype=torch.int32, device=dev),
+            send_pos=torch.zeros(max_tokens * topk, dtype=torch.int32,
+                                 device=dev),
+            send_base=torch.zeros(e, dtype=torch.int32, device=dev),
+            send_to_dst=torch.zeros(world, dtype=torch.int32, device=dev),
+            expert_base=torch.zeros(ctx.e_loc, dtype=torch.int32, device=dev),
+            expert_rows=torch.zeros(ctx.e_loc, dtype=torch.int32, device=dev),
+            recv_from_src=torch.zeros(world, dtype=torch.int32, device=dev),
+            recv_total=torch.zeros(1, dtype=torch.int32, device=dev),
+            arrive_d=torch.zeros(world, dtype=torch.int32, device=dev),
+            arrive_c=torch.zeros(world, dtype=torch.int32, device=dev),
+        )
+    return ctx
+
+
+def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
+                   topk_w: torch.Tensor, w_gate_up: torch.Tensor,
+                   w_down: torch.Tensor, ctx: EPContext,
+                   out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Full EP MoE layer for this rank's tokens.
+
+    x: [T, H] bf16; topk_ids: [T, K] int32 (global expert ids, -1 = drop);
+    topk_w: [T, K] fp32; w_gate_up: [E_loc, 2*I, H]; w_down: [E_loc, H, I].
+    Returns [T, H] bf16.
+    """
+    T, H = x.shape
+    K = ctx.topk
+    E, e_loc = ctx.n_experts, ctx.e_loc
+    world, rank = ctx.world, ctx.rank
+    assert T <= ctx.max_tokens and H == ctx.hidden
+    inter = w_down.shape[2]
+    if out is None:
+        out = torch.empty(T, H, dtype=torch.bfloat16, device=x.device)
+
+    if ctx.heap.backend == "cpu":
+        return _ep_moe_cpu(x, topk_ids, topk_w, w_gate_up, w_down, ctx, out)
+
+    _C = ctx.heap._C
+    heap = ctx.heap
+    stream = torch.cuda.current_stream()
+    s = stream.cuda_stream
+    L = ctx.local
+
+    # phase 0: reset + entry barrier (buffer reuse protection; graph-safe)
+    L["counts"].zero_()
+    L["arrive_d"].zero_()
+    L["arrive_c"].zero_()
+    _C.reset_flags(ctx.splits_flags.ptr(), world, 0, s)
+    _C.reset_flags(ctx.disp_flags.ptr(), world, 0, s)
+    _C.reset_flags(ctx.comb_flags.ptr(), world, 0, s)
+    heap.barrier_all_on_stream(stream)
+
+    # phase 1: routing histogram + slot assignment
+    _C.moe_count(topk_ids.data_ptr(), L["counts"].data_ptr(),
+                 L["send_pos"].data_ptr(), L["send_to_dst"].data_ptr(),
+                 T * K, E, e_loc, world, s)
+
+    # phase 2: exchange the splits matrix (SDMA push + flag per peer)
+    my_row_off = ctx.all_splits.offset + rank * E * 4
+    _C.memcpy_async(heap.ptr(rank, my_row_off), L["counts"].data_ptr(),
+                    E * 4, s)
+    _C.reset_flags(ctx.splits_flags.ptr() + rank * 4, 1, 1, s)
+    for i in range(world - 1):
+        peer = (rank + 1 + i) % world
+        _C.memcpy_async(heap.ptr(peer, my_row_off), L["counts"].data_ptr(),
+                        E * 4, s)
+        _C.memcpy_async(ctx.splits_flags.ptr(peer) + rank * 4,
+                        heap.one_src.ptr(), 4, s)
+    _C.wait_eq(ctx.splits_flags.ptr(), world, 1, s)
+
+    # phase 3: derive layouts
+    _C.moe_layout(ctx.all_splits.ptr(), rank, world, E, e_loc,
+                  L["send_base"].data_ptr(), L["expert_base"].data_ptr(),
+                  L["expert_rows"].data_ptr(), L["recv_from_src"].data_ptr(),
+                  L["recv_total"].data_ptr(), s)
+
+    # phase 4: dispatch (xGMI row push + per-dst completion signals)
+    _C.moe_dispatch(x.data_ptr(), topk_ids.data_ptr(),
+                    L["send_pos"].data_ptr(), L["send_base"].data_ptr(),
+                    L["send_to_dst"].data_ptr(), ctx.recv_x.offset,
+                    ctx.meta.offset, ctx.disp_flags.offset,
+                    L["arrive_d"].data_ptr(), T, K, H, e_loc, s)
+    _C.moe_wait_flags(ctx.disp_flags.ptr(), world, s)
+
+    # phase 5: grouped expert FFN
+    cap_tiles = (ctx.cap + 127) // 128
+    # +128-row slack everywhere an edge GEMM tile may over-read
+    expert_h = torch.empty(ctx.cap + 128, 2 * inter, dtype=torch.bfloat16,
+                           device=x.device)
+    _C.moe_grouped_gemm(ctx.recv_x.ptr(), w_gate_up.data_ptr(),
+                        expert_h.data_ptr(), L["expert_base"].data_ptr(),
+                        L["expert_rows"].data_ptr(), e_loc, cap_tiles,
+                        2 * inter, H, ctx.cap, s)
+    act = torch.empty(ctx.cap + 128, inter, dtype=torch.bfloat16,
+                      device=x.device)
+    _C.swiglu(expert_h.data_ptr(), act.data_ptr(), ctx.cap + 128, inter, s)
+    expert_out = torch.empty(ctx.cap + 128, H, dtype=torch.bfloat16,
+                             device=x.device)
+    _C.moe_grouped_gemm(act.data_ptr(), w_down.data_ptr(),
+                        expert_out.data_ptr(), L["expert_base"].data_ptr(),
+                        L["expert_rows"].data_ptr(), e_loc, cap_tiles, H,
+                        inter, ctx.cap, s)
+
+    # phase 6: combine (return rows + weighted reduce)
+    _C.moe_combine_send(expert_out.data_ptr(), ctx.meta.ptr(),
+                        L["recv_total"].data_ptr(),
+                        L["recv_from_src"].data_ptr(),
+                        ctx.combine_buf.offset, ctx.comb_flags.offset,
+                        L["arrive_c"].data_ptr(), ctx.cap, H, s)
+    _C.moe_combine_reduce(ctx.combine_buf.ptr(), topk_w.data_ptr(),
+                          topk_ids.data_ptr(), out.data_ptr(),
+                          ctx.comb_flags.ptr(), world, T, K, H, E, s)
+    return out
+
+
+def _ep_moe_cpu(x, topk_ids, topk_w, w_gate_up, w_down, ctx, out):
+    """CPU mock: same deterministic layout algebra over the shm heap."""
+    T, H = x.shape
+    K, E, e_loc = ctx.topk, ctx.n_experts, ctx.e_loc
+    world, rank = ctx.world, ctx.rank
+    inter = w_down.shape[2]
+    ctx.epoch += 1
+    heap = ctx.heap
+    heap.barrier_all()
+
+    ids = topk_ids.reshape(-1)
+    counts = torch.zeros(E, dtype=torch.int32)
+    send_pos = torch.full((T * K,), -1, dtype=torch.int32)
+    for i in range(T * K):
+        e = int(ids[i])
+        if 0 <= e < E:
+            send_pos[i] = counts[e]
+            counts[e] += 1
+    # splits exchange
+    for p in range(world):
+        ctx.all_splits.peer(p)[rank].copy_(counts)
+        cpu_shm.notify(ctx.splits_flags.peer(p), rank, ctx.epoch)
+    fl = ctx.splits_flags.local()
+    for sidx in range(world):
+        cpu_shm.wait_ge(fl, sidx, ctx.epoch)
+    splits = ctx.all_splits.local().clone()  # [world, E]
+
+    # layouts (same algebra as k_moe_layout)
+    send_base = torch.zeros(E, dtype=torch.int64)
+    expert_base = torch.zeros(e_loc, dtype=torch.int64)
+    expert_rows = torch.zeros(e_loc, dtype=torch.int64)
+    for d in range(world):
+        base = 0
+        for le in range(e_loc):
+            e = d * e_loc + le
+            off = 0
+            for src in range(world):
+                if src == rank:
+                    send_base[e] = base + off
+                off += int(splits[src, e])
+            if d == rank:
+                expert_base[le] = base
+                expert_rows[le] = off
+            base += off
+
+    # dispatch
+    for i in range(T * K):
+        e = int(ids[i])
+        if e < 0 or e >= E or send_pos[i] < 0:
+            continue
+        d = e // e_loc
+        slot = int(send_base[e]) + int(send_pos[i])
+        ctx.recv_x.peer(d)[slot].copy_(x[i // K])
+        ctx.meta.peer(d)[slot, 0] = rank
+        ctx.meta.peer(d)[slot, 1] = i
+    heap.barrier_all()
+
+    # grouped FFN
+    expert_out = torch.zeros(ctx.cap + 128, H, dtype=torch.bfloat16)
+    rx = ctx.recv_x.local()
+    for le in range(e_loc):
+        nb = int(expert_rows[le])
+        if nb == 0:
+            continue
+        b0 = int(expert_base[le])
+        xe = rx[b0:b0 + nb].float()
+        h = xe @ w_gate_up[le].float().t()
+        a = torch.nn.functional.silu(h[:, :inter]) * h[:, inter:]
+        y = a @ w_down[le].float().t()
+        expert_out[b0:b0 + nb] = y.to(torch.bfloat16)
+
+    # combine send
+    meta = ctx.meta.local()
+    total = int(expert_base[-1] + expert_rows[-1]) if e_loc else 0
+    for r in range(total):
+        src = int(meta[r, 0])
+        tok_k = int(meta[r, 1])
+        ctx.combine_buf.peer(src)[tok_k].copy_(expert_out[r])
+    heap.barrier_all()
+
+    # reduce
+    cb = ctx.combine_buf.local()
+    acc = torch.zeros(T, H, dtype=torch.float32)
+    for t in range(T):
+        for k in range(K):
+            e = int(ids[t * K + k])
+            if e < 0 or e >= E:
+                continue
+            acc[t] += float(topk_w[t, k]) * cb[t * K + k].float()
+    out.copy_(acc.to(torch.bfloat16))
+    heap.barrier_all()
+    return out
+
+
+def ep_moe_ref(x: torch.Tensor, topk_ids: torch.Tensor, topk_w: torch.Tensor,
+               full_w_gate_up: torch.Tensor, full_w_down: torch.Tensor,
+               group=None) -> torch.Tensor:
+    """Golden single-device reference for THIS rank's tokens: every rank has
+    the full expert weights [E, ...]; no communication needed."""
+    T, H = x.shape
+    E = full_w_gate_up.shape[0]
+    inter = full_w_down.shape[2]
+    K = topk_ids.shape[1]
+    acc = torch.zeros(T, H, dtype=torch.float32, device=x.device)
+    for t in range(T):
+        for k in range(K):
+            e = int(topk_ids[t, k])
+            if e < 0 or e >= E:
+                continue
+            h = x[t].float() @ full_w_gate_up[e].float().t()
+            a = torch.nn.functional.silu(h[:inter]) * h[inter:]
+            y = a @ full_w_down[e].float().t()
+            acc[t] += float(topk_w[t, k]) * y
+    return acc.to(x.dtype)
