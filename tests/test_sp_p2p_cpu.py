@@ -1,0 +1,74 @@
+"""CPU tests: SP flash-decode LSE merge, Ulysses a2a, p2p ring (gloo+shm)."""
+import torch
+
+from tests.conftest import run_distributed
+
+
+def _body_sp_decode(rank, world):
+    from triton_dist_amd.ops import (create_sp_flash_decode_context,
+                                     sp_flash_decode, sp_flash_decode_ref)
+    from triton_dist_amd.utils import assert_allclose
+
+    b, qh, kvh, d = 4, 4, 2, 128
+    chunk, maxlen = 16, 32
+    ctx = create_sp_flash_decode_context(max_batch=b, qh=qh)
+    g = torch.Generator().manual_seed(3)  # same full KV on all ranks
+    k_full = (torch.randn(b, world * chunk, kvh, d, generator=g) / 4).to(torch.bfloat16)
+    v_full = (torch.randn(b, world * chunk, kvh, d, generator=g) / 4).to(torch.bfloat16)
+    q = (torch.randn(b, qh * d, generator=g) / 4).to(torch.bfloat16)
+    # my shard
+    kc = torch.zeros(b, maxlen, kvh, d, dtype=torch.bfloat16)
+    vc = torch.zeros(b, maxlen, kvh, d, dtype=torch.bfloat16)
+    kc[:, :chunk] = k_full[:, rank * chunk:(rank + 1) * chunk]
+    vc[:, :chunk] = v_full[:, rank * chunk:(rank + 1) * chunk]
+    clen = torch.tensor(chunk, dtype=torch.int64)
+    out = sp_flash_decode(q, kc, vc, clen, ctx, qh, kvh)
+    ref = sp_flash_decode_ref(q, k_full, v_full, world * chunk, qh, kvh)
+    assert_allclose(out, ref, atol=5e-2, rtol=5e-2)
+
+
+def test_sp_flash_decode_cpu_2rank():
+    run_distributed(_body_sp_decode, world_size=2)
+
+
+def _body_ulysses(rank, world):
+    from triton_dist_amd.ops import (create_ulysses_context, ulysses_a2a,
+                                     ulysses_a2a_ref)
+    from triton_dist_amd.utils import assert_allclose
+
+    t_loc, heads, d = 8, 4 * world, 16
+    ctx = create_ulysses_context(max_tokens=t_loc, n_heads=heads, head_dim=d)
+    g = torch.Generator().manual_seed(10 + rank)
+    x = (torch.randn(t_loc, heads, d, generator=g)).to(torch.bfloat16)
+    out = ulysses_a2a(x, ctx)
+    ref = ulysses_a2a_ref(x)
+    assert torch.equal(out, ref), (out.shape, ref.shape)
+
+
+def test_ulysses_cpu_2rank():
+    run_distributed(_body_ulysses, world_size=2)
+
+
+def test_ulysses_cpu_4rank():
+    run_distributed(_body_ulysses, world_size=4)
+
+
+def _body_p2p(rank, world):
+    from triton_dist_amd.ops import create_p2p_context, p2p_recv, p2p_send
+
+    ctx = create_p2p_context(max_bytes=4096, depth=2)
+    nxt, prv = (rank + 1) % world, (rank - 1) % world
+    for i in range(6):  # exceed depth to exercise credits
+        x = torch.full((64,), float(rank * 100 + i)).to(torch.bfloat16)
+        p2p_send(x, nxt, ctx)
+        out = torch.empty(64, dtype=torch.bfloat16)
+        p2p_recv(out, prv, ctx)
+        assert (out.float() == prv * 100 + i).all(), (rank, i, out[0])
+
+
+def test_p2p_ring_cpu_2rank():
+    run_distributed(_body_p2p, world_size=2)
+
+
+def test_p2p_ring_cpu_4rank():
+    run_distributed(_body_p2p, world_size=4)

@@ -143,5 +143,13 @@ void launch_flash_decode(const void *q, const void *kcache,
                          const void *vcache, void *out, const void *offset,
                          int batch, int qh, int kvh, int max_len,
                          hipStream_t stream);
+void launch_flash_decode_partial(const void *q, const void *kcache,
+                                 const void *vcache, void *out_part,
+                                 void *lse, const void *chunk_len, int batch,
+                                 int qh, int kvh, int max_len,
+                                 hipStream_t stream);
+void launch_lse_combine(const void *parts, const void *lses, void *out,
+                        const void *flags, int world, int batch, int qh,
+                        hipStream_t stream);
 
 }  // namespace td

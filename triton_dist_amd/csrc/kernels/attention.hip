@@ -155,4 +155,179 @@ void launch_flash_decode(const void *q, const void *kcache,
                      (const long *)offset, qh, kvh, max_len, scale);
 }
 
+// ---------------------------------------------------------------------------
+// SP split-KV decode: same kernel body, but the KV chunk length comes from
+// `chunk_len` (device int64: number of valid positions in THIS rank's KV
+// shard) and the outputs are the UNNORMALIZED partial accumulator plus the
+// log-sum-exp, for the cross-rank combine (capability: reference
+// flash_decode.py:482-532 inter-rank LSE-merge).
+// out_part: [B, qh, 128] fp32 (acc / l), lse: [B, qh] fp32 (m + log l).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_flash_decode_partial(
+    const bf16 *__restrict__ q, const bf16 *__restrict__ kcache,
+    const bf16 *__restrict__ vcache, float *__restrict__ out_part,
+    float *__restrict__ lse, const long *__restrict__ chunk_len, int qh,
+    int kvh, int max_len, float scale) {
+  const int b = blockIdx.x;
+  const int kh = blockIdx.y;
+  const int G = qh / kvh;
+  const int tid = threadIdx.x;
+  const int g = tid >> 5;
+  const int t = tid & 31;
+  const long seqlen = *chunk_len;
+
+  __shared__ bf16 k_lds[kTile][kD];
+  __shared__ bf16 v_lds[kTile][kD];
+  __shared__ float p_lds[8][kTile];
+  __shared__ float m_lds[8], r_lds[8], l_lds[8];
+  __shared__ bf16 q_lds[8][kD];
+
+  for (int i = tid; i < 8 * kD / 8; i += 256) {
+    int hh = i / (kD / 8);
+    int c = (i % (kD / 8)) * 8;
+    bf16x8 v{};
+    if (hh < G)
+      v = *(const bf16x8 *)(q + (((size_t)b * qh) + kh * G + hh) * kD + c);
+    *(bf16x8 *)(&q_lds[hh][c]) = v;
+  }
+  if (tid < 8) {
+    m_lds[tid] = -1e30f;
+    l_lds[tid] = 0.f;
+  }
+  __syncthreads();
+
+  float acc[4] = {};
+  const int my_d0 = t * 4;
+  const long ntiles = (seqlen + kTile - 1) / kTile;
+  for (long tile = 0; tile < ntiles; ++tile) {
+    const long pos0 = tile * kTile;
+    __syncthreads();
+    for (int i = tid; i < kTile * kD / 8; i += 256) {
+      int r = i / (kD / 8);
+      int c = (i % (kD / 8)) * 8;
+      long pos = pos0 + r;
+      bf16x8 kv{}, vv{};
+      if (pos < seqlen) {
+        size_t base = (((size_t)b * max_len + pos) * kvh + kh) * kD + c;
+        kv = *(const bf16x8 *)(kcache + base);
+        vv = *(const bf16x8 *)(vcache + base);
+      }
+      *(bf16x8 *)(&k_lds[r][c]) = kv;
+      *(bf16x8 *)(&v_lds[r][c]) = vv;
+    }
+    __syncthreads();
+    float s = -1e30f;
+    if (g < G && pos0 + t < seqlen) {
+      float d = 0.f;
+#pragma unroll
+      for (int c = 0; c < kD / 8; ++c) {
+        bf16x8 qv = *(const bf16x8 *)(&q_lds[g][c * 8]);
+        bf16x8 kv = *(const bf16x8 *)(&k_lds[t][c * 8]);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) d += (float)qv[j] * (float)kv[j];
+      }
+      s = d * scale;
+    }
+    float mx = s;
+    for (int off = 16; off > 0; off >>= 1)
+      mx = fmaxf(mx, __shfl_xor(mx, off));
+    float m_old = m_lds[g];
+    float m_new = fmaxf(m_old, mx);
+    float p = (s > -1e29f) ? __expf(s - m_new) : 0.f;
+    p_lds[g][t] = p;
+    float psum = p;
+    for (int off = 16; off > 0; off >>= 1) psum += __shfl_xor(psum, off);
+    if (t == 0) {
+      float r = __expf(m_old - m_new);
+      r_lds[g] = r;
+      l_lds[g] = l_lds[g] * r + psum;
+      m_lds[g] = m_new;
+    }
+    __syncthreads();
+    const float r = r_lds[g];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[j] *= r;
+    for (int tt = 0; tt < kTile; ++tt) {
+      float p = p_lds[g][tt];
+      if (p != 0.f) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[j] += p * (float)v_lds[tt][my_d0 + j];
+      }
+    }
+  }
+  __syncthreads();
+  if (g < G) {
+    float l = l_lds[g];
+    float inv_l = l > 0.f ? 1.f / l : 0.f;
+    float *dst = out_part + (((size_t)b * qh) + kh * G + g) * kD + my_d0;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) dst[j] = acc[j] * inv_l;
+    if (t == 0) {
+      float m = m_lds[g];
+      lse[(size_t)b * qh + kh * G + g] =
+          (l > 0.f) ? m + __logf(l) : -1e30f;
+    }
+  }
+}
+
+void launch_flash_decode_partial(const void *q, const void *kcache,
+                                 const void *vcache, void *out_part,
+                                 void *lse, const void *chunk_len, int batch,
+                                 int qh, int kvh, int max_len,
+                                 hipStream_t stream) {
+  if (qh / kvh > 8 || qh % kvh)
+    throw std::runtime_error("flash_decode: qh/kvh must divide and be <= 8");
+  float scale = 1.f / sqrtf((float)kD);
+  hipLaunchKernelGGL(k_flash_decode_partial, dim3(batch, kvh), dim3(256), 0,
+                     stream, (const bf16 *)q, (const bf16 *)kcache,
+                     (const bf16 *)vcache, (float *)out_part, (float *)lse,
+                     (const long *)chunk_len, qh, kvh, max_len, scale);
+}
+
+// ---------------------------------------------------------------------------
+// Cross-rank LSE combine: partials [world, B, qh, 128] fp32 + lses
+// [world, B, qh] fp32 -> out [B, qh, 128] bf16, after waiting the per-rank
+// ready flags. Block per (b, h); 32 lanes per 128 dims x 4.
+// ---------------------------------------------------------------------------
+__global__ void k_lse_combine(const float *__restrict__ parts,
+                              const float *__restrict__ lses,
+                              bf16 *__restrict__ out, const int *flags,
+                              int world, int batch, int qh) {
+  const int b = blockIdx.x;
+  const int h = blockIdx.y;
+  if (threadIdx.x < (unsigned)world)
+    wait_ge_one<Scope::Sys>(flags + threadIdx.x, 1);
+  __syncthreads();
+  __shared__ float w_sh[kMaxRanks];
+  if (threadIdx.x == 0) {
+    float mx = -1e30f;
+    for (int r = 0; r < world; ++r)
+      mx = fmaxf(mx, lses[((size_t)r * batch + b) * qh + h]);
+    float denom = 0.f;
+    for (int r = 0; r < world; ++r) {
+      float w = __expf(lses[((size_t)r * batch + b) * qh + h] - mx);
+      w_sh[r] = w;
+      denom += w;
+    }
+    for (int r = 0; r < world; ++r) w_sh[r] /= denom;
+  }
+  __syncthreads();
+  for (int d = threadIdx.x; d < kD; d += blockDim.x) {
+    float acc = 0.f;
+    for (int r = 0; r < world; ++r)
+      acc += w_sh[r] *
+             parts[(((size_t)r * batch + b) * qh + h) * kD + d];
+    out[((size_t)b * qh + h) * kD + d] = (bf16)acc;
+  }
+}
+
+void launch_lse_combine(const void *parts, const void *lses, void *out,
+                        const void *flags, int world, int batch, int qh,
+                        hipStream_t stream) {
+  hipLaunchKernelGGL(k_lse_combine, dim3(batch, qh), dim3(128), 0, stream,
+                     (const float *)parts, (const float *)lses, (bf16 *)out,
+                     (const int *)flags, world, batch, qh);
+}
+
 }  // namespace td

@@ -507,6 +507,26 @@ static void flash_decode(uintptr_t q, uintptr_t kcache, uintptr_t vcache,
   TD_CHECK_HIP(hipGetLastError());
 }
 
+static void flash_decode_partial(uintptr_t q, uintptr_t kcache,
+                                 uintptr_t vcache, uintptr_t out_part,
+                                 uintptr_t lse, uintptr_t chunk_len,
+                                 int batch, int qh, int kvh, int max_len,
+                                 uintptr_t stream) {
+  launch_flash_decode_partial((void *)q, (void *)kcache, (void *)vcache,
+                              (void *)out_part, (void *)lse,
+                              (void *)chunk_len, batch, qh, kvh, max_len,
+                              as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
+static void lse_combine(uintptr_t parts, uintptr_t lses, uintptr_t out,
+                        uintptr_t flags, int world, int batch, int qh,
+                        uintptr_t stream) {
+  launch_lse_combine((void *)parts, (void *)lses, (void *)out, (void *)flags,
+                     world, batch, qh, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
 PYBIND11_MODULE(_C, m) {
   m.doc() = "triton_dist_amd native core: hipIpc symmetric heap + gfx950 kernels";
   m.def("heap_init", &heap_init, py::arg("rank"), py::arg("world"),
@@ -550,4 +570,6 @@ PYBIND11_MODULE(_C, m) {
   m.def("swiglu", &swiglu);
   m.def("qkv_prologue_decode", &qkv_prologue_decode);
   m.def("flash_decode", &flash_decode);
+  m.def("flash_decode_partial", &flash_decode_partial);
+  m.def("lse_combine", &lse_combine);
 }

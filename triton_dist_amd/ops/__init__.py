@@ -32,3 +32,23 @@ from .ep_moe import (  # noqa: F401
     ep_moe_forward,
     ep_moe_ref,
 )
+from .sp import (  # noqa: F401
+    SPFlashDecodeContext,
+    create_sp_flash_decode_context,
+    sp_flash_decode,
+    sp_flash_decode_ref,
+    UlyssesContext,
+    create_ulysses_context,
+    ulysses_a2a,
+    ulysses_a2a_ref,
+    SPAGAttnContext,
+    create_sp_ag_attn_context,
+    sp_ag_attention,
+)
+from .p2p import (  # noqa: F401
+    P2PContext,
+    create_p2p_context,
+    p2p_send,
+    p2p_recv,
+    PPCommLayer,
+)

@@ -1,0 +1,424 @@
+"""Sequence-parallel ops: distributed flash-decode (seq-sharded KV with
+cross-rank LSE merge), Ulysses head<->sequence all-to-all, and ring-AG
+prefill attention with online LSE merging.
+
+Capability parity (behavior only) with Triton-distributed:
+  kernels/nvidia/flash_decode.py:130-1132 + layers/nvidia/
+  sp_flash_decode_layer.py:44-149   — split-KV decode + inter-rank combine
+  kernels/nvidia/sp_ulysess_*.py, ulysses_sp_dispatch.py:470-606 — Ulysses
+  kernels/nvidia/sp_ag_attention_intra_node.py:106-428 — AG-attention
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import List, Optional
+
+import torch
+
+from ..runtime import cpu_shm
+from ..runtime.symm_mem import SymmBuffer, SymmHeap, get_heap
+
+
+# ---------------------------------------------------------------------------
+# Distributed flash-decode
+# ---------------------------------------------------------------------------
+@dataclass
+class SPFlashDecodeContext:
+    heap: SymmHeap
+    max_batch: int
+    qh: int
+    parts: SymmBuffer   # [world, B, qh, 128] fp32
+    lses: SymmBuffer    # [world, B, qh] fp32
+    flags: SymmBuffer   # [world] int32
+    epoch: int = 0
+
+    @property
+    def world(self):
+        return self.heap.world
+
+    @property
+    def rank(self):
+        return self.heap.rank
+
+
+def create_sp_flash_decode_context(max_batch: int, qh: int,
+                                   heap: Optional[SymmHeap] = None
+                                   ) -> SPFlashDecodeContext:
+    heap = heap or get_heap()
+    w = heap.world
+    return SPFlashDecodeContext(
+        heap, max_batch, qh,
+        parts=heap.alloc_buffer((w, max_batch, qh, 128), torch.float32),
+        lses=heap.alloc_buffer((w, max_batch, qh), torch.float32),
+        flags=heap.alloc_buffer((w,), torch.int32),
+    )
+
+
+def sp_flash_decode(q: torch.Tensor, kv_k: torch.Tensor, kv_v: torch.Tensor,
+                    chunk_len: torch.Tensor, ctx: SPFlashDecodeContext,
+                    qh: int, kvh: int,
+                    out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Sequence-parallel GQA decode: every rank holds a KV chunk of length
+    `chunk_len` (device int64) for ALL sequences; q [B, qh*128] replicated.
+    Each rank computes its split-KV partial + LSE, pushes both to every
+    peer, and the combine kernel merges. Returns [B, qh*128] bf16 on every
+    rank."""
+    b = q.shape[0]
+    world, rank = ctx.world, ctx.rank
+    assert b <= ctx.max_batch and qh == ctx.qh
+    heap = ctx.heap
+    if heap.backend == "cpu":
+        return _sp_flash_decode_cpu(q, kv_k, kv_v, chunk_len, ctx, qh, kvh,
+                                    out)
+    _C = heap._C
+    stream = torch.cuda.current_stream()
+    s = stream.cuda_stream
+    if out is None:
+        out = torch.empty_like(q)
+
+    _C.reset_flags(ctx.flags.ptr(), world, 0, s)
+    heap.barrier_all_on_stream(stream)
+
+    part_row = ctx.max_batch * qh * 128 * 4   # bytes per rank slot
+    lse_row = ctx.max_batch * qh * 4
+    my_part = ctx.parts.ptr() + rank * part_row
+    my_lse = ctx.lses.ptr() + rank * lse_row
+    _C.flash_decode_partial(q.data_ptr(), kv_k.data_ptr(), kv_v.data_ptr(),
+                            my_part, my_lse, chunk_len.data_ptr(), b, qh,
+                            kvh, kv_k.shape[1], s)
+    # push my partial+lse to peers (SDMA), then my own flag
+    for i in range(world - 1):
+        peer = (rank + 1 + i) % world
+        _C.memcpy_async(ctx.parts.ptr(peer) + rank * part_row, my_part,
+                        b * qh * 128 * 4, s)
+        _C.memcpy_async(ctx.lses.ptr(peer) + rank * lse_row, my_lse,
+                        b * qh * 4, s)
+        _C.memcpy_async(ctx.flags.ptr(peer) + rank * 4, heap.one_src.ptr(),
+                        4, s)
+    _C.reset_flags(ctx.flags.ptr() + rank * 4, 1, 1, s)
+    # combine waits all flags
+    # NOTE parts layout uses max_batch stride; kernel expects [world,b,qh]
+    # dense — valid since b rows are leading within each slot when
+    # b == max_batch; for b < max_batch we pass strides via max_batch.
+    assert b == ctx.max_batch, "v1: batch must equal ctx.max_batch"
+    _C.lse_combine(ctx.parts.ptr(), ctx.lses.ptr(), out.data_ptr(),
+                   ctx.flags.ptr(), world, b, qh, s)
+    return out
+
+
+def _sp_flash_decode_cpu(q, kv_k, kv_v, chunk_len, ctx, qh, kvh, out):
+    """CPU mock: torch partial attention with explicit LSE + shm exchange."""
+    b = q.shape[0]
+    d = 128
+    world, rank = ctx.world, ctx.rank
+    L = int(chunk_len)
+    ctx.epoch += 1
+    heap = ctx.heap
+    heap.barrier_all()
+    g = qh // kvh
+    qv = q.view(b, kvh, g, d).float()
+    ks = kv_k[:, :L].float()  # [b, L, kvh, d]
+    vs = kv_v[:, :L].float()
+    scores = torch.einsum("bhgd,blhd->bhgl", qv, ks) / (d ** 0.5)
+    m = scores.amax(-1, keepdim=True)
+    p = torch.exp(scores - m)
+    l = p.sum(-1, keepdim=True)
+    o = torch.einsum("bhgl,blhd->bhgd", p / l, vs)
+    lse = (m + torch.log(l)).squeeze(-1)  # [b, kvh, g]
+    for peer in range(world):
+        ctx.parts.peer(peer)[rank, :b].copy_(
+            o.reshape(b, qh, d))
+        ctx.lses.peer(peer)[rank, :b].copy_(lse.reshape(b, qh))
+        cpu_shm.notify(ctx.flags.peer(peer), rank, ctx.epoch)
+    fl = ctx.flags.local()
+    for r in range(world):
+        cpu_shm.wait_ge(fl, r, ctx.epoch)
+    parts = ctx.parts.local()[:, :b]   # [world, b, qh, d]
+    lses = ctx.lses.local()[:, :b]     # [world, b, qh]
+    mx = lses.amax(0)
+    w = torch.exp(lses - mx)
+    w = w / w.sum(0)
+    merged = (parts * w.unsqueeze(-1)).sum(0)
+    res = merged.reshape(b, qh * d).to(q.dtype)
+    if out is not None:
+        out.copy_(res)
+        return out
+    return res
+
+
+def sp_flash_decode_ref(q, kv_k_full, kv_v_full, total_len, qh, kvh):
+    """Golden: sdpa over the full (gathered) KV."""
+    import torch.nn.functional as F
+
+    b = q.shape[0]
+    qs = q.view(b, qh, 1, 128).float()
+    ks = kv_k_full[:, :total_len].transpose(1, 2).float()
+    vs = kv_v_full[:, :total_len].transpose(1, 2).float()
+    o = F.scaled_dot_product_attention(qs, ks, vs, enable_gqa=True)
+    return o.view(b, qh * 128).to(q.dtype)
+
+
+# ---------------------------------------------------------------------------
+# Ulysses all-to-all (tokens <-> heads resharding)
+# ---------------------------------------------------------------------------
+@dataclass
+class UlyssesContext:
+    heap: SymmHeap
+    max_tokens: int  # per-rank tokens
+    h_loc: int       # heads per rank after resharding
+    head_dim: int
+    recv: SymmBuffer  # [world, max_tokens, h_loc, D]
+    flags: SymmBuffer
+    epoch: int = 0
+
+    @property
+    def world(self):
+        return self.heap.world
+
+    @property
+    def rank(self):
+        return self.heap.rank
+
+
+def create_ulysses_context(max_tokens: int, n_heads: int, head_dim: int,
+                           heap: Optional[SymmHeap] = None) -> UlyssesContext:
+    heap = heap or get_heap()
+    w = heap.world
+    assert n_heads % w == 0
+    h_loc = n_heads // w
+    return UlyssesContext(
+        heap, max_tokens, h_loc, head_dim,
+        recv=heap.alloc_buffer((w, max_tokens, h_loc, head_dim),
+                               torch.bfloat16),
+        flags=heap.alloc_buffer((w,), torch.int32),
+    )
+
+
+def ulysses_a2a(x: torch.Tensor, ctx: UlyssesContext) -> torch.Tensor:
+    """[T_loc, n_heads, D] (seq-sharded, all heads) ->
+    [world*T_loc, h_loc, D] (all seq, head shard). The inverse resharding is
+    the same op with heads<->tokens roles swapped by the caller's reshape.
+    """
+    t_loc, n_heads, d = x.shape
+    world, rank = ctx.world, ctx.rank
+    h_loc = ctx.h_loc
+    assert n_heads == h_loc * world and d == ctx.head_dim
+    assert t_loc <= ctx.max_tokens
+    heap = ctx.heap
+
+    if heap.backend == "cpu":
+        ctx.epoch += 1
+        heap.barrier_all()
+        for peer in range(world):
+            seg = x[:, peer * h_loc:(peer + 1) * h_loc].contiguous()
+            ctx.recv.peer(peer)[rank, :t_loc].copy_(seg)
+            cpu_shm.notify(ctx.flags.peer(peer), rank, ctx.epoch)
+        fl = ctx.flags.local()
+        for r in range(world):
+            cpu_shm.wait_ge(fl, r, ctx.epoch)
+        return ctx.recv.local()[:, :t_loc].reshape(world * t_loc, h_loc, d) \
+            .clone()
+
+    _C = heap._C
+    stream = torch.cuda.current_stream()
+    s = stream.cuda_stream
+    _C.reset_flags(ctx.flags.ptr(), world, 0, s)
+    heap.barrier_all_on_stream(stream)
+    # contiguous per-peer head slices, then one SDMA per peer + flag
+    seg_bytes = t_loc * h_loc * d * 2
+    slot_bytes = ctx.max_tokens * h_loc * d * 2
+    # send buffer in [peer][t][h_loc][d] order (receiver's layout)
+    xs = x.view(t_loc, world, h_loc, d).permute(1, 0, 2, 3).contiguous()
+    for i in range(world - 1):
+        peer = (rank + 1 + i) % world
+        _C.memcpy_async(ctx.recv.ptr(peer) + rank * slot_bytes,
+                        xs[peer].data_ptr(), seg_bytes, s)
+        _C.memcpy_async(ctx.flags.ptr(peer) + rank * 4, heap.one_src.ptr(),
+                        4, s)
+    _C.memcpy_async(ctx.recv.ptr() + rank * slot_bytes,
+                    xs[rank].data_ptr(), seg_bytes, s)
+    _C.reset_flags(ctx.flags.ptr() + rank * 4, 1, 1, s)
+    _C.wait_eq(ctx.flags.ptr(), world, 1, s)
+    return ctx.recv.local()[:, :t_loc].reshape(world * t_loc, h_loc, d)
+
+
+def ulysses_a2a_ref(x: torch.Tensor, group=None) -> torch.Tensor:
+    """Golden: torch.distributed all_to_all of head slices (CPU-staged)."""
+    import torch.distributed as dist
+
+    world = dist.get_world_size(group)
+    rank = dist.get_rank(group)
+    t_loc, n_heads, d = x.shape
+    h_loc = n_heads // world
+    xs = x.view(t_loc, world, h_loc, d).permute(1, 0, 2, 3).contiguous().cpu()
+    # a2a via all_gather (gloo has no alltoall): every rank gathers all send
+    # matrices, then picks the column addressed to it.
+    gathered = [torch.empty_like(xs) for _ in range(world)]
+    dist.all_gather(gathered, xs, group=group)
+    outs = [gathered[src][rank] for src in range(world)]
+    return torch.cat(outs, 0).to(x.device)
+
+
+# ---------------------------------------------------------------------------
+# Ring-AG prefill attention: K/V chunks are all-gathered peer-by-peer on
+# comm streams while the consumer runs chunked flash attention with online
+# LSE merging (torch aten flash returns the logsumexp on ROCm/aotriton).
+# ---------------------------------------------------------------------------
+@dataclass
+class SPAGAttnContext:
+    heap: SymmHeap
+    max_tokens: int  # per-rank sequence chunk
+    kvh: int
+    head_dim: int
+    kbuf: SymmBuffer   # [world, B? folded] — [world, max_tokens, kvh, D]
+    vbuf: SymmBuffer
+    flags: SymmBuffer
+    comm_streams: List = field(default_factory=list)
+    ready_ev: Optional[object] = None
+    join_evs: List = field(default_factory=list)
+
+    @property
+    def world(self):
+        return self.heap.world
+
+    @property
+    def rank(self):
+        return self.heap.rank
+
+
+def create_sp_ag_attn_context(max_chunk_tokens: int, kvh: int, head_dim: int,
+                              heap: Optional[SymmHeap] = None
+                              ) -> SPAGAttnContext:
+    heap = heap or get_heap()
+    w = heap.world
+    ctx = SPAGAttnContext(
+        heap, max_chunk_tokens, kvh, head_dim,
+        kbuf=heap.alloc_buffer((w, max_chunk_tokens, kvh, head_dim),
+                               torch.bfloat16),
+        vbuf=heap.alloc_buffer((w, max_chunk_tokens, kvh, head_dim),
+                               torch.bfloat16),
+        flags=heap.alloc_buffer((w,), torch.int32),
+    )
+    if heap.backend == "hip":
+        n = max(w - 1, 1)
+        ctx.comm_streams = [torch.cuda.Stream() for _ in range(min(n, 7))]
+        ctx.ready_ev = torch.cuda.Event()
+        ctx.join_evs = [torch.cuda.Event() for _ in ctx.comm_streams]
+    return ctx
+
+
+def _flash_with_lse(q, k, v, causal):
+    """q/k/v: [B, H, S, D] bf16 -> (out [B,H,S,D], lse [B,H,S])."""
+    outs = torch.ops.aten._scaled_dot_product_flash_attention(
+        q, k, v, dropout_p=0.0, is_causal=causal)
+    return outs[0], outs[1]
+
+
+def _merge_lse(o1, l1, o2, l2):
+    """Merge two attention partials with their logsumexps."""
+    m = torch.maximum(l1, l2)
+    w1 = torch.exp(l1 - m)
+    w2 = torch.exp(l2 - m)
+    denom = w1 + w2
+    o = (o1 * (w1 / denom).unsqueeze(-1).to(o1.dtype)
+         + o2 * (w2 / denom).unsqueeze(-1).to(o2.dtype))
+    return o, m + torch.log(denom)
+
+
+def sp_ag_attention(q: torch.Tensor, k_chunk: torch.Tensor,
+                    v_chunk: torch.Tensor, ctx: SPAGAttnContext,
+                    qh: int) -> torch.Tensor:
+    """Causal self-attention with sequence sharding (zig-zag-free v1:
+    contiguous chunks; rank r's queries attend chunks 0..r).
+
+    q: [S_loc, qh, D] (this rank's query chunk, post-RoPE)
+    k_chunk/v_chunk: [S_loc, kvh, D] (this rank's KV chunk, post-RoPE)
+    Returns [S_loc, qh, D].
+    """
+    s_loc = q.shape[0]
+    world, rank = ctx.world, ctx.rank
+    kvh, d = ctx.kvh, ctx.head_dim
+    heap = ctx.heap
+    assert s_loc <= ctx.max_tokens
+
+    if heap.backend == "cpu":
+        return _sp_ag_attention_cpu(q, k_chunk, v_chunk, ctx, qh)
+
+    _C = heap._C
+    compute = torch.cuda.current_stream()
+    s = compute.cuda_stream
+    _C.reset_flags(ctx.flags.ptr(), world, 0, s)
+    heap.barrier_all_on_stream(compute)
+    slot = ctx.max_tokens * kvh * d * 2
+    nbytes = s_loc * kvh * d * 2
+    kc = k_chunk.contiguous()
+    vc = v_chunk.contiguous()
+    # local chunk into my slot + flag
+    _C.memcpy_async(ctx.kbuf.ptr() + rank * slot, kc.data_ptr(), nbytes, s)
+    _C.memcpy_async(ctx.vbuf.ptr() + rank * slot, vc.data_ptr(), nbytes, s)
+    _C.reset_flags(ctx.flags.ptr() + rank * 4, 1, 1, s)
+    # push to peers on comm streams (cp-engine producer)
+    ctx.ready_ev.record(compute)
+    ns = len(ctx.comm_streams)
+    for i in range(world - 1):
+        peer = (rank + 1 + i) % world
+        st = ctx.comm_streams[i % ns]
+        if i < ns:
+            st.wait_event(ctx.ready_ev)
+        _C.memcpy_async(ctx.kbuf.ptr(peer) + rank * slot, kc.data_ptr(),
+                        nbytes, st.cuda_stream)
+        _C.memcpy_async(ctx.vbuf.ptr(peer) + rank * slot, vc.data_ptr(),
+                        nbytes, st.cuda_stream)
+        _C.memcpy_async(ctx.flags.ptr(peer) + rank * 4, heap.one_src.ptr(),
+                        4, st.cuda_stream)
+
+    # consumer: chunks 0..rank (causal), waiting each chunk's flag
+    qt = q.permute(1, 0, 2).unsqueeze(0)  # [1, qh, S_loc, D]
+    o = lse = None
+    for src in range(rank + 1):
+        _C.wait_eq(ctx.flags.ptr() + src * 4, 1, 1, s)
+        kv_k = ctx.kbuf.local()[src, :s_loc].permute(1, 0, 2).unsqueeze(0)
+        kv_v = ctx.vbuf.local()[src, :s_loc].permute(1, 0, 2).unsqueeze(0)
+        causal = (src == rank)
+        oc, lc = _flash_with_lse(qt.contiguous(), kv_k.contiguous(),
+                                 kv_v.contiguous(), causal)
+        if o is None:
+            o, lse = oc, lc
+        else:
+            o, lse = _merge_lse(o, lse, oc, lc)
+    for i, ev in enumerate(ctx.join_evs):
+        ev.record(ctx.comm_streams[i])
+        compute.wait_event(ev)
+    return o.squeeze(0).permute(1, 0, 2).contiguous()
+
+
+def _sp_ag_attention_cpu(q, k_chunk, v_chunk, ctx, qh):
+    import torch.nn.functional as F
+
+    s_loc = q.shape[0]
+    world, rank = ctx.world, ctx.rank
+    ctx.epoch = getattr(ctx, "epoch", 0) + 1
+    heap = ctx.heap
+    heap.barrier_all()
+    for peer in range(world):
+        ctx.kbuf.peer(peer)[rank, :s_loc].copy_(k_chunk)
+        ctx.vbuf.peer(peer)[rank, :s_loc].copy_(v_chunk)
+        cpu_shm.notify(ctx.flags.peer(peer), rank, ctx.epoch)
+    fl = ctx.flags.local()
+    for r in range(world):
+        cpu_shm.wait_ge(fl, r, ctx.epoch)
+    # full causal attention over gathered prefix (CPU golden-ish path)
+    ks = ctx.kbuf.local()[:rank + 1, :s_loc].reshape((rank + 1) * s_loc, -1,
+                                                     ctx.head_dim)
+    vs = ctx.vbuf.local()[:rank + 1, :s_loc].reshape((rank + 1) * s_loc, -1,
+                                                     ctx.head_dim)
+    qt = q.permute(1, 0, 2).unsqueeze(0).float()
+    kt = ks.permute(1, 0, 2).unsqueeze(0).float()
+    vt = vs.permute(1, 0, 2).unsqueeze(0).float()
+    total = (rank + 1) * s_loc
+    qpos = rank * s_loc + torch.arange(s_loc)
+    mask = torch.arange(total).unsqueeze(0) <= qpos.unsqueeze(1)
+    o = F.scaled_dot_product_attention(qt, kt, vt,
+                                       attn_mask=mask.view(1, 1, s_loc, total),
+                                       enable_gqa=True)
+    return o.squeeze(0).permute(1, 0, 2).to(q.dtype)
