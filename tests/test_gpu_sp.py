@@ -1,0 +1,111 @@
+"""GPU SP + P2P tests (2 ranks sharing one GPU over hipIpc)."""
+import pytest
+import torch
+
+from tests.conftest import run_distributed
+
+pytestmark = pytest.mark.gpu
+
+
+def _body_sp_decode(rank, world):
+    from triton_dist_amd.ops import (create_sp_flash_decode_context,
+                                     sp_flash_decode, sp_flash_decode_ref)
+    from triton_dist_amd.utils import assert_allclose
+
+    b, qh, kvh, d = 16, 8, 2, 128
+    chunk, maxlen = 48, 64
+    ctx = create_sp_flash_decode_context(max_batch=b, qh=qh)
+    g = torch.Generator("cuda").manual_seed(3)
+    k_full = (torch.randn(b, world * chunk, kvh, d, device="cuda",
+                          generator=g) / 4).to(torch.bfloat16)
+    v_full = (torch.randn(b, world * chunk, kvh, d, device="cuda",
+                          generator=g) / 4).to(torch.bfloat16)
+    q = (torch.randn(b, qh * d, device="cuda", generator=g) / 4
+         ).to(torch.bfloat16)
+    kc = torch.zeros(b, maxlen, kvh, d, dtype=torch.bfloat16, device="cuda")
+    vc = torch.zeros(b, maxlen, kvh, d, dtype=torch.bfloat16, device="cuda")
+    kc[:, :chunk] = k_full[:, rank * chunk:(rank + 1) * chunk]
+    vc[:, :chunk] = v_full[:, rank * chunk:(rank + 1) * chunk]
+    clen = torch.tensor(chunk, dtype=torch.int64, device="cuda")
+    for _ in range(2):
+        out = sp_flash_decode(q, kc, vc, clen, ctx, qh, kvh)
+        torch.cuda.synchronize()
+        ref = sp_flash_decode_ref(q, k_full, v_full, world * chunk, qh, kvh)
+        assert_allclose(out, ref, atol=5e-2, rtol=5e-2)
+
+
+def test_sp_flash_decode_gpu_2rank():
+    run_distributed(_body_sp_decode, world_size=2)
+
+
+def _body_ulysses(rank, world):
+    from triton_dist_amd.ops import (create_ulysses_context, ulysses_a2a,
+                                     ulysses_a2a_ref)
+
+    t_loc, heads, d = 64, 8, 128
+    ctx = create_ulysses_context(max_tokens=t_loc, n_heads=heads, head_dim=d)
+    torch.manual_seed(10 + rank)
+    x = torch.randn(t_loc, heads, d, device="cuda").to(torch.bfloat16)
+    for _ in range(2):
+        out = ulysses_a2a(x, ctx)
+        torch.cuda.synchronize()
+        ref = ulysses_a2a_ref(x)
+        assert torch.equal(out.cpu(), ref.cpu())
+
+
+def test_ulysses_gpu_2rank():
+    run_distributed(_body_ulysses, world_size=2)
+
+
+def _body_ag_attn(rank, world):
+    import torch.nn.functional as F
+
+    from triton_dist_amd.ops import create_sp_ag_attn_context, sp_ag_attention
+    from triton_dist_amd.utils import assert_allclose
+
+    s_loc, qh, kvh, d = 128, 8, 2, 128
+    ctx = create_sp_ag_attn_context(max_chunk_tokens=s_loc, kvh=kvh,
+                                    head_dim=d)
+    g = torch.Generator("cuda").manual_seed(1)  # same full tensors everywhere
+    total = world * s_loc
+    q_full = (torch.randn(total, qh, d, device="cuda", generator=g) / 4
+              ).to(torch.bfloat16)
+    k_full = (torch.randn(total, kvh, d, device="cuda", generator=g) / 4
+              ).to(torch.bfloat16)
+    v_full = (torch.randn(total, kvh, d, device="cuda", generator=g) / 4
+              ).to(torch.bfloat16)
+    q = q_full[rank * s_loc:(rank + 1) * s_loc]
+    out = sp_ag_attention(q, k_full[rank * s_loc:(rank + 1) * s_loc],
+                          v_full[rank * s_loc:(rank + 1) * s_loc], ctx, qh)
+    torch.cuda.synchronize()
+    # golden: full causal attention, slice my rows
+    ref = F.scaled_dot_product_attention(
+        q_full.permute(1, 0, 2).unsqueeze(0).float(),
+        k_full.permute(1, 0, 2).unsqueeze(0).float(),
+        v_full.permute(1, 0, 2).unsqueeze(0).float(),
+        is_causal=True, enable_gqa=True)
+    ref = ref.squeeze(0).permute(1, 0, 2)[rank * s_loc:(rank + 1) * s_loc]
+    assert_allclose(out, ref, atol=5e-2, rtol=5e-2)
+
+
+def test_sp_ag_attention_gpu_2rank():
+    run_distributed(_body_ag_attn, world_size=2)
+
+
+def _body_p2p(rank, world):
+    from triton_dist_amd.ops import create_p2p_context, p2p_recv, p2p_send
+
+    ctx = create_p2p_context(max_bytes=1 << 20, depth=2)
+    nxt, prv = (rank + 1) % world, (rank - 1) % world
+    for i in range(6):
+        x = torch.full((4096,), float(rank * 10 + i), device="cuda"
+                       ).to(torch.bfloat16)
+        p2p_send(x, nxt, ctx)
+        out = torch.empty(4096, dtype=torch.bfloat16, device="cuda")
+        p2p_recv(out, prv, ctx)
+        torch.cuda.synchronize()
+        assert (out.float() == prv * 10 + i).all(), (rank, i)
+
+
+def test_p2p_ring_gpu_2rank():
+    run_distributed(_body_p2p, world_size=2)
