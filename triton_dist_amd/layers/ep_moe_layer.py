@@ -1,0 +1,96 @@
+"""Expert-parallel MoE layer: router (replicated) + EP dispatch/grouped
+FFN/combine over the symmetric heap.
+
+Capability parity with the reference's EP layers (Triton-distributed
+layers/amd/ep_a2a_layer.py:208-548 EPAll2AllLayer, ep_a2a_fused_layer.py,
+nvidia/ep_moe.py:70-247 — behavior only). Softmax top-k routing with
+optional renormalization (Qwen3-MoE style).
+"""
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+from ..ops.ep_moe import EPContext, create_ep_context, ep_moe_forward
+from ..runtime.symm_mem import SymmHeap, get_heap
+
+
+class EPMoELayer:
+    def __init__(self, hidden: int, moe_inter: int, n_experts: int,
+                 topk: int, norm_topk: bool = True,
+                 heap: Optional[SymmHeap] = None, device="cpu",
+                 dtype=torch.bfloat16):
+        self.heap = heap or get_heap()
+        self.world, self.rank = self.heap.world, self.heap.rank
+        assert n_experts % self.world == 0
+        self.hidden, self.inter = hidden, moe_inter
+        self.n_experts, self.topk = n_experts, topk
+        self.e_loc = n_experts // self.world
+        self.norm_topk = norm_topk
+        self.device, self.dtype = device, dtype
+        self.router = torch.empty(n_experts, hidden, device=device,
+                                  dtype=dtype)
+        self.w_gate_up = torch.empty(self.e_loc, 2 * moe_inter, hidden,
+                                     device=device, dtype=dtype)
+        self.w_down = torch.empty(self.e_loc, hidden, moe_inter,
+                                  device=device, dtype=dtype)
+        self.ctx: Optional[EPContext] = None
+
+    def init_ctx(self, max_tokens: int, ctx: Optional[EPContext] = None):
+        if ctx is None:
+            ctx = create_ep_context(max_tokens, self.hidden, self.n_experts,
+                                    self.topk, heap=self.heap)
+        self.ctx = ctx
+        return ctx
+
+    def route(self, x: torch.Tensor):
+        """softmax top-k router; returns (topk_ids int32, topk_w fp32)."""
+        logits = (x.float() @ self.router.float().t())
+        probs = torch.softmax(logits, dim=-1)
+        topk_w, topk_ids = torch.topk(probs, self.topk, dim=-1)
+        if self.norm_topk:
+            topk_w = topk_w / topk_w.sum(-1, keepdim=True)
+        return topk_ids.to(torch.int32).contiguous(), \
+            topk_w.float().contiguous()
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        """x: [T_local, H] (this rank's token shard) -> [T_local, H]."""
+        topk_ids, topk_w = self.route(x)
+        return ep_moe_forward(x, topk_ids, topk_w, self.w_gate_up,
+                              self.w_down, self.ctx)
+
+    __call__ = forward
+
+    def torch_fwd(self, x: torch.Tensor) -> torch.Tensor:
+        """Golden reference for REPLICATED x [M, H]: each rank computes its
+        local experts' weighted contribution for every token, then
+        all-reduces the partial sums."""
+        import torch.nn.functional as F
+
+        topk_ids, topk_w = self.route(x)
+        m = x.shape[0]
+        acc = torch.zeros(m, self.hidden, dtype=torch.float32,
+                          device=x.device)
+        lo, hi = self.rank * self.e_loc, (self.rank + 1) * self.e_loc
+        for le in range(self.e_loc):
+            e = lo + le
+            sel = (topk_ids == e)
+            if not sel.any():
+                continue
+            tok, kk = sel.nonzero(as_tuple=True)
+            xe = x[tok].float()
+            h = xe @ self.w_gate_up[le].float().t()
+            a = F.silu(h[:, :self.inter]) * h[:, self.inter:]
+            y = a @ self.w_down[le].float().t()
+            acc.index_add_(0, tok, y * topk_w[tok, kk].unsqueeze(1))
+        out = acc.to(self.dtype)
+        if dist.is_initialized() and self.world > 1:
+            if out.is_cuda and dist.get_backend() == "gloo":
+                cpu = out.cpu()
+                dist.all_reduce(cpu)
+                out = cpu.to(out.device)
+            else:
+                dist.all_reduce(out)
+        return out

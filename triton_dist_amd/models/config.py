@@ -23,7 +23,11 @@ class ModelConfig:
     tie_embeddings: bool = False
     max_length: int = 2048
     dtype: str = "bfloat16"
-    tp_mode: str = "ag_rs"  # ag_rs | allreduce | torch
+    tp_mode: str = "ag_rs"  # ag_rs | gemm_ar | allreduce | torch
+    # MoE fields (n_experts == 0 -> dense)
+    n_experts: int = 0
+    moe_topk: int = 8
+    moe_inter: int = 0
 
 
 PRESETS = {
@@ -40,6 +44,19 @@ PRESETS = {
     # at TP in {1, 2}
     "tiny-gpu": dict(hidden=512, intermediate=1024, n_layers=2, n_heads=4,
                      n_kv_heads=2, head_dim=128, vocab=1024),
+    # Qwen3-30B-A3B geometry (HF config)
+    "qwen3-30b-a3b": dict(hidden=2048, intermediate=6144, n_layers=48,
+                          n_heads=32, n_kv_heads=4, head_dim=128,
+                          vocab=151936, n_experts=128, moe_topk=8,
+                          moe_inter=768),
+    # tiny MoE for CPU tests
+    "tiny-moe": dict(hidden=64, intermediate=128, n_layers=2, n_heads=2,
+                     n_kv_heads=2, head_dim=32, vocab=256, n_experts=4,
+                     moe_topk=2, moe_inter=32),
+    # tiny MoE for GPU tests (128/64-tileable expert GEMMs)
+    "tiny-moe-gpu": dict(hidden=512, intermediate=1024, n_layers=2,
+                         n_heads=4, n_kv_heads=2, head_dim=128, vocab=1024,
+                         n_experts=4, moe_topk=2, moe_inter=128),
 }
 
 

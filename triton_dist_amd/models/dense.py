@@ -80,13 +80,15 @@ class DenseLLM:
                 wv[r * kvh * d:(r + 1) * kvh * d]]))
             wo = full((cfg.hidden, cfg.n_heads * d), s + 3)  # K-shard
             attn.w_o.copy_(wo[:, r * qh * d:(r + 1) * qh * d].contiguous())
-            i_s = mlp.inter_shard
-            wg = full((cfg.intermediate, cfg.hidden), s + 4)
-            wu = full((cfg.intermediate, cfg.hidden), s + 5)
-            mlp.w_gate_up.copy_(torch.cat([wg[r * i_s:(r + 1) * i_s],
-                                           wu[r * i_s:(r + 1) * i_s]]))
-            wd = full((cfg.hidden, cfg.intermediate), s + 6)
-            mlp.w_down.copy_(wd[:, r * i_s:(r + 1) * i_s].contiguous())
+            if hasattr(mlp, "inter_shard"):  # dense TP_MLP (MoE layers
+                # initialize their own expert weights in the subclass)
+                i_s = mlp.inter_shard
+                wg = full((cfg.intermediate, cfg.hidden), s + 4)
+                wu = full((cfg.intermediate, cfg.hidden), s + 5)
+                mlp.w_gate_up.copy_(torch.cat([wg[r * i_s:(r + 1) * i_s],
+                                               wu[r * i_s:(r + 1) * i_s]]))
+                wd = full((cfg.hidden, cfg.intermediate), s + 6)
+                mlp.w_down.copy_(wd[:, r * i_s:(r + 1) * i_s].contiguous())
 
     # ------------------------------------------------------------- contexts
     def init_dist_ctx(self, max_m_total: int):
@@ -96,14 +98,16 @@ class DenseLLM:
             ar0 = None
             for layer in self.layers:
                 ar0, _ = layer["attn"].init_ctx(max_m_total, ar0)
-                layer["mlp"].init_ctx(max_m_total, ar0)
+                if hasattr(layer["mlp"], "inter_shard"):
+                    layer["mlp"].init_ctx(max_m_total, ar0)
             return
         if self.mode != "ag_rs":
             return
         ag0 = rs0 = None
         for layer in self.layers:
             ag0, rs0 = layer["attn"].init_ctx(max_m_total, ag0, rs0)
-            layer["mlp"].init_ctx(max_m_total, ag0, rs0)
+            if hasattr(layer["mlp"], "inter_shard"):  # dense TP_MLP
+                layer["mlp"].init_ctx(max_m_total, ag0, rs0)
         self._decode_ag_ctx = ag0
 
     # -------------------------------------------------------------- forward
