@@ -1,0 +1,40 @@
+"""Autotuner: cache persistence + distributed-consistent choice (CPU)."""
+import os
+import time
+
+import torch
+
+
+def test_autotuner_cache(tmp_path, monkeypatch):
+    monkeypatch.setenv("TD_AUTOTUNE_DIR", str(tmp_path))
+    from triton_dist_amd.tune import AutoTuner
+
+    calls = {"n": 0}
+
+    def bench(cfg):
+        def run():
+            calls["n"] += 1
+            time.sleep(0.001 * cfg["x"])
+        return run
+
+    t = AutoTuner("unit", [{"x": 3}, {"x": 1}, {"x": 2}], warmup=1, iters=2)
+    best = t.tune("k1", bench)
+    assert best == {"x": 1}
+    n_after_first = calls["n"]
+    t2 = AutoTuner("unit", [{"x": 3}, {"x": 1}, {"x": 2}], warmup=1, iters=2)
+    best2 = t2.tune("k1", bench)
+    assert best2 == {"x": 1}
+    assert calls["n"] == n_after_first  # served from the JSON cache
+
+
+def test_autotune_decorator(tmp_path, monkeypatch):
+    monkeypatch.setenv("TD_AUTOTUNE_DIR", str(tmp_path))
+    from triton_dist_amd.tune import autotune
+
+    @autotune("deco", [{"s": 5}, {"s": 1}], key=lambda n, **kw: str(n))
+    def op(n, tune_config=None):
+        time.sleep(0.0005 * tune_config["s"])
+        return tune_config["s"]
+
+    assert op(7) == 1
+    assert op(7) == 1

@@ -4,6 +4,7 @@
 #include <stdint.h>
 
 #include "td/device.hpp"
+#include "td/profiler.hpp"
 
 namespace td {
 
@@ -55,6 +56,7 @@ struct AgGemmArgs {
   int world;
   int rank;
   int expect;           // flag value that means "ready"
+  KProf prof{};         // optional intra-kernel profiler (buf == nullptr: off)
 };
 void launch_ag_gemm_consumer_bf16(const AgGemmArgs &args, hipStream_t stream);
 

@@ -256,11 +256,14 @@ __global__ __launch_bounds__(NTHREADS) void k_ag_gemm_consumer_bf16(
   int rows_per_chunk = args.m_per_rank / args.chunks_per_rank;
   int c_lo = (pid_m * BM) / rows_per_chunk;
   int c_hi = (pid_m * BM + BM - 1) / rows_per_chunk;
+  unsigned long long tw0 = wallclock();
   if (threadIdx.x < 64) {
     for (int c = c_lo + (int)threadIdx.x; c <= c_hi; c += 64)
       wait_ge_one<Scope::Sys>(args.flags + c, args.expect);
   }
   __syncthreads();
+  unsigned long long tw1 = wallclock();
+  kprof_record(args.prof, 0 /*tile-wait*/, tw0, tw1);
 
   TileCtx t = tile_ctx();
   f32x4 acc[4][4] = {};
@@ -275,6 +278,7 @@ __global__ __launch_bounds__(NTHREADS) void k_ag_gemm_consumer_bf16(
   }
   bf16 *dst = (bf16 *)g.c + (size_t)pid_m * BM * g.ldc + pid_n * BN;
   epilogue_store(acc, t, lds_a, dst, g.ldc, nullptr, 0);
+  kprof_record(args.prof, 1 /*tile-compute*/, tw1, wallclock());
 }
 
 void launch_ag_gemm_consumer_bf16(const AgGemmArgs &args, hipStream_t stream) {

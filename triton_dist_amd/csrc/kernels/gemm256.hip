@@ -215,11 +215,14 @@ __global__ __launch_bounds__(NTH, 2) void k_ag_gemm256_consumer_bf16(
   int rows_per_chunk = args.m_per_rank / args.chunks_per_rank;
   int c_lo = (pid_m * BM) / rows_per_chunk;
   int c_hi = (pid_m * BM + BM - 1) / rows_per_chunk;
+  unsigned long long tw0 = wallclock();
   if (threadIdx.x < 64) {
     for (int c = c_lo + (int)threadIdx.x; c <= c_hi; c += 64)
       wait_ge_one<Scope::Sys>(args.flags + c, args.expect);
   }
   __syncthreads();
+  unsigned long long tw1 = wallclock();
+  kprof_record(args.prof, 0, tw0, tw1);
   WaveCtx w = wave_ctx();
   f32x4 acc[8][4] = {};
   const bf16 *ga = (const bf16 *)g.a + (size_t)pid_m * BM * g.lda;
@@ -227,6 +230,7 @@ __global__ __launch_bounds__(NTH, 2) void k_ag_gemm256_consumer_bf16(
   kloop(ga, gb, g.lda, g.ldb, g.k / BK, lds_a, lds_b, w, acc);
   bf16 *dst = (bf16 *)g.c + (size_t)pid_m * BM * g.ldc + pid_n * BN;
   epilogue(acc, w, lds_a, dst, g.ldc);
+  kprof_record(args.prof, 1, tw1, wallclock());
 }
 
 __global__ __launch_bounds__(NTH, 2) void k_gemm256_rs_producer_bf16(

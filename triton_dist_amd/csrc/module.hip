@@ -282,7 +282,9 @@ static void ag_gemm_consumer_bf16(uintptr_t a, uintptr_t b, uintptr_t c,
                                   int m, int n, int k, uintptr_t flags,
                                   int chunks_per_rank, int m_per_rank,
                                   int world, int rank, int expect,
-                                  uintptr_t stream) {
+                                  uintptr_t stream, uintptr_t prof_buf = 0,
+                                  uintptr_t prof_cursor = 0,
+                                  unsigned prof_cap = 0) {
   AgGemmArgs args;
   args.g = GemmArgs{reinterpret_cast<void *>(a), reinterpret_cast<void *>(b),
                     reinterpret_cast<void *>(c), nullptr, m, n, k, k, k, n};
@@ -292,6 +294,8 @@ static void ag_gemm_consumer_bf16(uintptr_t a, uintptr_t b, uintptr_t c,
   args.world = world;
   args.rank = rank;
   args.expect = expect;
+  args.prof = KProf{(unsigned long long *)prof_buf, (unsigned *)prof_cursor,
+                    prof_cap};
   launch_ag_gemm_consumer_bf16(args, as_stream(stream));
   TD_CHECK_HIP(hipGetLastError());
 }
@@ -550,7 +554,12 @@ PYBIND11_MODULE(_C, m) {
   m.def("put_signal", &put_signal);
   m.def("probe_mfma", &probe_mfma);
   m.def("gemm_bf16", &gemm_bf16);
-  m.def("ag_gemm_consumer_bf16", &ag_gemm_consumer_bf16);
+  m.def("ag_gemm_consumer_bf16", &ag_gemm_consumer_bf16, py::arg("a"),
+        py::arg("b"), py::arg("c"), py::arg("m"), py::arg("n"), py::arg("k"),
+        py::arg("flags"), py::arg("chunks_per_rank"), py::arg("m_per_rank"),
+        py::arg("world"), py::arg("rank"), py::arg("expect"),
+        py::arg("stream"), py::arg("prof_buf") = 0,
+        py::arg("prof_cursor") = 0, py::arg("prof_cap") = 0);
   m.def("gemm_rs_producer_bf16", &gemm_rs_producer_bf16);
   m.def("rs_reduce_bf16", &rs_reduce_bf16);
   m.def("gemm_splitk_bf16", &gemm_splitk_bf16);

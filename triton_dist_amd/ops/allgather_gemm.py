@@ -77,7 +77,7 @@ def create_ag_gemm_context(max_m_per_rank: int, k: int,
 
 def ag_gemm(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
             out: Optional[torch.Tensor] = None,
-            gathered_out: bool = False):
+            gathered_out: bool = False, profiler=None):
     """C[world*m, N] = AllGather(A[m, K]) @ W[N, K]^T.
 
     Returns C (and optionally the gathered A view for reuse).
@@ -152,10 +152,11 @@ def ag_gemm(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
             m_total, n, k, ctx.flags.ptr(), chunks, m, world, rank, 1,
             splits, compute.cuda_stream)
     else:
+        pb, pc, pcap = profiler.ptrs() if profiler is not None else (0, 0, 0)
         _C.ag_gemm_consumer_bf16(
             ctx.ws.ptr(), w.data_ptr(), out.data_ptr(), m_total, n, k,
             ctx.flags.ptr(), chunks, m, world, rank, 1,
-            compute.cuda_stream)
+            compute.cuda_stream, pb, pc, pcap)
     # join comm streams back into the compute stream (after the consumer
     # launch: no serialization, but graph capture requires joined forks)
     for s in range(min(ns, max(world - 1, 1))):

@@ -1,0 +1,135 @@
+"""Persistent, distributed-aware autotuner.
+
+Capability parity with Triton-distributed python/triton_dist/tune.py:280-426
+(AutoTuner: config space x key_fn, benchmarking under a distributed barrier,
+JSON cache under ~/.triton_dist/autotune/<hw_hash>/) and autotuner.py:43-105
+(contextual tuning: benchmark the WHOLE op including producer/consumer).
+
+All ranks benchmark together (barrier-aligned), timings are MAX-reduced
+across ranks, and rank 0's argmin is broadcast so every rank picks the same
+config — a per-rank argmin would deadlock ops whose producers and consumers
+must agree on chunking.
+"""
+from __future__ import annotations
+
+import functools
+import hashlib
+import json
+import os
+import time
+from pathlib import Path
+from typing import Any, Callable, Dict, List, Optional
+
+import torch
+import torch.distributed as dist
+
+
+def hardware_hash() -> str:
+    parts = [torch.__version__]
+    if torch.cuda.is_available():
+        p = torch.cuda.get_device_properties(0)
+        parts += [p.gcnArchName, str(p.multi_processor_count),
+                  str(torch.cuda.device_count())]
+    else:
+        parts += ["cpu"]
+    if dist.is_initialized():
+        parts.append(f"w{dist.get_world_size()}")
+    return hashlib.sha1("|".join(parts).encode()).hexdigest()[:12]
+
+
+def cache_dir() -> Path:
+    root = os.environ.get("TD_AUTOTUNE_DIR",
+                          os.path.expanduser("~/.triton_dist_amd/autotune"))
+    d = Path(root) / hardware_hash()
+    d.mkdir(parents=True, exist_ok=True)
+    return d
+
+
+class AutoTuner:
+    def __init__(self, name: str, configs: List[Dict[str, Any]],
+                 warmup: int = 3, iters: int = 10):
+        self.name = name
+        self.configs = configs
+        self.warmup = warmup
+        self.iters = iters
+        self._mem: Dict[str, Dict[str, Any]] = {}
+        self._path = cache_dir() / f"{name}.json"
+        self.always_tune = os.environ.get("TD_AUTOTUNE_ALWAYS_TUNE") == "1"
+        if self._path.exists() and not self.always_tune:
+            try:
+                self._mem = json.loads(self._path.read_text())
+            except Exception:
+                self._mem = {}
+
+    def _persist(self):
+        if dist.is_initialized() and dist.get_rank() != 0:
+            return
+        tmp = self._path.with_suffix(".tmp")
+        tmp.write_text(json.dumps(self._mem, indent=1, sort_keys=True))
+        tmp.replace(self._path)
+
+    def _time_one(self, fn: Callable[[], None]) -> float:
+        for _ in range(self.warmup):
+            fn()
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
+        if dist.is_initialized():
+            dist.barrier()
+        t0 = time.perf_counter()
+        for _ in range(self.iters):
+            fn()
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
+        ms = (time.perf_counter() - t0) * 1e3 / self.iters
+        if dist.is_initialized():
+            t = torch.tensor([ms])
+            dist.all_reduce(t, op=dist.ReduceOp.MAX)
+            ms = float(t.item())
+        return ms
+
+    def tune(self, key: str, bench_fn: Callable[[Dict[str, Any]], Callable],
+             ) -> Dict[str, Any]:
+        """bench_fn(config) -> zero-arg callable running the op once.
+        Returns the best config (cached)."""
+        if key in self._mem and not self.always_tune:
+            return self._mem[key]["config"]
+        results = []
+        for cfg in self.configs:
+            try:
+                run = bench_fn(cfg)
+                ms = self._time_one(run)
+            except Exception:
+                ms = float("inf")
+            results.append(ms)
+        # rank 0 decides; broadcast so all ranks agree
+        best_idx = int(min(range(len(results)), key=lambda i: results[i]))
+        if dist.is_initialized():
+            t = torch.tensor([best_idx], dtype=torch.int64)
+            dist.broadcast(t, src=0)
+            best_idx = int(t.item())
+        best = self.configs[best_idx]
+        self._mem[key] = {"config": best, "ms": results[best_idx],
+                          "all_ms": results}
+        self._persist()
+        return best
+
+
+def autotune(name: str, configs: List[Dict[str, Any]],
+             key: Callable[..., str]):
+    """Decorator: tunes over `configs`, passing the chosen one as
+    `tune_config=` kwarg; `key(*args, **kwargs)` maps a call to a cache key.
+    """
+    tuner = AutoTuner(name, configs)
+
+    def deco(fn):
+        @functools.wraps(fn)
+        def wrapped(*args, **kwargs):
+            k = key(*args, **kwargs)
+            cfg = tuner.tune(
+                k, lambda c: (lambda: fn(*args, tune_config=c, **kwargs)))
+            return fn(*args, tune_config=cfg, **kwargs)
+
+        wrapped.tuner = tuner
+        return wrapped
+
+    return deco
