@@ -95,12 +95,21 @@ TD_DEV WaveCtx wave_ctx() {
 }
 
 // The full pipelined K loop over `ktiles` tiles of BK.
+//
+// Staging is shifted one phase late relative to consumption: phase p of
+// tile t stages INTO region p-1 (freed when phase p-1's MFMAs consumed it
+// — each wave's lgkmcnt before its MFMA plus the barrier at phase-p entry
+// prove all reads retired). Phase 0 stages region 3 with the CURRENT
+// tile's last slice. One barrier per phase, no LDS-read drain before a
+// second barrier; ds_read latency hides under the stage issue + barrier.
+// Issue ledger: 4 loads/thread/phase, 12 in flight, vmcnt(8) steady
+// (drain 8/8/4/0 on the last tile).
 TD_DEV void kloop(const bf16 *ga, const bf16 *gb, int lda, int ldb,
                   int ktiles, bf16 *lds_a, bf16 *lds_b, const WaveCtx &w,
                   f32x4 acc[8][4]) {
-  // prologue: stage all 4 slices of tile 0
+  // prologue: stage slices 0..2 of tile 0 (slice 3 staged in phase 0)
 #pragma unroll
-  for (int p = 0; p < SLICES; ++p)
+  for (int p = 0; p < SLICES - 1; ++p)
     stage_slice(ga, gb, lda, ldb, p * SLICE_K, lds_a, lds_b, p);
 
   for (int t = 0; t < ktiles; ++t) {
@@ -109,17 +118,14 @@ TD_DEV void kloop(const bf16 *ga, const bf16 *gb, int lda, int ldb,
 #pragma unroll
     for (int p = 0; p < SLICES; ++p) {
       // retire the loads for slice p of tile t
-      if (has_next) {
-        asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
+      if (has_next || p <= 1) {
+        asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
       } else {
-        if (p == 0) asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
-        if (p == 1) asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
         if (p == 2) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
         if (p == 3) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       }
-      __builtin_amdgcn_s_barrier();  // slice p landed for ALL waves
+      __builtin_amdgcn_s_barrier();  // slice p landed; region p-1 free
       __builtin_amdgcn_sched_barrier(0);
-      // read fragments now; the barrier AFTER the reads frees the region
       const int jn = w.lane >> 4;
       bf16x8 af[8], bf[4];
 #pragma unroll
@@ -134,13 +140,16 @@ TD_DEV void kloop(const bf16 *ga, const bf16 *gb, int lda, int ldb,
         bf[j] = *(const bf16x8 *)(lds_b + p * SLICE_ELEMS + row * SLICE_K +
                                   swz(row, jn) * 8);
       }
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      __builtin_amdgcn_sched_barrier(0);
-      __builtin_amdgcn_s_barrier();  // all waves done READING slice p
-      __builtin_amdgcn_sched_barrier(0);
-      if (has_next)
-        stage_slice(ga + knext, gb + knext, lda, ldb, p * SLICE_K, lds_a,
-                    lds_b, p);
+      if (p == 0) {
+        stage_slice(ga + t * BK, gb + t * BK, lda, ldb, 3 * SLICE_K, lds_a,
+                    lds_b, 3);
+      } else if (has_next) {
+        stage_slice(ga + knext, gb + knext, lda, ldb, (p - 1) * SLICE_K,
+                    lds_a, lds_b, p - 1);
+      }
+      // no explicit lgkmcnt: the ds_reads are compiler-visible loads, so
+      // the backend inserts fine-grained lgkmcnt(N) per dependent MFMA
+      // (guide: explicit lgkmcnt(0) serializes the whole read burst)
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int i = 0; i < 8; ++i)
