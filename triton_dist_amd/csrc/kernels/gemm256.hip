@@ -126,6 +126,13 @@ TD_DEV void kloop(const bf16 *ga, const bf16 *gb, int lda, int ldb,
       }
       __builtin_amdgcn_s_barrier();  // slice p landed; region p-1 free
       __builtin_amdgcn_sched_barrier(0);
+      if (p == 0) {
+        stage_slice(ga + t * BK, gb + t * BK, lda, ldb, 3 * SLICE_K, lds_a,
+                    lds_b, 3);
+      } else if (has_next) {
+        stage_slice(ga + knext, gb + knext, lda, ldb, (p - 1) * SLICE_K,
+                    lds_a, lds_b, p - 1);
+      }
       const int jn = w.lane >> 4;
       bf16x8 af[8], bf[4];
 #pragma unroll
@@ -139,13 +146,6 @@ TD_DEV void kloop(const bf16 *ga, const bf16 *gb, int lda, int ldb,
         int row = w.wc * 64 + j * 16 + (w.lane & 15);
         bf[j] = *(const bf16x8 *)(lds_b + p * SLICE_ELEMS + row * SLICE_K +
                                   swz(row, jn) * 8);
-      }
-      if (p == 0) {
-        stage_slice(ga + t * BK, gb + t * BK, lda, ldb, 3 * SLICE_K, lds_a,
-                    lds_b, 3);
-      } else if (has_next) {
-        stage_slice(ga + knext, gb + knext, lda, ldb, (p - 1) * SLICE_K,
-                    lds_a, lds_b, p - 1);
       }
       // no explicit lgkmcnt: the ds_reads are compiler-visible loads, so
       // the backend inserts fine-grained lgkmcnt(N) per dependent MFMA
