@@ -107,9 +107,15 @@ void launch_moe_dispatch(const PeerTable &pt, const void *x,
                          const void *topk_ids, const void *send_pos,
                          const void *send_base, const void *send_to_dst,
                          size_t recv_x_off, size_t meta_off, size_t flags_off,
-                         unsigned *arrive, int T, int K, int H, int e_loc,
-                         hipStream_t stream);
-void launch_moe_wait_flags(const void *flags, int world, hipStream_t stream);
+                         unsigned *arrive, const void *val_cell, int T,
+                         int K, int H, int e_loc, hipStream_t stream);
+void launch_moe_wait_flags(const void *flags, int world, const void *cell,
+                           hipStream_t stream);
+void launch_bump_cell(void *cell, hipStream_t stream);
+void launch_wait_flags_ge_cell(const void *flags, int n, const void *cell,
+                               int delta, hipStream_t stream);
+void launch_signal_credit(const PeerTable &pt, size_t credit_off,
+                          const void *cell, hipStream_t stream);
 void launch_moe_grouped_gemm(const void *xin, const void *weights, void *out,
                              const void *expert_base, const void *expert_rows,
                              int e_loc, int cap_tiles_m, int n, int k,
@@ -117,11 +123,13 @@ void launch_moe_grouped_gemm(const void *xin, const void *weights, void *out,
 void launch_moe_combine_send(const PeerTable &pt, const void *expert_out,
                              const void *meta, const void *recv_total,
                              const void *recv_from_src, size_t combine_off,
-                             size_t cflags_off, unsigned *arrive, int cap,
-                             int H, hipStream_t stream);
+                             size_t cflags_off, unsigned *arrive,
+                             const void *val_cell, int cap, int H,
+                             hipStream_t stream);
 void launch_moe_combine_reduce(const void *combine_buf, const void *topk_w,
                                const void *topk_ids, void *out,
-                               const void *cflags, int world, int T, int K,
+                               const void *cflags, const void *val_cell,
+                               int world, int T, int K,
                                int H, int e_num, hipStream_t stream);
 
 // kernels/elementwise.hip ----------------------------------------------------

@@ -113,8 +113,9 @@ __global__ void k_moe_dispatch(PeerTable pt, const bf16 *__restrict__ x,
                                const int *__restrict__ send_base,
                                const int *__restrict__ send_to_dst,
                                size_t recv_x_off, size_t meta_off,
-                               size_t flags_off, unsigned *arrive, int T,
-                               int K, int H, int e_loc) {
+                               size_t flags_off, unsigned *arrive,
+                               const int *val_cell, int T, int K, int H,
+                               int e_loc) {
   const int i = blockIdx.x;  // copy index t*K+k
   const int e = topk_ids[i];
   const int pos = send_pos[i];
@@ -135,7 +136,7 @@ __global__ void k_moe_dispatch(PeerTable pt, const bf16 *__restrict__ x,
     unsigned prev = atomic_add<Scope::Gpu>(arrive + dst, 1u);
     if ((int)prev == send_to_dst[dst] - 1) {
       int *fl = (int *)((char *)pt.bases[dst] + flags_off);
-      st_release<Scope::Sys>(fl + pt.rank, 1);
+      st_release<Scope::Sys>(fl + pt.rank, val_cell ? *val_cell : 1);
     }
   }
 }
@@ -143,13 +144,36 @@ __global__ void k_moe_dispatch(PeerTable pt, const bf16 *__restrict__ x,
 // signal destinations that get ZERO copies from me (they still wait on my
 // flag), and likewise for the combine side.
 __global__ void k_moe_signal_empty(PeerTable pt, const int *__restrict__ cnt,
-                                   size_t flags_off) {
+                                   size_t flags_off,
+                                   const int *__restrict__ val_cell) {
   int d = threadIdx.x;
   if (d >= pt.world) return;
   if (cnt[d] == 0) {
     int *fl = (int *)((char *)pt.bases[d] + flags_off);
-    st_release<Scope::Sys>(fl + pt.rank, 1);
+    st_release<Scope::Sys>(fl + pt.rank, val_cell ? *val_cell : 1);
   }
+}
+
+// LL support: device call counter bump; value-from-cell waits; credits.
+__global__ void k_bump_cell(int *cell) {
+  if (threadIdx.x == 0) *cell = *cell + 1;
+}
+
+__global__ void k_wait_flags_ge_cell(const int *flags, int n,
+                                     const int *cell, int delta) {
+  int bound = *cell + delta;
+  if (bound <= 0) return;
+  if (threadIdx.x < (unsigned)n)
+    wait_ge_one<Scope::Sys>(flags + threadIdx.x, bound);
+  __syncthreads();
+}
+
+__global__ void k_signal_credit(PeerTable pt, size_t credit_off,
+                                const int *cell) {
+  int d = threadIdx.x;
+  if (d >= pt.world) return;
+  int *fl = (int *)((char *)pt.bases[d] + credit_off);
+  st_release<Scope::Sys>(fl + pt.rank, *cell);
 }
 
 // wait for all world dispatch flags (prefix kernel before the expert GEMM)
@@ -262,7 +286,8 @@ __global__ void k_moe_combine_send(PeerTable pt,
                                    const int *__restrict__ recv_total,
                                    const int *__restrict__ recv_from_src,
                                    size_t combine_off, size_t cflags_off,
-                                   unsigned *arrive, int H) {
+                                   unsigned *arrive, const int *val_cell,
+                                   int H) {
   const int r = blockIdx.x;
   if (r >= recv_total[0]) return;
   const int src = meta[r * 2];
@@ -277,7 +302,7 @@ __global__ void k_moe_combine_send(PeerTable pt,
     unsigned prev = atomic_add<Scope::Gpu>(arrive + src, 1u);
     if ((int)prev == recv_from_src[src] - 1) {
       int *fl = (int *)((char *)pt.bases[src] + cflags_off);
-      st_release<Scope::Sys>(fl + pt.rank, 1);
+      st_release<Scope::Sys>(fl + pt.rank, val_cell ? *val_cell : 1);
     }
   }
 }
@@ -288,10 +313,12 @@ __global__ void k_moe_combine_reduce(const bf16 *__restrict__ combine_buf,
                                      const float *__restrict__ topk_w,
                                      const int *__restrict__ topk_ids,
                                      bf16 *__restrict__ out,
-                                     const int *cflags, int world, int T,
+                                     const int *cflags,
+                                     const int *__restrict__ val_cell,
+                                     int world, int T,
                                      int K, int H, int e_num) {
   if (threadIdx.x < (unsigned)world)
-    wait_ge_one<Scope::Sys>(cflags + threadIdx.x, 1);
+    wait_ge_one<Scope::Sys>(cflags + threadIdx.x, val_cell ? *val_cell : 1);
   __syncthreads();
   const int t = blockIdx.x;
   if (t >= T) return;
@@ -342,21 +369,46 @@ void launch_moe_dispatch(const PeerTable &pt, const void *x,
                          const void *topk_ids, const void *send_pos,
                          const void *send_base, const void *send_to_dst,
                          size_t recv_x_off, size_t meta_off, size_t flags_off,
-                         unsigned *arrive, int T, int K, int H, int e_loc,
-                         hipStream_t stream) {
+                         unsigned *arrive, const void *val_cell, int T,
+                         int K, int H, int e_loc, hipStream_t stream) {
   if (H % 8) throw std::runtime_error("moe dispatch: H % 8 != 0");
   hipLaunchKernelGGL(k_moe_dispatch, dim3(T * K), dim3(256), 0, stream, pt,
                      (const bf16 *)x, (const int *)topk_ids,
                      (const int *)send_pos, (const int *)send_base,
                      (const int *)send_to_dst, recv_x_off, meta_off,
-                     flags_off, arrive, T, K, H, e_loc);
+                     flags_off, arrive, (const int *)val_cell, T, K, H,
+                     e_loc);
   hipLaunchKernelGGL(k_moe_signal_empty, dim3(1), dim3(kWave), 0, stream, pt,
-                     (const int *)send_to_dst, flags_off);
+                     (const int *)send_to_dst, flags_off,
+                     (const int *)val_cell);
 }
 
-void launch_moe_wait_flags(const void *flags, int world, hipStream_t stream) {
-  hipLaunchKernelGGL(k_moe_wait_flags, dim3(1), dim3(kWave), 0, stream,
-                     (const int *)flags, world);
+void launch_bump_cell(void *cell, hipStream_t stream) {
+  hipLaunchKernelGGL(k_bump_cell, dim3(1), dim3(1), 0, stream, (int *)cell);
+}
+
+void launch_wait_flags_ge_cell(const void *flags, int n, const void *cell,
+                               int delta, hipStream_t stream) {
+  hipLaunchKernelGGL(k_wait_flags_ge_cell, dim3(1), dim3(kWave), 0, stream,
+                     (const int *)flags, n, (const int *)cell, delta);
+}
+
+void launch_signal_credit(const PeerTable &pt, size_t credit_off,
+                          const void *cell, hipStream_t stream) {
+  hipLaunchKernelGGL(k_signal_credit, dim3(1), dim3(kWave), 0, stream, pt,
+                     credit_off, (const int *)cell);
+}
+
+void launch_moe_wait_flags(const void *flags, int world, const void *cell,
+                           hipStream_t stream) {
+  if (cell) {
+    hipLaunchKernelGGL(k_wait_flags_ge_cell, dim3(1), dim3(kWave), 0,
+                       stream, (const int *)flags, world, (const int *)cell,
+                       0);
+  } else {
+    hipLaunchKernelGGL(k_moe_wait_flags, dim3(1), dim3(kWave), 0, stream,
+                       (const int *)flags, world);
+  }
 }
 
 void launch_moe_grouped_gemm(const void *xin, const void *weights, void *out,
@@ -375,24 +427,28 @@ void launch_moe_grouped_gemm(const void *xin, const void *weights, void *out,
 void launch_moe_combine_send(const PeerTable &pt, const void *expert_out,
                              const void *meta, const void *recv_total,
                              const void *recv_from_src, size_t combine_off,
-                             size_t cflags_off, unsigned *arrive, int cap,
-                             int H, hipStream_t stream) {
+                             size_t cflags_off, unsigned *arrive,
+                             const void *val_cell, int cap, int H,
+                             hipStream_t stream) {
   hipLaunchKernelGGL(k_moe_combine_send, dim3(cap), dim3(256), 0, stream, pt,
                      (const bf16 *)expert_out, (const int *)meta,
                      (const int *)recv_total, (const int *)recv_from_src,
-                     combine_off, cflags_off, arrive, H);
+                     combine_off, cflags_off, arrive, (const int *)val_cell,
+                     H);
   hipLaunchKernelGGL(k_moe_signal_empty, dim3(1), dim3(kWave), 0, stream, pt,
-                     (const int *)recv_from_src, cflags_off);
+                     (const int *)recv_from_src, cflags_off,
+                     (const int *)val_cell);
 }
 
 void launch_moe_combine_reduce(const void *combine_buf, const void *topk_w,
                                const void *topk_ids, void *out,
-                               const void *cflags, int world, int T, int K,
+                               const void *cflags, const void *val_cell,
+                               int world, int T, int K,
                                int H, int e_num, hipStream_t stream) {
   hipLaunchKernelGGL(k_moe_combine_reduce, dim3(T), dim3(256), 0, stream,
                      (const bf16 *)combine_buf, (const float *)topk_w,
                      (const int *)topk_ids, (bf16 *)out, (const int *)cflags,
-                     world, T, K, H, e_num);
+                     (const int *)val_cell, world, T, K, H, e_num);
 }
 
 }  // namespace td

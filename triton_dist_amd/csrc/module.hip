@@ -418,17 +418,41 @@ static void moe_dispatch(uintptr_t x, uintptr_t topk_ids, uintptr_t send_pos,
                          uintptr_t send_base, uintptr_t send_to_dst,
                          size_t recv_x_off, size_t meta_off,
                          size_t flags_off, uintptr_t arrive, int T, int K,
-                         int H, int e_loc, uintptr_t stream) {
+                         int H, int e_loc, uintptr_t stream,
+                         uintptr_t val_cell = 0) {
   check_active();
   launch_moe_dispatch(g_heap.pt, (void *)x, (void *)topk_ids,
                       (void *)send_pos, (void *)send_base,
                       (void *)send_to_dst, recv_x_off, meta_off, flags_off,
-                      (unsigned *)arrive, T, K, H, e_loc, as_stream(stream));
+                      (unsigned *)arrive, (void *)val_cell, T, K, H, e_loc,
+                      as_stream(stream));
   TD_CHECK_HIP(hipGetLastError());
 }
 
-static void moe_wait_flags(uintptr_t flags, int world, uintptr_t stream) {
-  launch_moe_wait_flags((void *)flags, world, as_stream(stream));
+static void moe_wait_flags(uintptr_t flags, int world, uintptr_t stream,
+                           uintptr_t cell = 0) {
+  launch_moe_wait_flags((void *)flags, world, (void *)cell,
+                        as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
+static void bump_cell(uintptr_t cell, uintptr_t stream) {
+  launch_bump_cell((void *)cell, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
+static void wait_flags_ge_cell(uintptr_t flags, int n, uintptr_t cell,
+                               int delta, uintptr_t stream) {
+  launch_wait_flags_ge_cell((void *)flags, n, (void *)cell, delta,
+                            as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
+static void signal_credit(uintptr_t credit_off, uintptr_t cell,
+                          uintptr_t stream) {
+  check_active();
+  launch_signal_credit(g_heap.pt, (size_t)credit_off, (void *)cell,
+                       as_stream(stream));
   TD_CHECK_HIP(hipGetLastError());
 }
 
@@ -446,22 +470,24 @@ static void moe_combine_send(uintptr_t expert_out, uintptr_t meta,
                              uintptr_t recv_total, uintptr_t recv_from_src,
                              size_t combine_off, size_t cflags_off,
                              uintptr_t arrive, int cap, int H,
-                             uintptr_t stream) {
+                             uintptr_t stream, uintptr_t val_cell = 0) {
   check_active();
   launch_moe_combine_send(g_heap.pt, (void *)expert_out, (void *)meta,
                           (void *)recv_total, (void *)recv_from_src,
-                          combine_off, cflags_off, (unsigned *)arrive, cap,
-                          H, as_stream(stream));
+                          combine_off, cflags_off, (unsigned *)arrive,
+                          (void *)val_cell, cap, H, as_stream(stream));
   TD_CHECK_HIP(hipGetLastError());
 }
 
 static void moe_combine_reduce(uintptr_t combine_buf, uintptr_t topk_w,
                                uintptr_t topk_ids, uintptr_t out,
                                uintptr_t cflags, int world, int T, int K,
-                               int H, int e_num, uintptr_t stream) {
+                               int H, int e_num, uintptr_t stream,
+                               uintptr_t val_cell = 0) {
   launch_moe_combine_reduce((void *)combine_buf, (void *)topk_w,
                             (void *)topk_ids, (void *)out, (void *)cflags,
-                            world, T, K, H, e_num, as_stream(stream));
+                            (void *)val_cell, world, T, K, H, e_num,
+                            as_stream(stream));
   TD_CHECK_HIP(hipGetLastError());
 }
 
@@ -569,11 +595,27 @@ PYBIND11_MODULE(_C, m) {
   m.def("allreduce_twoshot", &allreduce_twoshot);
   m.def("moe_count", &moe_count);
   m.def("moe_layout", &moe_layout);
-  m.def("moe_dispatch", &moe_dispatch);
-  m.def("moe_wait_flags", &moe_wait_flags);
+  m.def("moe_dispatch", &moe_dispatch, py::arg("x"), py::arg("topk_ids"),
+        py::arg("send_pos"), py::arg("send_base"), py::arg("send_to_dst"),
+        py::arg("recv_x_off"), py::arg("meta_off"), py::arg("flags_off"),
+        py::arg("arrive"), py::arg("T"), py::arg("K"), py::arg("H"),
+        py::arg("e_loc"), py::arg("stream"), py::arg("val_cell") = 0);
+  m.def("bump_cell", &bump_cell);
+  m.def("wait_flags_ge_cell", &wait_flags_ge_cell);
+  m.def("signal_credit", &signal_credit);
+  m.def("moe_wait_flags", &moe_wait_flags, py::arg("flags"),
+        py::arg("world"), py::arg("stream"), py::arg("cell") = 0);
   m.def("moe_grouped_gemm", &moe_grouped_gemm);
-  m.def("moe_combine_send", &moe_combine_send);
-  m.def("moe_combine_reduce", &moe_combine_reduce);
+  m.def("moe_combine_send", &moe_combine_send, py::arg("expert_out"),
+        py::arg("meta"), py::arg("recv_total"), py::arg("recv_from_src"),
+        py::arg("combine_off"), py::arg("cflags_off"), py::arg("arrive"),
+        py::arg("cap"), py::arg("H"), py::arg("stream"),
+        py::arg("val_cell") = 0);
+  m.def("moe_combine_reduce", &moe_combine_reduce, py::arg("combine_buf"),
+        py::arg("topk_w"), py::arg("topk_ids"), py::arg("out"),
+        py::arg("cflags"), py::arg("world"), py::arg("T"), py::arg("K"),
+        py::arg("H"), py::arg("e_num"), py::arg("stream"),
+        py::arg("val_cell") = 0);
   m.def("rmsnorm", &rmsnorm);
   m.def("add_rmsnorm", &add_rmsnorm);
   m.def("swiglu", &swiglu);

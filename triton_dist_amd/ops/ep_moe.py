@@ -38,6 +38,9 @@ class EPContext:
     # local device state
     local: dict = field(default_factory=dict)
     epoch: int = 0
+    low_latency: bool = False
+    credit_flags: Optional[SymmBuffer] = None
+    call_no: int = 0
 
     @property
     def world(self):
@@ -53,24 +56,34 @@ class EPContext:
 
 
 def create_ep_context(max_tokens: int, hidden: int, n_experts: int,
-                      topk: int, heap: Optional[SymmHeap] = None
-                      ) -> EPContext:
+                      topk: int, heap: Optional[SymmHeap] = None,
+                      low_latency: bool = False) -> EPContext:
+    """low_latency=True allocates DOUBLE buffers (call-parity indexed) and
+    the op skips the entry barrier + flag resets: flags carry the running
+    call number and a credit flag bounds buffer reuse — the reference's LL
+    protocol (low_latency_all_to_all.py:36-170 act_pos=call%2 — behavior
+    only)."""
     heap = heap or get_heap()
     world = heap.world
     assert n_experts % world == 0
     cap = world * max_tokens * topk
+    nbuf = 2 if low_latency else 1
     ctx = EPContext(
         heap=heap, max_tokens=max_tokens, hidden=hidden,
         n_experts=n_experts, topk=topk, cap=cap,
-        all_splits=heap.alloc_buffer((world, n_experts), torch.int32),
-        splits_flags=heap.alloc_buffer((world,), torch.int32),
-        recv_x=heap.alloc_buffer((cap + 128, hidden), torch.bfloat16),
-        meta=heap.alloc_buffer((cap, 2), torch.int32),
-        combine_buf=heap.alloc_buffer((max_tokens * topk, hidden),
+        all_splits=heap.alloc_buffer((nbuf, world, n_experts), torch.int32),
+        splits_flags=heap.alloc_buffer((nbuf, world), torch.int32),
+        recv_x=heap.alloc_buffer((nbuf, cap + 128, hidden), torch.bfloat16),
+        meta=heap.alloc_buffer((nbuf, cap, 2), torch.int32),
+        combine_buf=heap.alloc_buffer((nbuf, max_tokens * topk, hidden),
                                       torch.bfloat16),
-        disp_flags=heap.alloc_buffer((world,), torch.int32),
-        comb_flags=heap.alloc_buffer((world,), torch.int32),
+        disp_flags=heap.alloc_buffer((nbuf, world), torch.int32),
+        comb_flags=heap.alloc_buffer((nbuf, world), torch.int32),
     )
+    ctx.low_latency = low_latency
+    ctx.credit_flags = heap.alloc_buffer((world,), torch.int32) \
+        if low_latency else None
+    ctx.call_no = 0
     if heap.backend == "hip":
         dev = "cuda"
         e = n_experts
@@ -86,6 +99,7 @@ def create_ep_context(max_tokens: int, hidden: int, n_experts: int,
             recv_total=torch.zeros(1, dtype=torch.int32, device=dev),
             arrive_d=torch.zeros(world, dtype=torch.int32, device=dev),
             arrive_c=torch.zeros(world, dtype=torch.int32, device=dev),
+            call_cell=torch.zeros(1, dtype=torch.int32, device=dev),
         )
     return ctx
 
@@ -117,15 +131,34 @@ def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
     stream = torch.cuda.current_stream()
     s = stream.cuda_stream
     L = ctx.local
+    nbuf = 2 if ctx.low_latency else 1
+    parity = ctx.call_no % nbuf
+    ctx.call_no += 1
+    # per-parity buffer offsets
+    splits_off = ctx.all_splits.offset + parity * world * E * 4
+    sflags_off = ctx.splits_flags.offset + parity * world * 4
+    recv_x_off = ctx.recv_x.offset + parity * (ctx.cap + 128) * H * 2
+    meta_off = ctx.meta.offset + parity * ctx.cap * 2 * 4
+    combine_off = ctx.combine_buf.offset + parity * ctx.max_tokens * K * H * 2
+    dflags_off = ctx.disp_flags.offset + parity * world * 4
+    cflags_off = ctx.comb_flags.offset + parity * world * 4
+    cell = 0
 
-    # phase 0: reset + entry barrier (buffer reuse protection; graph-safe)
+    # phase 0: buffer-reuse protection
     L["counts"].zero_()
     L["arrive_d"].zero_()
     L["arrive_c"].zero_()
-    _C.reset_flags(ctx.splits_flags.ptr(), world, 0, s)
-    _C.reset_flags(ctx.disp_flags.ptr(), world, 0, s)
-    _C.reset_flags(ctx.comb_flags.ptr(), world, 0, s)
-    heap.barrier_all_on_stream(stream)
+    if ctx.low_latency:
+        # LL protocol: no barrier, no flag resets — flags carry the call
+        # number; a credit flag bounds parity-buffer reuse to 2 calls
+        cell = L["call_cell"].data_ptr()
+        _C.bump_cell(cell, s)
+        _C.wait_flags_ge_cell(ctx.credit_flags.ptr(), world, cell, -2, s)
+    else:
+        _C.reset_flags(heap.ptr(rank, sflags_off), world, 0, s)
+        _C.reset_flags(heap.ptr(rank, dflags_off), world, 0, s)
+        _C.reset_flags(heap.ptr(rank, cflags_off), world, 0, s)
+        heap.barrier_all_on_stream(stream)
 
     # phase 1: routing histogram + slot assignment
     _C.moe_count(topk_ids.data_ptr(), L["counts"].data_ptr(),
@@ -133,20 +166,26 @@ def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
                  T * K, E, e_loc, world, s)
 
     # phase 2: exchange the splits matrix (SDMA push + flag per peer)
-    my_row_off = ctx.all_splits.offset + rank * E * 4
+    my_row_off = splits_off + rank * E * 4
     _C.memcpy_async(heap.ptr(rank, my_row_off), L["counts"].data_ptr(),
                     E * 4, s)
-    _C.reset_flags(ctx.splits_flags.ptr() + rank * 4, 1, 1, s)
+    if ctx.low_latency:
+        _C.memcpy_async(heap.ptr(rank, sflags_off) + rank * 4, cell, 4, s)
+    else:
+        _C.reset_flags(heap.ptr(rank, sflags_off) + rank * 4, 1, 1, s)
     for i in range(world - 1):
         peer = (rank + 1 + i) % world
         _C.memcpy_async(heap.ptr(peer, my_row_off), L["counts"].data_ptr(),
                         E * 4, s)
-        _C.memcpy_async(ctx.splits_flags.ptr(peer) + rank * 4,
-                        heap.one_src.ptr(), 4, s)
-    _C.wait_eq(ctx.splits_flags.ptr(), world, 1, s)
+        src = cell if ctx.low_latency else heap.one_src.ptr()
+        _C.memcpy_async(heap.ptr(peer, sflags_off) + rank * 4, src, 4, s)
+    if ctx.low_latency:
+        _C.wait_flags_ge_cell(heap.ptr(rank, sflags_off), world, cell, 0, s)
+    else:
+        _C.wait_eq(heap.ptr(rank, sflags_off), world, 1, s)
 
     # phase 3: derive layouts
-    _C.moe_layout(ctx.all_splits.ptr(), rank, world, E, e_loc,
+    _C.moe_layout(heap.ptr(rank, splits_off), rank, world, E, e_loc,
                   L["send_base"].data_ptr(), L["expert_base"].data_ptr(),
                   L["expert_rows"].data_ptr(), L["recv_from_src"].data_ptr(),
                   L["recv_total"].data_ptr(), s)
@@ -154,17 +193,17 @@ def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
     # phase 4: dispatch (xGMI row push + per-dst completion signals)
     _C.moe_dispatch(x.data_ptr(), topk_ids.data_ptr(),
                     L["send_pos"].data_ptr(), L["send_base"].data_ptr(),
-                    L["send_to_dst"].data_ptr(), ctx.recv_x.offset,
-                    ctx.meta.offset, ctx.disp_flags.offset,
-                    L["arrive_d"].data_ptr(), T, K, H, e_loc, s)
-    _C.moe_wait_flags(ctx.disp_flags.ptr(), world, s)
+                    L["send_to_dst"].data_ptr(), recv_x_off,
+                    meta_off, dflags_off,
+                    L["arrive_d"].data_ptr(), T, K, H, e_loc, s, cell)
+    _C.moe_wait_flags(heap.ptr(rank, dflags_off), world, s, cell)
 
     # phase 5: grouped expert FFN
     cap_tiles = (ctx.cap + 127) // 128
     # +128-row slack everywhere an edge GEMM tile may over-read
     expert_h = torch.empty(ctx.cap + 128, 2 * inter, dtype=torch.bfloat16,
                            device=x.device)
-    _C.moe_grouped_gemm(ctx.recv_x.ptr(), w_gate_up.data_ptr(),
+    _C.moe_grouped_gemm(heap.ptr(rank, recv_x_off), w_gate_up.data_ptr(),
                         expert_h.data_ptr(), L["expert_base"].data_ptr(),
                         L["expert_rows"].data_ptr(), e_loc, cap_tiles,
                         2 * inter, H, ctx.cap, s)
@@ -179,14 +218,17 @@ def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
                         inter, ctx.cap, s)
 
     # phase 6: combine (return rows + weighted reduce)
-    _C.moe_combine_send(expert_out.data_ptr(), ctx.meta.ptr(),
+    _C.moe_combine_send(expert_out.data_ptr(), heap.ptr(rank, meta_off),
                         L["recv_total"].data_ptr(),
                         L["recv_from_src"].data_ptr(),
-                        ctx.combine_buf.offset, ctx.comb_flags.offset,
-                        L["arrive_c"].data_ptr(), ctx.cap, H, s)
-    _C.moe_combine_reduce(ctx.combine_buf.ptr(), topk_w.data_ptr(),
+                        combine_off, cflags_off,
+                        L["arrive_c"].data_ptr(), ctx.cap, H, s, cell)
+    _C.moe_combine_reduce(heap.ptr(rank, combine_off), topk_w.data_ptr(),
                           topk_ids.data_ptr(), out.data_ptr(),
-                          ctx.comb_flags.ptr(), world, T, K, H, E, s)
+                          heap.ptr(rank, cflags_off), world, T, K, H, E, s,
+                          cell)
+    if ctx.low_latency:
+        _C.signal_credit(ctx.credit_flags.offset, cell, s)
     return out
 
 
@@ -210,12 +252,12 @@ def _ep_moe_cpu(x, topk_ids, topk_w, w_gate_up, w_down, ctx, out):
             counts[e] += 1
     # splits exchange
     for p in range(world):
-        ctx.all_splits.peer(p)[rank].copy_(counts)
-        cpu_shm.notify(ctx.splits_flags.peer(p), rank, ctx.epoch)
-    fl = ctx.splits_flags.local()
+        ctx.all_splits.peer(p)[0, rank].copy_(counts)
+        cpu_shm.notify(ctx.splits_flags.peer(p)[0], rank, ctx.epoch)
+    fl = ctx.splits_flags.local()[0]
     for sidx in range(world):
         cpu_shm.wait_ge(fl, sidx, ctx.epoch)
-    splits = ctx.all_splits.local().clone()  # [world, E]
+    splits = ctx.all_splits.local()[0].clone()  # [world, E]
 
     # layouts (same algebra as k_moe_layout)
     send_base = torch.zeros(E, dtype=torch.int64)
@@ -242,14 +284,14 @@ def _ep_moe_cpu(x, topk_ids, topk_w, w_gate_up, w_down, ctx, out):
             continue
         d = e // e_loc
         slot = int(send_base[e]) + int(send_pos[i])
-        ctx.recv_x.peer(d)[slot].copy_(x[i // K])
-        ctx.meta.peer(d)[slot, 0] = rank
-        ctx.meta.peer(d)[slot, 1] = i
+        ctx.recv_x.peer(d)[0, slot].copy_(x[i // K])
+        ctx.meta.peer(d)[0, slot, 0] = rank
+        ctx.meta.peer(d)[0, slot, 1] = i
     heap.barrier_all()
 
     # grouped FFN
     expert_out = torch.zeros(ctx.cap + 128, H, dtype=torch.bfloat16)
-    rx = ctx.recv_x.local()
+    rx = ctx.recv_x.local()[0]
     for le in range(e_loc):
         nb = int(expert_rows[le])
         if nb == 0:
@@ -262,16 +304,16 @@ def _ep_moe_cpu(x, topk_ids, topk_w, w_gate_up, w_down, ctx, out):
         expert_out[b0:b0 + nb] = y.to(torch.bfloat16)
 
     # combine send
-    meta = ctx.meta.local()
+    meta = ctx.meta.local()[0]
     total = int(expert_base[-1] + expert_rows[-1]) if e_loc else 0
     for r in range(total):
         src = int(meta[r, 0])
         tok_k = int(meta[r, 1])
-        ctx.combine_buf.peer(src)[tok_k].copy_(expert_out[r])
+        ctx.combine_buf.peer(src)[0, tok_k].copy_(expert_out[r])
     heap.barrier_all()
 
     # reduce
-    cb = ctx.combine_buf.local()
+    cb = ctx.combine_buf.local()[0]
     acc = torch.zeros(T, H, dtype=torch.float32)
     for t in range(T):
         for k in range(K):
