@@ -119,7 +119,8 @@ void launch_signal_credit(const PeerTable &pt, size_t credit_off,
 void launch_moe_grouped_gemm(const void *xin, const void *weights, void *out,
                              const void *expert_base, const void *expert_rows,
                              int e_loc, int cap_tiles_m, int n, int k,
-                             int cap_rows, hipStream_t stream);
+                             int cap_rows, hipStream_t stream,
+                             bool small_m = false);
 void launch_moe_combine_send(const PeerTable &pt, const void *expert_out,
                              const void *meta, const void *recv_total,
                              const void *recv_from_src, size_t combine_off,

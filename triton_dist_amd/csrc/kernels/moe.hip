@@ -276,6 +276,93 @@ __global__ __launch_bounds__(gg::NTH) void k_moe_grouped_gemm(
       }
 }
 
+// Small-M grouped GEMM: BM=32 tile (decode MoE: ~T*K*world/E rows per
+// expert, often 16-64) — 4 waves side by side on N (32x128 tile), no
+// M-waste. Same capacity-grid / masked-store contract as the 128 variant.
+__global__ __launch_bounds__(256) void k_moe_grouped_gemm_sm(
+    const bf16 *__restrict__ xin, const bf16 *__restrict__ weights,
+    bf16 *__restrict__ out, const int *__restrict__ expert_base,
+    const int *__restrict__ expert_rows, int cap_tiles_m, int n, int k,
+    int cap_rows) {
+  constexpr int BM = 32, BN = 128, BK = 64;
+  const int e = blockIdx.x / cap_tiles_m;
+  const int tm = blockIdx.x % cap_tiles_m;
+  const int tn = blockIdx.y;
+  const int rows = expert_rows[e];
+  if (tm * BM >= rows) return;
+  const int base = expert_base[e];
+
+  __shared__ bf16 lds_a[BM * BK];   // 4 KB
+  __shared__ bf16 lds_b[BN * BK];   // 16 KB
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  // wave w computes rows 0..31 x cols [w*32, w*32+32): 2x2 16x16 frags
+  f32x4 acc[2][2] = {};
+  const bf16 *ga = xin + (size_t)(base + tm * BM) * k;
+  const bf16 *gb = weights + (size_t)e * n * k + (size_t)tn * BN * k;
+  for (int k0 = 0; k0 < k; k0 += BK) {
+    // stage A (256 chunks) + B (1024 chunks): 1 + 4 loads per thread
+    {
+      int q = tid;  // A chunk
+      int row = q >> 3, kc = q & 7;
+      int wave_chunk0 = wave * 64;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int *)(
+              ga + (size_t)row * k + k0 + kc * 8),
+          (__attribute__((address_space(3))) unsigned int *)(lds_a +
+                                                             wave_chunk0 * 8),
+          16, 0, 0);
+#pragma unroll
+      for (int it = 0; it < 4; ++it) {
+        int qb = it * 256 + tid;
+        int rowb = qb >> 3, kcb = qb & 7;
+        int wave_chunk0b = it * 256 + wave * 64;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int *)(
+                gb + (size_t)rowb * k + k0 + kcb * 8),
+            (__attribute__((address_space(3))) unsigned int *)(
+                lds_b + wave_chunk0b * 8),
+            16, 0, 0);
+      }
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+#pragma unroll
+    for (int ks = 0; ks < BK / 32; ++ks) {
+      bf16x8 af[2], bfr[2];
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        int arow = i * 16 + (lane & 15);
+        int brow = wave * 32 + i * 16 + (lane & 15);
+        int kk = ks * 32 + (lane >> 4) * 8;
+        af[i] = *(const bf16x8 *)(lds_a + arow * BK + kk);
+        bfr[i] = *(const bf16x8 *)(lds_b + brow * BK + kk);
+      }
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[i], bfr[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+  const int row_lim = rows - tm * BM;
+  const int actual_base = base + tm * BM;
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = i * 16 + (lane >> 4) * 4 + r;
+        int col = wave * 32 + j * 16 + (lane & 15);
+        if (row < row_lim)
+          out[((size_t)actual_base + row) * n + (size_t)tn * BN + col] =
+              (bf16)acc[i][j][r];
+      }
+}
+
 // ---------------------------------------------------------------------------
 // Phase 4: combine send — return expert outputs to their source ranks.
 // Block per recv row; signals each source when all its rows are returned.
@@ -414,9 +501,19 @@ void launch_moe_wait_flags(const void *flags, int world, const void *cell,
 void launch_moe_grouped_gemm(const void *xin, const void *weights, void *out,
                              const void *expert_base, const void *expert_rows,
                              int e_loc, int cap_tiles_m, int n, int k,
-                             int cap_rows, hipStream_t stream) {
+                             int cap_rows, hipStream_t stream,
+                             bool small_m) {
   if (n % 128 || k % 64)
     throw std::runtime_error("grouped gemm: N%128/K%64 required");
+  if (small_m) {
+    hipLaunchKernelGGL(k_moe_grouped_gemm_sm,
+                       dim3(e_loc * cap_tiles_m, n / 128), dim3(256), 0,
+                       stream, (const bf16 *)xin, (const bf16 *)weights,
+                       (bf16 *)out, (const int *)expert_base,
+                       (const int *)expert_rows, cap_tiles_m, n, k,
+                       cap_rows);
+    return;
+  }
   hipLaunchKernelGGL(k_moe_grouped_gemm,
                      dim3(e_loc * cap_tiles_m, n / 128), dim3(gg::NTH), 0,
                      stream, (const bf16 *)xin, (const bf16 *)weights,

@@ -459,10 +459,12 @@ static void signal_credit(uintptr_t credit_off, uintptr_t cell,
 static void moe_grouped_gemm(uintptr_t xin, uintptr_t weights, uintptr_t out,
                              uintptr_t expert_base, uintptr_t expert_rows,
                              int e_loc, int cap_tiles_m, int n, int k,
-                             int cap_rows, uintptr_t stream) {
+                             int cap_rows, uintptr_t stream,
+                             bool small_m = false) {
   launch_moe_grouped_gemm((void *)xin, (void *)weights, (void *)out,
                           (void *)expert_base, (void *)expert_rows, e_loc,
-                          cap_tiles_m, n, k, cap_rows, as_stream(stream));
+                          cap_tiles_m, n, k, cap_rows, as_stream(stream),
+                          small_m);
   TD_CHECK_HIP(hipGetLastError());
 }
 
@@ -605,7 +607,11 @@ PYBIND11_MODULE(_C, m) {
   m.def("signal_credit", &signal_credit);
   m.def("moe_wait_flags", &moe_wait_flags, py::arg("flags"),
         py::arg("world"), py::arg("stream"), py::arg("cell") = 0);
-  m.def("moe_grouped_gemm", &moe_grouped_gemm);
+  m.def("moe_grouped_gemm", &moe_grouped_gemm, py::arg("xin"),
+        py::arg("weights"), py::arg("out"), py::arg("expert_base"),
+        py::arg("expert_rows"), py::arg("e_loc"), py::arg("cap_tiles_m"),
+        py::arg("n"), py::arg("k"), py::arg("cap_rows"), py::arg("stream"),
+        py::arg("small_m") = false);
   m.def("moe_combine_send", &moe_combine_send, py::arg("expert_out"),
         py::arg("meta"), py::arg("recv_total"), py::arg("recv_from_src"),
         py::arg("combine_off"), py::arg("cflags_off"), py::arg("arrive"),

@@ -45,7 +45,9 @@ class Qwen3MoE(DenseLLM):
         for li, layer in enumerate(self.layers):
             s = seed + 5000 + li * 10
             moe: EPMoELayer = layer["mlp"]
-            moe.router.copy_(full((cfg.n_experts, cfg.hidden), s))
+            # larger router std: near-uniform logits make top-k selection
+            # flip between bf16 paths, which is routing noise not a bug
+            moe.router.copy_(full((cfg.n_experts, cfg.hidden), s) * 25)
             gu = full((cfg.n_experts, 2 * cfg.moe_inter, cfg.hidden), s + 1)
             dn = full((cfg.n_experts, cfg.hidden, cfg.moe_inter), s + 2)
             moe.w_gate_up.copy_(gu[r * e_loc:(r + 1) * e_loc])

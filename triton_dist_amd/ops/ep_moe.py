@@ -198,15 +198,19 @@ def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
                     L["arrive_d"].data_ptr(), T, K, H, e_loc, s, cell)
     _C.moe_wait_flags(heap.ptr(rank, dflags_off), world, s, cell)
 
-    # phase 5: grouped expert FFN
-    cap_tiles = (ctx.cap + 127) // 128
+    # phase 5: grouped expert FFN (BM=32 tile when experts are lightly
+    # loaded — decode MoE averages T*K*world/E rows per expert)
+    avg_rows = max(1, (T * K * world) // E)
+    small_m = avg_rows <= 64
+    bm = 32 if small_m else 128
+    cap_tiles = (ctx.cap + bm - 1) // bm
     # +128-row slack everywhere an edge GEMM tile may over-read
     expert_h = torch.empty(ctx.cap + 128, 2 * inter, dtype=torch.bfloat16,
                            device=x.device)
     _C.moe_grouped_gemm(heap.ptr(rank, recv_x_off), w_gate_up.data_ptr(),
                         expert_h.data_ptr(), L["expert_base"].data_ptr(),
                         L["expert_rows"].data_ptr(), e_loc, cap_tiles,
-                        2 * inter, H, ctx.cap, s)
+                        2 * inter, H, ctx.cap, s, small_m)
     act = torch.empty(ctx.cap + 128, inter, dtype=torch.bfloat16,
                       device=x.device)
     _C.swiglu(expert_h.data_ptr(), act.data_ptr(), ctx.cap + 128, inter, s)
@@ -215,7 +219,7 @@ def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
     _C.moe_grouped_gemm(act.data_ptr(), w_down.data_ptr(),
                         expert_out.data_ptr(), L["expert_base"].data_ptr(),
                         L["expert_rows"].data_ptr(), e_loc, cap_tiles, H,
-                        inter, ctx.cap, s)
+                        inter, ctx.cap, s, small_m)
 
     # phase 6: combine (return rows + weighted reduce)
     _C.moe_combine_send(expert_out.data_ptr(), heap.ptr(rank, meta_off),
