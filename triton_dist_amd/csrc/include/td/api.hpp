@@ -102,7 +102,14 @@ void launch_moe_layout(const void *all_splits, int rank, int world,
                        int e_num, int e_loc, void *send_base,
                        void *expert_base, void *expert_rows,
                        void *recv_from_src, void *recv_total,
+                       void *work_items, void *work_count, int bm,
                        hipStream_t stream);
+void launch_moe_grouped_gemm_pq(const void *xin, const void *weights,
+                                void *out, const void *expert_base,
+                                const void *expert_rows,
+                                const void *work_items,
+                                const void *work_count, int n, int k,
+                                hipStream_t stream);
 void launch_moe_dispatch(const PeerTable &pt, const void *x,
                          const void *topk_ids, const void *send_pos,
                          const void *send_base, const void *send_to_dst,

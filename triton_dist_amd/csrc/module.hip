@@ -406,11 +406,26 @@ static void moe_count(uintptr_t topk_ids, uintptr_t counts,
 static void moe_layout(uintptr_t all_splits, int rank, int world, int e_num,
                        int e_loc, uintptr_t send_base, uintptr_t expert_base,
                        uintptr_t expert_rows, uintptr_t recv_from_src,
-                       uintptr_t recv_total, uintptr_t stream) {
+                       uintptr_t recv_total, uintptr_t stream,
+                       uintptr_t work_items = 0, uintptr_t work_count = 0,
+                       int bm = 128) {
   launch_moe_layout((void *)all_splits, rank, world, e_num, e_loc,
                     (void *)send_base, (void *)expert_base,
                     (void *)expert_rows, (void *)recv_from_src,
-                    (void *)recv_total, as_stream(stream));
+                    (void *)recv_total, (void *)work_items,
+                    (void *)work_count, bm, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
+static void moe_grouped_gemm_pq(uintptr_t xin, uintptr_t weights,
+                                uintptr_t out, uintptr_t expert_base,
+                                uintptr_t expert_rows, uintptr_t work_items,
+                                uintptr_t work_count, int n, int k,
+                                uintptr_t stream) {
+  launch_moe_grouped_gemm_pq((void *)xin, (void *)weights, (void *)out,
+                             (void *)expert_base, (void *)expert_rows,
+                             (void *)work_items, (void *)work_count, n, k,
+                             as_stream(stream));
   TD_CHECK_HIP(hipGetLastError());
 }
 
@@ -596,7 +611,13 @@ PYBIND11_MODULE(_C, m) {
   m.def("allreduce_oneshot", &allreduce_oneshot);
   m.def("allreduce_twoshot", &allreduce_twoshot);
   m.def("moe_count", &moe_count);
-  m.def("moe_layout", &moe_layout);
+  m.def("moe_layout", &moe_layout, py::arg("all_splits"), py::arg("rank"),
+        py::arg("world"), py::arg("e_num"), py::arg("e_loc"),
+        py::arg("send_base"), py::arg("expert_base"), py::arg("expert_rows"),
+        py::arg("recv_from_src"), py::arg("recv_total"), py::arg("stream"),
+        py::arg("work_items") = 0, py::arg("work_count") = 0,
+        py::arg("bm") = 128);
+  m.def("moe_grouped_gemm_pq", &moe_grouped_gemm_pq);
   m.def("moe_dispatch", &moe_dispatch, py::arg("x"), py::arg("topk_ids"),
         py::arg("send_pos"), py::arg("send_base"), py::arg("send_to_dst"),
         py::arg("recv_x_off"), py::arg("meta_off"), py::arg("flags_off"),

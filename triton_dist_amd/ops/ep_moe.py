@@ -100,6 +100,9 @@ def create_ep_context(max_tokens: int, hidden: int, n_experts: int,
             arrive_d=torch.zeros(world, dtype=torch.int32, device=dev),
             arrive_c=torch.zeros(world, dtype=torch.int32, device=dev),
             call_cell=torch.zeros(1, dtype=torch.int32, device=dev),
+            work_items=torch.zeros(ctx.e_loc + cap // 32 + 1,
+                                   dtype=torch.int32, device=dev),
+            work_count=torch.zeros(1, dtype=torch.int32, device=dev),
         )
     return ctx
 
@@ -184,11 +187,15 @@ def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
     else:
         _C.wait_eq(heap.ptr(rank, sflags_off), world, 1, s)
 
-    # phase 3: derive layouts
+    # phase 3: derive layouts (+ the grouped-GEMM work queue)
+    avg_rows = max(1, (T * K * world) // E)
+    small_m = avg_rows <= 64
+    bm = 32 if small_m else 128
     _C.moe_layout(heap.ptr(rank, splits_off), rank, world, E, e_loc,
                   L["send_base"].data_ptr(), L["expert_base"].data_ptr(),
                   L["expert_rows"].data_ptr(), L["recv_from_src"].data_ptr(),
-                  L["recv_total"].data_ptr(), s)
+                  L["recv_total"].data_ptr(), s,
+                  L["work_items"].data_ptr(), L["work_count"].data_ptr(), bm)
 
     # phase 4: dispatch (xGMI row push + per-dst completion signals)
     _C.moe_dispatch(x.data_ptr(), topk_ids.data_ptr(),
@@ -198,28 +205,41 @@ def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
                     L["arrive_d"].data_ptr(), T, K, H, e_loc, s, cell)
     _C.moe_wait_flags(heap.ptr(rank, dflags_off), world, s, cell)
 
-    # phase 5: grouped expert FFN (BM=32 tile when experts are lightly
-    # loaded — decode MoE averages T*K*world/E rows per expert)
-    avg_rows = max(1, (T * K * world) // E)
-    small_m = avg_rows <= 64
-    bm = 32 if small_m else 128
+    # phase 5: grouped expert FFN — persistent work-queue kernel when
+    # experts are lightly loaded (decode: ~T*K*world/E rows per expert)
     cap_tiles = (ctx.cap + bm - 1) // bm
     # +128-row slack everywhere an edge GEMM tile may over-read
     expert_h = torch.empty(ctx.cap + 128, 2 * inter, dtype=torch.bfloat16,
                            device=x.device)
-    _C.moe_grouped_gemm(heap.ptr(rank, recv_x_off), w_gate_up.data_ptr(),
-                        expert_h.data_ptr(), L["expert_base"].data_ptr(),
-                        L["expert_rows"].data_ptr(), e_loc, cap_tiles,
-                        2 * inter, H, ctx.cap, s, small_m)
+    if small_m:
+        _C.moe_grouped_gemm_pq(heap.ptr(rank, recv_x_off),
+                               w_gate_up.data_ptr(), expert_h.data_ptr(),
+                               L["expert_base"].data_ptr(),
+                               L["expert_rows"].data_ptr(),
+                               L["work_items"].data_ptr(),
+                               L["work_count"].data_ptr(), 2 * inter, H, s)
+    else:
+        _C.moe_grouped_gemm(heap.ptr(rank, recv_x_off), w_gate_up.data_ptr(),
+                            expert_h.data_ptr(), L["expert_base"].data_ptr(),
+                            L["expert_rows"].data_ptr(), e_loc, cap_tiles,
+                            2 * inter, H, ctx.cap, s, False)
     act = torch.empty(ctx.cap + 128, inter, dtype=torch.bfloat16,
                       device=x.device)
     _C.swiglu(expert_h.data_ptr(), act.data_ptr(), ctx.cap + 128, inter, s)
     expert_out = torch.empty(ctx.cap + 128, H, dtype=torch.bfloat16,
                              device=x.device)
-    _C.moe_grouped_gemm(act.data_ptr(), w_down.data_ptr(),
-                        expert_out.data_ptr(), L["expert_base"].data_ptr(),
-                        L["expert_rows"].data_ptr(), e_loc, cap_tiles, H,
-                        inter, ctx.cap, s, small_m)
+    if small_m:
+        _C.moe_grouped_gemm_pq(act.data_ptr(), w_down.data_ptr(),
+                               expert_out.data_ptr(),
+                               L["expert_base"].data_ptr(),
+                               L["expert_rows"].data_ptr(),
+                               L["work_items"].data_ptr(),
+                               L["work_count"].data_ptr(), H, inter, s)
+    else:
+        _C.moe_grouped_gemm(act.data_ptr(), w_down.data_ptr(),
+                            expert_out.data_ptr(), L["expert_base"].data_ptr(),
+                            L["expert_rows"].data_ptr(), e_loc, cap_tiles, H,
+                            inter, ctx.cap, s, False)
 
     # phase 6: combine (return rows + weighted reduce)
     _C.moe_combine_send(expert_out.data_ptr(), heap.ptr(rank, meta_off),
