@@ -140,6 +140,11 @@ void launch_moe_combine_reduce(const void *combine_buf, const void *topk_w,
                                int world, int T, int K,
                                int H, int e_num, hipStream_t stream);
 
+// kernels/megakernel.hip -----------------------------------------------------
+void launch_megakernel(const void *tasks, const void *queue,
+                       const void *queue_off, void *scoreboard, int n_wg,
+                       hipStream_t stream);
+
 // kernels/elementwise.hip ----------------------------------------------------
 void launch_rmsnorm(const void *x, const void *w, void *out, int rows,
                     int cols, float eps, hipStream_t stream);

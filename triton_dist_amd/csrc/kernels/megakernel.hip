@@ -1,0 +1,452 @@
+// Megakernel: one persistent kernel executes a whole decode step as a task
+// graph — per-workgroup task queues, op-level scoreboard dependencies, and
+// a dispatch switch over a fixed CDNA4 task vocabulary.
+//
+// Capability parity with the reference megakernel subsystem
+// (Triton-distributed mega_triton_kernel/core/{builder.py:34,
+// scheduler.py:31-157, code_generator.py:67-280} — behavior only). The
+// reference CODEGENS a Triton kernel per model; here the task vocabulary
+// is a fixed set of hand-written HIP device functions (RMSNorm,
+// add+RMSNorm, GEMM tile, SwiGLU, qkv-prologue, flash-decode, embed,
+// copy), so no JIT is needed — the model builder only emits descriptors.
+//
+// Scheduling contract (deadlock freedom): tasks are assigned to queues in
+// TOPOLOGICAL-LEVEL order, round-robin across workgroups. Every queue is
+// executed in order, so when a workgroup blocks on a level-L dependency,
+// every level<L task is either done or ahead of all blocked tasks in some
+// queue — progress is guaranteed. Scoreboard slots are per-op arrive
+// counters (system-scope not needed: single GPU, device scope).
+#include <stdexcept>
+
+#include "td/api.hpp"
+
+namespace td {
+
+using bf16 = __bf16;
+typedef __attribute__((ext_vector_type(8))) bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+namespace mk {
+
+enum TaskType : int {
+  T_RMSNORM = 0,      // a0=x, a1=w, a2=out, a3=rows, a4=cols, a5=row0
+  T_ADD_RMSNORM = 1,  // a0=delta, a1=resid_in, a2=resid_out, a3=w, a4=out,
+                      // a5=rows, a6=cols, a7=row0
+  T_GEMM_TILE = 2,    // a0=A, a1=B(NxK), a2=C, a3=m(valid rows), a4=n,
+                      // a5=k, a6=pid_m, a7=pid_n
+  T_SWIGLU = 3,       // a0=h, a1=out, a2=rows, a3=inter, a4=chunk, a5=nchunks
+  T_QKV_PROLOGUE = 4, // a0=qkv, a1=qout, a2=kcache, a3=vcache, a4=cos,
+                      // a5=sin, a6=qnw, a7=knw, a8=offset(i64 cell),
+                      // a9=b, a10=qh, a11=kvh, a12=maxlen
+  T_FLASH_DECODE = 5, // a0=q, a1=kc, a2=vc, a3=out, a4=offset, a5=b,
+                      // a6=kh, a7=qh, a8=kvh, a9=maxlen
+  T_EMBED = 6,        // a0=tokens(i64), a1=table, a2=out, a3=rows, a4=cols,
+                      // a5=row0
+  T_KV_ADVANCE = 7,   // a0=offset cell (i64): += 1
+};
+
+struct Task {
+  int type;
+  int score_slot;     // arrive counter this task bumps when done
+  int dep0, dep0_n;   // wait scoreboard[dep0] >= dep0_n   (-1 = none)
+  int dep1, dep1_n;
+  long long a[13];    // pointers / ints (pointers pre-resolved on host)
+};
+
+constexpr int NTH = 256;
+
+// ---------------------------------------------------------------------------
+// task bodies (256 threads each)
+// ---------------------------------------------------------------------------
+TD_DEV void t_rmsnorm(const Task &t, bool add) {
+  // one task = one row (grid of row tasks emitted by the builder)
+  const bf16 *x = (const bf16 *)t.a[0];
+  const int cols = (int)t.a[add ? 6 : 4];
+  const int row = (int)t.a[add ? 7 : 5];
+  const bf16 *xr = x + (size_t)row * cols;
+  const bf16 *rr = nullptr;
+  bf16 *ro = nullptr, *orow;
+  const bf16 *w;
+  if (add) {
+    rr = (const bf16 *)t.a[1] + (size_t)row * cols;
+    ro = (bf16 *)t.a[2] + (size_t)row * cols;
+    w = (const bf16 *)t.a[3];
+    orow = (bf16 *)t.a[4] + (size_t)row * cols;
+  } else {
+    w = (const bf16 *)t.a[1];
+    orow = (bf16 *)t.a[2] + (size_t)row * cols;
+  }
+  __shared__ float red[16];
+  float ss = 0.f;
+  const int nv = cols / 8;
+  for (int i = threadIdx.x; i < nv; i += NTH) {
+    bf16x8 v = *(const bf16x8 *)(xr + i * 8);
+    if (add) {
+      bf16x8 r = *(const bf16x8 *)(rr + i * 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v[j] = (bf16)((float)v[j] + (float)r[j]);
+      *(bf16x8 *)(ro + i * 8) = v;
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) ss += (float)v[j] * (float)v[j];
+  }
+  for (int off = 32; off > 0; off >>= 1) ss += __shfl_down(ss, off);
+  if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = ss;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float tt = 0.f;
+    for (int i = 0; i < NTH / 64; ++i) tt += red[i];
+    red[0] = rsqrtf(tt / cols + 1e-6f);
+  }
+  __syncthreads();
+  const float scale = red[0];
+  const bf16 *src = add ? ro : xr;
+  for (int i = threadIdx.x; i < nv; i += NTH) {
+    bf16x8 v = *(const bf16x8 *)(src + i * 8);
+    bf16x8 wv = *(const bf16x8 *)(w + i * 8);
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      o[j] = (bf16)((float)v[j] * scale * (float)wv[j]);
+    *(bf16x8 *)(orow + i * 8) = o;
+  }
+}
+
+TD_DEV void t_gemm_tile(const Task &t, bf16 *lds) {
+  // 128x128 C tile, 4 waves 2x2, BK=64, masked rows (A padded to 128 rows)
+  constexpr int BM = 128, BN = 128, BK = 64;
+  bf16 *lds_a = lds;            // 128*64
+  bf16 *lds_b = lds + BM * BK;  // 128*64
+  const bf16 *A = (const bf16 *)t.a[0];
+  const bf16 *B = (const bf16 *)t.a[1];
+  bf16 *C = (bf16 *)t.a[2];
+  const int m = (int)t.a[3], n = (int)t.a[4], k = (int)t.a[5];
+  const int pid_m = (int)t.a[6], pid_n = (int)t.a[7];
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int wr = wave >> 1, wc = wave & 1;
+  f32x4 acc[4][4] = {};
+  const bf16 *ga = A + (size_t)pid_m * BM * k;
+  const bf16 *gb = B + (size_t)pid_n * BN * k;
+  for (int k0 = 0; k0 < k; k0 += BK) {
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      int idx = it * NTH + tid;
+      int row = idx >> 3, kc = idx & 7;
+      int wave_chunk0 = it * NTH + wave * 64;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int *)(
+              ga + (size_t)row * k + k0 + kc * 8),
+          (__attribute__((address_space(3))) unsigned int *)(lds_a +
+                                                             wave_chunk0 * 8),
+          16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int *)(
+              gb + (size_t)row * k + k0 + kc * 8),
+          (__attribute__((address_space(3))) unsigned int *)(lds_b +
+                                                             wave_chunk0 * 8),
+          16, 0, 0);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+#pragma unroll
+    for (int ks = 0; ks < BK / 32; ++ks) {
+      bf16x8 af[4], bfr[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int arow = wr * 64 + i * 16 + (lane & 15);
+        int brow = wc * 64 + i * 16 + (lane & 15);
+        int kk = ks * 32 + (lane >> 4) * 8;
+        af[i] = *(const bf16x8 *)(lds_a + arow * BK + kk);
+        bfr[i] = *(const bf16x8 *)(lds_b + brow * BK + kk);
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[i], bfr[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+  const int row_lim = m - pid_m * BM;
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = wr * 64 + i * 16 + (lane >> 4) * 4 + r;
+        int col = wc * 64 + j * 16 + (lane & 15);
+        if (row < row_lim)
+          C[((size_t)pid_m * BM + row) * n + (size_t)pid_n * BN + col] =
+              (bf16)acc[i][j][r];
+      }
+}
+
+TD_DEV void t_swiglu(const Task &t) {
+  const bf16 *h = (const bf16 *)t.a[0];
+  bf16 *out = (bf16 *)t.a[1];
+  const size_t rows = (size_t)t.a[2];
+  const int inter = (int)t.a[3];
+  const int chunk = (int)t.a[4], nchunks = (int)t.a[5];
+  size_t total = rows * (size_t)inter / 8;
+  size_t per = (total + nchunks - 1) / nchunks;
+  size_t lo = chunk * per, hi = min(lo + per, total);
+  for (size_t v = lo + threadIdx.x; v < hi; v += NTH) {
+    size_t i = v * 8;
+    size_t r = i / inter, c = i % inter;
+    bf16x8 g = *(const bf16x8 *)(h + r * 2 * inter + c);
+    bf16x8 u = *(const bf16x8 *)(h + r * 2 * inter + inter + c);
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gf = (float)g[j];
+      o[j] = (bf16)(gf / (1.f + __expf(-gf)) * (float)u[j]);
+    }
+    *(bf16x8 *)(out + i) = o;
+  }
+}
+
+TD_DEV void t_qkv_prologue(const Task &t) {
+  // one task = one batch row, all (qh + 2*kvh) heads; 4 heads at a time
+  constexpr int D = 128;
+  const bf16 *qkv = (const bf16 *)t.a[0];
+  bf16 *q_out = (bf16 *)t.a[1];
+  bf16 *kcache = (bf16 *)t.a[2];
+  bf16 *vcache = (bf16 *)t.a[3];
+  const float *cos_t = (const float *)t.a[4];
+  const float *sin_t = (const float *)t.a[5];
+  const bf16 *qnw = (const bf16 *)t.a[6];
+  const bf16 *knw = (const bf16 *)t.a[7];
+  const long pos = *(const long *)t.a[8];
+  const int b = (int)t.a[9];
+  const int qh = (int)t.a[10], kvh = (int)t.a[11], maxlen = (int)t.a[12];
+  const int nh = qh + 2 * kvh;
+  const int lane = threadIdx.x & 63;
+  for (int h = threadIdx.x >> 6; h < nh; h += NTH / 64) {
+    const bf16 *src = qkv + ((size_t)b * nh + h) * D;
+    float v0 = (float)src[lane * 2];
+    float v1 = (float)src[lane * 2 + 1];
+    const bool is_q = h < qh;
+    const bool is_k = h >= qh && h < qh + kvh;
+    if (is_q || is_k) {
+      float ss = v0 * v0 + v1 * v1;
+      for (int off = 32; off > 0; off >>= 1) ss += __shfl_down(ss, off);
+      float scale = rsqrtf(__shfl(ss, 0) / D + 1e-6f);
+      const bf16 *nw = is_q ? qnw : knw;
+      v0 *= scale * (float)nw[lane * 2];
+      v1 *= scale * (float)nw[lane * 2 + 1];
+      float p0 = __shfl_xor(v0, 32);
+      float p1 = __shfl_xor(v1, 32);
+      int d2 = (lane & 31) * 2;
+      float c0 = cos_t[pos * (D / 2) + d2];
+      float s0 = sin_t[pos * (D / 2) + d2];
+      float c1 = cos_t[pos * (D / 2) + d2 + 1];
+      float s1 = sin_t[pos * (D / 2) + d2 + 1];
+      if (lane < 32) {
+        v0 = v0 * c0 - p0 * s0;
+        v1 = v1 * c1 - p1 * s1;
+      } else {
+        v0 = v0 * c0 + p0 * s0;
+        v1 = v1 * c1 + p1 * s1;
+      }
+    }
+    if (is_q) {
+      bf16 *dst = q_out + ((size_t)b * qh + h) * D;
+      dst[lane * 2] = (bf16)v0;
+      dst[lane * 2 + 1] = (bf16)v1;
+    } else {
+      const int kh = h - qh;
+      const bool k_side = kh < kvh;
+      const int hh = k_side ? kh : kh - kvh;
+      bf16 *cache = k_side ? kcache : vcache;
+      bf16 *dst = cache + (((size_t)b * maxlen + pos) * kvh + hh) * D;
+      dst[lane * 2] = (bf16)v0;
+      dst[lane * 2 + 1] = (bf16)v1;
+    }
+  }
+}
+
+TD_DEV void t_embed(const Task &t) {
+  const long *tokens = (const long *)t.a[0];
+  const bf16 *table = (const bf16 *)t.a[1];
+  bf16 *out = (bf16 *)t.a[2];
+  const int cols = (int)t.a[4];
+  const int row = (int)t.a[5];
+  const bf16 *src = table + (size_t)tokens[row] * cols;
+  bf16 *dst = out + (size_t)row * cols;
+  for (int i = threadIdx.x; i < cols / 8; i += NTH)
+    *(bf16x8 *)(dst + i * 8) = *(const bf16x8 *)(src + i * 8);
+}
+
+}  // namespace mk
+
+// flash-decode task shares the standalone kernel's body via a device
+// function (mirrors k_flash_decode in attention.hip, LDS passed in).
+TD_DEV void mk_flash_decode_body(const bf16 *q, const bf16 *kcache,
+                                 const bf16 *vcache, bf16 *out,
+                                 const long *offset, int b, int kh, int qh,
+                                 int kvh, int max_len, char *lds_raw) {
+  constexpr int kD = 128, kTile = 32;
+  const int G = qh / kvh;
+  const int tid = threadIdx.x;
+  const int g = tid >> 5;
+  const int t = tid & 31;
+  const long seqlen = *offset + 1;
+  bf16(*k_lds)[kD] = (bf16(*)[kD])lds_raw;
+  bf16(*v_lds)[kD] = (bf16(*)[kD])(lds_raw + kTile * kD * 2);
+  float(*p_lds)[kTile] = (float(*)[kTile])(lds_raw + 2 * kTile * kD * 2);
+  float *m_lds = (float *)(lds_raw + 2 * kTile * kD * 2 + 8 * kTile * 4);
+  float *r_lds = m_lds + 8;
+  float *l_lds = r_lds + 8;
+  bf16(*q_lds)[kD] = (bf16(*)[kD])(l_lds + 8);
+
+  for (int i = tid; i < 8 * kD / 8; i += 256) {
+    int hh = i / (kD / 8);
+    int c = (i % (kD / 8)) * 8;
+    bf16x8 v{};
+    if (hh < G)
+      v = *(const bf16x8 *)(q + (((size_t)b * qh) + kh * G + hh) * kD + c);
+    *(bf16x8 *)(&q_lds[hh][c]) = v;
+  }
+  if (tid < 8) {
+    m_lds[tid] = -1e30f;
+    l_lds[tid] = 0.f;
+  }
+  __syncthreads();
+  float acc[4] = {};
+  const int my_d0 = t * 4;
+  const long ntiles = (seqlen + kTile - 1) / kTile;
+  for (long tile = 0; tile < ntiles; ++tile) {
+    const long pos0 = tile * kTile;
+    __syncthreads();
+    for (int i = tid; i < kTile * kD / 8; i += 256) {
+      int r = i / (kD / 8);
+      int c = (i % (kD / 8)) * 8;
+      long pos = pos0 + r;
+      bf16x8 kv{}, vv{};
+      if (pos < seqlen) {
+        size_t base = (((size_t)b * max_len + pos) * kvh + kh) * kD + c;
+        kv = *(const bf16x8 *)(kcache + base);
+        vv = *(const bf16x8 *)(vcache + base);
+      }
+      *(bf16x8 *)(&k_lds[r][c]) = kv;
+      *(bf16x8 *)(&v_lds[r][c]) = vv;
+    }
+    __syncthreads();
+    float s = -1e30f;
+    if (g < G && pos0 + t < seqlen) {
+      float d = 0.f;
+#pragma unroll
+      for (int c = 0; c < kD / 8; ++c) {
+        bf16x8 qv = *(const bf16x8 *)(&q_lds[g][c * 8]);
+        bf16x8 kv = *(const bf16x8 *)(&k_lds[t][c * 8]);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) d += (float)qv[j] * (float)kv[j];
+      }
+      s = d * 0.08838834764831845f;  // 1/sqrt(128)
+    }
+    float mx = s;
+    for (int off = 16; off > 0; off >>= 1)
+      mx = fmaxf(mx, __shfl_xor(mx, off));
+    float m_old = m_lds[g];
+    float m_new = fmaxf(m_old, mx);
+    float pp = (s > -1e29f) ? __expf(s - m_new) : 0.f;
+    p_lds[g][t] = pp;
+    float psum = pp;
+    for (int off = 16; off > 0; off >>= 1) psum += __shfl_xor(psum, off);
+    if (t == 0) {
+      float r = __expf(m_old - m_new);
+      r_lds[g] = r;
+      l_lds[g] = l_lds[g] * r + psum;
+      m_lds[g] = m_new;
+    }
+    __syncthreads();
+    const float r = r_lds[g];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[j] *= r;
+    for (int tt = 0; tt < kTile; ++tt) {
+      float pp2 = p_lds[g][tt];
+      if (pp2 != 0.f) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[j] += pp2 * (float)v_lds[tt][my_d0 + j];
+      }
+    }
+  }
+  __syncthreads();
+  if (g < G) {
+    float inv_l = 1.f / l_lds[g];
+    bf16 *dst = out + (((size_t)b * qh) + kh * G + g) * kD + my_d0;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) dst[j] = (bf16)(acc[j] * inv_l);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// the persistent megakernel
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(mk::NTH) void k_megakernel(
+    const mk::Task *__restrict__ tasks, const int *__restrict__ queue,
+    const int *__restrict__ queue_off, int *__restrict__ scoreboard) {
+  __shared__ char lds[32768];  // union: gemm A/B tiles | flash-decode state
+  const int wg = blockIdx.x;
+  const int q_lo = queue_off[wg], q_hi = queue_off[wg + 1];
+  for (int qi = q_lo; qi < q_hi; ++qi) {
+    const mk::Task t = tasks[queue[qi]];
+    // dependency waits (device scope; decode-scale spins are short)
+    if (threadIdx.x == 0) {
+      if (t.dep0 >= 0) wait_ge_one<Scope::Gpu>(scoreboard + t.dep0, t.dep0_n);
+      if (t.dep1 >= 0) wait_ge_one<Scope::Gpu>(scoreboard + t.dep1, t.dep1_n);
+    }
+    __syncthreads();
+    switch (t.type) {
+      case mk::T_RMSNORM:
+        mk::t_rmsnorm(t, false);
+        break;
+      case mk::T_ADD_RMSNORM:
+        mk::t_rmsnorm(t, true);
+        break;
+      case mk::T_GEMM_TILE:
+        mk::t_gemm_tile(t, (bf16 *)lds);
+        break;
+      case mk::T_SWIGLU:
+        mk::t_swiglu(t);
+        break;
+      case mk::T_QKV_PROLOGUE:
+        mk::t_qkv_prologue(t);
+        break;
+      case mk::T_FLASH_DECODE:
+        mk_flash_decode_body((const bf16 *)t.a[0], (const bf16 *)t.a[1],
+                             (const bf16 *)t.a[2], (bf16 *)t.a[3],
+                             (const long *)t.a[4], (int)t.a[5], (int)t.a[6],
+                             (int)t.a[7], (int)t.a[8], (int)t.a[9], lds);
+        break;
+      case mk::T_EMBED:
+        mk::t_embed(t);
+        break;
+      case mk::T_KV_ADVANCE:
+        if (threadIdx.x == 0) *(long *)t.a[0] += 1;
+        break;
+      default:
+        break;
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      fence_release_sys();
+      atomic_add<Scope::Gpu>(scoreboard + t.score_slot, 1);
+    }
+    __syncthreads();
+  }
+}
+
+void launch_megakernel(const void *tasks, const void *queue,
+                       const void *queue_off, void *scoreboard, int n_wg,
+                       hipStream_t stream) {
+  hipLaunchKernelGGL(k_megakernel, dim3(n_wg), dim3(mk::NTH), 0, stream,
+                     (const mk::Task *)tasks, (const int *)queue,
+                     (const int *)queue_off, (int *)scoreboard);
+}
+
+}  // namespace td

@@ -508,6 +508,13 @@ static void moe_combine_reduce(uintptr_t combine_buf, uintptr_t topk_w,
   TD_CHECK_HIP(hipGetLastError());
 }
 
+static void megakernel(uintptr_t tasks, uintptr_t queue, uintptr_t queue_off,
+                       uintptr_t scoreboard, int n_wg, uintptr_t stream) {
+  launch_megakernel((void *)tasks, (void *)queue, (void *)queue_off,
+                    (void *)scoreboard, n_wg, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
 static void rmsnorm(uintptr_t x, uintptr_t w, uintptr_t out, int rows,
                     int cols, float eps, uintptr_t stream) {
   launch_rmsnorm((void *)x, (void *)w, (void *)out, rows, cols, eps,
@@ -643,6 +650,7 @@ PYBIND11_MODULE(_C, m) {
         py::arg("cflags"), py::arg("world"), py::arg("T"), py::arg("K"),
         py::arg("H"), py::arg("e_num"), py::arg("stream"),
         py::arg("val_cell") = 0);
+  m.def("megakernel", &megakernel);
   m.def("rmsnorm", &rmsnorm);
   m.def("add_rmsnorm", &add_rmsnorm);
   m.def("swiglu", &swiglu);

@@ -1,0 +1,111 @@
+"""Megakernel task-graph builder + levelized round-robin scheduler.
+
+Capability parity with the reference megakernel core (Triton-distributed
+mega_triton_kernel/core/{graph.py:59-134 Graph/Node, builder.py:48 task
+descriptors, scheduler.py:103-157 static round-robin scheduling} — the
+reference codegens a Triton kernel; here descriptors feed the fixed HIP
+task vocabulary in csrc/kernels/megakernel.hip).
+
+Task record layout (must match mk::Task):
+  int32 x 6: type, score_slot, dep0, dep0_n, dep1, dep1_n  (+8B pad)
+  int64 x 13: args (device pointers / scalars)
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import List, Optional, Tuple
+
+import torch
+
+# task types (mirror mk::TaskType)
+T_RMSNORM = 0
+T_ADD_RMSNORM = 1
+T_GEMM_TILE = 2
+T_SWIGLU = 3
+T_QKV_PROLOGUE = 4
+T_FLASH_DECODE = 5
+T_EMBED = 6
+T_KV_ADVANCE = 7
+
+TASK_INT64S = 13
+TASK_WORDS = 2 + 1 + TASK_INT64S  # 6 int32 = 3 int64 words (with pad)
+
+
+@dataclass
+class Op:
+    """One logical op = a scoreboard slot; its tasks arrive-count to it."""
+    slot: int
+    level: int
+    n_tasks: int = 0
+
+
+@dataclass
+class MegaGraph:
+    tasks: List[Tuple] = field(default_factory=list)  # (type, slot, deps, args)
+    ops: List[Op] = field(default_factory=list)
+    level: int = 0
+
+    def new_op(self) -> Op:
+        op = Op(slot=len(self.ops), level=self.level)
+        self.ops.append(op)
+        return op
+
+    def next_level(self):
+        self.level += 1
+
+    def add_task(self, ttype: int, op: Op, args: List[int],
+                 deps: List[Tuple[Op, int]] = ()):
+        d = [(-1, 0), (-1, 0)]
+        for i, (dop, need) in enumerate(deps):
+            d[i] = (dop.slot, need if need else dop.n_tasks)
+        assert len(args) <= TASK_INT64S
+        self.tasks.append((ttype, op.slot, d[0][0], d[0][1], d[1][0],
+                           d[1][1], list(args) + [0] * (TASK_INT64S - len(args)),
+                           self.level))
+        op.n_tasks += 1
+
+    # ---------------------------------------------------------------- build
+    def finalize(self, n_wg: int, device="cuda"):
+        """Levelized round-robin assignment -> device descriptor tensors."""
+        # encode tasks
+        n = len(self.tasks)
+        buf = torch.zeros(n, TASK_WORDS, dtype=torch.int64)
+        for i, (tt, slot, d0, d0n, d1, d1n, args, _lvl) in enumerate(self.tasks):
+            # pack int32 pairs into int64 words (little-endian)
+            buf[i, 0] = (tt & 0xFFFFFFFF) | ((slot & 0xFFFFFFFF) << 32)
+            buf[i, 1] = (d0 & 0xFFFFFFFF) | ((d0n & 0xFFFFFFFF) << 32)
+            buf[i, 2] = (d1 & 0xFFFFFFFF) | ((d1n & 0xFFFFFFFF) << 32)
+            for j, a in enumerate(args):
+                buf[i, 3 + j] = a
+        # levelized round-robin queues
+        order = sorted(range(n), key=lambda i: self.tasks[i][7])
+        queues = [[] for _ in range(n_wg)]
+        for pos, ti in enumerate(order):
+            queues[pos % n_wg].append(ti)
+        flat, offs = [], [0]
+        for q in queues:
+            flat.extend(q)
+            offs.append(len(flat))
+        return (buf.to(device),
+                torch.tensor(flat, dtype=torch.int32, device=device),
+                torch.tensor(offs, dtype=torch.int32, device=device),
+                torch.zeros(len(self.ops), dtype=torch.int32, device=device))
+
+
+class MegaRun:
+    """Owns the finalized descriptors; one launch per decode step."""
+
+    def __init__(self, graph: MegaGraph, n_wg: int = 512, device="cuda"):
+        self.n_wg = n_wg
+        (self.task_buf, self.queue, self.queue_off,
+         self.scoreboard) = graph.finalize(n_wg, device)
+        self.n_ops = len(graph.ops)
+
+    def launch(self, stream=None):
+        from .. import _C
+
+        s = stream or torch.cuda.current_stream()
+        self.scoreboard.zero_()
+        _C.megakernel(self.task_buf.data_ptr(), self.queue.data_ptr(),
+                      self.queue_off.data_ptr(), self.scoreboard.data_ptr(),
+                      self.n_wg, s.cuda_stream)
