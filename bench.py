@@ -39,7 +39,7 @@ def main():
     import torch.distributed as dist
 
     import triton_dist_amd as td
-    from triton_dist_amd.models import DenseLLM, Engine, get_config
+    from triton_dist_amd.models import AutoLLM, Engine, get_config
 
     td.initialize_distributed(seed=args.seed)
     world = td.world_size()
@@ -55,7 +55,7 @@ def main():
     max_len = args.ctx + args.warmup + args.steps + 16
 
     cfg = get_config(args.model, tp_mode=args.mode, max_length=max_len + 64)
-    model = DenseLLM(cfg, device=device)
+    model = AutoLLM(cfg, device=device)
     model.init_weights(seed=args.seed)
     if args.mode in ("ag_rs", "gemm_ar"):
         model.init_dist_ctx(max_m_total=batch)
