@@ -119,6 +119,16 @@ void launch_moe_dispatch(const PeerTable &pt, const void *x,
 void launch_moe_wait_flags(const void *flags, int world, const void *cell,
                            hipStream_t stream);
 void launch_bump_cell(void *cell, hipStream_t stream);
+void launch_moe_dispatch_fp8(const PeerTable &pt, const void *x,
+                             const void *topk_ids, const void *send_pos,
+                             const void *send_base, const void *send_to_dst,
+                             size_t recv_q_off, size_t recv_s_off,
+                             size_t meta_off, size_t flags_off,
+                             unsigned *arrive, const void *val_cell, int T,
+                             int K, int H, int e_loc, hipStream_t stream);
+void launch_moe_dequant(const void *rq, const void *rs, void *out,
+                        const void *recv_total, int cap, int H,
+                        hipStream_t stream);
 void launch_wait_flags_ge_cell(const void *flags, int n, const void *cell,
                                int delta, hipStream_t stream);
 void launch_signal_credit(const PeerTable &pt, size_t credit_off,

@@ -153,6 +153,97 @@ __global__ void k_moe_dispatch(PeerTable pt, const bf16 *__restrict__ x,
   }
 }
 
+// fp8 online-quant dispatch (reference LL layer capability: fp8 payload +
+// groupwise scales, layers/amd/ep_ll_a2a_layer.py:46-190 — behavior only).
+// Per 128-element group: scale = max|x|/448; payload OCP e4m3 via the
+// gfx950 cvt_pk_fp8_f32 instruction; halves the xGMI wire bytes.
+__global__ void k_moe_dispatch_fp8(PeerTable pt, const bf16 *__restrict__ x,
+                                   const int *__restrict__ topk_ids,
+                                   const int *__restrict__ send_pos,
+                                   const int *__restrict__ send_base,
+                                   const int *__restrict__ send_to_dst,
+                                   size_t recv_q_off, size_t recv_s_off,
+                                   size_t meta_off, size_t flags_off,
+                                   unsigned *arrive, const int *val_cell,
+                                   int T, int K, int H, int e_loc) {
+  const int i = blockIdx.x;
+  const int e = topk_ids[i];
+  const int pos = send_pos[i];
+  if (e < 0 || pos < 0) return;
+  const int dst = e / e_loc;
+  const int t = i / K;
+  const int slot = send_base[e] + pos;
+  unsigned char *rq =
+      (unsigned char *)((char *)pt.bases[dst] + recv_q_off) + (size_t)slot * H;
+  float *rs = (float *)((char *)pt.bases[dst] + recv_s_off) +
+              (size_t)slot * (H / 128);
+  const bf16 *src = x + (size_t)t * H;
+  const int lane16 = threadIdx.x & 15;  // 16 threads per 128-elem group
+  for (int c = threadIdx.x * 8; c < H; c += blockDim.x * 8) {
+    bf16x8 v = *(const bf16x8 *)(src + c);
+    float f[8];
+    float amax = 0.f;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      f[j] = (float)v[j];
+      amax = fmaxf(amax, fabsf(f[j]));
+    }
+    // group max across the 16 lanes covering this 128-elem group
+#pragma unroll
+    for (int off = 1; off < 16; off <<= 1)
+      amax = fmaxf(amax, __shfl_xor(amax, off));
+    float scale = amax / 448.f + 1e-12f;
+    float inv = 1.f / scale;
+    int lo = 0, hi = 0;
+    lo = __builtin_amdgcn_cvt_pk_fp8_f32(f[0] * inv, f[1] * inv, lo, false);
+    lo = __builtin_amdgcn_cvt_pk_fp8_f32(f[2] * inv, f[3] * inv, lo, true);
+    hi = __builtin_amdgcn_cvt_pk_fp8_f32(f[4] * inv, f[5] * inv, hi, false);
+    hi = __builtin_amdgcn_cvt_pk_fp8_f32(f[6] * inv, f[7] * inv, hi, true);
+    int2 packed = make_int2(lo, hi);
+    *(int2 *)(rq + c) = packed;
+    if (lane16 == 0) rs[c / 128] = scale;
+  }
+  if (threadIdx.x == 0) {
+    int *meta = (int *)((char *)pt.bases[dst] + meta_off);
+    meta[slot * 2] = pt.rank;
+    meta[slot * 2 + 1] = i;
+    fence_release_sys();
+    unsigned prev = atomic_add<Scope::Gpu>(arrive + dst, 1u);
+    if ((int)prev == send_to_dst[dst] - 1) {
+      int *fl = (int *)((char *)pt.bases[dst] + flags_off);
+      st_release<Scope::Sys>(fl + pt.rank, val_cell ? *val_cell : 1);
+    }
+  }
+}
+
+// dequantize received fp8 rows -> bf16 recv_x (capacity loop; garbage rows
+// beyond recv_total are skipped)
+__global__ void k_moe_dequant(const unsigned char *__restrict__ rq,
+                              const float *__restrict__ rs,
+                              bf16 *__restrict__ out,
+                              const int *__restrict__ recv_total, int H) {
+  const int r = blockIdx.x;
+  if (r >= recv_total[0]) return;
+  const unsigned char *q = rq + (size_t)r * H;
+  const float *sc = rs + (size_t)r * (H / 128);
+  bf16 *o = out + (size_t)r * H;
+  for (int c = threadIdx.x * 8; c < H; c += blockDim.x * 8) {
+    int2 packed = *(const int2 *)(q + c);
+    float scale = sc[c / 128];
+    typedef __attribute__((ext_vector_type(2))) float f32x2;
+    f32x2 f01 = __builtin_amdgcn_cvt_pk_f32_fp8(packed.x, false);
+    f32x2 f23 = __builtin_amdgcn_cvt_pk_f32_fp8(packed.x, true);
+    f32x2 f45 = __builtin_amdgcn_cvt_pk_f32_fp8(packed.y, false);
+    f32x2 f67 = __builtin_amdgcn_cvt_pk_f32_fp8(packed.y, true);
+    bf16x8 v;
+    v[0] = (bf16)(f01[0] * scale); v[1] = (bf16)(f01[1] * scale);
+    v[2] = (bf16)(f23[0] * scale); v[3] = (bf16)(f23[1] * scale);
+    v[4] = (bf16)(f45[0] * scale); v[5] = (bf16)(f45[1] * scale);
+    v[6] = (bf16)(f67[0] * scale); v[7] = (bf16)(f67[1] * scale);
+    *(bf16x8 *)(o + c) = v;
+  }
+}
+
 // signal destinations that get ZERO copies from me (they still wait on my
 // flag), and likewise for the combine side.
 __global__ void k_moe_signal_empty(PeerTable pt, const int *__restrict__ cnt,
@@ -602,6 +693,33 @@ void launch_moe_dispatch(const PeerTable &pt, const void *x,
   hipLaunchKernelGGL(k_moe_signal_empty, dim3(1), dim3(kWave), 0, stream, pt,
                      (const int *)send_to_dst, flags_off,
                      (const int *)val_cell);
+}
+
+void launch_moe_dispatch_fp8(const PeerTable &pt, const void *x,
+                             const void *topk_ids, const void *send_pos,
+                             const void *send_base, const void *send_to_dst,
+                             size_t recv_q_off, size_t recv_s_off,
+                             size_t meta_off, size_t flags_off,
+                             unsigned *arrive, const void *val_cell, int T,
+                             int K, int H, int e_loc, hipStream_t stream) {
+  if (H % 128) throw std::runtime_error("fp8 dispatch: H % 128 != 0");
+  hipLaunchKernelGGL(k_moe_dispatch_fp8, dim3(T * K), dim3(256), 0, stream,
+                     pt, (const bf16 *)x, (const int *)topk_ids,
+                     (const int *)send_pos, (const int *)send_base,
+                     (const int *)send_to_dst, recv_q_off, recv_s_off,
+                     meta_off, flags_off, arrive, (const int *)val_cell, T,
+                     K, H, e_loc);
+  hipLaunchKernelGGL(k_moe_signal_empty, dim3(1), dim3(kWave), 0, stream, pt,
+                     (const int *)send_to_dst, flags_off,
+                     (const int *)val_cell);
+}
+
+void launch_moe_dequant(const void *rq, const void *rs, void *out,
+                        const void *recv_total, int cap, int H,
+                        hipStream_t stream) {
+  hipLaunchKernelGGL(k_moe_dequant, dim3(cap), dim3(256), 0, stream,
+                     (const unsigned char *)rq, (const float *)rs,
+                     (bf16 *)out, (const int *)recv_total, H);
 }
 
 void launch_bump_cell(void *cell, hipStream_t stream) {

@@ -451,6 +451,31 @@ static void moe_wait_flags(uintptr_t flags, int world, uintptr_t stream,
   TD_CHECK_HIP(hipGetLastError());
 }
 
+static void moe_dispatch_fp8(uintptr_t x, uintptr_t topk_ids,
+                             uintptr_t send_pos, uintptr_t send_base,
+                             uintptr_t send_to_dst, size_t recv_q_off,
+                             size_t recv_s_off, size_t meta_off,
+                             size_t flags_off, uintptr_t arrive, int T,
+                             int K, int H, int e_loc, uintptr_t stream,
+                             uintptr_t val_cell = 0) {
+  check_active();
+  launch_moe_dispatch_fp8(g_heap.pt, (void *)x, (void *)topk_ids,
+                          (void *)send_pos, (void *)send_base,
+                          (void *)send_to_dst, recv_q_off, recv_s_off,
+                          meta_off, flags_off, (unsigned *)arrive,
+                          (void *)val_cell, T, K, H, e_loc,
+                          as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
+static void moe_dequant(uintptr_t rq, uintptr_t rs, uintptr_t out,
+                        uintptr_t recv_total, int cap, int H,
+                        uintptr_t stream) {
+  launch_moe_dequant((void *)rq, (void *)rs, (void *)out, (void *)recv_total,
+                     cap, H, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
 static void bump_cell(uintptr_t cell, uintptr_t stream) {
   launch_bump_cell((void *)cell, as_stream(stream));
   TD_CHECK_HIP(hipGetLastError());
@@ -630,6 +655,13 @@ PYBIND11_MODULE(_C, m) {
         py::arg("recv_x_off"), py::arg("meta_off"), py::arg("flags_off"),
         py::arg("arrive"), py::arg("T"), py::arg("K"), py::arg("H"),
         py::arg("e_loc"), py::arg("stream"), py::arg("val_cell") = 0);
+  m.def("moe_dispatch_fp8", &moe_dispatch_fp8, py::arg("x"),
+        py::arg("topk_ids"), py::arg("send_pos"), py::arg("send_base"),
+        py::arg("send_to_dst"), py::arg("recv_q_off"), py::arg("recv_s_off"),
+        py::arg("meta_off"), py::arg("flags_off"), py::arg("arrive"),
+        py::arg("T"), py::arg("K"), py::arg("H"), py::arg("e_loc"),
+        py::arg("stream"), py::arg("val_cell") = 0);
+  m.def("moe_dequant", &moe_dequant);
   m.def("bump_cell", &bump_cell);
   m.def("wait_flags_ge_cell", &wait_flags_ge_cell);
   m.def("signal_credit", &signal_credit);

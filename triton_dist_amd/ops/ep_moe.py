@@ -41,6 +41,9 @@ class EPContext:
     low_latency: bool = False
     credit_flags: Optional[SymmBuffer] = None
     call_no: int = 0
+    fp8: bool = False
+    recv_q: Optional[SymmBuffer] = None
+    recv_scale: Optional[SymmBuffer] = None
 
     @property
     def world(self):
@@ -57,7 +60,8 @@ class EPContext:
 
 def create_ep_context(max_tokens: int, hidden: int, n_experts: int,
                       topk: int, heap: Optional[SymmHeap] = None,
-                      low_latency: bool = False) -> EPContext:
+                      low_latency: bool = False,
+                      fp8: bool = False) -> EPContext:
     """low_latency=True allocates DOUBLE buffers (call-parity indexed) and
     the op skips the entry barrier + flag resets: flags carry the running
     call number and a credit flag bounds buffer reuse — the reference's LL
@@ -84,6 +88,12 @@ def create_ep_context(max_tokens: int, hidden: int, n_experts: int,
     ctx.credit_flags = heap.alloc_buffer((world,), torch.int32) \
         if low_latency else None
     ctx.call_no = 0
+    ctx.fp8 = fp8
+    if fp8:
+        assert hidden % 128 == 0
+        ctx.recv_q = heap.alloc_buffer((nbuf, cap, hidden), torch.uint8)
+        ctx.recv_scale = heap.alloc_buffer((nbuf, cap, hidden // 128),
+                                           torch.float32)
     if heap.backend == "hip":
         dev = "cuda"
         e = n_experts
@@ -197,13 +207,31 @@ def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
                   L["recv_total"].data_ptr(), s,
                   L["work_items"].data_ptr(), L["work_count"].data_ptr(), bm)
 
-    # phase 4: dispatch (xGMI row push + per-dst completion signals)
-    _C.moe_dispatch(x.data_ptr(), topk_ids.data_ptr(),
-                    L["send_pos"].data_ptr(), L["send_base"].data_ptr(),
-                    L["send_to_dst"].data_ptr(), recv_x_off,
-                    meta_off, dflags_off,
-                    L["arrive_d"].data_ptr(), T, K, H, e_loc, s, cell)
-    _C.moe_wait_flags(heap.ptr(rank, dflags_off), world, s, cell)
+    # phase 4: dispatch (xGMI row push + per-dst completion signals);
+    # fp8 mode quantizes the payload (groupwise scales) and dequantizes
+    # after the wait — halves the wire bytes
+    if ctx.fp8:
+        recv_q_off = ctx.recv_q.offset + parity * ctx.cap * H
+        recv_s_off = ctx.recv_scale.offset + parity * ctx.cap * (H // 128) * 4
+        _C.moe_dispatch_fp8(x.data_ptr(), topk_ids.data_ptr(),
+                            L["send_pos"].data_ptr(),
+                            L["send_base"].data_ptr(),
+                            L["send_to_dst"].data_ptr(), recv_q_off,
+                            recv_s_off, meta_off, dflags_off,
+                            L["arrive_d"].data_ptr(), T, K, H, e_loc, s,
+                            cell)
+        _C.moe_wait_flags(heap.ptr(rank, dflags_off), world, s, cell)
+        _C.moe_dequant(heap.ptr(rank, recv_q_off),
+                       heap.ptr(rank, recv_s_off),
+                       heap.ptr(rank, recv_x_off),
+                       L["recv_total"].data_ptr(), ctx.cap, H, s)
+    else:
+        _C.moe_dispatch(x.data_ptr(), topk_ids.data_ptr(),
+                        L["send_pos"].data_ptr(), L["send_base"].data_ptr(),
+                        L["send_to_dst"].data_ptr(), recv_x_off,
+                        meta_off, dflags_off,
+                        L["arrive_d"].data_ptr(), T, K, H, e_loc, s, cell)
+        _C.moe_wait_flags(heap.ptr(rank, dflags_off), world, s, cell)
 
     # phase 5: grouped expert FFN — persistent work-queue kernel when
     # experts are lightly loaded (decode: ~T*K*world/E rows per expert)
