@@ -194,13 +194,28 @@ TD_DEV void epilogue(f32x4 acc[8][4], const WaveCtx &w, bf16 *lds_c,
 // ---------------------------------------------------------------------------
 using namespace g256;
 
+// GROUP_M supertile mapping: consecutive wgids (which the XCD remap makes
+// contiguous per XCD) walk a GMx(tiles_n) band column-major, so an XCD's
+// window reuses GM A-panels and a narrow set of B-panels from L2/LLC
+// instead of streaming every B panel per tile row.
+TD_DEV void tile_coords(int wgid, int tiles_m, int tiles_n, int &pid_m,
+                        int &pid_n) {
+  constexpr int GM = 4;
+  int group = wgid / (GM * tiles_n);
+  int first_m = group * GM;
+  int gsz = min(tiles_m - first_m, GM);
+  pid_m = first_m + (wgid % (GM * tiles_n)) % gsz;
+  pid_n = (wgid % (GM * tiles_n)) / gsz;
+}
+
 __global__ __launch_bounds__(NTH, 2) void k_gemm256_bf16(GemmArgs args) {
   __shared__ bf16 lds_a[SLICES * SLICE_ELEMS];
   __shared__ bf16 lds_b[SLICES * SLICE_ELEMS];
   const int tiles_n = args.n / BN;
   const int tiles_m = args.m / BM;
   int wgid = xcd_remap(blockIdx.x, tiles_m * tiles_n);
-  int pid_m = wgid / tiles_n, pid_n = wgid % tiles_n;
+  int pid_m, pid_n;
+  tile_coords(wgid, tiles_m, tiles_n, pid_m, pid_n);
   WaveCtx w = wave_ctx();
   f32x4 acc[8][4] = {};
   const bf16 *ga = (const bf16 *)args.a + (size_t)pid_m * BM * args.lda;
@@ -219,7 +234,8 @@ __global__ __launch_bounds__(NTH, 2) void k_ag_gemm256_consumer_bf16(
   const int tiles_m = g.m / BM;
   const int tiles_per_rank = args.m_per_rank / BM;
   int wgid = xcd_remap(blockIdx.x, tiles_m * tiles_n);
-  int pid_m = wgid / tiles_n, pid_n = wgid % tiles_n;
+  int pid_m, pid_n;
+  tile_coords(wgid, tiles_m, tiles_n, pid_m, pid_n);
   pid_m = (pid_m + args.rank * tiles_per_rank) % tiles_m;
   int rows_per_chunk = args.m_per_rank / args.chunks_per_rank;
   int c_lo = (pid_m * BM) / rows_per_chunk;
@@ -251,7 +267,8 @@ __global__ __launch_bounds__(NTH, 2) void k_gemm256_rs_producer_bf16(
   const int tiles_m = g.m / BM;
   const int tiles_per_rank = args.m_per_rank / BM;
   int wgid = xcd_remap(blockIdx.x, tiles_m * tiles_n);
-  int pid_m = wgid / tiles_n, pid_n = wgid % tiles_n;
+  int pid_m, pid_n;
+  tile_coords(wgid, tiles_m, tiles_n, pid_m, pid_n);
   pid_m = (pid_m + (args.rank + 1) * tiles_per_rank) % tiles_m;
   WaveCtx w = wave_ctx();
   f32x4 acc[8][4] = {};
