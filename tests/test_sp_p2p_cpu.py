@@ -72,3 +72,21 @@ def test_p2p_ring_cpu_2rank():
 
 def test_p2p_ring_cpu_4rank():
     run_distributed(_body_p2p, world_size=4)
+
+
+def _body_ulysses_roundtrip(rank, world):
+    from triton_dist_amd.layers import UlyssesSPAllToAllLayer
+
+    t_loc, heads, d = 6, 4 * world, 16
+    layer = UlyssesSPAllToAllLayer(heads, d)
+    layer.init_ctx(t_loc)
+    g = torch.Generator().manual_seed(4 + rank)
+    x = torch.randn(t_loc, heads, d, generator=g).to(torch.bfloat16)
+    y = layer.pre_attn(x)            # [world*t_loc, heads/world, d]
+    assert y.shape == (world * t_loc, heads // world, d)
+    back = layer.post_attn(y.clone())
+    assert torch.equal(back, x), "ulysses pre+post must round-trip"
+
+
+def test_ulysses_layer_roundtrip_cpu_2rank():
+    run_distributed(_body_ulysses_roundtrip, world_size=2)
