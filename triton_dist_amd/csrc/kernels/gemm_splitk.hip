@@ -92,8 +92,8 @@ template <bool WAIT_FLAGS>
 __global__ __launch_bounds__(sk::NTH) void k_gemm_splitk_bf16(
     GemmArgs g, float *ws, int splits, const int *flags, int chunks_per_rank,
     int m_per_rank, int rank, int expect) {
-  __shared__ bf16 lds_a[3 * sk::BM * sk::BK];
-  __shared__ bf16 lds_b[3 * sk::BN * sk::BK];
+  __shared__ bf16 lds_a[sk::BM * sk::BK];
+  __shared__ bf16 lds_b[sk::BN * sk::BK];
   const int tiles_n = g.n / sk::BN;
   const int tiles_m = g.m / sk::BM;
   int wgid = xcd_remap(blockIdx.x, tiles_m * tiles_n);
@@ -118,30 +118,12 @@ __global__ __launch_bounds__(sk::NTH) void k_gemm_splitk_bf16(
   f32x4 acc[4][4] = {};
   const bf16 *ga = (const bf16 *)g.a + (size_t)pid_m * sk::BM * g.lda + k_lo;
   const bf16 *gb = (const bf16 *)g.b + (size_t)pid_n * sk::BN * g.ldb + k_lo;
-  // 3-buffer pipelined K loop (8 loads/thread/step, 2 steps in flight,
-  // vmcnt(8) steady — same discipline as the other pipelined kernels)
-  const int ksteps = kc / sk::BK;
-  sk_stage(ga, gb, g.lda, g.ldb, lds_a, lds_b);
-  if (ksteps > 1)
-    sk_stage(ga + sk::BK, gb + sk::BK, g.lda, g.ldb,
-             lds_a + sk::BM * sk::BK, lds_b + sk::BN * sk::BK);
-  for (int ks = 0; ks < ksteps; ++ks) {
-    const int buf = ks % 3;
-    if (ks + 1 < ksteps) {
-      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
-    } else {
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    }
-    __builtin_amdgcn_s_barrier();
-    __builtin_amdgcn_sched_barrier(0);
-    if (ks + 2 < ksteps)
-      sk_stage(ga + (ks + 2) * sk::BK, gb + (ks + 2) * sk::BK, g.lda, g.ldb,
-               lds_a + ((ks + 2) % 3) * sk::BM * sk::BK,
-               lds_b + ((ks + 2) % 3) * sk::BN * sk::BK);
-    sk_compute(lds_a + buf * sk::BM * sk::BK, lds_b + buf * sk::BN * sk::BK,
-               t, acc);
-    __builtin_amdgcn_s_barrier();
-    __builtin_amdgcn_sched_barrier(0);
+  for (int k0 = 0; k0 < kc; k0 += sk::BK) {
+    sk_stage(ga + k0, gb + k0, g.lda, g.ldb, lds_a, lds_b);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    sk_compute(lds_a, lds_b, t, acc);
+    __syncthreads();
   }
   float *wsp = ws + (size_t)split * g.m * g.n +
                (size_t)pid_m * sk::BM * g.n + pid_n * sk::BN;
