@@ -40,6 +40,17 @@ TD_DEV f32x4 sk_mfma16(bf16x8 a, bf16x8 b, f32x4 c) {
   return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
 }
 
+// GROUP_M supertile (see gemm256.hip tile_coords): L2 reuse of A/B panels
+TD_DEV void sk_tile_coords(int wgid, int tiles_m, int tiles_n, int &pid_m,
+                           int &pid_n) {
+  constexpr int GM = 4;
+  int group = wgid / (GM * tiles_n);
+  int first_m = group * GM;
+  int gsz = min(tiles_m - first_m, GM);
+  pid_m = first_m + (wgid % (GM * tiles_n)) % gsz;
+  pid_n = (wgid % (GM * tiles_n)) / gsz;
+}
+
 TD_DEV void sk_stage(const bf16 *ga, const bf16 *gb, int lda, int ldb,
                      bf16 *lds_a, bf16 *lds_b) {
   int tid = threadIdx.x;
@@ -97,7 +108,8 @@ __global__ __launch_bounds__(sk::NTH) void k_gemm_splitk_bf16(
   const int tiles_n = g.n / sk::BN;
   const int tiles_m = g.m / sk::BM;
   int wgid = xcd_remap(blockIdx.x, tiles_m * tiles_n);
-  int pid_m = wgid / tiles_n, pid_n = wgid % tiles_n;
+  int pid_m, pid_n;
+  sk_tile_coords(wgid, tiles_m, tiles_n, pid_m, pid_n);
   if (WAIT_FLAGS) {
     const int tiles_per_rank = m_per_rank / sk::BM;
     pid_m = (pid_m + rank * tiles_per_rank) % tiles_m;
