@@ -94,8 +94,14 @@ class SymmHeap:
             from .. import _C
             self._C = _C
             self.device = local_device()
+            # coarse-grained (hipMalloc) by default: fine-grained allocations
+            # bypass L2 on loads, which measured 1.8x slower AG-consumer
+            # GEMM A-panel reads. Flag correctness over xGMI relies on
+            # system-scope acquire/release (buffer_inv/wbl2), which works on
+            # coarse memory; SDMA writes are L2-coherent at the home node.
+            fine = os.environ.get("TD_HEAP_FINEGRAINED", "0") == "1"
             handle = _C.heap_init(self.rank, self.world, self.device,
-                                  self.size, True)
+                                  self.size, fine)
             handles = [None] * self.world
             if dist.is_initialized():
                 dist.all_gather_object(handles, bytes(handle), group=group)
