@@ -18,7 +18,7 @@ typedef __attribute__((ext_vector_type(8))) bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 namespace sk {
-constexpr int BM = 128, BN = 128, BK = 64;
+constexpr int BM = 128, BN = 128, BK = 32;
 constexpr int NTH = 256;
 }  // namespace sk
 
@@ -56,10 +56,10 @@ TD_DEV void sk_stage(const bf16 *ga, const bf16 *gb, int lda, int ldb,
   int tid = threadIdx.x;
   int wave = tid >> 6;
 #pragma unroll
-  for (int it = 0; it < 4; ++it) {
+  for (int it = 0; it < 2; ++it) {  // 128*32/8 = 512 chunks per matrix
     int idx = it * sk::NTH + tid;
-    int row = idx >> 3;
-    int kc = idx & 7;
+    int row = idx >> 2;             // BK/8 = 4 chunks per row
+    int kc = idx & 3;
     const bf16 *sa = ga + (size_t)row * lda + kc * 8;
     const bf16 *sb = gb + (size_t)row * ldb + kc * 8;
     int wave_chunk0 = it * sk::NTH + wave * 64;
@@ -78,22 +78,19 @@ TD_DEV void sk_stage(const bf16 *ga, const bf16 *gb, int lda, int ldb,
 
 TD_DEV void sk_compute(const bf16 *lds_a, const bf16 *lds_b,
                        const TileCtx &t, f32x4 acc[4][4]) {
+  bf16x8 af[4], bfr[4];
 #pragma unroll
-  for (int ks = 0; ks < sk::BK / 32; ++ks) {
-    bf16x8 af[4], bfr[4];
-#pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      int arow = t.wr * 64 + i * 16 + (t.lane & 15);
-      int brow = t.wc * 64 + i * 16 + (t.lane & 15);
-      int k0 = ks * 32 + (t.lane >> 4) * 8;
-      af[i] = *(const bf16x8 *)(lds_a + arow * sk::BK + k0);
-      bfr[i] = *(const bf16x8 *)(lds_b + brow * sk::BK + k0);
-    }
-#pragma unroll
-    for (int i = 0; i < 4; ++i)
-#pragma unroll
-      for (int j = 0; j < 4; ++j) acc[i][j] = sk_mfma16(af[i], bfr[j], acc[i][j]);
+  for (int i = 0; i < 4; ++i) {
+    int arow = t.wr * 64 + i * 16 + (t.lane & 15);
+    int brow = t.wc * 64 + i * 16 + (t.lane & 15);
+    int k0 = (t.lane >> 4) * 8;
+    af[i] = *(const bf16x8 *)(lds_a + arow * sk::BK + k0);
+    bfr[i] = *(const bf16x8 *)(lds_b + brow * sk::BK + k0);
   }
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = sk_mfma16(af[i], bfr[j], acc[i][j]);
 }
 
 // Body shared by plain and AG-consumer variants. Writes fp32 partials to
@@ -103,8 +100,8 @@ template <bool WAIT_FLAGS>
 __global__ __launch_bounds__(sk::NTH) void k_gemm_splitk_bf16(
     GemmArgs g, float *ws, int splits, const int *flags, int chunks_per_rank,
     int m_per_rank, int rank, int expect) {
-  __shared__ bf16 lds_a[sk::BM * sk::BK];
-  __shared__ bf16 lds_b[sk::BN * sk::BK];
+  __shared__ bf16 lds_a[3 * sk::BM * sk::BK];
+  __shared__ bf16 lds_b[3 * sk::BN * sk::BK];
   const int tiles_n = g.n / sk::BN;
   const int tiles_m = g.m / sk::BM;
   int wgid = xcd_remap(blockIdx.x, tiles_m * tiles_n);
@@ -130,12 +127,28 @@ __global__ __launch_bounds__(sk::NTH) void k_gemm_splitk_bf16(
   f32x4 acc[4][4] = {};
   const bf16 *ga = (const bf16 *)g.a + (size_t)pid_m * sk::BM * g.lda + k_lo;
   const bf16 *gb = (const bf16 *)g.b + (size_t)pid_n * sk::BN * g.ldb + k_lo;
-  for (int k0 = 0; k0 < kc; k0 += sk::BK) {
-    sk_stage(ga + k0, gb + k0, g.lda, g.ldb, lds_a, lds_b);
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __syncthreads();
-    sk_compute(lds_a, lds_b, t, acc);
-    __syncthreads();
+  // 3-buffer pipelined K loop: 4 loads/thread/step, 2 steps in flight,
+  // vmcnt(4) steady; 48KB LDS keeps 3 blocks/CU
+  constexpr int AB = sk::BM * sk::BK;
+  const int ksteps = kc / sk::BK;
+  sk_stage(ga, gb, g.lda, g.ldb, lds_a, lds_b);
+  if (ksteps > 1)
+    sk_stage(ga + sk::BK, gb + sk::BK, g.lda, g.ldb, lds_a + AB, lds_b + AB);
+  for (int ks = 0; ks < ksteps; ++ks) {
+    const int buf = ks % 3;
+    if (ks + 1 < ksteps) {
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+    __builtin_amdgcn_sched_barrier(0);
+    if (ks + 2 < ksteps)
+      sk_stage(ga + (ks + 2) * sk::BK, gb + (ks + 2) * sk::BK, g.lda, g.ldb,
+               lds_a + ((ks + 2) % 3) * AB, lds_b + ((ks + 2) % 3) * AB);
+    sk_compute(lds_a + buf * AB, lds_b + buf * AB, t, acc);
+    __builtin_amdgcn_s_barrier();
+    __builtin_amdgcn_sched_barrier(0);
   }
   float *wsp = ws + (size_t)split * g.m * g.n +
                (size_t)pid_m * sk::BM * g.n + pid_n * sk::BN;

@@ -31,7 +31,7 @@ def choose_splits(m: int, n: int, k: int) -> int:
     if grid >= 208:
         return 1
     for s in (8, 4, 2):
-        if (k // 64) % s == 0 and k // s >= 512 and grid * s <= 2048:
+        if (k // 32) % s == 0 and k // s >= 512 and grid * s <= 2048:
             return s
     return 1
 
