@@ -111,3 +111,25 @@ def _body_gemm_rs(rank, world):
 
 def test_gemm_rs_2rank():
     run_distributed(_body_gemm_rs, world_size=2)
+
+
+def _body_ag_pull(rank, world):
+    from triton_dist_amd.ops import allgather, create_ag_gemm_context
+    import torch.distributed as dist
+
+    m, k = 256, 512
+    ctx = create_ag_gemm_context(max_m_per_rank=m, k=k, chunks_per_rank=4)
+    torch.manual_seed(21 + rank)
+    a = (torch.randn(m, k, device="cuda") / 8).to(torch.bfloat16)
+    for _ in range(2):
+        g = allgather(a, ctx, method="pull")
+        torch.cuda.synchronize()
+        # golden
+        cpu = a.cpu()
+        full = torch.empty(world * m, k, dtype=torch.bfloat16)
+        dist.all_gather_into_tensor(full, cpu)
+        assert torch.equal(g.cpu(), full), "pull allgather mismatch"
+
+
+def test_allgather_pull_2rank():
+    run_distributed(_body_ag_pull, world_size=2)

@@ -28,6 +28,9 @@ void launch_put_signal(const PeerTable &pt, void *dst, const void *src,
                        size_t nbytes, int *flag, int val, int add,
                        hipStream_t stream);
 void launch_reset_flags(int *flags, int n, int val, hipStream_t stream);
+void launch_ag_pull(const PeerTable &pt, size_t ws_off, size_t flags_off,
+                    size_t seg_bytes, int chunks, int chunk_stride,
+                    hipStream_t stream);
 
 // kernels/gemm.hip -----------------------------------------------------------
 struct GemmArgs {

@@ -110,6 +110,43 @@ void launch_copy(void *dst, const void *src, size_t nbytes, hipStream_t stream) 
 }
 
 // ---------------------------------------------------------------------------
+// Pull-mode allgather: block (peer, chunk) spin-waits the SOURCE rank's
+// published chunk flag over xGMI, then pulls the chunk into the local
+// gathered workspace (cf. reference kernels/amd/allgather.py pull variants
+// :173-313 — behavior only). Pull reads ride the consumer GPU's own xGMI
+// links, which balances differently than push under contention.
+// ---------------------------------------------------------------------------
+__global__ void k_ag_pull(PeerTable pt, size_t ws_off, size_t flags_off,
+                          size_t seg_bytes, int chunks, int chunk_stride) {
+  const int pi = blockIdx.y;  // peer index (skip self)
+  const int peer = (pt.rank + 1 + pi) % pt.world;
+  const int c = blockIdx.x;
+  const size_t per = (seg_bytes + chunks - 1) / chunks;
+  const size_t lo = (size_t)c * per;
+  const size_t hi = min(lo + per, seg_bytes);
+  const int *pf = (const int *)((char *)pt.bases[peer] + flags_off);
+  if (threadIdx.x == 0)
+    wait_ge_one<Scope::Sys>(pf + peer * chunk_stride + c, 1);
+  __syncthreads();
+  const ulonglong2 *src =
+      (const ulonglong2 *)((char *)pt.bases[peer] + ws_off +
+                           (size_t)peer * seg_bytes + lo);
+  ulonglong2 *dst = (ulonglong2 *)((char *)pt.bases[pt.rank] + ws_off +
+                                   (size_t)peer * seg_bytes + lo);
+  for (size_t i = threadIdx.x; i * 16 < hi - lo; i += blockDim.x)
+    dst[i] = src[i];
+}
+
+void launch_ag_pull(const PeerTable &pt, size_t ws_off, size_t flags_off,
+                    size_t seg_bytes, int chunks, int chunk_stride,
+                    hipStream_t stream) {
+  if (pt.world <= 1) return;
+  hipLaunchKernelGGL(k_ag_pull, dim3(chunks, pt.world - 1), dim3(256), 0,
+                     stream, pt, ws_off, flags_off, seg_bytes, chunks,
+                     chunk_stride);
+}
+
+// ---------------------------------------------------------------------------
 // put_signal: multi-block copy; last block to arrive does a system release
 // fence and sets the flag. Self-contained (no stream-order dependence), the
 // analog of rocshmem_putmem_signal (shmem/rocshmem_bind/runtime/

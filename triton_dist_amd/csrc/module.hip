@@ -245,6 +245,14 @@ static void reset_flags(uintptr_t flags, int n, int val, uintptr_t stream) {
   TD_CHECK_HIP(hipGetLastError());
 }
 
+static void ag_pull(size_t ws_off, size_t flags_off, size_t seg_bytes,
+                    int chunks, int chunk_stride, uintptr_t stream) {
+  check_active();
+  launch_ag_pull(g_heap.pt, ws_off, flags_off, seg_bytes, chunks,
+                 chunk_stride, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
 static void copy_kernel(uintptr_t dst, uintptr_t src, size_t nbytes,
                         uintptr_t stream) {
   launch_copy(reinterpret_cast<void *>(dst), reinterpret_cast<void *>(src),
@@ -625,6 +633,7 @@ PYBIND11_MODULE(_C, m) {
   m.def("signal_set", &signal_set);
   m.def("wait_eq", &wait_eq_host);
   m.def("reset_flags", &reset_flags);
+  m.def("ag_pull", &ag_pull);
   m.def("copy_kernel", &copy_kernel);
   m.def("put_signal", &put_signal);
   m.def("probe_mfma", &probe_mfma);

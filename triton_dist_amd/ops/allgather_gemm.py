@@ -168,11 +168,18 @@ def ag_gemm(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
 
 
 def allgather(a: torch.Tensor, ctx: AGGemmContext,
-              out: Optional[torch.Tensor] = None) -> torch.Tensor:
-    """Standalone push all-gather through the ctx workspace (the consumer is
-    a wait kernel instead of the fused GEMM; cf. reference
-    kernels/amd/allgather.py capability). Returns [world*m, K]."""
+              out: Optional[torch.Tensor] = None,
+              method: str = "push") -> torch.Tensor:
+    """Standalone all-gather through the ctx workspace (cf. reference
+    kernels/amd/allgather.py push/pull variants :40-313 — capability).
+
+    push: SDMA producer streams + flag copies (default, CP-engine path)
+    pull: each rank publishes its shard locally; one kernel pulls every
+          peer segment over the CONSUMER's xGMI links (k_ag_pull).
+    Returns [world*m, K]."""
     m, k = a.shape
+    if method == "pull" and ctx.heap.backend == "hip" and ctx.world > 1:
+        return _allgather_pull(a, ctx, out)
     assert k == ctx.k and m == ctx.max_m_per_rank
     world, rank = ctx.world, ctx.rank
     chunks = ctx.chunks_per_rank
@@ -223,6 +230,29 @@ def allgather(a: torch.Tensor, ctx: AGGemmContext,
     for s in range(min(ns, max(world - 1, 1))):
         ctx.join_evs[s].record(ctx.comm_streams[s])
         compute.wait_event(ctx.join_evs[s])
+    gathered = ctx.ws.local().reshape(world * ctx.max_m_per_rank, k)
+    if out is not None:
+        out.copy_(gathered)
+        return out
+    return gathered
+
+
+def _allgather_pull(a: torch.Tensor, ctx: AGGemmContext,
+                    out: Optional[torch.Tensor]) -> torch.Tensor:
+    m, k = a.shape
+    world, rank = ctx.world, ctx.rank
+    heap, _C = ctx.heap, ctx.heap._C
+    compute = torch.cuda.current_stream()
+    s = compute.cuda_stream
+    chunks = ctx.chunks_per_rank
+    seg_bytes = ctx.max_m_per_rank * k * 2
+    # reset only MY flag row (sources publish their own chunks)
+    _C.reset_flags(ctx.flags.ptr() + rank * chunks * 4, chunks, 0, s)
+    heap.barrier_all_on_stream(compute)
+    _C.memcpy_async(ctx.ws.ptr() + rank * seg_bytes, a.data_ptr(), m * k * 2,
+                    s)
+    _C.reset_flags(ctx.flags.ptr() + rank * chunks * 4, chunks, 1, s)
+    _C.ag_pull(ctx.ws.offset, ctx.flags.offset, seg_bytes, chunks, chunks, s)
     gathered = ctx.ws.local().reshape(world * ctx.max_m_per_rank, k)
     if out is not None:
         out.copy_(gathered)
