@@ -34,21 +34,23 @@ __global__ __launch_bounds__(256) void k_flash_decode(
     const bf16 *__restrict__ vcache, bf16 *__restrict__ out,
     const long *__restrict__ offset, int qh, int kvh, int max_len,
     float scale) {
+  // K staged in LDS (16 reads/row); V read DIRECT from global in the PV
+  // loop — each half-wave reads a full 128-elem row coalesced and the 8
+  // head-groups hit L2 (guide: don't LDS-stage cache-resident data).
+  // 11 KB LDS -> high occupancy for latency hiding.
   const int b = blockIdx.x;
   const int kh = blockIdx.y;
-  const int G = qh / kvh;  // query heads per kv head (<= 8 supported)
+  const int G = qh / kvh;
   const int tid = threadIdx.x;
-  const int g = tid >> 5;       // 0..7 query-head slot
-  const int t = tid & 31;       // position within tile
+  const int g = tid >> 5;
+  const int t = tid & 31;
   const long seqlen = *offset + 1;
 
   __shared__ bf16 k_lds[kTile][kD];
-  __shared__ bf16 v_lds[kTile][kD];
   __shared__ float p_lds[8][kTile];
   __shared__ float m_lds[8], r_lds[8], l_lds[8];
   __shared__ bf16 q_lds[8][kD];
 
-  // load q for my block's G heads into LDS (pad missing heads with zeros)
   for (int i = tid; i < 8 * kD / 8; i += 256) {
     int hh = i / (kD / 8);
     int c = (i % (kD / 8)) * 8;
@@ -63,32 +65,24 @@ __global__ __launch_bounds__(256) void k_flash_decode(
   }
   __syncthreads();
 
-  // per-thread accumulator: 4 (g,d) slices — thread owns dims
-  // d = (tid%32)*4 ... for head slot tid/32
   float acc[4] = {};
   const int my_d0 = t * 4;
-
   const long ntiles = (seqlen + kTile - 1) / kTile;
   for (long tile = 0; tile < ntiles; ++tile) {
     const long pos0 = tile * kTile;
-    // stage K/V tile: 32 rows x 128 cols, 16B chunks: 32*16=512 chunks
     __syncthreads();
     for (int i = tid; i < kTile * kD / 8; i += 256) {
       int r = i / (kD / 8);
       int c = (i % (kD / 8)) * 8;
       long pos = pos0 + r;
-      bf16x8 kv{}, vv{};
+      bf16x8 kv{};
       if (pos < seqlen) {
         size_t base = (((size_t)b * max_len + pos) * kvh + kh) * kD + c;
         kv = *(const bf16x8 *)(kcache + base);
-        vv = *(const bf16x8 *)(vcache + base);
       }
       *(bf16x8 *)(&k_lds[r][c]) = kv;
-      *(bf16x8 *)(&v_lds[r][c]) = vv;
     }
     __syncthreads();
-
-    // scores: thread (g, t) -> dot(q[g], k[t])
     float s = -1e30f;
     if (g < G && pos0 + t < seqlen) {
       float d = 0.f;
@@ -101,11 +95,9 @@ __global__ __launch_bounds__(256) void k_flash_decode(
       }
       s = d * scale;
     }
-    // per-g max over the 32 positions (half-wave reduce)
     float mx = s;
     for (int off = 16; off > 0; off >>= 1)
       mx = fmaxf(mx, __shfl_xor(mx, off));
-    // online-softmax update (thread t==0 of each g publishes)
     float m_old = m_lds[g];
     float m_new = fmaxf(m_old, mx);
     float p = (s > -1e29f) ? __expf(s - m_new) : 0.f;
@@ -119,18 +111,18 @@ __global__ __launch_bounds__(256) void k_flash_decode(
       m_lds[g] = m_new;
     }
     __syncthreads();
-
-    // accumulate: acc[d] = acc[d]*r + sum_t p[t] * V[t][d]
     const float r = r_lds[g];
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[j] *= r;
-    for (int tt = 0; tt < kTile; ++tt) {
-      float p = p_lds[g][tt];
-      if (p != 0.f) {
+    const long lim = min((long)kTile, seqlen - pos0);
+    typedef __attribute__((ext_vector_type(4))) bf16 bf16x4;
+    const bf16 *vbase = vcache +
+        (((size_t)b * max_len + pos0) * kvh + kh) * kD + my_d0;
+    for (int tt = 0; tt < (int)lim; ++tt) {
+      float p2 = p_lds[g][tt];
+      bf16x4 vv = *(const bf16x4 *)(vbase + (size_t)tt * kvh * kD);
 #pragma unroll
-        for (int j = 0; j < 4; ++j)
-          acc[j] += p * (float)v_lds[tt][my_d0 + j];
-      }
+      for (int j = 0; j < 4; ++j) acc[j] += p2 * (float)vv[j];
     }
   }
   __syncthreads();
