@@ -34,6 +34,10 @@ PRESETS = {
     # Qwen3-32B geometry (HF Qwen/Qwen3-32B config.json)
     "qwen3-32b": dict(hidden=5120, intermediate=25600, n_layers=64,
                       n_heads=64, n_kv_heads=8, head_dim=128, vocab=151936),
+    # Seed-OSS-36B-class geometry (approximate: no HF access here)
+    "seed-oss-36b": dict(hidden=5120, intermediate=27648, n_layers=64,
+                         n_heads=80, n_kv_heads=8, head_dim=128,
+                         vocab=155136),
     # Qwen3-8B geometry
     "qwen3-8b": dict(hidden=4096, intermediate=12288, n_layers=36,
                      n_heads=32, n_kv_heads=8, head_dim=128, vocab=151936),
