@@ -49,15 +49,24 @@ def main():
         f"launch with torch.distributed.run for --gpus {args.gpus}"
     on_gpu = torch.cuda.is_available()
     device = "cuda" if on_gpu else "cpu"
-    heap = td.init_symm_heap()
-
     batch = args.batch_per_gpu * world
     max_len = args.ctx + args.warmup + args.steps + 16
 
     cfg = get_config(args.model, tp_mode=args.mode, max_length=max_len + 64)
+    # size the symmetric heap for fused-path PREFILL (M = batch*ctx tokens):
+    # AG workspace + RS scatter are each batch*ctx*hidden*2 bytes
+    prefill_m = batch * args.ctx
+    if args.mode == "ag_rs" and on_gpu:
+        need_mb = int(2.4 * prefill_m * cfg.hidden * 2 / 1e6) + 1024
+        heap = td.init_symm_heap(size_mb=max(need_mb, 4096))
+    else:
+        heap = td.init_symm_heap()
+
     model = AutoLLM(cfg, device=device)
     model.init_weights(seed=args.seed)
-    if args.mode in ("ag_rs", "gemm_ar"):
+    if args.mode == "ag_rs":
+        model.init_dist_ctx(max_m_total=prefill_m if on_gpu else batch)
+    elif args.mode == "gemm_ar":
         model.init_dist_ctx(max_m_total=batch)
 
     eng = Engine(model, batch=batch, max_len=max_len,

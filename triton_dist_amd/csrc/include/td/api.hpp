@@ -55,7 +55,8 @@ struct AgGemmArgs {
   GemmArgs g;           // a = symm workspace [M_total, K]
   const int *flags;     // per-chunk ready flags (local heap)
   int chunks_per_rank;  // flag granularity along M
-  int m_per_rank;
+  int m_per_rank;       // ACTUAL rows per rank this call (m % BM == 0)
+  int ws_stride;        // allocated rows per segment in the workspace
   int world;
   int rank;
   int expect;           // flag value that means "ready"
@@ -68,8 +69,9 @@ void launch_ag_gemm_consumer_bf16(const AgGemmArgs &args, hipStream_t stream);
 struct GemmRsArgs {
   GemmArgs g;       // c unused; m = world * m_per_rank
   PeerTable pt;
-  size_t scatter_off;  // offset of scatter buffer [world segments, m_per_rank, N]
-  int m_per_rank;
+  size_t scatter_off;  // offset of scatter buffer [world segments, ws_stride, N]
+  int m_per_rank;      // actual rows per rank this call
+  int ws_stride;       // allocated rows per segment
   int world;
   int rank;
 };
@@ -77,7 +79,8 @@ void launch_gemm_rs_producer_bf16(const GemmRsArgs &args, hipStream_t stream);
 
 // Reduce over world segments [world, m_per_rank, N] -> [m_per_rank, N].
 void launch_rs_reduce_bf16(const void *segments, void *out, int world,
-                           int rank, int m_per_rank, int n, hipStream_t stream);
+                           int rank, int m_per_rank, int ws_stride, int n,
+                           hipStream_t stream);
 
 // kernels/gemm_splitk.hip ---------------------------------------------------
 void launch_gemm_splitk_bf16(const GemmArgs &g, float *ws, int splits,

@@ -267,7 +267,12 @@ __global__ __launch_bounds__(NTHREADS) void k_ag_gemm_consumer_bf16(
 
   TileCtx t = tile_ctx();
   f32x4 acc[4][4] = {};
-  const bf16 *ga = (const bf16 *)g.a + (size_t)pid_m * BM * g.lda;
+  // segment-strided workspace: segment s of the gathered A lives at row
+  // s * ws_stride (allocation) but logical row s * m_per_rank
+  const int seg = (pid_m * BM) / args.m_per_rank;
+  const size_t arow0 =
+      (size_t)seg * args.ws_stride + (pid_m * BM - seg * args.m_per_rank);
+  const bf16 *ga = (const bf16 *)g.a + arow0 * g.lda;
   const bf16 *gb = (const bf16 *)g.b + (size_t)pid_n * BN * g.ldb;
   for (int k0 = 0; k0 < g.k; k0 += BK) {
     stage_tile(ga + k0, gb + k0, g.lda, g.ldb, lds_a, lds_b);
@@ -329,7 +334,7 @@ __global__ __launch_bounds__(NTHREADS) void k_gemm_rs_producer_bf16(
   int owner = (pid_m * BM) / args.m_per_rank;
   int local_row0 = pid_m * BM - owner * args.m_per_rank;
   bf16 *seg = (bf16 *)((char *)args.pt.bases[owner] + args.scatter_off) +
-              ((size_t)args.rank * args.m_per_rank + local_row0) * g.n + pid_n * BN;
+              ((size_t)args.rank * args.ws_stride + local_row0) * g.n + pid_n * BN;
   epilogue_store(acc, t, lds_a, seg, g.n, nullptr, 0);
 }
 
@@ -351,14 +356,14 @@ void launch_gemm_rs_producer_bf16(const GemmRsArgs &args, hipStream_t stream) {
 // starting at rank+1 (cf. gemm_reduce_scatter.py:230-284 semantics).
 // ---------------------------------------------------------------------------
 __global__ void k_rs_reduce_bf16(const bf16 *segments, bf16 *out, int world,
-                                 int rank, size_t elems) {
+                                 int rank, size_t elems, size_t seg_stride) {
   size_t i = ((size_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
   size_t stride = (size_t)gridDim.x * blockDim.x * 8;
   for (; i < elems; i += stride) {
     float acc[8] = {};
     for (int s = 0; s < world; ++s) {
       int r = (rank + 1 + s) % world;
-      bf16x8 v = *(const bf16x8 *)(segments + (size_t)r * elems + i);
+      bf16x8 v = *(const bf16x8 *)(segments + (size_t)r * seg_stride + i);
 #pragma unroll
       for (int j = 0; j < 8; ++j) acc[j] += (float)v[j];
     }
@@ -370,7 +375,7 @@ __global__ void k_rs_reduce_bf16(const bf16 *segments, bf16 *out, int world,
 }
 
 void launch_rs_reduce_bf16(const void *segments, void *out, int world,
-                           int rank, int m_per_rank, int n,
+                           int rank, int m_per_rank, int ws_stride, int n,
                            hipStream_t stream) {
   size_t elems = (size_t)m_per_rank * n;
   if (elems % 8) throw std::runtime_error("rs_reduce: elems % 8 != 0");
@@ -378,7 +383,8 @@ void launch_rs_reduce_bf16(const void *segments, void *out, int world,
   int blocks = (int)((work + 255) / 256);
   if (blocks > 2048) blocks = 2048;
   hipLaunchKernelGGL(k_rs_reduce_bf16, dim3(blocks), dim3(256), 0, stream,
-                     (const bf16 *)segments, (bf16 *)out, world, rank, elems);
+                     (const bf16 *)segments, (bf16 *)out, world, rank, elems,
+                     (size_t)ws_stride * n);
 }
 
 }  // namespace td

@@ -250,7 +250,10 @@ __global__ __launch_bounds__(NTH, 2) void k_ag_gemm256_consumer_bf16(
   kprof_record(args.prof, 0, tw0, tw1);
   WaveCtx w = wave_ctx();
   f32x4 acc[8][4] = {};
-  const bf16 *ga = (const bf16 *)g.a + (size_t)pid_m * BM * g.lda;
+  const int seg = (pid_m * BM) / args.m_per_rank;
+  const size_t arow0 =
+      (size_t)seg * args.ws_stride + (pid_m * BM - seg * args.m_per_rank);
+  const bf16 *ga = (const bf16 *)g.a + arow0 * g.lda;
   const bf16 *gb = (const bf16 *)g.b + (size_t)pid_n * BN * g.ldb;
   kloop(ga, gb, g.lda, g.ldb, g.k / BK, lds_a, lds_b, w, acc);
   bf16 *dst = (bf16 *)g.c + (size_t)pid_m * BM * g.ldc + pid_n * BN;
@@ -278,7 +281,7 @@ __global__ __launch_bounds__(NTH, 2) void k_gemm256_rs_producer_bf16(
   int owner = (pid_m * BM) / args.m_per_rank;
   int local_row0 = pid_m * BM - owner * args.m_per_rank;
   bf16 *seg = (bf16 *)((char *)args.pt.bases[owner] + args.scatter_off) +
-              ((size_t)args.rank * args.m_per_rank + local_row0) * g.n +
+              ((size_t)args.rank * args.ws_stride + local_row0) * g.n +
               pid_n * BN;
   epilogue(acc, w, lds_a, seg, g.n);
 }

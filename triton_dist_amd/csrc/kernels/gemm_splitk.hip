@@ -99,7 +99,7 @@ TD_DEV void sk_compute(const bf16 *lds_a, const bf16 *lds_b,
 template <bool WAIT_FLAGS>
 __global__ __launch_bounds__(sk::NTH) void k_gemm_splitk_bf16(
     GemmArgs g, float *ws, int splits, const int *flags, int chunks_per_rank,
-    int m_per_rank, int rank, int expect) {
+    int m_per_rank, int ws_stride, int rank, int expect) {
   __shared__ bf16 lds_a[3 * sk::BM * sk::BK];
   __shared__ bf16 lds_b[3 * sk::BN * sk::BK];
   const int tiles_n = g.n / sk::BN;
@@ -125,7 +125,12 @@ __global__ __launch_bounds__(sk::NTH) void k_gemm_splitk_bf16(
 
   TileCtx t = sk_tile_ctx();
   f32x4 acc[4][4] = {};
-  const bf16 *ga = (const bf16 *)g.a + (size_t)pid_m * sk::BM * g.lda + k_lo;
+  size_t arow0 = (size_t)pid_m * sk::BM;
+  if (WAIT_FLAGS) {  // gathered-A workspace is segment-strided
+    const int seg = (pid_m * sk::BM) / m_per_rank;
+    arow0 = (size_t)seg * ws_stride + (pid_m * sk::BM - seg * m_per_rank);
+  }
+  const bf16 *ga = (const bf16 *)g.a + arow0 * g.lda + k_lo;
   const bf16 *gb = (const bf16 *)g.b + (size_t)pid_n * sk::BN * g.ldb + k_lo;
   // 3-buffer pipelined K loop: 4 loads/thread/step, 2 steps in flight,
   // vmcnt(4) steady; 48KB LDS keeps 3 blocks/CU
@@ -188,7 +193,8 @@ __global__ void k_splitk_reduce(const float *ws, bf16 *c, const bf16 *bias,
 // segment (GEMM-RS small-M path).
 __global__ void k_splitk_reduce_scatter(const float *ws, PeerTable pt,
                                         size_t scatter_off, int m_per_rank,
-                                        int m, int n, int splits, int rank) {
+                                        int ws_stride, int m, int n,
+                                        int splits, int rank) {
   size_t mn = (size_t)m * n;
   size_t i = ((size_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
   size_t stride = (size_t)gridDim.x * blockDim.x * 8;
@@ -206,7 +212,7 @@ __global__ void k_splitk_reduce_scatter(const float *ws, PeerTable pt,
     int owner = row / m_per_rank;
     int lrow = row - owner * m_per_rank;
     bf16 *seg = (bf16 *)((char *)pt.bases[owner] + scatter_off) +
-                ((size_t)rank * m_per_rank + lrow) * n + col;
+                ((size_t)rank * ws_stride + lrow) * n + col;
     *(bf16x8 *)seg = o;
   }
 }
@@ -227,7 +233,7 @@ void launch_gemm_splitk_bf16(const GemmArgs &g, float *ws, int splits,
   int grid = (g.m / sk::BM) * (g.n / sk::BN);
   hipLaunchKernelGGL((k_gemm_splitk_bf16<false>), dim3(grid, splits),
                      dim3(sk::NTH), 0, stream, g, ws, splits, nullptr, 0, 0,
-                     0, 0);
+                     0, 0, 0);
   size_t mn = (size_t)g.m * g.n;
   hipLaunchKernelGGL(k_splitk_reduce, dim3(reduce_grid(mn / 4)), dim3(256),
                      0, stream, ws, (bf16 *)g.c, (const bf16 *)g.bias, mn,
@@ -242,7 +248,8 @@ void launch_ag_gemm_consumer_splitk_bf16(const AgGemmArgs &a, float *ws,
   int grid = (g.m / sk::BM) * (g.n / sk::BN);
   hipLaunchKernelGGL((k_gemm_splitk_bf16<true>), dim3(grid, splits),
                      dim3(sk::NTH), 0, stream, g, ws, splits, a.flags,
-                     a.chunks_per_rank, a.m_per_rank, a.rank, a.expect);
+                     a.chunks_per_rank, a.m_per_rank, a.ws_stride, a.rank,
+                     a.expect);
   size_t mn = (size_t)g.m * g.n;
   hipLaunchKernelGGL(k_splitk_reduce, dim3(reduce_grid(mn / 4)), dim3(256),
                      0, stream, ws, (bf16 *)g.c, nullptr, mn, g.n, splits);
@@ -256,11 +263,11 @@ void launch_gemm_rs_producer_splitk_bf16(const GemmRsArgs &a, float *ws,
   int grid = (g.m / sk::BM) * (g.n / sk::BN);
   hipLaunchKernelGGL((k_gemm_splitk_bf16<false>), dim3(grid, splits),
                      dim3(sk::NTH), 0, stream, g, ws, splits, nullptr, 0, 0,
-                     0, 0);
+                     0, 0, 0);
   size_t mn = (size_t)g.m * g.n;
   hipLaunchKernelGGL(k_splitk_reduce_scatter, dim3(reduce_grid(mn / 8)),
                      dim3(256), 0, stream, ws, a.pt, a.scatter_off,
-                     a.m_per_rank, g.m, g.n, splits, a.rank);
+                     a.m_per_rank, a.ws_stride, g.m, g.n, splits, a.rank);
 }
 
 }  // namespace td

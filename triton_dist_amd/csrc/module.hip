@@ -289,7 +289,8 @@ static void gemm_bf16(uintptr_t a, uintptr_t b, uintptr_t c, uintptr_t bias,
 static void ag_gemm_consumer_bf16(uintptr_t a, uintptr_t b, uintptr_t c,
                                   int m, int n, int k, uintptr_t flags,
                                   int chunks_per_rank, int m_per_rank,
-                                  int world, int rank, int expect,
+                                  int ws_stride, int world, int rank,
+                                  int expect,
                                   uintptr_t stream, uintptr_t prof_buf = 0,
                                   uintptr_t prof_cursor = 0,
                                   unsigned prof_cap = 0) {
@@ -299,6 +300,7 @@ static void ag_gemm_consumer_bf16(uintptr_t a, uintptr_t b, uintptr_t c,
   args.flags = reinterpret_cast<const int *>(flags);
   args.chunks_per_rank = chunks_per_rank;
   args.m_per_rank = m_per_rank;
+  args.ws_stride = ws_stride;
   args.world = world;
   args.rank = rank;
   args.expect = expect;
@@ -310,7 +312,8 @@ static void ag_gemm_consumer_bf16(uintptr_t a, uintptr_t b, uintptr_t c,
 
 static void gemm_rs_producer_bf16(uintptr_t a, uintptr_t b, int m, int n,
                                   int k, size_t scatter_off, int m_per_rank,
-                                  int world, int rank, uintptr_t stream) {
+                                  int ws_stride, int world, int rank,
+                                  uintptr_t stream) {
   check_active();
   GemmRsArgs args;
   args.g = GemmArgs{reinterpret_cast<void *>(a), reinterpret_cast<void *>(b),
@@ -318,6 +321,7 @@ static void gemm_rs_producer_bf16(uintptr_t a, uintptr_t b, int m, int n,
   args.pt = g_heap.pt;
   args.scatter_off = scatter_off;
   args.m_per_rank = m_per_rank;
+  args.ws_stride = ws_stride;
   args.world = world;
   args.rank = rank;
   launch_gemm_rs_producer_bf16(args, as_stream(stream));
@@ -325,10 +329,11 @@ static void gemm_rs_producer_bf16(uintptr_t a, uintptr_t b, int m, int n,
 }
 
 static void rs_reduce_bf16(uintptr_t segments, uintptr_t out, int world,
-                           int rank, int m_per_rank, int n, uintptr_t stream) {
+                           int rank, int m_per_rank, int ws_stride, int n,
+                           uintptr_t stream) {
   launch_rs_reduce_bf16(reinterpret_cast<void *>(segments),
                         reinterpret_cast<void *>(out), world, rank, m_per_rank,
-                        n, as_stream(stream));
+                        ws_stride, n, as_stream(stream));
   TD_CHECK_HIP(hipGetLastError());
 }
 
@@ -345,7 +350,8 @@ static void ag_gemm_consumer_splitk_bf16(uintptr_t a, uintptr_t b,
                                          uintptr_t c, uintptr_t ws, int m,
                                          int n, int k, uintptr_t flags,
                                          int chunks_per_rank, int m_per_rank,
-                                         int world, int rank, int expect,
+                                         int ws_stride, int world, int rank,
+                                         int expect,
                                          int splits, uintptr_t stream) {
   AgGemmArgs args;
   args.g = GemmArgs{(void *)a, (void *)b, (void *)c, nullptr,
@@ -353,6 +359,7 @@ static void ag_gemm_consumer_splitk_bf16(uintptr_t a, uintptr_t b,
   args.flags = (const int *)flags;
   args.chunks_per_rank = chunks_per_rank;
   args.m_per_rank = m_per_rank;
+  args.ws_stride = ws_stride;
   args.world = world;
   args.rank = rank;
   args.expect = expect;
@@ -364,7 +371,8 @@ static void ag_gemm_consumer_splitk_bf16(uintptr_t a, uintptr_t b,
 static void gemm_rs_producer_splitk_bf16(uintptr_t a, uintptr_t b,
                                          uintptr_t ws, int m, int n, int k,
                                          size_t scatter_off, int m_per_rank,
-                                         int world, int rank, int splits,
+                                         int ws_stride, int world, int rank,
+                                         int splits,
                                          uintptr_t stream) {
   check_active();
   GemmRsArgs args;
@@ -373,6 +381,7 @@ static void gemm_rs_producer_splitk_bf16(uintptr_t a, uintptr_t b,
   args.pt = g_heap.pt;
   args.scatter_off = scatter_off;
   args.m_per_rank = m_per_rank;
+  args.ws_stride = ws_stride;
   args.world = world;
   args.rank = rank;
   launch_gemm_rs_producer_splitk_bf16(args, (float *)ws, splits,
@@ -641,7 +650,8 @@ PYBIND11_MODULE(_C, m) {
   m.def("ag_gemm_consumer_bf16", &ag_gemm_consumer_bf16, py::arg("a"),
         py::arg("b"), py::arg("c"), py::arg("m"), py::arg("n"), py::arg("k"),
         py::arg("flags"), py::arg("chunks_per_rank"), py::arg("m_per_rank"),
-        py::arg("world"), py::arg("rank"), py::arg("expect"),
+        py::arg("ws_stride"), py::arg("world"), py::arg("rank"),
+        py::arg("expect"),
         py::arg("stream"), py::arg("prof_buf") = 0,
         py::arg("prof_cursor") = 0, py::arg("prof_cap") = 0);
   m.def("gemm_rs_producer_bf16", &gemm_rs_producer_bf16);

@@ -147,12 +147,28 @@ class DenseLLM:
         x_last = x.view(b, s, -1)[:, -1]
         return x_last @ self.lm_head.t()  # [B, vocab] bf16
 
+    def _dist_prefill_ok(self, m_total: int) -> bool:
+        """Can the fused ag_rs path run prefill at this token count?
+        (Replicated torch prefill all-reduces O(M*H) per layer — at
+        TP8/M=512K that is ~690 GB of RCCL traffic; the fused path shards
+        it and overlaps, so use it whenever the contexts cover M.)"""
+        if self.mode != "ag_rs" or self._decode_ag_ctx is None:
+            return False
+        ctx = self._decode_ag_ctx
+        if m_total % self.world:
+            return False
+        m = m_total // self.world
+        return (m <= ctx.max_m_per_rank and m % 128 == 0
+                and m % ctx.chunks_per_rank == 0)
+
     def prefill(self, tokens: torch.Tensor, kv: KVCache) -> torch.Tensor:
-        """Torch-eager prefill (cf. reference engine: prefill in torch mode,
-        decode in dist mode). Returns first sampled tokens [B]."""
+        """Prefill: fused ag_rs path when the contexts cover B*S tokens,
+        torch-eager otherwise (cf. reference engine's torch prefill).
+        Returns first sampled tokens [B]."""
         b, s = tokens.shape
         pos = torch.arange(s, device=tokens.device).expand(b, s)
-        logits = self.step(tokens, kv, pos, prefill=True, mode="torch")
+        mode = self.mode if self._dist_prefill_ok(b * s) else "torch"
+        logits = self.step(tokens, kv, pos, prefill=True, mode=mode)
         kv.advance(s)
         return logits.argmax(-1)
 

@@ -101,9 +101,10 @@ def ag_gemm(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
     heap, _C = ctx.heap, ctx.heap._C
     compute = torch.cuda.current_stream()
     chunks = ctx.chunks_per_rank
-    rows_per_chunk = ctx.max_m_per_rank // chunks
-    assert m == ctx.max_m_per_rank, "v1 consumer requires m == max_m_per_rank"
-    m_chunks = (m + rows_per_chunk - 1) // rows_per_chunk
+    assert m % chunks == 0 and m % 128 == 0, \
+        f"m={m} must divide chunks={chunks} and tile by 128"
+    rows_per_chunk = m // chunks
+    m_chunks = chunks
     chunk_bytes = rows_per_chunk * k * 2
 
     # 1. reset my flags, then entry barrier (workspace of the previous call
@@ -131,9 +132,8 @@ def ag_gemm(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
         dst_seg = ctx.ws.ptr(peer) + rank * ctx.max_m_per_rank * k * 2
         dst_flag = ctx.flags.ptr(peer) + rank * chunks * 4
         for c in range(m_chunks):
-            nbytes = min(chunk_bytes, (m - c * rows_per_chunk) * k * 2)
             _C.memcpy_async(dst_seg + c * chunk_bytes,
-                            a.data_ptr() + c * chunk_bytes, nbytes,
+                            a.data_ptr() + c * chunk_bytes, chunk_bytes,
                             stream.cuda_stream)
             _C.memcpy_async(dst_flag + c * 4, heap.one_src.ptr(), 4,
                             stream.cuda_stream)
@@ -149,13 +149,14 @@ def ag_gemm(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
         ws = splitk_ws(m_total, n, splits, a.device)
         _C.ag_gemm_consumer_splitk_bf16(
             ctx.ws.ptr(), w.data_ptr(), out.data_ptr(), ws.data_ptr(),
-            m_total, n, k, ctx.flags.ptr(), chunks, m, world, rank, 1,
+            m_total, n, k, ctx.flags.ptr(), chunks, m,
+            ctx.max_m_per_rank, world, rank, 1,
             splits, compute.cuda_stream)
     else:
         pb, pc, pcap = profiler.ptrs() if profiler is not None else (0, 0, 0)
         _C.ag_gemm_consumer_bf16(
             ctx.ws.ptr(), w.data_ptr(), out.data_ptr(), m_total, n, k,
-            ctx.flags.ptr(), chunks, m, world, rank, 1,
+            ctx.flags.ptr(), chunks, m, ctx.max_m_per_rank, world, rank, 1,
             compute.cuda_stream, pb, pc, pcap)
     # join comm streams back into the compute stream (after the consumer
     # launch: no serialization, but graph capture requires joined forks)
@@ -163,7 +164,10 @@ def ag_gemm(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
         ctx.join_evs[s].record(ctx.comm_streams[s])
         compute.wait_event(ctx.join_evs[s])
     if gathered_out:
-        return out, ctx.ws.local().reshape(world * ctx.max_m_per_rank, k)[:m_total]
+        g = ctx.ws.local()[:, :m].reshape(m_total, k) \
+            if m < ctx.max_m_per_rank \
+            else ctx.ws.local().reshape(m_total, k)
+        return out, g
     return out
 
 
@@ -203,7 +207,7 @@ def allgather(a: torch.Tensor, ctx: AGGemmContext,
 
     heap, _C = ctx.heap, ctx.heap._C
     compute = torch.cuda.current_stream()
-    rows_per_chunk = ctx.max_m_per_rank // chunks
+    rows_per_chunk = m // chunks
     chunk_bytes = rows_per_chunk * k * 2
     _C.reset_flags(ctx.flags.ptr(), world * chunks, 0, compute.cuda_stream)
     heap.barrier_all_on_stream(compute)
@@ -230,11 +234,22 @@ def allgather(a: torch.Tensor, ctx: AGGemmContext,
     for s in range(min(ns, max(world - 1, 1))):
         ctx.join_evs[s].record(ctx.comm_streams[s])
         compute.wait_event(ctx.join_evs[s])
-    gathered = ctx.ws.local().reshape(world * ctx.max_m_per_rank, k)
-    if out is not None:
-        out.copy_(gathered)
-        return out
-    return gathered
+    if m == ctx.max_m_per_rank:
+        gathered = ctx.ws.local().reshape(world * m, k)
+        if out is not None:
+            out.copy_(gathered)
+            return out
+        return gathered
+    # segment-strided workspace -> contiguous output
+    if out is None:
+        out = torch.empty(world * m, k, dtype=torch.bfloat16,
+                          device=a.device)
+    seg_bytes = ctx.max_m_per_rank * k * 2
+    for r in range(world):
+        _C.memcpy_async(out.data_ptr() + r * m * k * 2,
+                        ctx.ws.ptr() + r * seg_bytes, m * k * 2,
+                        compute.cuda_stream)
+    return out
 
 
 def _allgather_pull(a: torch.Tensor, ctx: AGGemmContext,

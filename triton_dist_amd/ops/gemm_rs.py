@@ -54,8 +54,9 @@ def gemm_rs(a: torch.Tensor, w: torch.Tensor, ctx: GemmRSContext,
     m, k = a.shape
     n = w.shape[0]
     world, rank = ctx.world, ctx.rank
-    assert n == ctx.n and m == world * ctx.max_m_per_rank
-    m_per_rank = ctx.max_m_per_rank
+    assert n == ctx.n and m % world == 0
+    m_per_rank = m // world
+    assert m_per_rank <= ctx.max_m_per_rank
 
     if ctx.heap.backend == "cpu":
         partial = (a.float() @ w.float().t())
@@ -78,6 +79,7 @@ def gemm_rs(a: torch.Tensor, w: torch.Tensor, ctx: GemmRSContext,
 
     heap, _C = ctx.heap, ctx.heap._C
     assert a.dtype == torch.bfloat16 and a.is_contiguous()
+    assert m_per_rank % 128 == 0, "HIP path needs m/world % 128 == 0"
     compute = torch.cuda.current_stream()
     # entry barrier: previous call's reduce has consumed the scatter bufs
     heap.barrier_all_on_stream(compute)
@@ -88,18 +90,20 @@ def gemm_rs(a: torch.Tensor, w: torch.Tensor, ctx: GemmRSContext,
         ws = splitk_ws(m, n, splits, a.device)
         _C.gemm_rs_producer_splitk_bf16(
             a.data_ptr(), w.data_ptr(), ws.data_ptr(), m, n, k,
-            ctx.scatter.offset, m_per_rank, world, rank, splits,
-            compute.cuda_stream)
+            ctx.scatter.offset, m_per_rank, ctx.max_m_per_rank, world, rank,
+            splits, compute.cuda_stream)
     else:
         _C.gemm_rs_producer_bf16(a.data_ptr(), w.data_ptr(), m, n, k,
-                                 ctx.scatter.offset, m_per_rank, world, rank,
+                                 ctx.scatter.offset, m_per_rank,
+                                 ctx.max_m_per_rank, world, rank,
                                  compute.cuda_stream)
     heap.barrier_all_on_stream(compute)
     if out is None:
         out = torch.empty(m_per_rank, n, dtype=torch.bfloat16,
                           device=a.device)
     _C.rs_reduce_bf16(ctx.scatter.ptr(), out.data_ptr(), world, rank,
-                      m_per_rank, n, compute.cuda_stream)
+                      m_per_rank, ctx.max_m_per_rank, n,
+                      compute.cuda_stream)
     return out
 
 
