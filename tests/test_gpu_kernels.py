@@ -123,7 +123,7 @@ def test_rs_reduce():
     world, m, n = 8, 256, 512
     segs = torch.randn(world, m, n, device="cuda").to(torch.bfloat16)
     out = torch.empty(m, n, device="cuda", dtype=torch.bfloat16)
-    _C.rs_reduce_bf16(segs.data_ptr(), out.data_ptr(), world, 3, m, n,
+    _C.rs_reduce_bf16(segs.data_ptr(), out.data_ptr(), world, 3, m, m, n,
                       _stream())
     torch.cuda.synchronize()
     ref = segs.float().sum(0)
