@@ -133,3 +133,44 @@ def _body_ag_pull(rank, world):
 
 def test_allgather_pull_2rank():
     run_distributed(_body_ag_pull, world_size=2)
+
+
+def _body_variable_m(rank, world):
+    """One ctx serves multiple m sizes (prefill + decode through the same
+    segment-strided workspace)."""
+    from triton_dist_amd.ops import (ag_gemm, ag_gemm_ref, allgather,
+                                     create_ag_gemm_context,
+                                     create_gemm_rs_context, gemm_rs,
+                                     gemm_rs_ref)
+    from triton_dist_amd.utils.testing import assert_allclose, bf16_gemm_tol
+    import torch.distributed as dist
+
+    k, n = 1024, 768
+    ctx = create_ag_gemm_context(max_m_per_rank=512, k=k, chunks_per_rank=4)
+    rs_ctx = create_gemm_rs_context(max_m_total=512 * world, n=n)
+    torch.manual_seed(31 + rank)
+    for m in (256, 512, 128):
+        a = (torch.randn(m, k, device="cuda") / 8).to(torch.bfloat16)
+        w = (torch.randn(n, k, device="cuda",
+                         generator=torch.Generator("cuda").manual_seed(m))
+             / 8).to(torch.bfloat16)
+        c = ag_gemm(a, w, ctx)
+        torch.cuda.synchronize()
+        ref = ag_gemm_ref(a, w)
+        assert_allclose(c, ref, **bf16_gemm_tol(k))
+        g = allgather(a, ctx)
+        torch.cuda.synchronize()
+        full = torch.empty(world * m, k, dtype=torch.bfloat16)
+        dist.all_gather_into_tensor(full, a.cpu())
+        assert torch.equal(g.cpu(), full), f"allgather m={m}"
+        a2 = (torch.randn(m * world, k, device="cuda") / 8).to(torch.bfloat16)
+        out = gemm_rs(a2, w, rs_ctx)
+        torch.cuda.synchronize()
+        ref2 = gemm_rs_ref(a2, w)
+        tol = bf16_gemm_tol(k)
+        assert_allclose(out, ref2, atol=tol["atol"] * world,
+                        rtol=tol["rtol"] * 2, msg=f"rs m={m}")
+
+
+def test_variable_m_2rank():
+    run_distributed(_body_variable_m, world_size=2)
