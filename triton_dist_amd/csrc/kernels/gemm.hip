@@ -217,6 +217,11 @@ __global__ __launch_bounds__(NTHREADS) void k_gemm_bf16(GemmArgs args) {
 // gemm256.hip perf tier (256^2/BK128 K-slice ring); these fall back to the
 // 128^2 kernel for shapes that don't tile by 256.
 bool gemm256_ok(int m, int n, int k);
+// the 256-tier runs 1 block/CU (128 KiB LDS): it needs >= ~224 workgroups
+// to fill the chip; below that the 128-tier's 4x denser grid wins
+static inline bool gemm256_fills(int m, int n) {
+  return (m / 256) * (n / 256) >= 224;
+}
 void launch_gemm256_bf16(const GemmArgs &args, hipStream_t stream);
 void launch_ag_gemm256_consumer_bf16(const AgGemmArgs &args,
                                      hipStream_t stream);
@@ -224,7 +229,8 @@ void launch_gemm256_rs_producer_bf16(const GemmRsArgs &args,
                                      hipStream_t stream);
 
 void launch_gemm_bf16(const GemmArgs &args, hipStream_t stream) {
-  if (gemm256_ok(args.m, args.n, args.k) && !args.bias) {
+  if (gemm256_ok(args.m, args.n, args.k) && !args.bias &&
+      gemm256_fills(args.m, args.n)) {
     launch_gemm256_bf16(args, stream);
     return;
   }
@@ -292,7 +298,8 @@ void launch_ag_gemm_consumer_bf16(const AgGemmArgs &args, hipStream_t stream) {
     throw std::runtime_error("ag_gemm: shape must tile by 128/128/64");
   if (args.m_per_rank % args.chunks_per_rank)
     throw std::runtime_error("ag_gemm: chunks_per_rank must divide m_per_rank");
-  if (gemm256_ok(g.m, g.n, g.k) && args.m_per_rank % 256 == 0) {
+  if (gemm256_ok(g.m, g.n, g.k) && args.m_per_rank % 256 == 0 &&
+      gemm256_fills(g.m, g.n)) {
     launch_ag_gemm256_consumer_bf16(args, stream);
     return;
   }
@@ -342,7 +349,8 @@ void launch_gemm_rs_producer_bf16(const GemmRsArgs &args, hipStream_t stream) {
   const GemmArgs &g = args.g;
   if (g.m % BM || g.n % BN || g.k % BK || args.m_per_rank % BM)
     throw std::runtime_error("gemm_rs: shape must tile by 128/128/64");
-  if (gemm256_ok(g.m, g.n, g.k) && args.m_per_rank % 256 == 0) {
+  if (gemm256_ok(g.m, g.n, g.k) && args.m_per_rank % 256 == 0 &&
+      gemm256_fills(g.m, g.n)) {
     launch_gemm256_rs_producer_bf16(args, stream);
     return;
   }
