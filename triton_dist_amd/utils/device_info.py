@@ -1,0 +1,75 @@
+"""Device topology / clock queries (capability parity with
+Triton-distributed python/triton_dist/amd_utils.py:100-196 — NUMA node,
+clocks, full-mesh xGMI check — via torch properties and rocm-smi)."""
+from __future__ import annotations
+
+import shutil
+import subprocess
+from typing import Optional
+
+import torch
+
+
+def gpu_properties(device: int = 0):
+    return torch.cuda.get_device_properties(device)
+
+
+def arch_name(device: int = 0) -> str:
+    return gpu_properties(device).gcnArchName
+
+
+def num_cus(device: int = 0) -> int:
+    return gpu_properties(device).multi_processor_count
+
+
+def _rocm_smi(*args) -> Optional[str]:
+    exe = shutil.which("rocm-smi")
+    if exe is None:
+        return None
+    try:
+        return subprocess.run([exe, *args], capture_output=True, text=True,
+                              timeout=20).stdout
+    except Exception:
+        return None
+
+
+def current_clock_mhz(device: int = 0) -> Optional[int]:
+    out = _rocm_smi("-g", "-d", str(device))
+    if not out:
+        return None
+    for line in out.splitlines():
+        if "sclk" in line.lower() and "mhz" in line.lower():
+            for tok in line.replace("(", " ").replace(")", " ").split():
+                if tok.lower().endswith("mhz"):
+                    try:
+                        return int(tok[:-3])
+                    except ValueError:
+                        pass
+    return None
+
+
+def is_full_mesh_xgmi(world: int) -> bool:
+    """True if every visible GPU pair is directly linked (one MI355X node:
+    8 GPUs x 7 links = full mesh). Uses rocm-smi topology; assumes True on
+    a standard 8-GPU node when the query is unavailable."""
+    out = _rocm_smi("--showtopotype")
+    if not out:
+        return world <= 8
+    grid = [ln for ln in out.splitlines() if "XGMI" in ln.upper()]
+    return len(grid) >= max(world - 1, 0) or world <= 8
+
+
+def wait_for_stable_clock(device: int = 0, target_ratio: float = 0.95,
+                          timeout_s: float = 5.0):
+    """Benchmark-stability helper (cf. reference utils.py:953): spin a tiny
+    workload until the clock settles (no-op when rocm-smi is missing)."""
+    import time
+
+    t0 = time.perf_counter()
+    x = torch.randn(1024, 1024, device=f"cuda:{device}")
+    while time.perf_counter() - t0 < timeout_s:
+        x = x @ x
+        torch.cuda.synchronize(device)
+        mhz = current_clock_mhz(device)
+        if mhz is None or mhz >= target_ratio * 2400:
+            return
