@@ -113,10 +113,14 @@ TD_DEV void t_rmsnorm(const Task &t, bool add) {
 }
 
 TD_DEV void t_gemm_tile(const Task &t, bf16 *lds) {
-  // 128x128 C tile, 4 waves 2x2, BK=64, masked rows (A padded to 128 rows)
-  constexpr int BM = 128, BN = 128, BK = 64;
-  bf16 *lds_a = lds;            // 128*64
-  bf16 *lds_b = lds + BM * BK;  // 128*64
+  // 32x128 C tile (decode-friendly: small-M waste bounded at 32 rows),
+  // 4 waves side-by-side on N, 3-buffer BK=64 pipelined K loop (counted
+  // vmcnt, one barrier pair per step — the measured-fast structure from
+  // the grouped-GEMM tier). Masked rows for M < 32-multiples.
+  constexpr int BM = 32, BN = 128, BK = 64;
+  constexpr int ABUF = BM * BK, BBUF = BN * BK;
+  bf16 *lds_a = lds;                 // 3 x 4 KB
+  bf16 *lds_b = lds + 3 * ABUF;      // 3 x 16 KB
   const bf16 *A = (const bf16 *)t.a[0];
   const bf16 *B = (const bf16 *)t.a[1];
   bf16 *C = (bf16 *)t.a[2];
@@ -124,60 +128,75 @@ TD_DEV void t_gemm_tile(const Task &t, bf16 *lds) {
   const int pid_m = (int)t.a[6], pid_n = (int)t.a[7];
   const int tid = threadIdx.x;
   const int wave = tid >> 6, lane = tid & 63;
-  const int wr = wave >> 1, wc = wave & 1;
-  f32x4 acc[4][4] = {};
+  f32x4 acc[2][2] = {};
   const bf16 *ga = A + (size_t)pid_m * BM * k;
   const bf16 *gb = B + (size_t)pid_n * BN * k;
-  for (int k0 = 0; k0 < k; k0 += BK) {
-#pragma unroll
-    for (int it = 0; it < 4; ++it) {
-      int idx = it * NTH + tid;
-      int row = idx >> 3, kc = idx & 7;
-      int wave_chunk0 = it * NTH + wave * 64;
+  const int ksteps = k / BK;
+  auto stage = [&](int ti, int buf) {
+    const int k0 = ti * BK;
+    {
+      int row = tid >> 3, kc = tid & 7;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int *)(
               ga + (size_t)row * k + k0 + kc * 8),
-          (__attribute__((address_space(3))) unsigned int *)(lds_a +
-                                                             wave_chunk0 * 8),
-          16, 0, 0);
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) unsigned int *)(
-              gb + (size_t)row * k + k0 + kc * 8),
-          (__attribute__((address_space(3))) unsigned int *)(lds_b +
-                                                             wave_chunk0 * 8),
+          (__attribute__((address_space(3))) unsigned int *)(
+              lds_a + buf * ABUF + (wave * 64) * 8),
           16, 0, 0);
     }
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __syncthreads();
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      int qb = it * 256 + tid;
+      int rowb = qb >> 3, kcb = qb & 7;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int *)(
+              gb + (size_t)rowb * k + k0 + kcb * 8),
+          (__attribute__((address_space(3))) unsigned int *)(
+              lds_b + buf * BBUF + (it * 256 + wave * 64) * 8),
+          16, 0, 0);
+    }
+  };
+  stage(0, 0);
+  if (ksteps > 1) stage(1, 1);
+  for (int ti = 0; ti < ksteps; ++ti) {
+    const int buf = ti % 3;
+    if (ti + 1 < ksteps) {
+      asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+    __builtin_amdgcn_sched_barrier(0);
+    if (ti + 2 < ksteps) stage(ti + 2, (ti + 2) % 3);
 #pragma unroll
     for (int ks = 0; ks < BK / 32; ++ks) {
-      bf16x8 af[4], bfr[4];
+      bf16x8 af[2], bfr[2];
 #pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        int arow = wr * 64 + i * 16 + (lane & 15);
-        int brow = wc * 64 + i * 16 + (lane & 15);
+      for (int i = 0; i < 2; ++i) {
+        int arow = i * 16 + (lane & 15);
+        int brow = wave * 32 + i * 16 + (lane & 15);
         int kk = ks * 32 + (lane >> 4) * 8;
-        af[i] = *(const bf16x8 *)(lds_a + arow * BK + kk);
-        bfr[i] = *(const bf16x8 *)(lds_b + brow * BK + kk);
+        af[i] = *(const bf16x8 *)(lds_a + buf * ABUF + arow * BK + kk);
+        bfr[i] = *(const bf16x8 *)(lds_b + buf * BBUF + brow * BK + kk);
       }
 #pragma unroll
-      for (int i = 0; i < 4; ++i)
+      for (int i = 0; i < 2; ++i)
 #pragma unroll
-        for (int j = 0; j < 4; ++j)
+        for (int j = 0; j < 2; ++j)
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[i], bfr[j], acc[i][j], 0, 0, 0);
     }
-    __syncthreads();
+    __builtin_amdgcn_s_barrier();
+    __builtin_amdgcn_sched_barrier(0);
   }
   const int row_lim = m - pid_m * BM;
 #pragma unroll
-  for (int i = 0; i < 4; ++i)
+  for (int i = 0; i < 2; ++i)
 #pragma unroll
-    for (int j = 0; j < 4; ++j)
+    for (int j = 0; j < 2; ++j)
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        int row = wr * 64 + i * 16 + (lane >> 4) * 4 + r;
-        int col = wc * 64 + j * 16 + (lane & 15);
+        int row = i * 16 + (lane >> 4) * 4 + r;
+        int col = wave * 32 + j * 16 + (lane & 15);
         if (row < row_lim)
           C[((size_t)pid_m * BM + row) * n + (size_t)pid_n * BN + col] =
               (bf16)acc[i][j][r];
@@ -390,7 +409,7 @@ TD_DEV void mk_flash_decode_body(const bf16 *q, const bf16 *kcache,
 __global__ __launch_bounds__(mk::NTH) void k_megakernel(
     const mk::Task *__restrict__ tasks, const int *__restrict__ queue,
     const int *__restrict__ queue_off, int *__restrict__ scoreboard) {
-  __shared__ char lds[32768];  // union: gemm A/B tiles | flash-decode state
+  __shared__ char lds[61440];  // union: gemm 3-buf A/B | flash-decode state
   const int wg = blockIdx.x;
   const int q_lo = queue_off[wg], q_hi = queue_off[wg + 1];
   for (int qi = q_lo; qi < q_hi; ++qi) {

@@ -27,7 +27,7 @@ class MegaQwen3Decode:
         assert cfg.head_dim == 128
         self.model, self.kv, self.b = model, kv, batch
         dev = model.device
-        bp = (batch + 127) // 128 * 128  # padded rows for GEMM tiles
+        bp = (batch + 31) // 32 * 32  # padded rows (32-row GEMM tiles)
         self.bp = bp
         H = cfg.hidden
         qh, kvh, d = cfg.n_heads, cfg.n_kv_heads, cfg.head_dim
@@ -53,7 +53,7 @@ class MegaQwen3Decode:
             t.zero_()  # padded rows stay zero
 
         g = MegaGraph()
-        tiles_m = bp // 128
+        tiles_m = bp // 32
 
         def gemm(a_buf, w, c_buf, n, k, dep):
             op = g.new_op()
