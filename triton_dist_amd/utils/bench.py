@@ -64,3 +64,59 @@ def perf_func_with_l2_reset(fn, iters=20, warmup=5, l2_size_mb=300):
         torch.cuda.synchronize()
         total_ms += start.elapsed_time(end)
     return out, total_ms / iters
+
+
+class group_profile:
+    """Merged multi-rank torch-profiler traces (capability parity with
+    Triton-distributed profiler_utils.py:205-303 group_profile): each rank
+    exports chrome traces under <dir>/rank<r>.json; rank 0 merges them into
+    one trace with per-rank pids after the barrier."""
+
+    def __init__(self, name="trace", do_prof=True, out_dir="prof_out",
+                 group=None):
+        self.name, self.do_prof, self.out_dir = name, do_prof, out_dir
+        self.group = group
+        self.prof = None
+
+    def __enter__(self):
+        if self.do_prof:
+            import torch.profiler as tp
+
+            acts = [tp.ProfilerActivity.CPU]
+            if torch.cuda.is_available():
+                acts.append(tp.ProfilerActivity.CUDA)
+            self.prof = tp.profile(activities=acts)
+            self.prof.__enter__()
+        return self
+
+    def __exit__(self, *exc):
+        if self.prof is None:
+            return False
+        import json
+        import os
+
+        import torch.distributed as dist
+
+        self.prof.__exit__(*exc)
+        rank = dist.get_rank(self.group) if dist.is_initialized() else 0
+        world = dist.get_world_size(self.group) if dist.is_initialized() else 1
+        os.makedirs(self.out_dir, exist_ok=True)
+        path = os.path.join(self.out_dir, f"{self.name}_rank{rank}.json")
+        self.prof.export_chrome_trace(path)
+        if dist.is_initialized():
+            dist.barrier(self.group)
+        if rank == 0 and world > 1:
+            events = []
+            for r in range(world):
+                p = os.path.join(self.out_dir, f"{self.name}_rank{r}.json")
+                try:
+                    data = json.load(open(p))
+                    for ev in data.get("traceEvents", []):
+                        ev["pid"] = f"rank{r}.{ev.get('pid', 0)}"
+                        events.append(ev)
+                except Exception:
+                    pass
+            json.dump({"traceEvents": events},
+                      open(os.path.join(self.out_dir,
+                                        f"{self.name}_merged.json"), "w"))
+        return False
