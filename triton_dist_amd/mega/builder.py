@@ -13,7 +13,7 @@ Task record layout (must match mk::Task):
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import List, Optional, Tuple
+from typing import List, Tuple
 
 import torch
 
