@@ -34,24 +34,26 @@ __global__ __launch_bounds__(256) void k_flash_decode(
     const bf16 *__restrict__ vcache, bf16 *__restrict__ out,
     const long *__restrict__ offset, int qh, int kvh, int max_len,
     float scale) {
-  // K staged in LDS (16 reads/row); V read DIRECT from global in the PV
-  // loop — each half-wave reads a full 128-elem row coalesced and the 8
-  // head-groups hit L2 (guide: don't LDS-stage cache-resident data).
-  // 11 KB LDS -> high occupancy for latency hiding.
+  // QK^T on MFMA: S[16 q-slots][32 pos] = Q[16,128] x K^T[128,32] — each
+  // of the 4 waves contributes one 32-dim k-chunk (2 mfma_16x16x32) and
+  // partials reduce through LDS. K staged in LDS; V read direct (L2).
   const int b = blockIdx.x;
   const int kh = blockIdx.y;
   const int G = qh / kvh;
   const int tid = threadIdx.x;
-  const int g = tid >> 5;
-  const int t = tid & 31;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int g = tid >> 5;   // softmax/PV role: query slot
+  const int t = tid & 31;   // softmax/PV role: position
   const long seqlen = *offset + 1;
 
   __shared__ bf16 k_lds[kTile][kD];
+  __shared__ bf16 q_lds[16][kD];          // rows >= G zero-padded
+  __shared__ float s_part[4][16][kTile];  // per-wave QK partials
   __shared__ float p_lds[8][kTile];
   __shared__ float m_lds[8], r_lds[8], l_lds[8];
-  __shared__ bf16 q_lds[8][kD];
 
-  for (int i = tid; i < 8 * kD / 8; i += 256) {
+  for (int i = tid; i < 16 * kD / 8; i += 256) {
     int hh = i / (kD / 8);
     int c = (i % (kD / 8)) * 8;
     bf16x8 v{};
@@ -64,6 +66,10 @@ __global__ __launch_bounds__(256) void k_flash_decode(
     l_lds[tid] = 0.f;
   }
   __syncthreads();
+
+  // hoist the Q fragment (constant across tiles): wave w covers k-chunk w
+  bf16x8 qfrag = *(const bf16x8 *)(
+      &q_lds[lane & 15][wave * 32 + (lane >> 4) * 8]);
 
   float acc[4] = {};
   const int my_d0 = t * 4;
@@ -83,17 +89,25 @@ __global__ __launch_bounds__(256) void k_flash_decode(
       *(bf16x8 *)(&k_lds[r][c]) = kv;
     }
     __syncthreads();
+
+    // MFMA QK^T: wave w, pos-half h: B[k][n=pos] = K[pos][k]
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      bf16x8 kfrag = *(const bf16x8 *)(
+          &k_lds[(lane & 15) + 16 * h][wave * 32 + (lane >> 4) * 8]);
+      f32x4 c4 = {0.f, 0.f, 0.f, 0.f};
+      c4 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag, kfrag, c4, 0, 0,
+                                                   0);
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        s_part[wave][(lane >> 4) * 4 + r][(lane & 15) + 16 * h] = c4[r];
+    }
+    __syncthreads();
+
     float s = -1e30f;
     if (g < G && pos0 + t < seqlen) {
-      float d = 0.f;
-#pragma unroll
-      for (int c = 0; c < kD / 8; ++c) {
-        bf16x8 qv = *(const bf16x8 *)(&q_lds[g][c * 8]);
-        bf16x8 kv = *(const bf16x8 *)(&k_lds[t][c * 8]);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) d += (float)qv[j] * (float)kv[j];
-      }
-      s = d * scale;
+      s = (s_part[0][g][t] + s_part[1][g][t] + s_part[2][g][t] +
+           s_part[3][g][t]) * scale;
     }
     float mx = s;
     for (int off = 16; off > 0; off >>= 1)
