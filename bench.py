@@ -56,18 +56,22 @@ def main():
     # size the symmetric heap for fused-path PREFILL (M = batch*ctx tokens):
     # AG workspace + RS scatter are each batch*ctx*hidden*2 bytes
     prefill_m = batch * args.ctx
+    # MoE models keep contexts at decode size (EP prefill buffers scale
+    # with M*topk; torch prefill has no collectives at world=1)
+    dense = cfg.n_experts == 0
+    ctx_m = prefill_m if (on_gpu and dense) else batch
     if args.mode == "ag_rs" and on_gpu:
-        need_mb = int(2.4 * prefill_m * cfg.hidden * 2 / 1e6) + 1024
-        heap = td.init_symm_heap(size_mb=max(need_mb, 4096))
+        need = 2.4 * ctx_m * cfg.hidden * 2
+        if not dense:  # EP recv/combine symm buffers
+            need += 4.6 * ctx_m * cfg.moe_topk * cfg.hidden * 2 / world
+        heap = td.init_symm_heap(size_mb=max(int(need / 1e6) + 1024, 4096))
     else:
         heap = td.init_symm_heap()
 
     model = AutoLLM(cfg, device=device)
     model.init_weights(seed=args.seed)
-    if args.mode == "ag_rs":
-        model.init_dist_ctx(max_m_total=prefill_m if on_gpu else batch)
-    elif args.mode == "gemm_ar":
-        model.init_dist_ctx(max_m_total=batch)
+    if args.mode in ("ag_rs", "gemm_ar"):
+        model.init_dist_ctx(max_m_total=ctx_m)
 
     eng = Engine(model, batch=batch, max_len=max_len,
                  use_graph=on_gpu and not args.no_graph)
