@@ -94,6 +94,15 @@ def ag_gemm(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
             return c, ctx.ws.local()[:, :m].reshape(world * m, k)
         return c
 
+    if world == 1:
+        # the gather is the identity: this is a PLAIN GEMM — route it to
+        # the fastest library implementation (hipBLASLt)
+        c = torch.matmul(a, w.t(), out=out) if out is not None \
+            else a @ w.t()
+        if gathered_out:
+            return c, a
+        return c
+
     # Every launch below takes only constant arguments and device-resident
     # state, so the whole op is hipGraph-capturable (the Engine captures the
     # decode step; cf. reference engine.py:75-105 requirement).
@@ -182,6 +191,11 @@ def allgather(a: torch.Tensor, ctx: AGGemmContext,
           peer segment over the CONSUMER's xGMI links (k_ag_pull).
     Returns [world*m, K]."""
     m, k = a.shape
+    if ctx.heap.backend == "hip" and ctx.world == 1:
+        if out is not None:
+            out.copy_(a)
+            return out
+        return a
     if method == "pull" and ctx.heap.backend == "hip" and ctx.world > 1:
         return _allgather_pull(a, ctx, out)
     assert k == ctx.k and m <= ctx.max_m_per_rank

@@ -77,6 +77,11 @@ def gemm_rs(a: torch.Tensor, w: torch.Tensor, ctx: GemmRSContext,
             return out
         return res
 
+    if world == 1:
+        # no scatter/reduce at world 1: plain GEMM via hipBLASLt
+        return torch.matmul(a, w.t(), out=out) if out is not None \
+            else a @ w.t()
+
     heap, _C = ctx.heap, ctx.heap._C
     assert a.dtype == torch.bfloat16 and a.is_contiguous()
     assert m_per_rank % 128 == 0, "HIP path needs m/world % 128 == 0"
