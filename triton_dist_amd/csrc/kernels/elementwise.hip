@@ -103,34 +103,35 @@ void launch_add_rmsnorm(const void *x, const void *resid_in, void *resid_out,
 // SwiGLU: h = [gate | up] rows of 2*inter; out = silu(gate) * up.
 // ---------------------------------------------------------------------------
 __global__ void k_swiglu(const bf16 *__restrict__ h, bf16 *__restrict__ out,
-                         size_t rows, int inter) {
-  size_t i = ((size_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
-  size_t stride = (size_t)gridDim.x * blockDim.x * 8;
-  size_t total = rows * (size_t)inter;
-  for (; i < total; i += stride) {
-    size_t r = i / inter;
-    size_t c = i % inter;
-    bf16x8 g = *(const bf16x8 *)(h + r * 2 * inter + c);
-    bf16x8 u = *(const bf16x8 *)(h + r * 2 * inter + inter + c);
-    bf16x8 o;
+                         int rows, int inter) {
+  // row-block mapping: no per-element 64-bit div/mod (the div version
+  // measured 0.77 TB/s — 8x off HBM)
+  for (int r = blockIdx.x; r < rows; r += gridDim.x) {
+    const bf16 *hg = h + (size_t)r * 2 * inter;
+    const bf16 *hu = hg + inter;
+    bf16 *orow = out + (size_t)r * inter;
+    for (int c = threadIdx.x * 8; c < inter; c += blockDim.x * 8) {
+      bf16x8 g = *(const bf16x8 *)(hg + c);
+      bf16x8 u = *(const bf16x8 *)(hu + c);
+      bf16x8 o;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      float gf = bf2f(g[j]);
-      float s = gf / (1.f + __expf(-gf));
-      o[j] = (bf16)(s * bf2f(u[j]));
+      for (int j = 0; j < 8; ++j) {
+        float gf = bf2f(g[j]);
+        float s = gf / (1.f + __expf(-gf));
+        o[j] = (bf16)(s * bf2f(u[j]));
+      }
+      *(bf16x8 *)(orow + c) = o;
     }
-    *(bf16x8 *)(out + i) = o;
   }
 }
 
 void launch_swiglu(const void *h, void *out, int rows, int inter,
                    hipStream_t stream) {
   if (inter % 8) throw std::runtime_error("swiglu: inter % 8 != 0");
-  size_t work = (size_t)rows * inter / 8;
-  int blocks = (int)((work + 255) / 256);
-  if (blocks > 2048) blocks = 2048;
+  int blocks = rows < 2048 ? rows : 2048;
+  if (blocks == 0) blocks = 1;
   hipLaunchKernelGGL(k_swiglu, dim3(blocks), dim3(256), 0, stream,
-                     (const bf16 *)h, (bf16 *)out, (size_t)rows, inter);
+                     (const bf16 *)h, (bf16 *)out, rows, inter);
 }
 
 // ---------------------------------------------------------------------------
