@@ -82,6 +82,12 @@ void launch_rs_reduce_bf16(const void *segments, void *out, int world,
                            int rank, int m_per_rank, int ws_stride, int n,
                            hipStream_t stream);
 
+// 256^2-tile split-K decode tier (gemm256.hip): fp32 atomic accumulation
+// into ws[M,N] + convert. Requires m%256==0, n%256==0, k%(128*sk)==0,
+// ldc==n.
+void launch_gemm256_sk_bf16(const GemmArgs &g, float *ws, int sk,
+                            hipStream_t stream);
+
 // kernels/gemm_splitk.hip ---------------------------------------------------
 void launch_gemm_splitk_bf16(const GemmArgs &g, float *ws, int splits,
                              hipStream_t stream);

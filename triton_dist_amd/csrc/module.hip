@@ -346,6 +346,15 @@ static void gemm_splitk_bf16(uintptr_t a, uintptr_t b, uintptr_t c,
   TD_CHECK_HIP(hipGetLastError());
 }
 
+static void gemm256_sk_bf16(uintptr_t a, uintptr_t b, uintptr_t c,
+                            uintptr_t bias, uintptr_t ws, int m, int n,
+                            int k, int sk, uintptr_t stream) {
+  GemmArgs args{(void *)a, (void *)b, (void *)c, (void *)bias,
+                m, n, k, k, k, n};
+  launch_gemm256_sk_bf16(args, (float *)ws, sk, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
 static void ag_gemm_consumer_splitk_bf16(uintptr_t a, uintptr_t b,
                                          uintptr_t c, uintptr_t ws, int m,
                                          int n, int k, uintptr_t flags,
@@ -657,6 +666,7 @@ PYBIND11_MODULE(_C, m) {
   m.def("gemm_rs_producer_bf16", &gemm_rs_producer_bf16);
   m.def("rs_reduce_bf16", &rs_reduce_bf16);
   m.def("gemm_splitk_bf16", &gemm_splitk_bf16);
+  m.def("gemm256_sk_bf16", &gemm256_sk_bf16);
   m.def("ag_gemm_consumer_splitk_bf16", &ag_gemm_consumer_splitk_bf16);
   m.def("gemm_rs_producer_splitk_bf16", &gemm_rs_producer_splitk_bf16);
   m.def("allreduce_oneshot", &allreduce_oneshot);

@@ -96,9 +96,9 @@ def ag_gemm(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
 
     if world == 1:
         # the gather is the identity: this is a PLAIN GEMM — route it to
-        # the fastest library implementation (hipBLASLt)
-        c = torch.matmul(a, w.t(), out=out) if out is not None \
-            else a @ w.t()
+        # the fastest plain-GEMM backend (ops.gemm.best_gemm)
+        from .gemm import best_gemm
+        c = best_gemm(a, w, out=out)
         if gathered_out:
             return c, a
         return c

@@ -40,6 +40,52 @@ def splitk_ws(m: int, n: int, splits: int, device) -> torch.Tensor:
     return torch.empty(splits, m, n, dtype=torch.float32, device=device)
 
 
+def sk256_pick(m: int, n: int, k: int) -> int:
+    """Split factor for the 256^2 fp32-atomic split-K tier, or 0 if the
+    shape doesn't profit. Measured (M=512 decode shapes, MI355X): wins only
+    in the K >> N regime (down-proj: 188 us vs hipBLASLt 262); loses when
+    N is large (atomic+convert traffic) or K is short (pipeline prologue).
+    Target ~200 workgroups (256 CUs, 1 wg/CU at 128 KiB LDS)."""
+    if m % 256 or n % 256 or k < 4 * n or k < 8192:
+        return 0
+    grid = (m // 256) * (n // 256)
+    best, best_d = 0, 1 << 30
+    for s in (2, 4, 5, 8, 10, 16, 20, 25):
+        if k % (128 * s) or (k // 128 // s) < 4:
+            continue
+        d = abs(grid * s - 200)
+        if d < best_d:
+            best, best_d = s, d
+    return best
+
+
+def best_gemm(a: torch.Tensor, w: torch.Tensor,
+              bias: torch.Tensor | None = None,
+              out: torch.Tensor | None = None) -> torch.Tensor:
+    """C = A @ W^T + bias via the fastest available plain-GEMM backend:
+    hipBLASLt (torch.matmul) in general, the in-house gemm256_sk tier for
+    K>>N decode shapes where it measures faster."""
+    m, k = a.shape
+    n = w.shape[0]
+    if a.is_cuda and a.dtype == torch.bfloat16:
+        sk = sk256_pick(m, n, k)
+        if sk:
+            if out is None:
+                out = torch.empty(m, n, dtype=torch.bfloat16,
+                                  device=a.device)
+            ws = torch.empty(m, n, dtype=torch.float32, device=a.device)
+            _native().gemm256_sk_bf16(
+                a.data_ptr(), w.data_ptr(), out.data_ptr(),
+                bias.data_ptr() if bias is not None else 0,
+                ws.data_ptr(), m, n, k, sk,
+                torch.cuda.current_stream().cuda_stream)
+            return out
+    c = torch.matmul(a, w.t(), out=out) if out is not None else a @ w.t()
+    if bias is not None:
+        c += bias
+    return c
+
+
 def gemm(a: torch.Tensor, w: torch.Tensor, bias: torch.Tensor | None = None,
          out: torch.Tensor | None = None) -> torch.Tensor:
     """C[M,N] = A[M,K] @ W[N,K]^T + bias."""

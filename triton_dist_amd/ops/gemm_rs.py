@@ -78,9 +78,10 @@ def gemm_rs(a: torch.Tensor, w: torch.Tensor, ctx: GemmRSContext,
         return res
 
     if world == 1:
-        # no scatter/reduce at world 1: plain GEMM via hipBLASLt
-        return torch.matmul(a, w.t(), out=out) if out is not None \
-            else a @ w.t()
+        # no scatter/reduce at world 1: plain GEMM (hipBLASLt, or the
+        # in-house split-K tier for K>>N shapes — see ops.gemm.best_gemm)
+        from .gemm import best_gemm
+        return best_gemm(a, w, out=out)
 
     heap, _C = ctx.heap, ctx.heap._C
     assert a.dtype == torch.bfloat16 and a.is_contiguous()
