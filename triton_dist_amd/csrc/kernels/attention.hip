@@ -36,22 +36,28 @@ __global__ __launch_bounds__(256) void k_flash_decode(
     float scale) {
   // QK^T on MFMA: S[16 q-slots][32 pos] = Q[16,128] x K^T[128,32] — each
   // of the 4 waves contributes one 32-dim k-chunk (2 mfma_16x16x32) and
-  // partials reduce through LDS. K staged in LDS; V read direct (L2).
+  // partials reduce through LDS. PV also on MFMA: P (bf16, standard
+  // flash-attention practice) x V[32,128], V staged row-major in LDS and
+  // consumed column-major via ds_read_b64_tr_b16 (the VALU PV loop this
+  // replaced was 63% of kernel time — see scripts/probe/probe_attn.hip).
   const int b = blockIdx.x;
   const int kh = blockIdx.y;
   const int G = qh / kvh;
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
   const int lane = tid & 63;
-  const int g = tid >> 5;   // softmax/PV role: query slot
-  const int t = tid & 31;   // softmax/PV role: position
+  const int g = tid >> 5;   // softmax role: query slot
+  const int t = tid & 31;   // softmax role: position
   const long seqlen = *offset + 1;
 
   __shared__ bf16 k_lds[kTile][kD];
+  // V rows padded +8 elems: tr_read's 4-rows-x-4-colchunks per 16-lane
+  // group land on rotated banks (row stride 272 B = 4-bank rotate)
+  __shared__ bf16 v_lds[kTile][kD + 8];
   __shared__ bf16 q_lds[16][kD];          // rows >= G zero-padded
+  __shared__ bf16 p_bf[16][kTile + 8];    // P tile as MFMA A operand
   __shared__ float s_part[4][16][kTile];  // per-wave QK partials
-  __shared__ float p_lds[8][kTile];
-  __shared__ float m_lds[8], r_lds[8], l_lds[8];
+  __shared__ float m_lds[8], r_lds[16], l_lds[8];
 
   for (int i = tid; i < 16 * kD / 8; i += 256) {
     int hh = i / (kD / 8);
@@ -65,14 +71,17 @@ __global__ __launch_bounds__(256) void k_flash_decode(
     m_lds[tid] = -1e30f;
     l_lds[tid] = 0.f;
   }
+  if (tid < 16) {
+    r_lds[tid] = 1.f;  // rows 8..15 never rescaled (P rows are zero)
+    for (int tt = 0; tt < kTile; ++tt) p_bf[tid][tt] = (bf16)0.f;
+  }
   __syncthreads();
 
   // hoist the Q fragment (constant across tiles): wave w covers k-chunk w
   bf16x8 qfrag = *(const bf16x8 *)(
       &q_lds[lane & 15][wave * 32 + (lane >> 4) * 8]);
 
-  float acc[4] = {};
-  const int my_d0 = t * 4;
+  f32x4 accPV[2] = {};
   const long ntiles = (seqlen + kTile - 1) / kTile;
   for (long tile = 0; tile < ntiles; ++tile) {
     const long pos0 = tile * kTile;
@@ -81,12 +90,14 @@ __global__ __launch_bounds__(256) void k_flash_decode(
       int r = i / (kD / 8);
       int c = (i % (kD / 8)) * 8;
       long pos = pos0 + r;
-      bf16x8 kv{};
+      bf16x8 kv{}, vv{};
       if (pos < seqlen) {
         size_t base = (((size_t)b * max_len + pos) * kvh + kh) * kD + c;
         kv = *(const bf16x8 *)(kcache + base);
+        vv = *(const bf16x8 *)(vcache + base);
       }
       *(bf16x8 *)(&k_lds[r][c]) = kv;
+      *(bf16x8 *)(&v_lds[r][c]) = vv;
     }
     __syncthreads();
 
@@ -115,7 +126,7 @@ __global__ __launch_bounds__(256) void k_flash_decode(
     float m_old = m_lds[g];
     float m_new = fmaxf(m_old, mx);
     float p = (s > -1e29f) ? __expf(s - m_new) : 0.f;
-    p_lds[g][t] = p;
+    p_bf[g][t] = (bf16)p;
     float psum = p;
     for (int off = 16; off > 0; off >>= 1) psum += __shfl_xor(psum, off);
     if (t == 0) {
@@ -125,27 +136,50 @@ __global__ __launch_bounds__(256) void k_flash_decode(
       m_lds[g] = m_new;
     }
     __syncthreads();
-    const float r = r_lds[g];
+
+    // MFMA PV: wave w owns output cols w*32..w*32+31 (2 col-groups of 16).
+    // tr_read semantics (measured, scripts/probe/probe_attn.hip): within a
+    // 16-lane group, lane k supplies an 8B-word address; lane l receives
+    // elem (l&3) of the words fetched by lanes ((l&15)>>2)+4j, j=0..3. So
+    // lane l pointing at V row k0+((l&15)>>2), col-chunk 4*(l&3) makes the
+    // instruction deliver V[k0+j][col l&15]: a column-major B-fragment
+    // from PLAIN row-major V.
+    bf16x8 afrag = *(const bf16x8 *)(&p_bf[lane & 15][(lane >> 4) * 8]);
 #pragma unroll
-    for (int j = 0; j < 4; ++j) acc[j] *= r;
-    const long lim = min((long)kTile, seqlen - pos0);
-    typedef __attribute__((ext_vector_type(4))) bf16 bf16x4;
-    const bf16 *vbase = vcache +
-        (((size_t)b * max_len + pos0) * kvh + kh) * kD + my_d0;
-    for (int tt = 0; tt < (int)lim; ++tt) {
-      float p2 = p_lds[g][tt];
-      bf16x4 vv = *(const bf16x4 *)(vbase + (size_t)tt * kvh * kD);
+    for (int h = 0; h < 2; ++h) {
+      const int cg = wave * 2 + h;
+      const int k0 = (lane >> 4) * 8;
+      unsigned addr = (unsigned)(uintptr_t)(
+          &v_lds[k0 + ((lane & 15) >> 2)][cg * 16 + 4 * (lane & 3)]);
+      unsigned long long lo, hi;
+      asm volatile(
+          "ds_read_b64_tr_b16 %0, %2\n"
+          "ds_read_b64_tr_b16 %1, %2 offset:%3\n"
+          "s_waitcnt lgkmcnt(0)"
+          : "=v"(lo), "=v"(hi)
+          : "v"(addr), "i"(4 * (kD + 8) * 2));
+      bf16x8 bfrag;
+      *(unsigned long long *)&bfrag = lo;
+      *((unsigned long long *)&bfrag + 1) = hi;
 #pragma unroll
-      for (int j = 0; j < 4; ++j) acc[j] += p2 * (float)vv[j];
+      for (int r = 0; r < 4; ++r)
+        accPV[h][r] *= r_lds[(lane >> 4) * 4 + r];
+      accPV[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag,
+                                                         accPV[h], 0, 0, 0);
     }
   }
   __syncthreads();
-  if (g < G) {
-    float inv_l = 1.f / l_lds[g];
-    bf16 *dst = out + (((size_t)b * qh) + kh * G + g) * kD + my_d0;
 #pragma unroll
-    for (int j = 0; j < 4; ++j) dst[j] = (bf16)(acc[j] * inv_l);
-  }
+  for (int h = 0; h < 2; ++h)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = (lane >> 4) * 4 + r;
+      if (row < G) {
+        float inv_l = 1.f / l_lds[row];
+        out[(((size_t)b * qh) + kh * G + row) * kD + wave * 32 + h * 16 +
+            (lane & 15)] = (bf16)(accPV[h][r] * inv_l);
+      }
+    }
 }
 
 void launch_flash_decode(const void *q, const void *kcache,

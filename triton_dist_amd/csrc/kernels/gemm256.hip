@@ -355,7 +355,7 @@ void launch_gemm256_sk_bf16(const GemmArgs &g, float *ws, int sk,
   if (g.k % (BK * sk))
     throw std::runtime_error("gemm256_sk: k % (128*sk) != 0");
   size_t elems = (size_t)g.m * g.n;
-  hipMemsetAsync(ws, 0, elems * sizeof(float), stream);
+  TD_CHECK_HIP(hipMemsetAsync(ws, 0, elems * sizeof(float), stream));
   int grid = (g.m / BM) * (g.n / BN) * sk;
   hipLaunchKernelGGL(k_gemm256_sk_bf16, dim3(grid), dim3(NTH), 0, stream, g,
                      ws, sk);
