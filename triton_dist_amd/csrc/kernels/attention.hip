@@ -212,14 +212,16 @@ __global__ __launch_bounds__(256) void k_flash_decode_partial(
   const int kh = blockIdx.y;
   const int G = qh / kvh;
   const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
   const int g = tid >> 5;
   const int t = tid & 31;
   const long seqlen = *chunk_len;
 
   __shared__ bf16 k_lds[kTile][kD];
-  __shared__ bf16 v_lds[kTile][kD];
-  __shared__ float p_lds[8][kTile];
-  __shared__ float m_lds[8], r_lds[8], l_lds[8];
+  __shared__ bf16 v_lds[kTile][kD + 8];  // padded for tr_read (see above)
+  __shared__ bf16 p_bf[16][kTile + 8];
+  __shared__ float m_lds[8], r_lds[16], l_lds[8];
   __shared__ bf16 q_lds[8][kD];
 
   for (int i = tid; i < 8 * kD / 8; i += 256) {
@@ -234,10 +236,13 @@ __global__ __launch_bounds__(256) void k_flash_decode_partial(
     m_lds[tid] = -1e30f;
     l_lds[tid] = 0.f;
   }
+  if (tid < 16) {
+    r_lds[tid] = 1.f;
+    for (int tt = 0; tt < kTile; ++tt) p_bf[tid][tt] = (bf16)0.f;
+  }
   __syncthreads();
 
-  float acc[4] = {};
-  const int my_d0 = t * 4;
+  f32x4 accPV[2] = {};
   const long ntiles = (seqlen + kTile - 1) / kTile;
   for (long tile = 0; tile < ntiles; ++tile) {
     const long pos0 = tile * kTile;
@@ -274,7 +279,7 @@ __global__ __launch_bounds__(256) void k_flash_decode_partial(
     float m_old = m_lds[g];
     float m_new = fmaxf(m_old, mx);
     float p = (s > -1e29f) ? __expf(s - m_new) : 0.f;
-    p_lds[g][t] = p;
+    p_bf[g][t] = (bf16)p;
     float psum = p;
     for (int off = 16; off > 0; off >>= 1) psum += __shfl_xor(psum, off);
     if (t == 0) {
@@ -284,31 +289,49 @@ __global__ __launch_bounds__(256) void k_flash_decode_partial(
       m_lds[g] = m_new;
     }
     __syncthreads();
-    const float r = r_lds[g];
+    // MFMA PV, same structure as k_flash_decode above
+    bf16x8 afrag = *(const bf16x8 *)(&p_bf[lane & 15][(lane >> 4) * 8]);
 #pragma unroll
-    for (int j = 0; j < 4; ++j) acc[j] *= r;
-    for (int tt = 0; tt < kTile; ++tt) {
-      float p = p_lds[g][tt];
-      if (p != 0.f) {
+    for (int h = 0; h < 2; ++h) {
+      const int cg = wave * 2 + h;
+      const int k0 = (lane >> 4) * 8;
+      unsigned addr = (unsigned)(uintptr_t)(
+          &v_lds[k0 + ((lane & 15) >> 2)][cg * 16 + 4 * (lane & 3)]);
+      unsigned long long lo, hi;
+      asm volatile(
+          "ds_read_b64_tr_b16 %0, %2\n"
+          "ds_read_b64_tr_b16 %1, %2 offset:%3\n"
+          "s_waitcnt lgkmcnt(0)"
+          : "=v"(lo), "=v"(hi)
+          : "v"(addr), "i"(4 * (kD + 8) * 2));
+      bf16x8 bfrag;
+      *(unsigned long long *)&bfrag = lo;
+      *((unsigned long long *)&bfrag + 1) = hi;
 #pragma unroll
-        for (int j = 0; j < 4; ++j)
-          acc[j] += p * (float)v_lds[tt][my_d0 + j];
-      }
+      for (int r = 0; r < 4; ++r)
+        accPV[h][r] *= r_lds[(lane >> 4) * 4 + r];
+      accPV[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag,
+                                                         accPV[h], 0, 0, 0);
     }
   }
   __syncthreads();
-  if (g < G) {
-    float l = l_lds[g];
-    float inv_l = l > 0.f ? 1.f / l : 0.f;
-    float *dst = out_part + (((size_t)b * qh) + kh * G + g) * kD + my_d0;
 #pragma unroll
-    for (int j = 0; j < 4; ++j) dst[j] = acc[j] * inv_l;
-    if (t == 0) {
-      float m = m_lds[g];
-      lse[(size_t)b * qh + kh * G + g] =
-          (l > 0.f) ? m + __logf(l) : -1e30f;
+  for (int h = 0; h < 2; ++h)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = (lane >> 4) * 4 + r;
+      if (row < G) {
+        float l = l_lds[row];
+        float inv_l = l > 0.f ? 1.f / l : 0.f;
+        out_part[(((size_t)b * qh) + kh * G + row) * kD + wave * 32 +
+                 h * 16 + (lane & 15)] = accPV[h][r] * inv_l;
+        if (h == 0 && wave == 0 && (lane & 15) == 0) {
+          float m = m_lds[row];
+          lse[(size_t)b * qh + kh * G + row] =
+              (l > 0.f) ? m + __logf(l) : -1e30f;
+        }
+      }
     }
-  }
 }
 
 void launch_flash_decode_partial(const void *q, const void *kcache,
