@@ -128,31 +128,47 @@ __global__ void k_moe_dispatch(PeerTable pt, const bf16 *__restrict__ x,
                                size_t flags_off, unsigned *arrive,
                                const int *val_cell, int T, int K, int H,
                                int e_loc) {
-  const int i = blockIdx.x;  // copy index t*K+k
-  const int e = topk_ids[i];
-  const int pos = send_pos[i];
-  if (e < 0 || pos < 0) return;
-  const int dst = e / e_loc;
-  const int t = i / K;
-  const int slot = send_base[e] + pos;
-  bf16 *rx = (bf16 *)((char *)pt.bases[dst] + recv_x_off) + (size_t)slot * H;
-  const bf16 *src = x + (size_t)t * H;
-  for (int c = threadIdx.x * 8; c < H; c += blockDim.x * 8)
-    *(bf16x8 *)(rx + c) = *(const bf16x8 *)(src + c);
+  // Grid-stride over copy indices with PER-BLOCK arrive batching: one
+  // acq_rel atomic per (block, dst) instead of one per row — T*K serialized
+  // RMWs on a single counter measured ~100 us/call at T*K=4096. Only the
+  // LAST arrival pays the system-release (L2 writeback) before publishing
+  // the flag; every earlier block's writes are ordered by its acq_rel
+  // device-scope arrive.
+  __shared__ int cnt[kMaxRanks];
+  if (threadIdx.x < kMaxRanks) cnt[threadIdx.x] = 0;
+  __syncthreads();
+  for (int i = blockIdx.x; i < T * K; i += gridDim.x) {
+    const int e = topk_ids[i];
+    const int pos = send_pos[i];
+    if (e < 0 || pos < 0) continue;
+    const int dst = e / e_loc;
+    const int t = i / K;
+    const int slot = send_base[e] + pos;
+    bf16 *rx =
+        (bf16 *)((char *)pt.bases[dst] + recv_x_off) + (size_t)slot * H;
+    const bf16 *src = x + (size_t)t * H;
+    for (int c = threadIdx.x * 8; c < H; c += blockDim.x * 8)
+      *(bf16x8 *)(rx + c) = *(const bf16x8 *)(src + c);
+    if (threadIdx.x == 0) {
+      int *meta = (int *)((char *)pt.bases[dst] + meta_off);
+      meta[slot * 2] = pt.rank;
+      meta[slot * 2 + 1] = i;  // t*K+k
+      ++cnt[dst];
+    }
+  }
+  __threadfence_block();
+  __syncthreads();
   if (threadIdx.x == 0) {
-    int *meta = (int *)((char *)pt.bases[dst] + meta_off);
-    meta[slot * 2] = pt.rank;
-    meta[slot * 2 + 1] = i;  // t*K+k
-    __threadfence_block();
-    // the acq_rel device-scope arrive orders every block's writes to the
-    // coherence point; only the LAST arrival pays the system-release
-    // (L2 writeback) before publishing the flag — a per-block sys fence
-    // measured 15x the data-movement floor on this kernel.
-    unsigned prev = atomic_add<Scope::Gpu>(arrive + dst, 1u);
-    if ((int)prev == send_to_dst[dst] - 1) {
-      fence_release_sys();
-      int *fl = (int *)((char *)pt.bases[dst] + flags_off);
-      st_release<Scope::Sys>(fl + pt.rank, val_cell ? *val_cell : 1);
+#pragma unroll
+    for (int dst = 0; dst < kMaxRanks; ++dst) {
+      int c = cnt[dst];
+      if (dst >= pt.world || c == 0) continue;
+      unsigned prev = atomic_add<Scope::Gpu>(arrive + dst, (unsigned)c);
+      if ((int)prev + c == send_to_dst[dst]) {
+        fence_release_sys();
+        int *fl = (int *)((char *)pt.bases[dst] + flags_off);
+        st_release<Scope::Sys>(fl + pt.rank, val_cell ? *val_cell : 1);
+      }
     }
   }
 }
@@ -603,22 +619,34 @@ __global__ void k_moe_combine_send(PeerTable pt,
                                    size_t combine_off, size_t cflags_off,
                                    unsigned *arrive, const int *val_cell,
                                    int H) {
-  const int r = blockIdx.x;
-  if (r >= recv_total[0]) return;
-  const int src = meta[r * 2];
-  const int tok_k = meta[r * 2 + 1];
-  bf16 *dst = (bf16 *)((char *)pt.bases[src] + combine_off) +
-              (size_t)tok_k * H;
-  const bf16 *row = expert_out + (size_t)r * H;
-  for (int c = threadIdx.x * 8; c < H; c += blockDim.x * 8)
-    *(bf16x8 *)(dst + c) = *(const bf16x8 *)(row + c);
+  // grid-stride + per-block arrive batching, same scheme as k_moe_dispatch
+  __shared__ int cnt[kMaxRanks];
+  if (threadIdx.x < kMaxRanks) cnt[threadIdx.x] = 0;
+  __syncthreads();
+  const int total = recv_total[0];
+  for (int r = blockIdx.x; r < total; r += gridDim.x) {
+    const int src = meta[r * 2];
+    const int tok_k = meta[r * 2 + 1];
+    bf16 *dst = (bf16 *)((char *)pt.bases[src] + combine_off) +
+                (size_t)tok_k * H;
+    const bf16 *row = expert_out + (size_t)r * H;
+    for (int c = threadIdx.x * 8; c < H; c += blockDim.x * 8)
+      *(bf16x8 *)(dst + c) = *(const bf16x8 *)(row + c);
+    if (threadIdx.x == 0) ++cnt[src];
+  }
+  __threadfence_block();
+  __syncthreads();
   if (threadIdx.x == 0) {
-    // see k_moe_dispatch: sys-release only on the last arrival
-    unsigned prev = atomic_add<Scope::Gpu>(arrive + src, 1u);
-    if ((int)prev == recv_from_src[src] - 1) {
-      fence_release_sys();
-      int *fl = (int *)((char *)pt.bases[src] + cflags_off);
-      st_release<Scope::Sys>(fl + pt.rank, val_cell ? *val_cell : 1);
+#pragma unroll
+    for (int src = 0; src < kMaxRanks; ++src) {
+      int c = cnt[src];
+      if (src >= pt.world || c == 0) continue;
+      unsigned prev = atomic_add<Scope::Gpu>(arrive + src, (unsigned)c);
+      if ((int)prev + c == recv_from_src[src]) {
+        fence_release_sys();
+        int *fl = (int *)((char *)pt.bases[src] + cflags_off);
+        st_release<Scope::Sys>(fl + pt.rank, val_cell ? *val_cell : 1);
+      }
     }
   }
 }
@@ -690,7 +718,8 @@ void launch_moe_dispatch(const PeerTable &pt, const void *x,
                          unsigned *arrive, const void *val_cell, int T,
                          int K, int H, int e_loc, hipStream_t stream) {
   if (H % 8) throw std::runtime_error("moe dispatch: H % 8 != 0");
-  hipLaunchKernelGGL(k_moe_dispatch, dim3(T * K), dim3(256), 0, stream, pt,
+  hipLaunchKernelGGL(k_moe_dispatch, dim3(T * K < 512 ? T * K : 512),
+                     dim3(256), 0, stream, pt,
                      (const bf16 *)x, (const int *)topk_ids,
                      (const int *)send_pos, (const int *)send_base,
                      (const int *)send_to_dst, recv_x_off, meta_off,
@@ -792,7 +821,8 @@ void launch_moe_combine_send(const PeerTable &pt, const void *expert_out,
                              size_t cflags_off, unsigned *arrive,
                              const void *val_cell, int cap, int H,
                              hipStream_t stream) {
-  hipLaunchKernelGGL(k_moe_combine_send, dim3(cap), dim3(256), 0, stream, pt,
+  hipLaunchKernelGGL(k_moe_combine_send, dim3(cap < 512 ? cap : 512),
+                     dim3(256), 0, stream, pt,
                      (const bf16 *)expert_out, (const int *)meta,
                      (const int *)recv_total, (const int *)recv_from_src,
                      combine_off, cflags_off, arrive, (const int *)val_cell,
