@@ -346,6 +346,14 @@ static void gemm_splitk_bf16(uintptr_t a, uintptr_t b, uintptr_t c,
   TD_CHECK_HIP(hipGetLastError());
 }
 
+static void moe_router(uintptr_t x, uintptr_t w, uintptr_t ids,
+                       uintptr_t tw, int T, int H, int E, int K, bool norm,
+                       uintptr_t stream) {
+  launch_moe_router((const void *)x, (const void *)w, (void *)ids,
+                    (void *)tw, T, H, E, K, norm, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
 static void gemm256_sk_bf16(uintptr_t a, uintptr_t b, uintptr_t c,
                             uintptr_t bias, uintptr_t ws, int m, int n,
                             int k, int sk, uintptr_t stream) {
@@ -667,6 +675,7 @@ PYBIND11_MODULE(_C, m) {
   m.def("rs_reduce_bf16", &rs_reduce_bf16);
   m.def("gemm_splitk_bf16", &gemm_splitk_bf16);
   m.def("gemm256_sk_bf16", &gemm256_sk_bf16);
+  m.def("moe_router", &moe_router);
   m.def("ag_gemm_consumer_splitk_bf16", &ag_gemm_consumer_splitk_bf16);
   m.def("gemm_rs_producer_splitk_bf16", &gemm_rs_producer_splitk_bf16);
   m.def("allreduce_oneshot", &allreduce_oneshot);

@@ -47,6 +47,18 @@ class EPMoELayer:
 
     def route(self, x: torch.Tensor):
         """softmax top-k router; returns (topk_ids int32, topk_w fp32)."""
+        if x.is_cuda and x.dtype == torch.bfloat16 and self.hidden <= 4096:
+            from .. import _C
+            t = x.shape[0]
+            ids = torch.empty(t, self.topk, dtype=torch.int32,
+                              device=x.device)
+            tw = torch.empty(t, self.topk, dtype=torch.float32,
+                             device=x.device)
+            _C.moe_router(x.data_ptr(), self.router.data_ptr(),
+                          ids.data_ptr(), tw.data_ptr(), t, self.hidden,
+                          self.n_experts, self.topk, self.norm_topk,
+                          torch.cuda.current_stream().cuda_stream)
+            return ids, tw
         logits = (x.float() @ self.router.float().t())
         probs = torch.softmax(logits, dim=-1)
         topk_w, topk_ids = torch.topk(probs, self.topk, dim=-1)
