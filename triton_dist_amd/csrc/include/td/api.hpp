@@ -96,9 +96,8 @@ void launch_ag_gemm_consumer_splitk_bf16(const AgGemmArgs &a, float *ws,
 void launch_gemm_rs_producer_splitk_bf16(const GemmRsArgs &a, float *ws,
                                          int splits, hipStream_t stream);
 
-void launch_moe_router(const void *x, const void *w, void *topk_ids,
-                       void *topk_w, int T, int H, int E, int K, bool norm,
-                       hipStream_t stream);
+void launch_moe_router(const void *logits, void *topk_ids, void *topk_w,
+                       int T, int E, int K, bool norm, hipStream_t stream);
 
 // kernels/allreduce.hip -----------------------------------------------------
 void launch_allreduce_oneshot(const PeerTable &pt, const void *x, void *out,

@@ -108,11 +108,16 @@ __global__ void k_moe_layout(const int *__restrict__ all_splits, int rank,
       recv_from_src[s] = r;
     }
     // work queue for the persistent grouped GEMM: one item per
-    // (expert, bm-row-tile) with nonzero rows; encoded le*65536 + tile
+    // (expert, bm-row-tile) with nonzero rows; encoded le*65536 + tile.
+    // rows come from the LDS copy — re-reading expert_rows[] just written
+    // to global was a serial ~400 ns round-trip per expert (41 us/call).
     if (work_items) {
       int n = 0;
       for (int le = 0; le < e_loc; ++le) {
-        int tiles = (expert_rows[le] + bm - 1) / bm;
+        int rows = 0;
+        for (int s2 = 0; s2 < world; ++s2)
+          rows += sp[s2 * e_num + rank * e_loc + le];
+        int tiles = (rows + bm - 1) / bm;
         for (int t2 = 0; t2 < tiles; ++t2) work_items[n++] = le * 65536 + t2;
       }
       work_count[0] = n;
@@ -128,62 +133,60 @@ __global__ void k_moe_layout(const int *__restrict__ all_splits, int rank,
 // ~0.4 ms). Block per token; W rows served from L2 (E*H*2 bytes total).
 // Reference behavior: Triton-distributed kernels' topk-gating (capability).
 // ---------------------------------------------------------------------------
-__global__ __launch_bounds__(256) void k_moe_router(
-    const bf16 *__restrict__ x, const bf16 *__restrict__ w,
-    int *__restrict__ topk_ids, float *__restrict__ topk_w, int H, int E,
-    int K, int norm) {
+__global__ __launch_bounds__(64) void k_moe_router(
+    const bf16 *__restrict__ logits, int *__restrict__ topk_ids,
+    float *__restrict__ topk_w, int E, int K, int norm) {
+  // One wave per token; logits come from a (fast, MFMA) bf16 matmul.
   constexpr int MAXE = 1024;
-  __shared__ bf16 xr[4096];
   __shared__ float lg[MAXE];
   const int t = blockIdx.x;
-  const int tid = threadIdx.x;
-  for (int c = tid * 8; c < H; c += blockDim.x * 8)
-    *(bf16x8 *)(&xr[c]) = *(const bf16x8 *)(x + (size_t)t * H + c);
+  const int lane = threadIdx.x;
+  for (int e = lane; e < E; e += 64)
+    lg[e] = (float)logits[(size_t)t * E + e];
   __syncthreads();
-  for (int e = tid; e < E; e += blockDim.x) {
-    const bf16 *wr = w + (size_t)e * H;
-    float d = 0.f;
-    for (int c = 0; c < H; c += 8) {
-      bf16x8 a = *(const bf16x8 *)(&xr[c]);
-      bf16x8 b = *(const bf16x8 *)(wr + c);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) d += (float)a[j] * (float)b[j];
+  // softmax denom (wave-parallel)
+  float mx = -1e30f;
+  for (int e = lane; e < E; e += 64) mx = fmaxf(mx, lg[e]);
+  for (int off = 32; off > 0; off >>= 1)
+    mx = fmaxf(mx, __shfl_xor(mx, off));
+  float den = 0.f;
+  for (int e = lane; e < E; e += 64) den += __expf(lg[e] - mx);
+  for (int off = 32; off > 0; off >>= 1) den += __shfl_xor(den, off);
+  // K wave-parallel argmax passes, first-index tie-break (torch.topk)
+  float wsum = 0.f;
+  for (int k = 0; k < K; ++k) {
+    float bv = -1e30f;
+    int bi = MAXE;
+    for (int e = lane; e < E; e += 64)
+      if (lg[e] > bv || (lg[e] == bv && e < bi)) { bv = lg[e]; bi = e; }
+    for (int off = 32; off > 0; off >>= 1) {
+      float ov = __shfl_xor(bv, off);
+      int oi = __shfl_xor(bi, off);
+      if (ov > bv || (ov == bv && oi < bi)) { bv = ov; bi = oi; }
     }
-    lg[e] = d;
-  }
-  __syncthreads();
-  if (tid == 0) {
-    float mx = -1e30f;
-    for (int e = 0; e < E; ++e) mx = fmaxf(mx, lg[e]);
-    float den = 0.f;
-    for (int e = 0; e < E; ++e) den += __expf(lg[e] - mx);
-    float wsum = 0.f;
-    for (int k = 0; k < K; ++k) {
-      int best = 0;
-      float bv = -1e30f;
-      for (int e = 0; e < E; ++e)
-        if (lg[e] > bv) { bv = lg[e]; best = e; }
-      float p = __expf(bv - mx) / den;
-      topk_ids[(size_t)t * K + k] = best;
+    float p = __expf(bv - mx) / den;
+    if (lane == 0) {
+      topk_ids[(size_t)t * K + k] = bi;
       topk_w[(size_t)t * K + k] = p;
-      wsum += p;
-      lg[best] = -1e30f;
     }
-    if (norm) {
-      float inv = 1.f / wsum;
-      for (int k = 0; k < K; ++k) topk_w[(size_t)t * K + k] *= inv;
-    }
+    wsum += p;
+    __syncthreads();
+    if (lane == (bi & 63)) lg[bi] = -1e30f;  // mask winner for next pass
+    __syncthreads();
+  }
+  if (norm && lane == 0) {
+    float inv = 1.f / wsum;
+    for (int k = 0; k < K; ++k) topk_w[(size_t)t * K + k] *= inv;
   }
 }
 
-void launch_moe_router(const void *x, const void *w, void *topk_ids,
-                       void *topk_w, int T, int H, int E, int K, bool norm,
-                       hipStream_t stream) {
-  if (H % 8 || H > 4096 || E > 1024 || K > 32)
-    throw std::runtime_error("moe_router: H%8, H<=4096, E<=1024, K<=32");
-  hipLaunchKernelGGL(k_moe_router, dim3(T), dim3(256), 0, stream,
-                     (const bf16 *)x, (const bf16 *)w, (int *)topk_ids,
-                     (float *)topk_w, H, E, K, norm ? 1 : 0);
+void launch_moe_router(const void *logits, void *topk_ids, void *topk_w,
+                       int T, int E, int K, bool norm, hipStream_t stream) {
+  if (E > 1024 || K > 32)
+    throw std::runtime_error("moe_router: E<=1024, K<=32");
+  hipLaunchKernelGGL(k_moe_router, dim3(T), dim3(64), 0, stream,
+                     (const bf16 *)logits, (int *)topk_ids, (float *)topk_w,
+                     E, K, norm ? 1 : 0);
 }
 
 // ---------------------------------------------------------------------------

@@ -346,11 +346,10 @@ static void gemm_splitk_bf16(uintptr_t a, uintptr_t b, uintptr_t c,
   TD_CHECK_HIP(hipGetLastError());
 }
 
-static void moe_router(uintptr_t x, uintptr_t w, uintptr_t ids,
-                       uintptr_t tw, int T, int H, int E, int K, bool norm,
-                       uintptr_t stream) {
-  launch_moe_router((const void *)x, (const void *)w, (void *)ids,
-                    (void *)tw, T, H, E, K, norm, as_stream(stream));
+static void moe_router(uintptr_t logits, uintptr_t ids, uintptr_t tw,
+                       int T, int E, int K, bool norm, uintptr_t stream) {
+  launch_moe_router((const void *)logits, (void *)ids, (void *)tw, T, E, K,
+                    norm, as_stream(stream));
   TD_CHECK_HIP(hipGetLastError());
 }
 

@@ -47,16 +47,16 @@ class EPMoELayer:
 
     def route(self, x: torch.Tensor):
         """softmax top-k router; returns (topk_ids int32, topk_w fp32)."""
-        if x.is_cuda and x.dtype == torch.bfloat16 and self.hidden <= 4096:
+        if x.is_cuda and x.dtype == torch.bfloat16:
             from .. import _C
             t = x.shape[0]
+            logits = torch.matmul(x, self.router.t())  # bf16 MFMA
             ids = torch.empty(t, self.topk, dtype=torch.int32,
                               device=x.device)
             tw = torch.empty(t, self.topk, dtype=torch.float32,
                              device=x.device)
-            _C.moe_router(x.data_ptr(), self.router.data_ptr(),
-                          ids.data_ptr(), tw.data_ptr(), t, self.hidden,
-                          self.n_experts, self.topk, self.norm_topk,
+            _C.moe_router(logits.data_ptr(), ids.data_ptr(), tw.data_ptr(),
+                          t, self.n_experts, self.topk, self.norm_topk,
                           torch.cuda.current_stream().cuda_stream)
             return ids, tw
         logits = (x.float() @ self.router.float().t())
