@@ -109,6 +109,17 @@ void launch_allreduce_twoshot(const PeerTable &pt, const void *x, void *out,
                               size_t flags_in_off, size_t flags_out_off,
                               size_t elems, int chunks, hipStream_t stream);
 
+// kernels/collectives.hip ---------------------------------------------------
+void launch_reduce_scatter(const PeerTable &pt, const void *x,
+                           size_t inbox_off, size_t flags_off,
+                           const void *local_inbox, const void *local_flags,
+                           void *out, size_t seg_elems, int chunks,
+                           const void *tag_cell, hipStream_t stream);
+void launch_ll_allgather(const PeerTable &pt, const void *x,
+                         size_t inbox_off, const void *local_inbox,
+                         void *out, int words, const void *tag_cell,
+                         hipStream_t stream);
+
 // kernels/moe.hip ------------------------------------------------------------
 void launch_moe_count(const void *topk_ids, void *counts, void *send_pos,
                       void *send_to_dst, int total, int e_num, int e_loc,

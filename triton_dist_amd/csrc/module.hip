@@ -346,6 +346,27 @@ static void gemm_splitk_bf16(uintptr_t a, uintptr_t b, uintptr_t c,
   TD_CHECK_HIP(hipGetLastError());
 }
 
+static void reduce_scatter_op(uintptr_t x, size_t inbox_off,
+                              size_t flags_off, uintptr_t local_inbox,
+                              uintptr_t local_flags, uintptr_t out,
+                              size_t seg_elems, int chunks,
+                              uintptr_t tag_cell, uintptr_t stream) {
+  launch_reduce_scatter(g_heap.pt, (const void *)x, inbox_off, flags_off,
+                        (const void *)local_inbox, (const void *)local_flags,
+                        (void *)out, seg_elems, chunks,
+                        (const void *)tag_cell, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
+static void ll_allgather_op(uintptr_t x, size_t inbox_off,
+                            uintptr_t local_inbox, uintptr_t out, int words,
+                            uintptr_t tag_cell, uintptr_t stream) {
+  launch_ll_allgather(g_heap.pt, (const void *)x, inbox_off,
+                      (const void *)local_inbox, (void *)out, words,
+                      (const void *)tag_cell, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
 static void moe_router(uintptr_t logits, uintptr_t ids, uintptr_t tw,
                        int T, int E, int K, bool norm, uintptr_t stream) {
   launch_moe_router((const void *)logits, (void *)ids, (void *)tw, T, E, K,
@@ -675,6 +696,8 @@ PYBIND11_MODULE(_C, m) {
   m.def("gemm_splitk_bf16", &gemm_splitk_bf16);
   m.def("gemm256_sk_bf16", &gemm256_sk_bf16);
   m.def("moe_router", &moe_router);
+  m.def("reduce_scatter", &reduce_scatter_op);
+  m.def("ll_allgather", &ll_allgather_op);
   m.def("ag_gemm_consumer_splitk_bf16", &ag_gemm_consumer_splitk_bf16);
   m.def("gemm_rs_producer_splitk_bf16", &gemm_rs_producer_splitk_bf16);
   m.def("allreduce_oneshot", &allreduce_oneshot);

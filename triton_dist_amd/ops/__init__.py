@@ -52,3 +52,10 @@ from .p2p import (  # noqa: F401
     p2p_recv,
     PPCommLayer,
 )
+from .collectives import (  # noqa: F401
+    CollContext,
+    create_coll_context,
+    reduce_scatter,
+    ll_all_gather,
+    reduce_scatter_ref,
+)
