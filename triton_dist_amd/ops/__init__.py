@@ -59,3 +59,9 @@ from .collectives import (  # noqa: F401
     ll_all_gather,
     reduce_scatter_ref,
 )
+from .moe_tp import (  # noqa: F401
+    moe_sort_tokens,
+    grouped_gemm,
+    tp_moe_forward,
+    tp_moe_ref,
+)

@@ -1,0 +1,157 @@
+"""TP-MoE: AllGather + grouped GEMM and grouped GEMM + topk-reduce +
+ReduceScatter — the tensor-parallel MoE path (experts replicated, each
+rank holding an intermediate-dim shard of EVERY expert).
+
+Capability parity (behavior only) with Triton-distributed's MoE-TP pair:
+  * AG + GroupGEMM  — kernels/nvidia/allgather_group_gemm.py /
+    group_gemm.py, plus the moe_ag_scatter_align sort
+    (csrc/lib/moe_utils.cu:61-356).
+  * GroupGEMM + RS  — kernels/nvidia/moe_reduce_rs.py:168-730
+    (grouped GEMM -> weighted topk reduce -> reduce_scatter).
+
+MI355X design: the gather rides the symmetric-heap push allgather, the
+expert-sorted grouped GEMMs ride the same persistent work-queue kernel
+as the EP path (csrc/kernels/moe.hip k_moe_grouped_gemm_pq), the final
+cross-rank sum of intermediate-shard partials rides the standalone
+reduce_scatter (flag-in-payload free: chunked release flags). The
+token sort itself is index arithmetic (argsort + index_select), not a
+custom kernel — CDNA4 has no cheap global atomics win over torch's sort
+at prefill sizes, and this op is not on the decode hot path.
+"""
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+import torch.distributed as dist
+
+from ..runtime.symm_mem import get_heap
+
+
+def moe_sort_tokens(x_full: torch.Tensor, topk_ids: torch.Tensor,
+                    n_experts: int, bm: int = 32
+                    ) -> Tuple[torch.Tensor, torch.Tensor, dict]:
+    """Expert-sort the gathered tokens: returns (x_sorted [M*K+slack, H],
+    tok_index [M*K] (source token of each sorted row), meta dict with the
+    grouped-GEMM descriptors expert_base/expert_rows/work_items/
+    work_count, all int32 on device)."""
+    mk = topk_ids.numel()
+    k = topk_ids.shape[1]
+    flat = topk_ids.reshape(-1).to(torch.int64)
+    order = torch.argsort(flat, stable=True)
+    tok = order // k
+    x_sorted = torch.zeros(mk + 128, x_full.shape[1], dtype=x_full.dtype,
+                           device=x_full.device)
+    x_sorted[:mk] = x_full.index_select(0, tok)
+    counts = torch.bincount(flat, minlength=n_experts)
+    base = torch.cumsum(counts, 0) - counts
+    cnt = counts.cpu().tolist()
+    items = []
+    for e, c in enumerate(cnt):
+        for t in range((c + bm - 1) // bm):
+            items.append(e * 65536 + t)
+    dev = x_full.device
+    meta = {
+        "expert_base": base.to(torch.int32).to(dev).contiguous(),
+        "expert_rows": counts.to(torch.int32).to(dev).contiguous(),
+        "work_items": torch.tensor(items or [0], dtype=torch.int32,
+                                   device=dev),
+        "work_count": torch.tensor([len(items)], dtype=torch.int32,
+                                   device=dev),
+        "order": order,
+    }
+    return x_sorted, tok, meta
+
+
+def grouped_gemm(x_sorted: torch.Tensor, weights: torch.Tensor,
+                 meta: dict, out: Optional[torch.Tensor] = None
+                 ) -> torch.Tensor:
+    """out[rows of expert e] = x_sorted[rows] @ weights[e]^T via the
+    persistent work-queue kernel. weights: [E, N, K]; N % 128 == 0,
+    K % 64 == 0."""
+    n = weights.shape[1]
+    kk = weights.shape[2]
+    if out is None:
+        out = torch.empty(x_sorted.shape[0], n, dtype=x_sorted.dtype,
+                          device=x_sorted.device)
+    if not x_sorted.is_cuda:
+        eb = meta["expert_base"].tolist()
+        er = meta["expert_rows"].tolist()
+        for e in range(weights.shape[0]):
+            lo, r = eb[e], er[e]
+            if r:
+                out[lo:lo + r] = (x_sorted[lo:lo + r].float()
+                                  @ weights[e].float().t()).to(out.dtype)
+        return out
+    from .. import _C
+    s = torch.cuda.current_stream().cuda_stream
+    _C.moe_grouped_gemm_pq(x_sorted.data_ptr(), weights.data_ptr(),
+                           out.data_ptr(), meta["expert_base"].data_ptr(),
+                           meta["expert_rows"].data_ptr(),
+                           meta["work_items"].data_ptr(),
+                           meta["work_count"].data_ptr(), n, kk, s)
+    return out
+
+
+def tp_moe_forward(x_shard: torch.Tensor, topk_ids: torch.Tensor,
+                   topk_w: torch.Tensor, w_gate_up: torch.Tensor,
+                   w_down: torch.Tensor, ag_ctx, coll_ctx) -> torch.Tensor:
+    """Full TP-MoE block: AG(x_shard) -> router-sorted grouped gate/up ->
+    SwiGLU -> grouped down (intermediate-shard partials) -> weighted topk
+    reduce -> reduce_scatter. topk_ids/topk_w are for the FULL gathered
+    batch [M, K] (router is replicated + deterministic).
+
+    w_gate_up: [E, 2*inter_shard, H] (my shard of gate rows then up rows),
+    w_down: [E, H, inter_shard]. Returns [m_local, H].
+    """
+    from .allgather_gemm import allgather
+    from .collectives import reduce_scatter
+    from .fused import swiglu_op
+
+    heap = get_heap()
+    world = heap.world
+    e_num = w_gate_up.shape[0]
+    inter_shard = w_gate_up.shape[1] // 2
+    h = x_shard.shape[1]
+    m_local = x_shard.shape[0]
+
+    x_full = allgather(x_shard, ag_ctx) if world > 1 else x_shard
+    mk = topk_ids.numel()
+    x_sorted, tok, meta = moe_sort_tokens(x_full, topk_ids, e_num)
+
+    hidden = grouped_gemm(x_sorted, w_gate_up, meta)
+    act = swiglu_op(hidden, inter_shard)
+    part = grouped_gemm(act, w_down, meta)
+
+    w_sorted = topk_w.reshape(-1)[meta["order"]].to(torch.float32)
+    y_full = torch.zeros(world * m_local, h, dtype=torch.float32,
+                         device=x_shard.device)
+    y_full.index_add_(0, tok, part[:mk].float() * w_sorted[:, None])
+    y_full = y_full.to(x_shard.dtype)
+
+    if world == 1:
+        return y_full
+    return reduce_scatter(y_full, coll_ctx)
+
+
+def tp_moe_ref(x_full: torch.Tensor, topk_ids: torch.Tensor,
+               topk_w: torch.Tensor, w_gate_up_full: torch.Tensor,
+               w_down_full: torch.Tensor, world: int,
+               rank: int) -> torch.Tensor:
+    """Golden single-process reference with FULL (unsharded) weights:
+    x_full [M, H] (the gathered batch), w_gate_up_full [E, 2*inter, H]
+    ([gate; up]), w_down_full [E, H, inter]. Returns rank's [M/world, H]
+    segment of the full MoE output."""
+    import torch.nn.functional as F
+
+    inter = w_down_full.shape[2]
+    y = torch.zeros(x_full.shape[0], x_full.shape[1], dtype=torch.float32)
+    for t in range(x_full.shape[0]):
+        for j in range(topk_ids.shape[1]):
+            e = int(topk_ids[t, j])
+            hv = x_full[t].float() @ w_gate_up_full[e].float().t()
+            a = F.silu(hv[:inter]) * hv[inter:]
+            y[t] += float(topk_w[t, j]) * (a @ w_down_full[e].float().t())
+    m_local = x_full.shape[0] // world
+    lo = rank * m_local
+    return y[lo:lo + m_local].to(x_full.dtype)
