@@ -38,3 +38,26 @@ def test_autotune_decorator(tmp_path, monkeypatch):
 
     assert op(7) == 1
     assert op(7) == 1
+
+
+def test_contextual_autotuner(tmp_path, monkeypatch):
+    import time as _time
+
+    monkeypatch.setenv("TD_AUTOTUNE_DIR", str(tmp_path))
+    from triton_dist_amd.tune import ContextualAutoTuner
+
+    calls = []
+
+    def make_composite(cfg):
+        def run():
+            calls.append(cfg["n"])
+            _time.sleep(0.0005 * cfg["n"])  # composite cost grows with n
+        return run
+
+    t = ContextualAutoTuner("demo", [{"n": 1}, {"n": 4}], warmup=1, iters=3)
+    best = t.tune("k0", make_composite)
+    assert best == {"n": 1}
+    # cached second call: no re-benchmark
+    before = len(calls)
+    assert t.tune("k0", make_composite) == {"n": 1}
+    assert len(calls) == before

@@ -133,3 +133,57 @@ def autotune(name: str, configs: List[Dict[str, Any]],
         return wrapped
 
     return deco
+
+
+class ContextualAutoTuner:
+    """Tunes an op IN CONTEXT: instead of timing the op alone, it times a
+    caller-supplied composite closure (producer + op + consumer), so the
+    chosen config accounts for stream contention, cache state, and
+    overlap with neighbors.
+
+    Capability parity with Triton-distributed's ContextualAutoTuner
+    (python/triton_dist/autotuner.py:43-105 — behavior only). Usage:
+
+        ctuner = ContextualAutoTuner("ag_gemm_ctx", configs)
+        cfg = ctuner.tune(key, make_composite)
+        # make_composite(config) -> zero-arg callable running the WHOLE
+        # surrounding region once with the op configured by `config`.
+
+    Shares the AutoTuner cache/broadcast machinery (distributed-safe:
+    MAX-reduced timings, rank-0 decision broadcast, JSON persistence
+    under ~/.triton_dist_amd/autotune/<hw_hash>/).
+    """
+
+    def __init__(self, name: str, configs: List[Dict[str, Any]],
+                 warmup: int = 2, iters: int = 5):
+        self._inner = AutoTuner(f"ctx_{name}", configs, warmup=warmup,
+                                iters=iters)
+
+    @property
+    def configs(self):
+        return self._inner.configs
+
+    def tune(self, key: str,
+             make_composite: Callable[[Dict[str, Any]], Callable]
+             ) -> Dict[str, Any]:
+        return self._inner.tune(key, make_composite)
+
+
+def contextual_autotune(name: str, configs: List[Dict[str, Any]],
+                        key: Callable[..., str]):
+    """Decorator form: the wrapped fn IS the composite region; the chosen
+    config is injected as `tune_config=`."""
+    tuner = ContextualAutoTuner(name, configs)
+
+    def deco(fn):
+        @functools.wraps(fn)
+        def wrapped(*args, **kwargs):
+            k = key(*args, **kwargs)
+            cfg = tuner.tune(
+                k, lambda c: (lambda: fn(*args, tune_config=c, **kwargs)))
+            return fn(*args, tune_config=cfg, **kwargs)
+
+        wrapped.tuner = tuner
+        return wrapped
+
+    return deco
