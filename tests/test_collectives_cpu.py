@@ -8,8 +8,8 @@ from tests.conftest import run_distributed
 
 def _worker_rs(rank, world):
     import triton_dist_amd as td
-    from triton_dist_amd.ops import (create_coll_context, reduce_scatter,
-                                     ll_all_gather)
+    from triton_dist_amd.ops import (all_to_all_single, create_coll_context,
+                                     reduce_scatter, ll_all_gather)
 
     td.init_symm_heap(size_mb=32)
     ctx = create_coll_context(max_seg_elems=4096, max_ll_words=1024)
@@ -28,6 +28,13 @@ def _worker_rs(rank, world):
     for r in range(world):
         exp = torch.arange(8, dtype=torch.float32).reshape(2, 4) + r * 100
         assert torch.equal(gathered[r * 2:(r + 1) * 2], exp)
+    z = ((torch.arange(world * 16 * 8, dtype=torch.float32) % 97)
+         .reshape(world * 16, 8).to(torch.bfloat16) + rank)
+    a2a = all_to_all_single(z, ctx)
+    for p in range(world):
+        exp = z.float().reshape(world, 16, 8)[rank] + (p - rank)
+        got = a2a.float().reshape(world, 16, 8)[p]
+        assert torch.equal(got, exp), (rank, p)
     td.shutdown_heap()
 
 

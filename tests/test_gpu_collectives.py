@@ -11,8 +11,8 @@ pytestmark = pytest.mark.gpu
 
 def _body_collectives(rank, world):
     import triton_dist_amd as td
-    from triton_dist_amd.ops import (create_coll_context, reduce_scatter,
-                                     ll_all_gather)
+    from triton_dist_amd.ops import (all_to_all_single, create_coll_context,
+                                     reduce_scatter, ll_all_gather)
 
     td.init_symm_heap(size_mb=64)
     ctx = create_coll_context(max_seg_elems=1 << 16, max_ll_words=4096)
@@ -38,6 +38,16 @@ def _body_collectives(rank, world):
             exp = (torch.arange(512, dtype=torch.float32).reshape(4, 128)
                    + r * 1000)
             assert torch.equal(got[r * 4:(r + 1) * 4], exp), (it, r)
+    z = ((torch.arange(world * 64 * 16, dtype=torch.float32) % 101)
+         .reshape(world * 64, 16).to(torch.bfloat16) + rank * 7).cuda()
+    for it in range(2):
+        a2a = all_to_all_single(z, ctx)
+        torch.cuda.synchronize()
+        got = a2a.float().cpu().reshape(world, 64, 16)
+        zc = z.float().cpu().reshape(world, 64, 16)
+        for p in range(world):
+            exp = zc[rank] + (p - rank) * 7
+            assert torch.equal(got[p], exp), (it, rank, p)
     td.shutdown_heap()
 
 

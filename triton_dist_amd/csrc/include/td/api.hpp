@@ -115,6 +115,10 @@ void launch_reduce_scatter(const PeerTable &pt, const void *x,
                            const void *local_inbox, const void *local_flags,
                            void *out, size_t seg_elems, int chunks,
                            const void *tag_cell, hipStream_t stream);
+void launch_all_to_all(const PeerTable &pt, const void *x, size_t inbox_off,
+                       size_t flags_off, const void *local_inbox,
+                       const void *local_flags, void *out, size_t seg_elems,
+                       int chunks, const void *tag_cell, hipStream_t stream);
 void launch_ll_allgather(const PeerTable &pt, const void *x,
                          size_t inbox_off, const void *local_inbox,
                          void *out, int words, const void *tag_cell,

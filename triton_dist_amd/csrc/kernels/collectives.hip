@@ -94,6 +94,53 @@ void launch_reduce_scatter(const PeerTable &pt, const void *x,
 }
 
 // --------------------------------------------------------------------------
+// all_to_all_single: the push half IS k_rs_push (segment p -> rank p's
+// inbox row [my rank]); this kernel is the receive half — flag-gated copy
+// of the inbox rows (+ the local segment straight from x) into out.
+// Parity: reference all_to_all_single_2d (kernels/nvidia a2a family —
+// behavior only).
+// --------------------------------------------------------------------------
+__global__ void k_a2a_recv(PeerTable pt, const bf16 *__restrict__ x,
+                           const bf16 *__restrict__ inbox,
+                           const int *__restrict__ flags,
+                           bf16 *__restrict__ out, size_t seg_elems,
+                           int chunks, const int *__restrict__ tag_cell) {
+  const int chunk = (int)blockIdx.x;
+  const int src = (int)blockIdx.y;
+  const int tag = *tag_cell;
+  if (src != pt.rank && threadIdx.x == 0)
+    wait_ge_one<Scope::Sys>(flags + src * chunks + chunk, tag);
+  __syncthreads();
+  const size_t per =
+      (((seg_elems + chunks - 1) / chunks) + 7) & ~(size_t)7;
+  const size_t lo = (size_t)chunk * per;
+  const size_t hi = lo + per < seg_elems ? lo + per : seg_elems;
+  const bf16 *srcp = src == pt.rank
+                         ? x + (size_t)pt.rank * seg_elems
+                         : inbox + (size_t)src * seg_elems;
+  bf16 *dst = out + (size_t)src * seg_elems;
+  typedef __attribute__((ext_vector_type(8))) bf16 v8;
+  for (size_t i = lo + threadIdx.x * 8; i + 8 <= hi; i += blockDim.x * 8)
+    *(v8 *)(dst + i) = *(const v8 *)(srcp + i);
+}
+
+void launch_all_to_all(const PeerTable &pt, const void *x, size_t inbox_off,
+                       size_t flags_off, const void *local_inbox,
+                       const void *local_flags, void *out, size_t seg_elems,
+                       int chunks, const void *tag_cell,
+                       hipStream_t stream) {
+  if (pt.world > 1) {
+    hipLaunchKernelGGL(k_rs_push, dim3(chunks, pt.world - 1), dim3(256), 0,
+                       stream, pt, (const bf16 *)x, inbox_off, flags_off,
+                       seg_elems, chunks, (const int *)tag_cell);
+  }
+  hipLaunchKernelGGL(k_a2a_recv, dim3(chunks, pt.world), dim3(256), 0,
+                     stream, pt, (const bf16 *)x, (const bf16 *)local_inbox,
+                     (const int *)local_flags, (bf16 *)out, seg_elems,
+                     chunks, (const int *)tag_cell);
+}
+
+// --------------------------------------------------------------------------
 // LL allgather
 // --------------------------------------------------------------------------
 

@@ -358,6 +358,17 @@ static void reduce_scatter_op(uintptr_t x, size_t inbox_off,
   TD_CHECK_HIP(hipGetLastError());
 }
 
+static void all_to_all_op(uintptr_t x, size_t inbox_off, size_t flags_off,
+                          uintptr_t local_inbox, uintptr_t local_flags,
+                          uintptr_t out, size_t seg_elems, int chunks,
+                          uintptr_t tag_cell, uintptr_t stream) {
+  launch_all_to_all(g_heap.pt, (const void *)x, inbox_off, flags_off,
+                    (const void *)local_inbox, (const void *)local_flags,
+                    (void *)out, seg_elems, chunks, (const void *)tag_cell,
+                    as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
 static void ll_allgather_op(uintptr_t x, size_t inbox_off,
                             uintptr_t local_inbox, uintptr_t out, int words,
                             uintptr_t tag_cell, uintptr_t stream) {
@@ -698,6 +709,7 @@ PYBIND11_MODULE(_C, m) {
   m.def("moe_router", &moe_router);
   m.def("reduce_scatter", &reduce_scatter_op);
   m.def("ll_allgather", &ll_allgather_op);
+  m.def("all_to_all", &all_to_all_op);
   m.def("ag_gemm_consumer_splitk_bf16", &ag_gemm_consumer_splitk_bf16);
   m.def("gemm_rs_producer_splitk_bf16", &gemm_rs_producer_splitk_bf16);
   m.def("allreduce_oneshot", &allreduce_oneshot);

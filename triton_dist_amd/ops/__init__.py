@@ -57,6 +57,7 @@ from .collectives import (  # noqa: F401
     create_coll_context,
     reduce_scatter,
     ll_all_gather,
+    all_to_all_single,
     reduce_scatter_ref,
 )
 from .moe_tp import (  # noqa: F401

@@ -129,6 +129,37 @@ def ll_all_gather(x: torch.Tensor, ctx: CollContext,
     return out
 
 
+def all_to_all_single(x: torch.Tensor, ctx: CollContext,
+                      out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """out[p-th segment] = rank p's x[my segment]: the classic
+    all_to_all_single over the symmetric heap (push + flag-gated gather).
+    x: [world * m, ...] bf16; segment elems must be a multiple of 8."""
+    world = ctx.world
+    assert x.shape[0] % world == 0
+    seg_elems = x.numel() // world
+    if out is None:
+        out = torch.empty_like(x)
+
+    if ctx.heap.backend != "hip":
+        # gloo has no all_to_all: all_gather everything, pick my column
+        gathered = [torch.empty_like(x) for _ in range(world)]
+        dist.all_gather(gathered, x.contiguous())
+        segs = [g.reshape(world, -1)[ctx.rank] for g in gathered]
+        out.copy_(torch.stack(segs).reshape(x.shape))
+        return out
+
+    assert x.dtype == torch.bfloat16 and x.is_contiguous()
+    assert seg_elems % 8 == 0 and seg_elems <= ctx.max_seg_elems
+    from .. import _C
+    stream = torch.cuda.current_stream()
+    _C.bump_cell(ctx.tag_cell.data_ptr(), stream.cuda_stream)
+    _C.all_to_all(x.data_ptr(), ctx.rs_inbox.offset, ctx.rs_flags.offset,
+                  ctx.rs_inbox.ptr(), ctx.rs_flags.ptr(), out.data_ptr(),
+                  seg_elems, ctx.chunks, ctx.tag_cell.data_ptr(),
+                  stream.cuda_stream)
+    return out
+
+
 def reduce_scatter_ref(x: torch.Tensor, world: int,
                        rank: int) -> torch.Tensor:
     """Golden single-process reference (fp32 accumulate)."""
