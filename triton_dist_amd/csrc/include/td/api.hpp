@@ -109,6 +109,12 @@ void launch_allreduce_twoshot(const PeerTable &pt, const void *x, void *out,
                               size_t flags_in_off, size_t flags_out_off,
                               size_t elems, int chunks, hipStream_t stream);
 
+// kernels/gdn.hip ------------------------------------------------------------
+void launch_gdn_decode(const void *q, const void *k, const void *v,
+                       const void *g, const void *beta, void *state,
+                       void *o, int B, int H, int K, int V, float scale,
+                       hipStream_t stream);
+
 // kernels/collectives.hip ---------------------------------------------------
 void launch_reduce_scatter(const PeerTable &pt, const void *x,
                            size_t inbox_off, size_t flags_off,

@@ -346,6 +346,15 @@ static void gemm_splitk_bf16(uintptr_t a, uintptr_t b, uintptr_t c,
   TD_CHECK_HIP(hipGetLastError());
 }
 
+static void gdn_decode(uintptr_t q, uintptr_t k, uintptr_t v, uintptr_t g,
+                       uintptr_t beta, uintptr_t state, uintptr_t o, int B,
+                       int H, int K, int V, float scale, uintptr_t stream) {
+  launch_gdn_decode((const void *)q, (const void *)k, (const void *)v,
+                    (const void *)g, (const void *)beta, (void *)state,
+                    (void *)o, B, H, K, V, scale, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
 static void reduce_scatter_op(uintptr_t x, size_t inbox_off,
                               size_t flags_off, uintptr_t local_inbox,
                               uintptr_t local_flags, uintptr_t out,
@@ -708,6 +717,7 @@ PYBIND11_MODULE(_C, m) {
   m.def("gemm256_sk_bf16", &gemm256_sk_bf16);
   m.def("moe_router", &moe_router);
   m.def("reduce_scatter", &reduce_scatter_op);
+  m.def("gdn_decode", &gdn_decode);
   m.def("ll_allgather", &ll_allgather_op);
   m.def("all_to_all", &all_to_all_op);
   m.def("ag_gemm_consumer_splitk_bf16", &ag_gemm_consumer_splitk_bf16);

@@ -66,3 +66,8 @@ from .moe_tp import (  # noqa: F401
     tp_moe_forward,
     tp_moe_ref,
 )
+from .gdn import (  # noqa: F401
+    gated_delta_rule_recurrent_ref,
+    chunk_gated_delta_rule_fwd,
+    gdn_decode_step,
+)
