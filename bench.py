@@ -14,6 +14,7 @@ from __future__ import annotations
 import argparse
 import json
 import os
+import sys
 import time
 
 import torch
@@ -49,6 +50,18 @@ def main():
         f"launch with torch.distributed.run for --gpus {args.gpus}"
     on_gpu = torch.cuda.is_available()
     device = "cuda" if on_gpu else "cpu"
+    if (on_gpu and world > torch.cuda.device_count()
+            and args.mode == "ag_rs"):
+        # Oversubscribed DEV topology (ranks sharing one GPU): the fused
+        # persistent spin-wait consumers of both ranks saturate the single
+        # GPU's CUs and the peer's flag-producing kernels starve. One rank
+        # per GPU (the benchmark topology) has no such cycle: every spin's
+        # producer is SDMA (CU-free) or stream-ordered before the waiter.
+        print("[bench] world > device_count: oversubscribed sharing "
+              "topology -> falling back to --mode allreduce --no-graph",
+              file=sys.stderr)
+        args.mode = "allreduce"
+        args.no_graph = True  # host collectives are not graph-capturable
     batch = args.batch_per_gpu * world
     max_len = args.ctx + args.warmup + args.steps + 16
 
