@@ -4,7 +4,11 @@ ll_all_gather (latency at small payloads) vs torch.distributed baselines.
 Run: bash scripts/launch.sh 8 benchmark/bench_collectives.py [--check]
 """
 import argparse
+import sys
 import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 
 import torch
 import torch.distributed as dist

@@ -3,7 +3,11 @@ HIP decode-step kernel (state-bandwidth bound).
 
 Run: python benchmark/bench_gdn.py
 """
+import sys
 import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 
 import torch
 import torch.nn.functional as F
