@@ -27,6 +27,7 @@ class ModelConfig:
     # MoE fields (n_experts == 0 -> dense)
     n_experts: int = 0
     moe_topk: int = 8
+    moe_impl: str = "ep"   # "ep" (expert-parallel) | "tp" (inter-sharded)
     moe_inter: int = 0
 
 

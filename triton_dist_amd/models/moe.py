@@ -15,7 +15,10 @@ from typing import Optional
 
 import torch
 
+import os
+
 from ..layers.ep_moe_layer import EPMoELayer
+from ..layers.tp_moe_layer import TPMoELayer
 from ..layers.tp_attn import TP_Attn
 from .config import ModelConfig
 from .dense import DenseLLM
@@ -26,9 +29,12 @@ class Qwen3MoE(DenseLLM):
                  dtype=torch.bfloat16, heap=None):
         assert cfg.n_experts > 0, "MoE config requires n_experts"
         super().__init__(cfg, device=device, dtype=dtype, heap=heap)
-        # replace each dense MLP with an EP MoE layer
+        # EP vs TP MoE: config field, overridable by env (reference parity:
+        # models/qwen_moe.py:75-81 selects EP_MoE/TP_MoE by env EP_MOE)
+        self.moe_impl = os.environ.get("TD_MOE_IMPL", cfg.moe_impl)
+        cls = TPMoELayer if self.moe_impl == "tp" else EPMoELayer
         for layer in self.layers:
-            layer["mlp"] = EPMoELayer(
+            layer["mlp"] = cls(
                 cfg.hidden, cfg.moe_inter, cfg.n_experts, cfg.moe_topk,
                 norm_topk=True, heap=self.heap, device=device, dtype=dtype)
 
@@ -44,14 +50,22 @@ class Qwen3MoE(DenseLLM):
         r, e_loc = self.rank, cfg.n_experts // self.world
         for li, layer in enumerate(self.layers):
             s = seed + 5000 + li * 10
-            moe: EPMoELayer = layer["mlp"]
+            moe = layer["mlp"]
             # larger router std: near-uniform logits make top-k selection
             # flip between bf16 paths, which is routing noise not a bug
             moe.router.copy_(full((cfg.n_experts, cfg.hidden), s) * 25)
             gu = full((cfg.n_experts, 2 * cfg.moe_inter, cfg.hidden), s + 1)
             dn = full((cfg.n_experts, cfg.hidden, cfg.moe_inter), s + 2)
-            moe.w_gate_up.copy_(gu[r * e_loc:(r + 1) * e_loc])
-            moe.w_down.copy_(dn[r * e_loc:(r + 1) * e_loc])
+            if self.moe_impl == "tp":
+                i_s = cfg.moe_inter // self.world
+                moe.w_gate_up.copy_(torch.cat(
+                    [gu[:, r * i_s:(r + 1) * i_s],
+                     gu[:, cfg.moe_inter + r * i_s:
+                        cfg.moe_inter + (r + 1) * i_s]], dim=1))
+                moe.w_down.copy_(dn[:, :, r * i_s:(r + 1) * i_s])
+            else:
+                moe.w_gate_up.copy_(gu[r * e_loc:(r + 1) * e_loc])
+                moe.w_down.copy_(dn[r * e_loc:(r + 1) * e_loc])
 
     def init_dist_ctx(self, max_m_total: int):
         super().init_dist_ctx(max_m_total)  # attention ag_rs contexts

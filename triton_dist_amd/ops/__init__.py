@@ -64,6 +64,7 @@ from .moe_tp import (  # noqa: F401
     moe_sort_tokens,
     grouped_gemm,
     tp_moe_forward,
+    tp_moe_from_full,
     tp_moe_ref,
 )
 from .gdn import (  # noqa: F401

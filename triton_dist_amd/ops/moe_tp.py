@@ -93,18 +93,12 @@ def grouped_gemm(x_sorted: torch.Tensor, weights: torch.Tensor,
     return out
 
 
-def tp_moe_forward(x_shard: torch.Tensor, topk_ids: torch.Tensor,
-                   topk_w: torch.Tensor, w_gate_up: torch.Tensor,
-                   w_down: torch.Tensor, ag_ctx, coll_ctx) -> torch.Tensor:
-    """Full TP-MoE block: AG(x_shard) -> router-sorted grouped gate/up ->
-    SwiGLU -> grouped down (intermediate-shard partials) -> weighted topk
-    reduce -> reduce_scatter. topk_ids/topk_w are for the FULL gathered
-    batch [M, K] (router is replicated + deterministic).
-
-    w_gate_up: [E, 2*inter_shard, H] (my shard of gate rows then up rows),
-    w_down: [E, H, inter_shard]. Returns [m_local, H].
-    """
-    from .allgather_gemm import allgather
+def tp_moe_from_full(x_full: torch.Tensor, topk_ids: torch.Tensor,
+                     topk_w: torch.Tensor, w_gate_up: torch.Tensor,
+                     w_down: torch.Tensor, coll_ctx) -> torch.Tensor:
+    """The compute half of TP-MoE, given the already-gathered batch:
+    sorted grouped gate/up -> SwiGLU -> grouped down (intermediate-shard
+    partials) -> weighted topk reduce -> reduce_scatter."""
     from .collectives import reduce_scatter
     from .fused import swiglu_op
 
@@ -112,10 +106,7 @@ def tp_moe_forward(x_shard: torch.Tensor, topk_ids: torch.Tensor,
     world = heap.world
     e_num = w_gate_up.shape[0]
     inter_shard = w_gate_up.shape[1] // 2
-    h = x_shard.shape[1]
-    m_local = x_shard.shape[0]
-
-    x_full = allgather(x_shard, ag_ctx) if world > 1 else x_shard
+    h = x_full.shape[1]
     mk = topk_ids.numel()
     x_sorted, tok, meta = moe_sort_tokens(x_full, topk_ids, e_num)
 
@@ -124,14 +115,32 @@ def tp_moe_forward(x_shard: torch.Tensor, topk_ids: torch.Tensor,
     part = grouped_gemm(act, w_down, meta)
 
     w_sorted = topk_w.reshape(-1)[meta["order"]].to(torch.float32)
-    y_full = torch.zeros(world * m_local, h, dtype=torch.float32,
-                         device=x_shard.device)
+    y_full = torch.zeros(x_full.shape[0], h, dtype=torch.float32,
+                         device=x_full.device)
     y_full.index_add_(0, tok, part[:mk].float() * w_sorted[:, None])
-    y_full = y_full.to(x_shard.dtype)
+    y_full = y_full.to(x_full.dtype)
 
     if world == 1:
         return y_full
     return reduce_scatter(y_full, coll_ctx)
+
+
+def tp_moe_forward(x_shard: torch.Tensor, topk_ids: torch.Tensor,
+                   topk_w: torch.Tensor, w_gate_up: torch.Tensor,
+                   w_down: torch.Tensor, ag_ctx, coll_ctx) -> torch.Tensor:
+    """Full TP-MoE block: AG(x_shard) -> tp_moe_from_full. topk_ids /
+    topk_w are for the FULL gathered batch [M, K] (router is replicated +
+    deterministic).
+
+    w_gate_up: [E, 2*inter_shard, H] (my shard of gate rows then up rows),
+    w_down: [E, H, inter_shard]. Returns [m_local, H].
+    """
+    from .allgather_gemm import allgather
+
+    world = get_heap().world
+    x_full = allgather(x_shard, ag_ctx) if world > 1 else x_shard
+    return tp_moe_from_full(x_full, topk_ids, topk_w, w_gate_up, w_down,
+                            coll_ctx)
 
 
 def tp_moe_ref(x_full: torch.Tensor, topk_ids: torch.Tensor,
