@@ -38,19 +38,7 @@ def main():
                            device="cuda")
     eng.serve(prompt, gen_len=3)  # prefill + capture + a couple replays
 
-    torch.cuda.synchronize()
-    with torch.profiler.profile(
-            activities=[torch.profiler.ProfilerActivity.CUDA]) as prof:
-        for _ in range(args.steps):
-            eng.graph.replay()
-        torch.cuda.synchronize()
-
-    rows = {}
-    for e in prof.key_averages():
-        t = getattr(e, "self_device_time_total", 0) or 0
-        if t <= 0:
-            continue
-        rows[e.key] = {"us_per_step": t / args.steps, "calls": e.count}
+    rows = eng.profile_decode(steps=args.steps)
     total = sum(r["us_per_step"] for r in rows.values())
     top = sorted(rows.items(), key=lambda kv: -kv[1]["us_per_step"])
     print(f"== decode step kernel totals ({args.model}, B={args.batch}) ==")

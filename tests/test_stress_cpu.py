@@ -44,3 +44,42 @@ def _body_stress(rank, world):
 
 def test_stress_fused_ops_cpu_2rank():
     run_distributed(_body_stress, world_size=2)
+
+
+def _worker_coll_stress(rank, world):
+    """Interleaved repeated collectives on ONE shared context: exercises
+    monotonic-tag reuse (no resets) across many calls and op mixes."""
+    import torch
+    import triton_dist_amd as td
+    from triton_dist_amd.ops import (all_to_all_single, create_coll_context,
+                                     ll_all_gather, reduce_scatter)
+
+    td.init_symm_heap(size_mb=32)
+    ctx = create_coll_context(max_seg_elems=2048, max_ll_words=512)
+    g = torch.Generator().manual_seed(100 + rank)
+    for it in range(12):
+        m = [8, 16, 4][it % 3]
+        x = (torch.randn(world * m, 16, generator=g) * 0.5).to(
+            torch.bfloat16)
+        xs = [torch.empty_like(x) for _ in range(world)]
+        import torch.distributed as dist
+        dist.all_gather(xs, x)
+        out = reduce_scatter(x, ctx)
+        ref = sum(t.float() for t in xs).reshape(world, m, 16)[rank]
+        assert (out.float() - ref).abs().max() < 0.3, it
+        y = ((torch.arange(32, dtype=torch.float32) % 50) + rank + it)
+        got = ll_all_gather(y, ctx)
+        for r in range(world):
+            exp = (torch.arange(32, dtype=torch.float32) % 50) + r + it
+            assert torch.equal(got[r * 32:(r + 1) * 32], exp), (it, r)
+        z = ((torch.arange(world * 8 * 8, dtype=torch.float32) % 60)
+             .reshape(world * 8, 8).to(torch.bfloat16) + rank)
+        a2a = all_to_all_single(z, ctx)
+        for p in range(world):
+            exp = z.float().reshape(world, 8, 8)[rank] + (p - rank)
+            assert torch.equal(a2a.float().reshape(world, 8, 8)[p], exp)
+    td.shutdown_heap()
+
+
+def test_collectives_stress_4rank():
+    run_distributed(_worker_coll_stress, world_size=4)

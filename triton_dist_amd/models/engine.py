@@ -62,6 +62,30 @@ class Engine:
             return self._token_buf
         return self.model.decode_step(tokens, self.kv)
 
+    def profile_decode(self, steps: int = 10,
+                       trace_path: str | None = None) -> dict:
+        """torch-profile `steps` decode replays and return per-kernel
+        totals for ONE step ({name: {us_per_step, calls}}); optionally
+        export a chrome trace (reference parity: Engine's
+        trace_static.json export, models/engine.py:153-179 — behavior
+        only). Requires a prior serve()/decode_once() so the graph is
+        captured."""
+        self._ensure_graph()
+        torch.cuda.synchronize()
+        with torch.profiler.profile(
+                activities=[torch.profiler.ProfilerActivity.CUDA]) as prof:
+            for _ in range(steps):
+                self.graph.replay()
+            torch.cuda.synchronize()
+        if trace_path:
+            prof.export_chrome_trace(trace_path)
+        rows = {}
+        for e in prof.key_averages():
+            t = getattr(e, "self_device_time_total", 0) or 0
+            if t > 0:
+                rows[e.key] = {"us_per_step": t / steps, "calls": e.count}
+        return rows
+
     def serve(self, input_ids: torch.Tensor, gen_len: int) -> torch.Tensor:
         """input_ids: [B, S] prompt -> [B, gen_len] generated (greedy)."""
         b, s = input_ids.shape
