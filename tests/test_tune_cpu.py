@@ -61,3 +61,15 @@ def test_contextual_autotuner(tmp_path, monkeypatch):
     before = len(calls)
     assert t.tune("k0", make_composite) == {"n": 1}
     assert len(calls) == before
+
+
+def test_perf_model_methods():
+    from triton_dist_amd.perf_model import (choose_ag_method,
+                                            ll_allgather_time_us,
+                                            reduce_scatter_time_us)
+
+    assert choose_ag_method(4 << 10, 8) == "ll"      # tiny: latency wins
+    assert choose_ag_method(64 << 20, 8) == "push"   # bulk: wire bytes win
+    assert ll_allgather_time_us(1 << 10, 8) < ll_allgather_time_us(1 << 24,
+                                                                   8)
+    assert reduce_scatter_time_us(1 << 20, 8) > 0

@@ -78,3 +78,42 @@ def allreduce_time_us(nbytes: int, world: int, method: str = "auto") -> float:
 def choose_ar_method(nbytes: int, world: int) -> str:
     return "one_shot" if allreduce_time_us(nbytes, world, "one_shot") <= \
         allreduce_time_us(nbytes, world, "two_shot") else "two_shot"
+
+
+# measured on-device protocol latencies (profiles/README.md; dev-box runs)
+XGMI_ONEWAY_LATENCY_US = 1.6     # small-put visible-at-peer latency
+KERNEL_LAUNCH_US = 2.0           # eager launch overhead (graphs: ~0)
+
+
+def ll_allgather_time_us(payload_bytes: int, world: int) -> float:
+    """Flag-in-payload allgather: one crossing carries data+signal, so
+    latency ~ max(link transfer of 2x payload, one xGMI latency). The 2x
+    is the (data, tag) interleave."""
+    if world <= 1:
+        return KERNEL_LAUNCH_US
+    wire = 2.0 * payload_bytes / (ARCH.xgmi_link_gbps * 1e9) * 1e6
+    return max(wire, XGMI_ONEWAY_LATENCY_US) + KERNEL_LAUNCH_US
+
+
+def choose_ag_method(payload_bytes: int, world: int) -> str:
+    """'ll' (flag-in-payload, 2x wire bytes, zero signal round-trips) for
+    small payloads; 'push' (SDMA bulk + flag copy) once the doubled wire
+    bytes cost more than the saved signal latency."""
+    if world <= 1:
+        return "push"
+    t_ll = ll_allgather_time_us(payload_bytes, world)
+    t_push = (ag_push_time_us(payload_bytes, world)
+              + XGMI_ONEWAY_LATENCY_US + 2 * KERNEL_LAUNCH_US)
+    return "ll" if t_ll < t_push else "push"
+
+
+def reduce_scatter_time_us(nbytes: int, world: int) -> float:
+    """Push-your-segments ((world-1)/world of the buffer leaves over
+    world-1 parallel links) + local reduce (world segments read, one
+    written)."""
+    if world <= 1:
+        return KERNEL_LAUNCH_US
+    seg = nbytes / world
+    t_push = seg / (ARCH.xgmi_link_gbps * 1e9) * 1e6
+    t_reduce = (nbytes + seg) / (ARCH.hbm_achievable_tbps * 1e12) * 1e6
+    return t_push + t_reduce + XGMI_ONEWAY_LATENCY_US
