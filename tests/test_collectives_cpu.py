@@ -44,3 +44,26 @@ def test_reduce_scatter_ll_allgather_2rank():
 
 def test_collectives_4rank():
     run_distributed(_worker_rs, world_size=4)
+
+
+def _worker_a2a_gemm(rank, world):
+    import triton_dist_amd as td
+    from triton_dist_amd.ops import a2a_gemm, create_ag_gemm_context
+
+    td.init_symm_heap(size_mb=32)
+    K, N, M = 32, 24, 8
+    ctx = create_ag_gemm_context(max_m_per_rank=M, k=K)
+    g = torch.Generator().manual_seed(5)
+    xs = [(torch.randn(world * M, K, generator=g) * 0.5).to(torch.bfloat16)
+          for _ in range(world)]
+    w = (torch.randn(N, K, generator=g) * 0.5).to(torch.bfloat16)
+    out = a2a_gemm(xs[rank], w, ctx)
+    mixed = torch.cat([xs[s].reshape(world, M, K)[rank].float()
+                       for s in range(world)])
+    ref = mixed @ w.float().t()
+    assert (out.float() - ref).abs().max() < 0.3
+    td.shutdown_heap()
+
+
+def test_a2a_gemm_2rank():
+    run_distributed(_worker_a2a_gemm, world_size=2)

@@ -174,3 +174,29 @@ def _body_variable_m(rank, world):
 
 def test_variable_m_2rank():
     run_distributed(_body_variable_m, world_size=2)
+
+
+def _body_a2a_gemm(rank, world):
+    import triton_dist_amd as td
+    from triton_dist_amd.ops import a2a_gemm, create_ag_gemm_context
+
+    td.init_symm_heap(size_mb=128)
+    K, N, M = 256, 256, 128
+    ctx = create_ag_gemm_context(max_m_per_rank=M, k=K)
+    g = torch.Generator().manual_seed(5)
+    xs = [(torch.randn(world * M, K, generator=g) * 0.5).to(torch.bfloat16)
+          for _ in range(world)]
+    w = (torch.randn(N, K, generator=g) * 0.5).to(torch.bfloat16).cuda()
+    out = a2a_gemm(xs[rank].cuda(), w, ctx)
+    torch.cuda.synchronize()
+    mixed = torch.cat([xs[s].reshape(world, M, K)[rank].float()
+                       for s in range(world)])
+    ref = mixed @ w.float().cpu().t()
+    err = (out.float().cpu() - ref).abs().max().item()
+    rel = err / ref.abs().max().item()
+    assert rel < 0.05, (rank, rel)
+    td.shutdown_heap()
+
+
+def test_a2a_gemm_gpu_2rank():
+    run_distributed(_body_a2a_gemm, world_size=2)

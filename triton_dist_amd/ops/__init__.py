@@ -1,5 +1,6 @@
 from .gemm import gemm, gemm_ref, gemm_supported  # noqa: F401
 from .allgather_gemm import (  # noqa: F401
+    a2a_gemm,
     AGGemmContext,
     create_ag_gemm_context,
     ag_gemm,

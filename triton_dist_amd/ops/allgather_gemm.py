@@ -180,6 +180,103 @@ def ag_gemm(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
     return out
 
 
+def a2a_gemm(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
+             out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Fused AllToAll + GEMM: a [world*m_seg, K] (segment p destined for
+    rank p) -> the a2a'd matrix [world*m_seg, K] (segment s = rank s's
+    p=rank segment) is consumed tile-by-tile by the same persistent
+    flag-waiting GEMM as ag_gemm, overlapping the SDMA segment exchange
+    with compute. Capability parity (behavior only): the reference's
+    a2a+GEMM overlap (kernels/nvidia/all_to_all_single_gemm.py).
+
+    Only the producer differs from ag_gemm: each push sources a DIFFERENT
+    segment of `a` per peer instead of the whole shard.
+    """
+    m_total, k = a.shape
+    n = w.shape[0]
+    world, rank = ctx.world, ctx.rank
+    assert m_total % world == 0
+    m = m_total // world          # rows per segment
+    assert k == ctx.k and m <= ctx.max_m_per_rank
+
+    if ctx.heap.backend != "hip":
+        from .collectives import all_to_all_single  # gloo-emulated a2a
+        from .collectives import CollContext  # noqa: F401 (doc)
+        import torch.distributed as dist
+        gathered = [torch.empty_like(a) for _ in range(world)]
+        dist.all_gather(gathered, a.contiguous())
+        mixed = torch.cat([g.reshape(world, m, k)[rank]
+                           for g in gathered])
+        c = (mixed.float() @ w.float().t()).to(a.dtype)
+        if out is not None:
+            out.copy_(c)
+            return out
+        return c
+
+    if world == 1:
+        from .gemm import best_gemm
+        return best_gemm(a, w, out=out)
+
+    assert a.dtype == torch.bfloat16 and a.is_contiguous()
+    heap, _C = ctx.heap, ctx.heap._C
+    compute = torch.cuda.current_stream()
+    chunks = ctx.chunks_per_rank
+    assert m % chunks == 0 and m % 128 == 0,         f"segment m={m} must divide chunks={chunks} and tile by 128"
+    rows_per_chunk = m // chunks
+    chunk_bytes = rows_per_chunk * k * 2
+    seg_bytes = m * k * 2
+
+    _C.reset_flags(ctx.flags.ptr(), world * chunks, 0, compute.cuda_stream)
+    heap.barrier_all_on_stream(compute)
+
+    # local: my own p=rank segment
+    my_seg_ptr = ctx.ws.ptr() + rank * ctx.max_m_per_rank * k * 2
+    _C.memcpy_async(my_seg_ptr, a.data_ptr() + rank * seg_bytes, seg_bytes,
+                    compute.cuda_stream)
+    _C.reset_flags(ctx.flags.ptr() + rank * chunks * 4, chunks, 1,
+                   compute.cuda_stream)
+
+    ctx.ready_ev.record(compute)
+    ns = len(ctx.comm_streams)
+    for s in range(ns):
+        ctx.comm_streams[s].wait_event(ctx.ready_ev)
+    for i in range(world - 1):
+        peer = (rank + 1 + i) % world
+        stream = ctx.comm_streams[i % ns]
+        dst_seg = ctx.ws.ptr(peer) + rank * ctx.max_m_per_rank * k * 2
+        dst_flag = ctx.flags.ptr(peer) + rank * chunks * 4
+        src = a.data_ptr() + peer * seg_bytes
+        for c in range(chunks):
+            _C.memcpy_async(dst_seg + c * chunk_bytes,
+                            src + c * chunk_bytes, chunk_bytes,
+                            stream.cuda_stream)
+            _C.memcpy_async(dst_flag + c * 4, heap.one_src.ptr(), 4,
+                            stream.cuda_stream)
+
+    if out is None:
+        out = torch.empty(m_total, n, dtype=torch.bfloat16,
+                          device=a.device)
+    from .gemm import choose_splits, splitk_ws
+
+    splits = choose_splits(m_total, n, k)
+    if splits > 1:
+        ws = splitk_ws(m_total, n, splits, a.device)
+        _C.ag_gemm_consumer_splitk_bf16(
+            ctx.ws.ptr(), w.data_ptr(), out.data_ptr(), ws.data_ptr(),
+            m_total, n, k, ctx.flags.ptr(), chunks, m,
+            ctx.max_m_per_rank, world, rank, 1,
+            splits, compute.cuda_stream)
+    else:
+        _C.ag_gemm_consumer_bf16(
+            ctx.ws.ptr(), w.data_ptr(), out.data_ptr(), m_total, n, k,
+            ctx.flags.ptr(), chunks, m, ctx.max_m_per_rank, world, rank, 1,
+            compute.cuda_stream, 0, 0, 0)
+    for s in range(min(ns, max(world - 1, 1))):
+        ctx.join_evs[s].record(ctx.comm_streams[s])
+        compute.wait_event(ctx.join_evs[s])
+    return out
+
+
 def allgather(a: torch.Tensor, ctx: AGGemmContext,
               out: Optional[torch.Tensor] = None,
               method: str = "push") -> torch.Tensor:
