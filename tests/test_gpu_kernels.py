@@ -212,6 +212,32 @@ def test_qkv_prologue_and_flash_decode():
     assert_allclose(out.view(b, qh, 1, d), ref, atol=4e-2, rtol=4e-2)
 
 
+@pytest.mark.parametrize("qh,kvh", [(10, 1), (16, 1), (12, 2)])
+def test_flash_decode_wide_group(qh, kvh):
+    """G = qh/kvh in (10, 16, 6): the 2-pass softmax sweep (seed-oss-36b
+    TP1 geometry has G=10)."""
+    import torch.nn.functional as F
+
+    from triton_dist_amd.ops.fused import flash_decode_op
+    from triton_dist_amd.utils.testing import assert_allclose
+
+    torch.manual_seed(qh * 31 + kvh)
+    b, d, maxlen, seq = 8, 128, 80, 51
+    q = (torch.randn(b, qh * d, device="cuda") / 4).to(torch.bfloat16)
+    kc = (torch.randn(b, maxlen, kvh, d, device="cuda") / 4).to(
+        torch.bfloat16)
+    vc = (torch.randn(b, maxlen, kvh, d, device="cuda") / 4).to(
+        torch.bfloat16)
+    offset = torch.tensor(seq - 1, dtype=torch.int64, device="cuda")
+    out = flash_decode_op(q, kc, vc, offset, qh, kvh)
+    torch.cuda.synchronize()
+    qs = q.view(b, qh, 1, d).float()
+    ks = kc[:, :seq].transpose(1, 2).float()
+    vs = vc[:, :seq].transpose(1, 2).float()
+    ref = F.scaled_dot_product_attention(qs, ks, vs, enable_gqa=True)
+    assert_allclose(out.view(b, qh, 1, d), ref, atol=4e-2, rtol=4e-2)
+
+
 @pytest.mark.parametrize("m,n,k", [(512, 1280, 5120), (512, 6400, 5120),
                                    (256, 5120, 1024), (128, 128, 4096)])
 def test_gemm_splitk(m, n, k):

@@ -57,7 +57,7 @@ __global__ __launch_bounds__(256) void k_flash_decode(
   __shared__ bf16 q_lds[16][kD];          // rows >= G zero-padded
   __shared__ bf16 p_bf[16][kTile + 8];    // P tile as MFMA A operand
   __shared__ float s_part[4][16][kTile];  // per-wave QK partials
-  __shared__ float m_lds[8], r_lds[16], l_lds[8];
+  __shared__ float m_lds[16], r_lds[16], l_lds[16];
 
   for (int i = tid; i < 16 * kD / 8; i += 256) {
     int hh = i / (kD / 8);
@@ -67,12 +67,10 @@ __global__ __launch_bounds__(256) void k_flash_decode(
       v = *(const bf16x8 *)(q + (((size_t)b * qh) + kh * G + hh) * kD + c);
     *(bf16x8 *)(&q_lds[hh][c]) = v;
   }
-  if (tid < 8) {
+  if (tid < 16) {
     m_lds[tid] = -1e30f;
     l_lds[tid] = 0.f;
-  }
-  if (tid < 16) {
-    r_lds[tid] = 1.f;  // rows 8..15 never rescaled (P rows are zero)
+    r_lds[tid] = 1.f;
     for (int tt = 0; tt < kTile; ++tt) p_bf[tid][tt] = (bf16)0.f;
   }
   __syncthreads();
@@ -115,25 +113,30 @@ __global__ __launch_bounds__(256) void k_flash_decode(
     }
     __syncthreads();
 
-    float s = -1e30f;
-    if (g < G && pos0 + t < seqlen) {
-      s = (s_part[0][g][t] + s_part[1][g][t] + s_part[2][g][t] +
-           s_part[3][g][t]) * scale;
-    }
-    float mx = s;
-    for (int off = 16; off > 0; off >>= 1)
-      mx = fmaxf(mx, __shfl_xor(mx, off));
-    float m_old = m_lds[g];
-    float m_new = fmaxf(m_old, mx);
-    float p = (s > -1e29f) ? __expf(s - m_new) : 0.f;
-    p_bf[g][t] = (bf16)p;
-    float psum = p;
-    for (int off = 16; off > 0; off >>= 1) psum += __shfl_xor(psum, off);
-    if (t == 0) {
-      float r = __expf(m_old - m_new);
-      r_lds[g] = r;
-      l_lds[g] = l_lds[g] * r + psum;
-      m_lds[g] = m_new;
+    // softmax: 8 thread-groups of 32 sweep up to 16 q-slots (2 passes —
+    // seed-oss-36b geometry has G = qh/kvh = 10)
+#pragma unroll
+    for (int gg = g; gg < 16; gg += 8) {
+      float s = -1e30f;
+      if (gg < G && pos0 + t < seqlen) {
+        s = (s_part[0][gg][t] + s_part[1][gg][t] + s_part[2][gg][t] +
+             s_part[3][gg][t]) * scale;
+      }
+      float mx = s;
+      for (int off = 16; off > 0; off >>= 1)
+        mx = fmaxf(mx, __shfl_xor(mx, off));
+      float m_old = m_lds[gg];
+      float m_new = fmaxf(m_old, mx);
+      float p = (s > -1e29f) ? __expf(s - m_new) : 0.f;
+      p_bf[gg][t] = (bf16)p;
+      float psum = p;
+      for (int off = 16; off > 0; off >>= 1) psum += __shfl_xor(psum, off);
+      if (t == 0) {
+        float r = __expf(m_old - m_new);
+        r_lds[gg] = r;
+        l_lds[gg] = l_lds[gg] * r + psum;
+        m_lds[gg] = m_new;
+      }
     }
     __syncthreads();
 
@@ -186,8 +189,8 @@ void launch_flash_decode(const void *q, const void *kcache,
                          const void *vcache, void *out, const void *offset,
                          int batch, int qh, int kvh, int max_len,
                          hipStream_t stream) {
-  if (qh / kvh > 8 || qh % kvh)
-    throw std::runtime_error("flash_decode: qh/kvh must divide and be <= 8");
+  if (qh / kvh > 16 || qh % kvh)
+    throw std::runtime_error("flash_decode: qh/kvh must divide and be <= 16");
   float scale = 1.f / sqrtf((float)kD);
   hipLaunchKernelGGL(k_flash_decode, dim3(batch, kvh), dim3(256), 0, stream,
                      (const bf16 *)q, (const bf16 *)kcache,
@@ -221,10 +224,10 @@ __global__ __launch_bounds__(256) void k_flash_decode_partial(
   __shared__ bf16 k_lds[kTile][kD];
   __shared__ bf16 v_lds[kTile][kD + 8];  // padded for tr_read (see above)
   __shared__ bf16 p_bf[16][kTile + 8];
-  __shared__ float m_lds[8], r_lds[16], l_lds[8];
-  __shared__ bf16 q_lds[8][kD];
+  __shared__ float m_lds[16], r_lds[16], l_lds[16];
+  __shared__ bf16 q_lds[16][kD];
 
-  for (int i = tid; i < 8 * kD / 8; i += 256) {
+  for (int i = tid; i < 16 * kD / 8; i += 256) {
     int hh = i / (kD / 8);
     int c = (i % (kD / 8)) * 8;
     bf16x8 v{};
@@ -232,11 +235,9 @@ __global__ __launch_bounds__(256) void k_flash_decode_partial(
       v = *(const bf16x8 *)(q + (((size_t)b * qh) + kh * G + hh) * kD + c);
     *(bf16x8 *)(&q_lds[hh][c]) = v;
   }
-  if (tid < 8) {
+  if (tid < 16) {
     m_lds[tid] = -1e30f;
     l_lds[tid] = 0.f;
-  }
-  if (tid < 16) {
     r_lds[tid] = 1.f;
     for (int tt = 0; tt < kTile; ++tt) p_bf[tid][tt] = (bf16)0.f;
   }
@@ -261,32 +262,35 @@ __global__ __launch_bounds__(256) void k_flash_decode_partial(
       *(bf16x8 *)(&v_lds[r][c]) = vv;
     }
     __syncthreads();
-    float s = -1e30f;
-    if (g < G && pos0 + t < seqlen) {
-      float d = 0.f;
 #pragma unroll
-      for (int c = 0; c < kD / 8; ++c) {
-        bf16x8 qv = *(const bf16x8 *)(&q_lds[g][c * 8]);
-        bf16x8 kv = *(const bf16x8 *)(&k_lds[t][c * 8]);
+    for (int gg = g; gg < 16; gg += 8) {
+      float s = -1e30f;
+      if (gg < G && pos0 + t < seqlen) {
+        float d = 0.f;
 #pragma unroll
-        for (int j = 0; j < 8; ++j) d += (float)qv[j] * (float)kv[j];
+        for (int c = 0; c < kD / 8; ++c) {
+          bf16x8 qv = *(const bf16x8 *)(&q_lds[gg][c * 8]);
+          bf16x8 kv = *(const bf16x8 *)(&k_lds[t][c * 8]);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) d += (float)qv[j] * (float)kv[j];
+        }
+        s = d * scale;
       }
-      s = d * scale;
-    }
-    float mx = s;
-    for (int off = 16; off > 0; off >>= 1)
-      mx = fmaxf(mx, __shfl_xor(mx, off));
-    float m_old = m_lds[g];
-    float m_new = fmaxf(m_old, mx);
-    float p = (s > -1e29f) ? __expf(s - m_new) : 0.f;
-    p_bf[g][t] = (bf16)p;
-    float psum = p;
-    for (int off = 16; off > 0; off >>= 1) psum += __shfl_xor(psum, off);
-    if (t == 0) {
-      float r = __expf(m_old - m_new);
-      r_lds[g] = r;
-      l_lds[g] = l_lds[g] * r + psum;
-      m_lds[g] = m_new;
+      float mx = s;
+      for (int off = 16; off > 0; off >>= 1)
+        mx = fmaxf(mx, __shfl_xor(mx, off));
+      float m_old = m_lds[gg];
+      float m_new = fmaxf(m_old, mx);
+      float p = (s > -1e29f) ? __expf(s - m_new) : 0.f;
+      p_bf[gg][t] = (bf16)p;
+      float psum = p;
+      for (int off = 16; off > 0; off >>= 1) psum += __shfl_xor(psum, off);
+      if (t == 0) {
+        float r = __expf(m_old - m_new);
+        r_lds[gg] = r;
+        l_lds[gg] = l_lds[gg] * r + psum;
+        m_lds[gg] = m_new;
+      }
     }
     __syncthreads();
     // MFMA PV, same structure as k_flash_decode above
@@ -339,8 +343,8 @@ void launch_flash_decode_partial(const void *q, const void *kcache,
                                  void *lse, const void *chunk_len, int batch,
                                  int qh, int kvh, int max_len,
                                  hipStream_t stream) {
-  if (qh / kvh > 8 || qh % kvh)
-    throw std::runtime_error("flash_decode: qh/kvh must divide and be <= 8");
+  if (qh / kvh > 16 || qh % kvh)
+    throw std::runtime_error("flash_decode: qh/kvh must divide and be <= 16");
   float scale = 1.f / sqrtf((float)kD);
   hipLaunchKernelGGL(k_flash_decode_partial, dim3(batch, kvh), dim3(256), 0,
                      stream, (const bf16 *)q, (const bf16 *)kcache,

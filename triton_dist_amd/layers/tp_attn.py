@@ -92,9 +92,7 @@ class TP_Attn:
         """qkv: [b*s, (qh+2kvh)*D] -> attention output [b*s, qh*D]."""
         if (native and not prefill and s == 1 and qkv.is_cuda
                 and self.head_dim == 128 and kv_cache is not None
-                and self.qh % self.kvh == 0 and self.qh // self.kvh <= 8):
-            # (G = qh/kvh > 8 — e.g. seed-oss-36b TP1 with G=10 — falls
-            # through to the torch path until the kernel grows G=16 slots)
+                and self.qh % self.kvh == 0 and self.qh // self.kvh <= 16):
             # fused HIP decode path: qk-norm + RoPE + cache append, then
             # GQA flash-decode (csrc/kernels/{elementwise,attention}.hip)
             from ..ops.fused import flash_decode_op, qkv_prologue_decode_op
