@@ -16,7 +16,10 @@ export HSA_ENABLE_IPC_MODE_LEGACY=${HSA_ENABLE_IPC_MODE_LEGACY:-0}
 # keep kernel args in device memory (fewer launch stalls)
 export HIP_FORCE_DEV_KERNARG=${HIP_FORCE_DEV_KERNARG:-1}
 
-exec python -m torch.distributed.run \
+# Sanitizer/debugger hook (reference parity: launch.sh:162-164 TORCHRUN
+# override): TD_LAUNCH_WRAPPER="rocgdb --batch -ex run --args" or
+# a sanitizer runner wraps every rank's interpreter.
+exec ${TD_LAUNCH_WRAPPER:-} python -m torch.distributed.run \
   --nnodes=1 --nproc-per-node "$NPROC" \
   --master-addr "$MASTER_ADDR" --master-port "$MASTER_PORT" \
   "$@"

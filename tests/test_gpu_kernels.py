@@ -212,6 +212,26 @@ def test_qkv_prologue_and_flash_decode():
     assert_allclose(out.view(b, qh, 1, d), ref, atol=4e-2, rtol=4e-2)
 
 
+@pytest.mark.parametrize("m,n,k", [(8192, 11008, 4096),
+                                   (8192, 4096, 12288)])
+def test_gemm_ci_shapes(m, n, k):
+    """The reference's own CI regression shapes (amd-ci.yml:
+    test_ag_gemm_intra_node 8192 11008 4096 / test_gemm_rs_intra_node
+    8192 4096 12288) as plain-GEMM numerics checks — the fused 2-rank
+    forms run the same consumer kernels with the same tile math (and are
+    covered at smaller shapes by test_gpu_dist; running THESE grids
+    2-ranks-on-1-GPU would oversubscribe the dev box)."""
+    from triton_dist_amd.ops import gemm
+    from triton_dist_amd.utils.testing import assert_allclose, bf16_gemm_tol
+
+    torch.manual_seed(n)
+    a = torch.randn(m, k, device="cuda").to(torch.bfloat16) / 8
+    w = torch.randn(n, k, device="cuda").to(torch.bfloat16) / 8
+    c = gemm(a, w)
+    ref = a.float() @ w.float().t()
+    assert_allclose(c, ref, **bf16_gemm_tol(k))
+
+
 @pytest.mark.parametrize("qh,kvh", [(10, 1), (16, 1), (12, 2)])
 def test_flash_decode_wide_group(qh, kvh):
     """G = qh/kvh in (10, 16, 6): the 2-pass softmax sweep (seed-oss-36b
