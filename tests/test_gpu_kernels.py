@@ -262,7 +262,6 @@ def test_flash_decode_wide_group(qh, kvh):
                                    (256, 5120, 1024), (128, 128, 4096)])
 def test_gemm_splitk(m, n, k):
     from triton_dist_amd.ops import gemm
-    from triton_dist_amd.ops.gemm import choose_splits
     from triton_dist_amd.utils.testing import assert_allclose, bf16_gemm_tol
 
     torch.manual_seed(m + n + k)

@@ -87,3 +87,48 @@ def test_loader_moe():
     with tempfile.TemporaryDirectory() as tmp:
         run_distributed(_worker_moe, world_size=1, args=(tmp,))
         run_distributed(_worker_moe, world_size=2, args=(tmp,))
+
+
+def _worker_index(rank, world, tmpdir):
+    """Sharded-checkpoint form: weight_map index over two files."""
+    import json
+    from pathlib import Path
+
+    import torch
+    import triton_dist_amd as td
+    from triton_dist_amd.models import (AutoLLM, get_config,
+                                        load_hf_weights, save_hf_weights)
+
+    td.init_symm_heap(size_mb=16)
+    cfg = get_config("tiny", tp_mode="torch")
+    a = AutoLLM(cfg, device="cpu")
+    a.init_weights(seed=11)
+    save_hf_weights(a, tmpdir)
+    # split the single file into two + an index
+    from safetensors import safe_open
+    from safetensors.torch import save_file
+    src = Path(tmpdir) / "model.safetensors"
+    with safe_open(str(src), framework="pt") as sf:
+        keys = sorted(sf.keys())
+        half = len(keys) // 2
+        t1 = {k: sf.get_tensor(k) for k in keys[:half]}
+        t2 = {k: sf.get_tensor(k) for k in keys[half:]}
+    src.unlink()
+    save_file(t1, str(Path(tmpdir) / "model-00001-of-00002.safetensors"))
+    save_file(t2, str(Path(tmpdir) / "model-00002-of-00002.safetensors"))
+    wm = {k: "model-00001-of-00002.safetensors" for k in keys[:half]}
+    wm.update({k: "model-00002-of-00002.safetensors" for k in keys[half:]})
+    (Path(tmpdir) / "model.safetensors.index.json").write_text(
+        json.dumps({"weight_map": wm}))
+
+    b = AutoLLM(cfg, device="cpu")
+    load_hf_weights(b, tmpdir)
+    assert torch.equal(a.embed, b.embed)
+    assert torch.equal(a.layers[1]["mlp"].w_down, b.layers[1]["mlp"].w_down)
+    td.shutdown_heap()
+
+
+def test_loader_sharded_index():
+    import tempfile
+    with tempfile.TemporaryDirectory() as tmp:
+        run_distributed(_worker_index, world_size=1, args=(tmp,))

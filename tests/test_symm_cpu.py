@@ -5,8 +5,6 @@ Mirrors the semantics of the reference's canonical primitive test
 (Triton-distributed test/amd/test_distributed-notify-wait.py:36-90 — a
 cross-rank SPSC queue over symmetric memory) on the CPU mock heap.
 """
-import numpy as np
-import pytest
 import torch
 
 from tests.conftest import run_distributed

@@ -2,8 +2,6 @@
 import os
 import time
 
-import torch
-
 
 def test_autotuner_cache(tmp_path, monkeypatch):
     monkeypatch.setenv("TD_AUTOTUNE_DIR", str(tmp_path))

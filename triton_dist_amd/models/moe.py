@@ -11,15 +11,12 @@ resharding between attention and MoE.
 """
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 
 import os
 
 from ..layers.ep_moe_layer import EPMoELayer
 from ..layers.tp_moe_layer import TPMoELayer
-from ..layers.tp_attn import TP_Attn
 from .config import ModelConfig
 from .dense import DenseLLM
 

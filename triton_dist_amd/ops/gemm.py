@@ -9,8 +9,6 @@ from __future__ import annotations
 
 import torch
 
-from ..utils.distributed import has_gpu
-
 
 def _native():
     from .. import _C

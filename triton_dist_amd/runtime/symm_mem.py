@@ -16,7 +16,7 @@ plumbing runs under gloo on CPU-only CI.
 from __future__ import annotations
 
 import os
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import List, Optional
 
 import torch
