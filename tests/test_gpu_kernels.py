@@ -3,6 +3,8 @@
 Every HIP kernel is compared against a plain PyTorch fp32 reference of the
 same op.
 """
+import os
+
 import pytest
 import torch
 
@@ -271,3 +273,22 @@ def test_gemm_splitk(m, n, k):
     c = gemm(a, w)
     ref = a.float() @ w.float().t()
     assert_allclose(c, ref, **bf16_gemm_tol(k))
+
+
+@pytest.mark.skipif(not os.environ.get("TD_EXPERIMENTAL"),
+                    reason="experimental kernel: set TD_EXPERIMENTAL=1")
+@pytest.mark.parametrize("m,n,k", [(256, 256, 128), (512, 768, 512),
+                                   (4096, 4096, 4096)])
+def test_gemm256_v2_experimental(m, n, k):
+    """BK=64 quadrant-phase template (gemm256_v2.hip) — round-2 WIP."""
+    from triton_dist_amd import _C
+    from triton_dist_amd.utils.testing import assert_allclose, bf16_gemm_tol
+
+    torch.manual_seed(k)
+    a = torch.randn(m, k, device="cuda").to(torch.bfloat16) / 8
+    w = torch.randn(n, k, device="cuda").to(torch.bfloat16) / 8
+    c = torch.empty(m, n, device="cuda", dtype=torch.bfloat16)
+    _C.gemm256_v2_bf16(a.data_ptr(), w.data_ptr(), c.data_ptr(), m, n, k,
+                       torch.cuda.current_stream().cuda_stream)
+    torch.cuda.synchronize()
+    assert_allclose(c, a.float() @ w.float().t(), **bf16_gemm_tol(k))

@@ -88,6 +88,10 @@ void launch_rs_reduce_bf16(const void *segments, void *out, int world,
 void launch_gemm256_sk_bf16(const GemmArgs &g, float *ws, int sk,
                             hipStream_t stream);
 
+// EXPERIMENTAL BK=64 quadrant-phase template (gemm256_v2.hip) — not in
+// any dispatch path; see the file header.
+void launch_gemm256_v2_bf16(const GemmArgs &args, hipStream_t stream);
+
 // kernels/gemm_splitk.hip ---------------------------------------------------
 void launch_gemm_splitk_bf16(const GemmArgs &g, float *ws, int splits,
                              hipStream_t stream);
