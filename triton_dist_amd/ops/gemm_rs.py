@@ -64,12 +64,13 @@ def gemm_rs(a: torch.Tensor, w: torch.Tensor, ctx: GemmRSContext,
         ctx.heap.barrier_all()
         for owner in range(world):
             seg = partial[owner * m_per_rank:(owner + 1) * m_per_rank]
-            ctx.scatter.peer(owner)[rank].copy_(seg.to(a.dtype))
+            ctx.scatter.peer(owner)[rank, :m_per_rank].copy_(
+                seg.to(a.dtype))
         ctx.heap.barrier_all()
         acc = torch.zeros(m_per_rank, n, dtype=torch.float32)
         for s in range(world):
             r = (rank + 1 + s) % world
-            acc += ctx.scatter.local()[r].float()
+            acc += ctx.scatter.local()[r, :m_per_rank].float()
         res = acc.to(a.dtype)
         ctx.heap.barrier_all()  # reduce done before anyone's next scatter
         if out is not None:
