@@ -49,5 +49,26 @@ def register_custom_ops():
     def _(q, kv_k, kv_v, offset, qh, kvh):
         return torch.empty_like(q)
 
+    @torch.library.custom_op("td::add_rms_norm", mutates_args=())
+    def add_rms_norm(x: torch.Tensor, resid: torch.Tensor, w: torch.Tensor,
+                     eps: float) -> tuple[torch.Tensor, torch.Tensor]:
+        return fused.add_rms_norm_op(x.contiguous(), resid.contiguous(), w,
+                                     eps)
+
+    @add_rms_norm.register_fake
+    def _(x, resid, w, eps):
+        return torch.empty_like(x), torch.empty_like(x)
+
+    @torch.library.custom_op("td::gdn_decode", mutates_args=("state",))
+    def gdn_decode(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                   g: torch.Tensor, beta: torch.Tensor, scale: float,
+                   state: torch.Tensor) -> torch.Tensor:
+        from .ops.gdn import gdn_decode_step
+        return gdn_decode_step(q, k, v, g, beta, scale, state)
+
+    @gdn_decode.register_fake
+    def _(q, k, v, g, beta, scale, state):
+        return q.new_empty(q.shape[0], q.shape[1], v.shape[-1])
+
     _REGISTERED = True
     return torch.ops.td
