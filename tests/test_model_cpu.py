@@ -63,3 +63,35 @@ def _body_engine_serve(rank, world):
 
 def test_engine_serve_2rank():
     run_distributed(_body_engine_serve, world_size=2)
+
+
+def test_paged_kv_cache():
+    """Paged pool semantics: append across block boundaries, gather
+    equals a dense reference cache."""
+    import torch
+
+    from triton_dist_amd.models.kv_cache import KVCache, PagedKVCache
+
+    L, B, MAX, KVH, D, BLK = 2, 3, 128, 2, 8, 32
+    dense = KVCache(L, B, MAX, KVH, D)
+    paged = PagedKVCache(L, B, MAX, KVH, D, block=BLK)
+    g = torch.Generator().manual_seed(0)
+    pos = 0
+    for step_len in (40, 1, 1, 25):  # crosses block boundaries
+        for layer in range(L):
+            k = torch.randn(B, step_len, KVH, D, generator=g).to(
+                torch.bfloat16)
+            v = torch.randn(B, step_len, KVH, D, generator=g).to(
+                torch.bfloat16)
+            dense.k[layer, :, pos:pos + step_len].copy_(k)
+            dense.v[layer, :, pos:pos + step_len].copy_(v)
+            paged.append(layer, k, v, pos)
+        pos += step_len
+    for layer in range(L):
+        pk, pv = paged.gather_layer(layer, pos)
+        assert torch.equal(pk, dense.k[layer, :, :pos])
+        assert torch.equal(pv, dense.v[layer, :, :pos])
+    # pool economy: only the blocks actually touched were allocated
+    assert paged._free_top == B * ((pos + BLK - 1) // BLK)
+    paged.reset()
+    assert (paged.block_table < 0).all()
