@@ -27,3 +27,14 @@ class FusedEpMoEFunction(torch.autograd.Function):
         raise NotImplementedError(
             "triton_dist_amd is inference-only (reference parity: "
             "function/amd/ep_moe_fused.py backward also raises)")
+
+
+def fused_tp_moe(x_shard: torch.Tensor, topk_ids: torch.Tensor,
+                 topk_w: torch.Tensor, w_gate_up: torch.Tensor,
+                 w_down: torch.Tensor, ag_ctx, coll_ctx) -> torch.Tensor:
+    """TP-MoE functional form: AG -> sorted grouped GEMMs -> topk reduce
+    -> reduce_scatter (see ops/moe_tp.py)."""
+    from .ops.moe_tp import tp_moe_forward
+
+    return tp_moe_forward(x_shard, topk_ids, topk_w, w_gate_up, w_down,
+                          ag_ctx, coll_ctx)
