@@ -1,0 +1,59 @@
+"""CPU tests for the megakernel BUILDER: task emission, scoreboard deps,
+and the opt-in K-split decomposition (TD_MK_KSPLIT)."""
+import torch
+
+from triton_dist_amd.mega.builder import (MegaGraph, T_GEMM_TILE,
+                                          T_GEMM_TILE_PART, T_TILE_REDUCE,
+                                          emit_gemm)
+
+
+def test_emit_gemm_plain():
+    g = MegaGraph()
+    dep = g.new_op()
+    g.next_level()
+    op = emit_gemm(g, 1, 2, 3, batch=64, n=256, k=512, dep=dep)
+    tiles = (64 // 32) * (256 // 128)
+    assert op.n_tasks == tiles
+    assert all(t[0] == T_GEMM_TILE for t in g.tasks)
+    # every task depends on `dep` with its full count
+    for t in g.tasks:
+        assert t[2] == dep.slot
+
+
+def test_emit_gemm_ksplit():
+    g = MegaGraph()
+    op = emit_gemm(g, 1, 2, 3, batch=32, n=256, k=512, dep=None, ksplit=4,
+                   ws_ptr=99)
+    parts = [t for t in g.tasks if t[0] == T_GEMM_TILE_PART]
+    reds = [t for t in g.tasks if t[0] == T_TILE_REDUCE]
+    tiles = 1 * 2
+    assert len(parts) == tiles * 4 and len(reds) == tiles
+    # K ranges tile the full K
+    ranges = sorted((t[6][8], t[6][9]) for t in parts if t[6][7] == 0
+                    and t[6][6] == 0)
+    assert ranges == [(0, 128), (128, 128), (256, 128), (384, 128)]
+    # reduce waits the parts op at FULL count, consumer op is the reduce
+    parts_slot = parts[0][1]
+    for t in reds:
+        assert t[2] == parts_slot and t[3] == len(parts)
+    assert op.slot == reds[0][1]
+    # parts at an earlier level than reduces
+    assert max(t[7] for t in parts) < min(t[7] for t in reds)
+
+
+def test_emit_gemm_ksplit_indivisible_falls_back():
+    g = MegaGraph()
+    op = emit_gemm(g, 1, 2, 3, batch=32, n=128, k=192, dep=None, ksplit=4,
+                   ws_ptr=99)  # 192 % 256 != 0
+    assert all(t[0] == T_GEMM_TILE for t in g.tasks)
+    assert op.n_tasks == 1
+
+
+def test_finalize_shapes():
+    g = MegaGraph()
+    emit_gemm(g, 1, 2, 3, batch=32, n=256, k=512, dep=None, ksplit=2,
+              ws_ptr=9)
+    buf, queue, offs, score = g.finalize(n_wg=4, device="cpu")
+    assert buf.shape[0] == len(g.tasks) and queue.numel() == len(g.tasks)
+    assert offs[-1].item() == len(g.tasks)
+    assert score.numel() == len(g.ops)

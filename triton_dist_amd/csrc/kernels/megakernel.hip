@@ -43,6 +43,10 @@ enum TaskType : int {
   T_EMBED = 6,        // a0=tokens(i64), a1=table, a2=out, a3=rows, a4=cols,
                       // a5=row0
   T_KV_ADVANCE = 7,   // a0=offset cell (i64): += 1
+  T_GEMM_TILE_PART = 8,  // a0=A, a1=B, a2=ws(f32), a3=m, a4=n, a5=k,
+                         // a6=pid_m, a7=pid_n, a8=k0, a9=klen, a10=sk
+  T_TILE_REDUCE = 9,     // a0=ws, a1=C, a2=m, a3=n, a4=pid_m, a5=pid_n,
+                         // a6=ksplit
 };
 
 struct Task {
@@ -201,6 +205,121 @@ TD_DEV void t_gemm_tile(const Task &t, bf16 *lds) {
           C[((size_t)pid_m * BM + row) * n + (size_t)pid_n * BN + col] =
               (bf16)acc[i][j][r];
       }
+}
+
+// K-split GEMM: one K-range partial of a 32x128 tile -> fp32 ws slice
+// [sk][m_pad][n]. Same pipelined structure as t_gemm_tile but over
+// [k0, k0+klen). Opt-in (TD_MK_KSPLIT): shortens the megakernel's per-op
+// critical path from a full-K tile to K/ksplit.
+TD_DEV void t_gemm_tile_part(const Task &t, bf16 *lds) {
+  constexpr int BM = 32, BN = 128, BK = 64;
+  constexpr int ABUF = BM * BK, BBUF = BN * BK;
+  bf16 *lds_a = lds;
+  bf16 *lds_b = lds + 3 * ABUF;
+  const bf16 *A = (const bf16 *)t.a[0];
+  const bf16 *B = (const bf16 *)t.a[1];
+  float *ws = (float *)t.a[2];
+  const int m = (int)t.a[3], n = (int)t.a[4], k = (int)t.a[5];
+  const int pid_m = (int)t.a[6], pid_n = (int)t.a[7];
+  const int k0 = (int)t.a[8], klen = (int)t.a[9], sk = (int)t.a[10];
+  const int m_pad = (m + BM - 1) / BM * BM;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  f32x4 acc[2][2] = {};
+  const bf16 *ga = A + (size_t)pid_m * BM * k + k0;
+  const bf16 *gb = B + (size_t)pid_n * BN * k + k0;
+  const int ksteps = klen / BK;
+  auto stage = [&](int ti, int buf) {
+    const int kk0 = ti * BK;
+    {
+      int row = tid >> 3, kc = tid & 7;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int *)(
+              ga + (size_t)row * k + kk0 + kc * 8),
+          (__attribute__((address_space(3))) unsigned int *)(
+              lds_a + buf * ABUF + (wave * 64) * 8),
+          16, 0, 0);
+    }
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      int qb = it * 256 + tid;
+      int rowb = qb >> 3, kcb = qb & 7;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int *)(
+              gb + (size_t)rowb * k + kk0 + kcb * 8),
+          (__attribute__((address_space(3))) unsigned int *)(
+              lds_b + buf * BBUF + (it * 256 + wave * 64) * 8),
+          16, 0, 0);
+    }
+  };
+  stage(0, 0);
+  if (ksteps > 1) stage(1, 1);
+  for (int ti = 0; ti < ksteps; ++ti) {
+    const int buf = ti % 3;
+    if (ti + 1 < ksteps) {
+      asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+    __builtin_amdgcn_sched_barrier(0);
+    if (ti + 2 < ksteps) stage(ti + 2, (ti + 2) % 3);
+#pragma unroll
+    for (int ks = 0; ks < BK / 32; ++ks) {
+      bf16x8 af[2], bfr[2];
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        int arow = i * 16 + (lane & 15);
+        int brow = wave * 32 + i * 16 + (lane & 15);
+        int kk = ks * 32 + (lane >> 4) * 8;
+        af[i] = *(const bf16x8 *)(lds_a + buf * ABUF + arow * BK + kk);
+        bfr[i] = *(const bf16x8 *)(lds_b + buf * BBUF + brow * BK + kk);
+      }
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[i], bfr[j], acc[i][j], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_barrier();
+    __builtin_amdgcn_sched_barrier(0);
+  }
+  const int row_lim = m - pid_m * BM;
+  float *wsl = ws + (size_t)sk * m_pad * n;
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = i * 16 + (lane >> 4) * 4 + r;
+        int col = wave * 32 + j * 16 + (lane & 15);
+        if (row < row_lim)
+          wsl[((size_t)pid_m * BM + row) * n + (size_t)pid_n * BN + col] =
+              acc[i][j][r];
+      }
+}
+
+// Sum the ksplit fp32 slices of one 32x128 tile -> bf16 C.
+TD_DEV void t_tile_reduce(const Task &t) {
+  constexpr int BM = 32, BN = 128;
+  const float *ws = (const float *)t.a[0];
+  bf16 *C = (bf16 *)t.a[1];
+  const int m = (int)t.a[2], n = (int)t.a[3];
+  const int pid_m = (int)t.a[4], pid_n = (int)t.a[5];
+  const int ksplit = (int)t.a[6];
+  const int m_pad = (m + BM - 1) / BM * BM;
+  const int row_lim = m - pid_m * BM;
+  for (int i = threadIdx.x; i < BM * BN; i += blockDim.x) {
+    int row = i / BN, col = i % BN;
+    if (row >= row_lim) continue;
+    size_t off = ((size_t)pid_m * BM + row) * n + (size_t)pid_n * BN + col;
+    float acc = 0.f;
+    for (int sk = 0; sk < ksplit; ++sk)
+      acc += ws[(size_t)sk * m_pad * n + off];
+    C[off] = (bf16)acc;
+  }
 }
 
 TD_DEV void t_swiglu(const Task &t) {
@@ -444,6 +563,12 @@ __global__ __launch_bounds__(mk::NTH) void k_megakernel(
         break;
       case mk::T_EMBED:
         mk::t_embed(t);
+        break;
+      case mk::T_GEMM_TILE_PART:
+        mk::t_gemm_tile_part(t, (bf16 *)lds);
+        break;
+      case mk::T_TILE_REDUCE:
+        mk::t_tile_reduce(t);
         break;
       case mk::T_KV_ADVANCE:
         if (threadIdx.x == 0) *(long *)t.a[0] += 1;

@@ -26,6 +26,8 @@ T_QKV_PROLOGUE = 4
 T_FLASH_DECODE = 5
 T_EMBED = 6
 T_KV_ADVANCE = 7
+T_GEMM_TILE_PART = 8   # K-range partial -> fp32 ws slice
+T_TILE_REDUCE = 9      # sum ws slices -> bf16 C tile
 
 TASK_INT64S = 13
 TASK_WORDS = 2 + 1 + TASK_INT64S  # 6 int32 = 3 int64 words (with pad)
@@ -90,6 +92,48 @@ class MegaGraph:
                 torch.tensor(flat, dtype=torch.int32, device=device),
                 torch.tensor(offs, dtype=torch.int32, device=device),
                 torch.zeros(len(self.ops), dtype=torch.int32, device=device))
+
+
+def emit_gemm(g: "MegaGraph", a_ptr: int, w_ptr: int, c_ptr: int,
+              batch: int, n: int, k: int, dep, ksplit: int = 1,
+              ws_ptr: int = 0):
+    """Emit one logical GEMM as scoreboard ops. ksplit == 1: one
+    T_GEMM_TILE per 32x128 C tile (the original scheme). ksplit > 1
+    (TD_MK_KSPLIT, opt-in): each tile becomes `ksplit` K-range partial
+    tasks writing fp32 ws slices plus one reduce task — shortens the
+    per-op critical path from a full-K tile to K/ksplit (the 59 ms vs
+    15 ms megakernel gap is op-chain latency; docs/ROADMAP.md #4).
+    Returns the op consumers must depend on."""
+    tiles_m = (batch + 31) // 32
+    tiles_n = n // 128
+    if ksplit <= 1 or k % (64 * ksplit):
+        op = g.new_op()
+        for pm in range(tiles_m):
+            for pn in range(tiles_n):
+                g.add_task(T_GEMM_TILE, op,
+                           [a_ptr, w_ptr, c_ptr, batch, n, k, pm, pn],
+                           [(dep, 0)] if dep else [])
+        g.next_level()
+        return op
+    assert ws_ptr, "ksplit > 1 needs an fp32 ws [ksplit, batch_pad, n]"
+    klen = k // ksplit
+    parts = g.new_op()
+    for pm in range(tiles_m):
+        for pn in range(tiles_n):
+            for sk in range(ksplit):
+                g.add_task(T_GEMM_TILE_PART, parts,
+                           [a_ptr, w_ptr, ws_ptr, batch, n, k, pm, pn,
+                            sk * klen, klen, sk],
+                           [(dep, 0)] if dep else [])
+    g.next_level()
+    red = g.new_op()
+    for pm in range(tiles_m):
+        for pn in range(tiles_n):
+            g.add_task(T_TILE_REDUCE, red,
+                       [ws_ptr, c_ptr, batch, n, pm, pn, ksplit],
+                       [(parts, 0)])
+    g.next_level()
+    return red
 
 
 class MegaRun:

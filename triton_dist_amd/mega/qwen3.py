@@ -11,7 +11,7 @@ import torch
 
 from ..models.dense import DenseLLM
 from ..models.kv_cache import KVCache
-from .builder import (MegaGraph, MegaRun, T_ADD_RMSNORM, T_EMBED,
+from .builder import (emit_gemm, MegaGraph, MegaRun, T_ADD_RMSNORM, T_EMBED,
                       T_FLASH_DECODE, T_GEMM_TILE, T_KV_ADVANCE,
                       T_QKV_PROLOGUE, T_RMSNORM, T_SWIGLU)
 
@@ -55,16 +55,22 @@ class MegaQwen3Decode:
         g = MegaGraph()
         tiles_m = bp // 32
 
+        import os
+        ksplit = int(os.environ.get("TD_MK_KSPLIT", "1"))
+        self._ws = {}
+
         def gemm(a_buf, w, c_buf, n, k, dep):
-            op = g.new_op()
-            for pm in range(tiles_m):
-                for pn in range(n // 128):
-                    g.add_task(T_GEMM_TILE, op,
-                               [a_buf.data_ptr(), w.data_ptr(),
-                                c_buf.data_ptr(), batch, n, k, pm, pn],
-                               [(dep, 0)] if dep else [])
-            g.next_level()
-            return op
+            ws_ptr = 0
+            if ksplit > 1 and k % (64 * ksplit) == 0:
+                key = (c_buf.data_ptr(), n)
+                if key not in self._ws:
+                    self._ws[key] = torch.zeros(ksplit, bp, n,
+                                                dtype=torch.float32,
+                                                device=dev)
+                ws_ptr = self._ws[key].data_ptr()
+            return emit_gemm(g, a_buf.data_ptr(), w.data_ptr(),
+                             c_buf.data_ptr(), batch, n, k, dep,
+                             ksplit=ksplit, ws_ptr=ws_ptr)
 
         # embed
         emb = g.new_op()
