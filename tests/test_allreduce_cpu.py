@@ -51,3 +51,7 @@ def _body_model_gemm_ar(rank, world):
 
 def test_model_gemm_ar_cpu_2rank():
     run_distributed(_body_model_gemm_ar, world_size=2)
+
+
+def test_allreduce_cpu_4rank():
+    run_distributed(_body_ar, world_size=4)

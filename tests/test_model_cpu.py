@@ -6,11 +6,11 @@ import torch
 from tests.conftest import run_distributed
 
 
-def _body_model_modes(rank, world):
+def _body_model_modes(rank, world, model_name="tiny"):
     from triton_dist_amd.models import DenseLLM, Engine, KVCache, get_config
     from triton_dist_amd.utils import assert_allclose
 
-    cfg = get_config("tiny", tp_mode="ag_rs", max_length=64)
+    cfg = get_config(model_name, tp_mode="ag_rs", max_length=64)
     model = DenseLLM(cfg, device="cpu")
     model.init_weights(seed=3)
     b, s = world * 2, 4
@@ -63,6 +63,14 @@ def _body_engine_serve(rank, world):
 
 def test_engine_serve_2rank():
     run_distributed(_body_engine_serve, world_size=2)
+
+
+def _body_model_modes4(rank, world):
+    _body_model_modes(rank, world, model_name="tiny4")
+
+
+def test_model_modes_cpu_4rank():
+    run_distributed(_body_model_modes4, world_size=4)
 
 
 def test_paged_kv_cache():

@@ -4,11 +4,11 @@ import torch
 from tests.conftest import run_distributed
 
 
-def _body_moe_model(rank, world):
+def _body_moe_model(rank, world, model_name="tiny-moe"):
     from triton_dist_amd.models import KVCache, Qwen3MoE, get_config
     from triton_dist_amd.utils import assert_allclose
 
-    cfg = get_config("tiny-moe", tp_mode="ag_rs", max_length=64)
+    cfg = get_config(model_name, tp_mode="ag_rs", max_length=64)
     model = Qwen3MoE(cfg, device="cpu")
     model.init_weights(seed=3)
     b, s = world * 2, 4
@@ -31,3 +31,11 @@ def _body_moe_model(rank, world):
 
 def test_moe_model_cpu_2rank():
     run_distributed(_body_moe_model, world_size=2)
+
+
+def _body_moe_model4(rank, world):
+    _body_moe_model(rank, world, model_name="tiny-moe4")
+
+
+def test_moe_model_cpu_4rank():
+    run_distributed(_body_moe_model4, world_size=4)

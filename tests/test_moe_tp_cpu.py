@@ -86,3 +86,7 @@ def _worker_model(rank, world):
 
 def test_tp_moe_model_wiring_2rank():
     run_distributed(_worker_model, world_size=2)
+
+
+def test_tp_moe_4rank():
+    run_distributed(_worker, world_size=4)

@@ -54,6 +54,12 @@ PRESETS = {
                           n_heads=32, n_kv_heads=4, head_dim=128,
                           vocab=151936, n_experts=128, moe_topk=8,
                           moe_inter=768),
+    # tiny configs divisible at TP=4 (4-rank gloo tests)
+    "tiny4": dict(hidden=256, intermediate=512, n_layers=2, n_heads=8,
+                  n_kv_heads=4, head_dim=64, vocab=512),
+    "tiny-moe4": dict(hidden=64, intermediate=128, n_layers=2, n_heads=4,
+                      n_kv_heads=4, head_dim=32, vocab=256, n_experts=4,
+                      moe_topk=2, moe_inter=32),
     # tiny MoE for CPU tests
     "tiny-moe": dict(hidden=64, intermediate=128, n_layers=2, n_heads=2,
                      n_kv_heads=2, head_dim=32, vocab=256, n_experts=4,
