@@ -62,3 +62,28 @@ def test_gemm_cpu_fallback():
     a = rand_tensor((64, 32), dtype=torch.bfloat16)
     w = rand_tensor((48, 32), dtype=torch.bfloat16)
     assert_allclose(gemm(a, w), gemm_ref(a, w), atol=5e-2, rtol=5e-2)
+
+
+def _body_allgather(rank, world):
+    from triton_dist_amd.ops import create_ag_gemm_context
+    from triton_dist_amd.ops.allgather_gemm import allgather
+    from triton_dist_amd.utils import rand_tensor
+
+    m, k = 32, 16
+    ctx = create_ag_gemm_context(max_m_per_rank=m, k=k, chunks_per_rank=4)
+    for mm in (m, m // 2):  # full + variable m
+        g = torch.Generator().manual_seed(3 + rank)
+        a = rand_tensor((mm, k), dtype=torch.bfloat16, generator=g)
+        out = allgather(a, ctx)
+        for r in range(world):
+            gr = torch.Generator().manual_seed(3 + r)
+            exp = rand_tensor((mm, k), dtype=torch.bfloat16, generator=gr)
+            assert torch.equal(out[r * mm:(r + 1) * mm], exp), (rank, r, mm)
+
+
+def test_allgather_cpu_2rank():
+    run_distributed(_body_allgather, world_size=2)
+
+
+def test_allgather_cpu_4rank():
+    run_distributed(_body_allgather, world_size=4)
