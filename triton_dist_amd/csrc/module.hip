@@ -394,6 +394,23 @@ static void moe_router(uintptr_t logits, uintptr_t ids, uintptr_t tw,
   TD_CHECK_HIP(hipGetLastError());
 }
 
+static pybind11::dict p2p_attributes(int dev, int peer) {
+  // Device-query parity: the reference checks P2P native-atomic support
+  // (utils.py:539-567) before picking cross-GPU atomic protocols.
+  pybind11::dict d;
+  int v = 0;
+  TD_CHECK_HIP(hipDeviceGetP2PAttribute(&v, hipDevP2PAttrAccessSupported,
+                                        dev, peer));
+  d["access"] = v;
+  TD_CHECK_HIP(hipDeviceGetP2PAttribute(
+      &v, hipDevP2PAttrNativeAtomicSupported, dev, peer));
+  d["native_atomics"] = v;
+  TD_CHECK_HIP(hipDeviceGetP2PAttribute(
+      &v, hipDevP2PAttrPerformanceRank, dev, peer));
+  d["performance_rank"] = v;
+  return d;
+}
+
 static void gemm256_v2_bf16(uintptr_t a, uintptr_t b, uintptr_t c, int mm,
                             int n, int k, uintptr_t stream) {
   GemmArgs args{(void *)a, (void *)b, (void *)c, nullptr,
@@ -724,6 +741,7 @@ PYBIND11_MODULE(_C, m) {
   m.def("gemm_splitk_bf16", &gemm_splitk_bf16);
   m.def("gemm256_sk_bf16", &gemm256_sk_bf16);
   m.def("gemm256_v2_bf16", &gemm256_v2_bf16);
+  m.def("p2p_attributes", &p2p_attributes);
   m.def("moe_router", &moe_router);
   m.def("reduce_scatter", &reduce_scatter_op);
   m.def("gdn_decode", &gdn_decode);

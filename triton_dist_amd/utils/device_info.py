@@ -73,3 +73,15 @@ def wait_for_stable_clock(device: int = 0, target_ratio: float = 0.95,
         mhz = current_clock_mhz(device)
         if mhz is None or mhz >= target_ratio * 2400:
             return
+
+
+def p2p_attributes(dev: int, peer: int) -> dict:
+    """P2P access / native-atomic / performance-rank attributes between two
+    devices (reference parity: utils.py:539-567 P2P native atomic check —
+    our one-shot AR and EP arrive protocols assume native atomics over
+    xGMI, which this verifies on real multi-GPU nodes)."""
+    from .. import _C
+
+    if _C is None or dev == peer:
+        return {"access": 1, "native_atomics": 1, "performance_rank": 0}
+    return dict(_C.p2p_attributes(dev, peer))
