@@ -292,3 +292,28 @@ def test_gemm256_v2_experimental(m, n, k):
                        torch.cuda.current_stream().cuda_stream)
     torch.cuda.synchronize()
     assert_allclose(c, a.float() @ w.float().t(), **bf16_gemm_tol(k))
+
+
+@pytest.mark.skipif(not os.environ.get("TD_EXPERIMENTAL"),
+                    reason="experimental kernel: set TD_EXPERIMENTAL=1")
+def test_flash_decode_paged_experimental():
+    """Paged pool + block table vs the contiguous kernel."""
+    from triton_dist_amd.models.kv_cache import PagedKVCache
+    from triton_dist_amd.ops.fused import (flash_decode_op,
+                                           flash_decode_paged_op)
+    from triton_dist_amd.utils.testing import assert_allclose
+
+    torch.manual_seed(9)
+    b, qh, kvh, d, maxlen, seq = 8, 16, 2, 128, 256, 77
+    paged = PagedKVCache(1, b, maxlen, kvh, d, block=64, device="cuda")
+    kc = (torch.randn(b, maxlen, kvh, d, device="cuda") / 4).to(
+        torch.bfloat16)
+    vc = (torch.randn(b, maxlen, kvh, d, device="cuda") / 4).to(
+        torch.bfloat16)
+    paged.append(0, kc[:, :seq], vc[:, :seq], 0)
+    q = (torch.randn(b, qh * d, device="cuda") / 4).to(torch.bfloat16)
+    offset = torch.tensor(seq - 1, dtype=torch.int64, device="cuda")
+    out_p = flash_decode_paged_op(q, paged, 0, offset, qh, kvh)
+    out_c = flash_decode_op(q, kc, vc, offset, qh, kvh)
+    torch.cuda.synchronize()
+    assert_allclose(out_p, out_c, atol=1e-3, rtol=1e-3)

@@ -216,6 +216,11 @@ void launch_flash_decode(const void *q, const void *kcache,
                          const void *vcache, void *out, const void *offset,
                          int batch, int qh, int kvh, int max_len,
                          hipStream_t stream);
+void launch_flash_decode_paged(const void *q, const void *k_pool,
+                               const void *v_pool, const void *block_table,
+                               int max_blocks, int block, void *out,
+                               const void *offset, int batch, int qh,
+                               int kvh, hipStream_t stream);
 void launch_flash_decode_partial(const void *q, const void *kcache,
                                  const void *vcache, void *out_part,
                                  void *lse, const void *chunk_len, int batch,

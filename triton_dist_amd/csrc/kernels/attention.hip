@@ -199,6 +199,174 @@ void launch_flash_decode(const void *q, const void *kcache,
 }
 
 // ---------------------------------------------------------------------------
+// EXPERIMENTAL paged flash-decode: identical math to k_flash_decode but K/V
+// come from a block pool [n_blocks, block, kvh, D] via a per-sequence block
+// table [B, max_blocks] (models/kv_cache.py PagedKVCache layout). Gated
+// behind TD_EXPERIMENTAL in tests; round 2 validates + wires an engine
+// path. Block size is a power of two (shift passed in).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_flash_decode_paged(
+    const bf16 *__restrict__ q, const bf16 *__restrict__ k_pool,
+    const bf16 *__restrict__ v_pool, const long *__restrict__ block_table,
+    int max_blocks, int blk_shift, bf16 *__restrict__ out,
+    const long *__restrict__ offset, int qh, int kvh, float scale) {
+  const int b = blockIdx.x;
+  const int kh = blockIdx.y;
+  const int G = qh / kvh;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int g = tid >> 5;
+  const int t = tid & 31;
+  const long seqlen = *offset + 1;
+  const long *bt = block_table + (size_t)b * max_blocks;
+  const int blk_mask = (1 << blk_shift) - 1;
+
+  __shared__ bf16 k_lds[kTile][kD];
+  __shared__ bf16 v_lds[kTile][kD + 8];
+  __shared__ bf16 q_lds[16][kD];
+  __shared__ bf16 p_bf[16][kTile + 8];
+  __shared__ float s_part[4][16][kTile];
+  __shared__ float m_lds[16], r_lds[16], l_lds[16];
+
+  for (int i = tid; i < 16 * kD / 8; i += 256) {
+    int hh = i / (kD / 8);
+    int c = (i % (kD / 8)) * 8;
+    bf16x8 v{};
+    if (hh < G)
+      v = *(const bf16x8 *)(q + (((size_t)b * qh) + kh * G + hh) * kD + c);
+    *(bf16x8 *)(&q_lds[hh][c]) = v;
+  }
+  if (tid < 16) {
+    m_lds[tid] = -1e30f;
+    l_lds[tid] = 0.f;
+    r_lds[tid] = 1.f;
+    for (int tt = 0; tt < kTile; ++tt) p_bf[tid][tt] = (bf16)0.f;
+  }
+  __syncthreads();
+  bf16x8 qfrag = *(const bf16x8 *)(
+      &q_lds[lane & 15][wave * 32 + (lane >> 4) * 8]);
+
+  f32x4 accPV[2] = {};
+  const long ntiles = (seqlen + kTile - 1) / kTile;
+  for (long tile = 0; tile < ntiles; ++tile) {
+    const long pos0 = tile * kTile;
+    __syncthreads();
+    for (int i = tid; i < kTile * kD / 8; i += 256) {
+      int r = i / (kD / 8);
+      int c = (i % (kD / 8)) * 8;
+      long pos = pos0 + r;
+      bf16x8 kv{}, vv{};
+      if (pos < seqlen) {
+        // pool addressing: block id from the table, row = pos % block
+        long pb = bt[pos >> blk_shift];
+        size_t base =
+            (((size_t)pb << blk_shift | (pos & blk_mask)) * kvh + kh) * kD +
+            c;
+        kv = *(const bf16x8 *)(k_pool + base);
+        vv = *(const bf16x8 *)(v_pool + base);
+      }
+      *(bf16x8 *)(&k_lds[r][c]) = kv;
+      *(bf16x8 *)(&v_lds[r][c]) = vv;
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      bf16x8 kfrag = *(const bf16x8 *)(
+          &k_lds[(lane & 15) + 16 * h][wave * 32 + (lane >> 4) * 8]);
+      f32x4 c4 = {0.f, 0.f, 0.f, 0.f};
+      c4 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag, kfrag, c4, 0, 0,
+                                                   0);
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        s_part[wave][(lane >> 4) * 4 + r][(lane & 15) + 16 * h] = c4[r];
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int gg = g; gg < 16; gg += 8) {
+      float s = -1e30f;
+      if (gg < G && pos0 + t < seqlen) {
+        s = (s_part[0][gg][t] + s_part[1][gg][t] + s_part[2][gg][t] +
+             s_part[3][gg][t]) * scale;
+      }
+      float mx = s;
+      for (int off = 16; off > 0; off >>= 1)
+        mx = fmaxf(mx, __shfl_xor(mx, off));
+      float m_old = m_lds[gg];
+      float m_new = fmaxf(m_old, mx);
+      float p = (s > -1e29f) ? __expf(s - m_new) : 0.f;
+      p_bf[gg][t] = (bf16)p;
+      float psum = p;
+      for (int off = 16; off > 0; off >>= 1) psum += __shfl_xor(psum, off);
+      if (t == 0) {
+        float r = __expf(m_old - m_new);
+        r_lds[gg] = r;
+        l_lds[gg] = l_lds[gg] * r + psum;
+        m_lds[gg] = m_new;
+      }
+    }
+    __syncthreads();
+
+    bf16x8 afrag = *(const bf16x8 *)(&p_bf[lane & 15][(lane >> 4) * 8]);
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      const int cg = wave * 2 + h;
+      const int k0 = (lane >> 4) * 8;
+      unsigned addr = (unsigned)(uintptr_t)(
+          &v_lds[k0 + ((lane & 15) >> 2)][cg * 16 + 4 * (lane & 3)]);
+      unsigned long long lo, hi;
+      asm volatile(
+          "ds_read_b64_tr_b16 %0, %2\n"
+          "ds_read_b64_tr_b16 %1, %2 offset:%3\n"
+          "s_waitcnt lgkmcnt(0)"
+          : "=v"(lo), "=v"(hi)
+          : "v"(addr), "i"(4 * (kD + 8) * 2));
+      bf16x8 bfrag;
+      *(unsigned long long *)&bfrag = lo;
+      *((unsigned long long *)&bfrag + 1) = hi;
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        accPV[h][r] *= r_lds[(lane >> 4) * 4 + r];
+      accPV[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag,
+                                                         accPV[h], 0, 0, 0);
+    }
+  }
+  __syncthreads();
+#pragma unroll
+  for (int h = 0; h < 2; ++h)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = (lane >> 4) * 4 + r;
+      if (row < G) {
+        float inv_l = 1.f / l_lds[row];
+        out[(((size_t)b * qh) + kh * G + row) * kD + wave * 32 + h * 16 +
+            (lane & 15)] = (bf16)(accPV[h][r] * inv_l);
+      }
+    }
+}
+
+void launch_flash_decode_paged(const void *q, const void *k_pool,
+                               const void *v_pool, const void *block_table,
+                               int max_blocks, int block, void *out,
+                               const void *offset, int batch, int qh,
+                               int kvh, hipStream_t stream) {
+  if (qh / kvh > 16 || qh % kvh)
+    throw std::runtime_error("flash_decode: qh/kvh must divide and be <= 16");
+  if (block & (block - 1))
+    throw std::runtime_error("flash_decode_paged: block must be a power of 2");
+  int shift = 0;
+  while ((1 << shift) < block) ++shift;
+  float scale = 1.f / sqrtf((float)kD);
+  hipLaunchKernelGGL(k_flash_decode_paged, dim3(batch, kvh), dim3(256), 0,
+                     stream, (const bf16 *)q, (const bf16 *)k_pool,
+                     (const bf16 *)v_pool, (const long *)block_table,
+                     max_blocks, shift, (bf16 *)out, (const long *)offset,
+                     qh, kvh, scale);
+}
+
+// ---------------------------------------------------------------------------
 // SP split-KV decode: same kernel body, but the KV chunk length comes from
 // `chunk_len` (device int64: number of valid positions in THIS rank's KV
 // shard) and the outputs are the UNNORMALIZED partial accumulator plus the

@@ -346,6 +346,17 @@ static void gemm_splitk_bf16(uintptr_t a, uintptr_t b, uintptr_t c,
   TD_CHECK_HIP(hipGetLastError());
 }
 
+static void flash_decode_paged(uintptr_t q, uintptr_t kp, uintptr_t vp,
+                               uintptr_t bt, int max_blocks, int block,
+                               uintptr_t out, uintptr_t off, int batch,
+                               int qh, int kvh, uintptr_t stream) {
+  launch_flash_decode_paged((const void *)q, (const void *)kp,
+                            (const void *)vp, (const void *)bt, max_blocks,
+                            block, (void *)out, (const void *)off, batch,
+                            qh, kvh, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
 static void gdn_decode(uintptr_t q, uintptr_t k, uintptr_t v, uintptr_t g,
                        uintptr_t beta, uintptr_t state, uintptr_t o, int B,
                        int H, int K, int V, float scale, uintptr_t stream) {
@@ -745,6 +756,7 @@ PYBIND11_MODULE(_C, m) {
   m.def("moe_router", &moe_router);
   m.def("reduce_scatter", &reduce_scatter_op);
   m.def("gdn_decode", &gdn_decode);
+  m.def("flash_decode_paged", &flash_decode_paged);
   m.def("ll_allgather", &ll_allgather_op);
   m.def("all_to_all", &all_to_all_op);
   m.def("ag_gemm_consumer_splitk_bf16", &ag_gemm_consumer_splitk_bf16);

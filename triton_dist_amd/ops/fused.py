@@ -89,3 +89,31 @@ def flash_decode_op(q: torch.Tensor, kv_k: torch.Tensor, kv_v: torch.Tensor,
                       out.data_ptr(), offset.data_ptr(), b, qh, kvh,
                       max_len, _s())
     return out
+
+
+def flash_decode_paged_op(q: torch.Tensor, paged_cache, layer: int,
+                          offset: torch.Tensor, qh: int,
+                          kvh: int) -> torch.Tensor:
+    """GQA flash-decode over a PagedKVCache (models/kv_cache.py): the
+    kernel indexes the block pool through the per-sequence table
+    (EXPERIMENTAL — csrc/kernels/attention.hip k_flash_decode_paged). On
+    CPU, gathers to contiguous and reuses the torch reference."""
+    b = q.shape[0]
+    if not q.is_cuda:
+        upto = int(offset.item()) + 1
+        kc, vc = paged_cache.gather_layer(layer, upto)
+        pad = torch.zeros(b, 1, kc.shape[2], kc.shape[3], dtype=kc.dtype)
+        kfull = torch.cat([kc, pad], 1)  # shape cover for offset+1 reads
+        return flash_decode_op(q, kfull, torch.cat([vc, pad], 1), offset,
+                               qh, kvh)
+    from .. import _C
+    out = torch.empty_like(q)
+    _C.flash_decode_paged(
+        q.contiguous().data_ptr(),
+        paged_cache.k_pool[layer].data_ptr(),
+        paged_cache.v_pool[layer].data_ptr(),
+        paged_cache.block_table.data_ptr(),
+        paged_cache.blocks_per_seq, paged_cache.block,
+        out.data_ptr(), offset.data_ptr(), b, qh, kvh,
+        torch.cuda.current_stream().cuda_stream)
+    return out
