@@ -103,3 +103,28 @@ def test_paged_kv_cache():
     assert paged._free_top == B * ((pos + BLK - 1) // BLK)
     paged.reset()
     assert (paged.block_table < 0).all()
+
+
+def test_flash_decode_paged_cpu_bridge():
+    """The paged op's CPU path (gather bridge + torch flash reference)
+    agrees with the contiguous op."""
+    import torch
+
+    from triton_dist_amd.models.kv_cache import PagedKVCache
+    from triton_dist_amd.ops.fused import flash_decode_paged_op
+
+    torch.manual_seed(4)
+    b, qh, kvh, d, maxlen, seq = 2, 4, 2, 32, 128, 37
+    paged = PagedKVCache(1, b, maxlen, kvh, d, block=32)
+    kc = (torch.randn(b, maxlen, kvh, d) / 4).to(torch.bfloat16)
+    vc = (torch.randn(b, maxlen, kvh, d) / 4).to(torch.bfloat16)
+    paged.append(0, kc[:, :seq], vc[:, :seq], 0)
+    q = (torch.randn(b, qh * d) / 4).to(torch.bfloat16)
+    offset = torch.tensor(seq - 1, dtype=torch.int64)
+    out_p = flash_decode_paged_op(q, paged, 0, offset, qh, kvh)
+    import torch.nn.functional as F
+    qs = q.view(b, qh, 1, d).float()
+    ks = kc[:, :seq].transpose(1, 2).float()
+    vs = vc[:, :seq].transpose(1, 2).float()
+    ref = F.scaled_dot_product_attention(qs, ks, vs, enable_gqa=True)
+    assert (out_p.float() - ref.view(b, qh * d)).abs().max() < 2e-2

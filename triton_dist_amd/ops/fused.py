@@ -100,12 +100,16 @@ def flash_decode_paged_op(q: torch.Tensor, paged_cache, layer: int,
     CPU, gathers to contiguous and reuses the torch reference."""
     b = q.shape[0]
     if not q.is_cuda:
+        import torch.nn.functional as F
+
         upto = int(offset.item()) + 1
         kc, vc = paged_cache.gather_layer(layer, upto)
-        pad = torch.zeros(b, 1, kc.shape[2], kc.shape[3], dtype=kc.dtype)
-        kfull = torch.cat([kc, pad], 1)  # shape cover for offset+1 reads
-        return flash_decode_op(q, kfull, torch.cat([vc, pad], 1), offset,
-                               qh, kvh)
+        d = kc.shape[3]
+        qs = q.view(b, qh, 1, d).float()
+        ks = kc.transpose(1, 2).float()
+        vs = vc.transpose(1, 2).float()
+        ref = F.scaled_dot_product_attention(qs, ks, vs, enable_gqa=True)
+        return ref.view(b, qh * d).to(q.dtype)
     from .. import _C
     out = torch.empty_like(q)
     _C.flash_decode_paged(
