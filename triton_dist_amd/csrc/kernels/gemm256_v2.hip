@@ -10,13 +10,15 @@
 // Ledger (sound variant; the guide's vmcnt(6)-at-phase-0 could not be
 // proven correct for the halves consumed in that same phase):
 //   pair P processes K-tiles (2P, 2P+1) from bufs (0, 1);
-//   phases 4-7 stage tile 2P+2 -> buf0 (freed after phase 3),
-//   phases 0-3 stage tile 2P+3 -> buf1 (freed after prev phase 7);
-//   phase-0 entry: vmcnt(0)  (tile 2P+2's 4 halves must have landed),
-//   phase-4 entry: vmcnt(8)  (bound the queue; nothing new needed).
-// One full drain per 128 MFMA/wave — half the frequency of a per-tile
-// drain; relaxing it further needs 3 buffers (LDS) or the wave-aligned
-// staging experiment (see ROADMAP).
+//   phases 0-3 stage tile 2P+1 -> buf1 (freed at the END of pair P-1;
+//     consumed THIS pair at phases 4-7),
+//   phases 4-7 stage tile 2P+2 -> buf0 (freed after phase 3; consumed
+//     at pair P+1 phases 0-3);
+//   with only 2 buffers the stage-to-consume gap is 4 phases on both
+//   streams, so BOTH half-pair boundaries drain: vmcnt(0) at phase 0
+//   and phase 4. That is one drain per 64 MFMA/wave — the knob round 2
+//   iterates on (3 narrower buffers, or wave-aligned staging, to get a
+//   counted non-zero wait; see ROADMAP).
 //
 // NOT wired into any dispatch path: reachable only via
 // _C.gemm256_v2_bf16 (scripts/bench_gemm_v2.py; tests env-gated by
@@ -109,13 +111,11 @@ __global__ __launch_bounds__(NTH, 2) void k_gemm256_v2(GemmArgs args) {
 
   f32x4 acc[8][4] = {};
 
-  // prologue: tiles 0 (buf0) and 1 (buf1)
+  // prologue: tile 0 only (tile 1 is staged during pair-0 phases 0-3)
 #pragma unroll
   for (int h = 0; h < 2; ++h) {
     stage_half(ga, args.lda, 0, lds_a, 0, h);
     stage_half(gb, args.ldb, 0, lds_b, 0, h);
-    stage_half(ga, args.lda, BK, lds_a, 1, h);
-    stage_half(gb, args.ldb, BK, lds_b, 1, h);
   }
 
   for (int p = 0; p < pairs; ++p) {
@@ -128,17 +128,15 @@ __global__ __launch_bounds__(NTH, 2) void k_gemm256_v2(GemmArgs args) {
       const int ih = (qq == 2 || qq == 3) ? 1 : 0;
       const int jh = (qq == 1 || qq == 2) ? 1 : 0;
 
-      if (ph == 0) {
+      if (ph == 0 || ph == 4) {
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      } else if (ph == 4) {
-        asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
       }
       __builtin_amdgcn_s_barrier();
       __builtin_amdgcn_sched_barrier(0);
 
-      // stage stream, offset 4 phases: phases 4-7 stage tile 2p+2 ->
-      // buf0; phases 0-3 stage tile 2p+3 -> buf1 (skipped on last pair)
-      const int stage_tile = (ph >= 4) ? 2 * p + 2 : 2 * p + 3;
+      // stage stream (see ledger above): phases 0-3 -> tile 2p+1 (buf1),
+      // phases 4-7 -> tile 2p+2 (buf0, skipped on the last pair)
+      const int stage_tile = (ph >= 4) ? 2 * p + 2 : 2 * p + 1;
       if (stage_tile < ktiles) {
         const int sph = ph & 3;        // 0: A h0, 1: A h1, 2: B h0, 3: B h1
         const int k0 = stage_tile * BK;
