@@ -72,11 +72,21 @@ def main():
     # MoE models keep contexts at decode size (EP prefill buffers scale
     # with M*topk; torch prefill has no collectives at world=1)
     dense = cfg.n_experts == 0
-    ctx_m = prefill_m if (on_gpu and dense) else batch
+    # ag_rs sizes contexts for the FUSED prefill (batch*ctx tokens);
+    # gemm_ar prefills via the torch path (replicated + RCCL AR), so its
+    # symmetric AR contexts only need the decode batch — prefill-sized
+    # replicated inboxes (world * M * H) would not fit the heap
+    ctx_m = prefill_m if (on_gpu and dense and args.mode == "ag_rs") \
+        else batch
     if args.mode == "ag_rs" and on_gpu:
         need = 2.4 * ctx_m * cfg.hidden * 2
         if not dense:  # EP recv/combine symm buffers
             need += 4.6 * ctx_m * cfg.moe_topk * cfg.hidden * 2 / world
+        heap = td.init_symm_heap(size_mb=max(int(need / 1e6) + 1024, 4096))
+    elif args.mode == "gemm_ar" and on_gpu:
+        # one-/two-shot AR inbox+outbox per layer-shared ctx: world
+        # replicated copies of the decode activation
+        need = 3.0 * world * batch * cfg.hidden * 2
         heap = td.init_symm_heap(size_mb=max(int(need / 1e6) + 1024, 4096))
     else:
         heap = td.init_symm_heap()
