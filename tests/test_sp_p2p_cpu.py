@@ -90,3 +90,7 @@ def _body_ulysses_roundtrip(rank, world):
 
 def test_ulysses_layer_roundtrip_cpu_2rank():
     run_distributed(_body_ulysses_roundtrip, world_size=2)
+
+
+def test_sp_decode_cpu_4rank():
+    run_distributed(_body_sp_decode, world_size=4)
