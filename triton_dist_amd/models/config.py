@@ -27,6 +27,12 @@ class ModelConfig:
     # MoE fields (n_experts == 0 -> dense)
     n_experts: int = 0
     moe_topk: int = 8
+    # hybrid GDN (Qwen3-Next geometry): 0 = pure attention; N > 0 keeps
+    # every N-th layer full attention, the rest are GDN mixers
+    gdn_period: int = 0
+    gdn_heads: int = 0
+    gdn_head_k: int = 128
+    gdn_head_v: int = 128
     moe_impl: str = "ep"   # "ep" (expert-parallel) | "tp" (inter-sharded)
     moe_inter: int = 0
 
@@ -54,6 +60,14 @@ PRESETS = {
                           n_heads=32, n_kv_heads=4, head_dim=128,
                           vocab=151936, n_experts=128, moe_topk=8,
                           moe_inter=768),
+    # hybrid GDN demo family (Qwen3-Next geometry; random-init only)
+    "qwen3-next-like": dict(hidden=2048, intermediate=5120, n_layers=48,
+                            n_heads=16, n_kv_heads=2, head_dim=128,
+                            vocab=151936, gdn_period=4, gdn_heads=32,
+                            gdn_head_k=128, gdn_head_v=128),
+    "tiny-gdn": dict(hidden=64, intermediate=128, n_layers=3, n_heads=2,
+                     n_kv_heads=2, head_dim=32, vocab=256, gdn_period=3,
+                     gdn_heads=4, gdn_head_k=16, gdn_head_v=16),
     # tiny configs divisible at TP=4 (4-rank gloo tests)
     "tiny4": dict(hidden=256, intermediate=512, n_layers=2, n_heads=8,
                   n_kv_heads=4, head_dim=64, vocab=512),

@@ -69,6 +69,9 @@ class DenseLLM:
         for li, layer in enumerate(self.layers):
             s = seed + 100 + li * 10
             attn, mlp = layer["attn"], layer["mlp"]
+            if not hasattr(attn, "w_qkv"):  # GDN mixer layers init
+                # their own weights (models/gdn_hybrid.py)
+                continue
             # qkv: rows grouped [Q(all heads); K; V] — shard heads
             wq = full((cfg.n_heads * d, cfg.hidden), s)
             wk = full((cfg.n_kv_heads * d, cfg.hidden), s + 1)
@@ -89,6 +92,13 @@ class DenseLLM:
                                                wu[r * i_s:(r + 1) * i_s]]))
                 wd = full((cfg.hidden, cfg.intermediate), s + 6)
                 mlp.w_down.copy_(wd[:, r * i_s:(r + 1) * i_s].contiguous())
+
+    def make_cache(self, batch: int, max_len: int) -> KVCache:
+        """Cache factory (hybrid models override with HybridCache)."""
+        return KVCache(self.cfg.n_layers, batch, max_len,
+                       max(self.cfg.n_kv_heads // self.world, 1),
+                       self.cfg.head_dim, device=self.device,
+                       dtype=self.dtype)
 
     # ------------------------------------------------------------- contexts
     def init_dist_ctx(self, max_m_total: int):

@@ -22,9 +22,7 @@ class Engine:
         self.batch = batch
         self.max_len = max_len
         cfg = model.cfg
-        self.kv = KVCache(cfg.n_layers, batch, max_len,
-                          cfg.n_kv_heads // model.world, cfg.head_dim,
-                          device=model.device, dtype=model.dtype)
+        self.kv = model.make_cache(batch, max_len)
         if use_graph is None:
             use_graph = torch.cuda.is_available()
         self.use_graph = use_graph

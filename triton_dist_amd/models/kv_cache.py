@@ -127,3 +127,28 @@ class PagedKVCache:
         v = self.v_pool[layer, blk]
         shp = (self.batch, nb * self.block, self.kvh, self.head_dim)
         return k.reshape(shp)[:, :upto_len], v.reshape(shp)[:, :upto_len]
+
+
+class HybridCache(KVCache):
+    """KVCache + per-GDN-layer recurrent state for hybrid (Qwen3-Next
+    style) stacks. GDN state is positionless fp32 [B, local_heads, K, V],
+    zeroed on reset; attention layers use the inherited paged-less KV
+    slots (GDN layers simply never touch theirs)."""
+
+    def __init__(self, *args, **kwargs):
+        super().__init__(*args, **kwargs)
+        self._gdn = {}
+
+    def gdn_state(self, layer: int, batch: int, lh: int, dk: int,
+                  dv: int) -> "torch.Tensor":
+        key = layer
+        if key not in self._gdn:
+            self._gdn[key] = torch.zeros(batch, lh, dk, dv,
+                                         dtype=torch.float32,
+                                         device=self.k.device)
+        return self._gdn[key]
+
+    def reset(self):
+        super().reset()
+        for t in self._gdn.values():
+            t.zero_()

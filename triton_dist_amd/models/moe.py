@@ -74,8 +74,11 @@ class Qwen3MoE(DenseLLM):
 
 
 def AutoLLM(cfg: ModelConfig, device="cpu", dtype=torch.bfloat16, heap=None):
-    """Dispatch dense vs MoE by config (reference models/__init__.py
-    AutoLLM.from_pretrained capability — random-init here)."""
+    """Dispatch dense vs MoE vs hybrid-GDN by config (reference
+    models/__init__.py AutoLLM capability — random-init here)."""
     if cfg.n_experts > 0:
         return Qwen3MoE(cfg, device=device, dtype=dtype, heap=heap)
+    if cfg.gdn_period > 0:
+        from .gdn_hybrid import HybridGDNLLM
+        return HybridGDNLLM(cfg, device=device, dtype=dtype, heap=heap)
     return DenseLLM(cfg, device=device, dtype=dtype, heap=heap)
