@@ -132,3 +132,37 @@ def test_loader_sharded_index():
     import tempfile
     with tempfile.TemporaryDirectory() as tmp:
         run_distributed(_worker_index, world_size=1, args=(tmp,))
+
+
+def _worker_hybrid(rank, world, tmpdir):
+    import torch
+    import triton_dist_amd as td
+    from triton_dist_amd.models import (AutoLLM, get_config,
+                                        load_hf_weights, save_hf_weights)
+
+    td.init_symm_heap(size_mb=16)
+    cfg = get_config("tiny-gdn", tp_mode="torch")
+    if world == 1:
+        a = AutoLLM(cfg, device="cpu")
+        a.init_weights(seed=21)
+        save_hf_weights(a, tmpdir)
+        b = AutoLLM(cfg, device="cpu")
+        load_hf_weights(b, tmpdir)
+        for la, lb in zip(a.layers, b.layers):
+            aa, ab = la["attn"], lb["attn"]
+            if hasattr(aa, "w_in"):
+                assert torch.equal(aa.w_in, ab.w_in)
+                assert torch.equal(aa.w_out, ab.w_out)
+            else:
+                assert torch.equal(aa.w_qkv, ab.w_qkv)
+    else:
+        m = AutoLLM(cfg, device="cpu")
+        load_hf_weights(m, tmpdir)  # sharded load of the world-1 save
+    td.shutdown_heap()
+
+
+def test_loader_hybrid_gdn():
+    import tempfile
+    with tempfile.TemporaryDirectory() as tmp:
+        run_distributed(_worker_hybrid, world_size=1, args=(tmp,))
+        run_distributed(_worker_hybrid, world_size=2, args=(tmp,))

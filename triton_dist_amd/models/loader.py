@@ -101,6 +101,10 @@ def load_hf_weights(model, path: str, strict: bool = True) -> int:
         put(layer["ln1"], take(p + "input_layernorm.weight"))
         put(layer["ln2"], take(p + "post_attention_layernorm.weight"))
 
+        if not hasattr(attn, "w_qkv"):  # GDN mixer layer (hybrid family)
+            _load_gdn_mixer(attn, ckpt, p, take, put)
+            _load_mlp(mlp, ckpt, p, r, take, put)
+            continue
         qh, kvh = attn.qh, attn.kvh
         wq = take(p + "self_attn.q_proj.weight")
         wk = take(p + "self_attn.k_proj.weight")
@@ -120,14 +124,7 @@ def load_hf_weights(model, path: str, strict: bool = True) -> int:
                 raise KeyError(qn)
 
         if hasattr(mlp, "inter_shard"):  # dense TP_MLP
-            i_s = mlp.inter_shard
-            wg = take(p + "mlp.gate_proj.weight")
-            wu = take(p + "mlp.up_proj.weight")
-            put(mlp.w_gate_up,
-                torch.cat([wg[r * i_s:(r + 1) * i_s],
-                           wu[r * i_s:(r + 1) * i_s]]))
-            wd = take(p + "mlp.down_proj.weight")
-            put(mlp.w_down, wd[:, r * i_s:(r + 1) * i_s].contiguous())
+            _load_mlp(mlp, ckpt, p, r, take, put)
         else:  # EPMoELayer
             put(mlp.router, take(p + "mlp.gate.weight"))
             lo = r * mlp.e_loc
@@ -139,6 +136,50 @@ def load_hf_weights(model, path: str, strict: bool = True) -> int:
                 put(mlp.w_gate_up[le], torch.cat([wg, wu]))
                 put(mlp.w_down[le], take(ep + "down_proj.weight"))
     return used
+
+
+def _save_mlp(out, p, mlp):
+    i_s = mlp.inter_shard
+    gu = mlp.w_gate_up.cpu()
+    out[p + "mlp.gate_proj.weight"] = gu[:i_s]
+    out[p + "mlp.up_proj.weight"] = gu[i_s:]
+    out[p + "mlp.down_proj.weight"] = mlp.w_down.cpu()
+
+
+def _load_mlp(mlp, ckpt, p, r, take, put):
+    import torch
+
+    i_s = mlp.inter_shard
+    wg = take(p + "mlp.gate_proj.weight")
+    wu = take(p + "mlp.up_proj.weight")
+    put(mlp.w_gate_up,
+        torch.cat([wg[r * i_s:(r + 1) * i_s],
+                   wu[r * i_s:(r + 1) * i_s]]))
+    wd = take(p + "mlp.down_proj.weight")
+    put(mlp.w_down, wd[:, r * i_s:(r + 1) * i_s].contiguous())
+
+
+def _load_gdn_mixer(attn, ckpt, p, take, put):
+    """GDN mixer shard: heads split across ranks; checkpoint names use a
+    `gdn.` block (our own serialization — no HF Qwen3-Next mapping is
+    attempted, this family is random-init/round-trip only)."""
+    import torch
+
+    r, lh = attn.rank, attn.lh
+    dk, dv = attn.dk, attn.dv
+    wq = take(p + "gdn.q_proj.weight")
+    wk = take(p + "gdn.k_proj.weight")
+    wv = take(p + "gdn.v_proj.weight")
+    wg = take(p + "gdn.g_proj.weight")
+    wb = take(p + "gdn.b_proj.weight")
+    wo = take(p + "gdn.o_proj.weight")
+    put(attn.w_in, torch.cat([
+        wq[r * lh * dk:(r + 1) * lh * dk],
+        wk[r * lh * dk:(r + 1) * lh * dk],
+        wv[r * lh * dv:(r + 1) * lh * dv],
+        wg[r * lh:(r + 1) * lh],
+        wb[r * lh:(r + 1) * lh]]))
+    put(attn.w_out, wo[:, r * lh * dv:(r + 1) * lh * dv].contiguous())
 
 
 def save_hf_weights(model, path: str):
@@ -162,6 +203,19 @@ def save_hf_weights(model, path: str):
         attn, mlp = layer["attn"], layer["mlp"]
         out[p + "input_layernorm.weight"] = layer["ln1"].cpu()
         out[p + "post_attention_layernorm.weight"] = layer["ln2"].cpu()
+        if not hasattr(attn, "w_qkv"):  # GDN mixer
+            lh, dk, dv = attn.lh, attn.dk, attn.dv
+            w_in = attn.w_in.cpu()
+            out[p + "gdn.q_proj.weight"] = w_in[:lh * dk]
+            out[p + "gdn.k_proj.weight"] = w_in[lh * dk:2 * lh * dk]
+            out[p + "gdn.v_proj.weight"] = \
+                w_in[2 * lh * dk:2 * lh * dk + lh * dv]
+            out[p + "gdn.g_proj.weight"] = \
+                w_in[2 * lh * dk + lh * dv:2 * lh * dk + lh * dv + lh]
+            out[p + "gdn.b_proj.weight"] = w_in[2 * lh * dk + lh * dv + lh:]
+            out[p + "gdn.o_proj.weight"] = attn.w_out.cpu()
+            _save_mlp(out, p, layer["mlp"])
+            continue
         qh, kvh = attn.qh, attn.kvh
         qkv = attn.w_qkv.cpu()
         out[p + "self_attn.q_proj.weight"] = qkv[:qh * d]
@@ -172,11 +226,7 @@ def save_hf_weights(model, path: str):
             out[p + "self_attn.q_norm.weight"] = attn.q_norm_w.cpu()
             out[p + "self_attn.k_norm.weight"] = attn.k_norm_w.cpu()
         if hasattr(mlp, "inter_shard"):
-            i_s = mlp.inter_shard
-            gu = mlp.w_gate_up.cpu()
-            out[p + "mlp.gate_proj.weight"] = gu[:i_s]
-            out[p + "mlp.up_proj.weight"] = gu[i_s:]
-            out[p + "mlp.down_proj.weight"] = mlp.w_down.cpu()
+            _save_mlp(out, p, mlp)
         else:
             out[p + "mlp.gate.weight"] = mlp.router.cpu()
             for e in range(mlp.e_loc):
