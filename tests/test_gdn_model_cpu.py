@@ -6,14 +6,14 @@ import torch
 from tests.conftest import run_distributed
 
 
-def _body_hybrid(rank, world):
+def _body_hybrid(rank, world, model_name="tiny-gdn"):
     import triton_dist_amd as td
     from triton_dist_amd.layers.gdn_layer import GDNMixer
     from triton_dist_amd.models import AutoLLM, HybridGDNLLM, get_config
     from triton_dist_amd.utils import assert_allclose
 
     td.init_symm_heap(size_mb=32)
-    cfg = get_config("tiny-gdn", tp_mode="ag_rs", max_length=64)
+    cfg = get_config(model_name, tp_mode="ag_rs", max_length=64)
     model = AutoLLM(cfg, device="cpu")
     assert isinstance(model, HybridGDNLLM)
     kinds = [type(l["attn"]).__name__ for l in model.layers]
@@ -73,3 +73,11 @@ def _body_engine(rank, world):
 
 def test_hybrid_gdn_engine_serve():
     run_distributed(_body_engine, world_size=2)
+
+
+def _body_hybrid4(rank, world):
+    _body_hybrid(rank, world, model_name="tiny-gdn4")
+
+
+def test_hybrid_gdn_model_4rank():
+    run_distributed(_body_hybrid4, world_size=4)

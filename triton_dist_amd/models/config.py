@@ -68,6 +68,9 @@ PRESETS = {
     "tiny-gdn": dict(hidden=64, intermediate=128, n_layers=3, n_heads=2,
                      n_kv_heads=2, head_dim=32, vocab=256, gdn_period=3,
                      gdn_heads=4, gdn_head_k=16, gdn_head_v=16),
+    "tiny-gdn4": dict(hidden=64, intermediate=128, n_layers=3, n_heads=4,
+                      n_kv_heads=4, head_dim=16, vocab=256, gdn_period=3,
+                      gdn_heads=4, gdn_head_k=16, gdn_head_v=16),
     # tiny configs divisible at TP=4 (4-rank gloo tests)
     "tiny4": dict(hidden=256, intermediate=512, n_layers=2, n_heads=8,
                   n_kv_heads=4, head_dim=64, vocab=512),

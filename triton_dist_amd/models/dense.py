@@ -69,20 +69,21 @@ class DenseLLM:
         for li, layer in enumerate(self.layers):
             s = seed + 100 + li * 10
             attn, mlp = layer["attn"], layer["mlp"]
-            if not hasattr(attn, "w_qkv"):  # GDN mixer layers init
-                # their own weights (models/gdn_hybrid.py)
-                continue
-            # qkv: rows grouped [Q(all heads); K; V] — shard heads
-            wq = full((cfg.n_heads * d, cfg.hidden), s)
-            wk = full((cfg.n_kv_heads * d, cfg.hidden), s + 1)
-            wv = full((cfg.n_kv_heads * d, cfg.hidden), s + 2)
-            qh, kvh = attn.qh, attn.kvh
-            attn.w_qkv.copy_(torch.cat([
-                wq[r * qh * d:(r + 1) * qh * d],
-                wk[r * kvh * d:(r + 1) * kvh * d],
-                wv[r * kvh * d:(r + 1) * kvh * d]]))
-            wo = full((cfg.hidden, cfg.n_heads * d), s + 3)  # K-shard
-            attn.w_o.copy_(wo[:, r * qh * d:(r + 1) * qh * d].contiguous())
+            if hasattr(attn, "w_qkv"):  # GDN mixer layers init their own
+                # attention-slot weights (models/gdn_hybrid.py); the MLP
+                # below is initialized for EVERY layer kind
+                # qkv: rows grouped [Q(all heads); K; V] — shard heads
+                wq = full((cfg.n_heads * d, cfg.hidden), s)
+                wk = full((cfg.n_kv_heads * d, cfg.hidden), s + 1)
+                wv = full((cfg.n_kv_heads * d, cfg.hidden), s + 2)
+                qh, kvh = attn.qh, attn.kvh
+                attn.w_qkv.copy_(torch.cat([
+                    wq[r * qh * d:(r + 1) * qh * d],
+                    wk[r * kvh * d:(r + 1) * kvh * d],
+                    wv[r * kvh * d:(r + 1) * kvh * d]]))
+                wo = full((cfg.hidden, cfg.n_heads * d), s + 3)  # K-shard
+                attn.w_o.copy_(
+                    wo[:, r * qh * d:(r + 1) * qh * d].contiguous())
             if hasattr(mlp, "inter_shard"):  # dense TP_MLP (MoE layers
                 # initialize their own expert weights in the subclass)
                 i_s = mlp.inter_shard
