@@ -7,7 +7,8 @@ import pytest
 
 from triton_dist_amd.models.config import get_config
 
-BENCH_MODELS = ["qwen3-32b", "qwen3-8b", "qwen3-30b-a3b", "seed-oss-36b"]
+BENCH_MODELS = ["qwen3-32b", "qwen3-8b", "qwen3-30b-a3b", "seed-oss-36b",
+                "qwen3-next-like"]
 BATCH_PER_GPU = 512
 CTX = 128
 
@@ -35,6 +36,16 @@ def _dense_shapes(cfg, world):
 @pytest.mark.parametrize("world", [1, 2, 4, 8])
 def test_fused_gemm_shapes(model, world):
     cfg = get_config(model, tp_mode="ag_rs")
+    if cfg.gdn_period:
+        # hybrid: GDN head shard must divide; mixer proj/out GEMMs ride
+        # the same ag/rs machinery with n = proj_dim / hidden
+        assert cfg.gdn_heads % world == 0
+        from triton_dist_amd.layers.gdn_layer import GDNMixer
+        lh = cfg.gdn_heads // world
+        used = lh * (2 * cfg.gdn_head_k + cfg.gdn_head_v + 2)
+        proj_dim = (used + 127) & ~127  # GDNMixer pads to 128
+        assert proj_dim % 128 == 0
+        assert (lh * cfg.gdn_head_v) % 64 == 0 or world == 0  # rs k-dim
     if cfg.n_experts:
         # MoE: attention GEMMs only (FFN goes through EP grouped kernels)
         d = cfg.head_dim

@@ -39,8 +39,12 @@ class GDNMixer:
         self.mode = mode
         self.device, self.dtype = device, dtype
         # fused qkvgb projection: rows [q(lh*dk); k(lh*dk); v(lh*dv);
-        # g(lh); beta(lh)] — one GEMM per step like TP_Attn's w_qkv
-        self.proj_dim = self.lh * (2 * self.dk + self.dv + 2)
+        # g(lh); beta(lh); zero-pad] — one GEMM per step like TP_Attn's
+        # w_qkv. Padded to a multiple of 128 so the fused AG-GEMM
+        # consumers' N-tiling constraint holds for any head count.
+        used = self.lh * (2 * self.dk + self.dv + 2)
+        self.proj_dim = (used + 127) & ~127
+        self._proj_used = used
         self.w_in = torch.empty(self.proj_dim, hidden, device=device,
                                 dtype=dtype)
         self.w_out = torch.empty(hidden, self.lh * self.dv, device=device,
@@ -67,12 +71,15 @@ class GDNMixer:
         wb = full((H, self.hidden))
         wo = full((self.hidden, H * dv))
         r, lh = self.rank, self.lh
+        pad = torch.zeros(self.proj_dim - self._proj_used, self.hidden,
+                          device=self.device, dtype=self.dtype)
         self.w_in.copy_(torch.cat([
             wq[r * lh * dk:(r + 1) * lh * dk],
             wk[r * lh * dk:(r + 1) * lh * dk],
             wv[r * lh * dv:(r + 1) * lh * dv],
             wg[r * lh:(r + 1) * lh],
-            wb[r * lh:(r + 1) * lh]]))
+            wb[r * lh:(r + 1) * lh],
+            pad]))
         self.w_out.copy_(wo[:, r * lh * dv:(r + 1) * lh * dv].contiguous())
 
     # ----------------------------------------------------------- contexts
@@ -99,7 +106,7 @@ class GDNMixer:
         q = proj[:, :lh * dk].reshape(m, lh, dk)
         k = proj[:, lh * dk:2 * lh * dk].reshape(m, lh, dk)
         v = proj[:, 2 * lh * dk:2 * lh * dk + lh * dv].reshape(m, lh, dv)
-        gb = proj[:, 2 * lh * dk + lh * dv:].float()
+        gb = proj[:, 2 * lh * dk + lh * dv:self._proj_used].float()
         g = F.logsigmoid(gb[:, :lh])
         beta = torch.sigmoid(gb[:, lh:])
         k = F.normalize(k.float(), p=2, dim=-1).to(k.dtype)

@@ -173,12 +173,14 @@ def _load_gdn_mixer(attn, ckpt, p, take, put):
     wg = take(p + "gdn.g_proj.weight")
     wb = take(p + "gdn.b_proj.weight")
     wo = take(p + "gdn.o_proj.weight")
+    pad = torch.zeros(attn.proj_dim - attn._proj_used, attn.hidden)
     put(attn.w_in, torch.cat([
         wq[r * lh * dk:(r + 1) * lh * dk],
         wk[r * lh * dk:(r + 1) * lh * dk],
         wv[r * lh * dv:(r + 1) * lh * dv],
         wg[r * lh:(r + 1) * lh],
-        wb[r * lh:(r + 1) * lh]]))
+        wb[r * lh:(r + 1) * lh],
+        pad]))
     put(attn.w_out, wo[:, r * lh * dv:(r + 1) * lh * dv].contiguous())
 
 
@@ -205,7 +207,7 @@ def save_hf_weights(model, path: str):
         out[p + "post_attention_layernorm.weight"] = layer["ln2"].cpu()
         if not hasattr(attn, "w_qkv"):  # GDN mixer
             lh, dk, dv = attn.lh, attn.dk, attn.dv
-            w_in = attn.w_in.cpu()
+            w_in = attn.w_in.cpu()[:attn._proj_used]  # drop alignment pad
             out[p + "gdn.q_proj.weight"] = w_in[:lh * dk]
             out[p + "gdn.k_proj.weight"] = w_in[lh * dk:2 * lh * dk]
             out[p + "gdn.v_proj.weight"] = \
