@@ -106,3 +106,28 @@ def _body_hybrid_armode(rank, world):
 
 def test_hybrid_gdn_allreduce_mode_2rank():
     run_distributed(_body_hybrid_armode, world_size=2)
+
+
+def _body_hybrid_gemm_ar(rank, world):
+    import triton_dist_amd as td
+    from triton_dist_amd.models import AutoLLM, get_config
+    from triton_dist_amd.utils import assert_allclose
+
+    td.init_symm_heap(size_mb=32)
+    cfg = get_config("tiny-gdn", tp_mode="gemm_ar", max_length=64)
+    model = AutoLLM(cfg, device="cpu")
+    model.init_weights(seed=5)
+    b, s = 4, 8
+    model.init_dist_ctx(max_m_total=b * s)
+    kv1 = model.make_cache(b, 32)
+    kv2 = model.make_cache(b, 32)
+    tokens = torch.randint(0, cfg.vocab, (b, s),
+                           generator=torch.Generator().manual_seed(3))
+    pos = torch.arange(s).expand(b, s)
+    l1 = model.step(tokens, kv1, pos, prefill=True)
+    l2 = model.step(tokens, kv2, pos, prefill=True, mode="torch")
+    assert_allclose(l1, l2, atol=1e-1, rtol=5e-2)
+
+
+def test_hybrid_gdn_gemm_ar_mode_2rank():
+    run_distributed(_body_hybrid_gemm_ar, world_size=2)

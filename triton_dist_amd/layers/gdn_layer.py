@@ -86,6 +86,14 @@ class GDNMixer:
     def init_ctx(self, max_m_total: int, ag_ctx=None, rs_ctx=None):
         """Same contract as TP_Attn.init_ctx: returns the shared (ag, rs)
         pair so DenseLLM's chain threads through mixer layers too."""
+        if self.mode == "gemm_ar":
+            from ..ops.allreduce import create_allreduce_context
+
+            if ag_ctx is None:
+                ag_ctx = create_allreduce_context(
+                    max_m_total * self.hidden, heap=self.heap)
+            self.ar_ctx = ag_ctx
+            return ag_ctx, None
         if self.mode != "ag_rs":
             return ag_ctx, rs_ctx
         from ..ops.allgather_gemm import create_ag_gemm_context
@@ -148,6 +156,11 @@ class GDNMixer:
 
         if self.mode == "ag_rs":
             return gemm_rs(o, self.w_out, self.rs_ctx)
+        if self.mode == "gemm_ar" and getattr(self, "ar_ctx", None) \
+                is not None:
+            from ..ops.allreduce import gemm_allreduce
+
+            return gemm_allreduce(o, self.w_out, self.ar_ctx)
         partial = best_gemm(o, self.w_out) if o.is_cuda \
             else (o.float() @ self.w_out.float().t()).to(self.dtype)
         if self.mode in ("allreduce", "gemm_ar") and self.world > 1:
