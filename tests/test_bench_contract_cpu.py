@@ -36,3 +36,11 @@ def test_bench_json_contract():
 def test_bench_json_moe_metric_name():
     j = _run(["--model", "tiny-moe", "--steps", "2", "--warmup", "1"])
     assert j["metric"] == "tiny_moe_tp_decode_tokens_per_s"
+
+
+def test_doctor_runs():
+    """`python -m triton_dist_amd.doctor` exits 0 on this host."""
+    r = subprocess.run([sys.executable, "-m", "triton_dist_amd.doctor"],
+                       capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "extension: LOADED" in r.stdout, r.stdout
