@@ -92,6 +92,12 @@ class SymmHeap:
 
         if backend == "hip":
             from .. import _C
+            if _C is None:
+                raise RuntimeError(
+                    "triton_dist_amd._C is not built but a GPU is present "
+                    "— run `python triton_dist_amd/build.py` (the HIP "
+                    "extension must load on GPU hosts; silent eager "
+                    "fallbacks are not allowed)")
             self._C = _C
             self.device = local_device()
             # coarse-grained (hipMalloc) by default: fine-grained allocations
