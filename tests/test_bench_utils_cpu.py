@@ -23,3 +23,28 @@ def test_perf_func_contract():
 def test_perf_func_l2_reset_cpu_fallback():
     out, ms = perf_func_with_l2_reset(lambda: 3, iters=2, warmup=1)
     assert out == 3 and ms >= 0.0
+
+
+def test_kernel_profiler_bookkeeping(tmp_path):
+    """KernelProfiler record decode + chrome export + summary (device
+    buffers faked on CPU; the device writer is csrc profiler.hpp)."""
+    from triton_dist_amd.tools.profiler import KernelProfiler
+
+    p = KernelProfiler(capacity=8, device="cpu")
+    # fake 3 records: (block, tag, t0, t1) at 100 MHz
+    p.buf[0] = torch.tensor([5, 0, 1000, 1200])   # tile_wait 2 us
+    p.buf[1] = torch.tensor([5, 1, 1200, 1700])   # tile_compute 5 us
+    p.buf[2] = torch.tensor([6, 0, 1000, 1100])   # tile_wait 1 us
+    p.cursor[0] = 3
+    s = p.summary()
+    assert s["tile_wait"]["count"] == 2
+    assert abs(s["tile_wait"]["total_us"] - 3.0) < 1e-6
+    assert abs(s["tile_compute"]["mean_us"] - 5.0) < 1e-6
+    out = tmp_path / "trace.json"
+    events = p.export_chrome_trace(str(out), rank=1)
+    assert len(events) == 3 and out.exists()
+    data = json.loads(out.read_text())
+    assert len(data["traceEvents"]) == 3
+    assert {e["name"] for e in events} == {"tile_wait", "tile_compute"}
+    p.reset()
+    assert len(p.records()) == 0
