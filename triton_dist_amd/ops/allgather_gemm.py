@@ -200,8 +200,6 @@ def a2a_gemm(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
     assert k == ctx.k and m <= ctx.max_m_per_rank
 
     if ctx.heap.backend != "hip":
-        from .collectives import all_to_all_single  # gloo-emulated a2a
-        from .collectives import CollContext  # noqa: F401 (doc)
         import torch.distributed as dist
         gathered = [torch.empty_like(a) for _ in range(world)]
         dist.all_gather(gathered, a.contiguous())
