@@ -16,6 +16,11 @@ from triton_dist_amd.ops import chunk_gated_delta_rule_fwd, gdn_decode_step
 
 
 def main():
+    import argparse
+
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--check", action="store_true")
+    args = ap.parse_args()
     dev = "cuda" if torch.cuda.is_available() else "cpu"
     B, T, H, K, V = (1, 8192, 16, 128, 128) if dev == "cuda" \
         else (1, 256, 4, 32, 32)
@@ -31,6 +36,21 @@ def main():
         if dev == "cuda":
             torch.cuda.synchronize()
 
+    if args.check:
+        from triton_dist_amd.ops import gated_delta_rule_recurrent_ref
+
+        Bs, Ts = 1, 96
+        o_ref, _ = gated_delta_rule_recurrent_ref(
+            q[:Bs, :Ts].float().cpu(), k[:Bs, :Ts].float().cpu(),
+            v[:Bs, :Ts].float().cpu(), g[:Bs, :Ts].cpu(),
+            beta[:Bs, :Ts].cpu(), scale)
+        o_c, _ = chunk_gated_delta_rule_fwd(q[:Bs, :Ts], k[:Bs, :Ts],
+                                            v[:Bs, :Ts], g[:Bs, :Ts],
+                                            beta[:Bs, :Ts], scale)
+        rel = ((o_c.float().cpu() - o_ref).abs().max()
+               / o_ref.abs().max()).item()
+        print(f"check vs recurrent ref: rel {rel:.2e} "
+              f"({'OK' if rel < 0.05 else 'FAIL'})")
     for _ in range(2):
         chunk_gated_delta_rule_fwd(q, k, v, g, beta, scale)
     sync()
