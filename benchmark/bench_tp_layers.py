@@ -47,6 +47,7 @@ def main():
     ap.add_argument("--kv-heads", type=int, default=8)
     ap.add_argument("--tokens", type=int, default=512)  # per rank
     ap.add_argument("--modes", default="ag_rs,allreduce")
+    ap.add_argument("--check", action="store_true")
     args = ap.parse_args()
     td.initialize_distributed()
     heap = td.init_symm_heap(
@@ -61,6 +62,18 @@ def main():
         mlp.init_ctx(max_m_total=m_total)
         x = (torch.randn(args.tokens, args.hidden, device=dev) / 8).to(
             torch.bfloat16)
+        if args.check and mode == "ag_rs":
+            # gather the shard stream and compare vs the replicated golden
+            xs = [torch.empty_like(x) for _ in range(world)]
+            dist.all_gather(xs, x.cpu() if dev == "cuda" else x)
+            full = torch.cat([t.to(dev) for t in xs])
+            y = mlp(x)
+            ref = mlp.torch_fwd(full)[rank * args.tokens:
+                                      (rank + 1) * args.tokens]
+            err = (y.float() - ref.float()).abs().max().item()
+            if rank == 0:
+                print(f"TP_MLP {mode} check: max err {err:.3f} "
+                      f"({'OK' if err < 0.5 else 'FAIL'})")
         us = timeit(lambda: mlp(x))
         gf = 2 * m_total * 3 * args.hidden * args.inter / world / 1e9
         if rank == 0:
