@@ -84,3 +84,48 @@ def test_gemm_dispatch_policies():
     assert choose_splits(4096, 4096, 4096) == 1
     # occupancy-starved decode shard: split-K on
     assert choose_splits(512, 1280, 5120) > 1
+
+
+def test_gemm256_v2_stage_read_consistency():
+    """Full addressing simulation of the experimental v2 kernel
+    (csrc/kernels/gemm256_v2.hip): emulate stage_half's global->LDS
+    placement and read_frag's swizzled reads, and check every fragment
+    element equals the matrix element the MFMA layout requires. Catches
+    swizzle/layout bugs before any GPU time is spent."""
+    BM, BK, NTH = 256, 64, 512
+    HALF_ELEMS = BM * BK // 2
+
+    # A[row][col] encoded as row * 1000 + col
+    def A(row, col):
+        return row * 1000 + col
+
+    lds = {}
+    # stage both halves of one K-tile (k0 = 0, buf offset ignored)
+    for h in range(2):
+        for it in range(2):
+            for tid in range(NTH):
+                q = it * NTH + tid
+                dst_byte = (h * HALF_ELEMS + q * 8) * 2
+                un = swz_off(dst_byte)
+                row = un // (BK * 2)
+                col = (un % (BK * 2)) // 2
+                # global_load_lds writes lane's 16B at the linear dest
+                for e in range(8):
+                    lds[h * HALF_ELEMS + q * 8 + e] = A(row, col + e)
+
+    assert len(lds) == BM * BK  # every slot written exactly once
+
+    def read_frag(row, ks, lane):
+        byte_off = (row * BK + ks * 32 + (lane >> 4) * 8) * 2
+        off = swz_off(byte_off) // 2
+        return [lds[off + e] for e in range(8)]
+
+    # every (row, ks, lane) fragment must deliver A[row][ks*32+(l>>4)*8+e]
+    for lane in range(64):
+        for base_row in range(0, BM, 16):
+            row = base_row + (lane & 15)
+            for ks in range(2):
+                frag = read_frag(row, ks, lane)
+                want = [A(row, ks * 32 + (lane >> 4) * 8 + e)
+                        for e in range(8)]
+                assert frag == want, (row, ks, lane, frag[:2], want[:2])
