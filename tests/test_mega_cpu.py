@@ -57,3 +57,41 @@ def test_finalize_shapes():
     assert buf.shape[0] == len(g.tasks) and queue.numel() == len(g.tasks)
     assert offs[-1].item() == len(g.tasks)
     assert score.numel() == len(g.ops)
+
+
+def test_ksplit_task_semantics():
+    """Execute emit_gemm's K-split task stream in numpy (each
+    T_GEMM_TILE_PART computes its (tile, k-range) partial into its ws
+    slice; each T_TILE_REDUCE sums slices) and compare against the full
+    matmul — validates the task-arg encoding and the splitting math
+    end-to-end on CPU."""
+    import numpy as np
+
+    BM, BN = 32, 128
+    batch, n, k, ksplit = 70, 256, 512, 4   # batch NOT a tile multiple
+    m_pad = (batch + BM - 1) // BM * BM
+    rng = np.random.default_rng(0)
+    A = rng.standard_normal((m_pad, k)).astype(np.float32)
+    B = rng.standard_normal((n, k)).astype(np.float32)
+    C = np.zeros((batch, n), dtype=np.float32)
+    ws = np.zeros((ksplit, m_pad, n), dtype=np.float32)
+
+    g = MegaGraph()
+    emit_gemm(g, 1, 2, 3, batch=batch, n=n, k=k, dep=None, ksplit=ksplit,
+              ws_ptr=9)
+    for (tt, _slot, _d0, _d0n, _d1, _d1n, args, _lvl) in g.tasks:
+        if tt == T_GEMM_TILE_PART:
+            (_a, _b, _w, m, nn, kk, pm, pn, k0, klen, sk) = args[:11]
+            rows = min(BM, m - pm * BM)
+            a_blk = A[pm * BM:pm * BM + rows, k0:k0 + klen]
+            b_blk = B[pn * BN:(pn + 1) * BN, k0:k0 + klen]
+            ws[sk, pm * BM:pm * BM + rows, pn * BN:(pn + 1) * BN] = \
+                a_blk @ b_blk.T
+        elif tt == T_TILE_REDUCE:
+            (_w, _c, m, nn, pm, pn, ks) = args[:7]
+            rows = min(BM, m - pm * BM)
+            C[pm * BM:pm * BM + rows, pn * BN:(pn + 1) * BN] = \
+                ws[:ks, pm * BM:pm * BM + rows,
+                   pn * BN:(pn + 1) * BN].sum(0)
+    ref = A[:batch] @ B.T
+    assert np.abs(C - ref).max() < 1e-3
