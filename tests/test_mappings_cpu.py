@@ -129,3 +129,30 @@ def test_gemm256_v2_stage_read_consistency():
                 want = [A(row, ks * 32 + (lane >> 4) * 8 + e)
                         for e in range(8)]
                 assert frag == want, (row, ks, lane, frag[:2], want[:2])
+
+
+@pytest.mark.parametrize("seg,chunks", [(4096, 8), (1000 * 8, 8), (64, 8),
+                                        (8, 8), (8192, 3), (24, 5)])
+def test_rs_chunk_ranges_cover_segment(seg, chunks):
+    """Mirror of the reduce_scatter/all_to_all chunk math
+    (csrc/kernels/collectives.hip): 8-aligned chunk boundaries computed
+    IDENTICALLY by pusher and reducer must tile [0, seg) exactly, with
+    trailing chunks allowed to be empty."""
+    per = ((seg + chunks - 1) // chunks + 7) & ~7
+    covered = []
+    for c in range(chunks):
+        lo = c * per
+        hi = min(lo + per, seg)
+        if lo >= seg:
+            continue  # empty trailing chunk (flag still published)
+        # vectorized path only: the kernels assume seg % 8 == 0, so when
+        # it holds every non-empty chunk is 8-aligned
+        if seg % 8 == 0:
+            assert lo % 8 == 0
+        covered.append((lo, hi))
+    # exact tiling, no overlap, no gap
+    pos = 0
+    for lo, hi in covered:
+        assert lo == pos and hi > lo
+        pos = hi
+    assert pos == seg
