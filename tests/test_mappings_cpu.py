@@ -156,3 +156,50 @@ def test_rs_chunk_ranges_cover_segment(seg, chunks):
         assert lo == pos and hi > lo
         pos = hi
     assert pos == seg
+
+
+def test_flash_decode_tr_read_bfrag_layout():
+    """Executable mirror of the ds_read_b64_tr_b16 semantics measured in
+    scripts/probe/probe_attn.hip (within a 16-lane group, lane k supplies
+    an 8-byte word address; lane l receives element l&3 of the words
+    fetched by lanes ((l&15)>>2)+4j, j=0..3) against the PV fragment
+    addressing in csrc/kernels/attention.hip — proves every lane's
+    B-fragment is V[k0+j][col] for its MFMA slot."""
+    kD, kTile = 128, 32
+    stride = kD + 8  # padded V row
+
+    def V(k, c):
+        return k * 1000 + c
+
+    # LDS contents: v_lds[k][c] row-major with padding
+    def lds_elem(addr_elems):
+        k, c = divmod(addr_elems, stride)
+        assert c < kD, "read touched the pad region"
+        return V(k, c)
+
+    def tr_read(addrs_bytes, lane):
+        """addrs_bytes[l] = per-lane byte address within the wave."""
+        grp = lane & ~15
+        out = []
+        for j in range(4):
+            word_lane = grp + (((lane & 15) >> 2) + 4 * j)
+            word_addr = addrs_bytes[word_lane]
+            assert word_addr % 8 == 0
+            out.append(lds_elem(word_addr // 2 + (lane & 3)))
+        return out
+
+    for wave in range(4):
+        for lane in range(64):
+            for h in range(2):
+                cg = wave * 2 + h
+                k0 = (lane >> 4) * 8
+                addrs = [((k0_ + ((ll & 15) >> 2)) * stride
+                          + cg * 16 + 4 * (ll & 3)) * 2
+                         for ll in range(64)
+                         for k0_ in [(ll >> 4) * 8]]
+                lo = tr_read(addrs, lane)
+                hi = tr_read([a + 4 * stride * 2 for a in addrs], lane)
+                col = cg * 16 + (lane & 15)
+                want = [V(k0 + j, col) for j in range(4)] + \
+                       [V(k0 + 4 + j, col) for j in range(4)]
+                assert lo + hi == want, (wave, lane, h)
