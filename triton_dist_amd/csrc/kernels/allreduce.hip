@@ -39,7 +39,7 @@ __global__ void k_ar_push(PeerTable pt, const bf16 *__restrict__ x,
   const int pi = blockIdx.y;  // peer index 0..world-2
   const int peer = (pt.rank + 1 + pi) % pt.world;
   const int c = blockIdx.x;
-  const size_t per = (elems + chunks - 1) / chunks;
+  const size_t per = (((elems + chunks - 1) / chunks) + 7) & ~(size_t)7;
   const size_t lo = (size_t)c * per;
   const size_t hi = min(lo + per, elems);
   bf16 *inbox = (bf16 *)((char *)pt.bases[peer] + inbox_off) +
@@ -62,7 +62,7 @@ __global__ void k_ar_reduce(PeerTable pt, const bf16 *__restrict__ x,
                             bf16 *__restrict__ out, const int *flags,
                             int chunks, size_t elems) {
   const int c = blockIdx.x;
-  const size_t per = (elems + chunks - 1) / chunks;
+  const size_t per = (((elems + chunks - 1) / chunks) + 7) & ~(size_t)7;
   const size_t lo = (size_t)c * per;
   const size_t hi = min(lo + per, elems);
   if (threadIdx.x < (unsigned)pt.world && (int)threadIdx.x != pt.rank) {
@@ -118,7 +118,7 @@ __global__ void k_ar2_scatter(PeerTable pt, const bf16 *__restrict__ x,
   const int pi = blockIdx.y;
   const int owner = (pt.rank + 1 + pi) % pt.world;
   const int c = blockIdx.x;
-  const size_t per = (slice + chunks - 1) / chunks;
+  const size_t per = (((slice + chunks - 1) / chunks) + 7) & ~(size_t)7;
   const size_t lo = (size_t)c * per;
   const size_t hi = min(lo + per, slice);
   bf16 *inbox = (bf16 *)((char *)pt.bases[owner] + inbox_off) +
@@ -140,7 +140,7 @@ __global__ void k_ar2_reduce_bcast(PeerTable pt, const bf16 *__restrict__ x,
                                    const int *flags_in, int chunks,
                                    size_t slice) {
   const int c = blockIdx.x;
-  const size_t per = (slice + chunks - 1) / chunks;
+  const size_t per = (((slice + chunks - 1) / chunks) + 7) & ~(size_t)7;
   const size_t lo = (size_t)c * per;
   const size_t hi = min(lo + per, slice);
   if (threadIdx.x < (unsigned)pt.world && (int)threadIdx.x != pt.rank)
@@ -183,7 +183,7 @@ __global__ void k_ar2_assemble(const bf16 *__restrict__ outbox,
                                int world, int chunks, size_t slice) {
   const int c = blockIdx.x;
   const int s = blockIdx.y;  // source slice owner
-  const size_t per = (slice + chunks - 1) / chunks;
+  const size_t per = (((slice + chunks - 1) / chunks) + 7) & ~(size_t)7;
   const size_t lo = (size_t)c * per;
   const size_t hi = min(lo + per, slice);
   if (threadIdx.x == 0) wait_ge_one<Scope::Sys>(flags + s * chunks + c, 1);
