@@ -121,7 +121,10 @@ __global__ void k_ag_pull(PeerTable pt, size_t ws_off, size_t flags_off,
   const int pi = blockIdx.y;  // peer index (skip self)
   const int peer = (pt.rank + 1 + pi) % pt.world;
   const int c = blockIdx.x;
-  const size_t per = (seg_bytes + chunks - 1) / chunks;
+  // 16B-aligned chunk boundaries: the copy is ulonglong2-vectorized and
+  // an unaligned tail would overrun the segment into the next one
+  const size_t per = (((seg_bytes + chunks - 1) / chunks) + 15) &
+                     ~(size_t)15;
   const size_t lo = (size_t)c * per;
   const size_t hi = min(lo + per, seg_bytes);
   const int *pf = (const int *)((char *)pt.bases[peer] + flags_off);
@@ -141,6 +144,8 @@ void launch_ag_pull(const PeerTable &pt, size_t ws_off, size_t flags_off,
                     size_t seg_bytes, int chunks, int chunk_stride,
                     hipStream_t stream) {
   if (pt.world <= 1) return;
+  if (seg_bytes % 16)
+    throw std::runtime_error("ag_pull: seg_bytes % 16 != 0");
   hipLaunchKernelGGL(k_ag_pull, dim3(chunks, pt.world - 1), dim3(256), 0,
                      stream, pt, ws_off, flags_off, seg_bytes, chunks,
                      chunk_stride);
