@@ -15,9 +15,14 @@ def _body_ar_gpu(rank, world):
 
     ctx = create_allreduce_context(max_elems=8 << 20)
     torch.manual_seed(5 + rank)
+    # the (1021, 8)/(1022, 16) rows pin the 8-aligned-chunk-boundary fix:
+    # ceil(elems/chunks) is NOT naturally 8-aligned for them, so the
+    # vectorized tail must not overrun the inbox row (allreduce.hip)
     for method, shape in (("one_shot", (512, 5120)),
                           ("two_shot", (1024, 5120)),
-                          ("one_shot", (8, 1024))):
+                          ("one_shot", (8, 1024)),
+                          ("one_shot", (1021, 8)),
+                          ("two_shot", (1022, 16))):
         x = (torch.randn(shape, device="cuda") / 4).to(torch.bfloat16)
         for _ in range(2):
             out = all_reduce(x, ctx, method=method)
