@@ -219,7 +219,8 @@ def a2a_gemm(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
     heap, _C = ctx.heap, ctx.heap._C
     compute = torch.cuda.current_stream()
     chunks = ctx.chunks_per_rank
-    assert m % chunks == 0 and m % 128 == 0,         f"segment m={m} must divide chunks={chunks} and tile by 128"
+    assert m % chunks == 0 and m % 128 == 0, \
+        f"segment m={m} must divide chunks={chunks} and tile by 128"
     rows_per_chunk = m // chunks
     chunk_bytes = rows_per_chunk * k * 2
     seg_bytes = m * k * 2
