@@ -58,8 +58,7 @@ class UlyssesSPAllToAllLayer:
                                               self.head_dim, heap=self.heap)
         # post direction: tokens play the role of heads and vice versa
         self.post_ctx = create_ulysses_context(
-            max_tokens_local * world // world, self.n_heads, self.head_dim,
-            heap=self.heap)
+            max_tokens_local, self.n_heads, self.head_dim, heap=self.heap)
         return self.pre_ctx
 
     def pre_attn(self, x: torch.Tensor) -> torch.Tensor:
