@@ -203,3 +203,37 @@ def test_flash_decode_tr_read_bfrag_layout():
                 want = [V(k0 + j, col) for j in range(4)] + \
                        [V(k0 + 4 + j, col) for j in range(4)]
                 assert lo + hi == want, (wave, lane, h)
+
+
+def test_gemm256_ring_stage_read_consistency():
+    """Addressing mirror of the PRODUCTION K-slice-ring kernel
+    (csrc/kernels/gemm256.hip stage_slice + kloop fragment reads): the
+    chunk-XOR swizzle applied to the global source at stage time must
+    cancel against the read-side swizzle for every fragment element."""
+    NTH, SLICE_K = 512, 32
+
+    def A(row, col):
+        return row * 1000 + col
+
+    lds = {}
+    # stage one slice (k0 = 0) of one matrix
+    for it in range(2):
+        for tid in range(NTH):
+            q = it * NTH + tid
+            row = q >> 2
+            jp = q & 3
+            jg = swz(row, jp)
+            # global_load_lds lands the lane's 16B at linear chunk q
+            for e in range(8):
+                lds[q * 8 + e] = A(row, jg * 8 + e)
+    assert len(lds) == 256 * SLICE_K
+
+    # kloop read: af[i] = lds[row * SLICE_K + swz(row, jn) * 8 .. +8)
+    for lane in range(64):
+        jn = lane >> 4
+        for base in range(0, 256, 16):
+            row = base + (lane & 15)
+            off = row * SLICE_K + swz(row, jn) * 8
+            frag = [lds[off + e] for e in range(8)]
+            want = [A(row, jn * 8 + e) for e in range(8)]
+            assert frag == want, (row, jn)
