@@ -237,3 +237,34 @@ def test_gemm256_ring_stage_read_consistency():
             frag = [lds[off + e] for e in range(8)]
             want = [A(row, jn * 8 + e) for e in range(8)]
             assert frag == want, (row, jn)
+
+
+def test_gemm256_accumulator_coverage():
+    """The 8-wave (2x4) accumulator layout of the 256^2 kernels must
+    cover every C-tile element exactly once: wave (wr, wc), frag (i, j),
+    reg r, lane l -> row = wr*128 + i*16 + (l>>4)*4 + r,
+    col = wc*64 + j*16 + (l&15). Also validates the v2 quadrant split
+    (ih/jh gray order) covers all 32 fragment positions."""
+    seen = set()
+    for wave in range(8):
+        wr, wc = wave >> 2, wave & 3
+        for lane in range(64):
+            for i in range(8):
+                for j in range(4):
+                    for r in range(4):
+                        row = wr * 128 + i * 16 + ((lane >> 4) * 4 + r)
+                        col = wc * 64 + j * 16 + (lane & 15)
+                        key = (row, col)
+                        assert key not in seen, key
+                        seen.add(key)
+    assert len(seen) == 256 * 256
+
+    # v2 gray-coded quadrants: phases 0..3 cover each (i, j) frag once
+    covered = set()
+    for qq in range(4):
+        ih = 1 if qq in (2, 3) else 0
+        jh = 1 if qq in (1, 2) else 0
+        for i in range(4):
+            for j in range(2):
+                covered.add((ih * 4 + i, jh * 2 + j))
+    assert covered == {(i, j) for i in range(8) for j in range(4)}
