@@ -268,3 +268,37 @@ def test_gemm256_accumulator_coverage():
             for j in range(2):
                 covered.add((ih * 4 + i, jh * 2 + j))
     assert covered == {(i, j) for i in range(8) for j in range(4)}
+
+
+def test_flash_decode_qk_partial_coverage():
+    """QK^T partial mapping (attention.hip): per wave and pos-half, lanes
+    write s_part[wave][(l>>4)*4+r][(l&15)+16h] — every (slot, pos) cell
+    of each wave's partial plane must be written exactly once."""
+    for wave in range(4):
+        seen = set()
+        for h in range(2):
+            for lane in range(64):
+                for r in range(4):
+                    slot = (lane >> 4) * 4 + r
+                    pos = (lane & 15) + 16 * h
+                    key = (slot, pos)
+                    assert key not in seen, key
+                    seen.add(key)
+        assert len(seen) == 16 * 32
+
+
+def test_flash_decode_pv_output_coverage():
+    """PV epilogue (attention.hip): wave w, halves h, regs r, lanes l ->
+    out[row][w*32 + h*16 + (l&15)] with row = (l>>4)*4 + r; the 4 waves
+    together must cover [16 rows] x [128 cols] exactly once."""
+    seen = set()
+    for wave in range(4):
+        for lane in range(64):
+            for h in range(2):
+                for r in range(4):
+                    row = (lane >> 4) * 4 + r
+                    col = wave * 32 + h * 16 + (lane & 15)
+                    key = (row, col)
+                    assert key not in seen, key
+                    seen.add(key)
+    assert len(seen) == 16 * 128
