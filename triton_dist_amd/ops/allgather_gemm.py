@@ -16,9 +16,14 @@ Mechanics here:
     flag memcpy carrying the call epoch (driver wait-value APIs are slow on
     ROCm; a 4B SDMA copy is ~µs).
   * consumer: k_ag_gemm_consumer_bf16 waits per-tile on exactly the chunk
-    flags its rows need (epoch >=) and starts at its own shard.
-  * epoch-counting flags: no reset pass; one cross-GPU barrier at op entry
-    protects workspace reuse.
+    flags its rows need and starts at its own shard.
+  * flag protocol (push paths): reset-my-flags -> cross-GPU entry barrier
+    -> publish. The barrier proves every peer consumed the previous call's
+    workspace before anyone overwrites flags or data, and peers only push
+    after their own barrier, so the reset cannot race their signals. The
+    pull path needs a barrier on BOTH sides of the reset because peers
+    read the flag row remotely (see _allgather_pull). The CPU backend
+    alone uses monotonic epoch tags (ctx.epoch).
 
 The CPU backend runs the same context/epoch/flag logic synchronously over
 the shared-memory mock (correct-by-construction testing on gloo).
@@ -366,12 +371,22 @@ def _allgather_pull(a: torch.Tensor, ctx: AGGemmContext,
                     out: Optional[torch.Tensor]) -> torch.Tensor:
     m, k = a.shape
     world, rank = ctx.world, ctx.rank
+    # pull always materializes world*max_m_per_rank rows, so partial
+    # shards would return stale workspace tail rows — require full shards
+    assert k == ctx.k and m == ctx.max_m_per_rank, \
+        f"pull method requires m == max_m_per_rank ({m} != {ctx.max_m_per_rank})"
     heap, _C = ctx.heap, ctx.heap._C
     compute = torch.cuda.current_stream()
     s = compute.cuda_stream
     chunks = ctx.chunks_per_rank
     seg_bytes = ctx.max_m_per_rank * k * 2
-    # reset only MY flag row (sources publish their own chunks)
+    # Peers read MY flag row remotely from inside k_ag_pull, so the reset
+    # must be fenced on BOTH sides: barrier #1 proves every peer finished
+    # the PREVIOUS call's pull kernel (stream order on each rank puts its
+    # barrier after its pull), then reset, then barrier #2 proves every
+    # rank's reset landed before anyone publishes fresh flags a peer could
+    # confuse with the previous call's.
+    heap.barrier_all_on_stream(compute)
     _C.reset_flags(ctx.flags.ptr() + rank * chunks * 4, chunks, 0, s)
     heap.barrier_all_on_stream(compute)
     _C.memcpy_async(ctx.ws.ptr() + rank * seg_bytes, a.data_ptr(), m * k * 2,
