@@ -35,17 +35,27 @@ def main():
         _C.gemm256_v2_bf16(a.data_ptr(), w.data_ptr(), c.data_ptr(),
                            m, n, k, s)
         torch.cuda.synchronize()
+        c3 = torch.empty_like(c)
+        _C.gemm256_v3_bf16(a.data_ptr(), w.data_ptr(), c3.data_ptr(),
+                           m, n, k, s)
+        torch.cuda.synchronize()
+        rel3 = ((c3.float() - ref).abs().max() / ref.abs().max()).item()
         rel = ((c.float() - ref).abs().max() / ref.abs().max()).item()
         status = "OK" if rel < 2e-2 else "FAIL"
-        line = f"{m}x{n}x{k}: numerics {status} (rel {rel:.1e})"
+        status3 = "OK" if rel3 < 2e-2 else "FAIL"
+        line = (f"{m}x{n}x{k}: v2 {status} (rel {rel:.1e})"
+                f" v3 {status3} (rel {rel3:.1e})")
         if status == "OK" and m >= 4096:
             gf = 2 * m * n * k / 1e9
             us_v2 = t(lambda: _C.gemm256_v2_bf16(
+                a.data_ptr(), w.data_ptr(), c.data_ptr(), m, n, k, s))
+            us_v3 = t(lambda: _C.gemm256_v3_bf16(
                 a.data_ptr(), w.data_ptr(), c.data_ptr(), m, n, k, s))
             us_v1 = t(lambda: _C.gemm_bf16(
                 a.data_ptr(), w.data_ptr(), c.data_ptr(), 0, m, n, k, s))
             us_blt = t(lambda: torch.matmul(a, w.t()))
             line += (f" | v2 {gf/us_v2*1e3:5.0f} TF"
+                     f" v3 {gf/us_v3*1e3:5.0f} TF"
                      f" ring {gf/us_v1*1e3:5.0f} TF"
                      f" blt {gf/us_blt*1e3:5.0f} TF")
         print(line)

@@ -294,6 +294,27 @@ def test_gemm256_v2_experimental(m, n, k):
     assert_allclose(c, a.float() @ w.float().t(), **bf16_gemm_tol(k))
 
 
+@pytest.mark.parametrize("m,n,k", [(256, 256, 128), (512, 768, 512),
+                                   (512, 768, 640), (4096, 4096, 4096)])
+def test_gemm256_v3(m, n, k):
+    """Faithful 8-phase template rebuild (gemm256_v3.hip): register-reuse
+    gray quadrant walk + counted vmcnt drains. Ledger CPU-proven in
+    test_mappings_cpu.test_gemm256_v3_pipeline_ledger."""
+    from triton_dist_amd import _C
+    from triton_dist_amd.utils.testing import assert_allclose, bf16_gemm_tol
+
+    torch.manual_seed(k)
+    a = torch.randn(m, k, device="cuda").to(torch.bfloat16) / 8
+    w = torch.randn(n, k, device="cuda").to(torch.bfloat16) / 8
+    c = torch.empty(m, n, device="cuda", dtype=torch.bfloat16)
+    for _ in range(3):  # re-run: race screen, not just one lucky pass
+        c.zero_()
+        _C.gemm256_v3_bf16(a.data_ptr(), w.data_ptr(), c.data_ptr(), m, n,
+                           k, torch.cuda.current_stream().cuda_stream)
+        torch.cuda.synchronize()
+        assert_allclose(c, a.float() @ w.float().t(), **bf16_gemm_tol(k))
+
+
 @pytest.mark.skipif(not os.environ.get("TD_EXPERIMENTAL"),
                     reason="experimental kernel: set TD_EXPERIMENTAL=1")
 def test_flash_decode_paged_experimental():

@@ -302,3 +302,91 @@ def test_flash_decode_pv_output_coverage():
                     assert key not in seen, key
                     seen.add(key)
     assert len(seen) == 16 * 128
+
+def test_gemm256_v3_pipeline_ledger():
+    """Discrete-event simulation of the v3 8-phase schedule
+    (csrc/kernels/gemm256_v3.hip): stage sequence [A0,B1,A1,B0] per tile,
+    7-half-tile prologue, one stage per phase, vmcnt(6) drains at
+    half-pair ends. Proves for every ktiles (even, up to 64):
+      1. every ds_read hits a half-tile already staged AND retired by a
+         drain that happened before the read's phase (vmcnt soundness),
+      2. no stage overwrites an LDS slot before the phase AFTER its
+         previous occupant's last ds_read issue (barrier separation),
+      3. register reuse is exact: on phases without a re-read, the held
+         register was loaded from the same (tile, half) the MFMA needs.
+    """
+    for ktiles in [2, 4, 6, 8, 16, 64]:
+        total = 4 * ktiles
+        pairs = ktiles // 2
+
+        def seq(s):
+            # -> (tile, matrix, half) ; matrix 0 = A, 1 = B
+            t = s >> 2
+            return [(t, 0, 0), (t, 1, 1), (t, 0, 1), (t, 1, 0)][s & 3]
+
+        # stage events: list of (phase_issued, s). Prologue = phase -1.
+        stage_phase = {}
+        for s in range(min(7, total)):
+            stage_phase[s] = -1
+        nxt = 7
+        # drains: list of (phase, vmcnt_halftiles) happening at END of
+        # that phase (before the barrier that opens the next phase).
+        drains = [(-1, 3 if total > 3 else 0)]
+        issued_at = {-1: min(7, total)}  # stages issued up to end of phase
+        reads = []  # (phase, tile, matrix, half)
+        regs = {}   # matrix -> (tile, half) currently in registers
+        for p in range(pairs):
+            for ph in range(8):
+                g = p * 8 + ph
+                tile = 2 * p + (ph >> 2)
+                qq = ph & 3
+                ih = 1 if qq >= 2 else 0
+                jh = 1 if qq in (1, 2) else 0
+                if qq in (0, 2):
+                    reads.append((g, tile, 0, ih))
+                    regs[0] = (tile, ih)
+                if qq != 2:
+                    reads.append((g, tile, 1, jh))
+                    regs[1] = (tile, jh)
+                # MFMA needs (tile, ih) in A regs and (tile, jh) in B regs
+                assert regs[0] == (tile, ih), (ktiles, g)
+                assert regs[1] == (tile, jh), (ktiles, g)
+                if nxt < total:
+                    stage_phase[nxt] = g
+                    nxt += 1
+                if qq == 3 and (p < pairs - 1 or ph == 3):
+                    n = 0 if nxt >= total else 3
+                    drains.append((g, n))
+                issued_at[g] = nxt
+
+        assert nxt == total  # staging exactly covers all half-tiles
+
+        # retirement: after drain (phase d, n), stages with index
+        # < issued_at[d] - n are certainly landed.
+        def landed_before(phase):
+            best = 0
+            for d, n in drains:
+                if d < phase:
+                    best = max(best, issued_at[d] - n)
+            return best  # stages [0, best) are retired
+
+        want = {}
+        for s in range(total):
+            want[seq(s)] = s
+        for g, tile, mat, half in reads:
+            s = want[(tile, mat, half)]
+            assert s in stage_phase and stage_phase[s] < g, (ktiles, g)
+            assert s < landed_before(g), \
+                f"ktiles={ktiles} phase {g} reads unretired stage {s}"
+
+        # slot-overwrite safety: stage of s overwrites slot of s-8 (same
+        # buf/matrix/half two tiles back); must be issued at a phase
+        # strictly after the last read-issue of the s-8 occupant.
+        last_read = {}
+        for g, tile, mat, half in reads:
+            t, m, h = tile, mat, half
+            last_read[(t, m, h)] = max(last_read.get((t, m, h), -2), g)
+        for s in range(8, total):
+            prev = seq(s - 8)
+            assert stage_phase[s] > last_read[prev], \
+                f"ktiles={ktiles} stage {s} overwrites live slot {prev}"

@@ -92,6 +92,10 @@ void launch_gemm256_sk_bf16(const GemmArgs &g, float *ws, int sk,
 // any dispatch path; see the file header.
 void launch_gemm256_v2_bf16(const GemmArgs &args, hipStream_t stream);
 
+// gemm256_v3.hip — faithful 8-phase template rebuild (register-reuse gray
+// quadrant walk + counted vmcnt drains); candidate production tier.
+void launch_gemm256_v3_bf16(const GemmArgs &args, hipStream_t stream);
+
 // kernels/gemm_splitk.hip ---------------------------------------------------
 void launch_gemm_splitk_bf16(const GemmArgs &g, float *ws, int splits,
                              hipStream_t stream);

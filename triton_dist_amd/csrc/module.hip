@@ -430,6 +430,14 @@ static void gemm256_v2_bf16(uintptr_t a, uintptr_t b, uintptr_t c, int mm,
   TD_CHECK_HIP(hipGetLastError());
 }
 
+static void gemm256_v3_bf16(uintptr_t a, uintptr_t b, uintptr_t c, int mm,
+                            int n, int k, uintptr_t stream) {
+  GemmArgs args{(void *)a, (void *)b, (void *)c, nullptr,
+                mm, n, k, k, k, n};
+  launch_gemm256_v3_bf16(args, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
 static void gemm256_sk_bf16(uintptr_t a, uintptr_t b, uintptr_t c,
                             uintptr_t bias, uintptr_t ws, int m, int n,
                             int k, int sk, uintptr_t stream) {
@@ -752,6 +760,7 @@ PYBIND11_MODULE(_C, m) {
   m.def("gemm_splitk_bf16", &gemm_splitk_bf16);
   m.def("gemm256_sk_bf16", &gemm256_sk_bf16);
   m.def("gemm256_v2_bf16", &gemm256_v2_bf16);
+  m.def("gemm256_v3_bf16", &gemm256_v3_bf16);
   m.def("p2p_attributes", &p2p_attributes);
   m.def("moe_router", &moe_router);
   m.def("reduce_scatter", &reduce_scatter_op);
