@@ -55,3 +55,34 @@ def test_model_gemm_ar_cpu_2rank():
 
 def test_allreduce_cpu_4rank():
     run_distributed(_body_ar, world_size=4)
+
+
+def _body_gemm_ar_tiled(rank, world):
+    from triton_dist_amd.ops import create_allreduce_context, gemm_allreduce
+    from triton_dist_amd.ops.allreduce import _n_owned
+    from triton_dist_amd.utils import assert_allclose, rand_tensor
+
+    # m=512, n=256 -> 2 tiles round-robin across ranks; k=128 hits the
+    # fused tile path's shape gate
+    ctx = create_allreduce_context(max_elems=512 * 256)
+    assert ctx.tile_scatter is not None
+    g = torch.Generator().manual_seed(11 + rank)
+    a = rand_tensor((512, 128), dtype=torch.bfloat16, generator=g) / 8
+    w = rand_tensor((256, 128), dtype=torch.bfloat16, generator=g) / 8
+    import torch.distributed as dist
+    for _ in range(2):  # back-to-back calls reuse the tile buffers
+        outd = gemm_allreduce(a, w, ctx)
+        ref = (a.float() @ w.float().t())
+        dist.all_reduce(ref)
+        assert_allclose(outd, ref.to(torch.bfloat16), atol=8e-2, rtol=5e-2)
+    # owner coverage arithmetic
+    tiles = (512 // 256) * (256 // 256)
+    assert sum(_n_owned(tiles, r, world) for r in range(world)) == tiles
+
+
+def test_gemm_ar_tiled_cpu_2rank():
+    run_distributed(_body_gemm_ar_tiled, world_size=2)
+
+
+def test_gemm_ar_tiled_cpu_4rank():
+    run_distributed(_body_gemm_ar_tiled, world_size=4)

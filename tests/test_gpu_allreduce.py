@@ -64,3 +64,30 @@ def _body_model_gemm_ar_gpu(rank, world):
 
 def test_model_gemm_ar_gpu_2rank():
     run_distributed(_body_model_gemm_ar_gpu, world_size=2)
+
+
+def _body_gemm_ar_tiled_gpu(rank, world):
+    from triton_dist_amd.ops import create_allreduce_context, gemm_allreduce
+    from triton_dist_amd.utils import assert_allclose
+    import torch.distributed as dist
+
+    ctx = create_allreduce_context(max_elems=2048 * 5120)
+    torch.manual_seed(17 + rank)
+    # (512, 5120, 3456): decode down-proj shard shape -> split-K producer;
+    # (2048, 5120, 1024): plain 256^2 producer path; (512, 512, 128):
+    # tiles < consumer-grid edge cases
+    for m, n, k in ((512, 5120, 3456), (2048, 5120, 1024),
+                    (512, 512, 128)):
+        a = (torch.randn(m, k, device="cuda") / 8).to(torch.bfloat16)
+        w = (torch.randn(n, k, device="cuda") / 8).to(torch.bfloat16)
+        for _ in range(2):  # back-to-back: flag reset/reuse protocol
+            out = gemm_allreduce(a, w, ctx)
+            torch.cuda.synchronize()
+            ref = a.float() @ w.float().t()
+            dist.all_reduce(ref)
+            assert_allclose(out, ref.to(torch.bfloat16), atol=2.5e-1,
+                            rtol=5e-2, msg=f"gemm_ar {m}x{n}x{k}")
+
+
+def test_gemm_ar_tiled_gpu_2rank():
+    run_distributed(_body_gemm_ar_tiled_gpu, world_size=2)

@@ -390,3 +390,68 @@ def test_gemm256_v3_pipeline_ledger():
             prev = seq(s - 8)
             assert stage_phase[s] > last_read[prev], \
                 f"ktiles={ktiles} stage {s} overwrites live slot {prev}"
+
+
+@pytest.mark.parametrize("band", [64, 32])
+def test_gemm256_v3_stage_quad_consistency(band):
+    """Addressing mirror of gemm256_v3.hip stage_quad<BAND>: the staged
+    quadrant-union (compact rows -> banded tile rows), the wave-uniform
+    LDS base + lane*16B hardware placement, and the swizzled source must
+    together land every LDS slot of the quad's rows exactly once with
+    the element read_frag expects."""
+    BM, BK, NTH = 256, 64, 512
+
+    def A(row, col):
+        return row * 1000 + col
+
+    lds = {}
+    for h in range(2):
+        for it in range(2):
+            for wave in range(8):
+                cr0 = it * 64 + wave * 8
+                row0 = (cr0 // band) * 2 * band + h * band + (cr0 % band)
+                base_elem = row0 * BK
+                for ln in range(64):
+                    tid = wave * 64 + ln
+                    q = it * NTH + tid
+                    cr = q >> 3
+                    row = (cr // band) * 2 * band + h * band + (cr % band)
+                    p_byte = (row * BK + (q & 7) * 8) * 2
+                    un = p_byte ^ (((p_byte >> 9) & 1) << 5)
+                    assert un // (BK * 2) == row  # swizzle is row-local
+                    col = (un % (BK * 2)) // 2
+                    # HW places lane ln's 16B at base + ln*16
+                    dst = base_elem + ln * 8
+                    assert dst * 2 == p_byte, (band, h, it, wave, ln)
+                    for e in range(8):
+                        lds[dst + e] = A(row, col + e)
+    # both halves staged -> the full tile is covered exactly once
+    assert len(lds) == BM * BK
+
+    def read_frag(row, ks, lane):
+        byte_off = (row * BK + ks * 32 + (lane >> 4) * 8) * 2
+        off = (byte_off ^ (((byte_off >> 9) & 1) << 5)) // 2
+        return [lds[off + e] for e in range(8)]
+
+    for lane in range(64):
+        for base_row in range(0, BM, 16):
+            row = base_row + (lane & 15)
+            for ks in range(2):
+                frag = read_frag(row, ks, lane)
+                want = [A(row, ks * 32 + (lane >> 4) * 8 + e)
+                        for e in range(8)]
+                assert frag == want, (band, row, ks, lane)
+
+    # quad-union row sets match exactly what the gray-walk reads consume:
+    # A (band 64): wave wr reads rows wr*128 + ih*64 + [0,64)
+    # B (band 32): wave wc reads rows wc*64 + jh*32 + [0,32)
+    for h in range(2):
+        staged = {(cr // band) * 2 * band + h * band + (cr % band)
+                  for cr in range(128)}
+        if band == 64:
+            want = {wr * 128 + h * 64 + r for wr in range(2)
+                    for r in range(64)}
+        else:
+            want = {wc * 64 + h * 32 + r for wc in range(4)
+                    for r in range(32)}
+        assert staged == want, (band, h)

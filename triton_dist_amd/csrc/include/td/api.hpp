@@ -107,6 +107,27 @@ void launch_gemm_rs_producer_splitk_bf16(const GemmRsArgs &a, float *ws,
 void launch_moe_router(const void *logits, void *topk_ids, void *topk_w,
                        int T, int E, int K, bool norm, hipStream_t stream);
 
+// kernels/gemm_ar.hip — tile-granular fused GEMM + AllReduce -----------------
+// Round-robin tile ownership (owner = linear_tile % world). Producer GEMM
+// pushes each C tile to the owner's scatter slot + bumps the owner's
+// arrive counter; the consumer (comm stream) reduces owned tiles as their
+// world arrivals land and broadcasts into every rank's symmetric out
+// buffer with per-tile flags. See the file header for the protocol.
+struct GemmArArgs {
+  GemmArgs g;          // a/b as usual; c unused (out lives in the heap)
+  PeerTable pt;
+  size_t scatter_off;  // [world_src][slots][256*256] bf16
+  size_t arrive_off;   // [slots] int32 per rank
+  size_t out_off;      // [M, N] bf16 symmetric out
+  size_t oflags_off;   // [tiles_m * tiles_n] int32 per rank
+  int slots;           // allocated slots per src segment
+};
+void launch_gemm256_ar_producer(const GemmArArgs &args, hipStream_t stream);
+void launch_gemm256_sk_ar_producer(const GemmArArgs &args, float *ws,
+                                   int *done, int sk, hipStream_t stream);
+void launch_ar_tile_consumer(const GemmArArgs &args, int n_owned,
+                             hipStream_t stream);
+
 // kernels/allreduce.hip -----------------------------------------------------
 void launch_allreduce_oneshot(const PeerTable &pt, const void *x, void *out,
                               size_t inbox_off, size_t flags_off,

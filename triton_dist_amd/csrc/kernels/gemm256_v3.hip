@@ -62,25 +62,41 @@ TD_DEV int swz_off(int byte_off) {
   return byte_off ^ (((byte_off >> 9) & 1) << 5);
 }
 
-// Stage one half-tile (rows h*128..h*128+127 of the K-tile at column k0)
-// into buffer `buf`: 2 x global_load_lds per thread, 16B each.
-TD_DEV void stage_half(const bf16 *g, int ld, int k0, bf16 *lds, int buf,
+// Stage one QUADRANT-union half-tile of the K-tile at column k0 into
+// buffer `buf`. The consumed unit of the gray walk is not a contiguous
+// 128-row block: wave wr reads A rows wr*128 + ih*64 + [0,64), so the
+// A-"half" ih is the union {0-63, 128-191} (ih=0) / {64-127, 192-255}
+// (ih=1) — band 64. B likewise with band 32 (wave wc reads rows
+// wc*64 + jh*32 + [0,32)). Staging exactly these unions is what makes
+// the [A0,B1,A1,B0] overwrite ledger sound (each stage lands one
+// barrier-separated phase after the replaced slot's last read issue).
+//
+// compact row cr (0..127) -> tile row (cr/band)*2*band + h*band +
+// cr%band. Each wave's 64 chunks cover 8 consecutive tile rows (cr
+// stays inside one band), so the wave-uniform LDS base + lane*16B
+// placement of global_load_lds lands every chunk at its natural
+// full-tile linear position; the st_16x32 swizzle is applied on the
+// SOURCE address (involution against the read side, row-preserving).
+template <int BAND>
+TD_DEV void stage_quad(const bf16 *g, int ld, int k0, bf16 *lds, int buf,
                        int h) {
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
 #pragma unroll
   for (int it = 0; it < 2; ++it) {
     int q = it * NTH + tid;  // 1024 chunks of 16B per half-tile
-    int dst_byte = (h * HALF_ELEMS + q * 8) * 2;
-    int un = swz_off(dst_byte);  // logical position for this slot
-    int row = un / (BK * 2);
+    int cr = q >> 3;         // compact row 0..127
+    int row = (cr / BAND) * 2 * BAND + h * BAND + (cr % BAND);
+    int p_byte = (row * BK + (q & 7) * 8) * 2;  // physical LDS byte
+    int un = swz_off(p_byte);                   // logical slot (same row)
     int col = (un % (BK * 2)) / 2;
     const bf16 *src = g + (size_t)row * ld + k0 + col;
-    int wave_chunk0 = it * NTH + wave * 64;  // wave-uniform LDS base
+    int cr0 = it * 64 + wave * 8;  // wave's first compact row
+    int row0 = (cr0 / BAND) * 2 * BAND + h * BAND + (cr0 % BAND);
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) unsigned int *)src,
         (__attribute__((address_space(3))) unsigned int *)(
-            lds + buf * TILE_ELEMS + h * HALF_ELEMS + wave_chunk0 * 8),
+            lds + buf * TILE_ELEMS + row0 * BK),
         16, 0, 0);
   }
 }
@@ -92,19 +108,21 @@ TD_DEV bf16x8 read_frag(const bf16 *lds, int buf, int row, int ks,
                            swz_off(byte_off));
 }
 
-// Per-tile stage order [A0, B1, A1, B0]: half idx -> (matrix, half).
-// Chosen so each stage issue trails the overwritten slot's last ds_read
-// issue by exactly one barrier-separated phase (see file header ledger).
+// Per-tile stage order [A0, B1, A1, B0] (quadrant unions): read issues
+// per window are A0@ph0, B1@ph1, A1@ph2, B0@ph0+ph3 — so each stage
+// trails the replaced slot's last read issue by exactly one
+// barrier-separated phase (A0 staged at +1, B1 at +2, A1 at +3, B0 at
+// +4 relative to the replaced tile's window start).
 TD_DEV void stage_seq(const bf16 *ga, const bf16 *gb, int lda, int ldb,
                       bf16 *lds_a, bf16 *lds_b, int s) {
   const int t = s >> 2;
   const int buf = t & 1;
   const int k0 = t * BK;
   switch (s & 3) {
-    case 0: stage_half(ga, lda, k0, lds_a, buf, 0); break;
-    case 1: stage_half(gb, ldb, k0, lds_b, buf, 1); break;
-    case 2: stage_half(ga, lda, k0, lds_a, buf, 1); break;
-    default: stage_half(gb, ldb, k0, lds_b, buf, 0); break;
+    case 0: stage_quad<64>(ga, lda, k0, lds_a, buf, 0); break;
+    case 1: stage_quad<32>(gb, ldb, k0, lds_b, buf, 1); break;
+    case 2: stage_quad<64>(ga, lda, k0, lds_a, buf, 1); break;
+    default: stage_quad<32>(gb, ldb, k0, lds_b, buf, 0); break;
   }
 }
 

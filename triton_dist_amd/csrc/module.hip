@@ -328,6 +328,49 @@ static void gemm_rs_producer_bf16(uintptr_t a, uintptr_t b, int m, int n,
   TD_CHECK_HIP(hipGetLastError());
 }
 
+static void gemm_ar_producer_bf16(uintptr_t a, uintptr_t b, int m, int n,
+                                  int k, size_t scatter_off,
+                                  size_t arrive_off, size_t out_off,
+                                  size_t oflags_off, int slots, int sk,
+                                  uintptr_t ws, uintptr_t done,
+                                  uintptr_t stream) {
+  check_active();
+  GemmArArgs args;
+  args.g = GemmArgs{reinterpret_cast<void *>(a), reinterpret_cast<void *>(b),
+                    nullptr, nullptr, m, n, k, k, k, n};
+  args.pt = g_heap.pt;
+  args.scatter_off = scatter_off;
+  args.arrive_off = arrive_off;
+  args.out_off = out_off;
+  args.oflags_off = oflags_off;
+  args.slots = slots;
+  if (sk > 1) {
+    launch_gemm256_sk_ar_producer(args, reinterpret_cast<float *>(ws),
+                                  reinterpret_cast<int *>(done), sk,
+                                  as_stream(stream));
+  } else {
+    launch_gemm256_ar_producer(args, as_stream(stream));
+  }
+  TD_CHECK_HIP(hipGetLastError());
+}
+
+static void ar_tile_consumer(int m, int n, size_t scatter_off,
+                             size_t arrive_off, size_t out_off,
+                             size_t oflags_off, int slots, int n_owned,
+                             uintptr_t stream) {
+  check_active();
+  GemmArArgs args;
+  args.g = GemmArgs{nullptr, nullptr, nullptr, nullptr, m, n, 0, 0, 0, n};
+  args.pt = g_heap.pt;
+  args.scatter_off = scatter_off;
+  args.arrive_off = arrive_off;
+  args.out_off = out_off;
+  args.oflags_off = oflags_off;
+  args.slots = slots;
+  launch_ar_tile_consumer(args, n_owned, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
 static void rs_reduce_bf16(uintptr_t segments, uintptr_t out, int world,
                            int rank, int m_per_rank, int ws_stride, int n,
                            uintptr_t stream) {
@@ -756,6 +799,8 @@ PYBIND11_MODULE(_C, m) {
         py::arg("stream"), py::arg("prof_buf") = 0,
         py::arg("prof_cursor") = 0, py::arg("prof_cap") = 0);
   m.def("gemm_rs_producer_bf16", &gemm_rs_producer_bf16);
+  m.def("gemm_ar_producer_bf16", &gemm_ar_producer_bf16);
+  m.def("ar_tile_consumer", &ar_tile_consumer);
   m.def("rs_reduce_bf16", &rs_reduce_bf16);
   m.def("gemm_splitk_bf16", &gemm_splitk_bf16);
   m.def("gemm256_sk_bf16", &gemm256_sk_bf16);
