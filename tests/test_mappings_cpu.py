@@ -304,92 +304,121 @@ def test_flash_decode_pv_output_coverage():
     assert len(seen) == 16 * 128
 
 def test_gemm256_v3_pipeline_ledger():
-    """Discrete-event simulation of the v3 8-phase schedule
-    (csrc/kernels/gemm256_v3.hip): stage sequence [A0,B1,A1,B0] per tile,
-    7-half-tile prologue, one stage per phase, vmcnt(6) drains at
-    half-pair ends. Proves for every ktiles (even, up to 64):
-      1. every ds_read hits a half-tile already staged AND retired by a
-         drain that happened before the read's phase (vmcnt soundness),
-      2. no stage overwrites an LDS slot before the phase AFTER its
-         previous occupant's last ds_read issue (barrier separation),
-      3. register reuse is exact: on phases without a re-read, the held
-         register was loaded from the same (tile, half) the MFMA needs.
+    """Discrete-event simulation of the PIPELINED v3 schedule
+    (csrc/kernels/gemm256_v3.hip): stage order [A0,B1,A1,B0] per tile,
+    7-half-tile prologue, one stage per phase TAIL, next-phase reads
+    pre-issued in the tail, boundary reads at the top, vmcnt(6) drains at
+    the tails of phases 3/7. Events are ordered (phase, seg) with seg
+    0=top, 1=tail; a barrier closes every phase. Proves for every even
+    ktiles up to 64:
+      1. register model: every MFMA phase holds the (tile, half) it
+         needs, loaded by a read that issued at or before its use,
+      2. drain soundness: every read's target half-tile was staged AND
+         retired by a drain whose certifying barrier precedes the read,
+      3. overwrite safety: every stage lands >= 1 barrier after the
+         replaced slot's last read issue — except the one latency-
+         protected class (boundary-top read vs same-phase tail stage,
+         separated by lgkmcnt(0) + 16 MFMAs in every wave), which must
+         be the ONLY violation and only at qq == 0 phases,
+      4. no stage targets a slot that a pre-read in the SAME tail reads.
     """
     for ktiles in [2, 4, 6, 8, 16, 64]:
         total = 4 * ktiles
         pairs = ktiles // 2
 
         def seq(s):
-            # -> (tile, matrix, half) ; matrix 0 = A, 1 = B
+            # -> (tile, matrix, half); matrix 0 = A, 1 = B
             t = s >> 2
             return [(t, 0, 0), (t, 1, 1), (t, 0, 1), (t, 1, 0)][s & 3]
 
-        # stage events: list of (phase_issued, s). Prologue = phase -1.
-        stage_phase = {}
+        # --- build event streams ---------------------------------------
+        stage_ev = {}      # s -> (phase, seg)
         for s in range(min(7, total)):
-            stage_phase[s] = -1
+            stage_ev[s] = (-1, 1)
+        reads = []         # (phase, seg, tile, matrix, half, use_phase)
+        drains = []        # (phase, seg, n_halftiles_allowed_outstanding)
+        issued_at = {(-1, 1): min(7, total)}
+        drains.append((-1, 1, 3 if total > 3 else 0))
         nxt = 7
-        # drains: list of (phase, vmcnt_halftiles) happening at END of
-        # that phase (before the barrier that opens the next phase).
-        drains = [(-1, 3 if total > 3 else 0)]
-        issued_at = {-1: min(7, total)}  # stages issued up to end of phase
-        reads = []  # (phase, tile, matrix, half)
-        regs = {}   # matrix -> (tile, half) currently in registers
+        regs = {}
+        reg_ok = True
         for p in range(pairs):
+            last = (p == pairs - 1)
             for ph in range(8):
                 g = p * 8 + ph
                 tile = 2 * p + (ph >> 2)
                 qq = ph & 3
                 ih = 1 if qq >= 2 else 0
                 jh = 1 if qq in (1, 2) else 0
-                if qq in (0, 2):
-                    reads.append((g, tile, 0, ih))
-                    regs[0] = (tile, ih)
-                if qq != 2:
-                    reads.append((g, tile, 1, jh))
-                    regs[1] = (tile, jh)
-                # MFMA needs (tile, ih) in A regs and (tile, jh) in B regs
-                assert regs[0] == (tile, ih), (ktiles, g)
-                assert regs[1] == (tile, jh), (ktiles, g)
-                if nxt < total:
-                    stage_phase[nxt] = g
-                    nxt += 1
-                if qq == 3 and (p < pairs - 1 or ph == 3):
+                if qq == 0:  # boundary top reads
+                    reads.append((g, 0, tile, 0, 0, g))
+                    reads.append((g, 0, tile, 1, 0, g))
+                    regs[0] = (tile, 0)
+                    regs[1] = (tile, 0)
+                # MFMA needs (tile, ih)/(tile, jh)
+                reg_ok &= regs.get(0) == (tile, ih)
+                reg_ok &= regs.get(1) == (tile, jh)
+                # tail: stage, then pre-reads, then drain
+                if (not last) or ph == 0:
+                    if nxt < total:
+                        stage_ev[nxt] = (g, 1)
+                        nxt += 1
+                if qq == 0:
+                    reads.append((g, 1, tile, 1, 1, g + 1))
+                    regs[1] = (tile, 1)
+                elif qq == 1:
+                    reads.append((g, 1, tile, 0, 1, g + 1))
+                    regs[0] = (tile, 1)
+                elif qq == 2:
+                    reads.append((g, 1, tile, 1, 0, g + 1))
+                    regs[1] = (tile, 0)
+                if qq == 3 and (ph == 3 or not last):
                     n = 0 if nxt >= total else 3
-                    drains.append((g, n))
-                issued_at[g] = nxt
+                    drains.append((g, 1, n))
+                issued_at[(g, 1)] = nxt
+        assert reg_ok
+        assert nxt == total
 
-        assert nxt == total  # staging exactly covers all half-tiles
-
-        # retirement: after drain (phase d, n), stages with index
-        # < issued_at[d] - n are certainly landed.
-        def landed_before(phase):
-            best = 0
-            for d, n in drains:
-                if d < phase:
-                    best = max(best, issued_at[d] - n)
-            return best  # stages [0, best) are retired
-
-        want = {}
+        slot_of = {}
         for s in range(total):
-            want[seq(s)] = s
-        for g, tile, mat, half in reads:
-            s = want[(tile, mat, half)]
-            assert s in stage_phase and stage_phase[s] < g, (ktiles, g)
-            assert s < landed_before(g), \
-                f"ktiles={ktiles} phase {g} reads unretired stage {s}"
+            slot_of[seq(s)] = s
 
-        # slot-overwrite safety: stage of s overwrites slot of s-8 (same
-        # buf/matrix/half two tiles back); must be issued at a phase
-        # strictly after the last read-issue of the s-8 occupant.
+        # --- 2. drain soundness ----------------------------------------
+        def landed_before(ev):
+            # certified-landed stages: drains whose phase closed (barrier)
+            # strictly before ev's phase
+            best = 0
+            for dg, dseg, n in drains:
+                if dg < ev[0]:
+                    best = max(best, issued_at[(dg, dseg)] - n)
+            return best
+
+        for g, seg, tile, mat, half, use in reads:
+            s = slot_of[(tile, mat, half)]
+            sg = stage_ev.get(s)
+            assert sg is not None and sg < (g, seg), (ktiles, g, seg)
+            assert s < landed_before((g, seg)), \
+                f"ktiles={ktiles} read@({g},{seg}) unretired stage {s}"
+
+        # --- 3. overwrite safety + 4. same-tail conflicts ----------------
         last_read = {}
-        for g, tile, mat, half in reads:
-            t, m, h = tile, mat, half
-            last_read[(t, m, h)] = max(last_read.get((t, m, h), -2), g)
+        for g, seg, tile, mat, half, use in reads:
+            key = (tile, mat, half)
+            last_read[key] = max(last_read.get(key, (-2, 0)), (g, seg))
         for s in range(8, total):
             prev = seq(s - 8)
-            assert stage_phase[s] > last_read[prev], \
-                f"ktiles={ktiles} stage {s} overwrites live slot {prev}"
+            sg, sseg = stage_ev[s]
+            lg, lseg = last_read[prev]
+            strictly_after = (sg > lg)
+            same_phase_protected = (sg == lg and lseg == 0 and sseg == 1
+                                    and (sg % 4) == 0)
+            assert strictly_after or same_phase_protected, \
+                f"ktiles={ktiles} stage {s} vs live slot {prev}"
+            # 4: same-tail pre-reads must not touch the staged slot
+            cur = seq(s)
+            for g2, seg2, tile2, mat2, half2, _ in reads:
+                if (g2, seg2) == (sg, 1):
+                    assert (tile2, mat2, half2) != cur, (ktiles, s)
 
 
 @pytest.mark.parametrize("band", [64, 32])

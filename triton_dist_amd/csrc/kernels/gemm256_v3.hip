@@ -1,38 +1,26 @@
-// gemm256_v3 — faithful rebuild of the CDNA4 guide's verified 256^2
-// 8-phase template (BK=64, double-buffered K-tile pair, one C-quadrant x
-// K=64 per phase, ONE half-tile staged per phase, counted vmcnt drains).
+// gemm256_v3 — 256^2 8-phase gray-quadrant GEMM (BK=64, double-buffered
+// K-tile pair), rebuilt from the CDNA4 guide's verified template and then
+// pipelined one step further: ds_reads for phase p+1 issue in phase p's
+// TAIL (after the MFMA cluster), so the LDS-read latency hides under the
+// MFMAs and the per-phase serial window shrinks to s_barrier +
+// s_waitcnt lgkmcnt(0) (nearly satisfied). One barrier per phase.
 //
-// What v2 got wrong (measured 712/737 TF on MI355X vs the template's
-// 1563/1728 and the production ring's 1035/1154):
-//   1. re-read all 12 fragments per phase. The template's gray-coded
-//      quadrant walk (0,0)(0,1)(1,1)(1,0) changes ONE operand half per
-//      phase, so only 4 (B) or 8 (A) ds_read_b128 are issued per phase
-//      and the other operand stays in registers (12 only at a buffer
-//      switch).
-//   2. vmcnt(0) full drains at phase 0/4. The sound counted schedule:
-//      with per-tile stage order [A0, B1, A1, B0] and a 7-half-tile
-//      prologue, the drain needed at the END of each half-pair (phases
-//      3/7, before the barrier that opens the next buffer's reads) is
-//      exactly vmcnt(6) = 3 half-tiles in flight. Ledger: at pair p
-//      phase 3, issued = 8p+11 half-tiles, consumption of tile 2p+1
-//      needs s <= 8p+7 landed -> 3 outstanding. Every slot overwrite
-//      (stage of tile t's half h at global phase 4t+idx(h)-7) lands one
-//      barrier-separated phase after the last ds_read issue of the slot
-//      it replaces (tile t-2's half h at phase 4t-8+idx(h)).
-//   3. sched_barrier(0) order-pinning (the guide measured that class of
-//      pinning as a regression).
-//
-// Phase body (per the guide template):
-//      ds_read the CHANGED operand half  (4/8/12 x ds_read_b128)
-//      stage one half-tile               (2 x global_load_lds, 16B)
-//      [phase 3/7] s_waitcnt vmcnt(6)    (vmcnt(0) on the last pair)
-//      s_barrier
-//      s_waitcnt lgkmcnt(0)
-//      s_setprio(1); 16 x mfma_f32_16x16x32_bf16; s_setprio(0)
-//      s_barrier
-//
-// The kloop is shared by the plain kernel and the fused AG-consumer /
-// RS-producer / split-K variants (same roles as gemm256.hip's ring).
+// Schedule (all positions locked by the CPU ledger test
+// tests/test_mappings_cpu.py::test_gemm256_v3_pipeline_ledger):
+//   * Staged unit = QUADRANT UNION, matching what the gray walk reads:
+//     A-half h = rows {(cr/64)*128 + h*64 + cr%64}, B-half h with band 32.
+//   * Stage order per tile [A0, B1, A1, B0], 7-half-tile prologue; phase
+//     g stages half-tile s = g+7 (variant (ph+3)&3), so each stage issues
+//     >= 1 barrier-separated phase after the replaced slot's last read.
+//   * Counted drains: vmcnt(6) at the tails of phases 3 and 7 (3
+//     half-tiles = 6 global_load_lds in flight); vmcnt(0) only at the
+//     last pair's phase 3 once staging is exhausted.
+//   * Phase body: [boundary only: read A+B at top] -> lgkmcnt(0) ->
+//     16 MFMA -> tail: stage one half-tile, pre-read next phase's
+//     changed operand, [drain], s_barrier.
+//   * All stage addresses are precomputed per thread and advanced by
+//     constants (+BK elems per use; LDS base XOR-toggles the buffer), so
+//     the phase window carries no 64-bit address VALU chains.
 #include <stdexcept>
 
 #include "td/api.hpp"
@@ -48,57 +36,86 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 constexpr int BM = 256, BN = 256, BK = 64;
 constexpr int NTH = 512;                    // 8 waves, 2 (M) x 4 (N)
 constexpr int TILE_ELEMS = BM * BK;         // per matrix per buffer
-constexpr int HALF_ELEMS = TILE_ELEMS / 2;  // rows 0-127 / 128-255
 
 TD_DEV f32x4 mfma16(bf16x8 a, bf16x8 b, f32x4 c) {
   return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
 }
 
-// st_16x32 swizzle on the tile-relative byte offset: XOR bit 9 into bit
-// 5. Applied to the global SOURCE address at stage time and to the
-// ds_read address (involution). Same mapping as v2 (CPU-simulated and
-// GPU-numerics-proven there).
+// st_16x32 swizzle: XOR bit 9 into bit 5 of the tile-relative byte
+// offset (row-preserving involution; source-side at stage, read-side in
+// read_frag).
 TD_DEV int swz_off(int byte_off) {
   return byte_off ^ (((byte_off >> 9) & 1) << 5);
 }
 
-// Stage one QUADRANT-union half-tile of the K-tile at column k0 into
-// buffer `buf`. The consumed unit of the gray walk is not a contiguous
-// 128-row block: wave wr reads A rows wr*128 + ih*64 + [0,64), so the
-// A-"half" ih is the union {0-63, 128-191} (ih=0) / {64-127, 192-255}
-// (ih=1) — band 64. B likewise with band 32 (wave wc reads rows
-// wc*64 + jh*32 + [0,32)). Staging exactly these unions is what makes
-// the [A0,B1,A1,B0] overwrite ledger sound (each stage lands one
-// barrier-separated phase after the replaced slot's last read issue).
-//
-// compact row cr (0..127) -> tile row (cr/band)*2*band + h*band +
-// cr%band. Each wave's 64 chunks cover 8 consecutive tile rows (cr
-// stays inside one band), so the wave-uniform LDS base + lane*16B
-// placement of global_load_lds lands every chunk at its natural
-// full-tile linear position; the st_16x32 swizzle is applied on the
-// SOURCE address (involution against the read side, row-preserving).
-template <int BAND>
-TD_DEV void stage_quad(const bf16 *g, int ld, int k0, bf16 *lds, int buf,
-                       int h) {
+// Per-variant banded row map: compact row cr (0..127) -> tile row.
+TD_DEV int quad_row(int cr, int band, int h) {
+  return (cr / band) * 2 * band + h * band + (cr % band);
+}
+
+// Precomputed stage state. Variants: 0=A h0 (band 64), 1=B h1 (band 32),
+// 2=A h1, 3=B h0. Each holds 2 per-thread global sources (it 0/1) that
+// advance by BK elements per use, a wave-uniform LDS element base, and
+// the buffer parity (XOR-toggled per use).
+struct StageState {
+  const bf16 *a0[2], *b1[2], *a1[2], *b0[2];
+  int da0[2], db1[2], da1[2], db0[2];  // LDS elem offsets (buf 0)
+  int pa0, pb1, pa1, pb0;              // buffer parity of next use
+};
+
+TD_DEV void init_variant(const bf16 *g, int ld, int band, int h,
+                         const bf16 *src[2], int dst[2]) {
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
 #pragma unroll
   for (int it = 0; it < 2; ++it) {
-    int q = it * NTH + tid;  // 1024 chunks of 16B per half-tile
-    int cr = q >> 3;         // compact row 0..127
-    int row = (cr / BAND) * 2 * BAND + h * BAND + (cr % BAND);
-    int p_byte = (row * BK + (q & 7) * 8) * 2;  // physical LDS byte
-    int un = swz_off(p_byte);                   // logical slot (same row)
+    int q = it * NTH + tid;
+    int cr = q >> 3;
+    int row = quad_row(cr, band, h);
+    int p_byte = (row * BK + (q & 7) * 8) * 2;
+    int un = swz_off(p_byte);  // same row, swizzled column chunk
     int col = (un % (BK * 2)) / 2;
-    const bf16 *src = g + (size_t)row * ld + k0 + col;
-    int cr0 = it * 64 + wave * 8;  // wave's first compact row
-    int row0 = (cr0 / BAND) * 2 * BAND + h * BAND + (cr0 % BAND);
-    __builtin_amdgcn_global_load_lds(
-        (const __attribute__((address_space(1))) unsigned int *)src,
-        (__attribute__((address_space(3))) unsigned int *)(
-            lds + buf * TILE_ELEMS + row0 * BK),
-        16, 0, 0);
+    src[it] = g + (size_t)row * ld + col;
+    int cr0 = it * 64 + wave * 8;
+    dst[it] = quad_row(cr0, band, h) * BK;
   }
+}
+
+TD_DEV void init_stage(StageState &st, const bf16 *ga, const bf16 *gb,
+                       int lda, int ldb) {
+  init_variant(ga, lda, 64, 0, st.a0, st.da0);
+  init_variant(gb, ldb, 32, 1, st.b1, st.db1);
+  init_variant(ga, lda, 64, 1, st.a1, st.da1);
+  init_variant(gb, ldb, 32, 0, st.b0, st.db0);
+  st.pa0 = st.pb1 = st.pa1 = st.pb0 = 0;
+}
+
+template <int V>
+TD_DEV void stage_step(StageState &st, bf16 *lds_a, bf16 *lds_b) {
+  const bf16 **src;
+  int *dst;
+  int *par;
+  bf16 *lds;
+  if constexpr (V == 0) {
+    src = st.a0; dst = st.da0; par = &st.pa0; lds = lds_a;
+  } else if constexpr (V == 1) {
+    src = st.b1; dst = st.db1; par = &st.pb1; lds = lds_b;
+  } else if constexpr (V == 2) {
+    src = st.a1; dst = st.da1; par = &st.pa1; lds = lds_a;
+  } else {
+    src = st.b0; dst = st.db0; par = &st.pb0; lds = lds_b;
+  }
+  const int boff = *par * TILE_ELEMS;
+#pragma unroll
+  for (int it = 0; it < 2; ++it) {
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int *)src[it],
+        (__attribute__((address_space(3))) unsigned int *)(
+            lds + boff + dst[it]),
+        16, 0, 0);
+    src[it] += BK;
+  }
+  *par ^= 1;
 }
 
 TD_DEV bf16x8 read_frag(const bf16 *lds, int buf, int row, int ks,
@@ -108,37 +125,33 @@ TD_DEV bf16x8 read_frag(const bf16 *lds, int buf, int row, int ks,
                            swz_off(byte_off));
 }
 
-// Per-tile stage order [A0, B1, A1, B0] (quadrant unions): read issues
-// per window are A0@ph0, B1@ph1, A1@ph2, B0@ph0+ph3 — so each stage
-// trails the replaced slot's last read issue by exactly one
-// barrier-separated phase (A0 staged at +1, B1 at +2, A1 at +3, B0 at
-// +4 relative to the replaced tile's window start).
-TD_DEV void stage_seq(const bf16 *ga, const bf16 *gb, int lda, int ldb,
-                      bf16 *lds_a, bf16 *lds_b, int s) {
-  const int t = s >> 2;
-  const int buf = t & 1;
-  const int k0 = t * BK;
-  switch (s & 3) {
-    case 0: stage_quad<64>(ga, lda, k0, lds_a, buf, 0); break;
-    case 1: stage_quad<32>(gb, ldb, k0, lds_b, buf, 1); break;
-    case 2: stage_quad<64>(ga, lda, k0, lds_a, buf, 1); break;
-    default: stage_quad<32>(gb, ldb, k0, lds_b, buf, 0); break;
+// dispatch helper: variant is compile-time under the unrolled ph loop
+TD_DEV void constexpr_stage(StageState &st, bf16 *lds_a, bf16 *lds_b,
+                            int ph) {
+  switch ((ph + 3) & 3) {
+    case 0: stage_step<0>(st, lds_a, lds_b); break;
+    case 1: stage_step<1>(st, lds_a, lds_b); break;
+    case 2: stage_step<2>(st, lds_a, lds_b); break;
+    default: stage_step<3>(st, lds_a, lds_b); break;
   }
 }
 
-// The 8-phase pipelined K loop. ktiles must be even (k % 128 == 0).
+// The pipelined 8-phase K loop. ktiles must be even (k % 128 == 0).
 TD_DEV void kloop(const bf16 *ga, const bf16 *gb, int lda, int ldb,
                   int ktiles, bf16 *lds_a, bf16 *lds_b, int wr, int wc,
                   int lane, f32x4 acc[8][4]) {
-  const int total_halves = 4 * ktiles;
   const int pairs = ktiles / 2;
-
-  // prologue: 7 half-tiles (tile 0 complete + tile 1 [A0,B1,A1]), then
-  // drain so tile 0 is resident: 3 half-tiles (6 loads) may stay in
-  // flight.
-  for (int s = 0; s < 7 && s < total_halves; ++s)
-    stage_seq(ga, gb, lda, ldb, lds_a, lds_b, s);
-  if (total_halves > 3) {
+  StageState st;
+  init_stage(st, ga, gb, lda, ldb);
+  // prologue: 7 half-tiles (tile 0 complete + tile 1 [A0,B1,A1])
+  stage_step<0>(st, lds_a, lds_b);
+  stage_step<1>(st, lds_a, lds_b);
+  stage_step<2>(st, lds_a, lds_b);
+  stage_step<3>(st, lds_a, lds_b);
+  stage_step<0>(st, lds_a, lds_b);
+  stage_step<1>(st, lds_a, lds_b);
+  stage_step<2>(st, lds_a, lds_b);
+  if (ktiles > 1) {
     asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
   } else {
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -146,58 +159,41 @@ TD_DEV void kloop(const bf16 *ga, const bf16 *gb, int lda, int ldb,
   __builtin_amdgcn_s_barrier();
 
   bf16x8 af[4][2], bfr[2][2];
-  int s = 7;  // next half-tile to stage
+
+  auto read_a = [&](int rbuf, int ih) {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int row = wr * 128 + (ih * 4 + i) * 16 + (lane & 15);
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks)
+        af[i][ks] = read_frag(lds_a, rbuf, row, ks, lane);
+    }
+  };
+  auto read_b = [&](int rbuf, int jh) {
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      int row = wc * 64 + (jh * 2 + j) * 16 + (lane & 15);
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks)
+        bfr[j][ks] = read_frag(lds_b, rbuf, row, ks, lane);
+    }
+  };
 
   for (int p = 0; p < pairs; ++p) {
+    const bool last = (p == pairs - 1);
 #pragma unroll
     for (int ph = 0; ph < 8; ++ph) {
-      const int rbuf = (2 * p + (ph >> 2)) & 1;
+      const int rbuf = ph >> 2;  // tile 2p is buf 0, 2p+1 is buf 1
       const int qq = ph & 3;
-      // gray order (ih, jh): (0,0) (0,1) (1,1) (1,0)
       const int ih = (qq >= 2) ? 1 : 0;
       const int jh = (qq == 1 || qq == 2) ? 1 : 0;
 
-      // ds_read only the operand half that changed this phase:
-      // A at qq 0 (buffer switch) and 2; B at qq 0 (switch), 1 and 3.
-      const bool new_a = (qq == 0 || qq == 2);
-      const bool new_b = (qq != 2);
-      if (new_a) {
-#pragma unroll
-        for (int i = 0; i < 4; ++i) {
-          int row = wr * 128 + (ih * 4 + i) * 16 + (lane & 15);
-#pragma unroll
-          for (int ks = 0; ks < 2; ++ks)
-            af[i][ks] = read_frag(lds_a, rbuf, row, ks, lane);
-        }
+      if (qq == 0) {
+        // boundary: fresh buffer — read both operands at the top (the
+        // previous tail's drain + this barrier certify residency)
+        read_a(rbuf, 0);
+        read_b(rbuf, 0);
       }
-      if (new_b) {
-#pragma unroll
-        for (int j = 0; j < 2; ++j) {
-          int row = wc * 64 + (jh * 2 + j) * 16 + (lane & 15);
-#pragma unroll
-          for (int ks = 0; ks < 2; ++ks)
-            bfr[j][ks] = read_frag(lds_b, rbuf, row, ks, lane);
-        }
-      }
-
-      // stage the next half-tile in sequence
-      if (s < total_halves) {
-        stage_seq(ga, gb, lda, ldb, lds_a, lds_b, s);
-        ++s;
-      }
-
-      // counted drain at half-pair ends: certifies the buffer the NEXT
-      // half-pair reads. vmcnt(6) steady state; vmcnt(0) once staging
-      // is exhausted (last pair's phase 3).
-      if (qq == 3 && (p < pairs - 1 || ph == 3)) {
-        if (s >= total_halves) {
-          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-        } else {
-          asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
-        }
-      }
-
-      __builtin_amdgcn_s_barrier();
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -209,6 +205,26 @@ TD_DEV void kloop(const bf16 *ga, const bf16 *gb, int lda, int ldb,
             acc[ih * 4 + i][jh * 2 + j] = mfma16(
                 af[i][ks], bfr[j][ks], acc[ih * 4 + i][jh * 2 + j]);
       __builtin_amdgcn_s_setprio(0);
+
+      // tail: stage half-tile s = 8p+ph+7 (variant (ph+3)&3), pre-read
+      // the NEXT phase's changed operand, drain at half-pair ends.
+      if (!last || ph == 0) {
+        constexpr_stage(st, lds_a, lds_b, ph);
+      }
+      if (qq == 0) {
+        read_b(rbuf, 1);        // phase +1 consumes B1
+      } else if (qq == 1) {
+        read_a(rbuf, 1);        // phase +2 consumes A1
+      } else if (qq == 2) {
+        read_b(rbuf, 0);        // phase +3 re-consumes B0
+      }
+      if (qq == 3 && (ph == 3 || !last)) {
+        if (last && ph == 3) {
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        } else {
+          asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+        }
+      }
       __builtin_amdgcn_s_barrier();
     }
   }

@@ -194,23 +194,30 @@ void launch_moe_router(const void *logits, void *topk_ids, void *topk_w,
 // into the owner's expert-sorted recv buffer, writes (src, tok_k) meta, and
 // signals the destination when the LAST of my copies to it lands.
 // ---------------------------------------------------------------------------
+constexpr int kMaxExperts = 256;  // shared-counter bound (DeepSeek-V3 = 256)
+
 __global__ void k_moe_dispatch(PeerTable pt, const bf16 *__restrict__ x,
                                const int *__restrict__ topk_ids,
                                const int *__restrict__ send_pos,
                                const int *__restrict__ send_base,
-                               const int *__restrict__ send_to_dst,
+                               const int *__restrict__ counts,
                                size_t recv_x_off, size_t meta_off,
-                               size_t flags_off, unsigned *arrive,
+                               size_t eflags_off, unsigned *arrive_e,
                                const int *val_cell, int T, int K, int H,
-                               int e_loc) {
-  // Grid-stride over copy indices with PER-BLOCK arrive batching: one
-  // acq_rel atomic per (block, dst) instead of one per row — T*K serialized
-  // RMWs on a single counter measured ~100 us/call at T*K=4096. Only the
-  // LAST arrival pays the system-release (L2 writeback) before publishing
-  // the flag; every earlier block's writes are ordered by its acq_rel
+                               int e_loc, int e_num) {
+  // Grid-stride over copy indices with PER-BLOCK PER-EXPERT arrive
+  // batching: one acq_rel atomic per (block, expert) — same cost class
+  // as the old per-dst scheme, but the completion signal is now
+  // per-(src, expert): eflags[src * e_loc + el] on the owner. The
+  // grouped GEMM gates each (expert, tile) work item on ITS expert's
+  // source flags, so FFN for early-complete experts starts while slow
+  // sources still stream (per-expert overlap; reference capability
+  // kernels/amd/ep_all2all_fused.py:316 — behavior only). Only the LAST
+  // arrival for an expert pays the system-release before publishing;
+  // every earlier block's writes are ordered by its acq_rel
   // device-scope arrive.
-  __shared__ int cnt[kMaxRanks];
-  if (threadIdx.x < kMaxRanks) cnt[threadIdx.x] = 0;
+  __shared__ int cnt[kMaxExperts];
+  for (int e = threadIdx.x; e < e_num; e += blockDim.x) cnt[e] = 0;
   __syncthreads();
   for (int i = blockIdx.x; i < T * K; i += gridDim.x) {
     const int e = topk_ids[i];
@@ -228,22 +235,21 @@ __global__ void k_moe_dispatch(PeerTable pt, const bf16 *__restrict__ x,
       int *meta = (int *)((char *)pt.bases[dst] + meta_off);
       meta[slot * 2] = pt.rank;
       meta[slot * 2 + 1] = i;  // t*K+k
-      ++cnt[dst];
+      ++cnt[e];
     }
   }
   __threadfence_block();
   __syncthreads();
-  if (threadIdx.x == 0) {
-#pragma unroll
-    for (int dst = 0; dst < kMaxRanks; ++dst) {
-      int c = cnt[dst];
-      if (dst >= pt.world || c == 0) continue;
-      unsigned prev = atomic_add<Scope::Gpu>(arrive + dst, (unsigned)c);
-      if ((int)prev + c == send_to_dst[dst]) {
-        fence_release_sys();
-        int *fl = (int *)((char *)pt.bases[dst] + flags_off);
-        st_release<Scope::Sys>(fl + pt.rank, val_cell ? *val_cell : 1);
-      }
+  for (int e = threadIdx.x; e < e_num; e += blockDim.x) {
+    int c = cnt[e];
+    if (c == 0) continue;
+    unsigned prev = atomic_add<Scope::Gpu>(arrive_e + e, (unsigned)c);
+    if ((int)prev + c == counts[e]) {
+      const int dst = e / e_loc;
+      fence_release_sys();
+      int *fl = (int *)((char *)pt.bases[dst] + eflags_off);
+      st_release<Scope::Sys>(fl + pt.rank * e_loc + e % e_loc,
+                             val_cell ? *val_cell : 1);
     }
   }
 }
@@ -350,6 +356,23 @@ __global__ void k_moe_signal_empty(PeerTable pt, const int *__restrict__ cnt,
   if (cnt[d] == 0) {
     int *fl = (int *)((char *)pt.bases[d] + flags_off);
     st_release<Scope::Sys>(fl + pt.rank, val_cell ? *val_cell : 1);
+  }
+}
+
+// per-expert empty signaling: experts I send NOTHING to still need my
+// flag on their owner (the GEMM gate waits all world sources per expert)
+__global__ void k_moe_signal_empty_e(PeerTable pt,
+                                     const int *__restrict__ counts,
+                                     size_t eflags_off,
+                                     const int *__restrict__ val_cell,
+                                     int e_loc, int e_num) {
+  for (int e = blockIdx.x * blockDim.x + threadIdx.x; e < e_num;
+       e += gridDim.x * blockDim.x) {
+    if (counts[e] != 0) continue;
+    const int dst = e / e_loc;
+    int *fl = (int *)((char *)pt.bases[dst] + eflags_off);
+    st_release<Scope::Sys>(fl + pt.rank * e_loc + e % e_loc,
+                           val_cell ? *val_cell : 1);
   }
 }
 
