@@ -174,13 +174,17 @@ void launch_moe_grouped_gemm_pq(const void *xin, const void *weights,
                                 const void *expert_rows,
                                 const void *work_items,
                                 const void *work_count, int n, int k,
-                                hipStream_t stream);
+                                hipStream_t stream,
+                                const void *eflags = nullptr,
+                                const void *val_cell = nullptr,
+                                int world = 0, int e_loc = 0);
 void launch_moe_dispatch(const PeerTable &pt, const void *x,
                          const void *topk_ids, const void *send_pos,
-                         const void *send_base, const void *send_to_dst,
-                         size_t recv_x_off, size_t meta_off, size_t flags_off,
-                         unsigned *arrive, const void *val_cell, int T,
-                         int K, int H, int e_loc, hipStream_t stream);
+                         const void *send_base, const void *counts,
+                         size_t recv_x_off, size_t meta_off,
+                         size_t eflags_off, unsigned *arrive_e,
+                         const void *val_cell, int T, int K, int H,
+                         int e_loc, int e_num, hipStream_t stream);
 void launch_moe_wait_flags(const void *flags, int world, const void *cell,
                            hipStream_t stream);
 void launch_bump_cell(void *cell, hipStream_t stream);
@@ -202,7 +206,9 @@ void launch_moe_grouped_gemm(const void *xin, const void *weights, void *out,
                              const void *expert_base, const void *expert_rows,
                              int e_loc, int cap_tiles_m, int n, int k,
                              int cap_rows, hipStream_t stream,
-                             bool small_m = false);
+                             bool small_m = false,
+                             const void *eflags = nullptr,
+                             const void *val_cell = nullptr, int world = 0);
 void launch_moe_combine_send(const PeerTable &pt, const void *expert_out,
                              const void *meta, const void *recv_total,
                              const void *recv_from_src, size_t combine_off,

@@ -418,11 +418,21 @@ __global__ __launch_bounds__(gg::NTH) void k_moe_grouped_gemm(
     const bf16 *__restrict__ xin, const bf16 *__restrict__ weights,
     bf16 *__restrict__ out, const int *__restrict__ expert_base,
     const int *__restrict__ expert_rows, int cap_tiles_m, int n, int k,
-    int cap_rows) {
+    int cap_rows, const int *__restrict__ eflags, const int *val_cell,
+    int world, int e_loc) {
   const int tiles_n = n / gg::BN;
   const int e = blockIdx.x / cap_tiles_m;
   const int tm = blockIdx.x % cap_tiles_m;
   const int tn = blockIdx.y;
+  // per-expert gate: wait THIS expert's world source flags (acquire),
+  // so tiles of early-complete experts compute while slow sources of
+  // other experts still stream
+  if (eflags) {
+    int v = val_cell ? *val_cell : 1;
+    if (threadIdx.x < (unsigned)world)
+      wait_ge_one<Scope::Sys>(eflags + threadIdx.x * e_loc + e, v);
+    __syncthreads();
+  }
   const int rows = expert_rows[e];
   if (tm * gg::BM >= rows) return;
   const int base = expert_base[e];
@@ -609,7 +619,9 @@ __global__ __launch_bounds__(256) void k_moe_grouped_gemm_pq(
     const bf16 *__restrict__ xin, const bf16 *__restrict__ weights,
     bf16 *__restrict__ out, const int *__restrict__ expert_base,
     const int *__restrict__ expert_rows, const int *__restrict__ work_items,
-    const int *__restrict__ work_count, int n, int k) {
+    const int *__restrict__ work_count, int n, int k,
+    const int *__restrict__ eflags, const int *val_cell, int world,
+    int e_loc) {
   constexpr int BM = 32, BN = 128, BK = 64;
   constexpr int ABUF = BM * BK, BBUF = BN * BK;
   __shared__ bf16 lds_a[3 * ABUF];
@@ -618,12 +630,19 @@ __global__ __launch_bounds__(256) void k_moe_grouped_gemm_pq(
   const int wave = tid >> 6, lane = tid & 63;
   const int tiles_n = n / BN;
   const int total = work_count[0] * tiles_n;
+  const int gate_v = (eflags && val_cell) ? *val_cell : 1;
 
   for (int wi = blockIdx.x; wi < total; wi += gridDim.x) {
     const int item = work_items[wi / tiles_n];
     const int tn = wi % tiles_n;
     const int e = item >> 16;
     const int tm = item & 0xFFFF;
+    if (eflags) {
+      // gate this work item on ITS expert's world source flags
+      if (tid < world)
+        wait_ge_one<Scope::Sys>(eflags + tid * e_loc + e, gate_v);
+      __syncthreads();
+    }
     const int rows = expert_rows[e];
     const int base = expert_base[e];
     f32x4 acc[2][2] = {};
@@ -811,21 +830,24 @@ void launch_moe_layout(const void *all_splits, int rank, int world,
 
 void launch_moe_dispatch(const PeerTable &pt, const void *x,
                          const void *topk_ids, const void *send_pos,
-                         const void *send_base, const void *send_to_dst,
-                         size_t recv_x_off, size_t meta_off, size_t flags_off,
-                         unsigned *arrive, const void *val_cell, int T,
-                         int K, int H, int e_loc, hipStream_t stream) {
+                         const void *send_base, const void *counts,
+                         size_t recv_x_off, size_t meta_off,
+                         size_t eflags_off, unsigned *arrive_e,
+                         const void *val_cell, int T, int K, int H,
+                         int e_loc, int e_num, hipStream_t stream) {
   if (H % 8) throw std::runtime_error("moe dispatch: H % 8 != 0");
+  if (e_num > kMaxExperts)
+    throw std::runtime_error("moe dispatch: e_num > 256");
   hipLaunchKernelGGL(k_moe_dispatch, dim3(T * K < 512 ? T * K : 512),
                      dim3(256), 0, stream, pt,
                      (const bf16 *)x, (const int *)topk_ids,
                      (const int *)send_pos, (const int *)send_base,
-                     (const int *)send_to_dst, recv_x_off, meta_off,
-                     flags_off, arrive, (const int *)val_cell, T, K, H,
-                     e_loc);
-  hipLaunchKernelGGL(k_moe_signal_empty, dim3(1), dim3(kWave), 0, stream, pt,
-                     (const int *)send_to_dst, flags_off,
-                     (const int *)val_cell);
+                     (const int *)counts, recv_x_off, meta_off,
+                     eflags_off, arrive_e, (const int *)val_cell, T, K, H,
+                     e_loc, e_num);
+  hipLaunchKernelGGL(k_moe_signal_empty_e, dim3(1), dim3(256), 0, stream,
+                     pt, (const int *)counts, eflags_off,
+                     (const int *)val_cell, e_loc, e_num);
 }
 
 void launch_moe_dispatch_fp8(const PeerTable &pt, const void *x,
@@ -888,29 +910,35 @@ void launch_moe_grouped_gemm_pq(const void *xin, const void *weights,
                                 const void *expert_rows,
                                 const void *work_items,
                                 const void *work_count, int n, int k,
-                                hipStream_t stream) {
+                                hipStream_t stream, const void *eflags,
+                                const void *val_cell, int world,
+                                int e_loc) {
   if (n % 128 || k % 64)
     throw std::runtime_error("grouped gemm pq: N%128/K%64 required");
   hipLaunchKernelGGL(k_moe_grouped_gemm_pq, dim3(1024), dim3(256), 0,
                      stream, (const bf16 *)xin, (const bf16 *)weights,
                      (bf16 *)out, (const int *)expert_base,
                      (const int *)expert_rows, (const int *)work_items,
-                     (const int *)work_count, n, k);
+                     (const int *)work_count, n, k, (const int *)eflags,
+                     (const int *)val_cell, world, e_loc);
 }
 
 void launch_moe_grouped_gemm(const void *xin, const void *weights, void *out,
                              const void *expert_base, const void *expert_rows,
                              int e_loc, int cap_tiles_m, int n, int k,
                              int cap_rows, hipStream_t stream,
-                             bool small_m) {
+                             bool small_m, const void *eflags,
+                             const void *val_cell, int world) {
   if (n % 128 || k % 64)
     throw std::runtime_error("grouped gemm: N%128/K%64 required");
-  (void)cap_tiles_m;
+  (void)small_m;
   hipLaunchKernelGGL(k_moe_grouped_gemm,
                      dim3(e_loc * cap_tiles_m, n / 128), dim3(gg::NTH), 0,
                      stream, (const bf16 *)xin, (const bf16 *)weights,
                      (bf16 *)out, (const int *)expert_base,
-                     (const int *)expert_rows, cap_tiles_m, n, k, cap_rows);
+                     (const int *)expert_rows, cap_tiles_m, n, k, cap_rows,
+                     (const int *)eflags, (const int *)val_cell, world,
+                     e_loc);
 }
 
 void launch_moe_combine_send(const PeerTable &pt, const void *expert_out,

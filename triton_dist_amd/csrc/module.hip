@@ -582,26 +582,29 @@ static void moe_grouped_gemm_pq(uintptr_t xin, uintptr_t weights,
                                 uintptr_t out, uintptr_t expert_base,
                                 uintptr_t expert_rows, uintptr_t work_items,
                                 uintptr_t work_count, int n, int k,
-                                uintptr_t stream) {
+                                uintptr_t stream, uintptr_t eflags = 0,
+                                uintptr_t val_cell = 0, int world = 0,
+                                int e_loc = 0) {
   launch_moe_grouped_gemm_pq((void *)xin, (void *)weights, (void *)out,
                              (void *)expert_base, (void *)expert_rows,
                              (void *)work_items, (void *)work_count, n, k,
-                             as_stream(stream));
+                             as_stream(stream), (void *)eflags,
+                             (void *)val_cell, world, e_loc);
   TD_CHECK_HIP(hipGetLastError());
 }
 
 static void moe_dispatch(uintptr_t x, uintptr_t topk_ids, uintptr_t send_pos,
-                         uintptr_t send_base, uintptr_t send_to_dst,
+                         uintptr_t send_base, uintptr_t counts,
                          size_t recv_x_off, size_t meta_off,
-                         size_t flags_off, uintptr_t arrive, int T, int K,
-                         int H, int e_loc, uintptr_t stream,
-                         uintptr_t val_cell = 0) {
+                         size_t eflags_off, uintptr_t arrive_e, int T,
+                         int K, int H, int e_loc, int e_num,
+                         uintptr_t stream, uintptr_t val_cell = 0) {
   check_active();
   launch_moe_dispatch(g_heap.pt, (void *)x, (void *)topk_ids,
                       (void *)send_pos, (void *)send_base,
-                      (void *)send_to_dst, recv_x_off, meta_off, flags_off,
-                      (unsigned *)arrive, (void *)val_cell, T, K, H, e_loc,
-                      as_stream(stream));
+                      (void *)counts, recv_x_off, meta_off, eflags_off,
+                      (unsigned *)arrive_e, (void *)val_cell, T, K, H,
+                      e_loc, e_num, as_stream(stream));
   TD_CHECK_HIP(hipGetLastError());
 }
 
@@ -661,11 +664,12 @@ static void moe_grouped_gemm(uintptr_t xin, uintptr_t weights, uintptr_t out,
                              uintptr_t expert_base, uintptr_t expert_rows,
                              int e_loc, int cap_tiles_m, int n, int k,
                              int cap_rows, uintptr_t stream,
-                             bool small_m = false) {
+                             bool small_m = false, uintptr_t eflags = 0,
+                             uintptr_t val_cell = 0, int world = 0) {
   launch_moe_grouped_gemm((void *)xin, (void *)weights, (void *)out,
                           (void *)expert_base, (void *)expert_rows, e_loc,
                           cap_tiles_m, n, k, cap_rows, as_stream(stream),
-                          small_m);
+                          small_m, (void *)eflags, (void *)val_cell, world);
   TD_CHECK_HIP(hipGetLastError());
 }
 
@@ -824,12 +828,18 @@ PYBIND11_MODULE(_C, m) {
         py::arg("recv_from_src"), py::arg("recv_total"), py::arg("stream"),
         py::arg("work_items") = 0, py::arg("work_count") = 0,
         py::arg("bm") = 128);
-  m.def("moe_grouped_gemm_pq", &moe_grouped_gemm_pq);
+  m.def("moe_grouped_gemm_pq", &moe_grouped_gemm_pq, py::arg("xin"),
+        py::arg("weights"), py::arg("out"), py::arg("expert_base"),
+        py::arg("expert_rows"), py::arg("work_items"),
+        py::arg("work_count"), py::arg("n"), py::arg("k"),
+        py::arg("stream"), py::arg("eflags") = 0, py::arg("val_cell") = 0,
+        py::arg("world") = 0, py::arg("e_loc") = 0);
   m.def("moe_dispatch", &moe_dispatch, py::arg("x"), py::arg("topk_ids"),
-        py::arg("send_pos"), py::arg("send_base"), py::arg("send_to_dst"),
-        py::arg("recv_x_off"), py::arg("meta_off"), py::arg("flags_off"),
-        py::arg("arrive"), py::arg("T"), py::arg("K"), py::arg("H"),
-        py::arg("e_loc"), py::arg("stream"), py::arg("val_cell") = 0);
+        py::arg("send_pos"), py::arg("send_base"), py::arg("counts"),
+        py::arg("recv_x_off"), py::arg("meta_off"), py::arg("eflags_off"),
+        py::arg("arrive_e"), py::arg("T"), py::arg("K"), py::arg("H"),
+        py::arg("e_loc"), py::arg("e_num"), py::arg("stream"),
+        py::arg("val_cell") = 0);
   m.def("moe_dispatch_fp8", &moe_dispatch_fp8, py::arg("x"),
         py::arg("topk_ids"), py::arg("send_pos"), py::arg("send_base"),
         py::arg("send_to_dst"), py::arg("recv_q_off"), py::arg("recv_s_off"),
@@ -846,7 +856,8 @@ PYBIND11_MODULE(_C, m) {
         py::arg("weights"), py::arg("out"), py::arg("expert_base"),
         py::arg("expert_rows"), py::arg("e_loc"), py::arg("cap_tiles_m"),
         py::arg("n"), py::arg("k"), py::arg("cap_rows"), py::arg("stream"),
-        py::arg("small_m") = false);
+        py::arg("small_m") = false, py::arg("eflags") = 0,
+        py::arg("val_cell") = 0, py::arg("world") = 0);
   m.def("moe_combine_send", &moe_combine_send, py::arg("expert_out"),
         py::arg("meta"), py::arg("recv_total"), py::arg("recv_from_src"),
         py::arg("combine_off"), py::arg("cflags_off"), py::arg("arrive"),

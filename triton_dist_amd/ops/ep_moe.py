@@ -33,8 +33,9 @@ class EPContext:
     recv_x: SymmBuffer          # [cap + 128, H] bf16 (tail slack for GEMM)
     meta: SymmBuffer            # [cap, 2] int32 (src, tok_k)
     combine_buf: SymmBuffer     # [max_tokens * topk, H] bf16
-    disp_flags: SymmBuffer      # [world] int32
+    disp_flags: SymmBuffer      # [world] int32 (fp8 path: per-dst)
     comb_flags: SymmBuffer      # [world] int32
+    eflags: Optional[SymmBuffer] = None  # [world, e_loc] int32 per-expert
     # local device state
     local: dict = field(default_factory=dict)
     epoch: int = 0
@@ -84,6 +85,8 @@ def create_ep_context(max_tokens: int, hidden: int, n_experts: int,
         disp_flags=heap.alloc_buffer((nbuf, world), torch.int32),
         comb_flags=heap.alloc_buffer((nbuf, world), torch.int32),
     )
+    ctx.eflags = heap.alloc_buffer((nbuf, world, n_experts // world),
+                                   torch.int32)
     ctx.low_latency = low_latency
     ctx.credit_flags = heap.alloc_buffer((world,), torch.int32) \
         if low_latency else None
@@ -108,12 +111,16 @@ def create_ep_context(max_tokens: int, hidden: int, n_experts: int,
             recv_from_src=torch.zeros(world, dtype=torch.int32, device=dev),
             recv_total=torch.zeros(1, dtype=torch.int32, device=dev),
             arrive_d=torch.zeros(world, dtype=torch.int32, device=dev),
+            arrive_e=torch.zeros(e, dtype=torch.int32, device=dev),
             arrive_c=torch.zeros(world, dtype=torch.int32, device=dev),
             call_cell=torch.zeros(1, dtype=torch.int32, device=dev),
             work_items=torch.zeros(ctx.e_loc + cap // 32 + 1,
                                    dtype=torch.int32, device=dev),
             work_count=torch.zeros(1, dtype=torch.int32, device=dev),
         )
+        ctx.local["comm_stream"] = torch.cuda.Stream()
+        ctx.local["ev_fork"] = torch.cuda.Event()
+        ctx.local["ev_join"] = torch.cuda.Event()
     return ctx
 
 
@@ -155,11 +162,13 @@ def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
     combine_off = ctx.combine_buf.offset + parity * ctx.max_tokens * K * H * 2
     dflags_off = ctx.disp_flags.offset + parity * world * 4
     cflags_off = ctx.comb_flags.offset + parity * world * 4
+    eflags_off = ctx.eflags.offset + parity * world * e_loc * 4
     cell = 0
 
     # phase 0: buffer-reuse protection
     L["counts"].zero_()
     L["arrive_d"].zero_()
+    L["arrive_e"].zero_()
     L["arrive_c"].zero_()
     if ctx.low_latency:
         # LL protocol: no barrier, no flag resets — flags carry the call
@@ -171,6 +180,7 @@ def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
         _C.reset_flags(heap.ptr(rank, sflags_off), world, 0, s)
         _C.reset_flags(heap.ptr(rank, dflags_off), world, 0, s)
         _C.reset_flags(heap.ptr(rank, cflags_off), world, 0, s)
+        _C.reset_flags(heap.ptr(rank, eflags_off), world * e_loc, 0, s)
         heap.barrier_all_on_stream(stream)
 
     # phase 1: routing histogram + slot assignment
@@ -226,12 +236,21 @@ def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
                        heap.ptr(rank, recv_x_off),
                        L["recv_total"].data_ptr(), ctx.cap, H, s)
     else:
+        # dispatch rides the comm stream; the expert GEMM is gated
+        # per-(expert, tile) on eflags, so FFN tiles of early-complete
+        # experts run while slow sources still stream (per-expert
+        # overlap; reference kernels/amd/ep_all2all_fused.py:316
+        # capability — behavior only)
+        comm = L["comm_stream"]
+        L["ev_fork"].record(stream)
+        comm.wait_event(L["ev_fork"])
         _C.moe_dispatch(x.data_ptr(), topk_ids.data_ptr(),
                         L["send_pos"].data_ptr(), L["send_base"].data_ptr(),
-                        L["send_to_dst"].data_ptr(), recv_x_off,
-                        meta_off, dflags_off,
-                        L["arrive_d"].data_ptr(), T, K, H, e_loc, s, cell)
-        _C.moe_wait_flags(heap.ptr(rank, dflags_off), world, s, cell)
+                        L["counts"].data_ptr(), recv_x_off,
+                        meta_off, eflags_off,
+                        L["arrive_e"].data_ptr(), T, K, H, e_loc, E,
+                        comm.cuda_stream, cell)
+        L["ev_join"].record(comm)
 
     # phase 5: grouped expert FFN — persistent work-queue kernel when
     # experts are lightly loaded (decode: ~T*K*world/E rows per expert)
@@ -239,18 +258,21 @@ def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
     # +128-row slack everywhere an edge GEMM tile may over-read
     expert_h = torch.empty(ctx.cap + 128, 2 * inter, dtype=torch.bfloat16,
                            device=x.device)
+    gate = 0 if ctx.fp8 else heap.ptr(rank, eflags_off)
     if small_m:
         _C.moe_grouped_gemm_pq(heap.ptr(rank, recv_x_off),
                                w_gate_up.data_ptr(), expert_h.data_ptr(),
                                L["expert_base"].data_ptr(),
                                L["expert_rows"].data_ptr(),
                                L["work_items"].data_ptr(),
-                               L["work_count"].data_ptr(), 2 * inter, H, s)
+                               L["work_count"].data_ptr(), 2 * inter, H, s,
+                               gate, cell, world, e_loc)
     else:
         _C.moe_grouped_gemm(heap.ptr(rank, recv_x_off), w_gate_up.data_ptr(),
                             expert_h.data_ptr(), L["expert_base"].data_ptr(),
                             L["expert_rows"].data_ptr(), e_loc, cap_tiles,
-                            2 * inter, H, ctx.cap, s, False)
+                            2 * inter, H, ctx.cap, s, False, gate, cell,
+                            world)
     act = torch.empty(ctx.cap + 128, inter, dtype=torch.bfloat16,
                       device=x.device)
     _C.swiglu(expert_h.data_ptr(), act.data_ptr(), ctx.cap + 128, inter, s)
@@ -269,7 +291,10 @@ def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
                             L["expert_rows"].data_ptr(), e_loc, cap_tiles, H,
                             inter, ctx.cap, s, False)
 
-    # phase 6: combine (return rows + weighted reduce)
+    # phase 6: combine (return rows + weighted reduce); join the
+    # dispatch fork first (graph hygiene — by now it long completed)
+    if not ctx.fp8:
+        stream.wait_event(L["ev_join"])
     _C.moe_combine_send(expert_out.data_ptr(), heap.ptr(rank, meta_off),
                         L["recv_total"].data_ptr(),
                         L["recv_from_src"].data_ptr(),
