@@ -243,6 +243,10 @@ void launch_qkv_prologue_decode(const void *qkv, void *q_out, void *kcache,
                                 hipStream_t stream);
 
 // kernels/attention.hip ------------------------------------------------------
+void launch_flash_prefill(const void *q, const void *k, const void *v,
+                          void *out, void *lse, int b, int s, int qh,
+                          int kvh, float scale, bool causal,
+                          hipStream_t stream);
 void launch_flash_decode(const void *q, const void *kcache,
                          const void *vcache, void *out, const void *offset,
                          int batch, int qh, int kvh, int max_len,

@@ -566,3 +566,200 @@ void launch_lse_combine(const void *parts, const void *lses, void *out,
 }
 
 }  // namespace td
+
+namespace td {
+
+// ---------------------------------------------------------------------------
+// Prefill flash-attention (FA2 forward, causal, GQA) — replaces torch sdpa
+// on the prefill hot path (capability parity with the reference's own
+// FA2-style consumer, kernels/nvidia/sp_ag_attention_intra_node.py:257-428
+// — behavior only).
+//
+//   grid = (ceil(s/128), qh, b); block = 256 threads (4 waves).
+//   Q tile 128 rows (wave w owns rows w*32..w*32+31), KV tiles of 32.
+//   Per kv tile: S = Q K^T on MFMA (Q fragments hoisted from global into
+//   registers once), online softmax with per-lane row state (each lane
+//   tracks its 8 C-layout rows), P staged to LDS as the MFMA A operand,
+//   PV via ds_read_b64_tr_b16 column-major V fragments from row-major
+//   LDS (same measured semantics as the decode kernel above).
+//   Causal mask: kv_pos > q_pos -> -inf (tiles fully above the diagonal
+//   are never visited).
+//
+// Layouts: q [b, s, qh, 128], k/v [b, s, kvh, 128], out [b, s, qh, 128]
+// — the layer's natural post-RoPE layout; no host-side transposes.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_flash_prefill(
+    const bf16 *__restrict__ q, const bf16 *__restrict__ k,
+    const bf16 *__restrict__ v, bf16 *__restrict__ out,
+    float *__restrict__ lse, int s, int qh, int kvh, float scale,
+    int causal) {
+  const int q0 = blockIdx.x * 128;
+  const int h = blockIdx.y;
+  const int b = blockIdx.z;
+  const int G = qh / kvh;
+  const int kh = h / G;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+
+  __shared__ bf16 k_lds[kTile][kD + 8];
+  __shared__ bf16 v_lds[kTile][kD + 8];
+  __shared__ bf16 p_lds[128][kTile + 8];
+
+  // hoist Q fragments: A operand rows = lane&15, 2 row-frags x 4 k-chunks
+  bf16x8 qf[2][4];
+  const int wrow0 = q0 + wave * 32;
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    int row = wrow0 + i * 16 + (lane & 15);
+    const bf16 *qr =
+        q + (((size_t)b * s + min(row, s - 1)) * qh + h) * kD;
+    bool ok = row < s;
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+      bf16x8 z{};
+      qf[i][ks] = ok ? *(const bf16x8 *)(qr + ks * 32 + (lane >> 4) * 8)
+                     : z;
+    }
+  }
+
+  float m_st[2][4], l_st[2][4];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      m_st[i][r] = -1e30f;
+      l_st[i][r] = 0.f;
+    }
+  f32x4 accO[2][8] = {};  // [row frag][d col group of 16]
+
+  // causal: this block's rows reach q0+127 -> kv tiles up to that row
+  const int kv_max = causal ? min(s, q0 + 128) : s;
+  const int ntiles = (kv_max + kTile - 1) / kTile;
+  for (int tile = 0; tile < ntiles; ++tile) {
+    const int pos0 = tile * kTile;
+    __syncthreads();
+    for (int i = tid; i < kTile * kD / 8; i += 256) {
+      int r = i / (kD / 8);
+      int c = (i % (kD / 8)) * 8;
+      int pos = pos0 + r;
+      bf16x8 kv8{}, vv8{};
+      if (pos < s) {
+        size_t base = (((size_t)b * s + pos) * kvh + kh) * kD + c;
+        kv8 = *(const bf16x8 *)(k + base);
+        vv8 = *(const bf16x8 *)(v + base);
+      }
+      *(bf16x8 *)(&k_lds[r][c]) = kv8;
+      *(bf16x8 *)(&v_lds[r][c]) = vv8;
+    }
+    __syncthreads();
+
+    // S = Q K^T: B operand cols = kv pos (lane&15 + j*16)
+    f32x4 accS[2][2] = {};
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+#pragma unroll
+      for (int ks = 0; ks < 4; ++ks) {
+        bf16x8 kf = *(const bf16x8 *)(
+            &k_lds[(lane & 15) + 16 * j][ks * 32 + (lane >> 4) * 8]);
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+          accS[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              qf[i][ks], kf, accS[i][j], 0, 0, 0);
+      }
+    }
+
+    // online softmax per C-layout row (i, r); cols j*16 + lane&15
+    float alpha[2][4];
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = wrow0 + i * 16 + (lane >> 4) * 4 + r;
+        float s0 = accS[i][0][r] * scale;
+        float s1 = accS[i][1][r] * scale;
+        int p0 = pos0 + (lane & 15);
+        bool ok0 = (p0 < s) && (!causal || p0 <= row);
+        bool ok1 = (p0 + 16 < s) && (!causal || p0 + 16 <= row);
+        s0 = ok0 ? s0 : -1e30f;
+        s1 = ok1 ? s1 : -1e30f;
+        float mx = fmaxf(s0, s1);
+#pragma unroll
+        for (int off = 8; off > 0; off >>= 1)
+          mx = fmaxf(mx, __shfl_xor(mx, off));
+        float m_new = fmaxf(m_st[i][r], mx);
+        float p0f = ok0 ? __expf(s0 - m_new) : 0.f;
+        float p1f = ok1 ? __expf(s1 - m_new) : 0.f;
+        float ps = p0f + p1f;
+#pragma unroll
+        for (int off = 8; off > 0; off >>= 1) ps += __shfl_xor(ps, off);
+        alpha[i][r] = __expf(m_st[i][r] - m_new);
+        l_st[i][r] = l_st[i][r] * alpha[i][r] + ps;
+        m_st[i][r] = m_new;
+        // store P to LDS (A-operand source)
+        int lrow = wave * 32 + i * 16 + (lane >> 4) * 4 + r;
+        p_lds[lrow][lane & 15] = (bf16)p0f;
+        p_lds[lrow][(lane & 15) + 16] = (bf16)p1f;
+      }
+    __syncthreads();
+
+    // PV: A = P[32 rows, 32 kv], B = V^T via tr_read; 8 d col-groups
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      bf16x8 af = *(const bf16x8 *)(
+          &p_lds[wave * 32 + i * 16 + (lane & 15)][(lane >> 4) * 8]);
+#pragma unroll
+      for (int cg = 0; cg < 8; ++cg) {
+        const int k0 = (lane >> 4) * 8;
+        unsigned addr = (unsigned)(uintptr_t)(
+            &v_lds[k0 + ((lane & 15) >> 2)][cg * 16 + 4 * (lane & 3)]);
+        unsigned long long lo, hi;
+        asm volatile(
+            "ds_read_b64_tr_b16 %0, %2\n"
+            "ds_read_b64_tr_b16 %1, %2 offset:%3\n"
+            "s_waitcnt lgkmcnt(0)"
+            : "=v"(lo), "=v"(hi)
+            : "v"(addr), "i"(4 * (kD + 8) * 2));
+        bf16x8 bf;
+        *(unsigned long long *)&bf = lo;
+        *((unsigned long long *)&bf + 1) = hi;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) accO[i][cg][r] *= alpha[i][r];
+        accO[i][cg] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af, bf, accO[i][cg], 0, 0, 0);
+      }
+    }
+    // NOTE: accO rescale uses alpha from THIS tile exactly once per
+    // cg (applied inside the cg loop before each MFMA).
+  }
+
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = wrow0 + i * 16 + (lane >> 4) * 4 + r;
+      if (row >= s) continue;
+      float inv_l = l_st[i][r] > 0.f ? 1.f / l_st[i][r] : 0.f;
+      bf16 *orow = out + (((size_t)b * s + row) * qh + h) * kD;
+#pragma unroll
+      for (int cg = 0; cg < 8; ++cg)
+        orow[cg * 16 + (lane & 15)] = (bf16)(accO[i][cg][r] * inv_l);
+      if (lse && (lane & 15) == 0)
+        lse[((size_t)b * s + row) * qh + h] =
+            m_st[i][r] + __logf(fmaxf(l_st[i][r], 1e-30f));
+    }
+}
+
+void launch_flash_prefill(const void *q, const void *k, const void *v,
+                          void *out, void *lse, int b, int s, int qh,
+                          int kvh, float scale, bool causal,
+                          hipStream_t stream) {
+  if (qh % kvh) throw std::runtime_error("flash_prefill: qh % kvh != 0");
+  dim3 grid((s + 127) / 128, qh, b);
+  hipLaunchKernelGGL(k_flash_prefill, grid, dim3(256), 0, stream,
+                     (const bf16 *)q, (const bf16 *)k, (const bf16 *)v,
+                     (bf16 *)out, (float *)lse, s, qh, kvh, scale,
+                     causal ? 1 : 0);
+}
+
+}  // namespace td

@@ -742,6 +742,16 @@ static void qkv_prologue_decode(uintptr_t qkv, uintptr_t q_out,
   TD_CHECK_HIP(hipGetLastError());
 }
 
+static void flash_prefill(uintptr_t q, uintptr_t k, uintptr_t v,
+                          uintptr_t out, int b, int s, int qh, int kvh,
+                          float scale, bool causal, uintptr_t stream,
+                          uintptr_t lse = 0) {
+  launch_flash_prefill((const void *)q, (const void *)k, (const void *)v,
+                       (void *)out, (void *)lse, b, s, qh, kvh, scale,
+                       causal, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
 static void flash_decode(uintptr_t q, uintptr_t kcache, uintptr_t vcache,
                          uintptr_t out, uintptr_t offset, int batch, int qh,
                          int kvh, int max_len, uintptr_t stream) {
@@ -815,6 +825,10 @@ PYBIND11_MODULE(_C, m) {
   m.def("reduce_scatter", &reduce_scatter_op);
   m.def("gdn_decode", &gdn_decode);
   m.def("flash_decode_paged", &flash_decode_paged);
+  m.def("flash_prefill", &flash_prefill, py::arg("q"), py::arg("k"),
+        py::arg("v"), py::arg("out"), py::arg("b"), py::arg("s"),
+        py::arg("qh"), py::arg("kvh"), py::arg("scale"), py::arg("causal"),
+        py::arg("stream"), py::arg("lse") = 0);
   m.def("ll_allgather", &ll_allgather_op);
   m.def("all_to_all", &all_to_all_op);
   m.def("ag_gemm_consumer_splitk_bf16", &ag_gemm_consumer_splitk_bf16);

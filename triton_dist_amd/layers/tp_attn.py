@@ -110,6 +110,19 @@ class TP_Attn:
             q = rms_norm(q, self.q_norm_w, self.rms_eps)
             k = rms_norm(k, self.k_norm_w, self.rms_eps)
         q, k = self.rotary.apply(q, k, pos)
+        # prefill attention: the in-house MFMA FA2 kernel consumes the
+        # natural [b, s, h, D] layout directly (no transposes); sdpa
+        # remains only for CPU / non-128 head dims / masked decode
+        use_fa2 = (q.is_cuda and self.head_dim == 128
+                   and (kv_cache is None or prefill))
+        if use_fa2:
+            from ..ops.fused import flash_prefill_op
+
+            if prefill and kv_cache is not None:
+                kv_cache.fill_prefill(layer_idx, k, v)
+            o = flash_prefill_op(q.to(self.dtype), k.to(self.dtype),
+                                 v.to(self.dtype), causal=(s > 1))
+            return o.reshape(b * s, self.qh * self.head_dim)
         q = q.transpose(1, 2)  # [b, qh, s, D]
         if kv_cache is None:
             ks, vs = k.transpose(1, 2), v.transpose(1, 2)

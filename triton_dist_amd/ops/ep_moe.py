@@ -165,6 +165,15 @@ def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
     eflags_off = ctx.eflags.offset + parity * world * e_loc * 4
     cell = 0
 
+    # world == 1 fast path: no peers — stream order replaces every
+    # reset/flag/barrier kernel (splits "exchange" = reading my own
+    # counts; dispatch is just the expert-sort permute; GEMMs ungated).
+    # Saves ~10 control-kernel launches per call (ROADMAP item 6).
+    if world == 1:
+        return _ep_moe_world1(x, topk_ids, topk_w, w_gate_up, w_down, ctx,
+                              out, parity, recv_x_off, meta_off,
+                              combine_off, cflags_off, eflags_off)
+
     # phase 0: buffer-reuse protection
     L["counts"].zero_()
     L["arrive_d"].zero_()
@@ -306,6 +315,84 @@ def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
                           cell)
     if ctx.low_latency:
         _C.signal_credit(ctx.credit_flags.offset, cell, s)
+    return out
+
+
+def _ep_moe_world1(x, topk_ids, topk_w, w_gate_up, w_down, ctx, out,
+                   parity, recv_x_off, meta_off, combine_off, cflags_off,
+                   eflags_off):
+    T, H = x.shape
+    K = ctx.topk
+    E, e_loc = ctx.n_experts, ctx.e_loc
+    inter = w_down.shape[2]
+    _C = ctx.heap._C
+    heap = ctx.heap
+    rank = 0
+    s = torch.cuda.current_stream().cuda_stream
+    L = ctx.local
+    L["counts"].zero_()
+    L["arrive_e"].zero_()
+    L["arrive_c"].zero_()
+    _C.moe_count(topk_ids.data_ptr(), L["counts"].data_ptr(),
+                 L["send_pos"].data_ptr(), L["send_to_dst"].data_ptr(),
+                 T * K, E, e_loc, 1, s)
+    avg_rows = max(1, T * K // E)
+    small_m = avg_rows <= 64
+    bm = 32 if small_m else 128
+    # splits matrix [1, E] IS my counts tensor — no exchange
+    _C.moe_layout(L["counts"].data_ptr(), 0, 1, E, e_loc,
+                  L["send_base"].data_ptr(), L["expert_base"].data_ptr(),
+                  L["expert_rows"].data_ptr(), L["recv_from_src"].data_ptr(),
+                  L["recv_total"].data_ptr(), s,
+                  L["work_items"].data_ptr(), L["work_count"].data_ptr(), bm)
+    # expert-sort permute (local copies; signals are cheap local stores
+    # and nothing waits on them)
+    _C.moe_dispatch(x.data_ptr(), topk_ids.data_ptr(),
+                    L["send_pos"].data_ptr(), L["send_base"].data_ptr(),
+                    L["counts"].data_ptr(), recv_x_off, meta_off,
+                    eflags_off, L["arrive_e"].data_ptr(), T, K, H, e_loc,
+                    E, s, 0)
+    cap_tiles = (ctx.cap + bm - 1) // bm
+    expert_h = torch.empty(ctx.cap + 128, 2 * inter, dtype=torch.bfloat16,
+                           device=x.device)
+    if small_m:
+        _C.moe_grouped_gemm_pq(heap.ptr(rank, recv_x_off),
+                               w_gate_up.data_ptr(), expert_h.data_ptr(),
+                               L["expert_base"].data_ptr(),
+                               L["expert_rows"].data_ptr(),
+                               L["work_items"].data_ptr(),
+                               L["work_count"].data_ptr(), 2 * inter, H, s)
+    else:
+        _C.moe_grouped_gemm(heap.ptr(rank, recv_x_off), w_gate_up.data_ptr(),
+                            expert_h.data_ptr(), L["expert_base"].data_ptr(),
+                            L["expert_rows"].data_ptr(), e_loc, cap_tiles,
+                            2 * inter, H, ctx.cap, s, False)
+    act = torch.empty(ctx.cap + 128, inter, dtype=torch.bfloat16,
+                      device=x.device)
+    _C.swiglu(expert_h.data_ptr(), act.data_ptr(), ctx.cap + 128, inter, s)
+    expert_out = torch.empty(ctx.cap + 128, H, dtype=torch.bfloat16,
+                             device=x.device)
+    if small_m:
+        _C.moe_grouped_gemm_pq(act.data_ptr(), w_down.data_ptr(),
+                               expert_out.data_ptr(),
+                               L["expert_base"].data_ptr(),
+                               L["expert_rows"].data_ptr(),
+                               L["work_items"].data_ptr(),
+                               L["work_count"].data_ptr(), H, inter, s)
+    else:
+        _C.moe_grouped_gemm(act.data_ptr(), w_down.data_ptr(),
+                            expert_out.data_ptr(),
+                            L["expert_base"].data_ptr(),
+                            L["expert_rows"].data_ptr(), e_loc, cap_tiles,
+                            H, inter, ctx.cap, s, False)
+    _C.moe_combine_send(expert_out.data_ptr(), heap.ptr(rank, meta_off),
+                        L["recv_total"].data_ptr(),
+                        L["recv_from_src"].data_ptr(),
+                        combine_off, cflags_off,
+                        L["arrive_c"].data_ptr(), ctx.cap, H, s, 0)
+    _C.moe_combine_reduce(heap.ptr(rank, combine_off), topk_w.data_ptr(),
+                          topk_ids.data_ptr(), out.data_ptr(),
+                          heap.ptr(rank, cflags_off), 1, T, K, H, E, s, 0)
     return out
 
 

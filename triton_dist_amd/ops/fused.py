@@ -79,6 +79,36 @@ def qkv_prologue_decode_op(qkv: torch.Tensor, kv_k: torch.Tensor,
     return q_out
 
 
+def flash_prefill_op(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                     causal: bool = True, return_lse: bool = False):
+    """FA2 forward on MFMA (csrc/kernels/attention.hip k_flash_prefill).
+
+    q: [b, s, qh, 128]; k/v: [b, s, kvh, 128] — the layer's natural
+    post-RoPE layout (no transposes). Returns [b, s, qh, 128].
+    Replaces torch sdpa on the prefill hot path (reference capability:
+    kernels/nvidia/sp_ag_attention_intra_node.py:257-428 FA2 consumer —
+    behavior only)."""
+    b, sq, qh, d = q.shape
+    kvh = k.shape[2]
+    assert d == 128 and k.shape[1] == sq
+    q = q.contiguous()
+    k = k.contiguous()
+    v = v.contiguous()
+    out = torch.empty_like(q)
+    scale = 1.0 / (d ** 0.5)
+    lse = None
+    lse_ptr = 0
+    if return_lse:
+        lse = torch.empty(b, sq, qh, dtype=torch.float32, device=q.device)
+        lse_ptr = lse.data_ptr()
+    _C().flash_prefill(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                       out.data_ptr(), b, sq, qh, kvh, scale, causal,
+                       _s(), lse_ptr)
+    if return_lse:
+        return out, lse
+    return out
+
+
 def flash_decode_op(q: torch.Tensor, kv_k: torch.Tensor, kv_v: torch.Tensor,
                     offset: torch.Tensor, qh: int, kvh: int) -> torch.Tensor:
     """q: [B, qh*128] post-RoPE -> out [B, qh*128]."""

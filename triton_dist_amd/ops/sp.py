@@ -308,10 +308,19 @@ def create_sp_ag_attn_context(max_chunk_tokens: int, kvh: int, head_dim: int,
 
 
 def _flash_with_lse(q, k, v, causal):
-    """q/k/v: [B, H, S, D] bf16 -> (out [B,H,S,D], lse [B,H,S])."""
+    """q/k/v: [B, S, H, D] bf16 -> (out [B,S,H,D], lse [B,S,H] fp32) via
+    the in-house MFMA FA2 kernel (csrc/kernels/attention.hip
+    k_flash_prefill); aten flash remains only for head_dim != 128."""
+    if q.is_cuda and q.shape[-1] == 128:
+        from .fused import flash_prefill_op
+
+        return flash_prefill_op(q, k, v, causal=causal, return_lse=True)
+    qt = q.permute(0, 2, 1, 3).contiguous()
+    kt = k.permute(0, 2, 1, 3).contiguous()
+    vt = v.permute(0, 2, 1, 3).contiguous()
     outs = torch.ops.aten._scaled_dot_product_flash_attention(
-        q, k, v, dropout_p=0.0, is_causal=causal)
-    return outs[0], outs[1]
+        qt, kt, vt, dropout_p=0.0, is_causal=causal)
+    return outs[0].permute(0, 2, 1, 3), outs[1].permute(0, 2, 1)
 
 
 def _merge_lse(o1, l1, o2, l2):
@@ -372,16 +381,16 @@ def sp_ag_attention(q: torch.Tensor, k_chunk: torch.Tensor,
         _C.memcpy_async(ctx.flags.ptr(peer) + rank * 4, heap.one_src.ptr(),
                         4, st.cuda_stream)
 
-    # consumer: chunks 0..rank (causal), waiting each chunk's flag
-    qt = q.permute(1, 0, 2).unsqueeze(0)  # [1, qh, S_loc, D]
+    # consumer: chunks 0..rank (causal), waiting each chunk's flag;
+    # the FA2 consumer takes the natural [1, S, H, D] layout directly
+    qt = q.unsqueeze(0)  # [1, S_loc, qh, D]
     o = lse = None
     for src in range(rank + 1):
         _C.wait_eq(ctx.flags.ptr() + src * 4, 1, 1, s)
-        kv_k = ctx.kbuf.local()[src, :s_loc].permute(1, 0, 2).unsqueeze(0)
-        kv_v = ctx.vbuf.local()[src, :s_loc].permute(1, 0, 2).unsqueeze(0)
+        kv_k = ctx.kbuf.local()[src, :s_loc].unsqueeze(0)
+        kv_v = ctx.vbuf.local()[src, :s_loc].unsqueeze(0)
         causal = (src == rank)
-        oc, lc = _flash_with_lse(qt.contiguous(), kv_k.contiguous(),
-                                 kv_v.contiguous(), causal)
+        oc, lc = _flash_with_lse(qt, kv_k, kv_v, causal)
         if o is None:
             o, lse = oc, lc
         else:
@@ -389,7 +398,7 @@ def sp_ag_attention(q: torch.Tensor, k_chunk: torch.Tensor,
     for i, ev in enumerate(ctx.join_evs):
         ev.record(ctx.comm_streams[i])
         compute.wait_event(ev)
-    return o.squeeze(0).permute(1, 0, 2).contiguous()
+    return o.squeeze(0).contiguous()
 
 
 def _sp_ag_attention_cpu(q, k_chunk, v_chunk, ctx, qh):
