@@ -77,11 +77,14 @@ def _body_ag_gemm(rank, world):
     a = (torch.randn(m, k, device="cuda") / 8).to(torch.bfloat16)
     torch.manual_seed(99)
     w = (torch.randn(n, k, device="cuda") / 8).to(torch.bfloat16)
-    for _ in range(3):
-        c = ag_gemm(a, w, ctx)
+    ref = ag_gemm_ref(a, w)
+    # push = SDMA stream-cooperative; fused = single-kernel paradigm
+    # (producer WGs + consumer GEMM in one launch); alternate to exercise
+    # cross-method buffer reuse too
+    for method in ("push", "fused", "push", "fused", "fused"):
+        c = ag_gemm(a, w, ctx, method=method)
         torch.cuda.synchronize()
-        ref = ag_gemm_ref(a, w)
-        assert_allclose(c, ref, **bf16_gemm_tol(k))
+        assert_allclose(c, ref, msg=method, **bf16_gemm_tol(k))
 
 
 def test_ag_gemm_2rank():

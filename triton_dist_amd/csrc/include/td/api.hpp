@@ -64,6 +64,16 @@ struct AgGemmArgs {
 };
 void launch_ag_gemm_consumer_bf16(const AgGemmArgs &args, hipStream_t stream);
 
+// Single-fused-kernel AG-GEMM (gemm256.hip): first comm_wgs workgroups
+// push my shard (incl. the self-copy) chunk-by-chunk into every peer's
+// workspace segment and signal per-chunk flags; the rest run the
+// flag-waiting 256^2 consumer. `arrive` = local [world*chunks] int32
+// sub-chunk counters (reset per call).
+void launch_ag_gemm256_fused(const AgGemmArgs &args, const PeerTable &pt,
+                             const void *src, size_t ws_off,
+                             size_t flags_off, int *arrive, int comm_wgs,
+                             int subsplit, hipStream_t stream);
+
 // GEMM-RS producer: GEMM whose epilogue scatters each output tile directly
 // into the owner rank's symmetric scatter buffer (remote store over xGMI).
 struct GemmRsArgs {
@@ -263,6 +273,6 @@ void launch_flash_decode_partial(const void *q, const void *kcache,
                                  hipStream_t stream);
 void launch_lse_combine(const void *parts, const void *lses, void *out,
                         const void *flags, int world, int batch, int qh,
-                        hipStream_t stream);
+                        int slot_batch, hipStream_t stream);
 
 }  // namespace td

@@ -528,7 +528,11 @@ void launch_flash_decode_partial(const void *q, const void *kcache,
 __global__ void k_lse_combine(const float *__restrict__ parts,
                               const float *__restrict__ lses,
                               bf16 *__restrict__ out, const int *flags,
-                              int world, int batch, int qh) {
+                              int world, int batch, int qh,
+                              int slot_batch) {
+  // slot_batch = allocated rows per rank slot (>= batch): peers write
+  // into [r, 0..batch) of a max_batch-strided slot, so b < max_batch
+  // calls index with the SLOT stride, not the call's batch
   const int b = blockIdx.x;
   const int h = blockIdx.y;
   if (threadIdx.x < (unsigned)world)
@@ -538,10 +542,10 @@ __global__ void k_lse_combine(const float *__restrict__ parts,
   if (threadIdx.x == 0) {
     float mx = -1e30f;
     for (int r = 0; r < world; ++r)
-      mx = fmaxf(mx, lses[((size_t)r * batch + b) * qh + h]);
+      mx = fmaxf(mx, lses[((size_t)r * slot_batch + b) * qh + h]);
     float denom = 0.f;
     for (int r = 0; r < world; ++r) {
-      float w = __expf(lses[((size_t)r * batch + b) * qh + h] - mx);
+      float w = __expf(lses[((size_t)r * slot_batch + b) * qh + h] - mx);
       w_sh[r] = w;
       denom += w;
     }
@@ -552,17 +556,17 @@ __global__ void k_lse_combine(const float *__restrict__ parts,
     float acc = 0.f;
     for (int r = 0; r < world; ++r)
       acc += w_sh[r] *
-             parts[(((size_t)r * batch + b) * qh + h) * kD + d];
+             parts[(((size_t)r * slot_batch + b) * qh + h) * kD + d];
     out[((size_t)b * qh + h) * kD + d] = (bf16)acc;
   }
 }
 
 void launch_lse_combine(const void *parts, const void *lses, void *out,
                         const void *flags, int world, int batch, int qh,
-                        hipStream_t stream) {
+                        int slot_batch, hipStream_t stream) {
   hipLaunchKernelGGL(k_lse_combine, dim3(batch, qh), dim3(128), 0, stream,
                      (const float *)parts, (const float *)lses, (bf16 *)out,
-                     (const int *)flags, world, batch, qh);
+                     (const int *)flags, world, batch, qh, slot_batch);
 }
 
 }  // namespace td

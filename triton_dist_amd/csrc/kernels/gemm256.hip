@@ -286,6 +286,96 @@ __global__ __launch_bounds__(NTH, 2) void k_gemm256_rs_producer_bf16(
   epilogue(acc, w, lds_a, seg, g.n);
 }
 
+// ---------------------------------------------------------------------------
+// Single-fused-kernel AG-GEMM (execution paradigm 2 of the reference,
+// kernels/amd/allgather_gemm.py:662-870 — behavior only): the first
+// `comm_wgs` workgroups push my shard's chunks to every peer's workspace
+// segment (16B vector copies over xGMI) and release-signal per-chunk
+// flags (sub-chunk WGs batch through a local arrive counter; the last
+// one signals); the REMAINING workgroups run the standard flag-waiting
+// persistent consumer GEMM. One launch, no comm streams, minimal launch
+// overhead (decode-friendly, graph-capturable). Producers sit at LOW
+// workgroup ids so the hardware dispatcher makes them resident before
+// the consumers start spinning.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(NTH, 2) void k_ag_gemm256_fused(
+    AgGemmArgs args, PeerTable pt, const bf16 *__restrict__ src,
+    size_t ws_off, size_t flags_off, int *__restrict__ arrive,
+    int comm_wgs, int subsplit) {
+  const int world = pt.world;
+  const int rank = pt.rank;
+  const int chunks = args.chunks_per_rank;
+  if ((int)blockIdx.x < comm_wgs) {
+    // producer role: items = (peer order i, chunk c, subchunk u)
+    const int items = world * chunks * subsplit;
+    const size_t rows_per_chunk = args.m_per_rank / chunks;
+    const size_t chunk_elems = rows_per_chunk * args.g.k;
+    const size_t sub_elems = chunk_elems / subsplit;
+    for (int it = blockIdx.x; it < items; it += comm_wgs) {
+      const int u = it % subsplit;
+      const int c = (it / subsplit) % chunks;
+      const int i = it / (subsplit * chunks);
+      const int peer = (rank + 1 + i) % world;  // i == world-1 -> self
+      bf16 *seg = (bf16 *)((char *)pt.bases[peer] + ws_off) +
+                  (size_t)rank * args.ws_stride * args.g.k;
+      const bf16 *s0 = src + c * chunk_elems + u * sub_elems;
+      bf16 *d0 = seg + c * chunk_elems + u * sub_elems;
+      for (size_t e = (size_t)threadIdx.x * 8; e < sub_elems;
+           e += (size_t)blockDim.x * 8)
+        *(bf16x8 *)(d0 + e) = *(const bf16x8 *)(s0 + e);
+      __syncthreads();
+      if (threadIdx.x == 0) {
+        int prev = atomic_add<Scope::Gpu>(arrive + i * chunks + c, 1);
+        if (prev + 1 == subsplit) {
+          fence_release_sys();
+          int *fl = (int *)((char *)pt.bases[peer] + flags_off);
+          st_release<Scope::Sys>(fl + rank * chunks + c, args.expect);
+        }
+      }
+    }
+    return;
+  }
+  // consumer role: standard per-tile flag-waiting persistent GEMM
+  __shared__ bf16 lds_a[SLICES * SLICE_ELEMS];
+  __shared__ bf16 lds_b[SLICES * SLICE_ELEMS];
+  GemmArgs &g = args.g;
+  const int tiles_n = g.n / BN;
+  const int tiles_m = g.m / BM;
+  const int tiles_per_rank = args.m_per_rank / BM;
+  int wgid = xcd_remap(blockIdx.x - comm_wgs, tiles_m * tiles_n);
+  int pid_m, pid_n;
+  tile_coords(wgid, tiles_m, tiles_n, pid_m, pid_n);
+  pid_m = (pid_m + args.rank * tiles_per_rank) % tiles_m;
+  int rows_per_chunk = args.m_per_rank / args.chunks_per_rank;
+  int c_lo = (pid_m * BM) / rows_per_chunk;
+  int c_hi = (pid_m * BM + BM - 1) / rows_per_chunk;
+  if (threadIdx.x < 64) {
+    for (int c = c_lo + (int)threadIdx.x; c <= c_hi; c += 64)
+      wait_ge_one<Scope::Sys>(args.flags + c, args.expect);
+  }
+  __syncthreads();
+  WaveCtx w = wave_ctx();
+  f32x4 acc[8][4] = {};
+  const int seg = (pid_m * BM) / args.m_per_rank;
+  const size_t arow0 =
+      (size_t)seg * args.ws_stride + (pid_m * BM - seg * args.m_per_rank);
+  const bf16 *ga = (const bf16 *)g.a + arow0 * g.lda;
+  const bf16 *gb = (const bf16 *)g.b + (size_t)pid_n * BN * g.ldb;
+  kloop(ga, gb, g.lda, g.ldb, g.k / BK, lds_a, lds_b, w, acc);
+  bf16 *dst = (bf16 *)g.c + (size_t)pid_m * BM * g.ldc + pid_n * BN;
+  epilogue(acc, w, lds_a, dst, g.ldc);
+}
+
+void launch_ag_gemm256_fused(const AgGemmArgs &args, const PeerTable &pt,
+                             const void *src, size_t ws_off,
+                             size_t flags_off, int *arrive, int comm_wgs,
+                             int subsplit, hipStream_t stream) {
+  int grid = comm_wgs + (args.g.m / BM) * (args.g.n / BN);
+  hipLaunchKernelGGL(k_ag_gemm256_fused, dim3(grid), dim3(NTH), 0, stream,
+                     args, pt, (const bf16 *)src, ws_off, flags_off,
+                     arrive, comm_wgs, subsplit);
+}
+
 // Split-K decode tier: each workgroup runs the same K-slice-ring pipeline
 // over a contiguous K/sk range and atomically accumulates its fp32 tile into
 // ws[M,N] (LLC-resident for decode shapes: 512x5120 fp32 = 10.5 MB). A

@@ -310,6 +310,33 @@ static void ag_gemm_consumer_bf16(uintptr_t a, uintptr_t b, uintptr_t c,
   TD_CHECK_HIP(hipGetLastError());
 }
 
+static void ag_gemm_fused_bf16(uintptr_t a, uintptr_t b, uintptr_t c,
+                               int m_total, int n, int k, size_t ws_off,
+                               size_t flags_off, int chunks,
+                               int m_per_rank, int ws_stride, int world,
+                               int rank, int expect, uintptr_t src,
+                               uintptr_t arrive, int comm_wgs,
+                               int subsplit, uintptr_t stream) {
+  check_active();
+  AgGemmArgs args;
+  args.g = GemmArgs{reinterpret_cast<void *>(a), reinterpret_cast<void *>(b),
+                    reinterpret_cast<void *>(c), nullptr,
+                    m_total, n, k, k, k, n};
+  args.flags = reinterpret_cast<const int *>(
+      (char *)g_heap.bases[g_heap.rank] + flags_off);
+  args.chunks_per_rank = chunks;
+  args.m_per_rank = m_per_rank;
+  args.ws_stride = ws_stride;
+  args.world = world;
+  args.rank = rank;
+  args.expect = expect;
+  launch_ag_gemm256_fused(args, g_heap.pt, reinterpret_cast<void *>(src),
+                          ws_off, flags_off,
+                          reinterpret_cast<int *>(arrive), comm_wgs,
+                          subsplit, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
 static void gemm_rs_producer_bf16(uintptr_t a, uintptr_t b, int m, int n,
                                   int k, size_t scatter_off, int m_per_rank,
                                   int ws_stride, int world, int rank,
@@ -775,9 +802,9 @@ static void flash_decode_partial(uintptr_t q, uintptr_t kcache,
 
 static void lse_combine(uintptr_t parts, uintptr_t lses, uintptr_t out,
                         uintptr_t flags, int world, int batch, int qh,
-                        uintptr_t stream) {
+                        int slot_batch, uintptr_t stream) {
   launch_lse_combine((void *)parts, (void *)lses, (void *)out, (void *)flags,
-                     world, batch, qh, as_stream(stream));
+                     world, batch, qh, slot_batch, as_stream(stream));
   TD_CHECK_HIP(hipGetLastError());
 }
 
@@ -813,6 +840,7 @@ PYBIND11_MODULE(_C, m) {
         py::arg("stream"), py::arg("prof_buf") = 0,
         py::arg("prof_cursor") = 0, py::arg("prof_cap") = 0);
   m.def("gemm_rs_producer_bf16", &gemm_rs_producer_bf16);
+  m.def("ag_gemm_fused_bf16", &ag_gemm_fused_bf16);
   m.def("gemm_ar_producer_bf16", &gemm_ar_producer_bf16);
   m.def("ar_tile_consumer", &ar_tile_consumer);
   m.def("rs_reduce_bf16", &rs_reduce_bf16);

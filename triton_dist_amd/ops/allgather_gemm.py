@@ -51,6 +51,7 @@ class AGGemmContext:
     comm_streams: List = field(default_factory=list)
     ready_ev: Optional[object] = None
     join_evs: List = field(default_factory=list)
+    ag_arrive: Optional[torch.Tensor] = None  # fused-kernel sub-chunk cnt
 
     @property
     def world(self) -> int:
@@ -77,12 +78,15 @@ def create_ag_gemm_context(max_m_per_rank: int, k: int,
         ctx.comm_streams = [torch.cuda.Stream() for _ in range(n_streams)]
         ctx.ready_ev = torch.cuda.Event()
         ctx.join_evs = [torch.cuda.Event() for _ in range(n_streams)]
+        ctx.ag_arrive = torch.zeros(world * chunks_per_rank,
+                                    dtype=torch.int32, device="cuda")
     return ctx
 
 
 def ag_gemm(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
             out: Optional[torch.Tensor] = None,
-            gathered_out: bool = False, profiler=None):
+            gathered_out: bool = False, profiler=None,
+            method: str = "auto"):
     """C[world*m, N] = AllGather(A[m, K]) @ W[N, K]^T.
 
     Returns C (and optionally the gathered A view for reuse).
@@ -117,6 +121,16 @@ def ag_gemm(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
     chunks = ctx.chunks_per_rank
     assert m % chunks == 0 and m % 128 == 0, \
         f"m={m} must divide chunks={chunks} and tile by 128"
+    # single-fused-kernel paradigm (reference allgather_gemm.py:662-870
+    # capability): producer workgroups + flag-waiting consumer GEMM in ONE
+    # launch. Wins at decode-sized m (3 launches instead of ~20 stream
+    # ops); the SDMA stream path stays for prefill (frees CUs).
+    if method == "auto":
+        method = "fused" if (m <= 1024 and (world * m) % 256 == 0
+                             and n % 256 == 0 and k % 128 == 0
+                             and ctx.ag_arrive is not None) else "push"
+    if method == "fused":
+        return _ag_gemm_fused(a, w, ctx, out, gathered_out)
     rows_per_chunk = m // chunks
     m_chunks = chunks
     chunk_bytes = rows_per_chunk * k * 2
@@ -177,6 +191,41 @@ def ag_gemm(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
     for s in range(min(ns, max(world - 1, 1))):
         ctx.join_evs[s].record(ctx.comm_streams[s])
         compute.wait_event(ctx.join_evs[s])
+    if gathered_out:
+        g = ctx.ws.local()[:, :m].reshape(m_total, k) \
+            if m < ctx.max_m_per_rank \
+            else ctx.ws.local().reshape(m_total, k)
+        return out, g
+    return out
+
+
+def _ag_gemm_fused(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
+                   out: Optional[torch.Tensor], gathered_out: bool):
+    m, k = a.shape
+    n = w.shape[0]
+    world, rank = ctx.world, ctx.rank
+    heap, _C = ctx.heap, ctx.heap._C
+    compute = torch.cuda.current_stream()
+    s = compute.cuda_stream
+    chunks = ctx.chunks_per_rank
+    m_total = world * m
+    # sub-split each chunk so enough producer WGs saturate the links
+    chunk_elems = (m // chunks) * k
+    subsplit = 1
+    while (subsplit < 4 and chunk_elems % (subsplit * 2 * 8) == 0
+           and chunk_elems // (subsplit * 2) >= 1 << 14):
+        subsplit *= 2
+    comm_wgs = min(world * chunks * subsplit, 64)
+    _C.reset_flags(ctx.flags.ptr(), world * chunks, 0, s)
+    _C.reset_flags(ctx.ag_arrive.data_ptr(), world * chunks, 0, s)
+    heap.barrier_all_on_stream(compute)
+    if out is None:
+        out = torch.empty(m_total, n, dtype=torch.bfloat16, device=a.device)
+    _C.ag_gemm_fused_bf16(
+        ctx.ws.ptr(), w.data_ptr(), out.data_ptr(), m_total, n, k,
+        ctx.ws.offset, ctx.flags.offset, chunks, m, ctx.max_m_per_rank,
+        world, rank, 1, a.data_ptr(), ctx.ag_arrive.data_ptr(), comm_wgs,
+        subsplit, s)
     if gathered_out:
         g = ctx.ws.local()[:, :m].reshape(m_total, k) \
             if m < ctx.max_m_per_rank \

@@ -96,13 +96,10 @@ def sp_flash_decode(q: torch.Tensor, kv_k: torch.Tensor, kv_v: torch.Tensor,
         _C.memcpy_async(ctx.flags.ptr(peer) + rank * 4, heap.one_src.ptr(),
                         4, s)
     _C.reset_flags(ctx.flags.ptr() + rank * 4, 1, 1, s)
-    # combine waits all flags
-    # NOTE parts layout uses max_batch stride; kernel expects [world,b,qh]
-    # dense — valid since b rows are leading within each slot when
-    # b == max_batch; for b < max_batch we pass strides via max_batch.
-    assert b == ctx.max_batch, "v1: batch must equal ctx.max_batch"
+    # combine waits all flags; slot stride = max_batch, so any
+    # b <= max_batch indexes correctly
     _C.lse_combine(ctx.parts.ptr(), ctx.lses.ptr(), out.data_ptr(),
-                   ctx.flags.ptr(), world, b, qh, s)
+                   ctx.flags.ptr(), world, b, qh, ctx.max_batch, s)
     return out
 
 
