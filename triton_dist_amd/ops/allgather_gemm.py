@@ -123,12 +123,18 @@ def ag_gemm(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
         f"m={m} must divide chunks={chunks} and tile by 128"
     # single-fused-kernel paradigm (reference allgather_gemm.py:662-870
     # capability): producer workgroups + flag-waiting consumer GEMM in ONE
-    # launch. Wins at decode-sized m (3 launches instead of ~20 stream
-    # ops); the SDMA stream path stays for prefill (frees CUs).
+    # launch vs the SDMA stream-cooperative path. method="auto" resolves
+    # via the contextual autotuner (measured on THIS hardware, cached);
+    # heuristic fallback while a graph capture is active and uncached.
     if method == "auto":
-        method = "fused" if (m <= 1024 and (world * m) % 256 == 0
-                             and n % 256 == 0 and k % 128 == 0
-                             and ctx.ag_arrive is not None) else "push"
+        eligible = ((world * m) % 256 == 0 and n % 256 == 0
+                    and k % 128 == 0 and ctx.ag_arrive is not None)
+        if not eligible:
+            method = "push"
+        elif torch.cuda.is_current_stream_capturing():
+            method = _ag_method_cached(m, n, k, world)
+        else:
+            method = _tune_ag_method(a, w, ctx, m, n, k, world)
     if method == "fused":
         return _ag_gemm_fused(a, w, ctx, out, gathered_out)
     rows_per_chunk = m // chunks
@@ -197,6 +203,40 @@ def ag_gemm(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
             else ctx.ws.local().reshape(m_total, k)
         return out, g
     return out
+
+
+_AG_METHOD_TUNER = None
+
+
+def _ag_tuner():
+    global _AG_METHOD_TUNER
+    if _AG_METHOD_TUNER is None:
+        from ..tune import ContextualAutoTuner
+        _AG_METHOD_TUNER = ContextualAutoTuner(
+            "ag_gemm_method",
+            [{"method": "push"}, {"method": "fused"}])
+    return _AG_METHOD_TUNER
+
+
+def _ag_key(m, n, k, world):
+    return f"m{m}_n{n}_k{k}_w{world}"
+
+
+def _ag_method_cached(m, n, k, world):
+    """Capture-safe lookup: cached tune result, else the size heuristic
+    (fused wins at decode-sized m: one launch vs ~20 stream ops)."""
+    t = _ag_tuner()._inner
+    entry = t._mem.get(_ag_key(m, n, k, world))
+    if entry:
+        return entry["config"]["method"]
+    return "fused" if m <= 1024 else "push"
+
+
+def _tune_ag_method(a, w, ctx, m, n, k, world):
+    cfg = _ag_tuner().tune(
+        _ag_key(m, n, k, world),
+        lambda c: (lambda: ag_gemm(a, w, ctx, method=c["method"])))
+    return cfg["method"]
 
 
 def _ag_gemm_fused(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,

@@ -194,7 +194,8 @@ def _n_owned(tiles: int, rank: int, world: int) -> int:
 
 def _gemm_ar_tiled_hip(a: torch.Tensor, w: torch.Tensor,
                        ctx: AllReduceContext,
-                       out: Optional[torch.Tensor]) -> torch.Tensor:
+                       out: Optional[torch.Tensor],
+                       sk: Optional[int] = None) -> torch.Tensor:
     assert a.dtype == torch.bfloat16 and a.is_contiguous()
     m, k = a.shape
     n = w.shape[0]
@@ -213,7 +214,8 @@ def _gemm_ar_tiled_hip(a: torch.Tensor, w: torch.Tensor,
     ctx.ev_fork.record(compute)
     ctx.comm_stream.wait_event(ctx.ev_fork)
 
-    sk = ar_sk_pick(m, n, k)
+    if sk is None:
+        sk = _gemm_ar_sk(a, w, ctx, m, n, k)
     ws_ptr = done_ptr = 0
     if sk > 1:
         if ctx.sk_ws is None or ctx.sk_ws.numel() < m * n:
@@ -242,6 +244,44 @@ def _gemm_ar_tiled_hip(a: torch.Tensor, w: torch.Tensor,
         out.copy_(res)
         return out
     return res
+
+
+_SK_TUNER = None
+
+
+def _sk_tuner():
+    global _SK_TUNER
+    if _SK_TUNER is None:
+        from ..tune import ContextualAutoTuner
+        _SK_TUNER = ContextualAutoTuner(
+            "gemm_ar_sk",
+            [{"sk": s} for s in (1, 2, 3, 4, 5, 6, 8, 9, 12, 16)])
+    return _SK_TUNER
+
+
+def _gemm_ar_sk(a, w, ctx, m, n, k) -> int:
+    """Producer split-K factor: tuned on hardware (invalid factors time
+    out to inf inside the tuner); heuristic when capturing uncached."""
+    key = f"m{m}_n{n}_k{k}_w{ctx.world}"
+    t = _sk_tuner()._inner
+    entry = t._mem.get(key)
+    if entry:
+        return entry["config"]["sk"]
+    if torch.cuda.is_current_stream_capturing():
+        return ar_sk_pick(m, n, k)
+
+    def mk(c):
+        sk = c["sk"]
+        if sk > 1 and (k % (128 * sk) or (k // 128 // sk) < 2):
+            raise ValueError("invalid sk for shape")
+        return lambda: _gemm_ar_tiled_hip_sk(a, w, ctx, sk)
+
+    return _sk_tuner().tune(key, mk)["sk"]
+
+
+def _gemm_ar_tiled_hip_sk(a, w, ctx, sk):
+    """Run the fused op with a FIXED sk (tuner closure)."""
+    _gemm_ar_tiled_hip(a, w, ctx, None, sk=sk)
 
 
 def _gemm_ar_tiled_cpu(a: torch.Tensor, w: torch.Tensor,
