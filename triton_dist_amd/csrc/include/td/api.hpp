@@ -117,6 +117,27 @@ void launch_gemm_rs_producer_splitk_bf16(const GemmRsArgs &a, float *ws,
 void launch_moe_router(const void *logits, void *topk_ids, void *topk_w,
                        int T, int E, int K, bool norm, hipStream_t stream);
 
+// Ulysses fused qkv-GEMM + head a2a (gemm256.hip): column-scatter
+// epilogue into the owning rank's recv buffer + per-src completion flags.
+struct UlyssesQkvArgs {
+  GemmArgs g;        // a [T_loc, H]; b = w_qkv [qkv_dim, H]; c unused
+  PeerTable pt;
+  size_t recv_off;   // [world_src][slot_rows][peer_cols] bf16 on each rank
+  size_t flags_off;  // [world] int32
+  int peer_cols;     // qkv_dim / world (multiple of 256)
+  int slot_rows;     // allocated rows per src slot (max_T)
+  int *arrive;       // local [world] int32 (reset per call)
+  int tiles_per_peer;
+  int expect;
+};
+void launch_gemm256_colscatter(const UlyssesQkvArgs &args,
+                               hipStream_t stream);
+// Accumulating GEMM (fp32 atomic ws, no memset/convert) + the converter.
+void launch_gemm256_acc_bf16(const GemmArgs &g, float *ws,
+                             hipStream_t stream);
+void launch_f32_to_bf16(const void *ws, void *c, const void *bias, int rows,
+                        int n, hipStream_t stream);
+
 // kernels/gemm_ar.hip — tile-granular fused GEMM + AllReduce -----------------
 // Round-robin tile ownership (owner = linear_tile % world). Producer GEMM
 // pushes each C tile to the owner's scatter slot + bumps the owner's

@@ -376,6 +376,113 @@ void launch_ag_gemm256_fused(const AgGemmArgs &args, const PeerTable &pt,
                      arrive, comm_wgs, subsplit);
 }
 
+// ---------------------------------------------------------------------------
+// Ulysses fused qkv-GEMM + head all-to-all (capability parity with the
+// reference kernels/nvidia/sp_ulysess_qkv_gemm_all2all.py:64-545 —
+// behavior only): the qkv projection's epilogue writes each 256^2 C tile
+// straight into the OWNER rank's recv buffer over xGMI, where the owner
+// is the rank whose head shard covers the tile's column block
+// (peer_cols = qkv_dim / world, tile-aligned). Per-owner arrive counters
+// (local, agent scope) release-signal the owner's per-src flag when my
+// last tile for it lands — the attention consumer waits world flags.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(NTH, 2) void k_gemm256_colscatter(
+    UlyssesQkvArgs args) {
+  __shared__ bf16 lds_a[SLICES * SLICE_ELEMS];
+  __shared__ bf16 lds_b[SLICES * SLICE_ELEMS];
+  GemmArgs &g = args.g;
+  const int tiles_n = g.n / BN;
+  const int tiles_m = g.m / BM;
+  int wgid = xcd_remap(blockIdx.x, tiles_m * tiles_n);
+  int pid_m, pid_n;
+  tile_coords(wgid, tiles_m, tiles_n, pid_m, pid_n);
+  WaveCtx w = wave_ctx();
+  f32x4 acc[8][4] = {};
+  const bf16 *ga = (const bf16 *)g.a + (size_t)pid_m * BM * g.lda;
+  const bf16 *gb = (const bf16 *)g.b + (size_t)pid_n * BN * g.ldb;
+  kloop(ga, gb, g.lda, g.ldb, g.k / BK, lds_a, lds_b, w, acc);
+  const int owner = (pid_n * BN) / args.peer_cols;
+  const int lcol = pid_n * BN - owner * args.peer_cols;
+  bf16 *dst = (bf16 *)((char *)args.pt.bases[owner] + args.recv_off) +
+              ((size_t)args.pt.rank * args.slot_rows + pid_m * BM) *
+                  args.peer_cols +
+              lcol;
+  epilogue(acc, w, lds_a, dst, args.peer_cols);
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    int prev = atomic_add<Scope::Gpu>(args.arrive + owner, 1);
+    if (prev + 1 == args.tiles_per_peer) {
+      fence_release_sys();
+      int *fl = (int *)((char *)args.pt.bases[owner] + args.flags_off);
+      st_release<Scope::Sys>(fl + args.pt.rank, args.expect);
+    }
+  }
+}
+
+void launch_gemm256_colscatter(const UlyssesQkvArgs &args,
+                               hipStream_t stream) {
+  if (args.g.m % BM || args.g.n % BN || args.g.k % BK ||
+      args.peer_cols % BN)
+    throw std::runtime_error("colscatter: 256/128 alignment required");
+  int grid = (args.g.m / BM) * (args.g.n / BN);
+  hipLaunchKernelGGL(k_gemm256_colscatter, dim3(grid), dim3(NTH), 0,
+                     stream, args);
+}
+
+// Accumulating GEMM: fp32 atomic adds into ws[M,N] (no memset, no
+// convert) — the Ulysses o-GEMM sums per-source K-block partials into a
+// shared ws as each source's a2a segment arrives.
+__global__ __launch_bounds__(NTH, 2) void k_gemm256_acc(GemmArgs g,
+                                                        float *ws) {
+  __shared__ bf16 lds_a[SLICES * SLICE_ELEMS];
+  __shared__ bf16 lds_b[SLICES * SLICE_ELEMS];
+  const int tiles_n = g.n / BN;
+  const int tiles_m = g.m / BM;
+  int wgid = xcd_remap(blockIdx.x, tiles_m * tiles_n);
+  int pid_m, pid_n;
+  tile_coords(wgid, tiles_m, tiles_n, pid_m, pid_n);
+  WaveCtx w = wave_ctx();
+  f32x4 acc[8][4] = {};
+  const bf16 *ga = (const bf16 *)g.a + (size_t)pid_m * BM * g.lda;
+  const bf16 *gb = (const bf16 *)g.b + (size_t)pid_n * BN * g.ldb;
+  kloop(ga, gb, g.lda, g.ldb, g.k / BK, lds_a, lds_b, w, acc);
+  float *wsb = ws + (size_t)pid_m * BM * g.ldc + pid_n * BN;
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = w.wr * 128 + i * 16 + ((w.lane >> 4) * 4 + r);
+        int col = w.wc * 64 + j * 16 + (w.lane & 15);
+        __hip_atomic_fetch_add(&wsb[(size_t)row * g.ldc + col],
+                               acc[i][j][r], __ATOMIC_RELAXED,
+                               __HIP_MEMORY_SCOPE_AGENT);
+      }
+}
+
+void launch_gemm256_acc_bf16(const GemmArgs &g, float *ws,
+                             hipStream_t stream) {
+  if (g.m % BM || g.n % BN || g.k % BK)
+    throw std::runtime_error("gemm256_acc: 256/128 alignment required");
+  int grid = (g.m / BM) * (g.n / BN);
+  hipLaunchKernelGGL(k_gemm256_acc, dim3(grid), dim3(NTH), 0, stream, g,
+                     ws);
+}
+
+__global__ void k_f32_to_bf16(const float *__restrict__ ws,
+                              bf16 *__restrict__ c,
+                              const bf16 *__restrict__ bias, int rows,
+                              int n);
+
+void launch_f32_to_bf16(const void *ws, void *c, const void *bias, int rows,
+                        int n, hipStream_t stream) {
+  int cgrid = rows < 2048 ? rows : 2048;
+  hipLaunchKernelGGL(k_f32_to_bf16, dim3(cgrid), dim3(256), 0, stream,
+                     (const float *)ws, (bf16 *)c, (const bf16 *)bias, rows,
+                     n);
+}
+
 // Split-K decode tier: each workgroup runs the same K-slice-ring pipeline
 // over a contiguous K/sk range and atomically accumulates its fp32 tile into
 // ws[M,N] (LLC-resident for decode shapes: 512x5120 fp32 = 10.5 MB). A

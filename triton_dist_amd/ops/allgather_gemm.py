@@ -127,7 +127,9 @@ def ag_gemm(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
     # via the contextual autotuner (measured on THIS hardware, cached);
     # heuristic fallback while a graph capture is active and uncached.
     if method == "auto":
-        eligible = ((world * m) % 256 == 0 and n % 256 == 0
+        # the fused consumer is the 256^2 kernel: a tile must lie inside
+        # ONE rank's segment, so m (per rank) must tile by 256
+        eligible = (m % 256 == 0 and n % 256 == 0
                     and k % 128 == 0 and ctx.ag_arrive is not None)
         if not eligible:
             method = "push"
@@ -243,6 +245,8 @@ def _ag_gemm_fused(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
                    out: Optional[torch.Tensor], gathered_out: bool):
     m, k = a.shape
     n = w.shape[0]
+    assert m % 256 == 0 and n % 256 == 0 and k % 128 == 0, \
+        "fused ag_gemm: m/n % 256, k % 128 required"
     world, rank = ctx.world, ctx.rank
     heap, _C = ctx.heap, ctx.heap._C
     compute = torch.cuda.current_stream()
