@@ -109,3 +109,48 @@ def _body_p2p(rank, world):
 
 def test_p2p_ring_gpu_2rank():
     run_distributed(_body_p2p, world_size=2)
+
+
+def _body_ulysses_fused_gpu(rank, world):
+    import torch.distributed as dist
+    from triton_dist_amd.ops import (create_ulysses_fused_context,
+                                     ulysses_a2a_o_gemm,
+                                     ulysses_qkv_gemm_a2a)
+    from triton_dist_amd.utils.testing import assert_allclose, bf16_gemm_tol
+
+    t_loc, hdim = 256, 1024
+    qkv_dim = world * 1280   # qwen3-32b-like per-rank block (5 x 256)
+    o_in = 1024
+    n_out = 1024
+    ctx = create_ulysses_fused_context(t_loc, qkv_dim, o_in)
+    torch.manual_seed(3 + rank)
+    x = (torch.randn(t_loc, hdim, device="cuda") / 8).to(torch.bfloat16)
+    torch.manual_seed(77)
+    w_qkv = (torch.randn(qkv_dim, hdim, device="cuda") / 8
+             ).to(torch.bfloat16)
+    w_o_split = (torch.randn(world, n_out, o_in, device="cuda") / 8
+                 ).to(torch.bfloat16)
+    for _ in range(2):
+        mine = ulysses_qkv_gemm_a2a(x, w_qkv, ctx)
+        torch.cuda.synchronize()
+        xs = [torch.empty_like(x) for _ in range(world)]
+        dist.all_gather(xs, x)
+        full = torch.cat(xs, 0).float() @ w_qkv.float().t()
+        pc = qkv_dim // world
+        ref = full[:, rank * pc:(rank + 1) * pc]
+        assert_allclose(mine, ref.to(torch.bfloat16), **bf16_gemm_tol(hdim))
+
+        attn = mine[:, :o_in].contiguous()
+        out = ulysses_a2a_o_gemm(attn, w_o_split, ctx)
+        torch.cuda.synchronize()
+        att_all = [torch.empty_like(attn) for _ in range(world)]
+        dist.all_gather(att_all, attn)
+        acc = torch.zeros(t_loc, n_out, dtype=torch.float32, device="cuda")
+        for src in range(world):
+            seg = att_all[src][rank * t_loc:(rank + 1) * t_loc].float()
+            acc += seg @ w_o_split[src].float().t()
+        assert_allclose(out, acc.to(torch.bfloat16), atol=2.5e-1, rtol=5e-2)
+
+
+def test_ulysses_fused_gpu_2rank():
+    run_distributed(_body_ulysses_fused_gpu, world_size=2)

@@ -94,3 +94,55 @@ def test_ulysses_layer_roundtrip_cpu_2rank():
 
 def test_sp_decode_cpu_4rank():
     run_distributed(_body_sp_decode, world_size=4)
+
+
+def _body_ulysses_fused(rank, world):
+    import torch
+    from triton_dist_amd.ops import (create_ulysses_fused_context,
+                                     ulysses_a2a_o_gemm,
+                                     ulysses_qkv_gemm_a2a)
+    from triton_dist_amd.utils import assert_allclose, rand_tensor
+
+    t_loc, hdim = 8, 64
+    qkv_dim = world * 512   # peer_cols = 512 (tiles by 256)
+    o_in = 512
+    n_out = 256
+    ctx = create_ulysses_fused_context(t_loc, qkv_dim, o_in)
+    g = torch.Generator().manual_seed(5 + rank)
+    x = rand_tensor((t_loc, hdim), dtype=torch.bfloat16, generator=g) / 8
+    gw = torch.Generator().manual_seed(42)
+    w_qkv = rand_tensor((qkv_dim, hdim), dtype=torch.bfloat16,
+                        generator=gw) / 8
+    w_o_split = rand_tensor((world, n_out, o_in), dtype=torch.bfloat16,
+                            generator=gw) / 8
+    import torch.distributed as dist
+    for _ in range(2):
+        mine = ulysses_qkv_gemm_a2a(x, w_qkv, ctx)
+        # golden: gather everyone's x, project, slice my column block
+        xs = [torch.empty_like(x) for _ in range(world)]
+        dist.all_gather(xs, x)
+        full = torch.cat(xs, 0).float() @ w_qkv.float().t()
+        pc = qkv_dim // world
+        ref = full[:, rank * pc:(rank + 1) * pc]
+        assert_allclose(mine, ref.to(torch.bfloat16), atol=8e-2, rtol=5e-2)
+
+        # o side: use `mine` (shape [world*t_loc, pc]) truncated to o_in
+        attn = mine[:, :o_in].contiguous()
+        out = ulysses_a2a_o_gemm(attn, w_o_split, ctx)
+        # golden: each rank's tokens gather their head shards from all
+        # ranks; o = sum_src attn_src(tokens of mine) @ w_o_split[src]^T
+        att_all = [torch.empty_like(attn) for _ in range(world)]
+        dist.all_gather(att_all, attn)
+        acc = torch.zeros(t_loc, n_out, dtype=torch.float32)
+        for src in range(world):
+            seg = att_all[src][rank * t_loc:(rank + 1) * t_loc].float()
+            acc += seg @ w_o_split[src].float().t()
+        assert_allclose(out, acc.to(torch.bfloat16), atol=8e-2, rtol=5e-2)
+
+
+def test_ulysses_fused_cpu_2rank():
+    run_distributed(_body_ulysses_fused, world_size=2)
+
+
+def test_ulysses_fused_cpu_4rank():
+    run_distributed(_body_ulysses_fused, world_size=4)

@@ -337,6 +337,45 @@ static void ag_gemm_fused_bf16(uintptr_t a, uintptr_t b, uintptr_t c,
   TD_CHECK_HIP(hipGetLastError());
 }
 
+static void gemm256_colscatter(uintptr_t a, uintptr_t b, int m, int n,
+                               int k, size_t recv_off, size_t flags_off,
+                               int peer_cols, int slot_rows,
+                               uintptr_t arrive, int tiles_per_peer,
+                               int expect, uintptr_t stream) {
+  check_active();
+  UlyssesQkvArgs args;
+  args.g = GemmArgs{reinterpret_cast<void *>(a), reinterpret_cast<void *>(b),
+                    nullptr, nullptr, m, n, k, k, k, n};
+  args.pt = g_heap.pt;
+  args.recv_off = recv_off;
+  args.flags_off = flags_off;
+  args.peer_cols = peer_cols;
+  args.slot_rows = slot_rows;
+  args.arrive = reinterpret_cast<int *>(arrive);
+  args.tiles_per_peer = tiles_per_peer;
+  args.expect = expect;
+  launch_gemm256_colscatter(args, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
+static void gemm256_acc_bf16(uintptr_t a, uintptr_t b, uintptr_t ws, int m,
+                             int n, int k, uintptr_t stream) {
+  GemmArgs args{reinterpret_cast<void *>(a), reinterpret_cast<void *>(b),
+                nullptr, nullptr, m, n, k, k, k, n};
+  launch_gemm256_acc_bf16(args, reinterpret_cast<float *>(ws),
+                          as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
+static void f32_to_bf16(uintptr_t ws, uintptr_t c, uintptr_t bias, int rows,
+                        int n, uintptr_t stream) {
+  launch_f32_to_bf16(reinterpret_cast<void *>(ws),
+                     reinterpret_cast<void *>(c),
+                     reinterpret_cast<void *>(bias), rows, n,
+                     as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
 static void gemm_rs_producer_bf16(uintptr_t a, uintptr_t b, int m, int n,
                                   int k, size_t scatter_off, int m_per_rank,
                                   int ws_stride, int world, int rank,
@@ -841,6 +880,9 @@ PYBIND11_MODULE(_C, m) {
         py::arg("prof_cursor") = 0, py::arg("prof_cap") = 0);
   m.def("gemm_rs_producer_bf16", &gemm_rs_producer_bf16);
   m.def("ag_gemm_fused_bf16", &ag_gemm_fused_bf16);
+  m.def("gemm256_colscatter", &gemm256_colscatter);
+  m.def("gemm256_acc_bf16", &gemm256_acc_bf16);
+  m.def("f32_to_bf16", &f32_to_bf16);
   m.def("gemm_ar_producer_bf16", &gemm_ar_producer_bf16);
   m.def("ar_tile_consumer", &ar_tile_consumer);
   m.def("rs_reduce_bf16", &rs_reduce_bf16);

@@ -42,6 +42,9 @@ from .sp import (  # noqa: F401
     create_ulysses_context,
     ulysses_a2a,
     ulysses_a2a_ref,
+    create_ulysses_fused_context,
+    ulysses_qkv_gemm_a2a,
+    ulysses_a2a_o_gemm,
     SPAGAttnContext,
     create_sp_ag_attn_context,
     sp_ag_attention,

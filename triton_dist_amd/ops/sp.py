@@ -428,3 +428,189 @@ def _sp_ag_attention_cpu(q, k_chunk, v_chunk, ctx, qh):
                                        attn_mask=mask.view(1, 1, s_loc, total),
                                        enable_gqa=True)
     return o.squeeze(0).permute(1, 0, 2).to(q.dtype)
+
+
+# ---------------------------------------------------------------------------
+# Fused Ulysses: qkv-GEMM whose epilogue IS the head all-to-all, and the
+# inverse a2a whose consumer is an arrival-ordered accumulating o-GEMM
+# (capability parity with the reference's fused pair,
+# kernels/nvidia/sp_ulysess_qkv_gemm_all2all.py:64-545 and
+# sp_ulysess_o_all2all_gemm.py — behavior only).
+# ---------------------------------------------------------------------------
+@dataclass
+class UlyssesFusedContext:
+    heap: SymmHeap
+    max_tokens: int      # per-rank tokens
+    qkv_dim: int         # (qh + 2*kvh) * head_dim (full, pre-shard)
+    o_in_dim: int        # qh_local_after_a2a... = qkv q-part per rank
+    recv_qkv: SymmBuffer  # [world_src, max_tokens, qkv_dim/world] bf16
+    qkv_flags: SymmBuffer  # [world] int32
+    recv_o: SymmBuffer   # [world_src, max_tokens, o_in_dim] bf16
+    o_flags: SymmBuffer  # [world] int32
+    arrive: Optional[torch.Tensor] = None  # local [world] int32
+
+    @property
+    def world(self):
+        return self.heap.world
+
+    @property
+    def rank(self):
+        return self.heap.rank
+
+    @property
+    def peer_cols(self):
+        return self.qkv_dim // self.world
+
+
+def create_ulysses_fused_context(max_tokens: int, qkv_dim: int,
+                                 o_in_dim: int,
+                                 heap: Optional[SymmHeap] = None
+                                 ) -> UlyssesFusedContext:
+    heap = heap or get_heap()
+    w = heap.world
+    assert qkv_dim % (w * 256) == 0, "peer column block must tile by 256"
+    ctx = UlyssesFusedContext(
+        heap, max_tokens, qkv_dim, o_in_dim,
+        recv_qkv=heap.alloc_buffer((w, max_tokens, qkv_dim // w),
+                                   torch.bfloat16),
+        qkv_flags=heap.alloc_buffer((w,), torch.int32),
+        recv_o=heap.alloc_buffer((w, max_tokens, o_in_dim), torch.bfloat16),
+        o_flags=heap.alloc_buffer((w,), torch.int32),
+    )
+    if heap.backend == "hip":
+        ctx.arrive = torch.zeros(w, dtype=torch.int32, device="cuda")
+    return ctx
+
+
+def ulysses_qkv_gemm_a2a(x: torch.Tensor, w_qkv: torch.Tensor,
+                         ctx: UlyssesFusedContext) -> torch.Tensor:
+    """[T_loc, H] @ w_qkv[qkv_dim, H]^T fused with the head a2a: the GEMM
+    epilogue pushes each output tile to the rank owning its column block.
+    Returns the gathered [world * T_loc, qkv_dim/world] view (all tokens,
+    my head shard)."""
+    t_loc, hdim = x.shape
+    world, rank = ctx.world, ctx.rank
+    assert w_qkv.shape[0] == ctx.qkv_dim and t_loc <= ctx.max_tokens
+    heap = ctx.heap
+    pc = ctx.peer_cols
+
+    if heap.backend == "cpu":
+        full = (x.float() @ w_qkv.float().t()).to(x.dtype)
+        heap.barrier_all()
+        for peer in range(world):
+            seg = full[:, peer * pc:(peer + 1) * pc]
+            ctx.recv_qkv.peer(peer)[rank, :t_loc].copy_(seg)
+        heap.barrier_all()
+        res = ctx.recv_qkv.local()[:, :t_loc].reshape(world * t_loc, pc)
+        out = res.clone()
+        heap.barrier_all()
+        return out
+
+    _C = heap._C
+    stream = torch.cuda.current_stream()
+    s = stream.cuda_stream
+    assert t_loc % 256 == 0 and hdim % 128 == 0
+    _C.reset_flags(ctx.qkv_flags.ptr(), world, 0, s)
+    _C.reset_flags(ctx.arrive.data_ptr(), world, 0, s)
+    heap.barrier_all_on_stream(stream)
+    tiles_per_peer = (t_loc // 256) * (pc // 256)
+    _C.gemm256_colscatter(x.data_ptr(), w_qkv.data_ptr(), t_loc,
+                          ctx.qkv_dim, hdim, ctx.recv_qkv.offset,
+                          ctx.qkv_flags.offset, pc, ctx.max_tokens,
+                          ctx.arrive.data_ptr(), tiles_per_peer, 1, s)
+    _C.wait_eq(ctx.qkv_flags.ptr(), world, 1, s)
+    if t_loc == ctx.max_tokens:
+        return ctx.recv_qkv.local().reshape(world * t_loc, pc)
+    return ctx.recv_qkv.local()[:, :t_loc].reshape(world * t_loc, pc)
+
+
+def ulysses_a2a_o_gemm(attn: torch.Tensor, w_o_split: torch.Tensor,
+                       ctx: UlyssesFusedContext,
+                       out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Inverse a2a fused with the o-projection: attn [world*T_loc, o_in]
+    (all tokens, my head shard) is pushed back per-destination (contiguous
+    row blocks, SDMA), and the o-GEMM accumulates per-SOURCE K-block
+    partials into an fp32 ws as each source's segment arrives (fixed
+    visit order, per-src flag waits) — overlap between late arrivals and
+    early partial GEMMs. w_o_split: [world, N, o_in] (w_o's K blocks,
+    pre-sliced contiguous at layer init)."""
+    world, rank = ctx.world, ctx.rank
+    t_total = attn.shape[0]
+    assert t_total % world == 0
+    t_loc = t_total // world
+    o_in = ctx.o_in_dim
+    n = w_o_split.shape[1]
+    heap = ctx.heap
+
+    if heap.backend == "cpu":
+        heap.barrier_all()
+        for peer in range(world):
+            seg = attn[peer * t_loc:(peer + 1) * t_loc]
+            ctx.recv_o.peer(peer)[rank, :t_loc].copy_(seg)
+        heap.barrier_all()
+        acc = torch.zeros(t_loc, n, dtype=torch.float32)
+        for src in range(world):
+            a_p = ctx.recv_o.local()[src, :t_loc].float()
+            acc += a_p @ w_o_split[src].float().t()
+        res = acc.to(attn.dtype)
+        heap.barrier_all()
+        if out is not None:
+            out.copy_(res)
+            return out
+        return res
+
+    _C = heap._C
+    stream = torch.cuda.current_stream()
+    s = stream.cuda_stream
+    assert attn.dtype == torch.bfloat16 and attn.is_contiguous()
+    assert t_loc % 256 == 0 and n % 256 == 0 and o_in % 128 == 0
+    _C.reset_flags(ctx.o_flags.ptr(), world, 0, s)
+    heap.barrier_all_on_stream(stream)
+    slot = ctx.max_tokens * o_in * 2
+    nbytes = t_loc * o_in * 2
+    for i in range(world - 1):
+        peer = (rank + 1 + i) % world
+        _C.memcpy_async(ctx.recv_o.ptr(peer) + rank * slot,
+                        attn.data_ptr() + peer * nbytes, nbytes, s)
+        _C.memcpy_async(ctx.o_flags.ptr(peer) + rank * 4,
+                        heap.one_src.ptr(), 4, s)
+    _C.memcpy_async(ctx.recv_o.ptr() + rank * slot,
+                    attn.data_ptr() + rank * nbytes, nbytes, s)
+    _C.reset_flags(ctx.o_flags.ptr() + rank * 4, 1, 1, s)
+    ws = torch.zeros(t_loc, n, dtype=torch.float32, device=attn.device)
+    for j in range(world):
+        src = (rank + j) % world  # self first (already local)
+        _C.wait_eq(ctx.o_flags.ptr() + src * 4, 1, 1, s)
+        _C.gemm256_acc_bf16(ctx.recv_o.ptr() + src * slot,
+                            w_o_split[src].data_ptr(), ws.data_ptr(),
+                            t_loc, n, o_in, s)
+    if out is None:
+        out = torch.empty(t_loc, n, dtype=torch.bfloat16,
+                          device=attn.device)
+    _C.f32_to_bf16(ws.data_ptr(), out.data_ptr(), 0, t_loc, n, s)
+    return out
+
+
+def ulysses_fused_ref(x: torch.Tensor, w_qkv: torch.Tensor,
+                      w_o: torch.Tensor, group=None):
+    """Golden for the fused pair composed: qkv-GEMM -> head a2a ->
+    (identity "attention") -> inverse a2a -> o-GEMM, via torch
+    collectives on CPU."""
+    import torch.distributed as dist
+
+    world = dist.get_world_size(group)
+    rank = dist.get_rank(group)
+    t_loc = x.shape[0]
+    full = (x.float() @ w_qkv.float().t())
+    pc = w_qkv.shape[0] // world
+    # a2a: my shard of every rank's tokens
+    segs = [full[:, p * pc:(p + 1) * pc].contiguous().cpu() for p in
+            range(world)]
+    gathered = [torch.empty_like(segs[0]) for _ in range(world)]
+    outs = []
+    for p in range(world):
+        dist.all_gather(gathered, segs[p], group=group)
+        if p == rank:
+            outs = [g.clone() for g in gathered]
+    mine = torch.cat(outs, 0)  # [world*t_loc, pc]
+    return mine.to(x.device)
