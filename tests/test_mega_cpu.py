@@ -95,3 +95,30 @@ def test_ksplit_task_semantics():
                    pn * BN:(pn + 1) * BN].sum(0)
     ref = A[:batch] @ B.T
     assert np.abs(C - ref).max() < 1e-3
+
+
+def test_emit_gemv_semantics():
+    """Execute emit_gemv's task stream in numpy: every 512-col chunk task
+    computes C[:, c0:c0+chunk] = A @ W[c0:c0+chunk].T; together they must
+    reproduce the full matmul, covering every column exactly once."""
+    import numpy as np
+
+    from triton_dist_amd.mega.builder import MegaGraph, T_GEMV, emit_gemv
+
+    rng = np.random.default_rng(3)
+    for batch, n, k in [(1, 1536, 512), (2, 1024, 256), (4, 2368, 128)]:
+        g = MegaGraph()
+        A = rng.standard_normal((batch, k)).astype(np.float32)
+        W = rng.standard_normal((n, k)).astype(np.float32)
+        C = np.zeros((batch, n), dtype=np.float32)
+        emit_gemv(g, 0, 0, 0, batch, n, k, None)
+        cols = []
+        for t in g.tasks:
+            tt, args = t[0], t[6]
+            assert tt == T_GEMV
+            _, _, _, m, nn, kk, c0, chunk = args
+            assert (m, nn, kk) == (batch, n, k)
+            C[:, c0:c0 + chunk] = A @ W[c0:c0 + chunk].T
+            cols += list(range(c0, c0 + chunk))
+        assert sorted(cols) == list(range(n))
+        np.testing.assert_allclose(C, A @ W.T, rtol=1e-5)

@@ -47,6 +47,12 @@ enum TaskType : int {
                          // a6=pid_m, a7=pid_n, a8=k0, a9=klen, a10=sk
   T_TILE_REDUCE = 9,     // a0=ws, a1=C, a2=m, a3=n, a4=pid_m, a5=pid_n,
                          // a6=ksplit
+  T_GEMV = 10,           // a0=A[m,K], a1=W[N,K], a2=C[m,N], a3=m, a4=n,
+                         // a5=k, a6=col0, a7=ncols — bsz<=4 decode GEMV:
+                         // one task = a 512-col chunk x full K (weight-
+                         // bandwidth-bound; ~25x fewer tasks than the
+                         // 32x128 tile scheme at bsz 1, which is what the
+                         // 0.25 us/task dispatch overhead demands)
 };
 
 struct Task {
@@ -522,6 +528,55 @@ TD_DEV void mk_flash_decode_body(const bf16 *q, const bf16 *kcache,
   }
 }
 
+namespace mk {
+
+// Decode GEMV: x rows staged in LDS (m <= 4), each thread owns
+// ncols/blockDim output columns and streams the weight rows with bf16x8
+// loads — HBM-bound by construction, the right regime for bsz<=4.
+TD_DEV void t_gemv(const Task &t, char *lds) {
+  const bf16 *A = (const bf16 *)t.a[0];
+  const bf16 *W = (const bf16 *)t.a[1];
+  bf16 *C = (bf16 *)t.a[2];
+  const int m = (int)t.a[3];
+  const int n = (int)t.a[4];
+  const int k = (int)t.a[5];
+  const int col0 = (int)t.a[6];
+  const int ncols = (int)t.a[7];
+  bf16 *x_lds = (bf16 *)lds;  // [m][k] (m*k*2 <= 61440 guaranteed by host)
+  for (int i = threadIdx.x * 8; i < m * k; i += blockDim.x * 8)
+    *(bf16x8 *)(x_lds + i) = *(const bf16x8 *)(A + i);
+  __syncthreads();
+  for (int c = col0 + threadIdx.x; c < col0 + ncols; c += blockDim.x) {
+    const bf16 *wr = W + (size_t)c * k;
+    float acc[4] = {0.f, 0.f, 0.f, 0.f};
+    for (int kk = 0; kk < k; kk += 8) {
+      bf16x8 w8 = *(const bf16x8 *)(wr + kk);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        if (r < m) {
+          bf16x8 x8 = *(const bf16x8 *)(x_lds + r * k + kk);
+#pragma unroll
+          for (int e = 0; e < 8; ++e)
+            acc[r] += (float)x8[e] * (float)w8[e];
+        }
+      }
+    }
+    for (int r = 0; r < m; ++r) C[(size_t)r * n + c] = (bf16)acc[r];
+  }
+}
+
+// relaxed spin (no per-iteration L2 invalidate); caller issues ONE
+// acquire fence after all deps are satisfied
+TD_DEV void spin_ge_relaxed(const int *flag, int bound) {
+  uint64_t t0 = wallclock();
+  while (ld_relaxed<Scope::Gpu>(flag) < bound) {
+    __builtin_amdgcn_s_sleep(1);
+    if (wallclock() - t0 > TD_SPIN_TIMEOUT_TICKS) __builtin_trap();
+  }
+}
+
+}  // namespace mk
+
 // ---------------------------------------------------------------------------
 // the persistent megakernel
 // ---------------------------------------------------------------------------
@@ -533,10 +588,14 @@ __global__ __launch_bounds__(mk::NTH) void k_megakernel(
   const int q_lo = queue_off[wg], q_hi = queue_off[wg + 1];
   for (int qi = q_lo; qi < q_hi; ++qi) {
     const mk::Task t = tasks[queue[qi]];
-    // dependency waits (device scope; decode-scale spins are short)
+    // dependency waits: relaxed spins + ONE acquire fence (a per-
+    // iteration agent-scope acquire invalidates the XCD L2 every spin —
+    // measured poison at 10k+ tasks/step)
     if (threadIdx.x == 0) {
-      if (t.dep0 >= 0) wait_ge_one<Scope::Gpu>(scoreboard + t.dep0, t.dep0_n);
-      if (t.dep1 >= 0) wait_ge_one<Scope::Gpu>(scoreboard + t.dep1, t.dep1_n);
+      if (t.dep0 >= 0) mk::spin_ge_relaxed(scoreboard + t.dep0, t.dep0_n);
+      if (t.dep1 >= 0) mk::spin_ge_relaxed(scoreboard + t.dep1, t.dep1_n);
+      if (t.dep0 >= 0 || t.dep1 >= 0)
+        __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
     }
     __syncthreads();
     switch (t.type) {
@@ -573,12 +632,17 @@ __global__ __launch_bounds__(mk::NTH) void k_megakernel(
       case mk::T_KV_ADVANCE:
         if (threadIdx.x == 0) *(long *)t.a[0] += 1;
         break;
+      case mk::T_GEMV:
+        mk::t_gemv(t, lds);
+        break;
       default:
         break;
     }
     __syncthreads();
     if (threadIdx.x == 0) {
-      fence_release_sys();
+      // agent scope: single GPU — the writeback makes this XCD's lines
+      // visible to consumer tasks on other XCDs
+      __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
       atomic_add<Scope::Gpu>(scoreboard + t.score_slot, 1);
     }
     __syncthreads();

@@ -94,6 +94,28 @@ class MegaGraph:
                 torch.zeros(len(self.ops), dtype=torch.int32, device=device))
 
 
+T_GEMV = 10            # bsz<=4 decode GEMV (512-col chunk x full K)
+
+
+def emit_gemv(g: "MegaGraph", a_ptr: int, w_ptr: int, c_ptr: int,
+              batch: int, n: int, k: int, dep):
+    """Decode-GEMV emission (batch <= 4): one task per 512-column chunk
+    streaming the full K — ~25x fewer tasks than 32x128 tiling at bsz 1,
+    which is what the ~0.25 us/task dispatch overhead demands. x rows
+    must fit the 60 KiB task LDS (m*k*2 <= 61440)."""
+    assert batch * k * 2 <= 61440
+    chunk = 512
+    while n % chunk:
+        chunk //= 2
+    op = g.new_op()
+    for c0 in range(0, n, chunk):
+        g.add_task(T_GEMV, op,
+                   [a_ptr, w_ptr, c_ptr, batch, n, k, c0, chunk],
+                   [(dep, 0)] if dep else [])
+    g.next_level()
+    return op
+
+
 def emit_gemm(g: "MegaGraph", a_ptr: int, w_ptr: int, c_ptr: int,
               batch: int, n: int, k: int, dep, ksplit: int = 1,
               ws_ptr: int = 0):

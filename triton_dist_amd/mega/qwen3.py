@@ -60,6 +60,13 @@ class MegaQwen3Decode:
         self._ws = {}
 
         def gemm(a_buf, w, c_buf, n, k, dep):
+            # bsz<=4: GEMV route (one task per 512-col chunk; weight-BW
+            # bound, ~25x fewer tasks). TD_MK_NO_GEMV=1 forces tiles.
+            if (batch <= 4 and batch * k * 2 <= 61440
+                    and not os.environ.get("TD_MK_NO_GEMV")):
+                from .builder import emit_gemv
+                return emit_gemv(g, a_buf.data_ptr(), w.data_ptr(),
+                                 c_buf.data_ptr(), batch, n, k, dep)
             ws_ptr = 0
             if ksplit > 1 and k % (64 * ksplit) == 0:
                 key = (c_buf.data_ptr(), n)
