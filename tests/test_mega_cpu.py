@@ -116,9 +116,9 @@ def test_emit_gemv_semantics():
         for t in g.tasks:
             tt, args = t[0], t[6]
             assert tt == T_GEMV
-            _, _, _, m, nn, kk, c0, chunk = args
+            _, _, _, m, nn, kk, c0, chunk = args[:8]
             assert (m, nn, kk) == (batch, n, k)
             C[:, c0:c0 + chunk] = A @ W[c0:c0 + chunk].T
             cols += list(range(c0, c0 + chunk))
         assert sorted(cols) == list(range(n))
-        np.testing.assert_allclose(C, A @ W.T, rtol=1e-5)
+        np.testing.assert_allclose(C, A @ W.T, rtol=1e-4, atol=1e-3)
