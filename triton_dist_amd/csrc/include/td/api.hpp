@@ -208,7 +208,8 @@ void launch_moe_grouped_gemm_pq(const void *xin, const void *weights,
                                 hipStream_t stream,
                                 const void *eflags = nullptr,
                                 const void *val_cell = nullptr,
-                                int world = 0, int e_loc = 0);
+                                int world = 0, int e_loc = 0,
+                                int fuse_swiglu = 0);
 void launch_moe_dispatch(const PeerTable &pt, const void *x,
                          const void *topk_ids, const void *send_pos,
                          const void *send_base, const void *counts,
@@ -239,7 +240,8 @@ void launch_moe_grouped_gemm(const void *xin, const void *weights, void *out,
                              int cap_rows, hipStream_t stream,
                              bool small_m = false,
                              const void *eflags = nullptr,
-                             const void *val_cell = nullptr, int world = 0);
+                             const void *val_cell = nullptr, int world = 0,
+                             int fuse_swiglu = 0);
 void launch_moe_combine_send(const PeerTable &pt, const void *expert_out,
                              const void *meta, const void *recv_total,
                              const void *recv_from_src, size_t combine_off,

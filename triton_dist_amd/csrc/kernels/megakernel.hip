@@ -546,10 +546,17 @@ TD_DEV void t_gemv(const Task &t, char *lds) {
   for (int i = threadIdx.x * 8; i < m * k; i += blockDim.x * 8)
     *(bf16x8 *)(x_lds + i) = *(const bf16x8 *)(A + i);
   __syncthreads();
-  for (int c = col0 + threadIdx.x; c < col0 + ncols; c += blockDim.x) {
+  // wave-per-column: 64 lanes stream one weight row's K contiguously
+  // (coalesced 1 KiB per load instruction), shuffle-tree reduce, lane 0
+  // stores — the column-per-thread layout was fully uncoalesced.
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int nwaves = blockDim.x >> 6;
+  for (int cb = wave; cb < ncols; cb += nwaves) {
+    const int c = col0 + cb;
     const bf16 *wr = W + (size_t)c * k;
     float acc[4] = {0.f, 0.f, 0.f, 0.f};
-    for (int kk = 0; kk < k; kk += 8) {
+    for (int kk = lane * 8; kk < k; kk += 64 * 8) {
       bf16x8 w8 = *(const bf16x8 *)(wr + kk);
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
@@ -561,7 +568,15 @@ TD_DEV void t_gemv(const Task &t, char *lds) {
         }
       }
     }
-    for (int r = 0; r < m; ++r) C[(size_t)r * n + c] = (bf16)acc[r];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      if (r < m) {
+#pragma unroll
+        for (int off = 32; off > 0; off >>= 1)
+          acc[r] += __shfl_xor(acc[r], off);
+        if (lane == 0) C[(size_t)r * n + c] = (bf16)acc[r];
+      }
+    }
   }
 }
 

@@ -419,7 +419,7 @@ __global__ __launch_bounds__(gg::NTH) void k_moe_grouped_gemm(
     bf16 *__restrict__ out, const int *__restrict__ expert_base,
     const int *__restrict__ expert_rows, int cap_tiles_m, int n, int k,
     int cap_rows, const int *__restrict__ eflags, const int *val_cell,
-    int world, int e_loc) {
+    int world, int e_loc, int fuse_swiglu) {
   const int tiles_n = n / gg::BN;
   const int e = blockIdx.x / cap_tiles_m;
   const int tm = blockIdx.x % cap_tiles_m;
@@ -490,7 +490,9 @@ __global__ __launch_bounds__(gg::NTH) void k_moe_grouped_gemm(
     }
     __syncthreads();
   }
-  // masked direct store (tail rows beyond the expert's count skipped)
+  // masked direct store (tail rows beyond the expert's count skipped);
+  // fuse_swiglu: interleaved gate/up columns pair via adjacent lanes —
+  // act = silu(gate) * up written at col/2 of the HALF-width output
   const int row_lim = rows - tm * gg::BM;
   const int actual_base = base + tm * gg::BM;
 #pragma unroll
@@ -501,113 +503,18 @@ __global__ __launch_bounds__(gg::NTH) void k_moe_grouped_gemm(
       for (int r = 0; r < 4; ++r) {
         int row = wr * 64 + i * 16 + (lane >> 4) * 4 + r;
         int col = wc * 64 + j * 16 + (lane & 15);
-        if (row < row_lim) {
+        if (fuse_swiglu) {
+          float g = acc[i][j][r];
+          float partner = __shfl_xor(g, 1);
+          if (((lane & 1) == 0) && row < row_lim) {
+            float silu = g / (1.f + __expf(-g));
+            out[((size_t)actual_base + row) * (n / 2) +
+                ((size_t)tn * gg::BN + col) / 2] = (bf16)(silu * partner);
+          }
+        } else if (row < row_lim) {
           out[((size_t)actual_base + row) * n + (size_t)tn * gg::BN + col] =
               (bf16)acc[i][j][r];
         }
-      }
-}
-
-// Small-M grouped GEMM: BM=32 tile (decode MoE: ~T*K*world/E rows per
-// expert). 3-buffer pipelined K-loop with counted vmcnt (5 loads/thread
-// per K-step; 2 steps in flight; one barrier per step) — same discipline
-// as gemm256.hip. 60 KB LDS -> 2 blocks/CU.
-__global__ __launch_bounds__(256) void k_moe_grouped_gemm_sm(
-    const bf16 *__restrict__ xin, const bf16 *__restrict__ weights,
-    bf16 *__restrict__ out, const int *__restrict__ expert_base,
-    const int *__restrict__ expert_rows, int cap_tiles_m, int n, int k,
-    int cap_rows) {
-  constexpr int BM = 32, BN = 128, BK = 64;
-  constexpr int ABUF = BM * BK, BBUF = BN * BK;
-  const int e = blockIdx.x / cap_tiles_m;
-  const int tm = blockIdx.x % cap_tiles_m;
-  const int tn = blockIdx.y;
-  const int rows = expert_rows[e];
-  if (tm * BM >= rows) return;
-  const int base = expert_base[e];
-
-  __shared__ bf16 lds_a[3 * ABUF];   // 3 x 4 KB
-  __shared__ bf16 lds_b[3 * BBUF];   // 3 x 16 KB
-  const int tid = threadIdx.x;
-  const int wave = tid >> 6, lane = tid & 63;
-  f32x4 acc[2][2] = {};
-  const bf16 *ga = xin + (size_t)(base + tm * BM) * k;
-  const bf16 *gb = weights + (size_t)e * n * k + (size_t)tn * BN * k;
-  const int ksteps = k / BK;
-
-  auto stage = [&](int t, int buf) {
-    const int k0 = t * BK;
-    {
-      int row = tid >> 3, kc = tid & 7;
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) unsigned int *)(
-              ga + (size_t)row * k + k0 + kc * 8),
-          (__attribute__((address_space(3))) unsigned int *)(
-              lds_a + buf * ABUF + (wave * 64) * 8),
-          16, 0, 0);
-    }
-#pragma unroll
-    for (int it = 0; it < 4; ++it) {
-      int qb = it * 256 + tid;
-      int rowb = qb >> 3, kcb = qb & 7;
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) unsigned int *)(
-              gb + (size_t)rowb * k + k0 + kcb * 8),
-          (__attribute__((address_space(3))) unsigned int *)(
-              lds_b + buf * BBUF + (it * 256 + wave * 64) * 8),
-          16, 0, 0);
-    }
-  };
-
-  stage(0, 0);
-  if (ksteps > 1) stage(1, 1);
-  for (int t = 0; t < ksteps; ++t) {
-    const int buf = t % 3;
-    // retire step t's 5 loads: outstanding = {t, t+1} -> vmcnt(5)
-    if (t + 1 < ksteps) {
-      asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
-    } else {
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    }
-    __builtin_amdgcn_s_barrier();
-    __builtin_amdgcn_sched_barrier(0);
-    if (t + 2 < ksteps) stage(t + 2, (t + 2) % 3);
-#pragma unroll
-    for (int ks = 0; ks < BK / 32; ++ks) {
-      bf16x8 af[2], bfr[2];
-#pragma unroll
-      for (int i = 0; i < 2; ++i) {
-        int arow = i * 16 + (lane & 15);
-        int brow = wave * 32 + i * 16 + (lane & 15);
-        int kk = ks * 32 + (lane >> 4) * 8;
-        af[i] = *(const bf16x8 *)(lds_a + buf * ABUF + arow * BK + kk);
-        bfr[i] = *(const bf16x8 *)(lds_b + buf * BBUF + brow * BK + kk);
-      }
-#pragma unroll
-      for (int i = 0; i < 2; ++i)
-#pragma unroll
-        for (int j = 0; j < 2; ++j)
-          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af[i], bfr[j], acc[i][j], 0, 0, 0);
-    }
-    // next iteration's barrier orders buffer reuse (reads retired by the
-    // compiler-inserted lgkmcnt before each MFMA)
-    __builtin_amdgcn_s_barrier();
-    __builtin_amdgcn_sched_barrier(0);
-  }
-  const int row_lim = rows - tm * BM;
-  const int actual_base = base + tm * BM;
-#pragma unroll
-  for (int i = 0; i < 2; ++i)
-#pragma unroll
-    for (int j = 0; j < 2; ++j)
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int row = i * 16 + (lane >> 4) * 4 + r;
-        int col = wave * 32 + j * 16 + (lane & 15);
-        if (row < row_lim)
-          out[((size_t)actual_base + row) * n + (size_t)tn * BN + col] =
-              (bf16)acc[i][j][r];
       }
 }
 
@@ -621,7 +528,7 @@ __global__ __launch_bounds__(256) void k_moe_grouped_gemm_pq(
     const int *__restrict__ expert_rows, const int *__restrict__ work_items,
     const int *__restrict__ work_count, int n, int k,
     const int *__restrict__ eflags, const int *val_cell, int world,
-    int e_loc) {
+    int e_loc, int fuse_swiglu) {
   constexpr int BM = 32, BN = 128, BK = 64;
   constexpr int ABUF = BM * BK, BBUF = BN * BK;
   __shared__ bf16 lds_a[3 * ABUF];
@@ -716,9 +623,24 @@ __global__ __launch_bounds__(256) void k_moe_grouped_gemm_pq(
         for (int r = 0; r < 4; ++r) {
           int row = i * 16 + (lane >> 4) * 4 + r;
           int col = wave * 32 + j * 16 + (lane & 15);
-          if (row < row_lim)
+          if (fuse_swiglu) {
+            // interleaved gate/up columns: even col = gate, odd = up
+            // (adjacent lanes) -> act = silu(gate) * up, written by the
+            // even lane at col/2 of the HALF-width output. Fuses the
+            // SwiGLU kernel into the epilogue and halves the activation
+            // round-trip (reference group_gemm epilogue-fusion
+            // capability — behavior only).
+            float g = acc[i][j][r];
+            float partner = __shfl_xor(g, 1);
+            if (((lane & 1) == 0) && row < row_lim) {
+              float silu = g / (1.f + __expf(-g));
+              out[((size_t)actual_base + row) * (n / 2) +
+                  ((size_t)tn * BN + col) / 2] = (bf16)(silu * partner);
+            }
+          } else if (row < row_lim) {
             out[((size_t)actual_base + row) * n + (size_t)tn * BN + col] =
                 (bf16)acc[i][j][r];
+          }
         }
     __syncthreads();  // LDS reuse across work items
   }
@@ -912,7 +834,7 @@ void launch_moe_grouped_gemm_pq(const void *xin, const void *weights,
                                 const void *work_count, int n, int k,
                                 hipStream_t stream, const void *eflags,
                                 const void *val_cell, int world,
-                                int e_loc) {
+                                int e_loc, int fuse_swiglu) {
   if (n % 128 || k % 64)
     throw std::runtime_error("grouped gemm pq: N%128/K%64 required");
   hipLaunchKernelGGL(k_moe_grouped_gemm_pq, dim3(1024), dim3(256), 0,
@@ -920,7 +842,7 @@ void launch_moe_grouped_gemm_pq(const void *xin, const void *weights,
                      (bf16 *)out, (const int *)expert_base,
                      (const int *)expert_rows, (const int *)work_items,
                      (const int *)work_count, n, k, (const int *)eflags,
-                     (const int *)val_cell, world, e_loc);
+                     (const int *)val_cell, world, e_loc, fuse_swiglu);
 }
 
 void launch_moe_grouped_gemm(const void *xin, const void *weights, void *out,
@@ -928,7 +850,8 @@ void launch_moe_grouped_gemm(const void *xin, const void *weights, void *out,
                              int e_loc, int cap_tiles_m, int n, int k,
                              int cap_rows, hipStream_t stream,
                              bool small_m, const void *eflags,
-                             const void *val_cell, int world) {
+                             const void *val_cell, int world,
+                             int fuse_swiglu) {
   if (n % 128 || k % 64)
     throw std::runtime_error("grouped gemm: N%128/K%64 required");
   (void)small_m;
@@ -938,7 +861,7 @@ void launch_moe_grouped_gemm(const void *xin, const void *weights, void *out,
                      (bf16 *)out, (const int *)expert_base,
                      (const int *)expert_rows, cap_tiles_m, n, k, cap_rows,
                      (const int *)eflags, (const int *)val_cell, world,
-                     e_loc);
+                     e_loc, fuse_swiglu);
 }
 
 void launch_moe_combine_send(const PeerTable &pt, const void *expert_out,

@@ -650,12 +650,12 @@ static void moe_grouped_gemm_pq(uintptr_t xin, uintptr_t weights,
                                 uintptr_t work_count, int n, int k,
                                 uintptr_t stream, uintptr_t eflags = 0,
                                 uintptr_t val_cell = 0, int world = 0,
-                                int e_loc = 0) {
+                                int e_loc = 0, int fuse_swiglu = 0) {
   launch_moe_grouped_gemm_pq((void *)xin, (void *)weights, (void *)out,
                              (void *)expert_base, (void *)expert_rows,
                              (void *)work_items, (void *)work_count, n, k,
                              as_stream(stream), (void *)eflags,
-                             (void *)val_cell, world, e_loc);
+                             (void *)val_cell, world, e_loc, fuse_swiglu);
   TD_CHECK_HIP(hipGetLastError());
 }
 
@@ -731,11 +731,13 @@ static void moe_grouped_gemm(uintptr_t xin, uintptr_t weights, uintptr_t out,
                              int e_loc, int cap_tiles_m, int n, int k,
                              int cap_rows, uintptr_t stream,
                              bool small_m = false, uintptr_t eflags = 0,
-                             uintptr_t val_cell = 0, int world = 0) {
+                             uintptr_t val_cell = 0, int world = 0,
+                             int fuse_swiglu = 0) {
   launch_moe_grouped_gemm((void *)xin, (void *)weights, (void *)out,
                           (void *)expert_base, (void *)expert_rows, e_loc,
                           cap_tiles_m, n, k, cap_rows, as_stream(stream),
-                          small_m, (void *)eflags, (void *)val_cell, world);
+                          small_m, (void *)eflags, (void *)val_cell, world,
+                          fuse_swiglu);
   TD_CHECK_HIP(hipGetLastError());
 }
 
@@ -917,7 +919,8 @@ PYBIND11_MODULE(_C, m) {
         py::arg("expert_rows"), py::arg("work_items"),
         py::arg("work_count"), py::arg("n"), py::arg("k"),
         py::arg("stream"), py::arg("eflags") = 0, py::arg("val_cell") = 0,
-        py::arg("world") = 0, py::arg("e_loc") = 0);
+        py::arg("world") = 0, py::arg("e_loc") = 0,
+        py::arg("fuse_swiglu") = 0);
   m.def("moe_dispatch", &moe_dispatch, py::arg("x"), py::arg("topk_ids"),
         py::arg("send_pos"), py::arg("send_base"), py::arg("counts"),
         py::arg("recv_x_off"), py::arg("meta_off"), py::arg("eflags_off"),
@@ -941,7 +944,8 @@ PYBIND11_MODULE(_C, m) {
         py::arg("expert_rows"), py::arg("e_loc"), py::arg("cap_tiles_m"),
         py::arg("n"), py::arg("k"), py::arg("cap_rows"), py::arg("stream"),
         py::arg("small_m") = false, py::arg("eflags") = 0,
-        py::arg("val_cell") = 0, py::arg("world") = 0);
+        py::arg("val_cell") = 0, py::arg("world") = 0,
+        py::arg("fuse_swiglu") = 0);
   m.def("moe_combine_send", &moe_combine_send, py::arg("expert_out"),
         py::arg("meta"), py::arg("recv_total"), py::arg("recv_from_src"),
         py::arg("combine_off"), py::arg("cflags_off"), py::arg("arrive"),

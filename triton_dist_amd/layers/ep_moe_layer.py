@@ -37,6 +37,7 @@ class EPMoELayer:
         self.w_down = torch.empty(self.e_loc, hidden, moe_inter,
                                   device=device, dtype=dtype)
         self.ctx: Optional[EPContext] = None
+        self._swiglu_fused = False  # w_gate_up rows interleaved g0,u0,...
 
     def init_ctx(self, max_tokens: int, ctx: Optional[EPContext] = None):
         if ctx is None:
@@ -67,11 +68,31 @@ class EPMoELayer:
         return topk_ids.to(torch.int32).contiguous(), \
             topk_w.float().contiguous()
 
+    def _fuse_weights(self):
+        """Interleave w_gate_up rows [g0..gI, u0..uI] -> [g0, u0, g1, u1,
+        ...] IN PLACE (per expert, transient temp only) so the grouped
+        GEMM's epilogue can compute silu(gate)*up from adjacent output
+        columns (fused SwiGLU — skips the 2*inter activation
+        round-trip). GPU path only; disable with TD_EP_NO_FUSE=1."""
+        import os
+        if self._swiglu_fused or os.environ.get("TD_EP_NO_FUSE"):
+            return
+        perm = torch.empty(2 * self.inter, dtype=torch.long,
+                           device=self.w_gate_up.device)
+        perm[0::2] = torch.arange(self.inter, device=perm.device)
+        perm[1::2] = torch.arange(self.inter, device=perm.device) +             self.inter
+        for e in range(self.e_loc):
+            self.w_gate_up[e] = self.w_gate_up[e][perm]
+        self._swiglu_fused = True
+
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         """x: [T_local, H] (this rank's token shard) -> [T_local, H]."""
+        if x.is_cuda and not self._swiglu_fused:
+            self._fuse_weights()
         topk_ids, topk_w = self.route(x)
         return ep_moe_forward(x, topk_ids, topk_w, self.w_gate_up,
-                              self.w_down, self.ctx)
+                              self.w_down, self.ctx,
+                              fused_swiglu=self._swiglu_fused)
 
     __call__ = forward
 
@@ -94,7 +115,10 @@ class EPMoELayer:
             tok, kk = sel.nonzero(as_tuple=True)
             xe = x[tok].float()
             h = xe @ self.w_gate_up[le].float().t()
-            a = F.silu(h[:, :self.inter]) * h[:, self.inter:]
+            if self._swiglu_fused:
+                a = F.silu(h[:, 0::2]) * h[:, 1::2]
+            else:
+                a = F.silu(h[:, :self.inter]) * h[:, self.inter:]
             y = a @ self.w_down[le].float().t()
             acc.index_add_(0, tok, y * topk_w[tok, kk].unsqueeze(1))
         out = acc.to(self.dtype)

@@ -127,7 +127,8 @@ def create_ep_context(max_tokens: int, hidden: int, n_experts: int,
 def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
                    topk_w: torch.Tensor, w_gate_up: torch.Tensor,
                    w_down: torch.Tensor, ctx: EPContext,
-                   out: Optional[torch.Tensor] = None) -> torch.Tensor:
+                   out: Optional[torch.Tensor] = None,
+                   fused_swiglu: bool = False) -> torch.Tensor:
     """Full EP MoE layer for this rank's tokens.
 
     x: [T, H] bf16; topk_ids: [T, K] int32 (global expert ids, -1 = drop);
@@ -172,7 +173,8 @@ def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
     if world == 1:
         return _ep_moe_world1(x, topk_ids, topk_w, w_gate_up, w_down, ctx,
                               out, parity, recv_x_off, meta_off,
-                              combine_off, cflags_off, eflags_off)
+                              combine_off, cflags_off, eflags_off,
+                              fused_swiglu)
 
     # phase 0: buffer-reuse protection
     L["counts"].zero_()
@@ -265,26 +267,30 @@ def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
     # experts are lightly loaded (decode: ~T*K*world/E rows per expert)
     cap_tiles = (ctx.cap + bm - 1) // bm
     # +128-row slack everywhere an edge GEMM tile may over-read
-    expert_h = torch.empty(ctx.cap + 128, 2 * inter, dtype=torch.bfloat16,
-                           device=x.device)
+    expert_h = None if fused_swiglu else torch.empty(
+        ctx.cap + 128, 2 * inter, dtype=torch.bfloat16, device=x.device)
     gate = 0 if ctx.fp8 else heap.ptr(rank, eflags_off)
+    act = torch.empty(ctx.cap + 128, inter, dtype=torch.bfloat16,
+                      device=x.device)
+    g1_out = act if fused_swiglu else expert_h
+    fs = 1 if fused_swiglu else 0
     if small_m:
         _C.moe_grouped_gemm_pq(heap.ptr(rank, recv_x_off),
-                               w_gate_up.data_ptr(), expert_h.data_ptr(),
+                               w_gate_up.data_ptr(), g1_out.data_ptr(),
                                L["expert_base"].data_ptr(),
                                L["expert_rows"].data_ptr(),
                                L["work_items"].data_ptr(),
                                L["work_count"].data_ptr(), 2 * inter, H, s,
-                               gate, cell, world, e_loc)
+                               gate, cell, world, e_loc, fs)
     else:
         _C.moe_grouped_gemm(heap.ptr(rank, recv_x_off), w_gate_up.data_ptr(),
-                            expert_h.data_ptr(), L["expert_base"].data_ptr(),
+                            g1_out.data_ptr(), L["expert_base"].data_ptr(),
                             L["expert_rows"].data_ptr(), e_loc, cap_tiles,
                             2 * inter, H, ctx.cap, s, False, gate, cell,
-                            world)
-    act = torch.empty(ctx.cap + 128, inter, dtype=torch.bfloat16,
-                      device=x.device)
-    _C.swiglu(expert_h.data_ptr(), act.data_ptr(), ctx.cap + 128, inter, s)
+                            world, fs)
+    if not fused_swiglu:
+        _C.swiglu(expert_h.data_ptr(), act.data_ptr(), ctx.cap + 128, inter,
+                  s)
     expert_out = torch.empty(ctx.cap + 128, H, dtype=torch.bfloat16,
                              device=x.device)
     if small_m:
@@ -320,7 +326,7 @@ def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
 
 def _ep_moe_world1(x, topk_ids, topk_w, w_gate_up, w_down, ctx, out,
                    parity, recv_x_off, meta_off, combine_off, cflags_off,
-                   eflags_off):
+                   eflags_off, fused_swiglu=False):
     T, H = x.shape
     K = ctx.topk
     E, e_loc = ctx.n_experts, ctx.e_loc
@@ -353,23 +359,28 @@ def _ep_moe_world1(x, topk_ids, topk_w, w_gate_up, w_down, ctx, out,
                     eflags_off, L["arrive_e"].data_ptr(), T, K, H, e_loc,
                     E, s, 0)
     cap_tiles = (ctx.cap + bm - 1) // bm
-    expert_h = torch.empty(ctx.cap + 128, 2 * inter, dtype=torch.bfloat16,
-                           device=x.device)
+    act = torch.empty(ctx.cap + 128, inter, dtype=torch.bfloat16,
+                      device=x.device)
+    expert_h = None if fused_swiglu else torch.empty(
+        ctx.cap + 128, 2 * inter, dtype=torch.bfloat16, device=x.device)
+    g1_out = act if fused_swiglu else expert_h
+    fs = 1 if fused_swiglu else 0
     if small_m:
         _C.moe_grouped_gemm_pq(heap.ptr(rank, recv_x_off),
-                               w_gate_up.data_ptr(), expert_h.data_ptr(),
+                               w_gate_up.data_ptr(), g1_out.data_ptr(),
                                L["expert_base"].data_ptr(),
                                L["expert_rows"].data_ptr(),
                                L["work_items"].data_ptr(),
-                               L["work_count"].data_ptr(), 2 * inter, H, s)
+                               L["work_count"].data_ptr(), 2 * inter, H, s,
+                               0, 0, 1, e_loc, fs)
     else:
         _C.moe_grouped_gemm(heap.ptr(rank, recv_x_off), w_gate_up.data_ptr(),
-                            expert_h.data_ptr(), L["expert_base"].data_ptr(),
+                            g1_out.data_ptr(), L["expert_base"].data_ptr(),
                             L["expert_rows"].data_ptr(), e_loc, cap_tiles,
-                            2 * inter, H, ctx.cap, s, False)
-    act = torch.empty(ctx.cap + 128, inter, dtype=torch.bfloat16,
-                      device=x.device)
-    _C.swiglu(expert_h.data_ptr(), act.data_ptr(), ctx.cap + 128, inter, s)
+                            2 * inter, H, ctx.cap, s, False, 0, 0, 1, fs)
+    if not fused_swiglu:
+        _C.swiglu(expert_h.data_ptr(), act.data_ptr(), ctx.cap + 128, inter,
+                  s)
     expert_out = torch.empty(ctx.cap + 128, H, dtype=torch.bfloat16,
                              device=x.device)
     if small_m:
