@@ -597,7 +597,8 @@ TD_DEV void spin_ge_relaxed(const int *flag, int bound) {
 // ---------------------------------------------------------------------------
 __global__ __launch_bounds__(mk::NTH) void k_megakernel(
     const mk::Task *__restrict__ tasks, const int *__restrict__ queue,
-    const int *__restrict__ queue_off, int *__restrict__ scoreboard) {
+    const int *__restrict__ queue_off, int *__restrict__ scoreboard,
+    int fence_mode) {
   __shared__ char lds[61440];  // union: gemm 3-buf A/B | flash-decode state
   const int wg = blockIdx.x;
   const int q_lo = queue_off[wg], q_hi = queue_off[wg + 1];
@@ -656,8 +657,11 @@ __global__ __launch_bounds__(mk::NTH) void k_megakernel(
     __syncthreads();
     if (threadIdx.x == 0) {
       // agent scope: single GPU — the writeback makes this XCD's lines
-      // visible to consumer tasks on other XCDs
-      __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+      // visible to consumer tasks on other XCDs. fence_mode 1 elides it
+      // (TD_MK_FENCE=1: A/B probe of the per-task wbl2 cost — NOT a
+      // correct configuration, diagnosis only).
+      if (fence_mode == 0)
+        __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
       atomic_add<Scope::Gpu>(scoreboard + t.score_slot, 1);
     }
     __syncthreads();
@@ -666,10 +670,11 @@ __global__ __launch_bounds__(mk::NTH) void k_megakernel(
 
 void launch_megakernel(const void *tasks, const void *queue,
                        const void *queue_off, void *scoreboard, int n_wg,
-                       hipStream_t stream) {
+                       hipStream_t stream, int fence_mode) {
   hipLaunchKernelGGL(k_megakernel, dim3(n_wg), dim3(mk::NTH), 0, stream,
                      (const mk::Task *)tasks, (const int *)queue,
-                     (const int *)queue_off, (int *)scoreboard);
+                     (const int *)queue_off, (int *)scoreboard,
+                     fence_mode);
 }
 
 }  // namespace td

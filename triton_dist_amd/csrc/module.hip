@@ -767,9 +767,11 @@ static void moe_combine_reduce(uintptr_t combine_buf, uintptr_t topk_w,
 }
 
 static void megakernel(uintptr_t tasks, uintptr_t queue, uintptr_t queue_off,
-                       uintptr_t scoreboard, int n_wg, uintptr_t stream) {
-  launch_megakernel((void *)tasks, (void *)queue, (void *)queue_off,
-                    (void *)scoreboard, n_wg, as_stream(stream));
+                       uintptr_t scoreboard, int n_wg, uintptr_t stream,
+                       int fence_mode = 0) {
+  launch_megakernel((const void *)tasks, (const void *)queue,
+                    (const void *)queue_off, (void *)scoreboard, n_wg,
+                    as_stream(stream), fence_mode);
   TD_CHECK_HIP(hipGetLastError());
 }
 
@@ -956,7 +958,9 @@ PYBIND11_MODULE(_C, m) {
         py::arg("cflags"), py::arg("world"), py::arg("T"), py::arg("K"),
         py::arg("H"), py::arg("e_num"), py::arg("stream"),
         py::arg("val_cell") = 0);
-  m.def("megakernel", &megakernel);
+  m.def("megakernel", &megakernel, py::arg("tasks"), py::arg("queue"),
+        py::arg("queue_off"), py::arg("scoreboard"), py::arg("n_wg"),
+        py::arg("stream"), py::arg("fence_mode") = 0);
   m.def("rmsnorm", &rmsnorm);
   m.def("add_rmsnorm", &add_rmsnorm);
   m.def("swiglu", &swiglu);

@@ -168,10 +168,13 @@ class MegaRun:
         self.n_ops = len(graph.ops)
 
     def launch(self, stream=None):
+        import os
+
         from .. import _C
 
         s = stream or torch.cuda.current_stream()
         self.scoreboard.zero_()
         _C.megakernel(self.task_buf.data_ptr(), self.queue.data_ptr(),
                       self.queue_off.data_ptr(), self.scoreboard.data_ptr(),
-                      self.n_wg, s.cuda_stream)
+                      self.n_wg, s.cuda_stream,
+                      int(os.environ.get("TD_MK_FENCE", "0")))

@@ -62,8 +62,12 @@ class MegaQwen3Decode:
         def gemm(a_buf, w, c_buf, n, k, dep):
             # bsz<=4: GEMV route (one task per 512-col chunk; weight-BW
             # bound, ~25x fewer tasks). TD_MK_NO_GEMV=1 forces tiles.
+            # GEMV route is opt-in: at 512-col chunks an op spans only
+            # ~12 workgroups, collapsing per-op parallelism (measured
+            # 137 ms vs 20.6 ms tiles at bsz 1 — per-op latency is
+            # parallelism-bound, not task-count-bound)
             if (batch <= 4 and batch * k * 2 <= 61440
-                    and not os.environ.get("TD_MK_NO_GEMV")):
+                    and os.environ.get("TD_MK_GEMV")):
                 from .builder import emit_gemv
                 return emit_gemv(g, a_buf.data_ptr(), w.data_ptr(),
                                  c_buf.data_ptr(), batch, n, k, dep)
