@@ -379,3 +379,32 @@ def test_flash_prefill(b, s, qh, kvh, causal):
                             > mask[None, None, :, None], float("-inf"))
     ref_lse = torch.logsumexp(sc, dim=-1).permute(0, 2, 1)  # [b,s,qh]
     assert_allclose(lse, ref_lse, atol=2e-2, rtol=2e-2)
+
+
+@pytest.mark.parametrize("m,n,k,fuse", [
+    (512, 1280, 5120, 0),
+    (32, 128, 64, 0),
+    (512, 512, 27648, 0),
+    (256, 1024, 2048, 1),    # fused swiglu: interleaved gate/up pairs
+])
+def test_gemm_skinny(m, n, k, fuse):
+    """Skinny-M decode tier (gemm_skinny.hip) vs fp32 reference, incl.
+    the fused-SwiGLU epilogue on interleaved columns."""
+    import torch.nn.functional as F
+    from triton_dist_amd import _C
+    from triton_dist_amd.utils.testing import assert_allclose, bf16_gemm_tol
+
+    torch.manual_seed(m + n)
+    a = (torch.randn(m, k, device="cuda") / 8).to(torch.bfloat16)
+    w = (torch.randn(n, k, device="cuda") / 8).to(torch.bfloat16)
+    ncols = n // 2 if fuse else n
+    c = torch.empty(m, ncols, device="cuda", dtype=torch.bfloat16)
+    _C.gemm_skinny_bf16(a.data_ptr(), w.data_ptr(), c.data_ptr(), 0,
+                        m, n, k, fuse, torch.cuda.current_stream().cuda_stream)
+    torch.cuda.synchronize()
+    full = a.float() @ w.float().t()
+    if fuse:
+        ref = F.silu(full[:, 0::2]) * full[:, 1::2]
+    else:
+        ref = full
+    assert_allclose(c, ref, **bf16_gemm_tol(k))

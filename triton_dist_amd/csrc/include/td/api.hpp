@@ -106,6 +106,11 @@ void launch_gemm256_v2_bf16(const GemmArgs &args, hipStream_t stream);
 // quadrant walk + counted vmcnt drains); candidate production tier.
 void launch_gemm256_v3_bf16(const GemmArgs &args, hipStream_t stream);
 
+// kernels/gemm_skinny.hip — BM=32 x BN=128 direct tier for decode shapes
+// (no split-K ws round trip); optional fused-SwiGLU epilogue.
+void launch_gemm_skinny(const GemmArgs &g, int fuse_swiglu,
+                        hipStream_t stream);
+
 // kernels/gemm_splitk.hip ---------------------------------------------------
 void launch_gemm_splitk_bf16(const GemmArgs &g, float *ws, int splits,
                              hipStream_t stream);

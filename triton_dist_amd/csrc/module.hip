@@ -446,6 +446,15 @@ static void rs_reduce_bf16(uintptr_t segments, uintptr_t out, int world,
   TD_CHECK_HIP(hipGetLastError());
 }
 
+static void gemm_skinny_bf16(uintptr_t a, uintptr_t b, uintptr_t c,
+                             uintptr_t bias, int m, int n, int k,
+                             int fuse_swiglu, uintptr_t stream) {
+  GemmArgs args{(void *)a, (void *)b, (void *)c, (void *)bias,
+                m, n, k, k, k, n};
+  launch_gemm_skinny(args, fuse_swiglu, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
 static void gemm_splitk_bf16(uintptr_t a, uintptr_t b, uintptr_t c,
                              uintptr_t bias, uintptr_t ws, int m, int n,
                              int k, int splits, uintptr_t stream) {
@@ -891,6 +900,7 @@ PYBIND11_MODULE(_C, m) {
   m.def("ar_tile_consumer", &ar_tile_consumer);
   m.def("rs_reduce_bf16", &rs_reduce_bf16);
   m.def("gemm_splitk_bf16", &gemm_splitk_bf16);
+  m.def("gemm_skinny_bf16", &gemm_skinny_bf16);
   m.def("gemm256_sk_bf16", &gemm256_sk_bf16);
   m.def("gemm256_v2_bf16", &gemm256_v2_bf16);
   m.def("gemm256_v3_bf16", &gemm256_v3_bf16);
