@@ -131,12 +131,15 @@ def ag_gemm(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
         # ONE rank's segment, so m (per rank) must tile by 256
         eligible = (m % 256 == 0 and n % 256 == 0
                     and k % 128 == 0 and ctx.ag_arrive is not None)
+        import os
         if not eligible:
             method = "push"
         elif torch.cuda.is_current_stream_capturing():
             method = _ag_method_cached(m, n, k, world)
-        else:
+        elif os.environ.get("TD_AUTOTUNE_METHODS") == "1":
             method = _tune_ag_method(a, w, ctx, m, n, k, world)
+        else:
+            method = _ag_method_cached(m, n, k, world)
     if method == "fused":
         return _ag_gemm_fused(a, w, ctx, out, gathered_out)
     rows_per_chunk = m // chunks
@@ -225,13 +228,16 @@ def _ag_key(m, n, k, world):
 
 
 def _ag_method_cached(m, n, k, world):
-    """Capture-safe lookup: cached tune result, else the size heuristic
-    (fused wins at decode-sized m: one launch vs ~20 stream ops)."""
+    """Capture-safe lookup: cached tune result, else a CONSERVATIVE
+    heuristic — fused only on configurations it has run on hardware
+    (<= 2 ranks); the stream path is the burn-in default for the
+    unattended 8-rank scaling run. Opt into live tuning with
+    TD_AUTOTUNE_METHODS=1 (cached results then win everywhere)."""
     t = _ag_tuner()._inner
     entry = t._mem.get(_ag_key(m, n, k, world))
     if entry:
         return entry["config"]["method"]
-    return "fused" if m <= 1024 else "push"
+    return "fused" if (m <= 1024 and world <= 2) else "push"
 
 
 def _tune_ag_method(a, w, ctx, m, n, k, world):
