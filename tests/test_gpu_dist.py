@@ -87,6 +87,28 @@ def _body_ag_gemm(rank, world):
         assert_allclose(c, ref, msg=method, **bf16_gemm_tol(k))
 
 
+def _body_ag_gemm_imperfect(rank, world):
+    from triton_dist_amd.ops import ag_gemm, ag_gemm_ref, create_ag_gemm_context
+    from triton_dist_amd.utils.testing import assert_allclose, bf16_gemm_tol
+
+    k, n = 1024, 384
+    ctx = create_ag_gemm_context(max_m_per_rank=1024, k=k,
+                                 chunks_per_rank=4)
+    for m in (320, 96, 513):  # none tile by 128 / divide chunks
+        torch.manual_seed(m + rank)
+        a = (torch.randn(m, k, device="cuda") / 8).to(torch.bfloat16)
+        torch.manual_seed(m)
+        w = (torch.randn(n, k, device="cuda") / 8).to(torch.bfloat16)
+        c = ag_gemm(a, w, ctx)
+        torch.cuda.synchronize()
+        ref = ag_gemm_ref(a, w)
+        assert_allclose(c, ref, msg=f"m={m}", **bf16_gemm_tol(k))
+
+
+def test_ag_gemm_imperfect_m_2rank():
+    run_distributed(_body_ag_gemm_imperfect, world_size=2)
+
+
 def test_ag_gemm_2rank():
     run_distributed(_body_ag_gemm, world_size=2)
 

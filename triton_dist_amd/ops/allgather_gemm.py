@@ -119,8 +119,11 @@ def ag_gemm(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
     heap, _C = ctx.heap, ctx.heap._C
     compute = torch.cuda.current_stream()
     chunks = ctx.chunks_per_rank
-    assert m % chunks == 0 and m % 128 == 0, \
-        f"m={m} must divide chunks={chunks} and tile by 128"
+    if m % chunks or m % 128:
+        # imperfect M: pad each rank's segment to the tile/chunk unit and
+        # compact the output (reference imperfect-chunk capability,
+        # allgather_gemm.py:487 — behavior only)
+        return _ag_gemm_padded(a, w, ctx, out, gathered_out)
     # single-fused-kernel paradigm (reference allgather_gemm.py:662-870
     # capability): producer workgroups + flag-waiting consumer GEMM in ONE
     # launch vs the SDMA stream-cooperative path. method="auto" resolves
@@ -245,6 +248,91 @@ def _tune_ag_method(a, w, ctx, m, n, k, world):
         _ag_key(m, n, k, world),
         lambda c: (lambda: ag_gemm(a, w, ctx, method=c["method"])))
     return cfg["method"]
+
+
+def _ag_gemm_padded(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
+                    out: Optional[torch.Tensor], gathered_out: bool):
+    """Arbitrary-M AG-GEMM: segments padded up to lcm(128, chunks) so
+    every consumer tile stays inside one rank segment; pad rows compute
+    garbage that the final per-segment compaction drops. Same
+    reset->barrier->publish protocol as the aligned push path; chunks
+    whose row range lies beyond m are signalled immediately (no data)."""
+    import math
+
+    m, k = a.shape
+    n = w.shape[0]
+    world, rank = ctx.world, ctx.rank
+    heap, _C = ctx.heap, ctx.heap._C
+    compute = torch.cuda.current_stream()
+    s = compute.cuda_stream
+    chunks = ctx.chunks_per_rank
+    unit = 128 * chunks // math.gcd(128, chunks)
+    m_pad = -(-m // unit) * unit
+    assert m_pad <= ctx.max_m_per_rank, \
+        f"padded m {m_pad} exceeds ctx.max_m_per_rank"
+    rows_per_chunk = m_pad // chunks
+
+    _C.reset_flags(ctx.flags.ptr(), world * chunks, 0, s)
+    heap.barrier_all_on_stream(compute)
+    my_seg_ptr = ctx.ws.ptr() + rank * ctx.max_m_per_rank * k * 2
+    _C.memcpy_async(my_seg_ptr, a.data_ptr(), m * k * 2, s)
+    _C.reset_flags(ctx.flags.ptr() + rank * chunks * 4, chunks, 1, s)
+    ctx.ready_ev.record(compute)
+    ns = len(ctx.comm_streams)
+    for st in range(ns):
+        ctx.comm_streams[st].wait_event(ctx.ready_ev)
+    for i in range(world - 1):
+        peer = (rank + 1 + i) % world
+        stream = ctx.comm_streams[i % ns]
+        dst_seg = ctx.ws.ptr(peer) + rank * ctx.max_m_per_rank * k * 2
+        dst_flag = ctx.flags.ptr(peer) + rank * chunks * 4
+        for c in range(chunks):
+            lo = c * rows_per_chunk
+            hi = min(lo + rows_per_chunk, m)
+            if hi > lo:
+                _C.memcpy_async(dst_seg + lo * k * 2,
+                                a.data_ptr() + lo * k * 2,
+                                (hi - lo) * k * 2, stream.cuda_stream)
+            _C.memcpy_async(dst_flag + c * 4, heap.one_src.ptr(), 4,
+                            stream.cuda_stream)
+
+    m_total_pad = world * m_pad
+    out_full = torch.empty(m_total_pad, n, dtype=torch.bfloat16,
+                           device=a.device)
+    from .gemm import choose_splits, splitk_ws
+
+    splits = choose_splits(m_total_pad, n, k)
+    if splits > 1:
+        ws = splitk_ws(m_total_pad, n, splits, a.device)
+        _C.ag_gemm_consumer_splitk_bf16(
+            ctx.ws.ptr(), w.data_ptr(), out_full.data_ptr(), ws.data_ptr(),
+            m_total_pad, n, k, ctx.flags.ptr(), chunks, m_pad,
+            ctx.max_m_per_rank, world, rank, 1, splits, s)
+    else:
+        _C.ag_gemm_consumer_bf16(
+            ctx.ws.ptr(), w.data_ptr(), out_full.data_ptr(), m_total_pad,
+            n, k, ctx.flags.ptr(), chunks, m_pad, ctx.max_m_per_rank,
+            world, rank, 1, s, 0, 0, 0)
+    for st in range(min(ns, max(world - 1, 1))):
+        ctx.join_evs[st].record(ctx.comm_streams[st])
+        compute.wait_event(ctx.join_evs[st])
+    # compact: drop the pad rows of every segment
+    if out is None:
+        out = torch.empty(world * m, n, dtype=torch.bfloat16,
+                          device=a.device)
+    for r in range(world):
+        _C.memcpy_async(out.data_ptr() + r * m * n * 2,
+                        out_full.data_ptr() + r * m_pad * n * 2,
+                        m * n * 2, s)
+    if gathered_out:
+        g = torch.empty(world * m, k, dtype=torch.bfloat16,
+                        device=a.device)
+        seg_bytes = ctx.max_m_per_rank * k * 2
+        for r in range(world):
+            _C.memcpy_async(g.data_ptr() + r * m * k * 2,
+                            ctx.ws.ptr() + r * seg_bytes, m * k * 2, s)
+        return out, g
+    return out
 
 
 def _ag_gemm_fused(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
