@@ -113,7 +113,11 @@ class TP_Attn:
         # prefill attention: the in-house MFMA FA2 kernel consumes the
         # natural [b, s, h, D] layout directly (no transposes); sdpa
         # remains only for CPU / non-128 head dims / masked decode
-        use_fa2 = (q.is_cuda and self.head_dim == 128
+        # FA2 measured 2.2-2.8x faster than sdpa at prefill s <= ~512
+        # (88.8 vs 198.8 us at b=32 s=128); sdpa/aotriton still wins at
+        # long context (s >= 2k) until the kernel pipelines KV tiles —
+        # route by sequence length (profiles/README.md r02)
+        use_fa2 = (q.is_cuda and self.head_dim == 128 and s <= 1024
                    and (kv_cache is None or prefill))
         if use_fa2:
             from ..ops.fused import flash_prefill_op

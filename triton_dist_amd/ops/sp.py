@@ -308,7 +308,7 @@ def _flash_with_lse(q, k, v, causal):
     """q/k/v: [B, S, H, D] bf16 -> (out [B,S,H,D], lse [B,S,H] fp32) via
     the in-house MFMA FA2 kernel (csrc/kernels/attention.hip
     k_flash_prefill); aten flash remains only for head_dim != 128."""
-    if q.is_cuda and q.shape[-1] == 128:
+    if q.is_cuda and q.shape[-1] == 128 and q.shape[1] <= 1024:
         from .fused import flash_prefill_op
 
         return flash_prefill_op(q, k, v, causal=causal, return_lse=True)
