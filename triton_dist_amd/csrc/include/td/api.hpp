@@ -284,7 +284,13 @@ void launch_qkv_prologue_decode(const void *qkv, void *q_out, void *kcache,
 void launch_flash_prefill(const void *q, const void *k, const void *v,
                           void *out, void *lse, int b, int s, int qh,
                           int kvh, float scale, bool causal,
-                          hipStream_t stream);
+                          hipStream_t stream, long kb_stride = 0);
+void launch_qkv_prologue_prefill(const void *qkv, void *q_out, void *kcache,
+                                 void *vcache, const void *cos_t,
+                                 const void *sin_t, const void *qnw,
+                                 const void *knw, int batch, int s, int qh,
+                                 int kvh, int max_len, float eps,
+                                 bool use_qk_norm, hipStream_t stream);
 void launch_flash_decode(const void *q, const void *kcache,
                          const void *vcache, void *out, const void *offset,
                          int batch, int qh, int kvh, int max_len,

@@ -596,7 +596,7 @@ __global__ __launch_bounds__(256) void k_flash_prefill(
     const bf16 *__restrict__ q, const bf16 *__restrict__ k,
     const bf16 *__restrict__ v, bf16 *__restrict__ out,
     float *__restrict__ lse, int s, int qh, int kvh, float scale,
-    int causal) {
+    int causal, long kb_stride) {
   const int q0 = blockIdx.x * 128;
   const int h = blockIdx.y;
   const int b = blockIdx.z;
@@ -649,7 +649,9 @@ __global__ __launch_bounds__(256) void k_flash_prefill(
       int pos = pos0 + r;
       bf16x8 kv8{}, vv8{};
       if (pos < s) {
-        size_t base = (((size_t)b * s + pos) * kvh + kh) * kD + c;
+        // kb_stride lets k/v live in the KV cache ([b, max_len, kvh, D])
+        size_t base =
+            (size_t)b * kb_stride + ((size_t)pos * kvh + kh) * kD + c;
         kv8 = *(const bf16x8 *)(k + base);
         vv8 = *(const bf16x8 *)(v + base);
       }
@@ -757,13 +759,14 @@ __global__ __launch_bounds__(256) void k_flash_prefill(
 void launch_flash_prefill(const void *q, const void *k, const void *v,
                           void *out, void *lse, int b, int s, int qh,
                           int kvh, float scale, bool causal,
-                          hipStream_t stream) {
+                          hipStream_t stream, long kb_stride) {
   if (qh % kvh) throw std::runtime_error("flash_prefill: qh % kvh != 0");
+  if (kb_stride == 0) kb_stride = (long)s * kvh * kD;
   dim3 grid((s + 127) / 128, qh, b);
   hipLaunchKernelGGL(k_flash_prefill, grid, dim3(256), 0, stream,
                      (const bf16 *)q, (const bf16 *)k, (const bf16 *)v,
                      (bf16 *)out, (float *)lse, s, qh, kvh, scale,
-                     causal ? 1 : 0);
+                     causal ? 1 : 0, kb_stride);
 }
 
 }  // namespace td

@@ -208,6 +208,87 @@ __global__ void k_qkv_prologue_decode(
   }
 }
 
+// Prefill variant: one block per (token row, head); position = the row's
+// index within its sequence (fresh prefill). q lands in the FA2-native
+// [b, s, qh, D] layout; k/v land straight in the KV cache (the FA2
+// consumer reads the cache through a batch stride) — replaces the eager
+// torch rmsnorm/rotary/copy soup on the prefill path.
+__global__ void k_qkv_prologue_prefill(
+    const bf16 *__restrict__ qkv, bf16 *__restrict__ q_out,
+    bf16 *__restrict__ kcache, bf16 *__restrict__ vcache,
+    const float *__restrict__ cos_t, const float *__restrict__ sin_t,
+    const bf16 *__restrict__ qnw, const bf16 *__restrict__ knw, int s,
+    int qh, int kvh, int max_len, float eps, int use_qk_norm) {
+  constexpr int D = 128;
+  const int row = blockIdx.x;     // b * s + t
+  const int b = row / s;
+  const long pos = row % s;
+  const int h = blockIdx.y;
+  const int lane = threadIdx.x;
+  const int nh = qh + 2 * kvh;
+  const bf16 *src = qkv + ((size_t)row * nh + h) * D;
+  float v0 = bf2f(src[lane * 2]);
+  float v1 = bf2f(src[lane * 2 + 1]);
+
+  const bool is_q = h < qh;
+  const bool is_k = h >= qh && h < qh + kvh;
+  if (is_q || is_k) {
+    if (use_qk_norm) {
+      float ss = v0 * v0 + v1 * v1;
+      for (int off = 32; off > 0; off >>= 1) ss += __shfl_down(ss, off);
+      float scale = rsqrtf(__shfl(ss, 0) / D + eps);
+      const bf16 *nw = is_q ? qnw : knw;
+      v0 *= scale * bf2f(nw[lane * 2]);
+      v1 *= scale * bf2f(nw[lane * 2 + 1]);
+    }
+    float p0 = __shfl_xor(v0, 32);
+    float p1 = __shfl_xor(v1, 32);
+    int d2 = (lane & 31) * 2;
+    float c0 = cos_t[pos * (D / 2) + d2];
+    float s0 = sin_t[pos * (D / 2) + d2];
+    float c1 = cos_t[pos * (D / 2) + d2 + 1];
+    float s1 = sin_t[pos * (D / 2) + d2 + 1];
+    float r0, r1;
+    if (lane < 32) {
+      r0 = v0 * c0 - p0 * s0;
+      r1 = v1 * c1 - p1 * s1;
+    } else {
+      r0 = v0 * c0 + p0 * s0;
+      r1 = v1 * c1 + p1 * s1;
+    }
+    v0 = r0;
+    v1 = r1;
+  }
+
+  if (is_q) {
+    bf16 *dst = q_out + ((size_t)row * qh + h) * D;
+    dst[lane * 2] = (bf16)v0;
+    dst[lane * 2 + 1] = (bf16)v1;
+  } else {
+    const int kh = h - qh;
+    const bool k_side = kh < kvh;
+    const int hh = k_side ? kh : kh - kvh;
+    bf16 *cache = k_side ? kcache : vcache;
+    bf16 *dst = cache + (((size_t)b * max_len + pos) * kvh + hh) * D;
+    dst[lane * 2] = (bf16)v0;
+    dst[lane * 2 + 1] = (bf16)v1;
+  }
+}
+
+void launch_qkv_prologue_prefill(const void *qkv, void *q_out, void *kcache,
+                                 void *vcache, const void *cos_t,
+                                 const void *sin_t, const void *qnw,
+                                 const void *knw, int batch, int s, int qh,
+                                 int kvh, int max_len, float eps,
+                                 bool use_qk_norm, hipStream_t stream) {
+  hipLaunchKernelGGL(k_qkv_prologue_prefill, dim3(batch * s, qh + 2 * kvh),
+                     dim3(64), 0, stream, (const bf16 *)qkv, (bf16 *)q_out,
+                     (bf16 *)kcache, (bf16 *)vcache, (const float *)cos_t,
+                     (const float *)sin_t, (const bf16 *)qnw,
+                     (const bf16 *)knw, s, qh, kvh, max_len, eps,
+                     use_qk_norm ? 1 : 0);
+}
+
 void launch_qkv_prologue_decode(const void *qkv, void *q_out, void *kcache,
                                 void *vcache, const void *cos_t,
                                 const void *sin_t, const void *qnw,

@@ -824,10 +824,25 @@ static void qkv_prologue_decode(uintptr_t qkv, uintptr_t q_out,
 static void flash_prefill(uintptr_t q, uintptr_t k, uintptr_t v,
                           uintptr_t out, int b, int s, int qh, int kvh,
                           float scale, bool causal, uintptr_t stream,
-                          uintptr_t lse = 0) {
+                          uintptr_t lse = 0, long kb_stride = 0) {
   launch_flash_prefill((const void *)q, (const void *)k, (const void *)v,
                        (void *)out, (void *)lse, b, s, qh, kvh, scale,
-                       causal, as_stream(stream));
+                       causal, as_stream(stream), kb_stride);
+  TD_CHECK_HIP(hipGetLastError());
+}
+
+static void qkv_prologue_prefill(uintptr_t qkv, uintptr_t q_out,
+                                 uintptr_t kcache, uintptr_t vcache,
+                                 uintptr_t cos_t, uintptr_t sin_t,
+                                 uintptr_t qnw, uintptr_t knw, int batch,
+                                 int s, int qh, int kvh, int max_len,
+                                 float eps, bool use_qk_norm,
+                                 uintptr_t stream) {
+  launch_qkv_prologue_prefill(
+      (const void *)qkv, (void *)q_out, (void *)kcache, (void *)vcache,
+      (const void *)cos_t, (const void *)sin_t, (const void *)qnw,
+      (const void *)knw, batch, s, qh, kvh, max_len, eps, use_qk_norm,
+      as_stream(stream));
   TD_CHECK_HIP(hipGetLastError());
 }
 
@@ -912,7 +927,8 @@ PYBIND11_MODULE(_C, m) {
   m.def("flash_prefill", &flash_prefill, py::arg("q"), py::arg("k"),
         py::arg("v"), py::arg("out"), py::arg("b"), py::arg("s"),
         py::arg("qh"), py::arg("kvh"), py::arg("scale"), py::arg("causal"),
-        py::arg("stream"), py::arg("lse") = 0);
+        py::arg("stream"), py::arg("lse") = 0, py::arg("kb_stride") = 0);
+  m.def("qkv_prologue_prefill", &qkv_prologue_prefill);
   m.def("ll_allgather", &ll_allgather_op);
   m.def("all_to_all", &all_to_all_op);
   m.def("ag_gemm_consumer_splitk_bf16", &ag_gemm_consumer_splitk_bf16);

@@ -105,6 +105,23 @@ class TP_Attn:
             return flash_decode_op(q_rot, kv_cache.k[layer_idx],
                                    kv_cache.v[layer_idx], kv_cache.offset,
                                    self.qh, self.kvh)
+        # fully fused prefill prologue + FA2 (fresh-prefill cache fill):
+        # one kernel does qk-norm + RoPE + cache fill; FA2 reads K/V from
+        # the cache through the batch stride — the torch rmsnorm/rotary/
+        # transpose/copy chain is gone from the hot path
+        if (qkv.is_cuda and self.head_dim == 128 and prefill
+                and kv_cache is not None and s > 1 and s <= 1024):
+            from ..ops.fused import flash_prefill_op, qkv_prologue_prefill_op
+
+            kc = kv_cache.k[layer_idx]
+            vc = kv_cache.v[layer_idx]
+            q4 = qkv_prologue_prefill_op(
+                qkv.to(self.dtype), kc, vc, self.rotary.cos,
+                self.rotary.sin, self.q_norm_w, self.k_norm_w, b, s,
+                self.qh, self.kvh, self.rms_eps, self.qk_norm)
+            kb = kc.shape[1] * self.kvh * self.head_dim
+            o = flash_prefill_op(q4, kc, vc, causal=True, kb_stride=kb)
+            return o.reshape(b * s, self.qh * self.head_dim)
         q, k, v = self._qkv_split(qkv, b, s)
         if self.qk_norm:
             q = rms_norm(q, self.q_norm_w, self.rms_eps)

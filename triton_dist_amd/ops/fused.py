@@ -79,8 +79,29 @@ def qkv_prologue_decode_op(qkv: torch.Tensor, kv_k: torch.Tensor,
     return q_out
 
 
+def qkv_prologue_prefill_op(qkv: torch.Tensor, kv_k: torch.Tensor,
+                            kv_v: torch.Tensor, cos_t: torch.Tensor,
+                            sin_t: torch.Tensor, qnw: torch.Tensor,
+                            knw: torch.Tensor, b: int, s: int, qh: int,
+                            kvh: int, eps: float,
+                            use_qk_norm: bool) -> torch.Tensor:
+    """Fused prefill prologue: per-head RMSNorm (optional) + RoPE + KV
+    cache fill, straight from the qkv projection. qkv: [b*s, (qh+2kvh)*
+    128]; kv_k/kv_v: [b, max_len, kvh, 128]. Returns q [b, s, qh, 128]
+    (the FA2-native layout). Replaces the eager torch rmsnorm/rotary/
+    transpose/copy chain on the prefill path."""
+    max_len = kv_k.shape[1]
+    q_out = torch.empty(b, s, qh, 128, dtype=qkv.dtype, device=qkv.device)
+    _C().qkv_prologue_prefill(
+        qkv.data_ptr(), q_out.data_ptr(), kv_k.data_ptr(), kv_v.data_ptr(),
+        cos_t.data_ptr(), sin_t.data_ptr(), qnw.data_ptr(), knw.data_ptr(),
+        b, s, qh, kvh, max_len, eps, use_qk_norm, _s())
+    return q_out
+
+
 def flash_prefill_op(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
-                     causal: bool = True, return_lse: bool = False):
+                     causal: bool = True, return_lse: bool = False,
+                     kb_stride: int = 0):
     """FA2 forward on MFMA (csrc/kernels/attention.hip k_flash_prefill).
 
     q: [b, s, qh, 128]; k/v: [b, s, kvh, 128] — the layer's natural
@@ -89,11 +110,14 @@ def flash_prefill_op(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     kernels/nvidia/sp_ag_attention_intra_node.py:257-428 FA2 consumer —
     behavior only)."""
     b, sq, qh, d = q.shape
-    kvh = k.shape[2]
-    assert d == 128 and k.shape[1] == sq
+    kvh = k.shape[-2]
+    assert d == 128
+    if kb_stride == 0:
+        assert k.shape[1] == sq
     q = q.contiguous()
-    k = k.contiguous()
-    v = v.contiguous()
+    if kb_stride == 0:
+        k = k.contiguous()
+        v = v.contiguous()
     out = torch.empty_like(q)
     scale = 1.0 / (d ** 0.5)
     lse = None
@@ -103,7 +127,7 @@ def flash_prefill_op(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
         lse_ptr = lse.data_ptr()
     _C().flash_prefill(q.data_ptr(), k.data_ptr(), v.data_ptr(),
                        out.data_ptr(), b, sq, qh, kvh, scale, causal,
-                       _s(), lse_ptr)
+                       _s(), lse_ptr, kb_stride)
     if return_lse:
         return out, lse
     return out
