@@ -222,6 +222,17 @@ void launch_moe_dispatch(const PeerTable &pt, const void *x,
                          size_t eflags_off, unsigned *arrive_e,
                          const void *val_cell, int T, int K, int H,
                          int e_loc, int e_num, hipStream_t stream);
+// Fused single-launch EP dispatch + per-expert-gated grouped GEMM
+// (reference ep_all2all_fused.py:316 mega-kernel — behavior only).
+void launch_moe_fused_dispatch_gemm(
+    const PeerTable &pt, const void *x, const void *topk_ids,
+    const void *send_pos, const void *send_base, const void *counts,
+    size_t recv_x_off, size_t meta_off, size_t eflags_off,
+    unsigned *arrive_e, const void *val_cell, int T, int K, int H,
+    int e_loc, int e_num, const void *weights, void *out,
+    const void *expert_base, const void *expert_rows,
+    const void *work_items, const void *work_count, int n, int k,
+    int fuse_swiglu, hipStream_t stream);
 void launch_moe_wait_flags(const void *flags, int world, const void *cell,
                            hipStream_t stream);
 void launch_bump_cell(void *cell, hipStream_t stream);

@@ -647,6 +647,216 @@ __global__ __launch_bounds__(256) void k_moe_grouped_gemm_pq(
 }
 
 // ---------------------------------------------------------------------------
+// Fused single-kernel EP dispatch + grouped GEMM (closes the reference's
+// mega-kernel row, kernels/amd/ep_all2all_fused.py:316 — behavior only):
+// the first `d_wgs` workgroups run the dispatch producer (grid-stride
+// token copies + per-expert arrive batching + empty-expert signalling);
+// the REMAINING workgroups run the per-expert-gated pq grouped GEMM in
+// the same launch. Producers sit at LOW workgroup ids so the dispatcher
+// makes them resident before consumers spin (same argument as the fused
+// AG-GEMM kernel).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_moe_fused_dispatch_gemm(
+    PeerTable pt, const bf16 *__restrict__ x,
+    const int *__restrict__ topk_ids, const int *__restrict__ send_pos,
+    const int *__restrict__ send_base, const int *__restrict__ counts,
+    size_t recv_x_off, size_t meta_off, size_t eflags_off,
+    unsigned *arrive_e, const int *val_cell, int T, int K, int H,
+    int e_loc, int e_num, int d_wgs,
+    // gemm side
+    const bf16 *__restrict__ weights, bf16 *__restrict__ out,
+    const int *__restrict__ expert_base, const int *__restrict__ expert_rows,
+    const int *__restrict__ work_items, const int *__restrict__ work_count,
+    int n, int k, int fuse_swiglu) {
+  if ((int)blockIdx.x < d_wgs) {
+    // ---- dispatch producer role ----
+    __shared__ int cnt[kMaxExperts];
+    for (int e = threadIdx.x; e < e_num; e += blockDim.x) cnt[e] = 0;
+    __syncthreads();
+    if (blockIdx.x == 0) {
+      // empty experts: signal every (dst, el) I send nothing to
+      for (int e = threadIdx.x; e < e_num; e += blockDim.x) {
+        if (counts[e] != 0) continue;
+        const int dst = e / e_loc;
+        int *fl = (int *)((char *)pt.bases[dst] + eflags_off);
+        st_release<Scope::Sys>(fl + pt.rank * e_loc + e % e_loc,
+                               val_cell ? *val_cell : 1);
+      }
+    }
+    for (int i = blockIdx.x; i < T * K; i += d_wgs) {
+      const int e = topk_ids[i];
+      const int pos = send_pos[i];
+      if (e < 0 || pos < 0) continue;
+      const int dst = e / e_loc;
+      const int t = i / K;
+      const int slot = send_base[e] + pos;
+      bf16 *rx =
+          (bf16 *)((char *)pt.bases[dst] + recv_x_off) + (size_t)slot * H;
+      const bf16 *src = x + (size_t)t * H;
+      for (int c = threadIdx.x * 8; c < H; c += blockDim.x * 8)
+        *(bf16x8 *)(rx + c) = *(const bf16x8 *)(src + c);
+      if (threadIdx.x == 0) {
+        int *meta = (int *)((char *)pt.bases[dst] + meta_off);
+        meta[slot * 2] = pt.rank;
+        meta[slot * 2 + 1] = i;
+        ++cnt[e];
+      }
+    }
+    __threadfence_block();
+    __syncthreads();
+    for (int e = threadIdx.x; e < e_num; e += blockDim.x) {
+      int c = cnt[e];
+      if (c == 0) continue;
+      unsigned prev = atomic_add<Scope::Gpu>(arrive_e + e, (unsigned)c);
+      if ((int)prev + c == counts[e]) {
+        const int dst = e / e_loc;
+        fence_release_sys();
+        int *fl = (int *)((char *)pt.bases[dst] + eflags_off);
+        st_release<Scope::Sys>(fl + pt.rank * e_loc + e % e_loc,
+                               val_cell ? *val_cell : 1);
+      }
+    }
+    return;
+  }
+  // ---- consumer role: per-expert-gated pq grouped GEMM ----
+  constexpr int BM = 32, BN = 128, BK = 64;
+  constexpr int ABUF = BM * BK, BBUF = BN * BK;
+  __shared__ bf16 lds_a[3 * ABUF];
+  __shared__ bf16 lds_b[3 * BBUF];
+  const bf16 *xin = (const bf16 *)((char *)pt.bases[pt.rank] + recv_x_off);
+  const int *eflags = (const int *)((char *)pt.bases[pt.rank] + eflags_off);
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int tiles_n = n / BN;
+  const int total = work_count[0] * tiles_n;
+  const int gate_v = val_cell ? *val_cell : 1;
+  const int gwgs = gridDim.x - d_wgs;
+
+  for (int wi = blockIdx.x - d_wgs; wi < total; wi += gwgs) {
+    const int item = work_items[wi / tiles_n];
+    const int tn = wi % tiles_n;
+    const int e = item >> 16;
+    const int tm = item & 0xFFFF;
+    if (tid < pt.world)
+      wait_ge_one<Scope::Sys>(eflags + tid * e_loc + e, gate_v);
+    __syncthreads();
+    const int rows = expert_rows[e];
+    const int base = expert_base[e];
+    f32x4 acc[2][2] = {};
+    const bf16 *ga = xin + (size_t)(base + tm * BM) * k;
+    const bf16 *gb = weights + (size_t)e * n * k + (size_t)tn * BN * k;
+    const int ksteps = k / BK;
+    auto stage = [&](int t, int buf) {
+      const int k0 = t * BK;
+      {
+        int row = tid >> 3, kc = tid & 7;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int *)(
+                ga + (size_t)row * k + k0 + kc * 8),
+            (__attribute__((address_space(3))) unsigned int *)(
+                lds_a + buf * ABUF + (wave * 64) * 8),
+            16, 0, 0);
+      }
+#pragma unroll
+      for (int it = 0; it < 4; ++it) {
+        int qb = it * 256 + tid;
+        int rowb = qb >> 3, kcb = qb & 7;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int *)(
+                gb + (size_t)rowb * k + k0 + kcb * 8),
+            (__attribute__((address_space(3))) unsigned int *)(
+                lds_b + buf * BBUF + (it * 256 + wave * 64) * 8),
+            16, 0, 0);
+      }
+    };
+    stage(0, 0);
+    if (ksteps > 1) stage(1, 1);
+    for (int t = 0; t < ksteps; ++t) {
+      const int buf = t % 3;
+      if (t + 1 < ksteps) {
+        asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
+      } else {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      }
+      __builtin_amdgcn_s_barrier();
+      __builtin_amdgcn_sched_barrier(0);
+      if (t + 2 < ksteps) stage(t + 2, (t + 2) % 3);
+#pragma unroll
+      for (int ks = 0; ks < BK / 32; ++ks) {
+        bf16x8 af[2], bfr[2];
+#pragma unroll
+        for (int i = 0; i < 2; ++i) {
+          int arow = i * 16 + (lane & 15);
+          int brow = wave * 32 + i * 16 + (lane & 15);
+          int kk = ks * 32 + (lane >> 4) * 8;
+          af[i] = *(const bf16x8 *)(lds_a + buf * ABUF + arow * BK + kk);
+          bfr[i] = *(const bf16x8 *)(lds_b + buf * BBUF + brow * BK + kk);
+        }
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+#pragma unroll
+          for (int j = 0; j < 2; ++j)
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[i], bfr[j], acc[i][j], 0, 0, 0);
+      }
+      __builtin_amdgcn_s_barrier();
+      __builtin_amdgcn_sched_barrier(0);
+    }
+    const int row_lim = rows - tm * BM;
+    const int actual_base = base + tm * BM;
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+      for (int j = 0; j < 2; ++j)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int row = i * 16 + (lane >> 4) * 4 + r;
+          int col = wave * 32 + j * 16 + (lane & 15);
+          if (fuse_swiglu) {
+            float g = acc[i][j][r];
+            float partner = __shfl_xor(g, 1);
+            if (((lane & 1) == 0) && row < row_lim) {
+              float silu = g / (1.f + __expf(-g));
+              out[((size_t)actual_base + row) * (n / 2) +
+                  ((size_t)tn * BN + col) / 2] = (bf16)(silu * partner);
+            }
+          } else if (row < row_lim) {
+            out[((size_t)actual_base + row) * n + (size_t)tn * BN + col] =
+                (bf16)acc[i][j][r];
+          }
+        }
+    __syncthreads();
+  }
+}
+
+void launch_moe_fused_dispatch_gemm(
+    const PeerTable &pt, const void *x, const void *topk_ids,
+    const void *send_pos, const void *send_base, const void *counts,
+    size_t recv_x_off, size_t meta_off, size_t eflags_off,
+    unsigned *arrive_e, const void *val_cell, int T, int K, int H,
+    int e_loc, int e_num, const void *weights, void *out,
+    const void *expert_base, const void *expert_rows,
+    const void *work_items, const void *work_count, int n, int k,
+    int fuse_swiglu, hipStream_t stream) {
+  if (H % 8 || n % 128 || k % 64)
+    throw std::runtime_error("moe fused: H%8/N%128/K%64 required");
+  if (e_num > kMaxExperts)
+    throw std::runtime_error("moe fused: e_num > 256");
+  int d_wgs = T * K < 256 ? T * K : 256;
+  if (d_wgs < 1) d_wgs = 1;
+  hipLaunchKernelGGL(k_moe_fused_dispatch_gemm, dim3(d_wgs + 768),
+                     dim3(256), 0, stream, pt, (const bf16 *)x,
+                     (const int *)topk_ids, (const int *)send_pos,
+                     (const int *)send_base, (const int *)counts,
+                     recv_x_off, meta_off, eflags_off, arrive_e,
+                     (const int *)val_cell, T, K, H, e_loc, e_num, d_wgs,
+                     (const bf16 *)weights, (bf16 *)out,
+                     (const int *)expert_base, (const int *)expert_rows,
+                     (const int *)work_items, (const int *)work_count, n,
+                     k, fuse_swiglu);
+}
+
+// ---------------------------------------------------------------------------
 // Phase 4: combine send — return expert outputs to their source ranks.
 // Block per recv row; signals each source when all its rows are returned.
 // ---------------------------------------------------------------------------

@@ -683,6 +683,25 @@ static void moe_dispatch(uintptr_t x, uintptr_t topk_ids, uintptr_t send_pos,
   TD_CHECK_HIP(hipGetLastError());
 }
 
+static void moe_fused_dispatch_gemm(
+    uintptr_t x, uintptr_t topk_ids, uintptr_t send_pos,
+    uintptr_t send_base, uintptr_t counts, size_t recv_x_off,
+    size_t meta_off, size_t eflags_off, uintptr_t arrive_e, int T, int K,
+    int H, int e_loc, int e_num, uintptr_t weights, uintptr_t out,
+    uintptr_t expert_base, uintptr_t expert_rows, uintptr_t work_items,
+    uintptr_t work_count, int n, int k, int fuse_swiglu, uintptr_t stream,
+    uintptr_t val_cell = 0) {
+  check_active();
+  launch_moe_fused_dispatch_gemm(
+      g_heap.pt, (void *)x, (void *)topk_ids, (void *)send_pos,
+      (void *)send_base, (void *)counts, recv_x_off, meta_off, eflags_off,
+      (unsigned *)arrive_e, (void *)val_cell, T, K, H, e_loc, e_num,
+      (void *)weights, (void *)out, (void *)expert_base,
+      (void *)expert_rows, (void *)work_items, (void *)work_count, n, k,
+      fuse_swiglu, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
 static void moe_wait_flags(uintptr_t flags, int world, uintptr_t stream,
                            uintptr_t cell = 0) {
   launch_moe_wait_flags((void *)flags, world, (void *)cell,
@@ -942,6 +961,16 @@ PYBIND11_MODULE(_C, m) {
         py::arg("recv_from_src"), py::arg("recv_total"), py::arg("stream"),
         py::arg("work_items") = 0, py::arg("work_count") = 0,
         py::arg("bm") = 128);
+  m.def("moe_fused_dispatch_gemm", &moe_fused_dispatch_gemm,
+        py::arg("x"), py::arg("topk_ids"), py::arg("send_pos"),
+        py::arg("send_base"), py::arg("counts"), py::arg("recv_x_off"),
+        py::arg("meta_off"), py::arg("eflags_off"), py::arg("arrive_e"),
+        py::arg("T"), py::arg("K"), py::arg("H"), py::arg("e_loc"),
+        py::arg("e_num"), py::arg("weights"), py::arg("out"),
+        py::arg("expert_base"), py::arg("expert_rows"),
+        py::arg("work_items"), py::arg("work_count"), py::arg("n"),
+        py::arg("k"), py::arg("fuse_swiglu"), py::arg("stream"),
+        py::arg("val_cell") = 0);
   m.def("moe_grouped_gemm_pq", &moe_grouped_gemm_pq, py::arg("xin"),
         py::arg("weights"), py::arg("out"), py::arg("expert_base"),
         py::arg("expert_rows"), py::arg("work_items"),

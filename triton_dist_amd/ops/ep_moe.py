@@ -246,7 +246,7 @@ def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
                        heap.ptr(rank, recv_s_off),
                        heap.ptr(rank, recv_x_off),
                        L["recv_total"].data_ptr(), ctx.cap, H, s)
-    else:
+    elif not (small_m and not ctx.low_latency):
         # dispatch rides the comm stream; the expert GEMM is gated
         # per-(expert, tile) on eflags, so FFN tiles of early-complete
         # experts run while slow sources still stream (per-expert
@@ -274,7 +274,19 @@ def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
                       device=x.device)
     g1_out = act if fused_swiglu else expert_h
     fs = 1 if fused_swiglu else 0
-    if small_m:
+    if small_m and not ctx.fp8 and not ctx.low_latency:
+        # SINGLE-LAUNCH mega-kernel: dispatch producer workgroups +
+        # per-expert-gated grouped GEMM in one kernel (closes the
+        # reference's ep_all2all_fused.py:316 row completely)
+        _C.moe_fused_dispatch_gemm(
+            x.data_ptr(), topk_ids.data_ptr(), L["send_pos"].data_ptr(),
+            L["send_base"].data_ptr(), L["counts"].data_ptr(), recv_x_off,
+            meta_off, eflags_off, L["arrive_e"].data_ptr(), T, K, H,
+            e_loc, E, w_gate_up.data_ptr(), g1_out.data_ptr(),
+            L["expert_base"].data_ptr(), L["expert_rows"].data_ptr(),
+            L["work_items"].data_ptr(), L["work_count"].data_ptr(),
+            2 * inter, H, fs, s, cell)
+    elif small_m:
         _C.moe_grouped_gemm_pq(heap.ptr(rank, recv_x_off),
                                w_gate_up.data_ptr(), g1_out.data_ptr(),
                                L["expert_base"].data_ptr(),
@@ -308,7 +320,7 @@ def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
 
     # phase 6: combine (return rows + weighted reduce); join the
     # dispatch fork first (graph hygiene — by now it long completed)
-    if not ctx.fp8:
+    if not ctx.fp8 and not (small_m and not ctx.low_latency):
         stream.wait_event(L["ev_join"])
     _C.moe_combine_send(expert_out.data_ptr(), heap.ptr(rank, meta_off),
                         L["recv_total"].data_ptr(),
