@@ -146,3 +146,44 @@ def test_ulysses_fused_cpu_2rank():
 
 def test_ulysses_fused_cpu_4rank():
     run_distributed(_body_ulysses_fused, world_size=4)
+
+
+def _body_zigzag(rank, world):
+    import torch
+    import torch.nn.functional as F
+    import torch.distributed as dist
+    from triton_dist_amd.ops import (create_sp_ag_attn_context,
+                                     sp_ag_attention_zigzag)
+    from triton_dist_amd.utils import assert_allclose
+
+    s_blk, qh, kvh, d = 8, 4, 2, 128
+    nb = 2 * world
+    ctx = create_sp_ag_attn_context(2 * s_blk, kvh, d)
+    torch.manual_seed(7)  # same full sequence everywhere
+    kf = (torch.randn(nb * s_blk, kvh, d) / 4).to(torch.bfloat16)
+    vf = (torch.randn(nb * s_blk, kvh, d) / 4).to(torch.bfloat16)
+    qf = (torch.randn(nb * s_blk, qh, d) / 4).to(torch.bfloat16)
+
+    def my_blocks(x):
+        b = x.reshape(nb, s_blk, x.shape[1], d)
+        return torch.stack([b[rank], b[nb - 1 - rank]], 0)
+
+    out = sp_ag_attention_zigzag(my_blocks(qf), my_blocks(kf),
+                                 my_blocks(vf), ctx, qh)
+    # golden: full causal attention, sliced at my two blocks
+    qt = qf.permute(1, 0, 2).unsqueeze(0).float()
+    kt = kf.permute(1, 0, 2).unsqueeze(0).float()
+    vt = vf.permute(1, 0, 2).unsqueeze(0).float()
+    ref = F.scaled_dot_product_attention(qt, kt, vt, is_causal=True,
+                                         enable_gqa=True)
+    ref = ref.squeeze(0).permute(1, 0, 2).reshape(nb, s_blk, qh, d)
+    want = torch.stack([ref[rank], ref[nb - 1 - rank]], 0)
+    assert_allclose(out, want.to(torch.bfloat16), atol=8e-2, rtol=5e-2)
+
+
+def test_sp_zigzag_cpu_2rank():
+    run_distributed(_body_zigzag, world_size=2)
+
+
+def test_sp_zigzag_cpu_4rank():
+    run_distributed(_body_zigzag, world_size=4)

@@ -48,6 +48,7 @@ from .sp import (  # noqa: F401
     SPAGAttnContext,
     create_sp_ag_attn_context,
     sp_ag_attention,
+    sp_ag_attention_zigzag,
 )
 from .p2p import (  # noqa: F401
     P2PContext,

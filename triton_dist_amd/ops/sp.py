@@ -614,3 +614,131 @@ def ulysses_fused_ref(x: torch.Tensor, w_qkv: torch.Tensor,
             outs = [g.clone() for g in gathered]
     mine = torch.cat(outs, 0)  # [world*t_loc, pc]
     return mine.to(x.device)
+
+
+def sp_ag_attention_zigzag(q: torch.Tensor, k_chunk: torch.Tensor,
+                           v_chunk: torch.Tensor, ctx: SPAGAttnContext,
+                           qh: int) -> torch.Tensor:
+    """Zig-zag-balanced causal SP attention (the reference's causal
+    load-balance capability, sp_ag_attention_intra_node.py:283 — behavior
+    only). The sequence is pre-sharded into 2*world blocks; rank r holds
+    blocks (r, 2*world-1-r) so every rank's attention work is the
+    constant 2*world+1 block-pairs instead of r+1.
+
+    q/k_chunk/v_chunk: [2, S_blk, heads, D] (this rank's two blocks,
+    ordered [block r, block 2W-1-r], post-RoPE). Returns [2, S_blk, qh,
+    D]. Requires 2*S_blk <= ctx.max_tokens (both blocks ride one slot).
+    """
+    two, s_blk, _, d = q.shape
+    assert two == 2 and d == ctx.head_dim
+    world, rank = ctx.world, ctx.rank
+    kvh = ctx.kvh
+    heap = ctx.heap
+    assert 2 * s_blk <= ctx.max_tokens
+
+    if heap.backend == "cpu":
+        return _sp_ag_attention_zigzag_cpu(q, k_chunk, v_chunk, ctx, qh)
+
+    _C = heap._C
+    compute = torch.cuda.current_stream()
+    s = compute.cuda_stream
+    _C.reset_flags(ctx.flags.ptr(), world, 0, s)
+    heap.barrier_all_on_stream(compute)
+    slot = ctx.max_tokens * kvh * d * 2
+    nbytes = 2 * s_blk * kvh * d * 2
+    kc = k_chunk.contiguous()
+    vc = v_chunk.contiguous()
+    _C.memcpy_async(ctx.kbuf.ptr() + rank * slot, kc.data_ptr(), nbytes, s)
+    _C.memcpy_async(ctx.vbuf.ptr() + rank * slot, vc.data_ptr(), nbytes, s)
+    _C.reset_flags(ctx.flags.ptr() + rank * 4, 1, 1, s)
+    ctx.ready_ev.record(compute)
+    ns = len(ctx.comm_streams)
+    for i in range(world - 1):
+        peer = (rank + 1 + i) % world
+        st = ctx.comm_streams[i % ns]
+        if i < ns:
+            st.wait_event(ctx.ready_ev)
+        _C.memcpy_async(ctx.kbuf.ptr(peer) + rank * slot, kc.data_ptr(),
+                        nbytes, st.cuda_stream)
+        _C.memcpy_async(ctx.vbuf.ptr(peer) + rank * slot, vc.data_ptr(),
+                        nbytes, st.cuda_stream)
+        _C.memcpy_async(ctx.flags.ptr(peer) + rank * 4, heap.one_src.ptr(),
+                        4, st.cuda_stream)
+
+    def kv_half(src, half):
+        base = ctx.kbuf.local()[src, :2 * s_blk].reshape(2, s_blk, kvh, d)
+        vb = ctx.vbuf.local()[src, :2 * s_blk].reshape(2, s_blk, kvh, d)
+        return base[half].unsqueeze(0), vb[half].unsqueeze(0)
+
+    outs = []
+    for half in range(2):
+        # q block's global index: half 0 -> rank; half 1 -> 2W-1-rank
+        qt = q[half].unsqueeze(0)
+        o = lse = None
+        # sources in (src, their half, causal?) order
+        plan = []
+        if half == 0:
+            for j in range(rank):
+                plan.append((j, 0, False))
+            plan.append((rank, 0, True))
+        else:
+            for j in range(world):
+                plan.append((j, 0, False))       # every A block
+            for j in range(rank + 1, world):
+                plan.append((j, 1, False))       # B blocks below mine
+            plan.append((rank, 1, True))
+        for src, kv_h, causal in plan:
+            _C.wait_eq(ctx.flags.ptr() + src * 4, 1, 1, s)
+            kk, vv = kv_half(src, kv_h)
+            oc, lc = _flash_with_lse(qt, kk, vv, causal)
+            o, lse = (oc, lc) if o is None else _merge_lse(o, lse, oc, lc)
+        outs.append(o.squeeze(0))
+    for i, ev in enumerate(ctx.join_evs):
+        ev.record(ctx.comm_streams[i])
+        compute.wait_event(ev)
+    return torch.stack(outs, 0).contiguous()
+
+
+def _sp_ag_attention_zigzag_cpu(q, k_chunk, v_chunk, ctx, qh):
+    import torch.nn.functional as F
+
+    two, s_blk, kvh, d = k_chunk.shape
+    world, rank = ctx.world, ctx.rank
+    heap = ctx.heap
+    ctx.epoch = getattr(ctx, "epoch", 0) + 1
+    heap.barrier_all()
+    flat_k = k_chunk.reshape(2 * s_blk, kvh, d)
+    flat_v = v_chunk.reshape(2 * s_blk, kvh, d)
+    for peer in range(world):
+        ctx.kbuf.peer(peer)[rank, :2 * s_blk].copy_(flat_k)
+        ctx.vbuf.peer(peer)[rank, :2 * s_blk].copy_(flat_v)
+        cpu_shm.notify(ctx.flags.peer(peer), rank, ctx.epoch)
+    fl = ctx.flags.local()
+    for r in range(world):
+        cpu_shm.wait_ge(fl, r, ctx.epoch)
+    # reassemble the FULL zig-zag sequence in global block order
+    nb = 2 * world
+    blocks_k = [None] * nb
+    blocks_v = [None] * nb
+    for src in range(world):
+        kb = ctx.kbuf.local()[src, :2 * s_blk].reshape(2, s_blk, kvh, d)
+        vb = ctx.vbuf.local()[src, :2 * s_blk].reshape(2, s_blk, kvh, d)
+        blocks_k[src] = kb[0]
+        blocks_k[nb - 1 - src] = kb[1]
+        blocks_v[src] = vb[0]
+        blocks_v[nb - 1 - src] = vb[1]
+    ks = torch.cat(blocks_k, 0)
+    vs = torch.cat(blocks_v, 0)
+    total = nb * s_blk
+    outs = []
+    for half, gidx in ((0, rank), (1, nb - 1 - rank)):
+        qt = q[half].permute(1, 0, 2).unsqueeze(0).float()
+        kt = ks.permute(1, 0, 2).unsqueeze(0).float()
+        vt = vs.permute(1, 0, 2).unsqueeze(0).float()
+        qpos = gidx * s_blk + torch.arange(s_blk)
+        mask = torch.arange(total).unsqueeze(0) <= qpos.unsqueeze(1)
+        o = F.scaled_dot_product_attention(
+            qt, kt, vt, attn_mask=mask.view(1, 1, s_blk, total),
+            enable_gqa=True)
+        outs.append(o.squeeze(0).permute(1, 0, 2).to(q.dtype))
+    return torch.stack(outs, 0)
