@@ -76,52 +76,85 @@ __global__ void k_moe_layout(const int *__restrict__ all_splits, int rank,
                              int *__restrict__ recv_total,
                              int *__restrict__ work_items,
                              int *__restrict__ work_count, int bm) {
-  // single block; e_num <= 1024 assumed. The scan itself is trivial; the
-  // cost was serial GLOBAL loads from one thread (~350 ns each, 46 us at
-  // E=128) — stage all_splits into LDS cooperatively first.
+  // single block, COOPERATIVE: the r01 thread-0 serial scan measured
+  // 37 us/call (1.8 ms/step at 48 layers); the Hillis-Steele scans
+  // below run the same algebra in ~2 us.
   __shared__ int sp[8 * 1024];
-  for (int i = threadIdx.x; i < world * e_num; i += blockDim.x)
+  __shared__ int off[1024];    // per-expert row totals (all sources)
+  __shared__ int pre[1024];    // exclusive prefix within each dst range
+  __shared__ int tmp[1024];
+  const int tid = threadIdx.x;
+  for (int i = tid; i < world * e_num; i += blockDim.x)
     sp[i] = all_splits[i];
   __syncthreads();
-  if (threadIdx.x == 0) {
-    for (int d = 0; d < world; ++d) {
-      int base = 0;  // running offset inside rank d's recv buffer
-      for (int le = 0; le < e_loc; ++le) {
-        int e = d * e_loc + le;
-        int off = 0;
-        for (int s = 0; s < world; ++s) {
-          if (s == rank) send_base[e] = base + off;
-          off += sp[s * e_num + e];
-        }
-        if (d == rank) {
-          expert_base[le] = base;
-          expert_rows[le] = off;
-        }
-        base += off;
-      }
-      if (d == rank) recv_total[0] = base;
+  for (int e = tid; e < e_num; e += blockDim.x) {
+    int o = 0;
+    for (int s2 = 0; s2 < world; ++s2) o += sp[s2 * e_num + e];
+    off[e] = o;
+  }
+  __syncthreads();
+  // exclusive prefix of off[] within each destination's e_loc range.
+  // Hillis-Steele over the FULL e_num then subtract range starts: since
+  // ranges are contiguous [d*e_loc, (d+1)*e_loc), a global exclusive
+  // scan minus the scan value at the range start gives the local one.
+  for (int e = tid; e < e_num; e += blockDim.x) tmp[e] = off[e];
+  __syncthreads();
+  for (int stride = 1; stride < e_num; stride <<= 1) {
+    for (int e = tid; e < e_num; e += blockDim.x)
+      pre[e] = e >= stride ? tmp[e] + tmp[e - stride] : tmp[e];
+    __syncthreads();
+    for (int e = tid; e < e_num; e += blockDim.x) tmp[e] = pre[e];
+    __syncthreads();
+  }
+  // tmp[e] = inclusive scan; exclusive local = tmp[e]-off[e]-tmp[d*e_loc-1]
+  for (int e = tid; e < e_num; e += blockDim.x) {
+    const int d = e / e_loc;
+    int range0 = d * e_loc == 0 ? 0 : tmp[d * e_loc - 1];
+    int excl = tmp[e] - off[e] - range0;
+    int below = 0;  // sources < rank within expert e's slot
+    for (int s2 = 0; s2 < rank; ++s2) below += sp[s2 * e_num + e];
+    send_base[e] = excl + below;
+    if (d == rank) {
+      expert_base[e - rank * e_loc] = excl;
+      expert_rows[e - rank * e_loc] = off[e];
     }
-    for (int s = 0; s < world; ++s) {
-      int r = 0;
-      for (int le = 0; le < e_loc; ++le)
-        r += sp[s * e_num + rank * e_loc + le];
-      recv_from_src[s] = r;
+  }
+  __syncthreads();
+  if (tid < world) {
+    int r = 0;
+    for (int le = 0; le < e_loc; ++le)
+      r += sp[tid * e_num + rank * e_loc + le];
+    recv_from_src[tid] = r;
+  }
+  if (tid == 0) {
+    int hi = (rank + 1) * e_loc - 1;
+    int lo0 = rank * e_loc == 0 ? 0 : tmp[rank * e_loc - 1];
+    recv_total[0] = tmp[hi] - lo0;
+  }
+  // work queue: one item per (expert, bm-row tile); tile counts prefix
+  if (work_items) {
+    __syncthreads();
+    for (int le = tid; le < e_loc; le += blockDim.x) {
+      int rows = off[rank * e_loc + le];
+      tmp[le] = (rows + bm - 1) / bm;
     }
-    // work queue for the persistent grouped GEMM: one item per
-    // (expert, bm-row-tile) with nonzero rows; encoded le*65536 + tile.
-    // rows come from the LDS copy — re-reading expert_rows[] just written
-    // to global was a serial ~400 ns round-trip per expert (41 us/call).
-    if (work_items) {
-      int n = 0;
-      for (int le = 0; le < e_loc; ++le) {
-        int rows = 0;
-        for (int s2 = 0; s2 < world; ++s2)
-          rows += sp[s2 * e_num + rank * e_loc + le];
-        int tiles = (rows + bm - 1) / bm;
-        for (int t2 = 0; t2 < tiles; ++t2) work_items[n++] = le * 65536 + t2;
-      }
-      work_count[0] = n;
+    __syncthreads();
+    // scan tile counts (reuse pre as scratch)
+    for (int stride = 1; stride < e_loc; stride <<= 1) {
+      for (int le = tid; le < e_loc; le += blockDim.x)
+        pre[le] = le >= stride ? tmp[le] + tmp[le - stride] : tmp[le];
+      __syncthreads();
+      for (int le = tid; le < e_loc; le += blockDim.x) tmp[le] = pre[le];
+      __syncthreads();
     }
+    for (int le = tid; le < e_loc; le += blockDim.x) {
+      int rows = off[rank * e_loc + le];
+      int tiles = (rows + bm - 1) / bm;
+      int base2 = tmp[le] - tiles;
+      for (int t2 = 0; t2 < tiles; ++t2)
+        work_items[base2 + t2] = le * 65536 + t2;
+    }
+    if (tid == 0) work_count[0] = e_loc == 0 ? 0 : tmp[e_loc - 1];
   }
 }
 
@@ -977,9 +1010,11 @@ void launch_moe_dispatch(const PeerTable &pt, const void *x,
                      (const int *)counts, recv_x_off, meta_off,
                      eflags_off, arrive_e, (const int *)val_cell, T, K, H,
                      e_loc, e_num);
-  hipLaunchKernelGGL(k_moe_signal_empty_e, dim3(1), dim3(256), 0, stream,
-                     pt, (const int *)counts, eflags_off,
-                     (const int *)val_cell, e_loc, e_num);
+  if (pt.world > 1) {
+    hipLaunchKernelGGL(k_moe_signal_empty_e, dim3(1), dim3(256), 0, stream,
+                       pt, (const int *)counts, eflags_off,
+                       (const int *)val_cell, e_loc, e_num);
+  }
 }
 
 void launch_moe_dispatch_fp8(const PeerTable &pt, const void *x,
