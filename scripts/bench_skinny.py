@@ -52,10 +52,15 @@ def main():
         skf = sk256_pick(m, n, k)
         if skf:
             ws = torch.empty(m, n, dtype=torch.float32, device="cuda")
-            us_256 = t(lambda: _C.gemm256_sk_bf16(
-                a.data_ptr(), w.data_ptr(), c.data_ptr(), 0, ws.data_ptr(),
-                m, n, k, skf, s))
-            line += f" sk256 {us_256:6.1f} us"
+            sweep = []
+            for sv in sorted({2, 3, 4, 5, skf, 2 * skf}):
+                if k % (128 * sv) or (k // 128 // sv) < 2:
+                    continue
+                us = t(lambda sv=sv: _C.gemm256_sk_bf16(
+                    a.data_ptr(), w.data_ptr(), c.data_ptr(), 0,
+                    ws.data_ptr(), m, n, k, sv, s))
+                sweep.append(f"sk{sv}:{us:.0f}")
+            line += " sk256[" + " ".join(sweep) + "]us"
         print(line)
 
 
