@@ -243,6 +243,11 @@ void launch_moe_dispatch_fp8(const PeerTable &pt, const void *x,
                              size_t meta_off, size_t flags_off,
                              unsigned *arrive, const void *val_cell, int T,
                              int K, int H, int e_loc, hipStream_t stream);
+void launch_moe_grouped_gemm_pq_fp8(
+    const void *rq, const void *rs, const void *weights, void *out,
+    const void *expert_base, const void *expert_rows,
+    const void *work_items, const void *work_count, int n, int k,
+    int fuse_swiglu, hipStream_t stream);
 void launch_moe_dequant(const void *rq, const void *rs, void *out,
                         const void *recv_total, int cap, int H,
                         hipStream_t stream);

@@ -680,6 +680,165 @@ __global__ __launch_bounds__(256) void k_moe_grouped_gemm_pq(
 }
 
 // ---------------------------------------------------------------------------
+// fp8 grouped GEMM: the A operand stays fp8 on the wire AND in LDS; the
+// per-frag dequant (cvt_pk_f32_fp8 + groupwise scale) happens in
+// registers right before the MFMA — the standalone dequant pass and its
+// bf16 recv_x round trip disappear (VERDICT: "fuse dequant into grouped
+// GEMM-1"; reference group_gemm epilogue/prologue fusion capability —
+// behavior only). B (weights) stays bf16.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_moe_grouped_gemm_pq_fp8(
+    const unsigned char *__restrict__ rq, const float *__restrict__ rs,
+    const bf16 *__restrict__ weights, bf16 *__restrict__ out,
+    const int *__restrict__ expert_base, const int *__restrict__ expert_rows,
+    const int *__restrict__ work_items, const int *__restrict__ work_count,
+    int n, int k, int fuse_swiglu) {
+  constexpr int BM = 32, BN = 128, BK = 64;
+  constexpr int ABUF = BM * BK;           // bytes (fp8)
+  constexpr int BBUF = BN * BK;           // elems (bf16)
+  __shared__ unsigned char lds_a8[3 * ABUF];
+  __shared__ bf16 lds_b[3 * BBUF];
+  __shared__ float lds_sc[3 * BM];        // per-row scale of the K-group
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int tiles_n = n / BN;
+  const int total = work_count[0] * tiles_n;
+
+  for (int wi = blockIdx.x; wi < total; wi += gridDim.x) {
+    const int item = work_items[wi / tiles_n];
+    const int tn = wi % tiles_n;
+    const int e = item >> 16;
+    const int tm = item & 0xFFFF;
+    const int rows = expert_rows[e];
+    const int base = expert_base[e];
+    f32x4 acc[2][2] = {};
+    const unsigned char *ga = rq + (size_t)(base + tm * BM) * k;
+    const float *gsc = rs + (size_t)(base + tm * BM) * (k / 128);
+    const bf16 *gb = weights + (size_t)e * n * k + (size_t)tn * BN * k;
+    const int ksteps = k / BK;
+
+    auto stage = [&](int t, int buf) {
+      const int k0 = t * BK;
+      // A: 32 rows x 64 fp8 = 2 KiB -> 128 x 16B chunks (tid < 128)
+      if (tid < 128) {
+        int row = tid >> 2, kc = tid & 3;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int *)(
+                ga + (size_t)row * k + k0 + kc * 16),
+            (__attribute__((address_space(3))) unsigned int *)(
+                lds_a8 + buf * ABUF + ((tid >> 6) * 64) * 16),
+            16, 0, 0);
+      }
+      // per-row scale for this K-group (BK=64 < 128: group = k0/128)
+      if (tid < BM) {
+        lds_sc[buf * BM + tid] =
+            gsc[(size_t)tid * (k / 128) + k0 / 128];
+      }
+#pragma unroll
+      for (int it = 0; it < 4; ++it) {
+        int qb = it * 256 + tid;
+        int rowb = qb >> 3, kcb = qb & 7;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int *)(
+                gb + (size_t)rowb * k + k0 + kcb * 8),
+            (__attribute__((address_space(3))) unsigned int *)(
+                lds_b + buf * BBUF + (it * 256 + wave * 64) * 8),
+            16, 0, 0);
+      }
+    };
+    stage(0, 0);
+    if (ksteps > 1) stage(1, 1);
+    for (int t = 0; t < ksteps; ++t) {
+      const int buf = t % 3;
+      if (t + 1 < ksteps) {
+        // per-thread stage op counts differ here (A: tid<128 only, +1
+        // scale load for tid<32, 4 B loads for all): vmcnt(4) is the
+        // uniform drain that retires ALL of stage t for every class
+        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      } else {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      }
+      __builtin_amdgcn_s_barrier();
+      __builtin_amdgcn_sched_barrier(0);
+      if (t + 2 < ksteps) stage(t + 2, (t + 2) % 3);
+#pragma unroll
+      for (int ks = 0; ks < BK / 32; ++ks) {
+        bf16x8 af[2], bfr[2];
+#pragma unroll
+        for (int i = 0; i < 2; ++i) {
+          int arow = i * 16 + (lane & 15);
+          int brow = wave * 32 + i * 16 + (lane & 15);
+          int kk = ks * 32 + (lane >> 4) * 8;
+          // registers: 8 fp8 -> 8 fp32 * scale -> bf16x8
+          int2 packed = *(const int2 *)(
+              lds_a8 + buf * ABUF + arow * BK + kk);
+          float scale = lds_sc[buf * BM + arow];
+          typedef __attribute__((ext_vector_type(2))) float f32x2;
+          f32x2 f01 = __builtin_amdgcn_cvt_pk_f32_fp8(packed.x, false);
+          f32x2 f23 = __builtin_amdgcn_cvt_pk_f32_fp8(packed.x, true);
+          f32x2 f45 = __builtin_amdgcn_cvt_pk_f32_fp8(packed.y, false);
+          f32x2 f67 = __builtin_amdgcn_cvt_pk_f32_fp8(packed.y, true);
+          bf16x8 v;
+          v[0] = (bf16)(f01[0] * scale); v[1] = (bf16)(f01[1] * scale);
+          v[2] = (bf16)(f23[0] * scale); v[3] = (bf16)(f23[1] * scale);
+          v[4] = (bf16)(f45[0] * scale); v[5] = (bf16)(f45[1] * scale);
+          v[6] = (bf16)(f67[0] * scale); v[7] = (bf16)(f67[1] * scale);
+          af[i] = v;
+          bfr[i] = *(const bf16x8 *)(lds_b + buf * BBUF + brow * BK + kk);
+        }
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+#pragma unroll
+          for (int j = 0; j < 2; ++j)
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[i], bfr[j], acc[i][j], 0, 0, 0);
+      }
+      __builtin_amdgcn_s_barrier();
+      __builtin_amdgcn_sched_barrier(0);
+    }
+    const int row_lim = rows - tm * BM;
+    const int actual_base = base + tm * BM;
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+      for (int j = 0; j < 2; ++j)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int row = i * 16 + (lane >> 4) * 4 + r;
+          int col = wave * 32 + j * 16 + (lane & 15);
+          if (fuse_swiglu) {
+            float g = acc[i][j][r];
+            float partner = __shfl_xor(g, 1);
+            if (((lane & 1) == 0) && row < row_lim) {
+              float silu = g / (1.f + __expf(-g));
+              out[((size_t)actual_base + row) * (n / 2) +
+                  ((size_t)tn * BN + col) / 2] = (bf16)(silu * partner);
+            }
+          } else if (row < row_lim) {
+            out[((size_t)actual_base + row) * n + (size_t)tn * BN + col] =
+                (bf16)acc[i][j][r];
+          }
+        }
+    __syncthreads();
+  }
+}
+
+void launch_moe_grouped_gemm_pq_fp8(
+    const void *rq, const void *rs, const void *weights, void *out,
+    const void *expert_base, const void *expert_rows,
+    const void *work_items, const void *work_count, int n, int k,
+    int fuse_swiglu, hipStream_t stream) {
+  if (n % 128 || k % 128)
+    throw std::runtime_error("grouped gemm pq fp8: N%128/K%128 required");
+  hipLaunchKernelGGL(k_moe_grouped_gemm_pq_fp8, dim3(1024), dim3(256), 0,
+                     stream, (const unsigned char *)rq, (const float *)rs,
+                     (const bf16 *)weights, (bf16 *)out,
+                     (const int *)expert_base, (const int *)expert_rows,
+                     (const int *)work_items, (const int *)work_count, n,
+                     k, fuse_swiglu);
+}
+
+// ---------------------------------------------------------------------------
 // Fused single-kernel EP dispatch + grouped GEMM (closes the reference's
 // mega-kernel row, kernels/amd/ep_all2all_fused.py:316 — behavior only):
 // the first `d_wgs` workgroups run the dispatch producer (grid-stride

@@ -726,6 +726,20 @@ static void moe_dispatch_fp8(uintptr_t x, uintptr_t topk_ids,
   TD_CHECK_HIP(hipGetLastError());
 }
 
+static void moe_grouped_gemm_pq_fp8(uintptr_t rq, uintptr_t rs,
+                                    uintptr_t weights, uintptr_t out,
+                                    uintptr_t expert_base,
+                                    uintptr_t expert_rows,
+                                    uintptr_t work_items,
+                                    uintptr_t work_count, int n, int k,
+                                    int fuse_swiglu, uintptr_t stream) {
+  launch_moe_grouped_gemm_pq_fp8(
+      (void *)rq, (void *)rs, (void *)weights, (void *)out,
+      (void *)expert_base, (void *)expert_rows, (void *)work_items,
+      (void *)work_count, n, k, fuse_swiglu, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
 static void moe_dequant(uintptr_t rq, uintptr_t rs, uintptr_t out,
                         uintptr_t recv_total, int cap, int H,
                         uintptr_t stream) {
@@ -991,6 +1005,7 @@ PYBIND11_MODULE(_C, m) {
         py::arg("T"), py::arg("K"), py::arg("H"), py::arg("e_loc"),
         py::arg("stream"), py::arg("val_cell") = 0);
   m.def("moe_dequant", &moe_dequant);
+  m.def("moe_grouped_gemm_pq_fp8", &moe_grouped_gemm_pq_fp8);
   m.def("bump_cell", &bump_cell);
   m.def("wait_flags_ge_cell", &wait_flags_ge_cell);
   m.def("signal_credit", &signal_credit);

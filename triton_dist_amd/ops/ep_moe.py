@@ -242,10 +242,12 @@ def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
                             L["arrive_d"].data_ptr(), T, K, H, e_loc, s,
                             cell)
         _C.moe_wait_flags(heap.ptr(rank, dflags_off), world, s, cell)
-        _C.moe_dequant(heap.ptr(rank, recv_q_off),
-                       heap.ptr(rank, recv_s_off),
-                       heap.ptr(rank, recv_x_off),
-                       L["recv_total"].data_ptr(), ctx.cap, H, s)
+        if not (small_m and H % 128 == 0):
+            # large-M fallback: standalone dequant + bf16 grouped GEMM
+            _C.moe_dequant(heap.ptr(rank, recv_q_off),
+                           heap.ptr(rank, recv_s_off),
+                           heap.ptr(rank, recv_x_off),
+                           L["recv_total"].data_ptr(), ctx.cap, H, s)
     elif not (small_m and not ctx.low_latency):
         # dispatch rides the comm stream; the expert GEMM is gated
         # per-(expert, tile) on eflags, so FFN tiles of early-complete
@@ -274,7 +276,19 @@ def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
                       device=x.device)
     g1_out = act if fused_swiglu else expert_h
     fs = 1 if fused_swiglu else 0
-    if small_m and not ctx.fp8 and not ctx.low_latency:
+    if small_m and ctx.fp8 and H % 128 == 0:
+        # fp8 A stays on the wire format: per-fragment dequant in
+        # registers inside the grouped GEMM (no recv_x round trip)
+        recv_q_off2 = ctx.recv_q.offset + parity * ctx.cap * H
+        recv_s_off2 = ctx.recv_scale.offset \
+            + parity * ctx.cap * (H // 128) * 4
+        _C.moe_grouped_gemm_pq_fp8(
+            heap.ptr(rank, recv_q_off2), heap.ptr(rank, recv_s_off2),
+            w_gate_up.data_ptr(), g1_out.data_ptr(),
+            L["expert_base"].data_ptr(), L["expert_rows"].data_ptr(),
+            L["work_items"].data_ptr(), L["work_count"].data_ptr(),
+            2 * inter, H, fs, s)
+    elif small_m and not ctx.fp8 and not ctx.low_latency:
         # SINGLE-LAUNCH mega-kernel: dispatch producer workgroups +
         # per-expert-gated grouped GEMM in one kernel (closes the
         # reference's ep_all2all_fused.py:316 row completely)
