@@ -50,10 +50,14 @@ __global__ __launch_bounds__(256) void k_flash_decode(
   const int t = tid & 31;   // softmax role: position
   const long seqlen = *offset + 1;
 
-  __shared__ bf16 k_lds[kTile][kD];
+  // K/V tiles double-buffered: tile t+1's global loads are issued into
+  // registers (4 x bf16x8 per thread) before computing tile t, so the
+  // HBM latency hides under the QK/softmax/PV work (the r01 kernel
+  // serialized load -> barrier -> compute each tile).
+  __shared__ bf16 k_lds[2][kTile][kD];
   // V rows padded +8 elems: tr_read's 4-rows-x-4-colchunks per 16-lane
   // group land on rotated banks (row stride 272 B = 4-bank rotate)
-  __shared__ bf16 v_lds[kTile][kD + 8];
+  __shared__ bf16 v_lds[2][kTile][kD + 8];
   __shared__ bf16 q_lds[16][kD];          // rows >= G zero-padded
   __shared__ bf16 p_bf[16][kTile + 8];    // P tile as MFMA A operand
   __shared__ float s_part[4][16][kTile];  // per-wave QK partials
@@ -81,29 +85,50 @@ __global__ __launch_bounds__(256) void k_flash_decode(
 
   f32x4 accPV[2] = {};
   const long ntiles = (seqlen + kTile - 1) / kTile;
-  for (long tile = 0; tile < ntiles; ++tile) {
-    const long pos0 = tile * kTile;
-    __syncthreads();
-    for (int i = tid; i < kTile * kD / 8; i += 256) {
+  // per-thread slice: 2 chunks of each matrix per tile
+  bf16x8 nk[2], nv[2];
+  auto load_tile = [&](long tile, bf16x8 *rk, bf16x8 *rv) {
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      int i = it * 256 + tid;
       int r = i / (kD / 8);
       int c = (i % (kD / 8)) * 8;
-      long pos = pos0 + r;
+      long pos = tile * kTile + r;
       bf16x8 kv{}, vv{};
       if (pos < seqlen) {
         size_t base = (((size_t)b * max_len + pos) * kvh + kh) * kD + c;
         kv = *(const bf16x8 *)(kcache + base);
         vv = *(const bf16x8 *)(vcache + base);
       }
-      *(bf16x8 *)(&k_lds[r][c]) = kv;
-      *(bf16x8 *)(&v_lds[r][c]) = vv;
+      rk[it] = kv;
+      rv[it] = vv;
     }
-    __syncthreads();
+  };
+  auto store_tile = [&](int buf, const bf16x8 *rk, const bf16x8 *rv) {
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      int i = it * 256 + tid;
+      int r = i / (kD / 8);
+      int c = (i % (kD / 8)) * 8;
+      *(bf16x8 *)(&k_lds[buf][r][c]) = rk[it];
+      *(bf16x8 *)(&v_lds[buf][r][c]) = rv[it];
+    }
+  };
+  load_tile(0, nk, nv);
+  store_tile(0, nk, nv);
+  for (long tile = 0; tile < ntiles; ++tile) {
+    const int cur = tile & 1;
+    const long pos0 = tile * kTile;
+    // issue next tile's global loads before this tile's compute
+    const bool has_next = tile + 1 < ntiles;
+    if (has_next) load_tile(tile + 1, nk, nv);
+    __syncthreads();  // cur buffer's stores (prev iter) are visible
 
     // MFMA QK^T: wave w, pos-half h: B[k][n=pos] = K[pos][k]
 #pragma unroll
     for (int h = 0; h < 2; ++h) {
       bf16x8 kfrag = *(const bf16x8 *)(
-          &k_lds[(lane & 15) + 16 * h][wave * 32 + (lane >> 4) * 8]);
+          &k_lds[cur][(lane & 15) + 16 * h][wave * 32 + (lane >> 4) * 8]);
       f32x4 c4 = {0.f, 0.f, 0.f, 0.f};
       c4 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag, kfrag, c4, 0, 0,
                                                    0);
@@ -153,7 +178,7 @@ __global__ __launch_bounds__(256) void k_flash_decode(
       const int cg = wave * 2 + h;
       const int k0 = (lane >> 4) * 8;
       unsigned addr = (unsigned)(uintptr_t)(
-          &v_lds[k0 + ((lane & 15) >> 2)][cg * 16 + 4 * (lane & 3)]);
+          &v_lds[cur][k0 + ((lane & 15) >> 2)][cg * 16 + 4 * (lane & 3)]);
       unsigned long long lo, hi;
       asm volatile(
           "ds_read_b64_tr_b16 %0, %2\n"
@@ -170,6 +195,9 @@ __global__ __launch_bounds__(256) void k_flash_decode(
       accPV[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag,
                                                          accPV[h], 0, 0, 0);
     }
+    // stage the prefetched tile into the other buffer; the loop-top
+    // barrier orders these stores against next iteration's reads
+    if (has_next) store_tile(cur ^ 1, nk, nv);
   }
   __syncthreads();
 #pragma unroll
