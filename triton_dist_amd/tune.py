@@ -55,11 +55,17 @@ class AutoTuner:
         self._mem: Dict[str, Dict[str, Any]] = {}
         self._path = cache_dir() / f"{name}.json"
         self.always_tune = os.environ.get("TD_AUTOTUNE_ALWAYS_TUNE") == "1"
-        if self._path.exists() and not self.always_tune:
-            try:
-                self._mem = json.loads(self._path.read_text())
-            except Exception:
-                self._mem = {}
+        if not self.always_tune:
+            # repo-shipped cache (committed results measured on MI355X
+            # boxes) seeds the table; the user-writable cache overrides
+            shipped = (Path(__file__).parent / "autotune_cache"
+                       / hardware_hash() / f"{name}.json")
+            for path in (shipped, self._path):
+                if path.exists():
+                    try:
+                        self._mem.update(json.loads(path.read_text()))
+                    except Exception:
+                        pass
 
     def _persist(self):
         if dist.is_initialized() and dist.get_rank() != 0:
