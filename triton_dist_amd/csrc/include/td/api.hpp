@@ -97,6 +97,10 @@ void launch_rs_reduce_bf16(const void *segments, void *out, int world,
 // ldc==n.
 void launch_gemm256_sk_bf16(const GemmArgs &g, float *ws, int sk,
                             hipStream_t stream);
+// two-stage variant: private fp32 slices ws[sk, M, N] + fused reduce
+// (no atomics; ws must hold sk*M*N floats)
+void launch_gemm256_sk2_bf16(const GemmArgs &g, float *ws, int sk,
+                             hipStream_t stream);
 
 // EXPERIMENTAL BK=64 quadrant-phase template (gemm256_v2.hip) — not in
 // any dispatch path; see the file header.

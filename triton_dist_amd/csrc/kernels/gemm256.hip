@@ -547,6 +547,87 @@ __global__ void k_f32_to_bf16(const float *__restrict__ ws,
   }
 }
 
+// Two-stage split-K variant: each split block STORES its fp32 partial
+// tile into a private ws slice [sk, M, N] (coalesced 64B-per-16-lane
+// stores), and one fused reduce kernel sums the slices + bias +
+// converts to bf16. Replaces the ~M*N*sk atomic RMWs of the atomic tier
+// (down-proj: 42M atomics/call) with plain streaming stores + one read
+// pass.
+__global__ __launch_bounds__(NTH, 2) void k_gemm256_sk2_bf16(GemmArgs g,
+                                                             float *ws,
+                                                             int sk) {
+  __shared__ bf16 lds_a[SLICES * SLICE_ELEMS];
+  __shared__ bf16 lds_b[SLICES * SLICE_ELEMS];
+  const int tiles_n = g.n / BN;
+  const int tiles_m = g.m / BM;
+  const int total = tiles_m * tiles_n * sk;
+  int wgid = xcd_remap(blockIdx.x, total);
+  const int tile = wgid / sk;
+  const int sid = wgid % sk;
+  int pid_m, pid_n;
+  tile_coords(tile, tiles_m, tiles_n, pid_m, pid_n);
+  const int ktiles_per = g.k / BK / sk;
+  const int k0 = sid * ktiles_per * BK;
+  WaveCtx w = wave_ctx();
+  f32x4 acc[8][4] = {};
+  const bf16 *ga = (const bf16 *)g.a + (size_t)pid_m * BM * g.lda + k0;
+  const bf16 *gb = (const bf16 *)g.b + (size_t)pid_n * BN * g.ldb + k0;
+  kloop(ga, gb, g.lda, g.ldb, ktiles_per, lds_a, lds_b, w, acc);
+  float *wsb = ws + ((size_t)sid * g.m + (size_t)pid_m * BM) * g.n +
+               (size_t)pid_n * BN;
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = w.wr * 128 + i * 16 + ((w.lane >> 4) * 4 + r);
+        int col = w.wc * 64 + j * 16 + (w.lane & 15);
+        wsb[(size_t)row * g.n + col] = acc[i][j][r];
+      }
+}
+
+__global__ void k_sk2_reduce(const float *__restrict__ ws,
+                             bf16 *__restrict__ c,
+                             const bf16 *__restrict__ bias, int rows,
+                             int n, int sk) {
+  typedef __attribute__((ext_vector_type(4))) float f4;
+  typedef __attribute__((ext_vector_type(4))) bf16 b4;
+  const size_t slice = (size_t)rows * n;
+  for (int r = blockIdx.x; r < rows; r += gridDim.x) {
+    const float *wr = ws + (size_t)r * n;
+    bf16 *cr = c + (size_t)r * n;
+    for (int col = threadIdx.x * 4; col < n; col += blockDim.x * 4) {
+      f4 acc = *(const f4 *)(wr + col);
+      for (int s2 = 1; s2 < sk; ++s2) {
+        f4 v = *(const f4 *)(wr + s2 * slice + col);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) acc[j] += v[j];
+      }
+      b4 o;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        float f = acc[j];
+        if (bias) f += (float)bias[col + j];
+        o[j] = (bf16)f;
+      }
+      *(b4 *)(cr + col) = o;
+    }
+  }
+}
+
+void launch_gemm256_sk2_bf16(const GemmArgs &g, float *ws, int sk,
+                             hipStream_t stream) {
+  if (g.k % (BK * sk))
+    throw std::runtime_error("gemm256_sk2: k % (128*sk) != 0");
+  int grid = (g.m / BM) * (g.n / BN) * sk;
+  hipLaunchKernelGGL(k_gemm256_sk2_bf16, dim3(grid), dim3(NTH), 0, stream,
+                     g, ws, sk);
+  int cgrid = g.m < 2048 ? g.m : 2048;
+  hipLaunchKernelGGL(k_sk2_reduce, dim3(cgrid), dim3(256), 0, stream, ws,
+                     (bf16 *)g.c, (const bf16 *)g.bias, g.m, g.n, sk);
+}
+
 void launch_gemm256_sk_bf16(const GemmArgs &g, float *ws, int sk,
                             hipStream_t stream) {
   if (g.k % (BK * sk))
