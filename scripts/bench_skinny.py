@@ -59,8 +59,13 @@ def main():
                 us = t(lambda sv=sv: _C.gemm256_sk_bf16(
                     a.data_ptr(), w.data_ptr(), c.data_ptr(), 0,
                     ws.data_ptr(), m, n, k, sv, s))
-                sweep.append(f"sk{sv}:{us:.0f}")
-            line += " sk256[" + " ".join(sweep) + "]us"
+                ws2 = torch.empty(sv, m, n, dtype=torch.float32,
+                                  device="cuda")
+                us2 = t(lambda sv=sv, ws2=ws2: _C.gemm256_sk2_bf16(
+                    a.data_ptr(), w.data_ptr(), c.data_ptr(), 0,
+                    ws2.data_ptr(), m, n, k, sv, s))
+                sweep.append(f"sk{sv}:{us:.0f}/{us2:.0f}")
+            line += " sk-atomic/2stage[" + " ".join(sweep) + "]us"
         print(line)
 
 

@@ -408,3 +408,24 @@ def test_gemm_skinny(m, n, k, fuse):
     else:
         ref = full
     assert_allclose(c, ref, **bf16_gemm_tol(k))
+
+
+def test_gemm256_sk2():
+    """Two-stage split-K tier (private fp32 slices + fused reduce) vs
+    fp32 reference and the atomic tier."""
+    from triton_dist_amd import _C
+    from triton_dist_amd.utils.testing import assert_allclose, bf16_gemm_tol
+
+    torch.manual_seed(9)
+    m, n, k = 512, 1280, 5120
+    a = (torch.randn(m, k, device="cuda") / 8).to(torch.bfloat16)
+    w = (torch.randn(n, k, device="cuda") / 8).to(torch.bfloat16)
+    s = torch.cuda.current_stream().cuda_stream
+    for sk in (2, 4, 8):
+        ws = torch.empty(sk, m, n, dtype=torch.float32, device="cuda")
+        c = torch.empty(m, n, device="cuda", dtype=torch.bfloat16)
+        _C.gemm256_sk2_bf16(a.data_ptr(), w.data_ptr(), c.data_ptr(), 0,
+                            ws.data_ptr(), m, n, k, sk, s)
+        torch.cuda.synchronize()
+        assert_allclose(c, a.float() @ w.float().t(), msg=f"sk={sk}",
+                        **bf16_gemm_tol(k))
