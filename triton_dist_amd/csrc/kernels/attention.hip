@@ -50,14 +50,16 @@ __global__ __launch_bounds__(256) void k_flash_decode(
   const int t = tid & 31;   // softmax role: position
   const long seqlen = *offset + 1;
 
-  // K/V tiles double-buffered: tile t+1's global loads are issued into
-  // registers (4 x bf16x8 per thread) before computing tile t, so the
-  // HBM latency hides under the QK/softmax/PV work (the r01 kernel
-  // serialized load -> barrier -> compute each tile).
-  __shared__ bf16 k_lds[2][kTile][kD];
+  // KV tiles prefetched through REGISTERS (4 x bf16x8 per thread): tile
+  // t+1's global loads issue before tile t's compute, so HBM latency
+  // hides under the QK/softmax/PV work; LDS buffers stay SINGLE (the
+  // double-LDS variant cost 5 -> 3 blocks/CU of occupancy). Stores land
+  // between the two loop-top barriers — same barrier count as the r01
+  // serial load -> compute structure.
+  __shared__ bf16 k_lds[kTile][kD];
   // V rows padded +8 elems: tr_read's 4-rows-x-4-colchunks per 16-lane
   // group land on rotated banks (row stride 272 B = 4-bank rotate)
-  __shared__ bf16 v_lds[2][kTile][kD + 8];
+  __shared__ bf16 v_lds[kTile][kD + 8];
   __shared__ bf16 q_lds[16][kD];          // rows >= G zero-padded
   __shared__ bf16 p_bf[16][kTile + 8];    // P tile as MFMA A operand
   __shared__ float s_part[4][16][kTile];  // per-wave QK partials
@@ -104,31 +106,30 @@ __global__ __launch_bounds__(256) void k_flash_decode(
       rv[it] = vv;
     }
   };
-  auto store_tile = [&](int buf, const bf16x8 *rk, const bf16x8 *rv) {
+  auto store_tile = [&](const bf16x8 *rk, const bf16x8 *rv) {
 #pragma unroll
     for (int it = 0; it < 2; ++it) {
       int i = it * 256 + tid;
       int r = i / (kD / 8);
       int c = (i % (kD / 8)) * 8;
-      *(bf16x8 *)(&k_lds[buf][r][c]) = rk[it];
-      *(bf16x8 *)(&v_lds[buf][r][c]) = rv[it];
+      *(bf16x8 *)(&k_lds[r][c]) = rk[it];
+      *(bf16x8 *)(&v_lds[r][c]) = rv[it];
     }
   };
   load_tile(0, nk, nv);
-  store_tile(0, nk, nv);
   for (long tile = 0; tile < ntiles; ++tile) {
-    const int cur = tile & 1;
     const long pos0 = tile * kTile;
-    // issue next tile's global loads before this tile's compute
+    __syncthreads();  // previous tile's LDS reads retired
+    store_tile(nk, nv);
     const bool has_next = tile + 1 < ntiles;
-    if (has_next) load_tile(tile + 1, nk, nv);
-    __syncthreads();  // cur buffer's stores (prev iter) are visible
+    if (has_next) load_tile(tile + 1, nk, nv);  // regs free after store
+    __syncthreads();  // this tile's stores visible
 
     // MFMA QK^T: wave w, pos-half h: B[k][n=pos] = K[pos][k]
 #pragma unroll
     for (int h = 0; h < 2; ++h) {
       bf16x8 kfrag = *(const bf16x8 *)(
-          &k_lds[cur][(lane & 15) + 16 * h][wave * 32 + (lane >> 4) * 8]);
+          &k_lds[(lane & 15) + 16 * h][wave * 32 + (lane >> 4) * 8]);
       f32x4 c4 = {0.f, 0.f, 0.f, 0.f};
       c4 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag, kfrag, c4, 0, 0,
                                                    0);
@@ -178,7 +179,7 @@ __global__ __launch_bounds__(256) void k_flash_decode(
       const int cg = wave * 2 + h;
       const int k0 = (lane >> 4) * 8;
       unsigned addr = (unsigned)(uintptr_t)(
-          &v_lds[cur][k0 + ((lane & 15) >> 2)][cg * 16 + 4 * (lane & 3)]);
+          &v_lds[k0 + ((lane & 15) >> 2)][cg * 16 + 4 * (lane & 3)]);
       unsigned long long lo, hi;
       asm volatile(
           "ds_read_b64_tr_b16 %0, %2\n"
@@ -195,9 +196,6 @@ __global__ __launch_bounds__(256) void k_flash_decode(
       accPV[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag,
                                                          accPV[h], 0, 0, 0);
     }
-    // stage the prefetched tile into the other buffer; the loop-top
-    // barrier orders these stores against next iteration's reads
-    if (has_next) store_tile(cur ^ 1, nk, nv);
   }
   __syncthreads();
 #pragma unroll
