@@ -220,9 +220,13 @@ __global__ void k_qkv_prologue_decode(
     float eps, int use_qk_norm) {
   constexpr int D = 128;
   const int b = blockIdx.x;
-  const int h = blockIdx.y;       // 0..qh+2*kvh-1: q heads, then k, then v
-  const int lane = threadIdx.x;   // 64 lanes, 2 elems each
+  // wave-per-head, 4 heads per 256-thread block (the 64-thread/1-wave
+  // launch measured 10 us/layer at decode; all reductions/shuffles below
+  // are wave-scoped, so packing 4 heads per block is free)
+  const int h = blockIdx.y * 4 + (threadIdx.x >> 6);
+  const int lane = threadIdx.x & 63;  // 64 lanes, 2 elems each
   const int nh = qh + 2 * kvh;
+  if (h >= nh) return;
   const long pos = *offset;
   const bf16 *src = qkv + ((size_t)b * nh + h) * D;
   float v0 = bf2f(src[lane * 2]);
@@ -292,9 +296,10 @@ __global__ void k_qkv_prologue_prefill(
   const int row = blockIdx.x;     // b * s + t
   const int b = row / s;
   const long pos = row % s;
-  const int h = blockIdx.y;
-  const int lane = threadIdx.x;
+  const int h = blockIdx.y * 4 + (threadIdx.x >> 6);  // wave-per-head
+  const int lane = threadIdx.x & 63;
   const int nh = qh + 2 * kvh;
+  if (h >= nh) return;
   const bf16 *src = qkv + ((size_t)row * nh + h) * D;
   float v0 = bf2f(src[lane * 2]);
   float v1 = bf2f(src[lane * 2 + 1]);
@@ -350,8 +355,9 @@ void launch_qkv_prologue_prefill(const void *qkv, void *q_out, void *kcache,
                                  const void *knw, int batch, int s, int qh,
                                  int kvh, int max_len, float eps,
                                  bool use_qk_norm, hipStream_t stream) {
-  hipLaunchKernelGGL(k_qkv_prologue_prefill, dim3(batch * s, qh + 2 * kvh),
-                     dim3(64), 0, stream, (const bf16 *)qkv, (bf16 *)q_out,
+  hipLaunchKernelGGL(k_qkv_prologue_prefill,
+                     dim3(batch * s, (qh + 2 * kvh + 3) / 4),
+                     dim3(256), 0, stream, (const bf16 *)qkv, (bf16 *)q_out,
                      (bf16 *)kcache, (bf16 *)vcache, (const float *)cos_t,
                      (const float *)sin_t, (const bf16 *)qnw,
                      (const bf16 *)knw, s, qh, kvh, max_len, eps,
@@ -365,8 +371,9 @@ void launch_qkv_prologue_decode(const void *qkv, void *q_out, void *kcache,
                                 int batch, int qh, int kvh, int max_len,
                                 float eps, bool use_qk_norm,
                                 hipStream_t stream) {
-  hipLaunchKernelGGL(k_qkv_prologue_decode, dim3(batch, qh + 2 * kvh),
-                     dim3(64), 0, stream, (const bf16 *)qkv, (bf16 *)q_out,
+  hipLaunchKernelGGL(k_qkv_prologue_decode,
+                     dim3(batch, (qh + 2 * kvh + 3) / 4),
+                     dim3(256), 0, stream, (const bf16 *)qkv, (bf16 *)q_out,
                      (bf16 *)kcache, (bf16 *)vcache, (const float *)cos_t,
                      (const float *)sin_t, (const bf16 *)qnw,
                      (const bf16 *)knw, (const long *)offset, qh, kvh,
