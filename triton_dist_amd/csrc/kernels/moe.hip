@@ -555,6 +555,7 @@ __global__ __launch_bounds__(gg::NTH) void k_moe_grouped_gemm(
 // (work_item x n_tile) pairs from the device-built queue — no empty-block
 // churn from capacity grids (measured 17x faster than the capacity grid at
 // DeepSeek decode loads).
+template <int NBUF>
 __global__ __launch_bounds__(256) void k_moe_grouped_gemm_pq(
     const bf16 *__restrict__ xin, const bf16 *__restrict__ weights,
     bf16 *__restrict__ out, const int *__restrict__ expert_base,
@@ -562,10 +563,13 @@ __global__ __launch_bounds__(256) void k_moe_grouped_gemm_pq(
     const int *__restrict__ work_count, int n, int k,
     const int *__restrict__ eflags, const int *val_cell, int world,
     int e_loc, int fuse_swiglu) {
+  // NBUF=3: 61 KiB LDS -> 2 blocks/CU, 2-step prefetch (r01 default).
+  // NBUF=2: 41 KiB -> 3 blocks/CU, 1-step prefetch — trades pipeline
+  // depth for +50% resident blocks (TD_MOE_PQ2=1 A/B).
   constexpr int BM = 32, BN = 128, BK = 64;
   constexpr int ABUF = BM * BK, BBUF = BN * BK;
-  __shared__ bf16 lds_a[3 * ABUF];
-  __shared__ bf16 lds_b[3 * BBUF];
+  __shared__ bf16 lds_a[NBUF * ABUF];
+  __shared__ bf16 lds_b[NBUF * BBUF];
   const int tid = threadIdx.x;
   const int wave = tid >> 6, lane = tid & 63;
   const int tiles_n = n / BN;
@@ -614,17 +618,18 @@ __global__ __launch_bounds__(256) void k_moe_grouped_gemm_pq(
       }
     };
     stage(0, 0);
-    if (ksteps > 1) stage(1, 1);
+    if (NBUF >= 3 && ksteps > 1) stage(1, 1);
     for (int t = 0; t < ksteps; ++t) {
-      const int buf = t % 3;
-      if (t + 1 < ksteps) {
+      const int buf = t % NBUF;
+      const int ahead = NBUF - 1;  // staged steps in flight
+      if (t + ahead - 1 < ksteps) {
         asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
       } else {
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       }
       __builtin_amdgcn_s_barrier();
       __builtin_amdgcn_sched_barrier(0);
-      if (t + 2 < ksteps) stage(t + 2, (t + 2) % 3);
+      if (t + ahead < ksteps) stage(t + ahead, (t + ahead) % NBUF);
 #pragma unroll
       for (int ks = 0; ks < BK / 32; ++ks) {
         bf16x8 af[2], bfr[2];
@@ -912,9 +917,10 @@ __global__ __launch_bounds__(256) void k_moe_fused_dispatch_gemm(
   }
   // ---- consumer role: per-expert-gated pq grouped GEMM ----
   constexpr int BM = 32, BN = 128, BK = 64;
+  constexpr int NBUF = 3;
   constexpr int ABUF = BM * BK, BBUF = BN * BK;
-  __shared__ bf16 lds_a[3 * ABUF];
-  __shared__ bf16 lds_b[3 * BBUF];
+  __shared__ bf16 lds_a[NBUF * ABUF];
+  __shared__ bf16 lds_b[NBUF * BBUF];
   const bf16 *xin = (const bf16 *)((char *)pt.bases[pt.rank] + recv_x_off);
   const int *eflags = (const int *)((char *)pt.bases[pt.rank] + eflags_off);
   const int tid = threadIdx.x;
@@ -962,17 +968,18 @@ __global__ __launch_bounds__(256) void k_moe_fused_dispatch_gemm(
       }
     };
     stage(0, 0);
-    if (ksteps > 1) stage(1, 1);
+    if (NBUF >= 3 && ksteps > 1) stage(1, 1);
     for (int t = 0; t < ksteps; ++t) {
-      const int buf = t % 3;
-      if (t + 1 < ksteps) {
+      const int buf = t % NBUF;
+      const int ahead = NBUF - 1;  // staged steps in flight
+      if (t + ahead - 1 < ksteps) {
         asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
       } else {
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       }
       __builtin_amdgcn_s_barrier();
       __builtin_amdgcn_sched_barrier(0);
-      if (t + 2 < ksteps) stage(t + 2, (t + 2) % 3);
+      if (t + ahead < ksteps) stage(t + ahead, (t + ahead) % NBUF);
 #pragma unroll
       for (int ks = 0; ks < BK / 32; ++ks) {
         bf16x8 af[2], bfr[2];
@@ -1241,7 +1248,20 @@ void launch_moe_grouped_gemm_pq(const void *xin, const void *weights,
                                 int e_loc, int fuse_swiglu) {
   if (n % 128 || k % 64)
     throw std::runtime_error("grouped gemm pq: N%128/K%64 required");
-  hipLaunchKernelGGL(k_moe_grouped_gemm_pq, dim3(1024), dim3(256), 0,
+  static const bool two_buf = [] {
+    const char *e = getenv("TD_MOE_PQ2");
+    return e && e[0] == '1';
+  }();
+  if (two_buf) {
+    hipLaunchKernelGGL((k_moe_grouped_gemm_pq<2>), dim3(1024), dim3(256),
+                       0, stream, (const bf16 *)xin, (const bf16 *)weights,
+                       (bf16 *)out, (const int *)expert_base,
+                       (const int *)expert_rows, (const int *)work_items,
+                       (const int *)work_count, n, k, (const int *)eflags,
+                       (const int *)val_cell, world, e_loc, fuse_swiglu);
+    return;
+  }
+  hipLaunchKernelGGL((k_moe_grouped_gemm_pq<3>), dim3(1024), dim3(256), 0,
                      stream, (const bf16 *)xin, (const bf16 *)weights,
                      (bf16 *)out, (const int *)expert_base,
                      (const int *)expert_rows, (const int *)work_items,
