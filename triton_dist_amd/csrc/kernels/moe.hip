@@ -622,7 +622,10 @@ __global__ __launch_bounds__(256) void k_moe_grouped_gemm_pq(
     for (int t = 0; t < ksteps; ++t) {
       const int buf = t % NBUF;
       const int ahead = NBUF - 1;  // staged steps in flight
-      if (t + ahead - 1 < ksteps) {
+      // NBUF=3: stage t fully landed once <=5 ops (stage t+1) remain.
+      // NBUF=2: stage t is the NEWEST in flight at this wait -> full
+      // drain (the overlap is stage t+1 issuing under compute t).
+      if (NBUF >= 3 && t + 1 < ksteps) {
         asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
       } else {
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -915,9 +918,10 @@ __global__ __launch_bounds__(256) void k_moe_fused_dispatch_gemm(
     }
     return;
   }
-  // ---- consumer role: per-expert-gated pq grouped GEMM ----
+  // ---- consumer role: per-expert-gated pq grouped GEMM (2-buffer:
+  // same occupancy trade as the standalone pq kernel) ----
   constexpr int BM = 32, BN = 128, BK = 64;
-  constexpr int NBUF = 3;
+  constexpr int NBUF = 2;
   constexpr int ABUF = BM * BK, BBUF = BN * BK;
   __shared__ bf16 lds_a[NBUF * ABUF];
   __shared__ bf16 lds_b[NBUF * BBUF];
@@ -972,7 +976,10 @@ __global__ __launch_bounds__(256) void k_moe_fused_dispatch_gemm(
     for (int t = 0; t < ksteps; ++t) {
       const int buf = t % NBUF;
       const int ahead = NBUF - 1;  // staged steps in flight
-      if (t + ahead - 1 < ksteps) {
+      // NBUF=3: stage t fully landed once <=5 ops (stage t+1) remain.
+      // NBUF=2: stage t is the NEWEST in flight at this wait -> full
+      // drain (the overlap is stage t+1 issuing under compute t).
+      if (NBUF >= 3 && t + 1 < ksteps) {
         asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
       } else {
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -1248,11 +1255,13 @@ void launch_moe_grouped_gemm_pq(const void *xin, const void *weights,
                                 int e_loc, int fuse_swiglu) {
   if (n % 128 || k % 64)
     throw std::runtime_error("grouped gemm pq: N%128/K%64 required");
-  static const bool two_buf = [] {
-    const char *e = getenv("TD_MOE_PQ2");
+  // 2-buffer default: 41 KiB LDS -> 3 blocks/CU measured 27.3 -> 25.8
+  // ms/step on qwen3-30b-a3b (TD_MOE_PQ3=1 reverts to the r01 3-buffer)
+  static const bool three_buf = [] {
+    const char *e = getenv("TD_MOE_PQ3");
     return e && e[0] == '1';
   }();
-  if (two_buf) {
+  if (!three_buf) {
     hipLaunchKernelGGL((k_moe_grouped_gemm_pq<2>), dim3(1024), dim3(256),
                        0, stream, (const bf16 *)xin, (const bf16 *)weights,
                        (bf16 *)out, (const int *)expert_base,
