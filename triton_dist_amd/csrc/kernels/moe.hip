@@ -702,11 +702,13 @@ __global__ __launch_bounds__(256) void k_moe_grouped_gemm_pq_fp8(
     const int *__restrict__ work_items, const int *__restrict__ work_count,
     int n, int k, int fuse_swiglu) {
   constexpr int BM = 32, BN = 128, BK = 64;
+  constexpr int NBUF = 2;  // 37 KiB LDS -> 4 blocks/CU (same occupancy
+                           // trade measured on the bf16 pq kernel)
   constexpr int ABUF = BM * BK;           // bytes (fp8)
   constexpr int BBUF = BN * BK;           // elems (bf16)
-  __shared__ unsigned char lds_a8[3 * ABUF];
-  __shared__ bf16 lds_b[3 * BBUF];
-  __shared__ float lds_sc[3 * BM];        // per-row scale of the K-group
+  __shared__ unsigned char lds_a8[NBUF * ABUF];
+  __shared__ bf16 lds_b[NBUF * BBUF];
+  __shared__ float lds_sc[NBUF * BM];     // per-row scale of the K-group
   const int tid = threadIdx.x;
   const int wave = tid >> 6, lane = tid & 63;
   const int tiles_n = n / BN;
@@ -755,20 +757,17 @@ __global__ __launch_bounds__(256) void k_moe_grouped_gemm_pq_fp8(
       }
     };
     stage(0, 0);
-    if (ksteps > 1) stage(1, 1);
+    if (NBUF >= 3 && ksteps > 1) stage(1, 1);
     for (int t = 0; t < ksteps; ++t) {
-      const int buf = t % 3;
-      if (t + 1 < ksteps) {
-        // per-thread stage op counts differ here (A: tid<128 only, +1
-        // scale load for tid<32, 4 B loads for all): vmcnt(4) is the
-        // uniform drain that retires ALL of stage t for every class
-        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-      } else {
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      }
+      const int buf = t % NBUF;
+      const int ahead = NBUF - 1;
+      // NBUF=2: stage t is the newest in flight -> full drain (overlap
+      // = stage t+1 issuing under compute t). NBUF=3 would need the
+      // class-uniform vmcnt(4) (see git history).
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       __builtin_amdgcn_s_barrier();
       __builtin_amdgcn_sched_barrier(0);
-      if (t + 2 < ksteps) stage(t + 2, (t + 2) % 3);
+      if (t + ahead < ksteps) stage(t + ahead, (t + ahead) % NBUF);
 #pragma unroll
       for (int ks = 0; ks < BK / 32; ++ks) {
         bf16x8 af[2], bfr[2];
