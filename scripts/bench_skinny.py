@@ -44,10 +44,16 @@ def main():
         us_sk = t(lambda: _C.gemm_skinny_bf16(
             a.data_ptr(), w.data_ptr(), c.data_ptr(), 0, m, n, k, 0, s))
         us_blt = t(lambda: torch.matmul(a, w.t(), out=c))
+        us_ring = None
+        if m % 256 == 0 and n % 256 == 0 and k % 128 == 0:
+            us_ring = t(lambda: _C.gemm_bf16(
+                a.data_ptr(), w.data_ptr(), c.data_ptr(), 0, m, n, k, s))
         floor = n * k * 2 / 8e12 * 1e6  # weight bytes / 8 TB/s
         line = (f"{name:8s} {m}x{n}x{k}: skinny {us_sk:6.1f} us"
                 f" blt {us_blt:6.1f} us floor {floor:5.1f} us"
                 f" rel {rel:.1e}")
+        if us_ring is not None:
+            line += f" ring {us_ring:6.1f} us"
         from triton_dist_amd.ops.gemm import sk256_pick
         skf = sk256_pick(m, n, k)
         if skf:
