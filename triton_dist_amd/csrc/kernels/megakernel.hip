@@ -129,8 +129,9 @@ TD_DEV void t_gemm_tile(const Task &t, bf16 *lds) {
   // the grouped-GEMM tier). Masked rows for M < 32-multiples.
   constexpr int BM = 32, BN = 128, BK = 64;
   constexpr int ABUF = BM * BK, BBUF = BN * BK;
-  bf16 *lds_a = lds;                 // 3 x 4 KB
-  bf16 *lds_b = lds + 3 * ABUF;      // 3 x 16 KB
+  bf16 *lds_a = lds;                 // 2 x 4 KB
+  bf16 *lds_b = lds + 2 * ABUF;      // 2 x 16 KB (2-buffer: the union
+                                     // drops to 40 KB -> 3 blocks/CU)
   const bf16 *A = (const bf16 *)t.a[0];
   const bf16 *B = (const bf16 *)t.a[1];
   bf16 *C = (bf16 *)t.a[2];
@@ -166,17 +167,15 @@ TD_DEV void t_gemm_tile(const Task &t, bf16 *lds) {
     }
   };
   stage(0, 0);
-  if (ksteps > 1) stage(1, 1);
   for (int ti = 0; ti < ksteps; ++ti) {
-    const int buf = ti % 3;
-    if (ti + 1 < ksteps) {
-      asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
-    } else {
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    }
+    const int buf = ti & 1;
+    // 2-buffer, 1-step lookahead: stage ti is the newest in flight at
+    // this wait -> full drain; the overlap is stage ti+1 issuing under
+    // compute ti
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
     __builtin_amdgcn_sched_barrier(0);
-    if (ti + 2 < ksteps) stage(ti + 2, (ti + 2) % 3);
+    if (ti + 1 < ksteps) stage(ti + 1, (ti + 1) & 1);
 #pragma unroll
     for (int ks = 0; ks < BK / 32; ++ks) {
       bf16x8 af[2], bfr[2];
@@ -221,7 +220,7 @@ TD_DEV void t_gemm_tile_part(const Task &t, bf16 *lds) {
   constexpr int BM = 32, BN = 128, BK = 64;
   constexpr int ABUF = BM * BK, BBUF = BN * BK;
   bf16 *lds_a = lds;
-  bf16 *lds_b = lds + 3 * ABUF;
+  bf16 *lds_b = lds + 2 * ABUF;
   const bf16 *A = (const bf16 *)t.a[0];
   const bf16 *B = (const bf16 *)t.a[1];
   float *ws = (float *)t.a[2];
@@ -259,17 +258,15 @@ TD_DEV void t_gemm_tile_part(const Task &t, bf16 *lds) {
     }
   };
   stage(0, 0);
-  if (ksteps > 1) stage(1, 1);
   for (int ti = 0; ti < ksteps; ++ti) {
-    const int buf = ti % 3;
-    if (ti + 1 < ksteps) {
-      asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
-    } else {
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    }
+    const int buf = ti & 1;
+    // 2-buffer, 1-step lookahead: stage ti is the newest in flight at
+    // this wait -> full drain; the overlap is stage ti+1 issuing under
+    // compute ti
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
     __builtin_amdgcn_sched_barrier(0);
-    if (ti + 2 < ksteps) stage(ti + 2, (ti + 2) % 3);
+    if (ti + 1 < ksteps) stage(ti + 1, (ti + 1) & 1);
 #pragma unroll
     for (int ks = 0; ks < BK / 32; ++ks) {
       bf16x8 af[2], bfr[2];
@@ -542,7 +539,7 @@ TD_DEV void t_gemv(const Task &t, char *lds) {
   const int k = (int)t.a[5];
   const int col0 = (int)t.a[6];
   const int ncols = (int)t.a[7];
-  bf16 *x_lds = (bf16 *)lds;  // [m][k] (m*k*2 <= 61440 guaranteed by host)
+  bf16 *x_lds = (bf16 *)lds;  // [m][k] (m*k*2 <= 40960 guaranteed by host)
   for (int i = threadIdx.x * 8; i < m * k; i += blockDim.x * 8)
     *(bf16x8 *)(x_lds + i) = *(const bf16x8 *)(A + i);
   __syncthreads();
@@ -599,7 +596,9 @@ __global__ __launch_bounds__(mk::NTH) void k_megakernel(
     const mk::Task *__restrict__ tasks, const int *__restrict__ queue,
     const int *__restrict__ queue_off, int *__restrict__ scoreboard,
     int fence_mode) {
-  __shared__ char lds[61440];  // union: gemm 3-buf A/B | flash-decode state
+  __shared__ char lds[40960];  // union: gemm 2-buf A/B | flash-decode
+                               // state (40 KB -> 3 blocks/CU; the 60 KB
+                               // 3-buf union capped residency at 2)
   const int wg = blockIdx.x;
   const int q_lo = queue_off[wg], q_hi = queue_off[wg + 1];
   for (int qi = q_lo; qi < q_hi; ++qi) {

@@ -102,8 +102,8 @@ def emit_gemv(g: "MegaGraph", a_ptr: int, w_ptr: int, c_ptr: int,
     """Decode-GEMV emission (batch <= 4): one task per 512-column chunk
     streaming the full K — ~25x fewer tasks than 32x128 tiling at bsz 1,
     which is what the ~0.25 us/task dispatch overhead demands. x rows
-    must fit the 60 KiB task LDS (m*k*2 <= 61440)."""
-    assert batch * k * 2 <= 61440
+    must fit the 40 KiB task LDS (m*k*2 <= 40960)."""
+    assert batch * k * 2 <= 40960
     chunk = 512
     while n % chunk:
         chunk //= 2

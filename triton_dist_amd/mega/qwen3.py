@@ -69,7 +69,7 @@ class MegaQwen3Decode:
             # ~12 workgroups, collapsing per-op parallelism (measured
             # 137 ms vs 20.6 ms tiles at bsz 1 — per-op latency is
             # parallelism-bound, not task-count-bound)
-            if (batch <= 4 and batch * k * 2 <= 61440
+            if (batch <= 4 and batch * k * 2 <= 40960
                     and os.environ.get("TD_MK_GEMV")):
                 from .builder import emit_gemv
                 return emit_gemv(g, a_buf.data_ptr(), w.data_ptr(),
