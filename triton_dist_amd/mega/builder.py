@@ -161,7 +161,7 @@ def emit_gemm(g: "MegaGraph", a_ptr: int, w_ptr: int, c_ptr: int,
 class MegaRun:
     """Owns the finalized descriptors; one launch per decode step."""
 
-    def __init__(self, graph: MegaGraph, n_wg: int = 512, device="cuda"):
+    def __init__(self, graph: MegaGraph, n_wg: int = 768, device="cuda"):
         self.n_wg = n_wg
         (self.task_buf, self.queue, self.queue_off,
          self.scoreboard) = graph.finalize(n_wg, device)

@@ -21,7 +21,7 @@ class MegaQwen3Decode:
     ONE kernel launch (+ a torch argmax for sampling)."""
 
     def __init__(self, model: DenseLLM, kv: KVCache, batch: int,
-                 n_wg: int = 512):
+                 n_wg: int = 768):
         assert model.world == 1, "megakernel v1 is single-GPU"
         cfg = model.cfg
         assert cfg.head_dim == 128
