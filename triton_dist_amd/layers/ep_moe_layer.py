@@ -96,6 +96,76 @@ class EPMoELayer:
 
     __call__ = forward
 
+    def prefill_fwd(self, x: torch.Tensor) -> torch.Tensor:
+        """Fast GPU prefill for REPLICATED x [M, H]: tokens routed to MY
+        experts are expert-sorted and run through the bf16 persistent
+        grouped GEMMs (fused SwiGLU when weights are interleaved), then
+        weighted fp32 scatter-add + all-reduce. Replaces the fp32
+        per-expert torch loop on the prefill path (~4x wall clock on the
+        profiled bench prefill); torch_fwd stays the fp32 golden."""
+        import torch.nn.functional as F
+
+        if not x.is_cuda:
+            return self.torch_fwd(x)
+        if not self._swiglu_fused:
+            self._fuse_weights()
+        from .. import _C
+
+        topk_ids, topk_w = self.route(x)
+        m = x.shape[0]
+        K = self.topk
+        lo = self.rank * self.e_loc
+        flat = topk_ids.reshape(-1).to(torch.int64)
+        sel = (flat >= lo) & (flat < lo + self.e_loc)
+        rows = sel.nonzero(as_tuple=True)[0]
+        ids_sel = flat[rows] - lo
+        order = torch.argsort(ids_sel, stable=True)
+        rows_sorted = rows[order]
+        tok = rows_sorted // K
+        nsel = rows_sorted.numel()
+        x_sorted = torch.zeros(nsel + 128, self.hidden, dtype=x.dtype,
+                               device=x.device)
+        x_sorted[:nsel] = x.index_select(0, tok)
+        counts = torch.bincount(ids_sel[order], minlength=self.e_loc)
+        base = (torch.cumsum(counts, 0) - counts).to(torch.int32)
+        cnt = counts.cpu().tolist()
+        bm = 32
+        items = []
+        for e, c in enumerate(cnt):
+            for t in range((c + bm - 1) // bm):
+                items.append(e * 65536 + t)
+        dev = x.device
+        eb = base.contiguous().to(dev)
+        er = counts.to(torch.int32).contiguous().to(dev)
+        wi = torch.tensor(items or [0], dtype=torch.int32, device=dev)
+        wc = torch.tensor([len(items)], dtype=torch.int32, device=dev)
+        st = torch.cuda.current_stream().cuda_stream
+        inter = self.inter
+        fs = 1 if self._swiglu_fused else 0
+        act = torch.empty(nsel + 128, inter, dtype=x.dtype, device=dev)
+        g1_out = act if fs else torch.empty(nsel + 128, 2 * inter,
+                                            dtype=x.dtype, device=dev)
+        _C.moe_grouped_gemm_pq(x_sorted.data_ptr(),
+                               self.w_gate_up.data_ptr(),
+                               g1_out.data_ptr(), eb.data_ptr(),
+                               er.data_ptr(), wi.data_ptr(), wc.data_ptr(),
+                               2 * inter, self.hidden, st, 0, 0, 0, 0, fs)
+        if not fs:
+            _C.swiglu(g1_out.data_ptr(), act.data_ptr(), nsel + 128,
+                      inter, st)
+        part = torch.empty(nsel + 128, self.hidden, dtype=x.dtype,
+                           device=dev)
+        _C.moe_grouped_gemm_pq(act.data_ptr(), self.w_down.data_ptr(),
+                               part.data_ptr(), eb.data_ptr(),
+                               er.data_ptr(), wi.data_ptr(), wc.data_ptr(),
+                               self.hidden, inter, st, 0, 0, 0, 0, 0)
+        w_sel = topk_w.reshape(-1)[rows_sorted].float()
+        y = torch.zeros(m, self.hidden, dtype=torch.float32, device=dev)
+        y.index_add_(0, tok, part[:nsel].float() * w_sel[:, None])
+        if dist.is_initialized() and self.world > 1:
+            dist.all_reduce(y)
+        return y.to(self.dtype)
+
     def torch_fwd(self, x: torch.Tensor) -> torch.Tensor:
         """Golden reference for REPLICATED x [M, H]: each rank computes its
         local experts' weighted contribution for every token, then

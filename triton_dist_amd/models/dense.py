@@ -150,7 +150,12 @@ class DenseLLM:
                 if mode == self.mode else \
                 attn.torch_fwd(h, kv, li, pos, b, s, prefill)
             x, h = add_rms_norm_op(a, x, layer["ln2"], eps)
-            pending = mlp.forward(h) if mode == self.mode else mlp.torch_fwd(h)
+            if mode == self.mode:
+                pending = mlp.forward(h)
+            elif (prefill and h.is_cuda and hasattr(mlp, "prefill_fwd")):
+                pending = mlp.prefill_fwd(h)
+            else:
+                pending = mlp.torch_fwd(h)
         _, x = add_rms_norm_op(pending, x, self.final_norm_w, eps)
         if sharded:
             x = allgather(x, self._decode_ag_ctx)
