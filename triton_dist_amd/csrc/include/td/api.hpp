@@ -282,7 +282,8 @@ void launch_moe_combine_reduce(const void *combine_buf, const void *topk_w,
 // kernels/megakernel.hip -----------------------------------------------------
 void launch_megakernel(const void *tasks, const void *queue,
                        const void *queue_off, void *scoreboard, int n_wg,
-                       hipStream_t stream, int fence_mode = 0);
+                       hipStream_t stream, int fence_mode = 0,
+                       void *prof = nullptr);
 
 // kernels/elementwise.hip ----------------------------------------------------
 void launch_rmsnorm(const void *x, const void *w, void *out, int rows,

@@ -595,14 +595,19 @@ TD_DEV void spin_ge_relaxed(const int *flag, int bound) {
 __global__ __launch_bounds__(mk::NTH) void k_megakernel(
     const mk::Task *__restrict__ tasks, const int *__restrict__ queue,
     const int *__restrict__ queue_off, int *__restrict__ scoreboard,
-    int fence_mode) {
+    int fence_mode, unsigned long long *__restrict__ prof) {
+  // prof (optional): [16 types x 2] wallclock accumulators —
+  // [type][0] += dependency-wait ticks, [type][1] += body ticks
   __shared__ char lds[40960];  // union: gemm 2-buf A/B | flash-decode
                                // state (40 KB -> 3 blocks/CU; the 60 KB
                                // 3-buf union capped residency at 2)
   const int wg = blockIdx.x;
   const int q_lo = queue_off[wg], q_hi = queue_off[wg + 1];
+  unsigned long long acc_wait[16] = {};
+  unsigned long long acc_body[16] = {};
   for (int qi = q_lo; qi < q_hi; ++qi) {
     const mk::Task t = tasks[queue[qi]];
+    const unsigned long long t0 = prof ? wallclock() : 0;
     // dependency waits: relaxed spins + ONE acquire fence (a per-
     // iteration agent-scope acquire invalidates the XCD L2 every spin —
     // measured poison at 10k+ tasks/step)
@@ -613,6 +618,7 @@ __global__ __launch_bounds__(mk::NTH) void k_megakernel(
         __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
     }
     __syncthreads();
+    const unsigned long long t1 = prof ? wallclock() : 0;
     switch (t.type) {
       case mk::T_RMSNORM:
         mk::t_rmsnorm(t, false);
@@ -662,18 +668,34 @@ __global__ __launch_bounds__(mk::NTH) void k_megakernel(
       if (fence_mode == 0)
         __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
       atomic_add<Scope::Gpu>(scoreboard + t.score_slot, 1);
+      if (prof) {
+        int ty = t.type & 15;
+        acc_wait[ty] += t1 - t0;
+        acc_body[ty] += wallclock() - t1;
+      }
     }
     __syncthreads();
+  }
+  if (prof && threadIdx.x == 0) {
+#pragma unroll
+    for (int ty = 0; ty < 16; ++ty) {
+      if (acc_wait[ty])
+        atomic_add<Scope::Gpu>((unsigned long long *)(prof + ty * 2),
+                               acc_wait[ty]);
+      if (acc_body[ty])
+        atomic_add<Scope::Gpu>((unsigned long long *)(prof + ty * 2 + 1),
+                               acc_body[ty]);
+    }
   }
 }
 
 void launch_megakernel(const void *tasks, const void *queue,
                        const void *queue_off, void *scoreboard, int n_wg,
-                       hipStream_t stream, int fence_mode) {
+                       hipStream_t stream, int fence_mode, void *prof) {
   hipLaunchKernelGGL(k_megakernel, dim3(n_wg), dim3(mk::NTH), 0, stream,
                      (const mk::Task *)tasks, (const int *)queue,
                      (const int *)queue_off, (int *)scoreboard,
-                     fence_mode);
+                     fence_mode, (unsigned long long *)prof);
 }
 
 }  // namespace td

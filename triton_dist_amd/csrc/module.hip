@@ -819,10 +819,10 @@ static void moe_combine_reduce(uintptr_t combine_buf, uintptr_t topk_w,
 
 static void megakernel(uintptr_t tasks, uintptr_t queue, uintptr_t queue_off,
                        uintptr_t scoreboard, int n_wg, uintptr_t stream,
-                       int fence_mode = 0) {
+                       int fence_mode = 0, uintptr_t prof = 0) {
   launch_megakernel((const void *)tasks, (const void *)queue,
                     (const void *)queue_off, (void *)scoreboard, n_wg,
-                    as_stream(stream), fence_mode);
+                    as_stream(stream), fence_mode, (void *)prof);
   TD_CHECK_HIP(hipGetLastError());
 }
 
@@ -1040,7 +1040,8 @@ PYBIND11_MODULE(_C, m) {
         py::arg("val_cell") = 0);
   m.def("megakernel", &megakernel, py::arg("tasks"), py::arg("queue"),
         py::arg("queue_off"), py::arg("scoreboard"), py::arg("n_wg"),
-        py::arg("stream"), py::arg("fence_mode") = 0);
+        py::arg("stream"), py::arg("fence_mode") = 0,
+        py::arg("prof") = 0);
   m.def("rmsnorm", &rmsnorm);
   m.def("add_rmsnorm", &add_rmsnorm);
   m.def("swiglu", &swiglu);

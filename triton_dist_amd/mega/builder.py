@@ -174,7 +174,13 @@ class MegaRun:
 
         s = stream or torch.cuda.current_stream()
         self.scoreboard.zero_()
+        prof_ptr = 0
+        if os.environ.get("TD_MK_PROF"):
+            if not hasattr(self, "prof"):
+                self.prof = torch.zeros(16, 2, dtype=torch.int64,
+                                        device="cuda")
+            prof_ptr = self.prof.data_ptr()
         _C.megakernel(self.task_buf.data_ptr(), self.queue.data_ptr(),
                       self.queue_off.data_ptr(), self.scoreboard.data_ptr(),
                       self.n_wg, s.cuda_stream,
-                      int(os.environ.get("TD_MK_FENCE", "0")))
+                      int(os.environ.get("TD_MK_FENCE", "0")), prof_ptr)
