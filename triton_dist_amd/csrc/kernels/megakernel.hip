@@ -603,8 +603,6 @@ __global__ __launch_bounds__(mk::NTH) void k_megakernel(
                                // 3-buf union capped residency at 2)
   const int wg = blockIdx.x;
   const int q_lo = queue_off[wg], q_hi = queue_off[wg + 1];
-  unsigned long long acc_wait[16] = {};
-  unsigned long long acc_body[16] = {};
   for (int qi = q_lo; qi < q_hi; ++qi) {
     const mk::Task t = tasks[queue[qi]];
     const unsigned long long t0 = prof ? wallclock() : 0;
@@ -669,23 +667,12 @@ __global__ __launch_bounds__(mk::NTH) void k_megakernel(
         __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
       atomic_add<Scope::Gpu>(scoreboard + t.score_slot, 1);
       if (prof) {
-        int ty = t.type & 15;
-        acc_wait[ty] += t1 - t0;
-        acc_body[ty] += wallclock() - t1;
+        const int ty = t.type & 15;
+        atomic_add<Scope::Gpu>(prof + ty * 2, t1 - t0);
+        atomic_add<Scope::Gpu>(prof + ty * 2 + 1, wallclock() - t1);
       }
     }
     __syncthreads();
-  }
-  if (prof && threadIdx.x == 0) {
-#pragma unroll
-    for (int ty = 0; ty < 16; ++ty) {
-      if (acc_wait[ty])
-        atomic_add<Scope::Gpu>((unsigned long long *)(prof + ty * 2),
-                               acc_wait[ty]);
-      if (acc_body[ty])
-        atomic_add<Scope::Gpu>((unsigned long long *)(prof + ty * 2 + 1),
-                               acc_body[ty]);
-    }
   }
 }
 
