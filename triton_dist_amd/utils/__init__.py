@@ -10,4 +10,5 @@ from .distributed import (  # noqa: F401
     env_local_rank,
 )
 from .testing import assert_allclose, rand_tensor, bf16_gemm_tol  # noqa: F401
-from .bench import perf_func, perf_func_with_l2_reset  # noqa: F401
+from .bench import (perf_func, perf_func_with_l2_reset,  # noqa: F401
+                    wait_stable_clock)

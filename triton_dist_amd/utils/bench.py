@@ -120,3 +120,45 @@ class group_profile:
                       open(os.path.join(self.out_dir,
                                         f"{self.name}_merged.json"), "w"))
         return False
+
+
+def wait_stable_clock(frac: float = 0.9, timeout_s: float = 5.0) -> bool:
+    """Spin a dummy workload until the GPU sclk reaches `frac` of its max
+    (capability parity with the reference's GPU-clock wait for stable
+    benchmarking, utils.py:953 — behavior only). Returns True when the
+    clock stabilized, False on timeout / no clock telemetry."""
+    if not torch.cuda.is_available():
+        return False
+    import re
+    import subprocess
+
+    def read_clock():
+        try:
+            out = subprocess.run(["rocm-smi", "--showgpuclocks"],
+                                 capture_output=True, text=True,
+                                 timeout=3).stdout
+            m = re.findall(r"sclk.*?\((\d+)Mhz\)", out)
+            return int(m[0]) if m else None
+        except Exception:
+            return None
+
+    mx = None
+    try:
+        out = subprocess.run(["rocm-smi", "-s"], capture_output=True,
+                             text=True, timeout=3).stdout
+        lv = re.findall(r"(\d+)Mhz", out)
+        mx = max(int(v) for v in lv) if lv else None
+    except Exception:
+        pass
+    if mx is None:
+        mx = 2400  # MI355X nominal peak
+    x = torch.randn(2048, 2048, device="cuda", dtype=torch.bfloat16)
+    t0 = time.perf_counter()
+    while time.perf_counter() - t0 < timeout_s:
+        for _ in range(10):
+            x = x @ x * 1e-3
+        torch.cuda.synchronize()
+        clk = read_clock()
+        if clk is not None and clk >= frac * mx:
+            return True
+    return False
