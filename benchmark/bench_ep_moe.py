@@ -24,6 +24,8 @@ def main():
     p.add_argument("--topk", type=int, default=8)
     p.add_argument("--iters", type=int, default=50)
     p.add_argument("--no-ll", action="store_true")
+    p.add_argument("--fp8", action="store_true",
+                   help="fp8 wire + in-register dequant grouped GEMM")
     p.add_argument("--check", action="store_true")
     args = p.parse_args()
 
@@ -39,7 +41,8 @@ def main():
     e_loc = E // world
     dev = "cuda" if torch.cuda.is_available() else "cpu"
     T, H, I, K = args.tokens, args.hidden, args.inter, args.topk
-    ctx = create_ep_context(T, H, E, K, low_latency=not args.no_ll)
+    ctx = create_ep_context(T, H, E, K, low_latency=not args.no_ll,
+                            fp8=args.fp8)
     g = torch.Generator(dev).manual_seed(7)
     full_gu = (torch.randn(E, 2 * I, H, device=dev, generator=g) * 0.02
                ).to(torch.bfloat16)
@@ -71,7 +74,8 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
     if rank == 0:
         print(f"EP MoE T={T}/rank H={H} I={I} E={E} topk={K} world={world} "
-              f"ll={not args.no_ll}: {float(t.item()) * 1e3:.1f} us/call")
+              f"ll={not args.no_ll} fp8={args.fp8}: "
+              f"{float(t.item()) * 1e3:.1f} us/call")
     td.finalize_distributed()
 
 
