@@ -1254,13 +1254,17 @@ void launch_moe_grouped_gemm_pq(const void *xin, const void *weights,
                                 int e_loc, int fuse_swiglu) {
   if (n % 128 || k % 64)
     throw std::runtime_error("grouped gemm pq: N%128/K%64 required");
-  // 2-buffer default: 41 KiB LDS -> 3 blocks/CU measured 27.3 -> 25.8
-  // ms/step on qwen3-30b-a3b (TD_MOE_PQ3=1 reverts to the r01 3-buffer)
-  static const bool three_buf = [] {
-    const char *e = getenv("TD_MOE_PQ3");
-    return e && e[0] == '1';
+  // buffer depth by K: short K (qwen3-30b H=2048, 32 steps) wins with
+  // 2 buffers / 3 blocks-per-CU (27.3 -> 25.8 ms/step); long K
+  // (DeepSeek-ish H=7168, 112 steps) pays the 2-buf full drain per step
+  // and prefers the counted 3-buffer pipeline. TD_MOE_PQ2/PQ3 force.
+  static const int force = [] {
+    const char *e2 = getenv("TD_MOE_PQ2");
+    const char *e3 = getenv("TD_MOE_PQ3");
+    return (e2 && e2[0] == '1') ? 2 : (e3 && e3[0] == '1') ? 3 : 0;
   }();
-  if (!three_buf) {
+  const bool use2 = force ? (force == 2) : (k < 4096);
+  if (use2) {
     hipLaunchKernelGGL((k_moe_grouped_gemm_pq<2>), dim3(1024), dim3(256),
                        0, stream, (const bf16 *)xin, (const bf16 *)weights,
                        (bf16 *)out, (const int *)expert_base,
