@@ -1254,16 +1254,15 @@ void launch_moe_grouped_gemm_pq(const void *xin, const void *weights,
                                 int e_loc, int fuse_swiglu) {
   if (n % 128 || k % 64)
     throw std::runtime_error("grouped gemm pq: N%128/K%64 required");
-  // buffer depth by K: short K (qwen3-30b H=2048, 32 steps) wins with
-  // 2 buffers / 3 blocks-per-CU (27.3 -> 25.8 ms/step); long K
-  // (DeepSeek-ish H=7168, 112 steps) pays the 2-buf full drain per step
-  // and prefers the counted 3-buffer pipeline. TD_MOE_PQ2/PQ3 force.
-  static const int force = [] {
-    const char *e2 = getenv("TD_MOE_PQ2");
-    const char *e3 = getenv("TD_MOE_PQ3");
-    return (e2 && e2[0] == '1') ? 2 : (e3 && e3[0] == '1') ? 3 : 0;
+  // 2-buffer / 3-blocks-per-CU wins at BOTH measured regimes (qwen3-30b
+  // K=2048: 27.3 -> 25.8 ms/step; DeepSeek-ish K=7168: 686 vs 740 us) —
+  // the occupancy gain beats the deeper counted pipeline even at long K.
+  // TD_MOE_PQ3=1 forces the 3-buffer variant for A/Bs.
+  static const bool three_buf = [] {
+    const char *e = getenv("TD_MOE_PQ3");
+    return e && e[0] == '1';
   }();
-  const bool use2 = force ? (force == 2) : (k < 4096);
+  const bool use2 = !three_buf;
   if (use2) {
     hipLaunchKernelGGL((k_moe_grouped_gemm_pq<2>), dim3(1024), dim3(256),
                        0, stream, (const bf16 *)xin, (const bf16 *)weights,
