@@ -663,16 +663,20 @@ __global__ __launch_bounds__(256) void k_flash_prefill(
     }
   f32x4 accO[2][8] = {};  // [row frag][d col group of 16]
 
-  // causal: this block's rows reach q0+127 -> kv tiles up to that row
+  // causal: this block's rows reach q0+127 -> kv tiles up to that row.
+  // KV tiles are prefetched through registers (2 chunks per matrix per
+  // thread): tile t+1's HBM loads issue before tile t's compute — the
+  // same single-LDS register-prefetch structure as the decode kernel.
   const int kv_max = causal ? min(s, q0 + 128) : s;
   const int ntiles = (kv_max + kTile - 1) / kTile;
-  for (int tile = 0; tile < ntiles; ++tile) {
-    const int pos0 = tile * kTile;
-    __syncthreads();
-    for (int i = tid; i < kTile * kD / 8; i += 256) {
+  bf16x8 nk[2], nv[2];
+  auto load_kv = [&](int tile, bf16x8 *rk, bf16x8 *rv) {
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      int i = it * 256 + tid;
       int r = i / (kD / 8);
       int c = (i % (kD / 8)) * 8;
-      int pos = pos0 + r;
+      int pos = tile * kTile + r;
       bf16x8 kv8{}, vv8{};
       if (pos < s) {
         // kb_stride lets k/v live in the KV cache ([b, max_len, kvh, D])
@@ -681,10 +685,27 @@ __global__ __launch_bounds__(256) void k_flash_prefill(
         kv8 = *(const bf16x8 *)(k + base);
         vv8 = *(const bf16x8 *)(v + base);
       }
-      *(bf16x8 *)(&k_lds[r][c]) = kv8;
-      *(bf16x8 *)(&v_lds[r][c]) = vv8;
+      rk[it] = kv8;
+      rv[it] = vv8;
     }
-    __syncthreads();
+  };
+  auto store_kv = [&](const bf16x8 *rk, const bf16x8 *rv) {
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      int i = it * 256 + tid;
+      int r = i / (kD / 8);
+      int c = (i % (kD / 8)) * 8;
+      *(bf16x8 *)(&k_lds[r][c]) = rk[it];
+      *(bf16x8 *)(&v_lds[r][c]) = rv[it];
+    }
+  };
+  load_kv(0, nk, nv);
+  for (int tile = 0; tile < ntiles; ++tile) {
+    const int pos0 = tile * kTile;
+    __syncthreads();  // previous tile's LDS reads retired
+    store_kv(nk, nv);
+    if (tile + 1 < ntiles) load_kv(tile + 1, nk, nv);
+    __syncthreads();  // stores visible
 
     // S = Q K^T: B operand cols = kv pos (lane&15 + j*16)
     f32x4 accS[2][2] = {};
