@@ -55,6 +55,13 @@ class MegaGraph:
     def next_level(self):
         self.level += 1
 
+    def set_level(self, lvl: int):
+        """Explicit level placement (pipelined chains interleave group
+        levels; the scheduler orders queues by task level, so any
+        monotone-with-deps assignment keeps the deadlock-freedom
+        argument intact)."""
+        self.level = lvl
+
     def add_task(self, ttype: int, op: Op, args: List[int],
                  deps: List[Tuple[Op, int]] = ()):
         d = [(-1, 0), (-1, 0)]

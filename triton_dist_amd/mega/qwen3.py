@@ -86,6 +86,87 @@ class MegaQwen3Decode:
                              c_buf.data_ptr(), batch, n, k, dep,
                              ksplit=ksplit, ws_ptr=ws_ptr)
 
+        # Tile-granular pipelined MLP chain (TD_MK_PIPE=1, bsz=1): the
+        # whole-op level-synchronized deps measured 70% WG wait
+        # (profiles/README.md r02). Here the chain is split into G
+        # column groups with K-sliced deps — gate_up tiles of group g
+        # feed ONLY group-g swiglu chunks, which feed ONLY the down
+        # K-slice s=g — and the emission levels interleave group chains,
+        # so each workgroup's queue is a software pipeline (group g+1
+        # parts run while group g's tail still drains).
+        GPIPE = 4
+        pipe = (os.environ.get("TD_MK_PIPE") == "1" and batch == 1
+                and ksplit in (1, GPIPE) and i_s % (GPIPE * 128) == 0
+                and (2 * i_s) % 128 == 0 and H % 128 == 0)
+
+        def emit_mlp_pipe(ml, dep):
+            I = i_s
+            gu_ptr, act_ptr, mo_ptr = (self.gu.data_ptr(),
+                                       self.act.data_ptr(),
+                                       self.mo.data_ptr())
+            # fp32 ws for both GEMMs' K-split parts
+            for key_ptr, n_out in ((gu_ptr, 2 * I), (mo_ptr, H)):
+                kkey = (key_ptr, n_out)
+                if kkey not in self._ws:
+                    self._ws[kkey] = torch.zeros(GPIPE, bp, n_out,
+                                                 dtype=torch.float32,
+                                                 device=dev)
+            ws_gu = self._ws[(gu_ptr, 2 * I)].data_ptr()
+            ws_mo = self._ws[(mo_ptr, H)].data_ptr()
+            h_ptr = self.h.data_ptr()
+            wgu = ml.w_gate_up.data_ptr()
+            wdn = ml.w_down.data_ptr()
+            tiles_gu = (2 * I) // 128
+            tiles_dn = H // 128
+            kgu = H          # gate_up K
+            klen_gu = kgu // GPIPE
+            klen_dn = I // GPIPE
+            nchunks = GPIPE * 4     # swiglu chunks (4 per group)
+            cpg = nchunks // GPIPE
+            RG = [g.new_op() for _ in range(GPIPE)]
+            SW = [g.new_op() for _ in range(GPIPE)]
+            DPS = [g.new_op() for _ in range(tiles_dn)]
+            base_lvl = g.level
+            # gate_up: per-tile part ops + group reduces
+            for tn in range(tiles_gu):
+                n0 = tn * 128
+                pc = n0 if n0 < I else n0 - I
+                grp = pc * GPIPE // I
+                g.set_level(base_lvl + grp * 5)
+                pt = g.new_op()
+                for sk2 in range(GPIPE):
+                    g.add_task(T_GEMM_TILE_PART, pt,
+                               [h_ptr, wgu, ws_gu, batch, 2 * I, kgu, 0,
+                                tn, sk2 * klen_gu, klen_gu, sk2],
+                               [(dep, 0)])
+                g.set_level(base_lvl + grp * 5 + 1)
+                g.add_task(T_TILE_REDUCE, RG[grp],
+                           [ws_gu, gu_ptr, batch, 2 * I, 0, tn, GPIPE],
+                           [(pt, 0)])
+            # swiglu per group
+            for grp in range(GPIPE):
+                g.set_level(base_lvl + grp * 5 + 2)
+                for c in range(grp * cpg, (grp + 1) * cpg):
+                    g.add_task(T_SWIGLU, SW[grp],
+                               [gu_ptr, act_ptr, batch, I, c, nchunks],
+                               [(RG[grp], 0)])
+            # down: per-tile K-slice parts (slice s deps only group s)
+            for grp in range(GPIPE):
+                g.set_level(base_lvl + grp * 5 + 3)
+                for dt in range(tiles_dn):
+                    g.add_task(T_GEMM_TILE_PART, DPS[dt],
+                               [act_ptr, wdn, ws_mo, batch, H, I, 0, dt,
+                                grp * klen_dn, klen_dn, grp],
+                               [(SW[grp], 0)])
+            g.set_level(base_lvl + (GPIPE - 1) * 5 + 4)
+            dr = g.new_op()
+            for dt in range(tiles_dn):
+                g.add_task(T_TILE_REDUCE, dr,
+                           [ws_mo, mo_ptr, batch, H, 0, dt, GPIPE],
+                           [(DPS[dt], 0)])
+            g.set_level(base_lvl + (GPIPE - 1) * 5 + 5)
+            return dr
+
         # embed
         emb = g.new_op()
         for r in range(batch):
@@ -141,15 +222,19 @@ class MegaQwen3Decode:
                             self.x.data_ptr(), ln2.data_ptr(),
                             self.h.data_ptr(), batch, H, r], [(o_op, 0)])
             g.next_level()
-            gu_op = gemm(self.h, ml.w_gate_up, self.gu, 2 * i_s, H, ar2)
-            sw = g.new_op()
-            nchunks = max(tiles_m * i_s // 1024, 8)
-            for c in range(nchunks):
-                g.add_task(T_SWIGLU, sw,
-                           [self.gu.data_ptr(), self.act.data_ptr(), batch,
-                            i_s, c, nchunks], [(gu_op, 0)])
-            g.next_level()
-            mo_op = gemm(self.act, ml.w_down, self.mo, H, i_s, sw)
+            if pipe:
+                mo_op = emit_mlp_pipe(ml, ar2)
+            else:
+                gu_op = gemm(self.h, ml.w_gate_up, self.gu, 2 * i_s, H,
+                             ar2)
+                sw = g.new_op()
+                nchunks = max(tiles_m * i_s // 1024, 8)
+                for c in range(nchunks):
+                    g.add_task(T_SWIGLU, sw,
+                               [self.gu.data_ptr(), self.act.data_ptr(),
+                                batch, i_s, c, nchunks], [(gu_op, 0)])
+                g.next_level()
+                mo_op = gemm(self.act, ml.w_down, self.mo, H, i_s, sw)
             pending, prev = mo_op, mo_op
 
         # final: xn = rms(x + mo); logits = xn @ lm_head^T; kv.offset += 1
