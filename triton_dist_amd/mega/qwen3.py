@@ -12,7 +12,8 @@ import torch
 from ..models.dense import DenseLLM
 from ..models.kv_cache import KVCache
 from .builder import (emit_gemm, MegaGraph, MegaRun, T_ADD_RMSNORM, T_EMBED,
-                      T_FLASH_DECODE, T_GEMM_TILE, T_KV_ADVANCE,
+                      T_FLASH_DECODE, T_GEMM_TILE, T_GEMM_TILE_PART,
+                      T_TILE_REDUCE, T_KV_ADVANCE,
                       T_QKV_PROLOGUE, T_RMSNORM, T_SWIGLU)
 
 
