@@ -603,53 +603,20 @@ __global__ __launch_bounds__(mk::NTH) void k_megakernel(
                                // 3-buf union capped residency at 2)
   const int wg = blockIdx.x;
   const int q_lo = queue_off[wg], q_hi = queue_off[wg + 1];
-  // In-queue OUT-OF-ORDER pickup: thread 0 scans a 16-task window of
-  // THIS workgroup's queue and picks the first not-yet-done task whose
-  // deps are satisfied (relaxed loads); if none is ready it blocks on
-  // the oldest undone task — the in-order fallback keeps the original
-  // deadlock-freedom argument intact, while ready later tasks (e.g. the
-  // next op's independent tiles) run instead of head-of-line stalling
-  // (measured 70% WG wait with strict in-order execution).
-  constexpr int WND = 16;
-  __shared__ int pick_sh;
-  unsigned done_mask = 0;  // bit w = task head+w done (thread0 state)
-  int head = q_lo;
-  while (head < q_hi) {
-    if (threadIdx.x == 0) {
-      const int wlim = min(WND, q_hi - head);
-      int pick = -1;
-      for (int w = 0; w < wlim; ++w) {
-        if (done_mask & (1u << w)) continue;
-        const mk::Task &tw = tasks[queue[head + w]];
-        bool ready =
-            (tw.dep0 < 0 ||
-             ld_relaxed<Scope::Gpu>(scoreboard + tw.dep0) >= tw.dep0_n) &&
-            (tw.dep1 < 0 ||
-             ld_relaxed<Scope::Gpu>(scoreboard + tw.dep1) >= tw.dep1_n);
-        if (ready) {
-          pick = w;
-          break;
-        }
-      }
-      if (pick < 0) {
-        // blocking fallback: oldest undone task
-        int w0 = 0;
-        while (done_mask & (1u << w0)) ++w0;
-        const mk::Task &tw = tasks[queue[head + w0]];
-        if (tw.dep0 >= 0)
-          mk::spin_ge_relaxed(scoreboard + tw.dep0, tw.dep0_n);
-        if (tw.dep1 >= 0)
-          mk::spin_ge_relaxed(scoreboard + tw.dep1, tw.dep1_n);
-        pick = w0;
-      }
-      __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
-      pick_sh = pick;
-    }
-    __syncthreads();
-    const int qi = head + pick_sh;
+  for (int qi = q_lo; qi < q_hi; ++qi) {
     const mk::Task t = tasks[queue[qi]];
     const unsigned long long t0 = prof ? wallclock() : 0;
-    const unsigned long long t1 = t0;
+    // dependency waits: relaxed spins + ONE acquire fence (a per-
+    // iteration agent-scope acquire invalidates the XCD L2 every spin —
+    // measured poison at 10k+ tasks/step)
+    if (threadIdx.x == 0) {
+      if (t.dep0 >= 0) mk::spin_ge_relaxed(scoreboard + t.dep0, t.dep0_n);
+      if (t.dep1 >= 0) mk::spin_ge_relaxed(scoreboard + t.dep1, t.dep1_n);
+      if (t.dep0 >= 0 || t.dep1 >= 0)
+        __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+    }
+    __syncthreads();
+    const unsigned long long t1 = prof ? wallclock() : 0;
     switch (t.type) {
       case mk::T_RMSNORM:
         mk::t_rmsnorm(t, false);
@@ -704,15 +671,7 @@ __global__ __launch_bounds__(mk::NTH) void k_megakernel(
         atomic_add<Scope::Gpu>(prof + ty * 2, t1 - t0);
         atomic_add<Scope::Gpu>(prof + ty * 2 + 1, wallclock() - t1);
       }
-      done_mask |= 1u << pick_sh;
-      while (done_mask & 1u) {
-        done_mask >>= 1;
-        ++head;
-      }
-      pick_sh = head;  // stash head for other threads via shared
     }
-    __syncthreads();
-    if (threadIdx.x != 0) head = pick_sh;  // keep heads in lockstep
     __syncthreads();
   }
 }
