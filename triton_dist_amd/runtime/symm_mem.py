@@ -85,7 +85,7 @@ class SymmHeap:
         self.backend = backend
         if size_mb is None:
             size_mb = int(os.environ.get(
-                "TD_SYMM_HEAP_MB", "4096" if backend == "hip" else "64"))
+                "TD_SYMM_HEAP_MB", "4096" if backend == "hip" else "256"))
         self.size = size_mb * 1024 * 1024
         self._offset = 0
         self._epoch = 0
