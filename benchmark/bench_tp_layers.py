@@ -64,8 +64,11 @@ def main():
             torch.bfloat16)
         if args.check and mode == "ag_rs":
             # gather the shard stream and compare vs the replicated golden
-            xs = [torch.empty_like(x) for _ in range(world)]
-            dist.all_gather(xs, x.cpu() if dev == "cuda" else x)
+            # stage via CPU: routes to gloo under both the gloo-only
+            # (ranks sharing one GPU) and cpu:gloo,cuda:nccl groups
+            x_cpu = x.cpu()
+            xs = [torch.empty_like(x_cpu) for _ in range(world)]
+            dist.all_gather(xs, x_cpu)
             full = torch.cat([t.to(dev) for t in xs])
             y = mlp(x)
             ref = mlp.torch_fwd(full)[rank * args.tokens:

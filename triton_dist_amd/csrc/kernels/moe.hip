@@ -418,8 +418,8 @@ __global__ void k_wait_flags_ge_cell(const int *flags, int n,
                                      const int *cell, int delta) {
   int bound = *cell + delta;
   if (bound <= 0) return;
-  if (threadIdx.x < (unsigned)n)
-    wait_ge_one<Scope::Sys>(flags + threadIdx.x, bound);
+  for (int i = threadIdx.x; i < n; i += blockDim.x)
+    wait_ge_one<Scope::Sys>(flags + i, bound);
   __syncthreads();
 }
 
@@ -433,8 +433,8 @@ __global__ void k_signal_credit(PeerTable pt, size_t credit_off,
 
 // wait for all world dispatch flags (prefix kernel before the expert GEMM)
 __global__ void k_moe_wait_flags(const int *flags, int world) {
-  if (threadIdx.x < (unsigned)world)
-    wait_ge_one<Scope::Sys>(flags + threadIdx.x, 1);
+  for (int i = threadIdx.x; i < world; i += blockDim.x)
+    wait_ge_one<Scope::Sys>(flags + i, 1);
 }
 
 // ---------------------------------------------------------------------------

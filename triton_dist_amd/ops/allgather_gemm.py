@@ -135,7 +135,15 @@ def ag_gemm(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
         eligible = (m % 256 == 0 and n % 256 == 0
                     and k % 128 == 0 and ctx.ag_arrive is not None)
         import os
-        if not eligible:
+        from ..utils.distributed import gpu_oversubscribed
+        grid_wgs = (world * m // 256) * ((n + 255) // 256)
+        if gpu_oversubscribed(world) and world * grid_wgs > 448:
+            # ranks sharing one GPU (validation) with a consumer grid too
+            # big for every process's spinners to co-reside: the peers'
+            # producer kernels starve behind the spin-wait grids until
+            # the 30 s watchdog traps. Use the non-spinning barrier path.
+            method = "barrier"
+        elif not eligible:
             method = "push"
         elif torch.cuda.is_current_stream_capturing():
             method = _ag_method_cached(m, n, k, world)
@@ -145,6 +153,8 @@ def ag_gemm(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
             method = _ag_method_cached(m, n, k, world)
     if method == "fused":
         return _ag_gemm_fused(a, w, ctx, out, gathered_out)
+    if method == "barrier":
+        return _ag_gemm_barrier(a, w, ctx, out, gathered_out, m, n, k)
     rows_per_chunk = m // chunks
     m_chunks = chunks
     chunk_bytes = rows_per_chunk * k * 2
@@ -205,6 +215,44 @@ def ag_gemm(a: torch.Tensor, w: torch.Tensor, ctx: AGGemmContext,
     for s in range(min(ns, max(world - 1, 1))):
         ctx.join_evs[s].record(ctx.comm_streams[s])
         compute.wait_event(ctx.join_evs[s])
+    if gathered_out:
+        g = ctx.ws.local()[:, :m].reshape(m_total, k) \
+            if m < ctx.max_m_per_rank \
+            else ctx.ws.local().reshape(m_total, k)
+        return out, g
+    return out
+
+
+def _ag_gemm_barrier(a, w, ctx, out, gathered_out, m, n, k):
+    """Non-spinning AG-GEMM for oversubscribed GPUs (several ranks on one
+    device — a validation topology, never production): every rank pushes
+    its shard on the COMPUTE stream, a symm barrier proves all segments
+    landed, then the tile consumer runs with expect=0 so its per-chunk
+    waits are satisfied immediately (a plain gathered-workspace GEMM).
+    No overlap, no spin-wait grids — immune to cross-process CU-slot
+    starvation."""
+    world, rank = ctx.world, ctx.rank
+    heap, _C = ctx.heap, ctx.heap._C
+    compute = torch.cuda.current_stream()
+    s = compute.cuda_stream
+    # barrier 1: peers finished consuming the previous call's workspace
+    heap.barrier_all_on_stream(compute)
+    seg_bytes = ctx.max_m_per_rank * k * 2
+    _C.memcpy_async(ctx.ws.ptr() + rank * seg_bytes, a.data_ptr(),
+                    m * k * 2, s)
+    for i in range(world - 1):
+        peer = (rank + 1 + i) % world
+        _C.memcpy_async(ctx.ws.ptr(peer) + rank * seg_bytes, a.data_ptr(),
+                        m * k * 2, s)
+    # barrier 2: every rank's pushes landed before anyone's GEMM reads
+    heap.barrier_all_on_stream(compute)
+    m_total = world * m
+    if out is None:
+        out = torch.empty(m_total, n, dtype=torch.bfloat16, device=a.device)
+    _C.ag_gemm_consumer_bf16(
+        ctx.ws.ptr(), w.data_ptr(), out.data_ptr(), m_total, n, k,
+        ctx.flags.ptr(), ctx.chunks_per_rank, m, ctx.max_m_per_rank,
+        world, rank, 0, s, 0, 0, 0)
     if gathered_out:
         g = ctx.ws.local()[:, :m].reshape(m_total, k) \
             if m < ctx.max_m_per_rank \

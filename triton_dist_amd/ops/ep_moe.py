@@ -17,6 +17,7 @@ import torch
 
 from ..runtime import cpu_shm
 from ..runtime.symm_mem import SymmBuffer, SymmHeap, get_heap
+from ..utils.distributed import gpu_oversubscribed
 
 
 @dataclass
@@ -231,6 +232,8 @@ def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
     # phase 4: dispatch (xGMI row push + per-dst completion signals);
     # fp8 mode quantizes the payload (groupwise scales) and dequantizes
     # after the wait — halves the wire bytes
+    safe = gpu_oversubscribed(world)
+    forked = False
     if ctx.fp8:
         recv_q_off = ctx.recv_q.offset + parity * ctx.cap * H
         recv_s_off = ctx.recv_scale.offset + parity * ctx.cap * (H // 128) * 4
@@ -248,12 +251,32 @@ def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
                            heap.ptr(rank, recv_s_off),
                            heap.ptr(rank, recv_x_off),
                            L["recv_total"].data_ptr(), ctx.cap, H, s)
+    elif safe:
+        # ranks sharing one GPU (validation boxes): the spin-gated
+        # overlap paths can cross-process starve — both ranks' GEMM
+        # grids occupy every CU slot spinning on eflags while the peer
+        # process's dispatch kernel waits for a slot, until the 30 s
+        # spin watchdog traps. Serialize instead: dispatch on the
+        # compute stream, one-workgroup wait on all world*e_loc
+        # (src, expert) flags, then UNGATED grouped GEMMs.
+        _C.moe_dispatch(x.data_ptr(), topk_ids.data_ptr(),
+                        L["send_pos"].data_ptr(), L["send_base"].data_ptr(),
+                        L["counts"].data_ptr(), recv_x_off,
+                        meta_off, eflags_off,
+                        L["arrive_e"].data_ptr(), T, K, H, e_loc, E, s,
+                        cell)
+        if ctx.low_latency:
+            _C.wait_flags_ge_cell(heap.ptr(rank, eflags_off),
+                                  world * e_loc, cell, 0, s)
+        else:
+            _C.wait_eq(heap.ptr(rank, eflags_off), world * e_loc, 1, s)
     elif not (small_m and not ctx.low_latency):
         # dispatch rides the comm stream; the expert GEMM is gated
         # per-(expert, tile) on eflags, so FFN tiles of early-complete
         # experts run while slow sources still stream (per-expert
         # overlap; reference kernels/amd/ep_all2all_fused.py:316
         # capability — behavior only)
+        forked = True
         comm = L["comm_stream"]
         L["ev_fork"].record(stream)
         comm.wait_event(L["ev_fork"])
@@ -271,7 +294,7 @@ def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
     # +128-row slack everywhere an edge GEMM tile may over-read
     expert_h = None if fused_swiglu else torch.empty(
         ctx.cap + 128, 2 * inter, dtype=torch.bfloat16, device=x.device)
-    gate = 0 if ctx.fp8 else heap.ptr(rank, eflags_off)
+    gate = 0 if (ctx.fp8 or safe) else heap.ptr(rank, eflags_off)
     act = torch.empty(ctx.cap + 128, inter, dtype=torch.bfloat16,
                       device=x.device)
     g1_out = act if fused_swiglu else expert_h
@@ -288,7 +311,7 @@ def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
             L["expert_base"].data_ptr(), L["expert_rows"].data_ptr(),
             L["work_items"].data_ptr(), L["work_count"].data_ptr(),
             2 * inter, H, fs, s)
-    elif small_m and not ctx.fp8 and not ctx.low_latency:
+    elif small_m and not ctx.fp8 and not ctx.low_latency and not safe:
         # SINGLE-LAUNCH mega-kernel: dispatch producer workgroups +
         # per-expert-gated grouped GEMM in one kernel (closes the
         # reference's ep_all2all_fused.py:316 row completely)
@@ -334,7 +357,7 @@ def ep_moe_forward(x: torch.Tensor, topk_ids: torch.Tensor,
 
     # phase 6: combine (return rows + weighted reduce); join the
     # dispatch fork first (graph hygiene — by now it long completed)
-    if not ctx.fp8 and not (small_m and not ctx.low_latency):
+    if forked:
         stream.wait_event(L["ev_join"])
     _C.moe_combine_send(expert_out.data_ptr(), heap.ptr(rank, meta_off),
                         L["recv_total"].data_ptr(),

@@ -46,6 +46,19 @@ def local_device() -> int:
     return env_local_rank() % max(n, 1)
 
 
+def gpu_oversubscribed(world: Optional[int] = None) -> bool:
+    """True when ranks outnumber visible GPUs (a validation box running a
+    2-rank world on one GPU). Two processes' large spin-wait grids can
+    then occupy every CU slot and starve each other's producer/dispatch
+    kernels until the 30 s spin watchdog traps — callers must route to
+    non-spinning (or single-workgroup-spin) fallbacks. On a one-rank-per-
+    GPU deployment this is always False and the overlap paths run."""
+    if not has_gpu():
+        return False
+    w = world if world is not None else env_world_size()
+    return w > torch.cuda.device_count()
+
+
 def initialize_distributed(seed: int = 42, timeout_s: int = 1800,
                            backend: Optional[str] = None):
     """Init the process group (gloo on CPU-only hosts, gloo+RCCL on GPU),
