@@ -87,3 +87,18 @@ def test_allgather_cpu_2rank():
 
 def test_allgather_cpu_4rank():
     run_distributed(_body_allgather, world_size=4)
+
+
+def test_gpu_oversubscribed_helper(monkeypatch):
+    """No GPU here -> always False; with a fake 1-GPU host, a 2-rank
+    world reports oversubscribed (the starvation-fallback trigger)."""
+    from triton_dist_amd.utils import distributed as d
+
+    assert d.gpu_oversubscribed() is False
+    assert d.gpu_oversubscribed(world=8) is False
+    monkeypatch.setattr(d, "has_gpu", lambda: True)
+    monkeypatch.setattr(d.torch.cuda, "device_count", lambda: 1)
+    assert d.gpu_oversubscribed(world=2) is True
+    assert d.gpu_oversubscribed(world=1) is False
+    monkeypatch.setattr(d.torch.cuda, "device_count", lambda: 8)
+    assert d.gpu_oversubscribed(world=8) is False
