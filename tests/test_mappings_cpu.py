@@ -484,3 +484,80 @@ def test_gemm256_v3_stage_quad_consistency(band):
             want = {wc * 64 + h * 32 + r for wc in range(4)
                     for r in range(32)}
         assert staged == want, (band, h)
+
+
+def _gs_swz(row, j):
+    return j ^ ((row >> 1) & 3)
+
+
+def test_gemm_stream_stage_read_consistency():
+    """gemm_stream.hip: each MFMA fragment read returns exactly the
+    logical (row, k-chunk) elements the stage wrote (swizzle applied on
+    both sides), for A (512x32) and B (64x32)."""
+    BK = 32
+    # A: instruction it in 0..3, wave, lane -> unit u; LDS elem off = u*8
+    lds_src = {}          # lds unit -> (row, logical chunk)
+    for it in range(4):
+        for wave in range(8):
+            for lane in range(64):
+                u = it * 512 + wave * 64 + lane
+                row, jp = u >> 2, u & 3
+                lds_src[u] = (row, _gs_swz(row, jp))
+    for wave in range(8):
+        for i in range(4):
+            for lane in range(64):
+                row = wave * 64 + i * 16 + (lane & 15)
+                kq = (lane >> 4) * 8
+                u = row * 4 + _gs_swz(row, kq >> 3)
+                srow, schunk = lds_src[u]
+                assert srow == row and schunk * 8 == kq, (wave, i, lane)
+    # B: one instruction per wave, lanes 0..31 -> unit wave*32+lane
+    lds_src = {}
+    for wave in range(8):
+        for lane in range(32):
+            u = wave * 32 + lane
+            row, jp = u >> 2, u & 3
+            lds_src[u] = (row, _gs_swz(row, jp))
+    assert len(lds_src) == 256  # full 64x32 slice staged exactly once
+    for wave in range(8):
+        for j in range(4):
+            for lane in range(64):
+                row = j * 16 + (lane & 15)
+                kq = (lane >> 4) * 8
+                u = row * 4 + _gs_swz(row, kq >> 3)
+                srow, schunk = lds_src[u]
+                assert srow == row and schunk * 8 == kq, (wave, j, lane)
+
+
+def test_gemm_stream_output_coverage():
+    """Epilogue (wave, i, j, r, lane) -> (row, col) covers the 512x64
+    tile exactly once; s-major grid layout groups one k-chunk per XCD."""
+    seen = set()
+    for wave in range(8):
+        for i in range(4):
+            for j in range(4):
+                for r in range(4):
+                    for lane in range(64):
+                        row = wave * 64 + i * 16 + (lane >> 4) * 4 + r
+                        col = j * 16 + (lane & 15)
+                        key = (row, col)
+                        assert key not in seen
+                        seen.add(key)
+    assert len(seen) == 512 * 64
+
+    # grid layout: wgid = s * tiles_n + tn with xcd_remap means each
+    # XCD's contiguous range stays within one s when tiles_n == grid/8
+    def xcd_remap(wgid, nwg, nxcd=8):
+        if nwg < nxcd:
+            return wgid
+        xcd, idx = wgid % nxcd, wgid // nxcd
+        q, r2 = divmod(nwg, nxcd)
+        return (xcd * (q + 1) if xcd < r2
+                else r2 * (q + 1) + (xcd - r2) * q) + idx
+    tiles_n, sk = 80, 8   # down-proj: n=5120, k=25600
+    per_xcd_s = {}
+    for b in range(tiles_n * sk):
+        w = xcd_remap(b, tiles_n * sk)
+        per_xcd_s.setdefault(b % 8, set()).add(w // tiles_n)
+    for xcd, ss in per_xcd_s.items():
+        assert len(ss) == 1, (xcd, ss)  # one A k-chunk per XCD

@@ -102,6 +102,10 @@ void launch_gemm256_sk_bf16(const GemmArgs &g, float *ws, int sk,
 void launch_gemm256_sk2_bf16(const GemmArgs &g, float *ws, int sk,
                              hipStream_t stream);
 
+// weight-streaming decode GEMM (gemm_stream.hip): BM=512, B read once
+void launch_gemm_stream_bf16(const GemmArgs &g, float *ws, int sk,
+                             hipStream_t stream);
+
 // EXPERIMENTAL BK=64 quadrant-phase template (gemm256_v2.hip) — not in
 // any dispatch path; see the file header.
 void launch_gemm256_v2_bf16(const GemmArgs &args, hipStream_t stream);

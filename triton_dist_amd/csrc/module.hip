@@ -574,6 +574,15 @@ static void gemm256_sk2_bf16(uintptr_t a, uintptr_t b, uintptr_t c,
   TD_CHECK_HIP(hipGetLastError());
 }
 
+static void gemm_stream_bf16(uintptr_t a, uintptr_t b, uintptr_t c,
+                             uintptr_t bias, uintptr_t ws, int m, int n,
+                             int k, int sk, uintptr_t stream) {
+  GemmArgs args{(void *)a, (void *)b, (void *)c, (void *)bias,
+                m, n, k, k, k, n};
+  launch_gemm_stream_bf16(args, (float *)ws, sk, as_stream(stream));
+  TD_CHECK_HIP(hipGetLastError());
+}
+
 static void ag_gemm_consumer_splitk_bf16(uintptr_t a, uintptr_t b,
                                          uintptr_t c, uintptr_t ws, int m,
                                          int n, int k, uintptr_t flags,
@@ -960,6 +969,7 @@ PYBIND11_MODULE(_C, m) {
   m.def("gemm_skinny_bf16", &gemm_skinny_bf16);
   m.def("gemm256_sk_bf16", &gemm256_sk_bf16);
   m.def("gemm256_sk2_bf16", &gemm256_sk2_bf16);
+  m.def("gemm_stream_bf16", &gemm_stream_bf16);
   m.def("gemm256_v2_bf16", &gemm256_v2_bf16);
   m.def("gemm256_v3_bf16", &gemm256_v3_bf16);
   m.def("p2p_attributes", &p2p_attributes);
