@@ -23,11 +23,12 @@
 // Layout per WG (8 waves, 512 threads): wave w owns rows [64w, 64w+64)
 // of the 512 x 64 output; per BK=32 slice each wave runs 16
 // mfma_f32_16x16x32_bf16. LDS per buffer: A 512x32 + B 64x32 bf16 =
-// 36 KB; double buffered = 72 KB -> 2 blocks/CU. Staging uses
-// global_load_lds (16 B/lane, wave-uniform LDS base) with the g256
+// 36 KB; 4-slice ring = 144 KB -> 1 block/CU (depth beats occupancy
+// for latency hiding, same trade the 256^2 ring kernel makes). Staging
+// uses global_load_lds (16 B/lane, wave-uniform LDS base) with the g256
 // 2-bit XOR chunk swizzle on both sides; ledger: 5 loads/wave/slice
-// (4 A + 1 B with lanes 0-31), 2-buffer 1-step lookahead -> steady
-// s_waitcnt vmcnt(5), final slice vmcnt(0).
+// (4 A + 1 B with lanes 0-31), 3 slices in flight -> steady
+// s_waitcnt vmcnt(10), drain 10/5/0.
 #include <stdexcept>
 
 #include "td/api.hpp"
@@ -62,11 +63,11 @@ TD_DEV int swz(int row, int j) { return j ^ ((row >> 1) & 3); }
 
 }  // namespace gs
 
-__global__ __launch_bounds__(gs::NTH, 2) void k_gemm_stream_bf16(
+__global__ __launch_bounds__(gs::NTH, 1) void k_gemm_stream_bf16(
     GemmArgs g, float *__restrict__ ws, int sk) {
   using namespace gs;
-  __shared__ bf16 lds_a[2 * ABUF];
-  __shared__ bf16 lds_b[2 * BBUF];
+  __shared__ bf16 lds_a[4 * ABUF];
+  __shared__ bf16 lds_b[4 * BBUF];
   const int tiles_n = g.n / BN;
   // s-major layout: wgid = s * tiles_n + tn, so xcd_remap's contiguous
   // per-XCD ranges keep one k-chunk's n-tiles (one A chunk) per XCD
@@ -112,20 +113,30 @@ __global__ __launch_bounds__(gs::NTH, 2) void k_gemm_stream_bf16(
     }
   };
 
+  // 4-slice ring pipeline (the proven g256 kloop cadence): slices t..t+2
+  // in flight (15 loads/wave), ONE barrier per slice. The entry barrier
+  // of slice t proves (a) every wave's slice-t loads landed (each waited
+  // its own vmcnt before arriving) and (b) every wave consumed ring slot
+  // (t-1)%4 in the previous slice (its MFMAs — which register-consume
+  // the ds_reads — precede its barrier arrival), so staging slice t+3
+  // into that slot after the barrier is safe.
   f32x4 acc[4][4] = {};
   stage(0, 0);
+  if (ksl > 1) stage(1, 1);
+  if (ksl > 2) stage(2, 2);
   for (int t = 0; t < ksl; ++t) {
-    const int buf = t & 1;
-    if (t + 1 < ksl) {
-      stage(t + 1, buf ^ 1);
-      // 2-buffer, 1-step lookahead: slice t is newest-but-one; exactly
-      // the 5 loads of slice t+1 may remain in flight
+    const int buf = t & 3;
+    // retire slice t: 5 loads per newer in-flight slice may remain
+    if (t + 2 < ksl) {
+      asm volatile("s_waitcnt vmcnt(10)" ::: "memory");
+    } else if (t + 1 < ksl) {
       asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
     } else {
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     }
     __builtin_amdgcn_s_barrier();
     __builtin_amdgcn_sched_barrier(0);
+    if (t + 3 < ksl) stage(t + 3, (t + 3) & 3);
     const int kq = (lane >> 4) * 8;  // k offset of my frag quarter
     bf16x8 af[4], bfr[4];
 #pragma unroll
@@ -147,8 +158,6 @@ __global__ __launch_bounds__(gs::NTH, 2) void k_gemm_stream_bf16(
       for (int j = 0; j < 4; ++j)
         acc[i][j] = mfma16(af[i], bfr[j], acc[i][j]);
     __builtin_amdgcn_s_setprio(0);
-    __builtin_amdgcn_s_barrier();
-    __builtin_amdgcn_sched_barrier(0);
   }
 
   // epilogue: sk == 1 -> bf16 C (+bias); sk > 1 -> fp32 ws slice s
