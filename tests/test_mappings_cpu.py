@@ -561,3 +561,60 @@ def test_gemm_stream_output_coverage():
         per_xcd_s.setdefault(b % 8, set()).add(w // tiles_n)
     for xcd, ss in per_xcd_s.items():
         assert len(ss) == 1, (xcd, ss)  # one A k-chunk per XCD
+
+
+import pytest as _pytest
+
+
+@_pytest.mark.parametrize("nbuf,ksteps", [(2, 1), (2, 5), (3, 1), (3, 2),
+                                          (3, 7), (4, 1), (4, 2), (4, 3),
+                                          (4, 9)])
+def test_moe_pq_pipeline_ledger(nbuf, ksteps):
+    """moe.hip k_moe_grouped_gemm_pq<NBUF>: simulate the stage/wait/
+    compute schedule and check (a) every compute reads a fully-landed
+    stage, (b) no stage overwrites a buffer before its last reader, for
+    the generalized NBUF in {2,3,4} waits (0 / 5 / 10)."""
+    LOADS = 5
+    issued = []          # (step, buf) in issue order
+    landed_upto = 0      # prefix of `issued` guaranteed landed
+
+    def stage(step, buf):
+        issued.append((step, buf))
+
+    def wait_vmcnt(allowed):
+        nonlocal landed_upto
+        landed_upto = max(landed_upto, len(issued) * LOADS - allowed)
+
+    stage(0, 0)
+    if nbuf >= 3 and ksteps > 1:
+        stage(1, 1)
+    if nbuf >= 4 and ksteps > 2:
+        stage(2, 2)
+    buf_owner = {}
+    for s, b in issued:
+        buf_owner[b] = s
+    last_read = {}
+    for t in range(ksteps):
+        buf = t % nbuf
+        if nbuf >= 4 and t + 2 < ksteps:
+            wait_vmcnt(10)
+        elif nbuf >= 3 and t + 1 < ksteps:
+            wait_vmcnt(5)
+        else:
+            wait_vmcnt(0)
+        # barrier here; then stage t+ahead, then compute t
+        ahead = nbuf - 1
+        if t + ahead < ksteps:
+            b2 = (t + ahead) % nbuf
+            # overwrite safety: previous occupant's compute finished
+            prev = buf_owner.get(b2)
+            if prev is not None:
+                assert last_read.get(prev, -1) == prev, (t, b2, prev)
+            stage(t + ahead, b2)
+            buf_owner[b2] = t + ahead
+        # compute t: stage t must be fully landed
+        idx = issued.index((t, buf))
+        assert (idx + 1) * LOADS <= landed_upto, \
+            f"stage {t} not landed: {(idx+1)*LOADS} > {landed_upto}"
+        last_read[t] = t
+    assert len(last_read) == ksteps

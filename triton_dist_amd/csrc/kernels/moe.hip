@@ -619,13 +619,17 @@ __global__ __launch_bounds__(256) void k_moe_grouped_gemm_pq(
     };
     stage(0, 0);
     if (NBUF >= 3 && ksteps > 1) stage(1, 1);
+    if (NBUF >= 4 && ksteps > 2) stage(2, 2);
     for (int t = 0; t < ksteps; ++t) {
       const int buf = t % NBUF;
       const int ahead = NBUF - 1;  // staged steps in flight
-      // NBUF=3: stage t fully landed once <=5 ops (stage t+1) remain.
-      // NBUF=2: stage t is the NEWEST in flight at this wait -> full
-      // drain (the overlap is stage t+1 issuing under compute t).
-      if (NBUF >= 3 && t + 1 < ksteps) {
+      // At the wait of step t the newest staged step is t + NBUF - 2
+      // (t + NBUF - 1 is staged AFTER this wait): allow 5 ops per newer
+      // in-flight step. NBUF=2: stage t is the NEWEST in flight at this
+      // wait -> full drain (the overlap is stage t+1 under compute t).
+      if (NBUF >= 4 && t + 2 < ksteps) {
+        asm volatile("s_waitcnt vmcnt(10)" ::: "memory");
+      } else if (NBUF >= 3 && t + 1 < ksteps) {
         asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
       } else {
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -1257,27 +1261,38 @@ void launch_moe_grouped_gemm_pq(const void *xin, const void *weights,
   // 2-buffer / 3-blocks-per-CU wins at BOTH measured regimes (qwen3-30b
   // K=2048: 27.3 -> 25.8 ms/step; DeepSeek-ish K=7168: 686 vs 740 us) —
   // the occupancy gain beats the deeper counted pipeline even at long K.
-  // TD_MOE_PQ3=1 forces the 3-buffer variant for A/Bs.
-  static const bool three_buf = [] {
-    const char *e = getenv("TD_MOE_PQ3");
-    return e && e[0] == '1';
+  // TD_MOE_PQ3=1 / TD_MOE_PQ4=1 force the 3-/4-buffer variants for A/Bs
+  // (4-buffer: NBUF=3's 2-blocks-per-CU occupancy but a 3-step counted
+  // pipeline — vmcnt(10) steady instead of a full drain).
+  static const int nbuf = [] {
+    const char *e3 = getenv("TD_MOE_PQ3");
+    const char *e4 = getenv("TD_MOE_PQ4");
+    if (e4 && e4[0] == '1') return 4;
+    if (e3 && e3[0] == '1') return 3;
+    return 2;
   }();
-  const bool use2 = !three_buf;
-  if (use2) {
+  if (nbuf == 4) {
+    hipLaunchKernelGGL((k_moe_grouped_gemm_pq<4>), dim3(1024), dim3(256),
+                       0, stream, (const bf16 *)xin, (const bf16 *)weights,
+                       (bf16 *)out, (const int *)expert_base,
+                       (const int *)expert_rows, (const int *)work_items,
+                       (const int *)work_count, n, k, (const int *)eflags,
+                       (const int *)val_cell, world, e_loc, fuse_swiglu);
+  } else if (nbuf == 3) {
+    hipLaunchKernelGGL((k_moe_grouped_gemm_pq<3>), dim3(1024), dim3(256),
+                       0, stream, (const bf16 *)xin, (const bf16 *)weights,
+                       (bf16 *)out, (const int *)expert_base,
+                       (const int *)expert_rows, (const int *)work_items,
+                       (const int *)work_count, n, k, (const int *)eflags,
+                       (const int *)val_cell, world, e_loc, fuse_swiglu);
+  } else {
     hipLaunchKernelGGL((k_moe_grouped_gemm_pq<2>), dim3(1024), dim3(256),
                        0, stream, (const bf16 *)xin, (const bf16 *)weights,
                        (bf16 *)out, (const int *)expert_base,
                        (const int *)expert_rows, (const int *)work_items,
                        (const int *)work_count, n, k, (const int *)eflags,
                        (const int *)val_cell, world, e_loc, fuse_swiglu);
-    return;
   }
-  hipLaunchKernelGGL((k_moe_grouped_gemm_pq<3>), dim3(1024), dim3(256), 0,
-                     stream, (const bf16 *)xin, (const bf16 *)weights,
-                     (bf16 *)out, (const int *)expert_base,
-                     (const int *)expert_rows, (const int *)work_items,
-                     (const int *)work_count, n, k, (const int *)eflags,
-                     (const int *)val_cell, world, e_loc, fuse_swiglu);
 }
 
 void launch_moe_grouped_gemm(const void *xin, const void *weights, void *out,
