@@ -46,17 +46,42 @@ def local_device() -> int:
     return env_local_rank() % max(n, 1)
 
 
+_SHARED_GPU: Optional[bool] = None
+
+
 def gpu_oversubscribed(world: Optional[int] = None) -> bool:
-    """True when ranks outnumber visible GPUs (a validation box running a
-    2-rank world on one GPU). Two processes' large spin-wait grids can
-    then occupy every CU slot and starve each other's producer/dispatch
-    kernels until the 30 s spin watchdog traps — callers must route to
-    non-spinning (or single-workgroup-spin) fallbacks. On a one-rank-per-
-    GPU deployment this is always False and the overlap paths run."""
+    """True when two ranks share one PHYSICAL GPU (a validation box
+    running a 2-rank world on one device). Two processes' large
+    spin-wait grids can then occupy every CU slot and starve each
+    other's producer/dispatch kernels until the 30 s spin watchdog
+    traps — callers must route to non-spinning (or single-workgroup-
+    spin) fallbacks. On a one-rank-per-GPU deployment this is always
+    False and the overlap paths run.
+
+    `world > device_count` alone is ambiguous: a launcher that pins one
+    visible device per rank (HIP_VISIBLE_DEVICES) also shows count 1 at
+    world 8 without any sharing. When the group is up, ranks exchange
+    their device's (host, PCI domain/bus/device) once over gloo and
+    look for duplicates; the env heuristic is only the fallback."""
     if not has_gpu():
         return False
     w = world if world is not None else env_world_size()
-    return w > torch.cuda.device_count()
+    if w <= torch.cuda.device_count():
+        return False
+    if not dist.is_initialized():
+        return True  # conservative: cannot verify isolation
+    global _SHARED_GPU
+    if _SHARED_GPU is None:
+        import socket
+
+        p = torch.cuda.get_device_properties(torch.cuda.current_device())
+        mine = (socket.gethostname(), getattr(p, "pci_domain_id", -1),
+                getattr(p, "pci_bus_id", -1),
+                getattr(p, "pci_device_id", -1))
+        allv = [None] * dist.get_world_size()
+        dist.all_gather_object(allv, mine)
+        _SHARED_GPU = len(set(allv)) < len(allv)
+    return _SHARED_GPU
 
 
 def initialize_distributed(seed: int = 42, timeout_s: int = 1800,
@@ -98,7 +123,8 @@ def initialize_distributed(seed: int = 42, timeout_s: int = 1800,
 
 
 def finalize_distributed():
-    global _INITIALIZED
+    global _INITIALIZED, _SHARED_GPU
+    _SHARED_GPU = None
     from ..runtime.symm_mem import shutdown_heap
 
     shutdown_heap()
