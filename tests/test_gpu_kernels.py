@@ -429,3 +429,72 @@ def test_gemm256_sk2():
         torch.cuda.synchronize()
         assert_allclose(c, a.float() @ w.float().t(), msg=f"sk={sk}",
                         **bf16_gemm_tol(k))
+
+
+def test_gemm_stream():
+    """Weight-streaming decode GEMM (BM=512 register-K tier; kept as a
+    measured-negative experiment, profiles/README.md) vs fp32 ref:
+    full and partial m, bias, and the split-K reduce path."""
+    from triton_dist_amd import _C
+    from triton_dist_amd.utils.testing import assert_allclose, bf16_gemm_tol
+
+    torch.manual_seed(11)
+    s = torch.cuda.current_stream().cuda_stream
+    for m, n, k, sk, use_bias in [(512, 1536, 2048, 1, False),
+                                  (512, 1536, 2048, 4, False),
+                                  (300, 640, 1024, 1, True),
+                                  (300, 640, 1024, 2, True)]:
+        a = (torch.randn(m, k, device="cuda") / 8).to(torch.bfloat16)
+        w = (torch.randn(n, k, device="cuda") / 8).to(torch.bfloat16)
+        bias = (torch.randn(n, device="cuda") / 4).to(torch.bfloat16) \
+            if use_bias else None
+        c = torch.empty(m, n, device="cuda", dtype=torch.bfloat16)
+        ws = torch.empty(sk, m, n, dtype=torch.float32, device="cuda") \
+            if sk > 1 else c
+        _C.gemm_stream_bf16(a.data_ptr(), w.data_ptr(), c.data_ptr(),
+                            bias.data_ptr() if use_bias else 0,
+                            ws.data_ptr(), m, n, k, sk, s)
+        torch.cuda.synchronize()
+        ref = a.float() @ w.float().t()
+        if use_bias:
+            ref = ref + bias.float()
+        assert_allclose(c, ref, msg=f"m={m} sk={sk}", **bf16_gemm_tol(k))
+
+
+def test_moe_pq_nbuf_variants():
+    """k_moe_grouped_gemm_pq<3>/<4> (env-gated A/B variants) match <2>
+    numerics on a grouped decode shape."""
+    import subprocess
+    import sys
+    code = r'''
+import torch
+from triton_dist_amd import _C
+torch.manual_seed(5)
+E, rows_e, n, k = 8, 32, 256, 512
+rows = E * rows_e
+x = (torch.randn(rows + 128, k, device="cuda") / 8).to(torch.bfloat16)
+w = (torch.randn(E, n, k, device="cuda") / 8).to(torch.bfloat16)
+base = torch.arange(E, dtype=torch.int32, device="cuda") * rows_e
+erows = torch.full((E,), rows_e, dtype=torch.int32, device="cuda")
+items = torch.tensor([(e << 16) for e in range(E)], dtype=torch.int32,
+                     device="cuda")
+cnt = torch.tensor([E], dtype=torch.int32, device="cuda")
+out = torch.zeros(rows + 128, n, dtype=torch.bfloat16, device="cuda")
+s = torch.cuda.current_stream().cuda_stream
+_C.moe_grouped_gemm_pq(x.data_ptr(), w.data_ptr(), out.data_ptr(),
+                       base.data_ptr(), erows.data_ptr(),
+                       items.data_ptr(), cnt.data_ptr(), n, k, s)
+torch.cuda.synchronize()
+ref = torch.cat([x[e * rows_e:(e + 1) * rows_e].float()
+                 @ w[e].float().t() for e in range(E)])
+err = (out[:rows].float() - ref).abs().max().item()
+assert err < 0.1, err
+print("OK")
+'''
+    import os as _os
+    for env in ({}, {"TD_MOE_PQ3": "1"}, {"TD_MOE_PQ4": "1"}):
+        r = subprocess.run([sys.executable, "-c", code],
+                           env={**_os.environ, **env},
+                           capture_output=True, text=True, timeout=180)
+        assert r.returncode == 0 and "OK" in r.stdout, \
+            (env, r.stdout[-500:], r.stderr[-500:])
