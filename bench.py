@@ -91,6 +91,11 @@ def main():
     else:
         heap = td.init_symm_heap()
 
+    if on_gpu:
+        # shipped hipBLASLt/rocBLAS algo picks for the matmul-routed
+        # GEMMs (read-only TunableOp file; +6-7% on the qwen3-32b step)
+        from triton_dist_amd.tune import maybe_enable_tunableop
+        maybe_enable_tunableop()
     model = AutoLLM(cfg, device=device)
     model.init_weights(seed=args.seed)
     if args.mode in ("ag_rs", "gemm_ar"):

@@ -71,3 +71,17 @@ def test_perf_model_methods():
     assert ll_allgather_time_us(1 << 10, 8) < ll_allgather_time_us(1 << 24,
                                                                    8)
     assert reduce_scatter_time_us(1 << 20, 8) > 0
+
+
+def test_maybe_enable_tunableop_cpu_noop():
+    """No GPU here: the shipped TunableOp file must not be activated
+    (and a live tuning session must never be overridden)."""
+    from triton_dist_amd.tune import maybe_enable_tunableop
+
+    assert maybe_enable_tunableop() is False
+    import os
+    os.environ["PYTORCH_TUNABLEOP_TUNING"] = "1"
+    try:
+        assert maybe_enable_tunableop() is False
+    finally:
+        del os.environ["PYTORCH_TUNABLEOP_TUNING"]

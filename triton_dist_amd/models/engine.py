@@ -21,6 +21,9 @@ class Engine:
         self.model = model
         self.batch = batch
         self.max_len = max_len
+        if torch.cuda.is_available():
+            from ..tune import maybe_enable_tunableop
+            maybe_enable_tunableop()  # shipped library-GEMM algo picks
         cfg = model.cfg
         self.kv = model.make_cache(batch, max_len)
         if use_graph is None:

@@ -193,3 +193,33 @@ def contextual_autotune(name: str, configs: List[Dict[str, Any]],
         return wrapped
 
     return deco
+
+
+def maybe_enable_tunableop() -> bool:
+    """Activate the repo-shipped hipBLASLt/rocBLAS algorithm selections
+    (a PyTorch TunableOp result file tuned on an MI355X: decode-shaped
+    torch.matmul GEMMs measured +6-7%% on the qwen3-32b step —
+    profiles/README.md) in READ-ONLY mode. The file's own validators
+    (torch / hipBLASLt / rocBLAS / gfx arch) make a stale file a no-op,
+    and tuning stays off so the choice is hipGraph-capture-safe. A live
+    tuning session (PYTORCH_TUNABLEOP_TUNING=1) manages itself and is
+    left alone. Returns True when the file was loaded.
+
+    Counterpart of the reference's persistent AutoTuner cache idea
+    (tune.py:294,384-426 — behavior only) applied to the library-GEMM
+    backend itself."""
+    if os.environ.get("PYTORCH_TUNABLEOP_TUNING") == "1":
+        return False
+    t = getattr(torch.cuda, "tunable", None)
+    if t is None or not torch.cuda.is_available():
+        return False
+    path = Path(__file__).parent / "autotune_cache" / "tunableop_gfx950.csv"
+    if not path.exists():
+        return False
+    try:
+        t.enable(True)
+        t.tuning_enable(False)
+        t.read_file(str(path))
+    except Exception:
+        return False
+    return True
