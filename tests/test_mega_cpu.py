@@ -122,3 +122,60 @@ def test_emit_gemv_semantics():
             cols += list(range(c0, c0 + chunk))
         assert sorted(cols) == list(range(n))
         np.testing.assert_allclose(C, A @ W.T, rtol=1e-4, atol=1e-3)
+
+
+def test_fused_graph_structure():
+    """TD_MK_FUSE=1: the three per-layer serialization hops disappear —
+    no standalone rmsnorm / prologue / flash-decode tasks; norm-fused
+    GEMM partials and fused prologue+decode tasks appear instead, and
+    every dep slot points at a valid op."""
+    import os
+
+    import torch
+
+    from triton_dist_amd.mega.builder import (
+        T_ADD_RMSNORM, T_FLASH_DECODE, T_GEMM_TILE_PART,
+        T_GEMM_TILE_PART_NR, T_PRO_FLASH_DECODE, T_QKV_PROLOGUE, T_RMSNORM)
+    from triton_dist_amd.mega.qwen3 import MegaQwen3Decode
+    from triton_dist_amd.models import DenseLLM, KVCache, get_config
+
+    cfg = get_config("tiny-gpu", max_length=64)
+    model = DenseLLM(cfg, device="cpu")
+    model.init_weights(seed=3)
+    kv = KVCache(cfg.n_layers, 8, 64, cfg.n_kv_heads, cfg.head_dim,
+                 device="cpu")
+
+    def types(meg):
+        return (meg.run.task_buf[:, 0] & 0xFFFFFFFF).tolist()
+
+    os.environ["TD_MK_FUSE"] = "1"
+    try:
+        fused = MegaQwen3Decode(model, kv, batch=8)
+    finally:
+        del os.environ["TD_MK_FUSE"]
+    assert fused.fused
+    tf = types(fused)
+    assert tf.count(T_RMSNORM) == 0
+    assert tf.count(T_QKV_PROLOGUE) == 0
+    assert tf.count(T_FLASH_DECODE) == 0
+    # one fused prologue+decode per (b, kh) per layer
+    assert tf.count(T_PRO_FLASH_DECODE) == 8 * cfg.n_kv_heads * cfg.n_layers
+    # qkv + gate_up partials are norm-fused; o/down/lm stay plain
+    assert tf.count(T_GEMM_TILE_PART_NR) > 0
+    assert tf.count(T_GEMM_TILE_PART) > 0
+    # residual updates ride parallel ADD_RMSNORM tasks (ping-pong x)
+    assert tf.count(T_ADD_RMSNORM) > 0
+    # dep slots reference valid ops
+    n_ops = fused.run.n_ops
+    d0 = (fused.run.task_buf[:, 1] & 0xFFFFFFFF).to(torch.int32)
+    d1 = (fused.run.task_buf[:, 2] & 0xFFFFFFFF).to(torch.int32)
+    assert int(d0.max()) < n_ops and int(d1.max()) < n_ops
+
+    kv2 = KVCache(cfg.n_layers, 8, 64, cfg.n_kv_heads, cfg.head_dim,
+                  device="cpu")
+    plain = MegaQwen3Decode(model, kv2, batch=8)
+    assert not plain.fused
+    tp = types(plain)
+    assert tp.count(T_PRO_FLASH_DECODE) == 0
+    assert tp.count(T_GEMM_TILE_PART_NR) == 0
+    assert tp.count(T_FLASH_DECODE) == 8 * cfg.n_kv_heads * cfg.n_layers

@@ -53,6 +53,19 @@ enum TaskType : int {
                          // bandwidth-bound; ~25x fewer tasks than the
                          // 32x128 tile scheme at bsz 1, which is what the
                          // 0.25 us/task dispatch overhead demands)
+  T_PRO_FLASH_DECODE = 11,  // hop fusion (TD_MK_FUSE): qkv prologue for
+                            // ONE (b, kh)'s heads + flash decode in one
+                            // task — removes the 1-task-per-row prologue
+                            // serialization hop. a0=qkv, a1=q, a2=kc,
+                            // a3=vc, a4=cos, a5=sin, a6=qnw, a7=knw,
+                            // a8=offset, a9=attn_out, a10=pack(b,kh),
+                            // a11=pack(qh,kvh), a12=maxlen
+  T_GEMM_TILE_PART_NR = 12,  // hop fusion: K-range GEMM partial whose A
+                             // operand is rmsnorm(x [+ res]) computed on
+                             // the fly — removes the [add+]rmsnorm hop
+                             // before each projection. Args of
+                             // T_GEMM_TILE_PART with a0=x, a11=ln_w,
+                             // a12=res (0 = none)
 };
 
 struct Task {
@@ -267,6 +280,142 @@ TD_DEV void t_gemm_tile_part(const Task &t, bf16 *lds) {
     __builtin_amdgcn_s_barrier();
     __builtin_amdgcn_sched_barrier(0);
     if (ti + 1 < ksteps) stage(ti + 1, (ti + 1) & 1);
+#pragma unroll
+    for (int ks = 0; ks < BK / 32; ++ks) {
+      bf16x8 af[2], bfr[2];
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        int arow = i * 16 + (lane & 15);
+        int brow = wave * 32 + i * 16 + (lane & 15);
+        int kk = ks * 32 + (lane >> 4) * 8;
+        af[i] = *(const bf16x8 *)(lds_a + buf * ABUF + arow * BK + kk);
+        bfr[i] = *(const bf16x8 *)(lds_b + buf * BBUF + brow * BK + kk);
+      }
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[i], bfr[j], acc[i][j], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_barrier();
+    __builtin_amdgcn_sched_barrier(0);
+  }
+  const int row_lim = m - pid_m * BM;
+  float *wsl = ws + (size_t)sk * m_pad * n;
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = i * 16 + (lane >> 4) * 4 + r;
+        int col = wave * 32 + j * 16 + (lane & 15);
+        if (row < row_lim)
+          wsl[((size_t)pid_m * BM + row) * n + (size_t)pid_n * BN + col] =
+              acc[i][j][r];
+      }
+}
+
+// Hop-fused K-split partial: A = rmsnorm(x [+ res]) computed in
+// registers at stage time (per-row inv-rms over the FULL row first, then
+// VALU-normalized A staging; B keeps the async global_load_lds path).
+// Removes the one-task-per-row [add+]rmsnorm serialization hop that
+// gated every projection. The residual UPDATE (x' = x + res) runs as a
+// parallel off-critical-path task writing a ping-pong buffer, so no
+// reader ever sees a half-updated x.
+TD_DEV void t_gemm_tile_part_nr(const Task &t, bf16 *lds, float *rs) {
+  constexpr int BM = 32, BN = 128, BK = 64;
+  constexpr int ABUF = BM * BK, BBUF = BN * BK;
+  bf16 *lds_a = lds;
+  bf16 *lds_b = lds + 2 * ABUF;
+  const bf16 *X = (const bf16 *)t.a[0];
+  const bf16 *B = (const bf16 *)t.a[1];
+  float *ws = (float *)t.a[2];
+  const int m = (int)t.a[3], n = (int)t.a[4], k = (int)t.a[5];
+  const int pid_m = (int)t.a[6], pid_n = (int)t.a[7];
+  const int k0 = (int)t.a[8], klen = (int)t.a[9], sk = (int)t.a[10];
+  const bf16 *lnw = (const bf16 *)t.a[11];
+  const bf16 *res = (const bf16 *)t.a[12];
+  const int m_pad = (m + BM - 1) / BM * BM;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int row_g0 = pid_m * BM;
+  // per-row inv-rms of (x [+ res]) over the full row: 8 threads per row
+  {
+    const int r = tid >> 3, tsub = tid & 7;
+    int rr = row_g0 + r;
+    if (rr >= m) rr = m > 0 ? m - 1 : 0;
+    const bf16 *xr = X + (size_t)rr * k;
+    const bf16 *dr = res ? res + (size_t)rr * k : nullptr;
+    float ss = 0.f;
+    for (int c = tsub * 8; c < k; c += 64) {
+      bf16x8 v = *(const bf16x8 *)(xr + c);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = (float)v[j];
+        if (dr) f += (float)*(dr + c + j);
+        ss += f * f;
+      }
+    }
+#pragma unroll
+    for (int off = 1; off < 8; off <<= 1) ss += __shfl_xor(ss, off);
+    if (tsub == 0) rs[r] = rsqrtf(ss / k + 1e-6f);
+  }
+  __syncthreads();
+  f32x4 acc[2][2] = {};
+  const bf16 *gb = B + (size_t)pid_n * BN * k + k0;
+  const int ksteps = klen / BK;
+  auto stage_b = [&](int ti, int buf) {
+    const int kk0 = ti * BK;
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      int qb = it * 256 + tid;
+      int rowb = qb >> 3, kcb = qb & 7;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int *)(
+              gb + (size_t)rowb * k + kk0 + kcb * 8),
+          (__attribute__((address_space(3))) unsigned int *)(
+              lds_b + buf * BBUF + (it * 256 + wave * 64) * 8),
+          16, 0, 0);
+    }
+  };
+  auto stage_a = [&](int ti, int buf) {
+    // same 32x64 LDS layout the glds path produced (elem off = tid*8)
+    const int row = tid >> 3;
+    const int col = k0 + ti * BK + (tid & 7) * 8;
+    int rr = row_g0 + row;
+    if (rr >= m) rr = m > 0 ? m - 1 : 0;
+    bf16x8 v = *(const bf16x8 *)(X + (size_t)rr * k + col);
+    bf16x8 wv = *(const bf16x8 *)(lnw + col);
+    const float sc = rs[row];
+    bf16x8 o;
+    if (res) {
+      bf16x8 d = *(const bf16x8 *)(res + (size_t)rr * k + col);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        o[j] = (bf16)(((float)v[j] + (float)d[j]) * sc * (float)wv[j]);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        o[j] = (bf16)((float)v[j] * sc * (float)wv[j]);
+    }
+    *(bf16x8 *)(lds_a + buf * ABUF + tid * 8) = o;
+  };
+  stage_b(0, 0);
+  stage_a(0, 0);
+  for (int ti = 0; ti < ksteps; ++ti) {
+    const int buf = ti & 1;
+    // B: 2-buffer full drain (stage ti is newest in flight); A: my
+    // ds_writes for ti must land before peers read them post-barrier
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    __builtin_amdgcn_sched_barrier(0);
+    if (ti + 1 < ksteps) {
+      stage_b(ti + 1, (ti + 1) & 1);
+      stage_a(ti + 1, (ti + 1) & 1);
+    }
 #pragma unroll
     for (int ks = 0; ks < BK / 32; ++ks) {
       bf16x8 af[2], bfr[2];
@@ -525,6 +674,79 @@ TD_DEV void mk_flash_decode_body(const bf16 *q, const bf16 *kcache,
   }
 }
 
+// Hop-fused prologue + flash decode for ONE (b, kh): qk-norm + RoPE +
+// cache append for exactly this task's heads (q slots kh*G..kh*G+G-1,
+// k/v head kh — the (b, kh) partition of the prologue is exact, so no
+// cross-task dependency exists), then the standard flash-decode body
+// reading the q / cache rows this task just wrote (vmcnt-drained +
+// block-synced before the read).
+TD_DEV void mk_pro_flash_decode(const mk::Task &t, char *lds_raw) {
+  constexpr int D = 128;
+  const bf16 *qkv = (const bf16 *)t.a[0];
+  bf16 *q_out = (bf16 *)t.a[1];
+  bf16 *kcache = (bf16 *)t.a[2];
+  bf16 *vcache = (bf16 *)t.a[3];
+  const float *cos_t = (const float *)t.a[4];
+  const float *sin_t = (const float *)t.a[5];
+  const bf16 *qnw = (const bf16 *)t.a[6];
+  const bf16 *knw = (const bf16 *)t.a[7];
+  const long pos = *(const long *)t.a[8];
+  bf16 *attn_out = (bf16 *)t.a[9];
+  const int b = (int)(t.a[10] & 0xFFFFFFFF);
+  const int kh = (int)(t.a[10] >> 32);
+  const int qh = (int)(t.a[11] & 0xFFFFFFFF);
+  const int kvh = (int)(t.a[11] >> 32);
+  const int maxlen = (int)t.a[12];
+  const int G = qh / kvh;
+  const int lane = threadIdx.x & 63;
+  for (int hl = threadIdx.x >> 6; hl < G + 2; hl += mk::NTH / 64) {
+    // local head -> global qkv head: q slots first, then k, then v
+    const bool is_q = hl < G;
+    const bool is_k = hl == G;
+    const int h = is_q ? kh * G + hl : (is_k ? qh + kh : qh + kvh + kh);
+    const bf16 *src = qkv + ((size_t)b * (qh + 2 * kvh) + h) * D;
+    float v0 = (float)src[lane * 2];
+    float v1 = (float)src[lane * 2 + 1];
+    if (is_q || is_k) {
+      float ss = v0 * v0 + v1 * v1;
+      for (int off = 32; off > 0; off >>= 1) ss += __shfl_down(ss, off);
+      float scale = rsqrtf(__shfl(ss, 0) / D + 1e-6f);
+      const bf16 *nw = is_q ? qnw : knw;
+      v0 *= scale * (float)nw[lane * 2];
+      v1 *= scale * (float)nw[lane * 2 + 1];
+      float p0 = __shfl_xor(v0, 32);
+      float p1 = __shfl_xor(v1, 32);
+      int d2 = (lane & 31) * 2;
+      float c0 = cos_t[pos * (D / 2) + d2];
+      float s0 = sin_t[pos * (D / 2) + d2];
+      float c1 = cos_t[pos * (D / 2) + d2 + 1];
+      float s1 = sin_t[pos * (D / 2) + d2 + 1];
+      if (lane < 32) {
+        v0 = v0 * c0 - p0 * s0;
+        v1 = v1 * c1 - p1 * s1;
+      } else {
+        v0 = v0 * c0 + p0 * s0;
+        v1 = v1 * c1 + p1 * s1;
+      }
+    }
+    bf16 *dst;
+    if (is_q) {
+      dst = q_out + ((size_t)b * qh + h) * D;
+    } else {
+      bf16 *cache = is_k ? kcache : vcache;
+      dst = cache + (((size_t)b * maxlen + pos) * kvh + kh) * D;
+    }
+    dst[lane * 2] = (bf16)v0;
+    dst[lane * 2 + 1] = (bf16)v1;
+  }
+  // my global stores must be readable below (same block, other threads)
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  mk_flash_decode_body(q_out, kcache, vcache, attn_out,
+                       (const long *)t.a[8], b, kh, qh, kvh, maxlen,
+                       lds_raw);
+}
+
 namespace mk {
 
 // Decode GEMV: x rows staged in LDS (m <= 4), each thread owns
@@ -598,9 +820,11 @@ __global__ __launch_bounds__(mk::NTH) void k_megakernel(
     int fence_mode, unsigned long long *__restrict__ prof) {
   // prof (optional): [16 types x 2] wallclock accumulators —
   // [type][0] += dependency-wait ticks, [type][1] += body ticks
-  __shared__ char lds[40960];  // union: gemm 2-buf A/B | flash-decode
-                               // state (40 KB -> 3 blocks/CU; the 60 KB
-                               // 3-buf union capped residency at 2)
+  __shared__ char lds[41088];  // union: gemm 2-buf A/B | flash-decode
+                               // state, + 128 B row-scale scratch for
+                               // the norm-fused GEMM parts (40.1 KB ->
+                               // still 3 blocks/CU; the 60 KB 3-buf
+                               // union capped residency at 2)
   const int wg = blockIdx.x;
   const int q_lo = queue_off[wg], q_hi = queue_off[wg + 1];
   for (int qi = q_lo; qi < q_hi; ++qi) {
@@ -653,6 +877,12 @@ __global__ __launch_bounds__(mk::NTH) void k_megakernel(
         break;
       case mk::T_GEMV:
         mk::t_gemv(t, lds);
+        break;
+      case mk::T_PRO_FLASH_DECODE:
+        mk_pro_flash_decode(t, lds);
+        break;
+      case mk::T_GEMM_TILE_PART_NR:
+        mk::t_gemm_tile_part_nr(t, (bf16 *)lds, (float *)(lds + 40960));
         break;
       default:
         break;

@@ -123,25 +123,35 @@ def emit_gemv(g: "MegaGraph", a_ptr: int, w_ptr: int, c_ptr: int,
     return op
 
 
+T_PRO_FLASH_DECODE = 11  # hop fusion: qkv prologue slice + flash decode
+T_GEMM_TILE_PART_NR = 12  # hop fusion: K-partial with on-the-fly rmsnorm A
+
+
 def emit_gemm(g: "MegaGraph", a_ptr: int, w_ptr: int, c_ptr: int,
               batch: int, n: int, k: int, dep, ksplit: int = 1,
-              ws_ptr: int = 0):
+              ws_ptr: int = 0, norm=None, dep2=None):
     """Emit one logical GEMM as scoreboard ops. ksplit == 1: one
     T_GEMM_TILE per 32x128 C tile (the original scheme). ksplit > 1
     (TD_MK_KSPLIT, opt-in): each tile becomes `ksplit` K-range partial
     tasks writing fp32 ws slices plus one reduce task — shortens the
     per-op critical path from a full-K tile to K/ksplit (the 59 ms vs
     15 ms megakernel gap is op-chain latency; docs/ROADMAP.md #4).
+    norm=(lnw_ptr, res_ptr) (TD_MK_FUSE, ksplit>1 only): the A operand
+    is rmsnorm(a [+ res]) computed inside each partial — the
+    one-task-per-row norm hop disappears from the critical path. dep2:
+    optional second dependency (the ping-pong residual update).
     Returns the op consumers must depend on."""
     tiles_m = (batch + 31) // 32
     tiles_n = n // 128
+    deps = ([(dep, 0)] if dep else []) + ([(dep2, 0)] if dep2 else [])
     if ksplit <= 1 or k % (64 * ksplit):
+        assert norm is None, "norm fusion requires the ksplit path"
         op = g.new_op()
         for pm in range(tiles_m):
             for pn in range(tiles_n):
                 g.add_task(T_GEMM_TILE, op,
                            [a_ptr, w_ptr, c_ptr, batch, n, k, pm, pn],
-                           [(dep, 0)] if dep else [])
+                           deps)
         g.next_level()
         return op
     assert ws_ptr, "ksplit > 1 needs an fp32 ws [ksplit, batch_pad, n]"
@@ -150,10 +160,15 @@ def emit_gemm(g: "MegaGraph", a_ptr: int, w_ptr: int, c_ptr: int,
     for pm in range(tiles_m):
         for pn in range(tiles_n):
             for sk in range(ksplit):
-                g.add_task(T_GEMM_TILE_PART, parts,
-                           [a_ptr, w_ptr, ws_ptr, batch, n, k, pm, pn,
-                            sk * klen, klen, sk],
-                           [(dep, 0)] if dep else [])
+                if norm is None:
+                    g.add_task(T_GEMM_TILE_PART, parts,
+                               [a_ptr, w_ptr, ws_ptr, batch, n, k, pm, pn,
+                                sk * klen, klen, sk], deps)
+                else:
+                    g.add_task(T_GEMM_TILE_PART_NR, parts,
+                               [a_ptr, w_ptr, ws_ptr, batch, n, k, pm, pn,
+                                sk * klen, klen, sk, norm[0], norm[1]],
+                               deps)
     g.next_level()
     red = g.new_op()
     for pm in range(tiles_m):

@@ -13,7 +13,7 @@ from ..models.dense import DenseLLM
 from ..models.kv_cache import KVCache
 from .builder import (emit_gemm, MegaGraph, MegaRun, T_ADD_RMSNORM, T_EMBED,
                       T_FLASH_DECODE, T_GEMM_TILE, T_GEMM_TILE_PART,
-                      T_TILE_REDUCE, T_KV_ADVANCE,
+                      T_PRO_FLASH_DECODE, T_TILE_REDUCE, T_KV_ADVANCE,
                       T_QKV_PROLOGUE, T_RMSNORM, T_SWIGLU)
 
 
@@ -100,6 +100,33 @@ class MegaQwen3Decode:
                 and ksplit in (1, GPIPE) and i_s % (GPIPE * 128) == 0
                 and (2 * i_s) % 128 == 0 and H % 128 == 0)
 
+        # Hop fusion (TD_MK_FUSE=1): the three 1-task-per-row
+        # serialization hops per layer — [add+]rmsnorm before qkv,
+        # qkv-prologue, [add+]rmsnorm before gate_up — fold into their
+        # consumers (norm into the GEMM K-partials' A staging, prologue
+        # into each (b, kh) flash-decode task). The residual UPDATE runs
+        # as a parallel task writing a PING-PONG x buffer (readers of
+        # (x, delta) never see a half-updated row); TD_MK_PROF measured
+        # 70% WG wait, and these hops are the per-layer critical path.
+        fuse = (os.environ.get("TD_MK_FUSE", "0") == "1" and batch <= 32
+                and ksplit > 1 and H % (64 * ksplit) == 0)
+        self.fused = fuse
+        if fuse:
+            self.x2 = e(bp, H, dtype=bf, device=dev)
+            self.x2.zero_()
+
+        def gemm_nr(x_t, lnw_t, res_ptr, w, c_buf, n, k, dep, dep2):
+            key = (c_buf.data_ptr(), n)
+            if key not in self._ws:
+                self._ws[key] = torch.zeros(ksplit, bp, n,
+                                            dtype=torch.float32,
+                                            device=dev)
+            return emit_gemm(g, x_t.data_ptr(), w.data_ptr(),
+                             c_buf.data_ptr(), batch, n, k, dep,
+                             ksplit=ksplit,
+                             ws_ptr=self._ws[key].data_ptr(),
+                             norm=(lnw_t.data_ptr(), res_ptr), dep2=dep2)
+
         def emit_mlp_pipe(ml, dep):
             I = i_s
             gu_ptr, act_ptr, mo_ptr = (self.gu.data_ptr(),
@@ -178,56 +205,103 @@ class MegaQwen3Decode:
 
         pending = None  # (op, delta_buf) awaiting the fused residual add
         prev = emb
+        xup_prev = None            # ping-pong x-update op (fused mode)
+        x_cur, x_alt = self.x, (self.x2 if fuse else self.x)
         for li, layer in enumerate(model.layers):
             at, ml = layer["attn"], layer["mlp"]
             ln1, ln2 = layer["ln1"], layer["ln2"]
-            # h = rms(x + pending?)
-            nrm = g.new_op()
-            for r in range(batch):
-                if pending is None:
-                    g.add_task(T_RMSNORM, nrm,
-                               [self.x.data_ptr(), ln1.data_ptr(),
-                                self.h.data_ptr(), batch, H, r], [(prev, 0)])
-                else:
-                    g.add_task(T_ADD_RMSNORM, nrm,
-                               [self.mo.data_ptr(), self.x.data_ptr(),
-                                self.x.data_ptr(), ln1.data_ptr(),
-                                self.h.data_ptr(), batch, H, r], [(prev, 0)])
-            g.next_level()
-            qkv_op = gemm(self.h, at.w_qkv, self.qkv, qkv_dim, H, nrm)
-            pro = g.new_op()
-            for b in range(batch):
-                g.add_task(T_QKV_PROLOGUE, pro,
-                           [self.qkv.data_ptr(), self.q.data_ptr(),
-                            kv.k[li].data_ptr(), kv.v[li].data_ptr(),
-                            model.rotary.cos.data_ptr(),
-                            model.rotary.sin.data_ptr(),
-                            at.q_norm_w.data_ptr(), at.k_norm_w.data_ptr(),
-                            kv.offset.data_ptr(), b, qh, kvh, kv.max_len],
-                           [(qkv_op, 0)])
-            g.next_level()
-            fd = g.new_op()
-            for b in range(batch):
-                for kh in range(kvh):
-                    g.add_task(T_FLASH_DECODE, fd,
-                               [self.q.data_ptr(), kv.k[li].data_ptr(),
-                                kv.v[li].data_ptr(), self.attn.data_ptr(),
-                                kv.offset.data_ptr(), b, kh, qh, kvh,
-                                kv.max_len], [(pro, 0)])
-            g.next_level()
-            o_op = gemm(self.attn, at.w_o, self.attn_o, H, qh * d, fd)
-            ar2 = g.new_op()
-            for r in range(batch):
-                g.add_task(T_ADD_RMSNORM, ar2,
-                           [self.attn_o.data_ptr(), self.x.data_ptr(),
-                            self.x.data_ptr(), ln2.data_ptr(),
-                            self.h.data_ptr(), batch, H, r], [(o_op, 0)])
-            g.next_level()
-            if pipe:
-                mo_op = emit_mlp_pipe(ml, ar2)
+            if fuse:
+                res_ptr = self.mo.data_ptr() if pending is not None else 0
+                if pending is not None:
+                    # parallel (off critical path): x_alt = x_cur + mo
+                    xup = g.new_op()
+                    for r in range(batch):
+                        g.add_task(
+                            T_ADD_RMSNORM, xup,
+                            [self.mo.data_ptr(), x_cur.data_ptr(),
+                             x_alt.data_ptr(), ln1.data_ptr(),
+                             self.h.data_ptr(), batch, H, r],
+                            [(prev, 0)] + ([(xup_prev, 0)] if xup_prev
+                                           else []))
+                qkv_op = gemm_nr(x_cur, ln1, res_ptr, at.w_qkv, self.qkv,
+                                 qkv_dim, H, prev, xup_prev)
+                if pending is not None:
+                    xup_prev = xup
+                    x_cur, x_alt = x_alt, x_cur
             else:
-                gu_op = gemm(self.h, ml.w_gate_up, self.gu, 2 * i_s, H,
-                             ar2)
+                # h = rms(x + pending?)
+                nrm = g.new_op()
+                for r in range(batch):
+                    if pending is None:
+                        g.add_task(T_RMSNORM, nrm,
+                                   [self.x.data_ptr(), ln1.data_ptr(),
+                                    self.h.data_ptr(), batch, H, r],
+                                   [(prev, 0)])
+                    else:
+                        g.add_task(T_ADD_RMSNORM, nrm,
+                                   [self.mo.data_ptr(), self.x.data_ptr(),
+                                    self.x.data_ptr(), ln1.data_ptr(),
+                                    self.h.data_ptr(), batch, H, r],
+                                   [(prev, 0)])
+                g.next_level()
+                qkv_op = gemm(self.h, at.w_qkv, self.qkv, qkv_dim, H, nrm)
+            if fuse:
+                # prologue folded into each (b, kh) flash-decode task
+                fd = g.new_op()
+                for b in range(batch):
+                    for kh in range(kvh):
+                        g.add_task(
+                            T_PRO_FLASH_DECODE, fd,
+                            [self.qkv.data_ptr(), self.q.data_ptr(),
+                             kv.k[li].data_ptr(), kv.v[li].data_ptr(),
+                             model.rotary.cos.data_ptr(),
+                             model.rotary.sin.data_ptr(),
+                             at.q_norm_w.data_ptr(),
+                             at.k_norm_w.data_ptr(),
+                             kv.offset.data_ptr(), self.attn.data_ptr(),
+                             (kh << 32) | b, (kvh << 32) | qh,
+                             kv.max_len], [(qkv_op, 0)])
+                g.next_level()
+            else:
+                pro = g.new_op()
+                for b in range(batch):
+                    g.add_task(T_QKV_PROLOGUE, pro,
+                               [self.qkv.data_ptr(), self.q.data_ptr(),
+                                kv.k[li].data_ptr(), kv.v[li].data_ptr(),
+                                model.rotary.cos.data_ptr(),
+                                model.rotary.sin.data_ptr(),
+                                at.q_norm_w.data_ptr(),
+                                at.k_norm_w.data_ptr(),
+                                kv.offset.data_ptr(), b, qh, kvh,
+                                kv.max_len],
+                               [(qkv_op, 0)])
+                g.next_level()
+                fd = g.new_op()
+                for b in range(batch):
+                    for kh in range(kvh):
+                        g.add_task(T_FLASH_DECODE, fd,
+                                   [self.q.data_ptr(), kv.k[li].data_ptr(),
+                                    kv.v[li].data_ptr(),
+                                    self.attn.data_ptr(),
+                                    kv.offset.data_ptr(), b, kh, qh, kvh,
+                                    kv.max_len], [(pro, 0)])
+                g.next_level()
+            o_op = gemm(self.attn, at.w_o, self.attn_o, H, qh * d, fd)
+            if fuse:
+                xup2 = g.new_op()  # parallel: x_alt = x_cur + attn_o
+                for r in range(batch):
+                    g.add_task(
+                        T_ADD_RMSNORM, xup2,
+                        [self.attn_o.data_ptr(), x_cur.data_ptr(),
+                         x_alt.data_ptr(), ln2.data_ptr(),
+                         self.h.data_ptr(), batch, H, r],
+                        [(o_op, 0)] + ([(xup_prev, 0)] if xup_prev
+                                       else []))
+                gu_op = gemm_nr(x_cur, ln2, self.attn_o.data_ptr(),
+                                ml.w_gate_up, self.gu, 2 * i_s, H, o_op,
+                                xup_prev)
+                xup_prev = xup2
+                x_cur, x_alt = x_alt, x_cur
                 sw = g.new_op()
                 nchunks = max(tiles_m * i_s // 1024, 8)
                 for c in range(nchunks):
@@ -236,15 +310,39 @@ class MegaQwen3Decode:
                                 batch, i_s, c, nchunks], [(gu_op, 0)])
                 g.next_level()
                 mo_op = gemm(self.act, ml.w_down, self.mo, H, i_s, sw)
+            else:
+                ar2 = g.new_op()
+                for r in range(batch):
+                    g.add_task(T_ADD_RMSNORM, ar2,
+                               [self.attn_o.data_ptr(), self.x.data_ptr(),
+                                self.x.data_ptr(), ln2.data_ptr(),
+                                self.h.data_ptr(), batch, H, r],
+                               [(o_op, 0)])
+                g.next_level()
+                if pipe:
+                    mo_op = emit_mlp_pipe(ml, ar2)
+                else:
+                    gu_op = gemm(self.h, ml.w_gate_up, self.gu, 2 * i_s,
+                                 H, ar2)
+                    sw = g.new_op()
+                    nchunks = max(tiles_m * i_s // 1024, 8)
+                    for c in range(nchunks):
+                        g.add_task(T_SWIGLU, sw,
+                                   [self.gu.data_ptr(),
+                                    self.act.data_ptr(),
+                                    batch, i_s, c, nchunks], [(gu_op, 0)])
+                    g.next_level()
+                    mo_op = gemm(self.act, ml.w_down, self.mo, H, i_s, sw)
             pending, prev = mo_op, mo_op
 
         # final: xn = rms(x + mo); logits = xn @ lm_head^T; kv.offset += 1
         fin = g.new_op()
         for r in range(batch):
             g.add_task(T_ADD_RMSNORM, fin,
-                       [self.mo.data_ptr(), self.x.data_ptr(),
-                        self.x.data_ptr(), model.final_norm_w.data_ptr(),
-                        self.xn.data_ptr(), batch, H, r], [(prev, 0)])
+                       [self.mo.data_ptr(), x_cur.data_ptr(),
+                        x_alt.data_ptr(), model.final_norm_w.data_ptr(),
+                        self.xn.data_ptr(), batch, H, r],
+                       [(prev, 0)] + ([(xup_prev, 0)] if xup_prev else []))
         g.next_level()
         vocab_pad = cfg.vocab // 128 * 128
         lm = gemm(self.xn, model.lm_head, self.logits, vocab_pad, H, fin)
