@@ -41,15 +41,18 @@ def _worker(rank, world):
     w = torch.softmax(torch.randn(world * M_LOC, K, generator=g2), dim=-1)
 
     x_shard = x_full[rank * M_LOC:(rank + 1) * M_LOC].cuda().contiguous()
-    out = tp_moe_forward(x_shard, ids.cuda(), w.cuda(), w_gate_up, w_down,
-                         ag_ctx, coll_ctx)
-    torch.cuda.synchronize()
     ref = tp_moe_ref(x_full, ids, w,
                      torch.cat([gate, up], dim=1).to(torch.bfloat16),
                      down.to(torch.bfloat16), world, rank)
-    err = (out.float().cpu() - ref.float()).abs().max().item()
-    rel = err / ref.float().abs().max().item()
-    assert rel < 0.05, (rank, err, rel)
+    # both paths: gather-then-compute and the segment-progressive
+    # AG<->grouped-GEMM overlap (rank-staggered arrival order)
+    for overlap in (False, True):
+        out = tp_moe_forward(x_shard, ids.cuda(), w.cuda(), w_gate_up,
+                             w_down, ag_ctx, coll_ctx, overlap=overlap)
+        torch.cuda.synchronize()
+        err = (out.float().cpu() - ref.float()).abs().max().item()
+        rel = err / ref.float().abs().max().item()
+        assert rel < 0.05, (rank, overlap, err, rel)
     td.shutdown_heap()
 
 

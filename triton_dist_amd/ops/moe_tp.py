@@ -127,20 +127,126 @@ def tp_moe_from_full(x_full: torch.Tensor, topk_ids: torch.Tensor,
 
 def tp_moe_forward(x_shard: torch.Tensor, topk_ids: torch.Tensor,
                    topk_w: torch.Tensor, w_gate_up: torch.Tensor,
-                   w_down: torch.Tensor, ag_ctx, coll_ctx) -> torch.Tensor:
+                   w_down: torch.Tensor, ag_ctx, coll_ctx,
+                   overlap: Optional[bool] = None) -> torch.Tensor:
     """Full TP-MoE block: AG(x_shard) -> tp_moe_from_full. topk_ids /
     topk_w are for the FULL gathered batch [M, K] (router is replicated +
     deterministic).
 
     w_gate_up: [E, 2*inter_shard, H] (my shard of gate rows then up rows),
     w_down: [E, H, inter_shard]. Returns [m_local, H].
-    """
+
+    overlap (default on for hip, world > 1): segment-progressive
+    AG↔grouped-GEMM overlap — see _tp_moe_forward_overlapped."""
     from .allgather_gemm import allgather
 
-    world = get_heap().world
+    heap = get_heap()
+    world = heap.world
+    if overlap is None:
+        overlap = heap.backend == "hip" and world > 1
+    if overlap and heap.backend == "hip" and world > 1:
+        return _tp_moe_forward_overlapped(x_shard, topk_ids, topk_w,
+                                          w_gate_up, w_down, ag_ctx,
+                                          coll_ctx)
     x_full = allgather(x_shard, ag_ctx) if world > 1 else x_shard
     return tp_moe_from_full(x_full, topk_ids, topk_w, w_gate_up, w_down,
                             coll_ctx)
+
+
+def _tp_moe_forward_overlapped(x_shard, topk_ids, topk_w, w_gate_up,
+                               w_down, ag_ctx, coll_ctx):
+    """Rank-staggered AG + grouped-GEMM overlap — the reference's AG-MoE
+    threadblock-swizzle capability (threadblock_swizzle_ag_moe.cu:248
+    rank-staggered tile visit order after AG — behavior only), as an
+    MI355X segment-progressive design: the shard push is the standard
+    SDMA producer over the comm-stream pool; the expert compute runs in
+    arrival-preference order (own shard first, then (rank+1)%world, ...)
+    with each segment gated by one single-workgroup flag-wait kernel on
+    the compute stream. Grouped GEMMs of arrived segments overlap the
+    xGMI transfer of later ones; the router is replicated, so every
+    segment's expert-sort metadata is built before any data lands. The
+    only spin is the 1-WG wait (peer flags come from SDMA, no CU
+    dependency), so this is safe even when ranks share one GPU."""
+    from .collectives import reduce_scatter
+    from .fused import swiglu_op
+    from .. import _C
+
+    heap = get_heap()
+    world, rank = heap.world, heap.rank
+    m, H = x_shard.shape
+    E = w_gate_up.shape[0]
+    inter = w_down.shape[2]
+    ctx = ag_ctx
+    chunks = ctx.chunks_per_rank
+    assert H == ctx.k and m <= ctx.max_m_per_rank and m % chunks == 0
+    dev = x_shard.device
+    compute = torch.cuda.current_stream()
+    s = compute.cuda_stream
+    # push phase: the allgather SDMA producer, WITHOUT the global wait
+    _C.reset_flags(ctx.flags.ptr(), world * chunks, 0, s)
+    heap.barrier_all_on_stream(compute)
+    seg_bytes = ctx.max_m_per_rank * H * 2
+    _C.memcpy_async(ctx.ws.ptr() + rank * seg_bytes, x_shard.data_ptr(),
+                    m * H * 2, s)
+    _C.reset_flags(ctx.flags.ptr() + rank * chunks * 4, chunks, 1, s)
+    ctx.ready_ev.record(compute)
+    ns = len(ctx.comm_streams)
+    chunk_bytes = (m // chunks) * H * 2
+    for st in range(ns):
+        ctx.comm_streams[st].wait_event(ctx.ready_ev)
+    for i in range(world - 1):
+        peer = (rank + 1 + i) % world
+        stream = ctx.comm_streams[i % ns]
+        dst_seg = ctx.ws.ptr(peer) + rank * seg_bytes
+        dst_flag = ctx.flags.ptr(peer) + rank * chunks * 4
+        for c in range(chunks):
+            _C.memcpy_async(dst_seg + c * chunk_bytes,
+                            x_shard.data_ptr() + c * chunk_bytes,
+                            chunk_bytes, stream.cuda_stream)
+            _C.memcpy_async(dst_flag + c * 4, heap.one_src.ptr(), 4,
+                            stream.cuda_stream)
+    # per-segment sort metadata from the replicated router (no data dep)
+    K = topk_ids.shape[1]
+    metas = []
+    for src in range(world):
+        flat = topk_ids[src * m:(src + 1) * m].reshape(-1).to(torch.int64)
+        order = torch.argsort(flat, stable=True)
+        counts = torch.bincount(flat, minlength=E)
+        base = torch.cumsum(counts, 0) - counts
+        cnt = counts.cpu().tolist()
+        items = [e * 65536 + t for e, c in enumerate(cnt)
+                 for t in range((c + 31) // 32)]
+        metas.append(dict(
+            expert_base=base.to(torch.int32).to(dev).contiguous(),
+            expert_rows=counts.to(torch.int32).to(dev).contiguous(),
+            work_items=torch.tensor(items or [0], dtype=torch.int32,
+                                    device=dev),
+            work_count=torch.tensor([len(items)], dtype=torch.int32,
+                                    device=dev),
+            order=order, tok=order // K))
+    # segment-progressive expert compute
+    ws_view = ctx.ws.local()  # [world, max_m, H]
+    y_full = torch.zeros(world * m, H, dtype=torch.float32, device=dev)
+    mk = m * K
+    for i in range(world):
+        src = (rank + i) % world
+        if src != rank:  # own segment is compute-stream-ordered already
+            _C.wait_eq(ctx.flags.ptr() + src * chunks * 4, chunks, 1, s)
+        meta = metas[src]
+        x_sorted = torch.zeros(mk + 128, H, dtype=x_shard.dtype,
+                               device=dev)
+        x_sorted[:mk] = ws_view[src, :m].index_select(0, meta["tok"])
+        hidden = grouped_gemm(x_sorted, w_gate_up, meta)
+        act = swiglu_op(hidden, inter)
+        part = grouped_gemm(act, w_down, meta)
+        w_sorted = topk_w[src * m:(src + 1) * m].reshape(-1)[
+            meta["order"]].to(torch.float32)
+        y_full.index_add_(0, src * m + meta["tok"],
+                          part[:mk].float() * w_sorted[:, None])
+    for st in range(min(ns, max(world - 1, 1))):
+        ctx.join_evs[st].record(ctx.comm_streams[st])
+        compute.wait_event(ctx.join_evs[st])
+    return reduce_scatter(y_full.to(x_shard.dtype), coll_ctx)
 
 
 def tp_moe_ref(x_full: torch.Tensor, topk_ids: torch.Tensor,
