@@ -90,3 +90,31 @@ def test_tp_moe_model_wiring_2rank():
 
 def test_tp_moe_4rank():
     run_distributed(_worker, world_size=4)
+
+
+def test_segment_sort_meta_matches_global_sort():
+    """Per-segment metadata (built from the replicated router for the
+    AG-overlap path) must agree with moe_sort_tokens on that segment's
+    ids: same base/rows/items, and the row->token mapping reindexes the
+    same gathered rows."""
+    import torch
+
+    from triton_dist_amd.ops.moe_tp import moe_sort_tokens, segment_sort_meta
+
+    g = torch.Generator().manual_seed(11)
+    E, m, K, H = 8, 24, 3, 16
+    ids = torch.randint(0, E, (m, K), generator=g, dtype=torch.int32)
+    x = torch.randn(m, H, generator=g).to(torch.bfloat16)
+
+    meta = segment_sort_meta(ids, E, "cpu")
+    x_sorted_ref, tok_ref, meta_ref = moe_sort_tokens(x, ids, E)
+    assert torch.equal(meta["expert_base"], meta_ref["expert_base"])
+    assert torch.equal(meta["expert_rows"], meta_ref["expert_rows"])
+    assert torch.equal(meta["work_items"], meta_ref["work_items"])
+    assert torch.equal(meta["work_count"], meta_ref["work_count"])
+    assert torch.equal(meta["tok"], tok_ref)
+    # reconstructing x_sorted from the meta matches moe_sort_tokens
+    mk = m * K
+    x_sorted = torch.zeros(mk + 128, H, dtype=x.dtype)
+    x_sorted[:mk] = x.index_select(0, meta["tok"])
+    assert torch.equal(x_sorted, x_sorted_ref)

@@ -153,6 +153,30 @@ def tp_moe_forward(x_shard: torch.Tensor, topk_ids: torch.Tensor,
                             coll_ctx)
 
 
+def segment_sort_meta(topk_ids_seg: torch.Tensor, n_experts: int,
+                      dev, bm: int = 32) -> dict:
+    """Expert-sort descriptors for ONE gathered segment, built purely
+    from the replicated router output (no activation data needed):
+    expert_base/rows, pq work items, the sorted row order, and each
+    sorted row's source token within the segment."""
+    k = topk_ids_seg.shape[1]
+    flat = topk_ids_seg.reshape(-1).to(torch.int64)
+    order = torch.argsort(flat, stable=True)
+    counts = torch.bincount(flat, minlength=n_experts)
+    base = torch.cumsum(counts, 0) - counts
+    cnt = counts.cpu().tolist()
+    items = [e * 65536 + t for e, c in enumerate(cnt)
+             for t in range((c + bm - 1) // bm)]
+    return dict(
+        expert_base=base.to(torch.int32).to(dev).contiguous(),
+        expert_rows=counts.to(torch.int32).to(dev).contiguous(),
+        work_items=torch.tensor(items or [0], dtype=torch.int32,
+                                device=dev),
+        work_count=torch.tensor([len(items)], dtype=torch.int32,
+                                device=dev),
+        order=order, tok=order // k)
+
+
 def _tp_moe_forward_overlapped(x_shard, topk_ids, topk_w, w_gate_up,
                                w_down, ag_ctx, coll_ctx):
     """Rank-staggered AG + grouped-GEMM overlap — the reference's AG-MoE
@@ -207,23 +231,8 @@ def _tp_moe_forward_overlapped(x_shard, topk_ids, topk_w, w_gate_up,
                             stream.cuda_stream)
     # per-segment sort metadata from the replicated router (no data dep)
     K = topk_ids.shape[1]
-    metas = []
-    for src in range(world):
-        flat = topk_ids[src * m:(src + 1) * m].reshape(-1).to(torch.int64)
-        order = torch.argsort(flat, stable=True)
-        counts = torch.bincount(flat, minlength=E)
-        base = torch.cumsum(counts, 0) - counts
-        cnt = counts.cpu().tolist()
-        items = [e * 65536 + t for e, c in enumerate(cnt)
-                 for t in range((c + 31) // 32)]
-        metas.append(dict(
-            expert_base=base.to(torch.int32).to(dev).contiguous(),
-            expert_rows=counts.to(torch.int32).to(dev).contiguous(),
-            work_items=torch.tensor(items or [0], dtype=torch.int32,
-                                    device=dev),
-            work_count=torch.tensor([len(items)], dtype=torch.int32,
-                                    device=dev),
-            order=order, tok=order // K))
+    metas = [segment_sort_meta(topk_ids[src * m:(src + 1) * m], E, dev)
+             for src in range(world)]
     # segment-progressive expert compute
     ws_view = ctx.ws.local()  # [world, max_m, H]
     y_full = torch.zeros(world * m, H, dtype=torch.float32, device=dev)
