@@ -106,15 +106,24 @@ TD_DEV WaveCtx wave_ctx() {
 // (drain 8/8/4/0 on the last tile).
 TD_DEV void kloop(const bf16 *ga, const bf16 *gb, int lda, int ldb,
                   int ktiles, bf16 *lds_a, bf16 *lds_b, const WaveCtx &w,
-                  f32x4 acc[8][4]) {
-  // prologue: stage slices 0..2 of tile 0 (slice 3 staged in phase 0)
+                  f32x4 acc[8][4], int kt0 = 0) {
+  // kt0: staggered K start (StaggerU, adopted from the hipBLASLt SK3
+  // ISA study — profiles/hipblaslt_sk3_isa_study.md): the walk covers
+  // tiles kt0, kt0+1, ... mod ktiles, so concurrent WGs hit different
+  // HBM channels at any instant instead of all camping on k=0 together.
+  // fp32 accumulation order changes; sum is reassociation-equivalent.
+  // prologue: stage slices 0..2 of tile kt0 (slice 3 staged in phase 0)
 #pragma unroll
   for (int p = 0; p < SLICES - 1; ++p)
-    stage_slice(ga, gb, lda, ldb, p * SLICE_K, lds_a, lds_b, p);
+    stage_slice(ga + (size_t)kt0 * BK, gb + (size_t)kt0 * BK, lda, ldb,
+                p * SLICE_K, lds_a, lds_b, p);
 
   for (int t = 0; t < ktiles; ++t) {
+    int tt = kt0 + t;
+    if (tt >= ktiles) tt -= ktiles;
     const bool has_next = (t + 1) < ktiles;
-    const int knext = (t + 1) * BK;
+    int tn = tt + 1 == ktiles ? 0 : tt + 1;
+    const size_t knext = (size_t)tn * BK;
 #pragma unroll
     for (int p = 0; p < SLICES; ++p) {
       // retire the loads for slice p of tile t
@@ -127,8 +136,8 @@ TD_DEV void kloop(const bf16 *ga, const bf16 *gb, int lda, int ldb,
       __builtin_amdgcn_s_barrier();  // slice p landed; region p-1 free
       __builtin_amdgcn_sched_barrier(0);
       if (p == 0) {
-        stage_slice(ga + t * BK, gb + t * BK, lda, ldb, 3 * SLICE_K, lds_a,
-                    lds_b, 3);
+        stage_slice(ga + (size_t)tt * BK, gb + (size_t)tt * BK, lda, ldb,
+                    3 * SLICE_K, lds_a, lds_b, 3);
       } else if (has_next) {
         stage_slice(ga + knext, gb + knext, lda, ldb, (p - 1) * SLICE_K,
                     lds_a, lds_b, p - 1);
@@ -208,7 +217,8 @@ TD_DEV void tile_coords(int wgid, int tiles_m, int tiles_n, int &pid_m,
   pid_n = (wgid % (GM * tiles_n)) / gsz;
 }
 
-__global__ __launch_bounds__(NTH, 2) void k_gemm256_bf16(GemmArgs args) {
+__global__ __launch_bounds__(NTH, 2) void k_gemm256_bf16(GemmArgs args,
+                                                          int stagger) {
   __shared__ bf16 lds_a[SLICES * SLICE_ELEMS];
   __shared__ bf16 lds_b[SLICES * SLICE_ELEMS];
   const int tiles_n = args.n / BN;
@@ -220,7 +230,9 @@ __global__ __launch_bounds__(NTH, 2) void k_gemm256_bf16(GemmArgs args) {
   f32x4 acc[8][4] = {};
   const bf16 *ga = (const bf16 *)args.a + (size_t)pid_m * BM * args.lda;
   const bf16 *gb = (const bf16 *)args.b + (size_t)pid_n * BN * args.ldb;
-  kloop(ga, gb, args.lda, args.ldb, args.k / BK, lds_a, lds_b, w, acc);
+  const int ktiles = args.k / BK;
+  const int kt0 = stagger ? wgid % ktiles : 0;
+  kloop(ga, gb, args.lda, args.ldb, ktiles, lds_a, lds_b, w, acc, kt0);
   bf16 *dst = (bf16 *)args.c + (size_t)pid_m * BM * args.ldc + pid_n * BN;
   epilogue(acc, w, lds_a, dst, args.ldc);
 }
@@ -491,7 +503,8 @@ void launch_f32_to_bf16(const void *ws, void *c, const void *bias, int rows,
 // restores occupancy while keeping the proven per-CU schedule.
 __global__ __launch_bounds__(NTH, 2) void k_gemm256_sk_bf16(GemmArgs g,
                                                             float *ws,
-                                                            int sk) {
+                                                            int sk,
+                                                            int stagger) {
   __shared__ bf16 lds_a[SLICES * SLICE_ELEMS];
   __shared__ bf16 lds_b[SLICES * SLICE_ELEMS];
   const int tiles_n = g.n / BN;
@@ -508,7 +521,8 @@ __global__ __launch_bounds__(NTH, 2) void k_gemm256_sk_bf16(GemmArgs g,
   f32x4 acc[8][4] = {};
   const bf16 *ga = (const bf16 *)g.a + (size_t)pid_m * BM * g.lda + k0;
   const bf16 *gb = (const bf16 *)g.b + (size_t)pid_n * BN * g.ldb + k0;
-  kloop(ga, gb, g.lda, g.ldb, ktiles_per, lds_a, lds_b, w, acc);
+  const int kt0 = stagger ? wgid % ktiles_per : 0;
+  kloop(ga, gb, g.lda, g.ldb, ktiles_per, lds_a, lds_b, w, acc, kt0);
   float *wsb = ws + (size_t)pid_m * BM * g.ldc + pid_n * BN;
 #pragma unroll
   for (int i = 0; i < 8; ++i)
@@ -555,7 +569,8 @@ __global__ void k_f32_to_bf16(const float *__restrict__ ws,
 // pass.
 __global__ __launch_bounds__(NTH, 2) void k_gemm256_sk2_bf16(GemmArgs g,
                                                              float *ws,
-                                                             int sk) {
+                                                             int sk,
+                                                             int stagger) {
   __shared__ bf16 lds_a[SLICES * SLICE_ELEMS];
   __shared__ bf16 lds_b[SLICES * SLICE_ELEMS];
   const int tiles_n = g.n / BN;
@@ -572,7 +587,8 @@ __global__ __launch_bounds__(NTH, 2) void k_gemm256_sk2_bf16(GemmArgs g,
   f32x4 acc[8][4] = {};
   const bf16 *ga = (const bf16 *)g.a + (size_t)pid_m * BM * g.lda + k0;
   const bf16 *gb = (const bf16 *)g.b + (size_t)pid_n * BN * g.ldb + k0;
-  kloop(ga, gb, g.lda, g.ldb, ktiles_per, lds_a, lds_b, w, acc);
+  const int kt0 = stagger ? wgid % ktiles_per : 0;
+  kloop(ga, gb, g.lda, g.ldb, ktiles_per, lds_a, lds_b, w, acc, kt0);
   float *wsb = ws + ((size_t)sid * g.m + (size_t)pid_m * BM) * g.n +
                (size_t)pid_n * BN;
 #pragma unroll
@@ -621,8 +637,12 @@ void launch_gemm256_sk2_bf16(const GemmArgs &g, float *ws, int sk,
   if (g.k % (BK * sk))
     throw std::runtime_error("gemm256_sk2: k % (128*sk) != 0");
   int grid = (g.m / BM) * (g.n / BN) * sk;
+  static const int stagger = [] {
+    const char *e = getenv("TD_GEMM_STAGGER");
+    return (e && e[0] == '0') ? 0 : 1;
+  }();
   hipLaunchKernelGGL(k_gemm256_sk2_bf16, dim3(grid), dim3(NTH), 0, stream,
-                     g, ws, sk);
+                     g, ws, sk, stagger);
   int cgrid = g.m < 2048 ? g.m : 2048;
   hipLaunchKernelGGL(k_sk2_reduce, dim3(cgrid), dim3(256), 0, stream, ws,
                      (bf16 *)g.c, (const bf16 *)g.bias, g.m, g.n, sk);
@@ -635,8 +655,12 @@ void launch_gemm256_sk_bf16(const GemmArgs &g, float *ws, int sk,
   size_t elems = (size_t)g.m * g.n;
   TD_CHECK_HIP(hipMemsetAsync(ws, 0, elems * sizeof(float), stream));
   int grid = (g.m / BM) * (g.n / BN) * sk;
+  static const int stagger = [] {
+    const char *e = getenv("TD_GEMM_STAGGER");
+    return (e && e[0] == '0') ? 0 : 1;
+  }();
   hipLaunchKernelGGL(k_gemm256_sk_bf16, dim3(grid), dim3(NTH), 0, stream, g,
-                     ws, sk);
+                     ws, sk, stagger);
   int cgrid = g.m < 2048 ? g.m : 2048;
   hipLaunchKernelGGL(k_f32_to_bf16, dim3(cgrid), dim3(256), 0, stream, ws,
                      (bf16 *)g.c, (const bf16 *)g.bias, g.m, g.n);
@@ -651,7 +675,13 @@ bool gemm256_ok(int m, int n, int k) {
 
 void launch_gemm256_bf16(const GemmArgs &args, hipStream_t stream) {
   int grid = (args.m / BM) * (args.n / BN);
-  hipLaunchKernelGGL(k_gemm256_bf16, dim3(grid), dim3(NTH), 0, stream, args);
+  // TD_GEMM_STAGGER=0 disables the StaggerU K walk (A/B)
+  static const int stagger = [] {
+    const char *e = getenv("TD_GEMM_STAGGER");
+    return (e && e[0] == '0') ? 0 : 1;
+  }();
+  hipLaunchKernelGGL(k_gemm256_bf16, dim3(grid), dim3(NTH), 0, stream, args,
+                     stagger);
 }
 
 void launch_ag_gemm256_consumer_bf16(const AgGemmArgs &args,
