@@ -637,9 +637,12 @@ void launch_gemm256_sk2_bf16(const GemmArgs &g, float *ws, int sk,
   if (g.k % (BK * sk))
     throw std::runtime_error("gemm256_sk2: k % (128*sk) != 0");
   int grid = (g.m / BM) * (g.n / BN) * sk;
+  // StaggerU measured NEGATIVE here (profiles/README.md r02 addendum):
+  // per-WG K offsets destroy the GROUP_M L2 panel sharing. Default off;
+  // TD_GEMM_STAGGER=1 reproduces the experiment.
   static const int stagger = [] {
     const char *e = getenv("TD_GEMM_STAGGER");
-    return (e && e[0] == '0') ? 0 : 1;
+    return (e && e[0] == '1') ? 1 : 0;
   }();
   hipLaunchKernelGGL(k_gemm256_sk2_bf16, dim3(grid), dim3(NTH), 0, stream,
                      g, ws, sk, stagger);
@@ -655,9 +658,12 @@ void launch_gemm256_sk_bf16(const GemmArgs &g, float *ws, int sk,
   size_t elems = (size_t)g.m * g.n;
   TD_CHECK_HIP(hipMemsetAsync(ws, 0, elems * sizeof(float), stream));
   int grid = (g.m / BM) * (g.n / BN) * sk;
+  // StaggerU measured NEGATIVE here (profiles/README.md r02 addendum):
+  // per-WG K offsets destroy the GROUP_M L2 panel sharing. Default off;
+  // TD_GEMM_STAGGER=1 reproduces the experiment.
   static const int stagger = [] {
     const char *e = getenv("TD_GEMM_STAGGER");
-    return (e && e[0] == '0') ? 0 : 1;
+    return (e && e[0] == '1') ? 1 : 0;
   }();
   hipLaunchKernelGGL(k_gemm256_sk_bf16, dim3(grid), dim3(NTH), 0, stream, g,
                      ws, sk, stagger);
@@ -676,9 +682,12 @@ bool gemm256_ok(int m, int n, int k) {
 void launch_gemm256_bf16(const GemmArgs &args, hipStream_t stream) {
   int grid = (args.m / BM) * (args.n / BN);
   // TD_GEMM_STAGGER=0 disables the StaggerU K walk (A/B)
+  // StaggerU measured NEGATIVE here (profiles/README.md r02 addendum):
+  // per-WG K offsets destroy the GROUP_M L2 panel sharing. Default off;
+  // TD_GEMM_STAGGER=1 reproduces the experiment.
   static const int stagger = [] {
     const char *e = getenv("TD_GEMM_STAGGER");
-    return (e && e[0] == '0') ? 0 : 1;
+    return (e && e[0] == '1') ? 1 : 0;
   }();
   hipLaunchKernelGGL(k_gemm256_bf16, dim3(grid), dim3(NTH), 0, stream, args,
                      stagger);
