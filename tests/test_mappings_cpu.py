@@ -618,3 +618,33 @@ def test_moe_pq_pipeline_ledger(nbuf, ksteps):
             f"stage {t} not landed: {(idx+1)*LOADS} > {landed_upto}"
         last_read[t] = t
     assert len(last_read) == ksteps
+
+
+@pytest.mark.parametrize("ktiles,kt0", [(1, 0), (4, 0), (4, 3), (32, 7),
+                                        (40, 39)])
+def test_kloop_stagger_walk(ktiles, kt0):
+    """gemm256 kloop StaggerU walk: tiles kt0, kt0+1, ... mod ktiles —
+    every K tile consumed exactly once, every stage lands before its
+    consume, and the staged (region, slice) sequence matches the ring's
+    p-1 staging discipline."""
+    consumed = []
+    staged = {}  # region -> (tile, slice)
+    # prologue: slices 0..2 of tile kt0 into regions 0..2
+    for p in range(3):
+        staged[p] = (kt0, p)
+    for t in range(ktiles):
+        tt = kt0 + t
+        if tt >= ktiles:
+            tt -= ktiles
+        tn = 0 if tt + 1 == ktiles else tt + 1
+        has_next = (t + 1) < ktiles
+        for p in range(4):
+            if p == 0:
+                staged[3] = (tt, 3)  # current tile's last slice
+            elif has_next:
+                staged[p - 1] = (tn, p - 1)
+            # consume (tile tt, slice p) from region p
+            assert staged.get(p) == (tt, p), (t, p, staged.get(p))
+        consumed.append(tt)
+    assert sorted(consumed) == list(range(ktiles))
+    assert consumed[0] == kt0 % ktiles
