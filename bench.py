@@ -50,13 +50,16 @@ def main():
         f"launch with torch.distributed.run for --gpus {args.gpus}"
     on_gpu = torch.cuda.is_available()
     device = "cuda" if on_gpu else "cpu"
-    if (on_gpu and world > torch.cuda.device_count()
-            and args.mode == "ag_rs"):
-        # Oversubscribed DEV topology (ranks sharing one GPU): the fused
-        # persistent spin-wait consumers of both ranks saturate the single
-        # GPU's CUs and the peer's flag-producing kernels starve. One rank
-        # per GPU (the benchmark topology) has no such cycle: every spin's
-        # producer is SDMA (CU-free) or stream-ordered before the waiter.
+    from triton_dist_amd.utils import gpu_oversubscribed
+    if on_gpu and args.mode == "ag_rs" and gpu_oversubscribed(world):
+        # Oversubscribed DEV topology (ranks REALLY sharing one physical
+        # GPU — verified by a PCI-id exchange, so a launcher that pins
+        # one visible device per rank at world 8 does NOT trip this):
+        # the fused persistent spin-wait consumers of both ranks saturate
+        # the single GPU's CUs and the peer's flag-producing kernels
+        # starve. One rank per GPU (the benchmark topology) has no such
+        # cycle: every spin's producer is SDMA (CU-free) or
+        # stream-ordered before the waiter.
         print("[bench] world > device_count: oversubscribed sharing "
               "topology -> falling back to --mode allreduce --no-graph",
               file=sys.stderr)
