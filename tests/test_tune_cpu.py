@@ -85,3 +85,35 @@ def test_maybe_enable_tunableop_cpu_noop():
         assert maybe_enable_tunableop() is False
     finally:
         del os.environ["PYTORCH_TUNABLEOP_TUNING"]
+
+
+def test_maybe_enable_tunableop_loads_shipped_file(monkeypatch):
+    """With a (faked) GPU and tunable API present, the loader activates
+    the shipped result file read-only: enable(True), tuning off, and
+    read_file pointed at autotune_cache/tunableop_gfx950.csv."""
+    from pathlib import Path
+
+    import triton_dist_amd.tune as tn
+
+    calls = {}
+
+    class FakeTunable:
+        @staticmethod
+        def enable(v):
+            calls["enable"] = v
+
+        @staticmethod
+        def tuning_enable(v):
+            calls["tuning"] = v
+
+        @staticmethod
+        def read_file(p):
+            calls["file"] = p
+
+    monkeypatch.setattr(tn.torch.cuda, "tunable", FakeTunable,
+                        raising=False)
+    monkeypatch.setattr(tn.torch.cuda, "is_available", lambda: True)
+    assert tn.maybe_enable_tunableop() is True
+    assert calls["enable"] is True and calls["tuning"] is False
+    assert calls["file"].endswith("tunableop_gfx950.csv")
+    assert Path(calls["file"]).exists()  # the shipped file is real
