@@ -648,3 +648,26 @@ def test_kloop_stagger_walk(ktiles, kt0):
         consumed.append(tt)
     assert sorted(consumed) == list(range(ktiles))
     assert consumed[0] == kt0 % ktiles
+
+
+@pytest.mark.parametrize("m_total,m_per,chunks", [(4096, 2048, 4),
+                                                  (1024, 512, 4),
+                                                  (2048, 256, 2)])
+def test_ag_consumer_chunk_wait_ranges(m_total, m_per, chunks):
+    """k_ag_gemm256_consumer flag indexing: every 256-row tile waits on
+    exactly the global chunk flags covering its rows, and all indices
+    stay inside the world*chunks flag array."""
+    world = m_total // m_per
+    rows_per_chunk = m_per // chunks
+    covered = set()
+    for pid_m in range(m_total // 256):
+        c_lo = (pid_m * 256) // rows_per_chunk
+        c_hi = (pid_m * 256 + 255) // rows_per_chunk
+        assert 0 <= c_lo <= c_hi < world * chunks, (pid_m, c_lo, c_hi)
+        for c in range(c_lo, c_hi + 1):
+            covered.add(c)
+            # chunk c's rows overlap the tile's rows
+            r0, r1 = c * rows_per_chunk, (c + 1) * rows_per_chunk
+            t0, t1 = pid_m * 256, pid_m * 256 + 256
+            assert r0 < t1 and t0 < r1
+    assert covered == set(range(world * chunks))
