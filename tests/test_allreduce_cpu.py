@@ -53,10 +53,6 @@ def test_model_gemm_ar_cpu_2rank():
     run_distributed(_body_model_gemm_ar, world_size=2)
 
 
-def test_allreduce_cpu_4rank():
-    run_distributed(_body_ar, world_size=4)
-
-
 def _body_gemm_ar_tiled(rank, world):
     from triton_dist_amd.ops import create_allreduce_context, gemm_allreduce
     from triton_dist_amd.ops.allreduce import _n_owned
