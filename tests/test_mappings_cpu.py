@@ -671,3 +671,23 @@ def test_ag_consumer_chunk_wait_ranges(m_total, m_per, chunks):
             t0, t1 = pid_m * 256, pid_m * 256 + 256
             assert r0 < t1 and t0 < r1
     assert covered == set(range(world * chunks))
+
+
+def test_split_pick_divisibility_invariants():
+    """Every split factor either tuner returns must divide the K-step
+    grid it will be launched with (the launchers throw otherwise), over
+    a broad shape sweep."""
+    from triton_dist_amd.ops.gemm import choose_splits, sk256_pick
+
+    for m in (128, 256, 512, 1024, 4096):
+        for n in (256, 1280, 5120, 14336, 51200):
+            for k in (512, 2048, 5120, 8192, 25600, 27648):
+                s = choose_splits(m, n, k)
+                assert s >= 1
+                if s > 1:
+                    assert (k // 32) % s == 0 and k // s >= 512, (m, n, k, s)
+                sk = sk256_pick(m, n, k)
+                if sk:
+                    assert m % 256 == 0 and n % 256 == 0
+                    assert k % (128 * sk) == 0, (m, n, k, sk)
+                    assert (k // 128) // sk >= 4
