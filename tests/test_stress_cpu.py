@@ -83,3 +83,53 @@ def _worker_coll_stress(rank, world):
 
 def test_collectives_stress_4rank():
     run_distributed(_worker_coll_stress, world_size=4)
+
+
+def test_ep_ll_credit_window_simulation():
+    """EP low-latency protocol lock (docs/MEMORY_ORDERING.md §11): with
+    2 parity buffers and the credit rule `wait credit[r] >= N-2 before
+    writing call N`, no rank ever overwrites a peer's parity slot while
+    the peer is still consuming the previous same-parity call — over
+    randomized asynchronous interleavings of rank progress."""
+    import random
+
+    for seed in range(40):
+        rng = random.Random(seed)
+        world, calls = rng.choice([2, 4, 8]), 12
+        # per-rank program counter: (call, phase); phases per call:
+        # 0 wait-credits, 1 write-to-peers, 2 read/compute, 3 signal
+        pc = [[1, 0] for _ in range(world)]
+        credit = [[0] * world for _ in range(world)]  # credit[dst][src]
+        read_done = [[True] * (calls + 3) for _ in range(world)]
+        for r in range(world):
+            for n in range(1, calls + 1):
+                read_done[r][n] = False
+        finished = 0
+        steps = 0
+        while finished < world and steps < 100000:
+            steps += 1
+            r = rng.randrange(world)
+            n, ph = pc[r]
+            if n > calls:
+                continue
+            if ph == 0:
+                # can pass only when every peer granted credit >= n-2
+                if all(credit[r][s] >= n - 2 for s in range(world)):
+                    pc[r][1] = 1
+            elif ph == 1:
+                # writing call n into every peer's parity slot n%2:
+                # the peer must have finished reading call n-2
+                for dst in range(world):
+                    if n - 2 >= 1:
+                        assert read_done[dst][n - 2], (seed, r, n, dst)
+                pc[r][1] = 2
+            elif ph == 2:
+                read_done[r][n] = True
+                pc[r][1] = 3
+            else:
+                for dst in range(world):
+                    credit[dst][r] = n
+                pc[r] = [n + 1, 0]
+                if n == calls:
+                    finished += 1
+        assert finished == world, f"deadlock at seed {seed}"
