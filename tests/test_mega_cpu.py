@@ -179,3 +179,52 @@ def test_fused_graph_structure():
     assert tp.count(T_PRO_FLASH_DECODE) == 0
     assert tp.count(T_GEMM_TILE_PART_NR) == 0
     assert tp.count(T_FLASH_DECODE) == 8 * cfg.n_kv_heads * cfg.n_layers
+
+
+import pytest
+
+
+@pytest.mark.parametrize("env", [{}, {"TD_MK_FUSE": "1"},
+                                 {"TD_MK_PIPE": "1"}])
+def test_graph_levels_monotone_with_deps(env, monkeypatch):
+    """Scheduler deadlock-freedom precondition: queues execute in level
+    order, so every task's dependencies must be PRODUCED at strictly
+    lower levels than the task's own level — for the default, hop-fused
+    and pipelined emission modes."""
+    import os
+
+    import torch
+
+    from triton_dist_amd.mega import builder as B
+    from triton_dist_amd.mega.qwen3 import MegaQwen3Decode
+    from triton_dist_amd.models import DenseLLM, KVCache, get_config
+
+    captured = {}
+    orig = B.MegaGraph.finalize
+
+    def capture(self, n_wg, device="cuda"):
+        captured["tasks"] = list(self.tasks)
+        captured["n_ops"] = len(self.ops)
+        return orig(self, n_wg, device)
+
+    monkeypatch.setattr(B.MegaGraph, "finalize", capture)
+    for k, v in env.items():
+        monkeypatch.setenv(k, v)
+    batch = 1 if env.get("TD_MK_PIPE") else 8
+    cfg = get_config("tiny-gpu", max_length=64)
+    model = DenseLLM(cfg, device="cpu")
+    model.init_weights(seed=2)
+    kv = KVCache(cfg.n_layers, batch, 64, cfg.n_kv_heads, cfg.head_dim,
+                 device="cpu")
+    MegaQwen3Decode(model, kv, batch=batch)
+    tasks = captured["tasks"]
+    assert tasks
+    # per-op max level of its producing tasks
+    op_level = {}
+    for (tt, slot, d0, d0n, d1, d1n, args, lvl) in tasks:
+        op_level[slot] = max(op_level.get(slot, -1), lvl)
+    for (tt, slot, d0, d0n, d1, d1n, args, lvl) in tasks:
+        for d in (d0, d1):
+            if d >= 0:
+                assert op_level[d] < lvl, (env, tt, slot, d, op_level[d],
+                                           lvl)
